@@ -1,0 +1,357 @@
+"""
+citus_amd — MI355X-native implementation of Citus's columnar scan +
+partial-aggregate hot path (chunk-group read -> LZ4 decode -> qual filter ->
+partial aggregate -> combine), behind the reference's scan/combine surfaces
+restated as a C ABI (include/cstripe.h).
+
+This package is ctypes plumbing over libcstripe.so (hand-written HIP/CDNA4
+kernels + host C++). The GPU path requires a visible MI355X and fails loudly
+without one; the CPU reference restatement lives in oracle/ and is test
+infrastructure only.
+"""
+import ctypes as C
+import os
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_HERE, "libcstripe.so")
+
+if not os.path.exists(_LIB_PATH):
+    raise ImportError(
+        "libcstripe.so not built — run `make -C citus_amd/csrc` "
+        "(or __graft_entry__.build())")
+
+_lib = C.CDLL(_LIB_PATH)
+
+# ---- enums (include/cstripe.h) ----
+I8, I16, I32, I64, F32, F64 = 1, 2, 3, 4, 5, 6
+COMP_NONE, COMP_PGLZ, COMP_LZ4, COMP_ZSTD = 0, 1, 2, 3
+PRED_LT, PRED_LE, PRED_GT, PRED_GE, PRED_EQ, PRED_NE = range(6)
+(AGG_COUNT_STAR, AGG_COUNT_COL, AGG_SUM_I64, AGG_SUM_F64,
+ AGG_MIN_I64, AGG_MAX_I64, AGG_MIN_F64, AGG_MAX_F64,
+ AGG_SUM_PROD_I64, AGG_SUM_DISC_I64, AGG_SUM_DISC_TAX_I64) = range(11)
+
+OK, ERR, ERR_IO, ERR_FORMAT, ERR_NOGPU, ERR_ARG, END = 0, -1, -2, -3, -4, -5, 1
+
+
+class ColDef(C.Structure):
+    _fields_ = [("name", C.c_char * 32), ("type", C.c_uint8),
+                ("scale", C.c_uint8), ("_pad", C.c_uint8 * 6)]
+
+
+class Options(C.Structure):
+    _fields_ = [("stripe_row_limit", C.c_uint64),
+                ("chunk_group_row_limit", C.c_uint32),
+                ("compression", C.c_uint8), ("compression_level", C.c_int8),
+                ("lz4_seg_target_kb", C.c_uint16)]
+
+
+class Pred(C.Structure):
+    _fields_ = [("column", C.c_uint32), ("op", C.c_uint32),
+                ("ival", C.c_int64), ("fval", C.c_double)]
+
+
+class AggSpec(C.Structure):
+    _fields_ = [("kind", C.c_uint32), ("col_a", C.c_int32), ("col_b", C.c_int32),
+                ("col_c", C.c_int32), ("one", C.c_int64)]
+
+
+class Partial(C.Structure):
+    _fields_ = [("i128_lo", C.c_int64), ("i128_hi", C.c_int64), ("f64", C.c_double),
+                ("count", C.c_int64), ("is_null", C.c_uint8), ("_pad", C.c_uint8 * 7)]
+
+    @property
+    def i128(self):
+        return (self.i128_hi << 64) | (self.i128_lo & ((1 << 64) - 1))
+
+    def as_dict(self):
+        return {"i128": self.i128, "f64": self.f64, "count": self.count,
+                "is_null": bool(self.is_null)}
+
+
+class Batch(C.Structure):
+    _fields_ = [("n_rows", C.c_uint32), ("first_row_number", C.c_uint64),
+                ("col_values", C.POINTER(C.c_void_p)),
+                ("col_nulls", C.POINTER(C.POINTER(C.c_uint8)))]
+
+
+class GroupResult(C.Structure):
+    _fields_ = [("n_groups", C.c_uint32), ("keys", C.c_uint16 * 64)]
+
+
+def _sig(name, res, args):
+    f = getattr(_lib, name)
+    f.restype = res
+    f.argtypes = args
+    return f
+
+
+_errmsg = _sig("cstripe_errmsg", C.c_char_p, [])
+_default_options = _sig("cstripe_default_options", None, [C.POINTER(Options)])
+_write_begin = _sig("cstripe_write_begin", C.c_void_p,
+                    [C.c_char_p, C.POINTER(ColDef), C.c_uint32, C.POINTER(Options)])
+_write_rows = _sig("cstripe_write_rows", C.c_int,
+                   [C.c_void_p, C.c_uint64, C.POINTER(C.c_void_p), C.POINTER(C.c_void_p)])
+_write_end = _sig("cstripe_write_end", C.c_int, [C.c_void_p])
+_open = _sig("cstripe_open", C.c_void_p, [C.c_char_p])
+_close = _sig("cstripe_close", None, [C.c_void_p])
+_row_count = _sig("cstripe_row_count", C.c_uint64, [C.c_void_p])
+_column_count = _sig("cstripe_column_count", C.c_uint32, [C.c_void_p])
+_stripe_count = _sig("cstripe_stripe_count", C.c_uint32, [C.c_void_p])
+_column_def = _sig("cstripe_column_def", C.c_int, [C.c_void_p, C.c_uint32, C.POINTER(ColDef)])
+_scan_begin = _sig("cstripe_scan_begin", C.c_void_p,
+                   [C.c_void_p, C.c_uint64, C.POINTER(Pred), C.c_uint32])
+_scan_end = _sig("cstripe_scan_end", None, [C.c_void_p])
+_filtered = _sig("cstripe_scan_chunk_groups_filtered", C.c_int64, [C.c_void_p])
+_gpu_stage = _sig("cstripe_gpu_stage", C.c_int, [C.c_void_p, C.c_int])
+_staged_bytes = _sig("cstripe_gpu_staged_bytes", C.c_uint64, [C.c_void_p])
+_scan_agg = _sig("cstripe_scan_agg", C.c_int,
+                 [C.c_void_p, C.POINTER(AggSpec), C.c_uint32, C.POINTER(Partial)])
+_scan_agg_grouped = _sig("cstripe_scan_agg_grouped", C.c_int,
+                         [C.c_void_p, C.POINTER(AggSpec), C.c_uint32,
+                          C.POINTER(C.c_uint32), C.c_uint32, C.POINTER(GroupResult),
+                          C.POINTER(Partial)])
+_next_batch = _sig("cstripe_scan_next_batch", C.c_int, [C.c_void_p, C.POINTER(Batch)])
+_rewind = _sig("cstripe_scan_rewind", C.c_int, [C.c_void_p])
+_last_kernel_ms = _sig("cstripe_scan_last_kernel_ms", C.c_double, [C.c_void_p])
+_last_decode_ms = _sig("cstripe_scan_last_decode_kernel_ms", C.c_double, [C.c_void_p])
+_last_agg_ms = _sig("cstripe_scan_last_agg_kernel_ms", C.c_double, [C.c_void_p])
+_combine = _sig("cagg_combine", C.c_int,
+                [C.POINTER(AggSpec), C.c_uint32, C.POINTER(Partial), C.c_uint32,
+                 C.POINTER(Partial)])
+_gpu_available = _sig("cstripe_gpu_available", C.c_int, [])
+_gen_lineitem = _sig("csbench_gen_lineitem", C.c_int,
+                     [C.c_char_p, C.c_uint64, C.c_uint64, C.c_int, C.c_int, C.c_int,
+                      C.c_uint64, C.c_uint32])
+
+
+def errmsg():
+    return _errmsg().decode()
+
+
+def gpu_available():
+    return bool(_gpu_available())
+
+
+class CStripeError(RuntimeError):
+    pass
+
+
+def _check(rc, what):
+    if rc != OK:
+        raise CStripeError(f"{what} failed (rc={rc}): {errmsg()}")
+
+
+def default_options(**kw):
+    o = Options()
+    _default_options(C.byref(o))
+    for k, v in kw.items():
+        setattr(o, k, v)
+    return o
+
+
+def make_coldefs(defs):
+    """defs: list of (name, type, scale)"""
+    arr = (ColDef * len(defs))()
+    for i, (name, typ, scale) in enumerate(defs):
+        arr[i].name = name.encode()
+        arr[i].type = typ
+        arr[i].scale = scale
+    return arr
+
+
+def write_table(path, defs, columns, nulls=None, **opt_kw):
+    """Write numpy columns to a stripe file. columns: list of np arrays
+    (dtype matching coldef types); nulls: optional list of uint8 arrays."""
+    import numpy as np
+    opts = default_options(**opt_kw)
+    cols = make_coldefs(defs)
+    w = _write_begin(path.encode(), cols, len(defs), C.byref(opts))
+    if not w:
+        raise CStripeError("write_begin: " + errmsg())
+    n = len(columns[0])
+    vals = (C.c_void_p * len(defs))()
+    for i, a in enumerate(columns):
+        a = np.ascontiguousarray(a)
+        columns[i] = a
+        vals[i] = a.ctypes.data_as(C.c_void_p).value
+    nl = None
+    if nulls is not None:
+        nl = (C.c_void_p * len(defs))()
+        for i, a in enumerate(nulls):
+            nl[i] = a.ctypes.data_as(C.c_void_p).value if a is not None else None
+    _check(_write_rows(w, n, vals, nl), "write_rows")
+    _check(_write_end(w), "write_end")
+
+
+def gen_lineitem(path, n_rows, seed=42, compression=COMP_LZ4, level=3, seg_kb=0,
+                 stripe_rows=0, chunk_rows=0):
+    _check(_gen_lineitem(path.encode(), n_rows, seed, compression, level, seg_kb,
+                         stripe_rows, chunk_rows), "gen_lineitem")
+
+
+LINEITEM_COLS = {"l_orderkey": 0, "l_quantity": 1, "l_extendedprice": 2,
+                 "l_discount": 3, "l_tax": 4, "l_shipdate": 5,
+                 "l_returnflag": 6, "l_linestatus": 7}
+
+
+class Reader:
+    def __init__(self, path):
+        self._h = _open(path.encode())
+        if not self._h:
+            raise CStripeError("open: " + errmsg())
+        self.path = path
+
+    def close(self):
+        if self._h:
+            _close(self._h)
+            self._h = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+    @property
+    def row_count(self):
+        return _row_count(self._h)
+
+    @property
+    def column_count(self):
+        return _column_count(self._h)
+
+    @property
+    def stripe_count(self):
+        return _stripe_count(self._h)
+
+    def column_def(self, i):
+        d = ColDef()
+        _check(_column_def(self._h, i, C.byref(d)), "column_def")
+        return d.name.decode(), d.type, d.scale
+
+    def scan(self, cols_mask=0, preds=()):
+        return Scan(self, cols_mask, preds)
+
+
+def make_preds(preds):
+    arr = (Pred * max(1, len(preds)))()
+    for i, p in enumerate(preds):
+        arr[i].column = p[0]
+        arr[i].op = p[1]
+        v = p[2]
+        if isinstance(v, float):
+            arr[i].fval = v
+            arr[i].ival = 0
+        else:
+            arr[i].ival = int(v)
+            arr[i].fval = 0.0
+    return arr
+
+
+def make_aggs(aggs):
+    """aggs: list of (kind, col_a[, col_b[, col_c[, one]]])"""
+    arr = (AggSpec * len(aggs))()
+    for i, a in enumerate(aggs):
+        a = tuple(a) + (-1, -1, -1, 0)[len(a) - 1:]
+        arr[i].kind, arr[i].col_a, arr[i].col_b, arr[i].col_c, arr[i].one = a[:5]
+    return arr
+
+
+class Scan:
+    def __init__(self, reader, cols_mask, preds):
+        self._preds = make_preds(preds)
+        self._h = _scan_begin(reader._h, cols_mask, self._preds, len(preds))
+        if not self._h:
+            raise CStripeError("scan_begin: " + errmsg())
+        self.reader = reader
+
+    def end(self):
+        if self._h:
+            _scan_end(self._h)
+            self._h = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.end()
+
+    @property
+    def chunk_groups_filtered(self):
+        return _filtered(self._h)
+
+    @property
+    def staged_bytes(self):
+        return _staged_bytes(self._h)
+
+    @property
+    def last_kernel_ms(self):
+        return _last_kernel_ms(self._h)
+
+    @property
+    def last_decode_ms(self):
+        return _last_decode_ms(self._h)
+
+    @property
+    def last_agg_ms(self):
+        return _last_agg_ms(self._h)
+
+    def stage(self, device=-1):
+        _check(_gpu_stage(self._h, device), "gpu_stage")
+
+    def agg(self, aggs):
+        """aggs: list of (kind, col_a[, col_b[, col_c[, one]]]) -> [Partial]"""
+        arr = make_aggs(aggs)
+        out = (Partial * len(aggs))()
+        _check(_scan_agg(self._h, arr, len(aggs), out), "scan_agg")
+        return list(out)
+
+    def agg_grouped(self, aggs, group_cols):
+        arr = make_aggs(aggs)
+        gc = (C.c_uint32 * len(group_cols))(*group_cols)
+        gr = GroupResult()
+        out = (Partial * (64 * len(aggs)))()
+        _check(_scan_agg_grouped(self._h, arr, len(aggs), gc, len(group_cols),
+                                 C.byref(gr), out), "scan_agg_grouped")
+        res = {}
+        for g in range(gr.n_groups):
+            key = (gr.keys[g] & 0xFF, gr.keys[g] >> 8)
+            res[key] = [out[g * len(aggs) + a] for a in range(len(aggs))]
+        return res
+
+    def next_batch(self, col_arrays, null_arrays=None):
+        """col_arrays: dict col_index -> numpy array (capacity chunk rows).
+        Returns (n_rows, first_row_number) or None at end."""
+        ncols = self.reader.column_count
+        b = Batch()
+        vals = (C.c_void_p * ncols)()
+        for ci, a in col_arrays.items():
+            vals[ci] = a.ctypes.data_as(C.c_void_p).value
+        b.col_values = vals
+        if null_arrays:
+            nl = (C.POINTER(C.c_uint8) * ncols)()
+            for ci, a in null_arrays.items():
+                nl[ci] = a.ctypes.data_as(C.POINTER(C.c_uint8))
+            b.col_nulls = nl
+        rc = _next_batch(self._h, C.byref(b))
+        if rc == END:
+            return None
+        _check(rc, "next_batch")
+        return b.n_rows, b.first_row_number
+
+    def rewind(self):
+        _check(_rewind(self._h), "rewind")
+
+
+def combine(aggs, parts_list):
+    """parts_list: list (per shard/GPU) of lists of Partial -> [Partial].
+    The coordinator-merge step (cagg_combine)."""
+    n_aggs = len(aggs)
+    arr = make_aggs(aggs)
+    flat = (Partial * (n_aggs * len(parts_list)))()
+    for p, parts in enumerate(parts_list):
+        for a in range(n_aggs):
+            flat[p * n_aggs + a] = parts[a]
+    out = (Partial * n_aggs)()
+    _check(_combine(arr, n_aggs, flat, len(parts_list), out), "combine")
+    return list(out)

@@ -1,0 +1,75 @@
+/*
+ * bench_gen.cpp — seeded synthetic lineitem-shaped data generator
+ * (bench/test infrastructure; value distributions per SURVEY.md §8d:
+ * xorshift64 seed 42; quantity fixed-point [100,5000]; price [900,105000];
+ * discount [0,10]; tax [0,8]; shipdate uniform over 7 years of days
+ * (1992-01-01=8035 .. 1998-12-31=10591); returnflag A/N/R 25/25/50;
+ * linestatus O/F 50/50). Physical schema (fixed-point int64 scale 2 for
+ * decimal(15,2) measures; i8 categorical codes for char(1) flags):
+ *   0 l_orderkey i64, 1 l_quantity i64/s2, 2 l_extendedprice i64/s2,
+ *   3 l_discount i64/s2, 4 l_tax i64/s2, 5 l_shipdate i64 (days),
+ *   6 l_returnflag i8 (0=A,1=N,2=R), 7 l_linestatus i8 (0=O,1=F)
+ */
+#include "../../include/cstripe.h"
+
+#include <cstdint>
+#include <cstring>
+#include <vector>
+
+static inline uint64_t xs64(uint64_t &x)
+{
+    x ^= x << 13;
+    x ^= x >> 7;
+    x ^= x << 17;
+    return x;
+}
+
+extern "C" int csbench_gen_lineitem(const char *path, uint64_t n_rows, uint64_t seed,
+                                    int compression, int level, int seg_kb,
+                                    uint64_t stripe_rows, uint32_t chunk_rows)
+{
+    cstripe_coldef cols[8] = {};
+    const char *names[8] = {"l_orderkey", "l_quantity", "l_extendedprice", "l_discount",
+                            "l_tax", "l_shipdate", "l_returnflag", "l_linestatus"};
+    for (int i = 0; i < 8; i++) {
+        strncpy(cols[i].name, names[i], 31);
+        cols[i].type = (i < 6) ? CSTRIPE_I64 : CSTRIPE_I8;
+        cols[i].scale = (i >= 1 && i <= 4) ? 2 : 0;
+    }
+    cstripe_options opts;
+    cstripe_default_options(&opts);
+    if (stripe_rows) opts.stripe_row_limit = stripe_rows;
+    if (chunk_rows) opts.chunk_group_row_limit = chunk_rows;
+    opts.compression = (uint8_t)compression;
+    opts.compression_level = (int8_t)level;
+    if (seg_kb > 0) opts.lz4_seg_target_kb = (uint16_t)seg_kb;
+
+    cstripe_writer *w = cstripe_write_begin(path, cols, 8, &opts);
+    if (!w) return CSTRIPE_ERR;
+
+    uint64_t x = seed ? seed : 42;
+    const uint64_t BATCH = 1u << 20;
+    std::vector<int64_t> c0(BATCH), c1(BATCH), c2(BATCH), c3(BATCH), c4(BATCH), c5(BATCH);
+    std::vector<int8_t> c6(BATCH), c7(BATCH);
+    uint64_t done = 0, orderkey = 1;
+    while (done < n_rows) {
+        uint64_t n = n_rows - done < BATCH ? n_rows - done : BATCH;
+        for (uint64_t i = 0; i < n; i++) {
+            c0[i] = (int64_t)orderkey++;
+            c1[i] = 100 + (int64_t)(xs64(x) % 4901);            /* quantity 1.00..50.00 */
+            c2[i] = 900 + (int64_t)(xs64(x) % 104101);          /* price 9.00..1050.00 */
+            c3[i] = (int64_t)(xs64(x) % 11);                    /* discount 0.00..0.10 */
+            c4[i] = (int64_t)(xs64(x) % 9);                     /* tax 0.00..0.08 */
+            c5[i] = 8035 + (int64_t)(xs64(x) % 2557);           /* shipdate 1992..1998 */
+            uint64_t rf = xs64(x) % 4;                          /* A/N 25/25, R 50 */
+            c6[i] = (int8_t)(rf == 0 ? 0 : (rf == 1 ? 1 : 2));
+            c7[i] = (int8_t)(xs64(x) % 2);
+        }
+        const void *vals[8] = {c0.data(), c1.data(), c2.data(), c3.data(),
+                               c4.data(), c5.data(), c6.data(), c7.data()};
+        int rc = cstripe_write_rows(w, n, vals, nullptr);
+        if (rc != CSTRIPE_OK) { cstripe_write_abort(w); return rc; }
+        done += n;
+    }
+    return cstripe_write_end(w);
+}

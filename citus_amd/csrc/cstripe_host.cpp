@@ -1,0 +1,823 @@
+/*
+ * cstripe_host.cpp — host side of the MI355X-native columnar scan path:
+ * format producer (writer), flat-file stripe directory (reader/footer),
+ * chunk-group min/max pruning, and the combine surface.
+ *
+ * Reference semantics restated (never copied) from:
+ *   writer:  columnar_writer.c  ColumnarWriteRow :168-266, FlushStripe
+ *            :391-516, SerializeBoolArray :523-545, SerializeSingleDatum
+ *            :555-585, SerializeChunkData :592-654, UpdateChunkSkipNodeMinMax
+ *            :663-718
+ *   codec:   columnar_compression.c CompressBuffer :62-158 (lz4 :71-94,
+ *            zstd :97-121), DecompressBuffer :165-270
+ *   pruning: columnar_reader.c SelectedChunkMask :1132-1187 (range
+ *            refutation per qual var; count a chunk once)
+ *   combine: distributed/utils/aggregate_utils.c coord_combine_agg core
+ *            :820-1021 (strict: skip NULL partials, first non-NULL
+ *            initializes); COUNT NULL->0 COALESCE
+ *            multi_logical_optimizer.c:1831-1885
+ */
+#include "internal.h"
+#include "compress.h"
+
+#include <cstdarg>
+#include <cstdio>
+#include <cstring>
+#include <cstdlib>
+#include <algorithm>
+
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+/* ============================ errors ============================ */
+
+static thread_local char g_errbuf[512] = "";
+
+void cs_set_err(const char *fmt, ...)
+{
+    va_list ap;
+    va_start(ap, fmt);
+    vsnprintf(g_errbuf, sizeof(g_errbuf), fmt, ap);
+    va_end(ap);
+}
+
+extern "C" const char *cstripe_errmsg(void) { return g_errbuf; }
+extern "C" uint32_t cstripe_abi_version(void) { return CSTRIPE_ABI_VERSION; }
+
+extern "C" void cstripe_default_options(cstripe_options *o)
+{
+    /* defaults mirror columnar.c:30-44 (stripe 150000, chunk 10000, level 3;
+     * default codec here is LZ4 — the GPU decode path's native codec) */
+    o->stripe_row_limit = 150000;
+    o->chunk_group_row_limit = 10000;
+    o->compression = CSTRIPE_COMP_LZ4;
+    o->compression_level = 3;
+    o->lz4_seg_target_kb = 16;
+}
+
+static uint32_t type_width(uint8_t t) { return csf_type_width(t); }
+
+/* ============================ writer ============================ */
+
+namespace {
+
+struct ChunkOut {
+    std::vector<uint8_t> exists_packed;
+    std::vector<uint8_t> value_comp;      /* compressed (or raw) value stream */
+    std::vector<csf_seg> segs;
+    csf_skipnode node;                    /* offsets filled at flush */
+};
+
+struct ColCur {
+    std::vector<uint8_t> values;          /* packed present values, current chunk */
+    std::vector<uint8_t> exists;          /* 1 byte per row, current chunk */
+    bool has_min_max = false;
+    int64_t min_i = 0, max_i = 0;         /* physical encoding (f64 via bit pattern) */
+    std::vector<uint8_t> raw_pending;     /* finalized raw value streams, pending compress */
+};
+
+struct PendingChunk {                     /* raw chunk awaiting compression at flush */
+    uint32_t col;
+    uint32_t chunk;
+    std::vector<uint8_t> raw_values;
+    std::vector<uint8_t> exists_packed;
+    csf_skipnode node;                    /* min/max/row_count/n_present/decomp filled */
+};
+
+} /* namespace */
+
+struct cstripe_writer {
+    FILE *f = nullptr;
+    std::string path;
+    std::vector<csf_coldef> cols;
+    cstripe_options opts{};
+    uint64_t file_pos = 0;
+    uint64_t total_rows = 0;
+
+    /* current stripe */
+    uint64_t stripe_first_row = 0;
+    uint32_t stripe_rows = 0;             /* rows accumulated in current stripe */
+    std::vector<ColCur> cur;              /* per column, current chunk state */
+    std::vector<PendingChunk> pending;    /* finalized raw chunks of current stripe */
+    std::vector<uint32_t> chunk_rows;     /* rows per finalized chunk */
+
+    /* footer accumulation */
+    std::vector<cs_stripe_info> stripes;
+};
+
+/* typed min/max update over a span of present values */
+static void update_minmax(ColCur &c, uint8_t type, const uint8_t *vals, size_t n)
+{
+    if (n == 0) return;
+    switch (type) {
+        case CSTRIPE_I8: {
+            const int8_t *v = (const int8_t *)vals;
+            int8_t mn = v[0], mx = v[0];
+            for (size_t i = 1; i < n; i++) { if (v[i] < mn) mn = v[i]; if (v[i] > mx) mx = v[i]; }
+            if (!c.has_min_max) { c.min_i = mn; c.max_i = mx; c.has_min_max = true; }
+            else { if (mn < (int8_t)c.min_i) c.min_i = mn; if (mx > (int8_t)c.max_i) c.max_i = mx; }
+            break;
+        }
+        case CSTRIPE_I16: {
+            const int16_t *v = (const int16_t *)vals;
+            int16_t mn = v[0], mx = v[0];
+            for (size_t i = 1; i < n; i++) { if (v[i] < mn) mn = v[i]; if (v[i] > mx) mx = v[i]; }
+            if (!c.has_min_max) { c.min_i = mn; c.max_i = mx; c.has_min_max = true; }
+            else { if (mn < (int16_t)c.min_i) c.min_i = mn; if (mx > (int16_t)c.max_i) c.max_i = mx; }
+            break;
+        }
+        case CSTRIPE_I32: {
+            const int32_t *v = (const int32_t *)vals;
+            int32_t mn = v[0], mx = v[0];
+            for (size_t i = 1; i < n; i++) { if (v[i] < mn) mn = v[i]; if (v[i] > mx) mx = v[i]; }
+            if (!c.has_min_max) { c.min_i = mn; c.max_i = mx; c.has_min_max = true; }
+            else { if (mn < (int32_t)c.min_i) c.min_i = mn; if (mx > (int32_t)c.max_i) c.max_i = mx; }
+            break;
+        }
+        case CSTRIPE_I64: {
+            const int64_t *v = (const int64_t *)vals;
+            int64_t mn = v[0], mx = v[0];
+            for (size_t i = 1; i < n; i++) { if (v[i] < mn) mn = v[i]; if (v[i] > mx) mx = v[i]; }
+            if (!c.has_min_max) { c.min_i = mn; c.max_i = mx; c.has_min_max = true; }
+            else { if (mn < c.min_i) c.min_i = mn; if (mx > c.max_i) c.max_i = mx; }
+            break;
+        }
+        case CSTRIPE_F32: {
+            const float *v = (const float *)vals;
+            float mn = v[0], mx = v[0];
+            for (size_t i = 1; i < n; i++) { if (v[i] < mn) mn = v[i]; if (v[i] > mx) mx = v[i]; }
+            double dmn = mn, dmx = mx;
+            if (!c.has_min_max) { memcpy(&c.min_i, &dmn, 8); memcpy(&c.max_i, &dmx, 8); c.has_min_max = true; }
+            else {
+                double omn, omx; memcpy(&omn, &c.min_i, 8); memcpy(&omx, &c.max_i, 8);
+                if (dmn < omn) memcpy(&c.min_i, &dmn, 8);
+                if (dmx > omx) memcpy(&c.max_i, &dmx, 8);
+            }
+            break;
+        }
+        case CSTRIPE_F64: {
+            const double *v = (const double *)vals;
+            double mn = v[0], mx = v[0];
+            for (size_t i = 1; i < n; i++) { if (v[i] < mn) mn = v[i]; if (v[i] > mx) mx = v[i]; }
+            if (!c.has_min_max) { memcpy(&c.min_i, &mn, 8); memcpy(&c.max_i, &mx, 8); c.has_min_max = true; }
+            else {
+                double omn, omx; memcpy(&omn, &c.min_i, 8); memcpy(&omx, &c.max_i, 8);
+                if (mn < omn) memcpy(&c.min_i, &mn, 8);
+                if (mx > omx) memcpy(&c.max_i, &mx, 8);
+            }
+            break;
+        }
+    }
+}
+
+/* SerializeBoolArray (columnar_writer.c:523-545): bit i of byte i/8, LSB first */
+static std::vector<uint8_t> pack_bools(const std::vector<uint8_t> &b)
+{
+    size_t n = b.size();
+    std::vector<uint8_t> out((n + 7) / 8, 0);
+    for (size_t i = 0; i < n; i++)
+        if (b[i]) out[i / 8] |= (uint8_t)(1u << (i % 8));
+    return out;
+}
+
+extern "C" cstripe_writer *cstripe_write_begin(const char *path, const cstripe_coldef *cols,
+                                               uint32_t n_cols, const cstripe_options *opts)
+{
+    if (!path || !cols || n_cols == 0 || n_cols > 64) { cs_set_err("write_begin: bad args"); return nullptr; }
+    FILE *f = fopen(path, "wb");
+    if (!f) { cs_set_err("write_begin: cannot open %s", path); return nullptr; }
+
+    auto *w = new cstripe_writer();
+    w->f = f;
+    w->path = path;
+    if (opts) w->opts = *opts; else cstripe_default_options(&w->opts);
+    if (w->opts.lz4_seg_target_kb == 0) w->opts.lz4_seg_target_kb = 16;
+    for (uint32_t i = 0; i < n_cols; i++) {
+        csf_coldef d{};
+        memcpy(d.name, cols[i].name, sizeof(d.name));
+        d.name[31] = 0;
+        d.type = cols[i].type;
+        d.scale = cols[i].scale;
+        if (type_width(d.type) == 0) { cs_set_err("write_begin: bad column type %d", d.type); fclose(f); delete w; return nullptr; }
+        w->cols.push_back(d);
+    }
+    w->cur.resize(n_cols);
+
+    uint8_t hdr[CSF_HEADER_SIZE] = {0};
+    memcpy(hdr, CSF_MAGIC, 8);
+    uint32_t ver = CSF_VERSION;
+    memcpy(hdr + 8, &ver, 4);
+    fwrite(hdr, 1, CSF_HEADER_SIZE, f);
+    w->file_pos = CSF_HEADER_SIZE;
+    return w;
+}
+
+/* finalize the current chunk of every column into w->pending (raw; compression
+ * deferred to flush so it can run in parallel across chunks) */
+static void finalize_chunk(cstripe_writer *w, uint32_t rows_in_chunk)
+{
+    uint32_t chunk_index = (uint32_t)w->chunk_rows.size();
+    for (uint32_t c = 0; c < w->cols.size(); c++) {
+        ColCur &cc = w->cur[c];
+        PendingChunk pc;
+        pc.col = c;
+        pc.chunk = chunk_index;
+        pc.raw_values = std::move(cc.values);
+        pc.exists_packed = pack_bools(cc.exists);
+        memset(&pc.node, 0, sizeof(pc.node));
+        pc.node.min_i = cc.min_i;
+        pc.node.max_i = cc.max_i;
+        pc.node.has_min_max = cc.has_min_max ? 1 : 0;
+        pc.node.row_count = rows_in_chunk;
+        uint32_t width = type_width(w->cols[c].type);
+        pc.node.n_present = (uint32_t)(pc.raw_values.size() / width);
+        pc.node.decompressed_size = pc.raw_values.size();
+        w->pending.push_back(std::move(pc));
+        cc.values.clear();
+        cc.exists.clear();
+        cc.has_min_max = false;
+        cc.min_i = cc.max_i = 0;
+    }
+    w->chunk_rows.push_back(rows_in_chunk);
+}
+
+/* compress one pending chunk's value stream into out (segments); mirrors
+ * SerializeChunkData/CompressBuffer: lz4/zstd per segment, raw when the codec
+ * is NONE or fails */
+static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, ChunkOut &out)
+{
+    out.node = pc.node;
+    out.exists_packed = pc.exists_packed;
+    const std::vector<uint8_t> &raw = pc.raw_values;
+    uint8_t codec = w->opts.compression;
+    size_t target = (size_t)w->opts.lz4_seg_target_kb * 1024;
+    uint32_t width = type_width(w->cols[pc.col].type);
+
+    if (raw.empty() || codec == CSTRIPE_COMP_NONE) {
+        out.value_comp = raw;
+        csf_seg s{0, (uint32_t)raw.size(), 0, (uint32_t)raw.size()};
+        out.segs.push_back(s);
+        out.node.comp_type = CSTRIPE_COMP_NONE;
+        out.node.comp_level = 0;
+        out.node.n_segs = 1;
+        out.node.value_len = raw.size();
+        return;
+    }
+
+    /* segment boundaries aligned to value width */
+    size_t n = raw.size();
+    size_t nseg = (n + target - 1) / target;
+    if (nseg < 1) nseg = 1;
+    size_t per = ((n / nseg) / width) * width;
+    if (per == 0) { per = n; nseg = 1; }
+
+    out.value_comp.clear();
+    out.segs.clear();
+    size_t off = 0;
+    bool ok = true;
+    while (off < n) {
+        size_t len = std::min(per, n - off);
+        if (n - off - len < width) len = n - off;   /* absorb the tail */
+        csf_seg s{};
+        s.decomp_off = (uint32_t)off;
+        s.decomp_len = (uint32_t)len;
+        s.comp_off = (uint32_t)out.value_comp.size();
+        if (codec == CSTRIPE_COMP_LZ4) {
+            int bound = LZ4_compressBound((int)len);
+            size_t base = out.value_comp.size();
+            out.value_comp.resize(base + (size_t)bound);
+            int csz = LZ4_compress_default((const char *)raw.data() + off,
+                                           (char *)out.value_comp.data() + base,
+                                           (int)len, bound);
+            if (csz <= 0) { ok = false; break; }
+            out.value_comp.resize(base + (size_t)csz);
+            s.comp_len = (uint32_t)csz;
+        } else if (codec == CSTRIPE_COMP_ZSTD) {
+            size_t bound = ZSTD_compressBound(len);
+            size_t base = out.value_comp.size();
+            out.value_comp.resize(base + bound);
+            size_t csz = ZSTD_compress(out.value_comp.data() + base, bound,
+                                       raw.data() + off, len, w->opts.compression_level);
+            if (ZSTD_isError(csz)) { ok = false; break; }
+            out.value_comp.resize(base + csz);
+            s.comp_len = (uint32_t)csz;
+        } else {
+            ok = false; break;
+        }
+        out.segs.push_back(s);
+        off += len;
+    }
+
+    if (!ok) {  /* CompressBuffer returned false -> keep uncompressed */
+        out.value_comp = raw;
+        out.segs.clear();
+        csf_seg s{0, (uint32_t)raw.size(), 0, (uint32_t)raw.size()};
+        out.segs.push_back(s);
+        out.node.comp_type = CSTRIPE_COMP_NONE;
+        out.node.comp_level = 0;
+    } else {
+        out.node.comp_type = codec;
+        out.node.comp_level = w->opts.compression_level;
+    }
+    out.node.n_segs = (uint16_t)out.segs.size();
+    out.node.value_len = out.value_comp.size();
+}
+
+/* FlushStripe (columnar_writer.c:391-516): compress pending chunks, lay out
+ * per column all exists streams then all value streams, offsets relative to
+ * stripe data start; append stripe metadata for the footer */
+static int flush_stripe(cstripe_writer *w)
+{
+    if (w->stripe_rows == 0) return CSTRIPE_OK;
+    /* finalize a partial last chunk (FlushStripe :419-422) */
+    uint32_t done_rows = 0;
+    for (uint32_t r : w->chunk_rows) done_rows += r;
+    if (done_rows < w->stripe_rows)
+        finalize_chunk(w, w->stripe_rows - done_rows);
+
+    uint32_t n_cols = (uint32_t)w->cols.size();
+    uint32_t n_chunks = (uint32_t)w->chunk_rows.size();
+
+    /* compress all pending chunks (parallel; independent per column-chunk) */
+    std::vector<ChunkOut> outs(w->pending.size());
+    #pragma omp parallel for schedule(dynamic)
+    for (long i = 0; i < (long)w->pending.size(); i++)
+        compress_chunk(w, w->pending[i], outs[i]);
+
+    /* outs indexed by pending order: chunk-major (all columns of chunk 0, then
+     * chunk 1, ...). Reorder access as [col][chunk]. */
+    auto out_at = [&](uint32_t col, uint32_t chunk) -> ChunkOut & {
+        return outs[(size_t)chunk * n_cols + col];
+    };
+
+    /* compute offsets: per column, exists streams then value streams
+     * (FlushStripe :425-458) */
+    uint64_t stripe_size = 0;
+    for (uint32_t c = 0; c < n_cols; c++) {
+        for (uint32_t k = 0; k < n_chunks; k++) {
+            ChunkOut &o = out_at(c, k);
+            o.node.exists_off = stripe_size;
+            o.node.exists_len = o.exists_packed.size();
+            stripe_size += o.exists_packed.size();
+        }
+        for (uint32_t k = 0; k < n_chunks; k++) {
+            ChunkOut &o = out_at(c, k);
+            o.node.value_off = stripe_size;
+            stripe_size += o.node.value_len;
+        }
+    }
+
+    uint64_t stripe_file_off = w->file_pos;
+
+    /* write data: per column all exists, then all values (FlushStripe :477-502) */
+    for (uint32_t c = 0; c < n_cols; c++) {
+        for (uint32_t k = 0; k < n_chunks; k++) {
+            ChunkOut &o = out_at(c, k);
+            if (!o.exists_packed.empty() &&
+                fwrite(o.exists_packed.data(), 1, o.exists_packed.size(), w->f) != o.exists_packed.size())
+                { cs_set_err("flush: short write"); return CSTRIPE_ERR_IO; }
+        }
+        for (uint32_t k = 0; k < n_chunks; k++) {
+            ChunkOut &o = out_at(c, k);
+            if (!o.value_comp.empty() &&
+                fwrite(o.value_comp.data(), 1, o.value_comp.size(), w->f) != o.value_comp.size())
+                { cs_set_err("flush: short write"); return CSTRIPE_ERR_IO; }
+        }
+    }
+    w->file_pos += stripe_size;
+
+    /* record stripe for the footer */
+    cs_stripe_info si;
+    si.meta.file_offset = stripe_file_off;
+    si.meta.data_size = stripe_size;
+    si.meta.first_row_number = w->stripe_first_row;
+    si.meta.row_count = w->stripe_rows;
+    si.meta.chunk_count = n_chunks;
+    si.meta.reserved = 0;
+    si.group_rows = w->chunk_rows;
+    si.nodes.resize(n_cols);
+    for (uint32_t c = 0; c < n_cols; c++) {
+        si.nodes[c].resize(n_chunks);
+        for (uint32_t k = 0; k < n_chunks; k++) {
+            si.nodes[c][k].n = out_at(c, k).node;
+            si.nodes[c][k].segs = out_at(c, k).segs;
+        }
+    }
+    w->stripes.push_back(std::move(si));
+
+    /* reset stripe state */
+    w->pending.clear();
+    w->chunk_rows.clear();
+    w->stripe_first_row += w->stripe_rows;
+    w->stripe_rows = 0;
+    return CSTRIPE_OK;
+}
+
+extern "C" int cstripe_write_rows(cstripe_writer *w, uint64_t n_rows,
+                                  const void *const *values, const uint8_t *const *nulls)
+{
+    if (!w || !values) { cs_set_err("write_rows: bad args"); return CSTRIPE_ERR_ARG; }
+    uint32_t n_cols = (uint32_t)w->cols.size();
+    const uint32_t chunk_limit = w->opts.chunk_group_row_limit;
+    const uint64_t stripe_limit = w->opts.stripe_row_limit;
+
+    uint64_t done = 0;
+    while (done < n_rows) {
+        uint32_t in_chunk = w->stripe_rows % chunk_limit;
+        uint64_t to_chunk = chunk_limit - in_chunk;
+        uint64_t to_stripe = stripe_limit - w->stripe_rows;
+        uint64_t span = std::min({n_rows - done, to_chunk, to_stripe});
+
+        for (uint32_t c = 0; c < n_cols; c++) {
+            ColCur &cc = w->cur[c];
+            uint32_t width = type_width(w->cols[c].type);
+            const uint8_t *src = (const uint8_t *)values[c] + done * width;
+            const uint8_t *nl = (nulls && nulls[c]) ? nulls[c] + done : nullptr;
+            size_t base = cc.exists.size();
+            cc.exists.resize(base + span);
+            if (!nl) {
+                memset(cc.exists.data() + base, 1, span);
+                size_t vbase = cc.values.size();
+                cc.values.resize(vbase + span * width);
+                memcpy(cc.values.data() + vbase, src, span * width);
+                update_minmax(cc, w->cols[c].type, src, span);
+            } else {
+                for (uint64_t i = 0; i < span; i++) {
+                    if (nl[i]) { cc.exists[base + i] = 0; continue; }
+                    cc.exists[base + i] = 1;
+                    size_t vbase = cc.values.size();
+                    cc.values.resize(vbase + width);
+                    memcpy(cc.values.data() + vbase, src + i * width, width);
+                    update_minmax(cc, w->cols[c].type, src + i * width, 1);
+                }
+            }
+        }
+
+        w->stripe_rows += (uint32_t)span;
+        w->total_rows += span;
+        done += span;
+
+        if (w->stripe_rows % chunk_limit == 0)
+            finalize_chunk(w, chunk_limit);
+        if (w->stripe_rows >= stripe_limit) {
+            int rc = flush_stripe(w);
+            if (rc != CSTRIPE_OK) return rc;
+        }
+    }
+    return CSTRIPE_OK;
+}
+
+static void footer_append(std::vector<uint8_t> &buf, const void *p, size_t n)
+{
+    const uint8_t *b = (const uint8_t *)p;
+    buf.insert(buf.end(), b, b + n);
+}
+
+extern "C" int cstripe_write_end(cstripe_writer *w)
+{
+    if (!w) return CSTRIPE_ERR_ARG;
+    int rc = flush_stripe(w);
+    if (rc != CSTRIPE_OK) { cstripe_write_abort(w); return rc; }
+
+    std::vector<uint8_t> fb;
+    csf_footer_head head{};
+    head.version = CSF_VERSION;
+    head.column_count = (uint32_t)w->cols.size();
+    head.stripe_count = (uint32_t)w->stripes.size();
+    head.chunk_row_limit = w->opts.chunk_group_row_limit;
+    head.stripe_row_limit = w->opts.stripe_row_limit;
+    head.total_rows = w->total_rows;
+    head.compression = w->opts.compression;
+    head.compression_level = w->opts.compression_level;
+    head.lz4_seg_target_kb = w->opts.lz4_seg_target_kb;
+    footer_append(fb, &head, sizeof(head));
+    for (auto &c : w->cols) footer_append(fb, &c, sizeof(c));
+    for (auto &s : w->stripes) {
+        footer_append(fb, &s.meta, sizeof(s.meta));
+        footer_append(fb, s.group_rows.data(), s.group_rows.size() * 4);
+        for (auto &colnodes : s.nodes) {
+            for (auto &nd : colnodes) {
+                footer_append(fb, &nd.n, sizeof(nd.n));
+                footer_append(fb, nd.segs.data(), nd.segs.size() * sizeof(csf_seg));
+            }
+        }
+    }
+    uint64_t foff = w->file_pos;
+    if (fwrite(fb.data(), 1, fb.size(), w->f) != fb.size()) { cs_set_err("footer: short write"); cstripe_write_abort(w); return CSTRIPE_ERR_IO; }
+    fwrite(&foff, 1, 8, w->f);
+    fwrite(CSF_FOOT_MAGIC, 1, 8, w->f);
+    int frc = fclose(w->f);
+    w->f = nullptr;
+    delete w;
+    if (frc != 0) { cs_set_err("close failed"); return CSTRIPE_ERR_IO; }
+    return CSTRIPE_OK;
+}
+
+extern "C" void cstripe_write_abort(cstripe_writer *w)
+{
+    if (!w) return;
+    if (w->f) fclose(w->f);
+    delete w;
+}
+
+/* ============================ reader ============================ */
+
+static int footer_read(cstripe_reader *r)
+{
+    if (r->map_size < CSF_HEADER_SIZE + 16) { cs_set_err("file too small"); return CSTRIPE_ERR_FORMAT; }
+    if (memcmp(r->map, CSF_MAGIC, 8) != 0) { cs_set_err("bad header magic"); return CSTRIPE_ERR_FORMAT; }
+    if (memcmp(r->map + r->map_size - 8, CSF_FOOT_MAGIC, 8) != 0) { cs_set_err("bad footer magic"); return CSTRIPE_ERR_FORMAT; }
+    uint64_t foff;
+    memcpy(&foff, r->map + r->map_size - 16, 8);
+    if (foff + 16 > r->map_size) { cs_set_err("bad footer offset"); return CSTRIPE_ERR_FORMAT; }
+
+    const uint8_t *p = r->map + foff;
+    const uint8_t *end = r->map + r->map_size - 16;
+    auto need = [&](size_t n) -> bool { return (size_t)(end - p) >= n; };
+
+    if (!need(sizeof(csf_footer_head))) { cs_set_err("truncated footer"); return CSTRIPE_ERR_FORMAT; }
+    memcpy(&r->head, p, sizeof(r->head));
+    p += sizeof(r->head);
+    if (r->head.version != CSF_VERSION) { cs_set_err("bad version %u", r->head.version); return CSTRIPE_ERR_FORMAT; }
+    if (r->head.column_count == 0 || r->head.column_count > 64) { cs_set_err("bad column count"); return CSTRIPE_ERR_FORMAT; }
+
+    r->cols.resize(r->head.column_count);
+    for (auto &c : r->cols) {
+        if (!need(sizeof(c))) { cs_set_err("truncated coldefs"); return CSTRIPE_ERR_FORMAT; }
+        memcpy(&c, p, sizeof(c));
+        p += sizeof(c);
+    }
+    r->stripes.resize(r->head.stripe_count);
+    for (auto &s : r->stripes) {
+        if (!need(sizeof(s.meta))) { cs_set_err("truncated stripe meta"); return CSTRIPE_ERR_FORMAT; }
+        memcpy(&s.meta, p, sizeof(s.meta));
+        p += sizeof(s.meta);
+        if (s.meta.chunk_count > 1u << 22) { cs_set_err("bad chunk count"); return CSTRIPE_ERR_FORMAT; }
+        s.group_rows.resize(s.meta.chunk_count);
+        if (!need(s.group_rows.size() * 4)) { cs_set_err("truncated group rows"); return CSTRIPE_ERR_FORMAT; }
+        memcpy(s.group_rows.data(), p, s.group_rows.size() * 4);
+        p += s.group_rows.size() * 4;
+        s.nodes.resize(r->head.column_count);
+        for (auto &cn : s.nodes) {
+            cn.resize(s.meta.chunk_count);
+            for (auto &nd : cn) {
+                if (!need(sizeof(nd.n))) { cs_set_err("truncated skipnode"); return CSTRIPE_ERR_FORMAT; }
+                memcpy(&nd.n, p, sizeof(nd.n));
+                p += sizeof(nd.n);
+                if (nd.n.n_segs == 0 || nd.n.n_segs > 4096) { cs_set_err("bad n_segs"); return CSTRIPE_ERR_FORMAT; }
+                nd.segs.resize(nd.n.n_segs);
+                if (!need(nd.segs.size() * sizeof(csf_seg))) { cs_set_err("truncated segs"); return CSTRIPE_ERR_FORMAT; }
+                memcpy(nd.segs.data(), p, nd.segs.size() * sizeof(csf_seg));
+                p += nd.segs.size() * sizeof(csf_seg);
+            }
+        }
+    }
+    return CSTRIPE_OK;
+}
+
+extern "C" cstripe_reader *cstripe_open(const char *path)
+{
+    int fd = open(path, O_RDONLY);
+    if (fd < 0) { cs_set_err("open %s failed", path); return nullptr; }
+    struct stat st;
+    if (fstat(fd, &st) != 0 || st.st_size <= 0) { cs_set_err("stat failed"); close(fd); return nullptr; }
+    void *m = mmap(nullptr, (size_t)st.st_size, PROT_READ, MAP_PRIVATE, fd, 0);
+    if (m == MAP_FAILED) { cs_set_err("mmap failed"); close(fd); return nullptr; }
+    auto *r = new cstripe_reader();
+    r->fd = fd;
+    r->map = (const uint8_t *)m;
+    r->map_size = (size_t)st.st_size;
+    if (footer_read(r) != CSTRIPE_OK) { cstripe_close(r); return nullptr; }
+    return r;
+}
+
+extern "C" void cstripe_close(cstripe_reader *r)
+{
+    if (!r) return;
+    if (r->map) munmap((void *)r->map, r->map_size);
+    if (r->fd >= 0) close(r->fd);
+    delete r;
+}
+
+extern "C" uint64_t cstripe_row_count(const cstripe_reader *r) { return r ? r->head.total_rows : 0; }
+extern "C" uint32_t cstripe_column_count(const cstripe_reader *r) { return r ? r->head.column_count : 0; }
+extern "C" uint32_t cstripe_stripe_count(const cstripe_reader *r) { return r ? r->head.stripe_count : 0; }
+
+extern "C" int cstripe_column_def(const cstripe_reader *r, uint32_t col, cstripe_coldef *out)
+{
+    if (!r || col >= r->cols.size() || !out) return CSTRIPE_ERR_ARG;
+    memcpy(out->name, r->cols[col].name, 32);
+    out->type = r->cols[col].type;
+    out->scale = r->cols[col].scale;
+    return CSTRIPE_OK;
+}
+
+/* ============================ pruning / scan ============================ */
+
+/* range refutation of one predicate against a chunk's [min,max] — the
+ * BuildBaseConstraint/UpdateConstraint/predicate_refuted_by combination of
+ * SelectedChunkMask (columnar_reader.c:1132-1187) specialized to the
+ * pushdownable family. Returns true if NO row in [min,max] can satisfy. */
+static bool pred_refutes(const cstripe_pred &p, uint8_t type, int64_t min_i, int64_t max_i)
+{
+    if (type == CSTRIPE_F32 || type == CSTRIPE_F64) {
+        double mn, mx;
+        memcpy(&mn, &min_i, 8);
+        memcpy(&mx, &max_i, 8);
+        double c = p.fval;
+        switch (p.op) {
+            case CSTRIPE_PRED_LT: return mn >= c;
+            case CSTRIPE_PRED_LE: return mn > c;
+            case CSTRIPE_PRED_GT: return mx <= c;
+            case CSTRIPE_PRED_GE: return mx < c;
+            case CSTRIPE_PRED_EQ: return c < mn || c > mx;
+            case CSTRIPE_PRED_NE: return mn == c && mx == c;
+        }
+        return false;
+    }
+    int64_t c = p.ival;
+    switch (p.op) {
+        case CSTRIPE_PRED_LT: return min_i >= c;
+        case CSTRIPE_PRED_LE: return min_i > c;
+        case CSTRIPE_PRED_GT: return max_i <= c;
+        case CSTRIPE_PRED_GE: return max_i < c;
+        case CSTRIPE_PRED_EQ: return c < min_i || c > max_i;
+        case CSTRIPE_PRED_NE: return min_i == c && max_i == c;
+    }
+    return false;
+}
+
+extern "C" cstripe_scan *cstripe_scan_begin(cstripe_reader *r, uint64_t cols_mask,
+                                            const cstripe_pred *preds, uint32_t n_preds)
+{
+    if (!r) { cs_set_err("scan_begin: null reader"); return nullptr; }
+    auto *s = new cstripe_scan();
+    s->r = r;
+    s->cols_mask = cols_mask;
+    for (uint32_t i = 0; i < n_preds; i++) {
+        if (preds[i].column >= r->cols.size()) { cs_set_err("pred column out of range"); delete s; return nullptr; }
+        s->preds.push_back(preds[i]);
+        s->cols_mask |= 1ull << preds[i].column;   /* pred columns must be read */
+    }
+    for (uint32_t si = 0; si < r->stripes.size(); si++) {
+        const cs_stripe_info &st = r->stripes[si];
+        for (uint32_t k = 0; k < st.meta.chunk_count; k++) {
+            bool selected = true;
+            for (const auto &p : s->preds) {
+                const csf_skipnode &nd = st.nodes[p.column][k].n;
+                /* all-NULL chunks have no min/max and are never refuted
+                 * (columnar_reader.c:1160-1166) */
+                if (!nd.has_min_max) continue;
+                if (pred_refutes(p, r->cols[p.column].type, nd.min_i, nd.max_i)) {
+                    selected = false;
+                    break;      /* count once (:1178-1182) */
+                }
+            }
+            if (selected) s->sel.push_back({si, k});
+            else s->chunk_groups_filtered++;
+        }
+    }
+    return s;
+}
+
+extern "C" void cstripe_scan_end(cstripe_scan *s)
+{
+    if (!s) return;
+    csgpu_release(s);
+    delete s;
+}
+
+extern "C" int64_t cstripe_scan_chunk_groups_filtered(const cstripe_scan *s)
+{
+    return s ? s->chunk_groups_filtered : 0;
+}
+
+extern "C" int cstripe_scan_rewind(cstripe_scan *s)
+{
+    if (!s) return CSTRIPE_ERR_ARG;
+    s->batch_pos = 0;
+    return CSTRIPE_OK;
+}
+
+extern "C" double cstripe_scan_last_kernel_ms(const cstripe_scan *s) { return s ? s->last_kernel_ms : 0; }
+extern "C" double cstripe_scan_last_decode_kernel_ms(const cstripe_scan *s) { return s ? s->last_decode_ms : 0; }
+extern "C" double cstripe_scan_last_agg_kernel_ms(const cstripe_scan *s) { return s ? s->last_agg_ms : 0; }
+
+extern "C" int cstripe_gpu_stage(cstripe_scan *s, int device_id)
+{
+    if (!s) return CSTRIPE_ERR_ARG;
+    return csgpu_stage(s, device_id);
+}
+
+extern "C" uint64_t cstripe_gpu_staged_bytes(const cstripe_scan *s)
+{
+    return s ? csgpu_staged_bytes(s) : 0;
+}
+
+extern "C" int cstripe_scan_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
+                                cstripe_partial *out)
+{
+    if (!s || !aggs || n_aggs == 0 || !out) { cs_set_err("scan_agg: bad args"); return CSTRIPE_ERR_ARG; }
+    return csgpu_agg(s, aggs, n_aggs, nullptr, 0, nullptr, out);
+}
+
+extern "C" int cstripe_scan_agg_grouped(cstripe_scan *s, const cstripe_agg_spec *aggs,
+                                        uint32_t n_aggs, const uint32_t *group_cols,
+                                        uint32_t n_group_cols, cstripe_group_result *gr,
+                                        cstripe_partial *out)
+{
+    if (!s || !aggs || n_aggs == 0 || !out || !gr || n_group_cols == 0 ||
+        n_group_cols > CSTRIPE_MAX_GROUP_COLS) { cs_set_err("scan_agg_grouped: bad args"); return CSTRIPE_ERR_ARG; }
+    return csgpu_agg(s, aggs, n_aggs, group_cols, n_group_cols, gr, out);
+}
+
+extern "C" int cstripe_scan_next_batch(cstripe_scan *s, cstripe_batch *batch)
+{
+    if (!s || !batch) return CSTRIPE_ERR_ARG;
+    return csgpu_next_batch(s, batch);
+}
+
+/* ============================ combine ============================ */
+
+static bool agg_is_count(uint32_t k)
+{
+    return k == CSTRIPE_AGG_COUNT_STAR || k == CSTRIPE_AGG_COUNT_COL;
+}
+
+static bool agg_is_i128_sum(uint32_t k)
+{
+    return k == CSTRIPE_AGG_SUM_I64 || k == CSTRIPE_AGG_SUM_PROD_I64 ||
+           k == CSTRIPE_AGG_SUM_DISC_I64 || k == CSTRIPE_AGG_SUM_DISC_TAX_I64;
+}
+
+extern "C" int cagg_combine(const cstripe_agg_spec *aggs, uint32_t n_aggs,
+                            const cstripe_partial *parts, uint32_t n_parts,
+                            cstripe_partial *out)
+{
+    if (!aggs || !parts || !out || n_aggs == 0) { cs_set_err("combine: bad args"); return CSTRIPE_ERR_ARG; }
+    for (uint32_t a = 0; a < n_aggs; a++) {
+        cstripe_partial acc{};
+        acc.is_null = 1;
+        for (uint32_t p = 0; p < n_parts; p++) {
+            const cstripe_partial &in = parts[p * n_aggs + a];
+            if (in.is_null) continue;    /* strict combine skips NULL partials
+                                          * (aggregate_utils.c:976-1000) */
+            if (acc.is_null) { acc = in; continue; }
+            switch (aggs[a].kind) {
+                case CSTRIPE_AGG_COUNT_STAR:
+                case CSTRIPE_AGG_COUNT_COL:
+                    acc.count += in.count;
+                    break;
+                case CSTRIPE_AGG_SUM_I64:
+                case CSTRIPE_AGG_SUM_PROD_I64:
+                case CSTRIPE_AGG_SUM_DISC_I64:
+                case CSTRIPE_AGG_SUM_DISC_TAX_I64: {
+                    __int128 x = ((__int128)acc.i128_hi << 64) | (unsigned long long)acc.i128_lo;
+                    __int128 y = ((__int128)in.i128_hi << 64) | (unsigned long long)in.i128_lo;
+                    x += y;
+                    acc.i128_lo = (int64_t)(uint64_t)x;
+                    acc.i128_hi = (int64_t)(x >> 64);
+                    acc.count += in.count;
+                    break;
+                }
+                case CSTRIPE_AGG_SUM_F64:
+                    acc.f64 += in.f64;
+                    acc.count += in.count;
+                    break;
+                case CSTRIPE_AGG_MIN_I64:
+                    acc.i128_lo = std::min(acc.i128_lo, in.i128_lo);
+                    acc.count += in.count;
+                    break;
+                case CSTRIPE_AGG_MAX_I64:
+                    acc.i128_lo = std::max(acc.i128_lo, in.i128_lo);
+                    acc.count += in.count;
+                    break;
+                case CSTRIPE_AGG_MIN_F64:
+                    acc.f64 = std::min(acc.f64, in.f64);
+                    acc.count += in.count;
+                    break;
+                case CSTRIPE_AGG_MAX_F64:
+                    acc.f64 = std::max(acc.f64, in.f64);
+                    acc.count += in.count;
+                    break;
+                default:
+                    cs_set_err("combine: bad agg kind %u", aggs[a].kind);
+                    return CSTRIPE_ERR_ARG;
+            }
+        }
+        /* COUNT: NULL -> 0 (COALESCE, multi_logical_optimizer.c:1874-1884) */
+        if (acc.is_null && agg_is_count(aggs[a].kind)) {
+            acc.is_null = 0;
+            acc.count = 0;
+        }
+        (void)agg_is_i128_sum;
+        out[a] = acc;
+    }
+    return CSTRIPE_OK;
+}

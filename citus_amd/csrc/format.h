@@ -1,0 +1,130 @@
+/*
+ * format.h — on-disk stripe format, shared by writer and reader host code.
+ *
+ * Data layout follows the reference byte semantics:
+ *  - per stripe, per column: all exists streams (chunk 0..n-1), then all
+ *    value streams (chunk 0..n-1); offsets relative to the stripe's data
+ *    start (FlushStripe, columnar_writer.c:391-516, esp. :425-458).
+ *  - exists stream: bit-packed, bit i = row i present, LSB-first within a
+ *    byte (SerializeBoolArray, columnar_writer.c:523-545); never compressed
+ *    (SerializeChunkData :606-614 compresses only value buffers).
+ *  - value stream: attlen-sized, att_align_nominal-aligned packed values of
+ *    PRESENT rows only (SerializeSingleDatum, columnar_writer.c:555-585;
+ *    DeserializeDatumArray, columnar_reader.c:1542-1572); for the fixed-width
+ *    types supported here, a dense array of the non-null values.
+ *  - value stream compression: whole-chunk LZ4/ZSTD (columnar_compression.c:
+ *    62-158), except that a chunk MAY be stored as N independently decodable
+ *    segments (each its own LZ4/ZSTD block over a slice of the decompressed
+ *    stream, concatenated) with a segment table in the footer. N=1 is byte-
+ *    compatible with the reference's whole-chunk block. Compressed-byte
+ *    parity is unpinned by the reference's own tests;
+ *    decompressed-content parity is what is pinned and checked.
+ *
+ * The footer replaces the columnar.stripe / chunk_group / chunk catalogs
+ * (columnar_metadata.c:604-832; schema sql/columnar--9.5-1--10.0-1.sql:18-63)
+ * with a flat directory, per SURVEY.md §2.
+ *
+ * All integers little-endian. This is a private format of this framework;
+ * the drop-in boundary is include/cstripe.h, not these bytes.
+ */
+#ifndef CSTRIPE_FORMAT_H
+#define CSTRIPE_FORMAT_H
+
+#include <stdint.h>
+
+#define CSF_MAGIC      "CSTRIPE1"      /* 8 bytes, file offset 0 */
+#define CSF_FOOT_MAGIC "CSTRFOOT"      /* 8 bytes, end of file   */
+#define CSF_HEADER_SIZE 16             /* magic + u32 version + u32 pad */
+#define CSF_VERSION     1
+
+/* footer tail: [u64 footer_offset][CSF_FOOT_MAGIC] — last 16 bytes */
+
+/* ---- fixed-size footer records (packed, written/read as structs) ---- */
+
+#pragma pack(push, 1)
+
+typedef struct csf_footer_head {
+    uint32_t version;
+    uint32_t column_count;
+    uint32_t stripe_count;
+    uint32_t chunk_row_limit;      /* columnar.chunk_group_row_limit default 10000 */
+    uint64_t stripe_row_limit;     /* columnar.stripe_row_limit default 150000 */
+    uint64_t total_rows;
+    uint8_t  compression;          /* requested codec (per-chunk actual in skip node) */
+    int8_t   compression_level;
+    uint16_t lz4_seg_target_kb;
+    uint32_t reserved;
+} csf_footer_head;
+
+typedef struct csf_coldef {
+    char     name[32];
+    uint8_t  type;                 /* cstripe_type */
+    uint8_t  scale;
+    uint8_t  pad[6];
+} csf_coldef;
+
+typedef struct csf_stripe_meta {
+    uint64_t file_offset;          /* absolute offset of stripe data start */
+    uint64_t data_size;
+    uint64_t first_row_number;
+    uint64_t row_count;
+    uint32_t chunk_count;
+    uint32_t reserved;
+} csf_stripe_meta;
+
+/* mirror of ColumnChunkSkipNode (columnar.h:85-111) + our extensions
+ * (n_present; segment table follows this record when n_segs > 1) */
+typedef struct csf_skipnode {
+    int64_t  min_i;                /* min/max in physical encoding; f64 via bit */
+    int64_t  max_i;                /*   pattern (compare as double for F32/F64) */
+    uint64_t row_count;            /* rows in chunk incl. nulls */
+    uint64_t value_off;            /* relative to stripe data start */
+    uint64_t value_len;            /* compressed length (sum of segments) */
+    uint64_t exists_off;
+    uint64_t exists_len;
+    uint64_t decompressed_size;    /* raw value stream bytes */
+    uint32_t n_present;            /* non-null rows (popcount of exists) */
+    uint8_t  has_min_max;
+    uint8_t  comp_type;            /* cstripe_compression actually applied */
+    int8_t   comp_level;
+    uint8_t  reserved;
+    uint16_t n_segs;               /* >=1; seg table follows iff n_segs >= 1 */
+    uint16_t reserved2;
+} csf_skipnode;
+
+/* one independently decodable compressed segment of a chunk's value stream */
+typedef struct csf_seg {
+    uint32_t comp_off;             /* relative to chunk's value_off */
+    uint32_t comp_len;
+    uint32_t decomp_off;           /* relative to chunk's decompressed stream */
+    uint32_t decomp_len;
+} csf_seg;
+
+#pragma pack(pop)
+
+/* footer layout, starting at footer_offset:
+ *   csf_footer_head
+ *   csf_coldef[column_count]
+ *   for each stripe:
+ *     csf_stripe_meta
+ *     uint32 chunk_group_row_counts[chunk_count]
+ *     for each column (0..column_count-1):
+ *       for each chunk (0..chunk_count-1):
+ *         csf_skipnode
+ *         csf_seg[n_segs]          (always present, n_segs entries)
+ */
+
+static inline uint32_t csf_type_width(uint8_t t)
+{
+    switch (t) {
+        case 1: return 1;          /* I8  */
+        case 2: return 2;          /* I16 */
+        case 3: return 4;          /* I32 */
+        case 4: return 8;          /* I64 */
+        case 5: return 4;          /* F32 */
+        case 6: return 8;          /* F64 */
+        default: return 0;
+    }
+}
+
+#endif /* CSTRIPE_FORMAT_H */

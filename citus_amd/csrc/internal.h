@@ -1,0 +1,67 @@
+/*
+ * internal.h — shared host-side structures between cstripe_host.cpp (writer,
+ * reader, pruning, combine) and cstripe_gpu.hip (staging + kernels).
+ */
+#ifndef CSTRIPE_INTERNAL_H
+#define CSTRIPE_INTERNAL_H
+
+#include <cstdint>
+#include <cstddef>
+#include <vector>
+#include <string>
+
+#include "../../include/cstripe.h"
+#include "format.h"
+
+void cs_set_err(const char *fmt, ...);
+
+struct cs_skipnode {
+    csf_skipnode n;
+    std::vector<csf_seg> segs;
+};
+
+struct cs_stripe_info {
+    csf_stripe_meta meta;
+    std::vector<uint32_t> group_rows;            /* [chunk] */
+    std::vector<std::vector<cs_skipnode>> nodes; /* [col][chunk] */
+};
+
+struct cstripe_reader {
+    int fd = -1;
+    const uint8_t *map = nullptr;
+    size_t map_size = 0;
+    csf_footer_head head{};
+    std::vector<csf_coldef> cols;
+    std::vector<cs_stripe_info> stripes;
+};
+
+struct cs_selchunk {
+    uint32_t stripe;
+    uint32_t chunk;
+};
+
+struct cs_gpu_state;   /* defined in cstripe_gpu.hip */
+
+struct cstripe_scan {
+    cstripe_reader *r = nullptr;
+    uint64_t cols_mask = 0;
+    std::vector<cstripe_pred> preds;
+    std::vector<cs_selchunk> sel;     /* surviving chunk groups, scan order */
+    int64_t chunk_groups_filtered = 0;
+    cs_gpu_state *gpu = nullptr;
+    size_t batch_pos = 0;             /* next_batch cursor into sel */
+    double last_kernel_ms = 0.0;
+    double last_decode_ms = 0.0;
+    double last_agg_ms = 0.0;
+};
+
+/* implemented in cstripe_gpu.hip */
+int  csgpu_stage(cstripe_scan *s, int device_id);
+void csgpu_release(cstripe_scan *s);
+uint64_t csgpu_staged_bytes(const cstripe_scan *s);
+int  csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
+               const uint32_t *group_cols, uint32_t n_group_cols,
+               cstripe_group_result *gr, cstripe_partial *out);
+int  csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch);
+
+#endif
