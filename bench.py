@@ -1,0 +1,239 @@
+#!/usr/bin/env python3
+"""
+bench.py — the BASELINE.json headline: TPC-H Q6 filter+SUM over synthetic
+lineitem-shaped LZ4 columnar stripes, rows/s + achieved HBM GB/s on MI355X.
+
+A step = one pass of the hot path over the staged stripes:
+  host chunk-group pruning -> LZ4 segment decode kernel -> fused
+  filter+partial-aggregate kernel -> device-wide reduce -> combine
+  (all_gather over RCCL at N>1 — the coordinator-merge replacement).
+Data is device-resident when the timed region starts (staged HtoD untimed);
+decode+filter+aggregate re-run every step — no work is skipped.
+
+N=1 workload is BASELINE config 2 (the largest single-GPU config): 100M-row
+lineitem, lz4, stripe 150k / chunk group 10k (reference defaults). N>1:
+per-rank shards of the same size (weak scaling; config-3 shape).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--rows R]
+Multi-GPU via: python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+               --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+ALG_BYTES_PER_ROW = 32.5   # 4 projected int64 cols + 4 exists bits
+HBM_PEAK_GBPS = 8000.0     # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def build_if_needed():
+    if not os.path.exists(os.path.join(REPO, "citus_amd", "libcstripe.so")):
+        import subprocess
+        subprocess.check_call(["make", "-C", os.path.join(REPO, "citus_amd", "csrc")])
+    if not os.path.exists(os.path.join(REPO, "oracle", "liboracle.so")):
+        import subprocess
+        subprocess.check_call(["make", "-C", os.path.join(REPO, "oracle")])
+
+
+def q6_preds(ca):
+    # l_shipdate >= 1994-01-01 (day 8766) AND < 1995-01-01 (9131)
+    # AND l_discount BETWEEN 0.05 AND 0.07 AND l_quantity < 24
+    return [(5, ca.PRED_GE, 8766), (5, ca.PRED_LT, 9131),
+            (3, ca.PRED_GE, 5), (3, ca.PRED_LE, 7), (1, ca.PRED_LT, 2400)]
+
+
+def q6_aggs(ca):
+    return [(ca.AGG_SUM_PROD_I64, 2, 3), (ca.AGG_COUNT_STAR, -1)]
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--rows", type=int, default=100_000_000,
+                    help="rows per GPU (config 2: 100M)")
+    ap.add_argument("--compression", default="lz4", choices=["lz4", "none", "zstd"])
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    build_if_needed()
+    import citus_amd as ca
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    dist = world > 1
+
+    import torch
+    if dist:
+        import torch.distributed as td
+        from citus_amd.dist import all_gather_combine
+        torch.cuda.set_device(local_rank)
+        td.init_process_group("nccl")
+
+    comp = {"lz4": ca.COMP_LZ4, "none": ca.COMP_NONE, "zstd": ca.COMP_ZSTD}[args.compression]
+
+    # ---- setup (untimed): generate per-rank shard, open, prune, stage ----
+    cache = os.environ.get("CSTRIPE_BENCH_DIR", "/tmp/cstripe_bench")
+    os.makedirs(cache, exist_ok=True)
+    shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_r{rank}.cs")
+    t0 = time.time()
+    if not os.path.exists(shard):
+        ca.gen_lineitem(shard, args.rows, seed=42 + rank, compression=comp)
+    gen_s = time.time() - t0
+
+    reader = ca.Reader(shard)
+    scan = reader.scan(preds=q6_preds(ca))
+    t0 = time.time()
+    scan.stage(local_rank if dist else -1)
+    stage_s = time.time() - t0
+    staged_gb = scan.staged_bytes / 1e9
+    aggs = q6_aggs(ca)
+
+    def step():
+        parts = scan.agg(aggs)
+        if dist:
+            return all_gather_combine(aggs, parts, device="cuda")
+        return ca.combine(aggs, [parts])
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        result = step()
+
+    decode_ms = []
+    agg_ms = []
+
+    if dist:
+        import torch.distributed as td
+        td.barrier()
+    torch.cuda.synchronize()
+    t_start = time.time()
+    for _ in range(args.steps):
+        result = step()
+        decode_ms.append(scan.last_decode_ms)
+        agg_ms.append(scan.last_agg_ms)
+    if dist:
+        td.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.time() - t_start
+
+    if dist:
+        t = torch.tensor([elapsed], device="cuda")
+        td.all_reduce(t, op=td.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank != 0:
+        scan.end()
+        reader.close()
+        if dist:
+            td.destroy_process_group()
+        return
+
+    total_rows = args.rows * world
+    value = total_rows * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    avg_decode = sum(decode_ms) / len(decode_ms)
+    avg_agg = sum(agg_ms) / len(agg_ms)
+    dominant = "lz4_decode_kernel" if avg_decode >= avg_agg else "filter_agg_kernel"
+    dominant_ms = max(avg_decode, avg_agg)
+    alg_bytes = args.rows * ALG_BYTES_PER_ROW          # per launch (this rank)
+    achieved_gbps = alg_bytes / (dominant_ms / 1e3) / 1e9 if dominant_ms > 0 else None
+
+    # PMC-measured HBM traffic for the dominant kernel, if a committed profile
+    # exists (profiles/traffic_r*.json, written from rocprofv3 --pmc runs)
+    traffic = None
+    try:
+        import glob
+        profs = sorted(glob.glob(os.path.join(REPO, "profiles", "traffic_r*.json")))
+        if profs:
+            tj = json.load(open(profs[-1]))
+            if tj.get("workload_rows") == args.rows and tj.get("kernel") == dominant:
+                traffic = tj.get("hbm_bytes_per_launch")
+    except Exception:
+        traffic = None
+
+    # ---- CPU baseline (oracle restatement, rank 0, N=1 only) ----
+    cpu_baseline = None
+    parity = None
+    if world == 1 and not args.no_cpu_baseline:
+        import oracle
+        sample_rows = args.rows
+        sample_path = shard
+        if args.rows > 150_000_000:
+            sample_rows = 100_000_000
+            sample_path = os.path.join(cache, f"li_{sample_rows}_{args.compression}_r0.cs")
+            if not os.path.exists(sample_path):
+                ca.gen_lineitem(sample_path, sample_rows, seed=42, compression=comp)
+        with oracle.OracleTable(sample_path) as t:
+            t0 = time.time()
+            cpu_parts, _ = t.scan_agg(q6_preds(ca), aggs)
+            cpu_s = time.time() - t0
+        cpu_baseline = {
+            "value": sample_rows / cpu_s,
+            "unit": "rows/s",
+            "cores": 1,
+            "kind": "port",
+            "sample": f"full {sample_rows / 1e6:.0f}M-row Q6 pass, single thread "
+                      f"({cpu_s:.1f}s)",
+        }
+        if sample_path == shard:
+            parity = ("bit-exact" if (cpu_parts[0].i128 == result[0].i128 and
+                                      cpu_parts[1].count == result[1].count)
+                      else "MISMATCH")
+
+    out = {
+        "metric": "columnar_rows_per_s_q6",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": ms_per_step,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int64",
+        "data": "synthetic",
+        "config": {
+            "workload": f"q6_lineitem_{args.rows // 1_000_000}m_{args.compression}",
+            "query": "tpch_q6",
+            "rows_per_gpu": args.rows,
+            "compression": args.compression,
+            "stripe_rows": 150000,
+            "chunk_group_rows": 10000,
+            "parallelism": f"shard-dp{world}",
+            "staged_gb_compressed": round(staged_gb, 3),
+            "gen_s": round(gen_s, 1),
+            "stage_s": round(stage_s, 2),
+            "kernel_ms": {"decode": round(avg_decode, 3), "filter_agg": round(avg_agg, 3)},
+            "parity_vs_oracle": parity,
+            "q6_revenue_scale4": result[0].i128 if world == 1 else None,
+        },
+        "roofline": {
+            "bound": "hbm",
+            "achieved": round(achieved_gbps, 1) if achieved_gbps else None,
+            "peak": HBM_PEAK_GBPS,
+            "unit": "GB/s",
+            "frac": round(achieved_gbps / HBM_PEAK_GBPS, 4) if achieved_gbps else None,
+            "traffic": traffic,
+            "kernel": dominant,
+        },
+        "cpu_baseline": cpu_baseline,
+    }
+    print(json.dumps(out))
+
+    scan.end()
+    reader.close()
+    if dist:
+        td.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -61,6 +61,7 @@ class Partial(C.Structure):
 
     @property
     def i128(self):
+        """signed 128-bit view (sums; min/max are sign-extended into hi)"""
         return (self.i128_hi << 64) | (self.i128_lo & ((1 << 64) - 1))
 
     def as_dict(self):

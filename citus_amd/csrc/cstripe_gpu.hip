@@ -764,6 +764,11 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             case CSTRIPE_AGG_MAX_F64:
                 o.f64 = c.f;
                 break;
+            case CSTRIPE_AGG_MIN_I64:
+            case CSTRIPE_AGG_MAX_I64:
+                o.i128_lo = c.lo;
+                o.i128_hi = c.lo < 0 ? -1 : 0;
+                break;
             default:
                 o.i128_lo = c.lo;
                 o.i128_hi = c.hi;

@@ -792,10 +792,12 @@ extern "C" int cagg_combine(const cstripe_agg_spec *aggs, uint32_t n_aggs,
                     break;
                 case CSTRIPE_AGG_MIN_I64:
                     acc.i128_lo = std::min(acc.i128_lo, in.i128_lo);
+                    acc.i128_hi = acc.i128_lo < 0 ? -1 : 0;
                     acc.count += in.count;
                     break;
                 case CSTRIPE_AGG_MAX_I64:
                     acc.i128_lo = std::max(acc.i128_lo, in.i128_lo);
+                    acc.i128_hi = acc.i128_lo < 0 ? -1 : 0;
                     acc.count += in.count;
                     break;
                 case CSTRIPE_AGG_MIN_F64:

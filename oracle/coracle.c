@@ -309,6 +309,7 @@ static void oacc_to_partial(const oacc *a, uint32_t kind, cstripe_partial *o)
         case CSTRIPE_AGG_MIN_I64:
         case CSTRIPE_AGG_MAX_I64:
             o->i128_lo = a->minmax_i;
+            o->i128_hi = a->minmax_i < 0 ? -1 : 0;
             break;
         case CSTRIPE_AGG_MIN_F64:
         case CSTRIPE_AGG_MAX_F64:

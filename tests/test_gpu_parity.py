@@ -1,0 +1,208 @@
+"""
+GPU parity tests (@pytest.mark.gpu — run on a real MI355X): the HIP path
+(LZ4 decode kernel + fused filter/aggregate kernels), called through the
+C ABI, against the CPU oracle on identical inputs.
+
+Bar (BASELINE.json): bit-exact COUNT and integer/fixed-point SUM/MIN/MAX;
+<= 1e-6 relative for float SUM (order of f64 additions differs).
+"""
+import os
+
+import numpy as np
+import pytest
+
+import citus_amd as ca
+import oracle
+
+from conftest import q6_preds
+
+pytestmark = pytest.mark.gpu
+
+RNG = np.random.default_rng(7)
+
+
+@pytest.fixture(scope="module", autouse=True)
+def require_gpu():
+    assert ca.gpu_available(), "MI355X required for these tests"
+
+
+def both(path, preds, aggs):
+    with oracle.OracleTable(path) as t:
+        op, ofilt = t.scan_agg(preds, aggs)
+    with ca.Reader(path) as r, r.scan(preds=preds) as s:
+        s.stage()
+        gp = s.agg(aggs)
+        gfilt = s.chunk_groups_filtered
+    return op, ofilt, gp, gfilt
+
+
+def assert_parity(op, gp, aggs):
+    for i, a in enumerate(aggs):
+        kind = a[0]
+        assert op[i].is_null == gp[i].is_null, f"agg {i} null mismatch"
+        assert op[i].count == gp[i].count, f"agg {i} count mismatch"
+        if kind == ca.AGG_SUM_F64 or kind in (ca.AGG_MIN_F64, ca.AGG_MAX_F64):
+            if not op[i].is_null:
+                if op[i].f64 == 0:
+                    assert abs(gp[i].f64) < 1e-9
+                else:
+                    assert abs(gp[i].f64 - op[i].f64) <= 1e-6 * abs(op[i].f64)
+        else:
+            assert op[i].i128 == gp[i].i128, f"agg {i} i128 mismatch: {op[i].i128} vs {gp[i].i128}"
+
+
+@pytest.mark.parametrize("variant", ["lz4", "none", "zstd"])
+def test_q6_golden_fixture(golden_dir, expected, variant):
+    path = os.path.join(golden_dir, f"lineitem12k_{variant}.cs")
+    preds = q6_preds(ca, expected)
+    aggs = [(ca.AGG_SUM_PROD_I64, 2, 3), (ca.AGG_COUNT_STAR, -1)]
+    op, ofilt, gp, gfilt = both(path, preds, aggs)
+    assert gp[0].i128 == expected["q6"]["revenue_scale4"]
+    assert ofilt == gfilt
+    assert_parity(op, gp, aggs)
+
+
+@pytest.mark.parametrize("comp", [ca.COMP_NONE, ca.COMP_LZ4, ca.COMP_ZSTD])
+@pytest.mark.parametrize("n", [1, 999, 10000, 123457])
+def test_agg_kinds_random(tmp_path, comp, n):
+    a = RNG.integers(-10**6, 10**6, n).astype(np.int64)
+    b = RNG.integers(0, 10**4, n).astype(np.int64)
+    f = RNG.normal(size=n)
+    path = str(tmp_path / f"r{comp}_{n}.cs")
+    ca.write_table(path, [("a", ca.I64, 0), ("b", ca.I64, 0), ("f", ca.F64, 0)],
+                   [a, b, f], compression=comp,
+                   stripe_row_limit=50000, chunk_group_row_limit=5000)
+    preds = [(0, ca.PRED_GT, -500000), (0, ca.PRED_LE, 700000)]
+    aggs = [(ca.AGG_COUNT_STAR, -1), (ca.AGG_COUNT_COL, 1),
+            (ca.AGG_SUM_I64, 0), (ca.AGG_SUM_I64, 1),
+            (ca.AGG_MIN_I64, 0), (ca.AGG_MAX_I64, 0),
+            (ca.AGG_SUM_F64, 2), (ca.AGG_MIN_F64, 2), (ca.AGG_MAX_F64, 2),
+            (ca.AGG_SUM_PROD_I64, 0, 1)]
+    op, ofilt, gp, gfilt = both(path, preds, aggs)
+    assert ofilt == gfilt
+    assert_parity(op, gp, aggs)
+
+
+def test_nulls_parity(tmp_path):
+    n = 34567
+    a = RNG.integers(-1000, 1000, n).astype(np.int64)
+    na = (RNG.random(n) < 0.35).astype(np.uint8)
+    b = RNG.integers(0, 100, n).astype(np.int64)
+    nb = (RNG.random(n) < 0.05).astype(np.uint8)
+    path = str(tmp_path / "nulls.cs")
+    ca.write_table(path, [("a", ca.I64, 0), ("b", ca.I64, 0)], [a, b],
+                   nulls=[na, nb], compression=ca.COMP_LZ4,
+                   chunk_group_row_limit=3000)
+    preds = [(1, ca.PRED_GE, 10)]      # NULL b rows must fail the qual
+    aggs = [(ca.AGG_COUNT_STAR, -1), (ca.AGG_COUNT_COL, 0),
+            (ca.AGG_SUM_I64, 0), (ca.AGG_MIN_I64, 0), (ca.AGG_MAX_I64, 0),
+            (ca.AGG_SUM_PROD_I64, 0, 1)]
+    op, ofilt, gp, gfilt = both(path, preds, aggs)
+    assert_parity(op, gp, aggs)
+
+
+def test_all_null_column(tmp_path):
+    n = 5000
+    a = np.zeros(n, dtype=np.int64)
+    na = np.ones(n, dtype=np.uint8)
+    path = str(tmp_path / "allnull.cs")
+    ca.write_table(path, [("a", ca.I64, 0)], [a], nulls=[na],
+                   compression=ca.COMP_LZ4)
+    aggs = [(ca.AGG_COUNT_STAR, -1), (ca.AGG_COUNT_COL, 0), (ca.AGG_SUM_I64, 0)]
+    op, _, gp, _ = both(path, [], aggs)
+    assert gp[0].count == n
+    assert gp[1].count == 0
+    assert gp[2].is_null
+    assert_parity(op, gp, aggs)
+
+
+def test_pruning_parity(tmp_path):
+    """simple_chunk_filtering vector on device: count 111111, 12 pruned."""
+    n = 234568
+    arr = np.arange(n, dtype=np.int64)
+    path = str(tmp_path / "simple.cs")
+    ca.write_table(path, [("i", ca.I64, 0)], [arr],
+                   compression=ca.COMP_LZ4, chunk_group_row_limit=10000)
+    with ca.Reader(path) as r, r.scan(preds=[(0, ca.PRED_GT, 123456)]) as s:
+        s.stage()
+        gp = s.agg([(ca.AGG_COUNT_STAR, -1)])
+        assert gp[0].count == 111111
+        assert s.chunk_groups_filtered == 12
+
+
+def test_decode_parity_batches(tmp_path):
+    """GPU LZ4 decode byte-parity: next_batch row-aligned output equals the
+    oracle's DeserializeChunkData restatement for every chunk."""
+    n = 47000
+    a = RNG.integers(-2**40, 2**40, n).astype(np.int64)
+    na = (RNG.random(n) < 0.2).astype(np.uint8)
+    b = RNG.normal(size=n).astype(np.float32)
+    path = str(tmp_path / "dec.cs")
+    ca.write_table(path, [("a", ca.I64, 0), ("b", ca.F32, 0)], [a, b],
+                   nulls=[na, None], compression=ca.COMP_LZ4,
+                   stripe_row_limit=20000, chunk_group_row_limit=3000)
+
+    with ca.Reader(path) as r, r.scan(cols_mask=0b11) as s:
+        s.stage()
+        va = np.zeros(3000, dtype=np.int64)
+        vb = np.zeros(3000, dtype=np.float32)
+        ea = np.zeros(3000, dtype=np.uint8)
+        eb = np.zeros(3000, dtype=np.uint8)
+        ova = np.zeros(3000, dtype=np.int64)
+        ovb = np.zeros(3000, dtype=np.float32)
+        oea = np.zeros(3000, dtype=np.uint8)
+        oeb = np.zeros(3000, dtype=np.uint8)
+        with oracle.OracleTable(path) as t:
+            stripe = chunk = 0
+            rows_left_in_stripe = min(20000, n)
+            seen = 0
+            while True:
+                res = s.next_batch({0: va, 1: vb}, {0: ea, 1: eb})
+                if res is None:
+                    break
+                nr, first = res
+                t.read_chunk(stripe, chunk, 0, ova, oea)
+                t.read_chunk(stripe, chunk, 1, ovb, oeb)
+                np.testing.assert_array_equal(va[:nr], ova[:nr])
+                np.testing.assert_array_equal(vb[:nr], ovb[:nr])
+                np.testing.assert_array_equal(ea[:nr], 1 - oea[:nr])  # nulls vs exists
+                seen += nr
+                chunk += 1
+                rows_left_in_stripe -= nr
+                if rows_left_in_stripe == 0:
+                    stripe += 1
+                    chunk = 0
+                    rows_left_in_stripe = min(20000, n - seen)
+            assert seen == n
+
+
+def test_repeated_agg_calls_stable(golden_dir, expected):
+    """rescan semantics: re-invoking scan_agg re-runs decode+filter+agg and
+    returns identical results."""
+    path = os.path.join(golden_dir, "lineitem12k_lz4.cs")
+    preds = q6_preds(ca, expected)
+    aggs = [(ca.AGG_SUM_PROD_I64, 2, 3)]
+    with ca.Reader(path) as r, r.scan(preds=preds) as s:
+        s.stage()
+        r1 = s.agg(aggs)
+        r2 = s.agg(aggs)
+        assert r1[0].i128 == r2[0].i128 == expected["q6"]["revenue_scale4"]
+        assert s.last_kernel_ms > 0
+
+
+def test_q6_large_synthetic(tmp_path):
+    """config-1-shaped: 1M-row lineitem, count(*) WHERE l_quantity < 24 —
+    plus full Q6 aggregate; GPU vs oracle bit-exact."""
+    path = str(tmp_path / "li1m.cs")
+    ca.gen_lineitem(path, 1_000_000)
+    preds_c1 = [(1, ca.PRED_LT, 2400)]
+    aggs_c1 = [(ca.AGG_COUNT_STAR, -1)]
+    op, _, gp, _ = both(path, preds_c1, aggs_c1)
+    assert_parity(op, gp, aggs_c1)
+
+    preds_q6 = [(5, ca.PRED_GE, 8766), (5, ca.PRED_LT, 9131),
+                (3, ca.PRED_GE, 5), (3, ca.PRED_LE, 7), (1, ca.PRED_LT, 2400)]
+    aggs_q6 = [(ca.AGG_SUM_PROD_I64, 2, 3), (ca.AGG_COUNT_STAR, -1)]
+    op, ofilt, gp, gfilt = both(path, preds_q6, aggs_q6)
+    assert ofilt == gfilt
+    assert_parity(op, gp, aggs_q6)
