@@ -89,12 +89,12 @@ def main():
     gen_s = time.time() - t0
 
     reader = ca.Reader(shard)
-    scan = reader.scan(preds=q6_preds(ca))
+    aggs = q6_aggs(ca)
+    scan = reader.scan(cols_mask=ca.agg_cols_mask(aggs), preds=q6_preds(ca))
     t0 = time.time()
     scan.stage(local_rank if dist else -1)
     stage_s = time.time() - t0
     staged_gb = scan.staged_bytes / 1e9
-    aggs = q6_aggs(ca)
 
     def step():
         parts = scan.agg(aggs)

@@ -249,6 +249,18 @@ def make_preds(preds):
     return arr
 
 
+def agg_cols_mask(aggs):
+    """Projection bitmask covering every column the agg specs read — the
+    caller-side analog of ColumnarAttrNeeded (columnar_customscan.c:1813-1851):
+    projection is fixed at scan_begin, before aggs are bound."""
+    m = 0
+    for a in aggs:
+        for c in a[1:4]:
+            if isinstance(c, int) and c >= 0:
+                m |= 1 << c
+    return m
+
+
 def make_aggs(aggs):
     """aggs: list of (kind, col_a[, col_b[, col_c[, one]]])"""
     arr = (AggSpec * len(aggs))()

@@ -665,6 +665,8 @@ extern "C" cstripe_scan *cstripe_scan_begin(cstripe_reader *r, uint64_t cols_mas
         s->preds.push_back(preds[i]);
         s->cols_mask |= 1ull << preds[i].column;   /* pred columns must be read */
     }
+    if (s->cols_mask == 0)
+        s->cols_mask = 1;   /* pure count(*): still scan one column's chunks */
     for (uint32_t si = 0; si < r->stripes.size(); si++) {
         const cs_stripe_info &st = r->stripes[si];
         for (uint32_t k = 0; k < st.meta.chunk_count; k++) {

@@ -29,7 +29,7 @@ def require_gpu():
 def both(path, preds, aggs):
     with oracle.OracleTable(path) as t:
         op, ofilt = t.scan_agg(preds, aggs)
-    with ca.Reader(path) as r, r.scan(preds=preds) as s:
+    with ca.Reader(path) as r, r.scan(cols_mask=ca.agg_cols_mask(aggs), preds=preds) as s:
         s.stage()
         gp = s.agg(aggs)
         gfilt = s.chunk_groups_filtered
@@ -182,7 +182,7 @@ def test_repeated_agg_calls_stable(golden_dir, expected):
     path = os.path.join(golden_dir, "lineitem12k_lz4.cs")
     preds = q6_preds(ca, expected)
     aggs = [(ca.AGG_SUM_PROD_I64, 2, 3)]
-    with ca.Reader(path) as r, r.scan(preds=preds) as s:
+    with ca.Reader(path) as r, r.scan(cols_mask=ca.agg_cols_mask(aggs), preds=preds) as s:
         s.stage()
         r1 = s.agg(aggs)
         r2 = s.agg(aggs)
