@@ -112,7 +112,8 @@ def main():
     if dist:
         import torch.distributed as td
         td.barrier()
-    torch.cuda.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
     t_start = time.time()
     for _ in range(args.steps):
         result = step()
@@ -120,7 +121,8 @@ def main():
         agg_ms.append(scan.last_agg_ms)
     if dist:
         td.barrier()
-    torch.cuda.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
     elapsed = time.time() - t_start
 
     if dist:

@@ -752,6 +752,8 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         const AccCell &c = h_final[a];
         o.count = c.cnt;
         o.is_null = (c.cnt == 0) ? 1 : 0;
+        if (o.is_null && aggs[a].kind != CSTRIPE_AGG_COUNT_STAR &&
+            aggs[a].kind != CSTRIPE_AGG_COUNT_COL) { out[a] = o; continue; }
         switch (aggs[a].kind) {
             case CSTRIPE_AGG_COUNT_STAR:
             case CSTRIPE_AGG_COUNT_COL:

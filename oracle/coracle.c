@@ -297,6 +297,8 @@ static void oacc_to_partial(const oacc *a, uint32_t kind, cstripe_partial *o)
     memset(o, 0, sizeof(*o));
     o->count = a->cnt;
     o->is_null = (a->cnt == 0);
+    if (o->is_null && kind != CSTRIPE_AGG_COUNT_STAR && kind != CSTRIPE_AGG_COUNT_COL)
+        return;   /* NULL partial: value fields zeroed */
     switch (kind) {
         case CSTRIPE_AGG_COUNT_STAR:
         case CSTRIPE_AGG_COUNT_COL:
