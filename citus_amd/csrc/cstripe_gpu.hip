@@ -105,6 +105,8 @@ struct cs_gpu_state {
     uint64_t data_bytes = 0;
     uint64_t scratch_bytes = 0;
     uint32_t n_segs = 0;
+    uint32_t max_seg_comp = 0;
+    uint32_t max_seg_dlen = 0;
     uint32_t n_groups = 0;
     uint32_t n_proj = 0;
     uint32_t max_blocks = 0;
@@ -134,19 +136,23 @@ extern "C" int cstripe_gpu_available(void)
  * window via modulo, so one parallel pass is race-free.
  * Implements the LZ4 block format consumed by LZ4_decompress_safe
  * (columnar_compression.c:183); any valid block decodes, segmented or not.
+ *
+ * Two variants:
+ *  - lz4_decode_lds_kernel: compressed segment is first staged into LDS with
+ *    coalesced u32 loads and the OUTPUT is decoded in LDS too, so the
+ *    byte-serial parse and all match copies run at LDS latency instead of
+ *    chained dependent HBM round trips; the finished segment is flushed to
+ *    global with coalesced 16 B stores. Needs in+out <= 64 KiB of LDS —
+ *    the writer's default 8 KiB segments give ~9 resident waves/CU.
+ *  - lz4_decode_kernel: the global-memory fallback for oversized segments
+ *    (e.g. single-block whole-chunk compatibility mode).
  * ===================================================================== */
 
-__global__ __launch_bounds__(WAVE) void lz4_decode_kernel(
-    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
-    const SegDesc *__restrict__ segs, int *__restrict__ err)
+template <bool IN_LDS>
+__device__ inline void lz4_body(const uint8_t *__restrict__ src, uint32_t slen,
+                                uint8_t *__restrict__ dst, uint32_t dlen,
+                                uint32_t lane, int *__restrict__ err)
 {
-    const SegDesc s = segs[blockIdx.x];
-    const uint8_t *src = data + s.src_off;
-    uint8_t *dst = scratch + s.dst_off;
-    const uint32_t slen = s.comp_len;
-    const uint32_t dlen = s.decomp_len;
-    const uint32_t lane = threadIdx.x;
-
     uint32_t ip = 0, op = 0;
     while (ip < slen) {
         const uint32_t token = src[ip++];
@@ -184,6 +190,50 @@ __global__ __launch_bounds__(WAVE) void lz4_decode_kernel(
         op += mlen;
     }
     if (op != dlen) { if (lane == 0) atomicOr(err, 2); }
+}
+
+__global__ __launch_bounds__(WAVE) void lz4_decode_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, int *__restrict__ err)
+{
+    const SegDesc s = segs[blockIdx.x];
+    lz4_body<false>(data + s.src_off, s.comp_len, scratch + s.dst_off,
+                    s.decomp_len, threadIdx.x, err);
+}
+
+__global__ __launch_bounds__(WAVE) void lz4_decode_lds_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, uint32_t in_cap, int *__restrict__ err)
+{
+    extern __shared__ uint8_t sbuf[];      /* [in_cap compressed][output] */
+    const SegDesc s = segs[blockIdx.x];
+    const uint32_t lane = threadIdx.x;
+
+    /* stage compressed bytes: aligned u32 loads from (src_off & ~3) */
+    const uint64_t base = s.src_off & ~3ull;
+    const uint32_t shift = (uint32_t)(s.src_off - base);
+    const uint32_t words = (s.comp_len + shift + 3) >> 2;
+    const uint32_t *gsrc = (const uint32_t *)(data + base);
+    uint32_t *lin = (uint32_t *)sbuf;
+    for (uint32_t j = lane; j < words; j += WAVE) lin[j] = gsrc[j];
+    __syncthreads();
+
+    uint8_t *out = sbuf + in_cap;
+    lz4_body<true>(sbuf + shift, s.comp_len, out, s.decomp_len, lane, err);
+    __syncthreads();
+
+    /* coalesced flush LDS -> global */
+    uint8_t *dst = scratch + s.dst_off;
+    const uint32_t dlen = s.decomp_len;
+    if (((uintptr_t)dst & 15) == 0) {
+        uint32_t vec = dlen >> 4;
+        uint4 *d4 = (uint4 *)dst;
+        const uint4 *o4 = (const uint4 *)out;
+        for (uint32_t j = lane; j < vec; j += WAVE) d4[j] = o4[j];
+        for (uint32_t j = (vec << 4) + lane; j < dlen; j += WAVE) dst[j] = out[j];
+    } else {
+        for (uint32_t j = lane; j < dlen; j += WAVE) dst[j] = out[j];
+    }
 }
 
 /* =====================================================================
@@ -384,11 +434,15 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
 
     for (uint32_t row = row_start + threadIdx.x; row < row_end; row += AGG_BLOCK) {
         bool pass = true;
+        int last_proj = -1;
+        int64_t liv = 0; double lfv = 0; bool lok = false;
         for (uint32_t p = 0; p < params.n_preds; p++) {
             const PredD &pr = params.preds[p];
-            int64_t iv; double fv;
-            if (!col_value(data, scratch, rank, cols[pr.proj], row, iv, fv) ||
-                !pred_eval(pr, iv, fv)) { pass = false; break; }
+            if ((int)pr.proj != last_proj) {   /* BETWEEN reuses the load */
+                lok = col_value(data, scratch, rank, cols[pr.proj], row, liv, lfv);
+                last_proj = (int)pr.proj;
+            }
+            if (!lok || !pred_eval(pr, liv, lfv)) { pass = false; break; }
         }
         if (!pass) continue;
         for (uint32_t a = 0; a < params.n_aggs; a++)
@@ -618,6 +672,8 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                     sd.dst_off = spos + sg.decomp_off;
                     sd.comp_len = sg.comp_len;
                     sd.decomp_len = sg.decomp_len;
+                    if (sg.comp_len > g->max_seg_comp) g->max_seg_comp = sg.comp_len;
+                    if (sg.decomp_len > g->max_seg_dlen) g->max_seg_dlen = sg.decomp_len;
                     h_segs.push_back(sd);
                 }
                 dpos += align_up(nd.n.value_len, 16);
@@ -660,6 +716,24 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     HIP_TRY(hipMemcpyAsync(g->d_colloc, h_colloc.data(), h_colloc.size() * sizeof(ColLoc), hipMemcpyHostToDevice, g->stream));
     HIP_TRY(hipStreamSynchronize(g->stream));
     return CSTRIPE_OK;
+}
+
+/* launch the decode grid: LDS-staged variant when every segment fits the
+ * 64 KiB budget (writer default 8 KiB segments -> ~9 waves/CU), else the
+ * global-memory fallback */
+static void launch_decode(cs_gpu_state *g)
+{
+    if (g->n_segs == 0) return;
+    uint32_t in_cap = (g->max_seg_comp + 8 + 15) & ~15u;
+    uint32_t out_cap = (g->max_seg_dlen + 15) & ~15u;
+    if (in_cap + out_cap <= 64 * 1024) {
+        hipLaunchKernelGGL(lz4_decode_lds_kernel, dim3(g->n_segs), dim3(WAVE),
+                           in_cap + out_cap, g->stream,
+                           g->d_data, g->d_scratch, g->d_segs, in_cap, g->d_error);
+    } else {
+        hipLaunchKernelGGL(lz4_decode_kernel, dim3(g->n_segs), dim3(WAVE), 0, g->stream,
+                           g->d_data, g->d_scratch, g->d_segs, g->d_error);
+    }
 }
 
 /* =====================================================================
@@ -722,9 +796,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
 
     HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
     HIP_TRY(hipEventRecord(g->ev0, g->stream));
-    if (g->n_segs > 0)
-        hipLaunchKernelGGL(lz4_decode_kernel, dim3(g->n_segs), dim3(WAVE), 0, g->stream,
-                           g->d_data, g->d_scratch, g->d_segs, g->d_error);
+    launch_decode(g);
     HIP_TRY(hipEventRecord(g->ev1, g->stream));
     hipLaunchKernelGGL(filter_agg_kernel, dim3(n_blocks), dim3(AGG_BLOCK), 0, g->stream,
                        g->d_data, g->d_scratch, g->d_rank, g->d_groups, g->d_colloc,
@@ -798,8 +870,7 @@ int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
     /* make sure scratch holds decoded data (decode everything once per rewind) */
     if (g->n_segs > 0 && s->batch_pos == 0) {
         HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
-        hipLaunchKernelGGL(lz4_decode_kernel, dim3(g->n_segs), dim3(WAVE), 0, g->stream,
-                           g->d_data, g->d_scratch, g->d_segs, g->d_error);
+        launch_decode(g);
         int h_err = 0;
         HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
         HIP_TRY(hipStreamSynchronize(g->stream));

@@ -58,7 +58,7 @@ extern "C" void cstripe_default_options(cstripe_options *o)
     o->chunk_group_row_limit = 10000;
     o->compression = CSTRIPE_COMP_LZ4;
     o->compression_level = 3;
-    o->lz4_seg_target_kb = 16;
+    o->lz4_seg_target_kb = 8;
 }
 
 static uint32_t type_width(uint8_t t) { return csf_type_width(t); }
@@ -270,11 +270,13 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
         return;
     }
 
-    /* segment boundaries aligned to value width */
+    /* segment boundaries 16 B-aligned (pure byte split of the raw stream;
+     * lets the GPU flush decoded segments with aligned 16 B stores) */
+    (void)width;
     size_t n = raw.size();
     size_t nseg = (n + target - 1) / target;
     if (nseg < 1) nseg = 1;
-    size_t per = ((n / nseg) / width) * width;
+    size_t per = ((n / nseg) / 16) * 16;
     if (per == 0) { per = n; nseg = 1; }
 
     out.value_comp.clear();
@@ -283,7 +285,7 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     bool ok = true;
     while (off < n) {
         size_t len = std::min(per, n - off);
-        if (n - off - len < width) len = n - off;   /* absorb the tail */
+        if (n - off - len < 16) len = n - off;      /* absorb the tail */
         csf_seg s{};
         s.decomp_off = (uint32_t)off;
         s.decomp_len = (uint32_t)len;

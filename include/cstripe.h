@@ -105,7 +105,7 @@ typedef struct cstripe_options {
     int8_t      compression_level;     /* default 3 (zstd) */
     uint16_t    lz4_seg_target_kb;     /* decompressed bytes per independently
                                         * decodable LZ4 segment (parallel decode
-                                        * sidecar; 0 => default 16 KiB; one
+                                        * sidecar; 0 => default 8 KiB; one
                                         * segment == plain whole-chunk block,
                                         * exactly the reference layout) */
 } cstripe_options;
