@@ -42,7 +42,8 @@ class Options(C.Structure):
     _fields_ = [("stripe_row_limit", C.c_uint64),
                 ("chunk_group_row_limit", C.c_uint32),
                 ("compression", C.c_uint8), ("compression_level", C.c_int8),
-                ("lz4_seg_target_kb", C.c_uint16)]
+                ("lz4_seg_target_kb", C.c_uint16),
+                ("lz4_seg_target_bytes", C.c_uint32)]
 
 
 class Pred(C.Structure):
@@ -145,6 +146,8 @@ def _check(rc, what):
 def default_options(**kw):
     o = Options()
     _default_options(C.byref(o))
+    if "lz4_seg_target_kb" in kw and "lz4_seg_target_bytes" not in kw:
+        kw["lz4_seg_target_bytes"] = 0   # the coarse knob overrides the default
     for k, v in kw.items():
         setattr(o, k, v)
     return o

@@ -107,6 +107,7 @@ struct cs_gpu_state {
     uint32_t n_segs = 0;
     uint32_t max_seg_comp = 0;
     uint32_t max_seg_dlen = 0;
+    bool segs_16aligned = true;
     uint32_t n_groups = 0;
     uint32_t n_proj = 0;
     uint32_t max_blocks = 0;
@@ -199,6 +200,122 @@ __global__ __launch_bounds__(WAVE) void lz4_decode_kernel(
     const SegDesc s = segs[blockIdx.x];
     lz4_body<false>(data + s.src_off, s.comp_len, scratch + s.dst_off,
                     s.decomp_len, threadIdx.x, err);
+}
+
+/* =====================================================================
+ * Lane-parallel LZ4 decode — ONE LANE per micro-segment (writer default
+ * 512 B decompressed). Why: this data compresses to ~one sequence per
+ * 8-byte value (measured ~8 B/sequence), so a wave-cooperative decoder is
+ * serialized on the per-sequence parse chain and moves ~8 B per chain step.
+ * Here every wave advances 64 independent segment streams per step:
+ *  - input: per-lane 16-byte register window over the compressed stream,
+ *    refilled with aligned u64 loads (one dependent global load per 8
+ *    compressed bytes instead of per byte)
+ *  - output: per-lane LDS region (stride 528 = 16-byte aligned, 4-bank
+ *    skew per lane so equal-progress lanes hit different banks); match
+ *    copies are plain per-lane sequential byte moves (memmove-forward
+ *    semantics handles overlap for free)
+ *  - finish: block-cooperative 16 B coalesced flush LDS -> scratch
+ * ===================================================================== */
+
+struct ByteStream {
+    const uint8_t *base;    /* 8B-aligned */
+    uint64_t w0, w1;        /* current 16-byte window */
+    uint32_t woff;          /* byte offset of w0 within base */
+    uint32_t pos;           /* current byte position (>= initial shift) */
+};
+
+__device__ inline void bs_init(ByteStream &b, const uint8_t *data, uint64_t off)
+{
+    uint64_t a = off & ~7ull;
+    b.base = data + a;
+    b.pos = (uint32_t)(off - a);
+    b.woff = 0;
+    b.w0 = *(const uint64_t *)(b.base);
+    b.w1 = *(const uint64_t *)(b.base + 8);
+}
+
+__device__ inline uint8_t bs_get(ByteStream &b)
+{
+    uint32_t rel = b.pos - b.woff;
+    if (rel >= 8) {
+        b.w0 = b.w1;
+        b.woff += 8;
+        b.w1 = *(const uint64_t *)(b.base + b.woff + 8);
+        rel -= 8;
+    }
+    b.pos++;
+    return (uint8_t)(b.w0 >> (8 * rel));
+}
+
+__global__ void lz4_decode_lane_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, uint32_t n_segs, uint32_t stride,
+    int *__restrict__ err)
+{
+    extern __shared__ uint8_t sout_all[];
+    const uint32_t tid = threadIdx.x;
+    const uint32_t first = blockIdx.x * blockDim.x;
+    const uint32_t gseg = first + tid;
+    uint8_t *sout = sout_all + (size_t)tid * stride;
+
+    if (gseg < n_segs) {
+        const SegDesc s = segs[gseg];
+        ByteStream bs;
+        bs_init(bs, data, s.src_off);
+        const uint32_t send = bs.pos + s.comp_len;   /* end position */
+        const uint32_t dlen = s.decomp_len;
+        uint32_t op = 0;
+        bool bad = false;
+        while (bs.pos < send) {
+            const uint32_t token = bs_get(bs);
+            uint32_t litlen = token >> 4;
+            if (litlen == 15) {
+                uint32_t b;
+                do { b = bs_get(bs); litlen += b; } while (b == 255 && bs.pos < send);
+            }
+            if (bs.pos + litlen > send || op + litlen > dlen) { bad = true; break; }
+            for (uint32_t j = 0; j < litlen; j++) sout[op + j] = bs_get(bs);
+            op += litlen;
+            if (bs.pos >= send) break;       /* last sequence: literals only */
+
+            if (bs.pos + 2 > send) { bad = true; break; }
+            uint32_t offset = (uint32_t)bs_get(bs);
+            offset |= (uint32_t)bs_get(bs) << 8;
+            uint32_t mlen = token & 15;
+            if (mlen == 15) {
+                uint32_t b;
+                do { b = bs_get(bs); mlen += b; } while (b == 255 && bs.pos < send);
+            }
+            mlen += 4;
+            if (offset == 0 || offset > op || op + mlen > dlen) { bad = true; break; }
+            /* sequential per-lane copy: forward memmove handles overlap */
+            const uint8_t *msrc = sout + op - offset;
+            for (uint32_t j = 0; j < mlen; j++) sout[op + j] = msrc[j];
+            op += mlen;
+        }
+        if (bad || op != dlen) atomicOr(err, 4);
+    }
+    __syncthreads();
+
+    /* cooperative coalesced flush: flat element index over the block's
+     * segments, stride/16 u128 slots per segment */
+    const uint32_t eps = stride >> 4;                 /* elements per segment */
+    const uint32_t in_block = min(blockDim.x, n_segs - first);
+    const uint32_t total = in_block * eps;
+    for (uint32_t f = tid; f < total; f += blockDim.x) {
+        const uint32_t sidx = f / eps;
+        const uint32_t boff = (f % eps) << 4;
+        const SegDesc s = segs[first + sidx];
+        if (boff >= s.decomp_len) continue;
+        uint8_t *dst = scratch + s.dst_off + boff;
+        const uint8_t *ls = sout_all + (size_t)sidx * stride + boff;
+        if (boff + 16 <= s.decomp_len) {
+            *(uint4 *)dst = *(const uint4 *)ls;       /* dst 16B-aligned (writer) */
+        } else {
+            for (uint32_t j = 0; j < s.decomp_len - boff; j++) dst[j] = ls[j];
+        }
+    }
 }
 
 __global__ __launch_bounds__(WAVE) void lz4_decode_lds_kernel(
@@ -596,6 +713,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     HIP_TRY(hipEventCreate(&g->ev0));
     HIP_TRY(hipEventCreate(&g->ev1));
     HIP_TRY(hipEventCreate(&g->ev2));
+    data_bytes += 16;               /* register-window read slack */
     if (data_bytes) HIP_TRY(hipMalloc(&g->d_data, data_bytes));
     if (scratch_bytes) HIP_TRY(hipMalloc(&g->d_scratch, scratch_bytes));
     if (rank_words) HIP_TRY(hipMalloc(&g->d_rank, rank_words * 4));
@@ -674,6 +792,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                     sd.decomp_len = sg.decomp_len;
                     if (sg.comp_len > g->max_seg_comp) g->max_seg_comp = sg.comp_len;
                     if (sg.decomp_len > g->max_seg_dlen) g->max_seg_dlen = sg.decomp_len;
+                    if (sg.decomp_off % 16 != 0) g->segs_16aligned = false;
                     h_segs.push_back(sd);
                 }
                 dpos += align_up(nd.n.value_len, 16);
@@ -724,6 +843,17 @@ int csgpu_stage(cstripe_scan *s, int device_id)
 static void launch_decode(cs_gpu_state *g)
 {
     if (g->n_segs == 0) return;
+    /* lane-parallel path for micro-segments (one lane per segment) */
+    if (g->segs_16aligned && g->max_seg_dlen <= 1039) {
+        uint32_t block = (g->max_seg_dlen <= 527) ? 256 : 128;
+        uint32_t stride = (g->max_seg_dlen <= 527) ? 528 : 1040;
+        uint32_t grid = (g->n_segs + block - 1) / block;
+        hipLaunchKernelGGL(lz4_decode_lane_kernel, dim3(grid), dim3(block),
+                           block * stride, g->stream,
+                           g->d_data, g->d_scratch, g->d_segs, g->n_segs, stride,
+                           g->d_error);
+        return;
+    }
     uint32_t in_cap = (g->max_seg_comp + 8 + 15) & ~15u;
     uint32_t out_cap = (g->max_seg_dlen + 15) & ~15u;
     if (in_cap + out_cap <= 64 * 1024) {

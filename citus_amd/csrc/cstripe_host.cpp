@@ -58,7 +58,8 @@ extern "C" void cstripe_default_options(cstripe_options *o)
     o->chunk_group_row_limit = 10000;
     o->compression = CSTRIPE_COMP_LZ4;
     o->compression_level = 3;
-    o->lz4_seg_target_kb = 8;
+    o->lz4_seg_target_kb = 0;
+    o->lz4_seg_target_bytes = 512;
 }
 
 static uint32_t type_width(uint8_t t) { return csf_type_width(t); }
@@ -197,7 +198,8 @@ extern "C" cstripe_writer *cstripe_write_begin(const char *path, const cstripe_c
     w->f = f;
     w->path = path;
     if (opts) w->opts = *opts; else cstripe_default_options(&w->opts);
-    if (w->opts.lz4_seg_target_kb == 0) w->opts.lz4_seg_target_kb = 16;
+    if (w->opts.lz4_seg_target_bytes == 0 && w->opts.lz4_seg_target_kb == 0)
+        w->opts.lz4_seg_target_bytes = 512;
     for (uint32_t i = 0; i < n_cols; i++) {
         csf_coldef d{};
         memcpy(d.name, cols[i].name, sizeof(d.name));
@@ -256,7 +258,8 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     out.exists_packed = pc.exists_packed;
     const std::vector<uint8_t> &raw = pc.raw_values;
     uint8_t codec = w->opts.compression;
-    size_t target = (size_t)w->opts.lz4_seg_target_kb * 1024;
+    size_t target = w->opts.lz4_seg_target_bytes ? w->opts.lz4_seg_target_bytes
+                                                 : (size_t)w->opts.lz4_seg_target_kb * 1024;
     uint32_t width = type_width(w->cols[pc.col].type);
 
     if (raw.empty() || codec == CSTRIPE_COMP_NONE) {

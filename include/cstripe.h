@@ -103,11 +103,13 @@ typedef struct cstripe_options {
     uint32_t    chunk_group_row_limit; /* default 10000  */
     uint8_t     compression;           /* cstripe_compression, default LZ4 */
     int8_t      compression_level;     /* default 3 (zstd) */
-    uint16_t    lz4_seg_target_kb;     /* decompressed bytes per independently
-                                        * decodable LZ4 segment (parallel decode
-                                        * sidecar; 0 => default 8 KiB; one
-                                        * segment == plain whole-chunk block,
-                                        * exactly the reference layout) */
+    uint16_t    lz4_seg_target_kb;     /* decompressed KiB per independently
+                                        * decodable LZ4 segment (coarse knob;
+                                        * one segment == plain whole-chunk
+                                        * block, exactly the reference layout) */
+    uint32_t    lz4_seg_target_bytes;  /* fine knob, overrides kb when nonzero;
+                                        * default 512 B = one GPU lane per
+                                        * segment (lane-parallel decode) */
 } cstripe_options;
 
 void cstripe_default_options(cstripe_options *opts);
