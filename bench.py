@@ -59,6 +59,8 @@ def main():
     ap.add_argument("--rows", type=int, default=100_000_000,
                     help="rows per GPU (config 2: 100M)")
     ap.add_argument("--compression", default="lz4", choices=["lz4", "none", "zstd"])
+    ap.add_argument("--seg-bytes", type=int, default=0,
+                    help="LZ4 micro-segment size override (bytes)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -82,10 +84,11 @@ def main():
     # ---- setup (untimed): generate per-rank shard, open, prune, stage ----
     cache = os.environ.get("CSTRIPE_BENCH_DIR", "/tmp/cstripe_bench")
     os.makedirs(cache, exist_ok=True)
-    shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_r{rank}.cs")
+    shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}_r{rank}.cs")
     t0 = time.time()
     if not os.path.exists(shard):
-        ca.gen_lineitem(shard, args.rows, seed=42 + rank, compression=comp)
+        ca.gen_lineitem(shard, args.rows, seed=42 + rank, compression=comp,
+                        seg_bytes=args.seg_bytes)
     gen_s = time.time() - t0
 
     reader = ca.Reader(shard)

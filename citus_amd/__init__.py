@@ -188,7 +188,9 @@ def write_table(path, defs, columns, nulls=None, **opt_kw):
 
 
 def gen_lineitem(path, n_rows, seed=42, compression=COMP_LZ4, level=3, seg_kb=0,
-                 stripe_rows=0, chunk_rows=0):
+                 stripe_rows=0, chunk_rows=0, seg_bytes=0):
+    if seg_bytes:
+        seg_kb = -int(seg_bytes)      # negative seg_kb = bytes (csbench_gen_lineitem)
     _check(_gen_lineitem(path.encode(), n_rows, seed, compression, level, seg_kb,
                          stripe_rows, chunk_rows), "gen_lineitem")
 

@@ -43,6 +43,7 @@ extern "C" int csbench_gen_lineitem(const char *path, uint64_t n_rows, uint64_t 
     opts.compression = (uint8_t)compression;
     opts.compression_level = (int8_t)level;
     if (seg_kb > 0) { opts.lz4_seg_target_kb = (uint16_t)seg_kb; opts.lz4_seg_target_bytes = 0; }
+    else if (seg_kb < 0) { opts.lz4_seg_target_kb = 0; opts.lz4_seg_target_bytes = (uint32_t)(-seg_kb); }
 
     cstripe_writer *w = cstripe_write_begin(path, cols, 8, &opts);
     if (!w) return CSTRIPE_ERR;

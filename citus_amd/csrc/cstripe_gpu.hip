@@ -258,9 +258,14 @@ __global__ void lz4_decode_lane_kernel(
     const uint32_t first = blockIdx.x * blockDim.x;
     const uint32_t gseg = first + tid;
     uint8_t *sout = sout_all + (size_t)tid * stride;
+    /* per-segment (dst_off, decomp_len) cached in LDS for the flush loop */
+    uint64_t *sdst = (uint64_t *)(sout_all + (size_t)blockDim.x * stride);
+    uint32_t *sdlen = (uint32_t *)(sdst + blockDim.x);
 
     if (gseg < n_segs) {
         const SegDesc s = segs[gseg];
+        sdst[tid] = s.dst_off;
+        sdlen[tid] = s.decomp_len;
         ByteStream bs;
         bs_init(bs, data, s.src_off);
         const uint32_t send = bs.pos + s.comp_len;   /* end position */
@@ -306,14 +311,14 @@ __global__ void lz4_decode_lane_kernel(
     for (uint32_t f = tid; f < total; f += blockDim.x) {
         const uint32_t sidx = f / eps;
         const uint32_t boff = (f % eps) << 4;
-        const SegDesc s = segs[first + sidx];
-        if (boff >= s.decomp_len) continue;
-        uint8_t *dst = scratch + s.dst_off + boff;
+        const uint32_t dl = sdlen[sidx];
+        if (boff >= dl) continue;
+        uint8_t *dst = scratch + sdst[sidx] + boff;
         const uint8_t *ls = sout_all + (size_t)sidx * stride + boff;
-        if (boff + 16 <= s.decomp_len) {
+        if (boff + 16 <= dl) {
             *(uint4 *)dst = *(const uint4 *)ls;       /* dst 16B-aligned (writer) */
         } else {
-            for (uint32_t j = 0; j < s.decomp_len - boff; j++) dst[j] = ls[j];
+            for (uint32_t j = 0; j < dl - boff; j++) dst[j] = ls[j];
         }
     }
 }
@@ -843,13 +848,17 @@ int csgpu_stage(cstripe_scan *s, int device_id)
 static void launch_decode(cs_gpu_state *g)
 {
     if (g->n_segs == 0) return;
-    /* lane-parallel path for micro-segments (one lane per segment) */
+    /* lane-parallel path for micro-segments (one lane per segment).
+     * stride: 16 B-multiple, (stride/4)%64 != 0 so equal-progress lanes land
+     * on different banks; LDS/wave = 64*stride -> waves/CU 9 / 4 / 2. */
     if (g->segs_16aligned && g->max_seg_dlen <= 1039) {
-        uint32_t block = (g->max_seg_dlen <= 527) ? 256 : 128;
-        uint32_t stride = (g->max_seg_dlen <= 527) ? 528 : 1040;
+        uint32_t block, stride;
+        if (g->max_seg_dlen <= 271)      { block = 256; stride = 272; }
+        else if (g->max_seg_dlen <= 527) { block = 256; stride = 528; }
+        else                             { block = 128; stride = 1040; }
         uint32_t grid = (g->n_segs + block - 1) / block;
         hipLaunchKernelGGL(lz4_decode_lane_kernel, dim3(grid), dim3(block),
-                           block * stride, g->stream,
+                           block * stride + block * 16, g->stream,
                            g->d_data, g->d_scratch, g->d_segs, g->n_segs, stride,
                            g->d_error);
         return;

@@ -59,7 +59,7 @@ extern "C" void cstripe_default_options(cstripe_options *o)
     o->compression = CSTRIPE_COMP_LZ4;
     o->compression_level = 3;
     o->lz4_seg_target_kb = 0;
-    o->lz4_seg_target_bytes = 512;
+    o->lz4_seg_target_bytes = 256;
 }
 
 static uint32_t type_width(uint8_t t) { return csf_type_width(t); }
@@ -199,7 +199,7 @@ extern "C" cstripe_writer *cstripe_write_begin(const char *path, const cstripe_c
     w->path = path;
     if (opts) w->opts = *opts; else cstripe_default_options(&w->opts);
     if (w->opts.lz4_seg_target_bytes == 0 && w->opts.lz4_seg_target_kb == 0)
-        w->opts.lz4_seg_target_bytes = 512;
+        w->opts.lz4_seg_target_bytes = 256;
     for (uint32_t i = 0; i < n_cols; i++) {
         csf_coldef d{};
         memcpy(d.name, cols[i].name, sizeof(d.name));

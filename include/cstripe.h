@@ -108,7 +108,7 @@ typedef struct cstripe_options {
                                         * one segment == plain whole-chunk
                                         * block, exactly the reference layout) */
     uint32_t    lz4_seg_target_bytes;  /* fine knob, overrides kb when nonzero;
-                                        * default 512 B = one GPU lane per
+                                        * default 256 B = one GPU lane per
                                         * segment (lane-parallel decode) */
 } cstripe_options;
 
