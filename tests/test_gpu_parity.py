@@ -208,3 +208,21 @@ def test_q6_large_synthetic(tmp_path):
     op, ofilt, gp, gfilt = both(path, preds_q6, aggs_q6)
     assert ofilt == gfilt
     assert_parity(op, gp, aggs_q6)
+
+
+@pytest.mark.parametrize("seg", [("bytes", 0), ("kb", 8), ("kb", 1024)])
+def test_decode_kernel_variants(tmp_path, seg):
+    """All three decode kernels give identical, oracle-exact results:
+    lane-parallel (default 256B micro-segments), wave-cooperative LDS
+    (8KB segments), global fallback (single whole-chunk block)."""
+    kind, val = seg
+    path = str(tmp_path / f"v{val}.cs")
+    kw = {"seg_kb": val} if kind == "kb" else {}
+    ca.gen_lineitem(path, 300_000, **kw)
+    preds = [(5, ca.PRED_GE, 8766), (5, ca.PRED_LT, 9131),
+             (3, ca.PRED_GE, 5), (3, ca.PRED_LE, 7), (1, ca.PRED_LT, 2400)]
+    aggs = [(ca.AGG_SUM_PROD_I64, 2, 3), (ca.AGG_COUNT_STAR, -1),
+            (ca.AGG_MIN_I64, 2), (ca.AGG_MAX_I64, 2)]
+    op, ofilt, gp, gfilt = both(path, preds, aggs)
+    assert ofilt == gfilt
+    assert_parity(op, gp, aggs)
