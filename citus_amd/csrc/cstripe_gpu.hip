@@ -220,7 +220,7 @@ __global__ __launch_bounds__(WAVE) void lz4_decode_kernel(
 
 struct ByteStream {
     const uint8_t *base;    /* 8B-aligned */
-    uint64_t w0, w1;        /* current 16-byte window */
+    uint64_t w0, w1, w2;    /* 24-byte window; w2 is ~2 rotations of prefetch */
     uint32_t woff;          /* byte offset of w0 within base */
     uint32_t pos;           /* current byte position (>= initial shift) */
 };
@@ -233,14 +233,16 @@ __device__ inline void bs_init(ByteStream &b, const uint8_t *data, uint64_t off)
     b.woff = 0;
     b.w0 = *(const uint64_t *)(b.base);
     b.w1 = *(const uint64_t *)(b.base + 8);
+    b.w2 = *(const uint64_t *)(b.base + 16);
 }
 
 __device__ inline void bs_norm(ByteStream &b)
 {
     while (b.pos - b.woff >= 8) {
         b.w0 = b.w1;
+        b.w1 = b.w2;
         b.woff += 8;
-        b.w1 = *(const uint64_t *)(b.base + b.woff + 8);
+        b.w2 = *(const uint64_t *)(b.base + b.woff + 16);
     }
 }
 
@@ -250,14 +252,6 @@ __device__ inline uint8_t bs_get(ByteStream &b)
     uint32_t rel = b.pos - b.woff;
     b.pos++;
     return (uint8_t)(b.w0 >> (8 * rel));
-}
-
-/* next 8 bytes at pos as a little-endian u64 (rel kept in [0,8) by bs_norm) */
-__device__ inline uint64_t bs_peek64(ByteStream &b)
-{
-    bs_norm(b);
-    uint32_t s = 8 * (b.pos - b.woff);
-    return s ? ((b.w0 >> s) | (b.w1 << (64 - s))) : b.w0;
 }
 
 __global__ void lz4_decode_lane_kernel(
@@ -282,132 +276,37 @@ __global__ void lz4_decode_lane_kernel(
         bs_init(bs, data, s.src_off);
         const uint32_t send = bs.pos + s.comp_len;   /* end position */
         const uint32_t dlen = s.decomp_len;
-        /* word-granular output: W = this lane's LDS region as u64 words;
-         * obuf buffers the bytes of the current (incomplete) word — every
-         * byte below op & ~7 is already in LDS */
-        uint64_t *W = (uint64_t *)sout;              /* sout is 16B-aligned */
-        uint64_t obuf = 0;
         uint32_t op = 0;
         bool bad = false;
-
-        #define PUT_BYTE(bv) do { \
-            obuf |= (uint64_t)(uint8_t)(bv) << (8 * (op & 7)); \
-            op++; \
-            if (!(op & 7)) { W[(op >> 3) - 1] = obuf; obuf = 0; } \
-        } while (0)
-        /* append 8 bytes from v (always completes exactly one word) */
-        #define PUT8(v64) do { \
-            uint32_t _s = 8 * (op & 7); \
-            uint64_t _v = (v64); \
-            W[op >> 3] = _s ? (obuf | (_v << _s)) : _v; \
-            obuf = _s ? (_v >> (64 - _s)) : 0; \
-            op += 8; \
-        } while (0)
-
         while (bs.pos < send) {
-            uint64_t h = bs_peek64(bs);
-            const uint32_t token = (uint32_t)h & 0xff;
+            const uint32_t token = bs_get(bs);
             uint32_t litlen = token >> 4;
-            uint32_t mlen = token & 15;
-
-            if (litlen <= 5 && mlen != 15 && bs.pos + 3 + litlen <= send) {
-                /* fast path: token + literals + offset all inside h */
-                uint64_t lits = h >> 8;
-                uint32_t offset = (uint32_t)(h >> (8 * (1 + litlen))) & 0xFFFF;
-                bs.pos += 3 + litlen;
-                if (op + litlen > dlen) { bad = true; break; }
-                for (uint32_t j = 0; j < litlen; j++) { PUT_BYTE(lits); lits >>= 8; }
-                mlen += 4;
-                if (offset == 0 || offset > op || op + mlen > dlen) { bad = true; break; }
-                if ((offset & 7) == 0) {
-                    /* word-phase match: src and dst share the byte phase, so
-                     * copy whole words k = offset/8 back; k==1 is a constant
-                     * repeating word (the common offset-8 value pattern) */
-                    const uint32_t k = offset >> 3;
-                    const uint32_t phase = op & 7;
-                    uint32_t wi = op >> 3;
-                    const uint32_t opn = op + mlen;
-                    const uint32_t last_wi = (opn - 1) >> 3;
-                    uint64_t lowmask = phase ? ((1ull << (8 * phase)) - 1) : 0;
-                    uint64_t w = (obuf & lowmask) | (W[wi - k] & ~lowmask);
-                    W[wi] = w;
-                    if (k == 1) {
-                        for (uint32_t i = wi + 1; i <= last_wi; i++) W[i] = w;
-                    } else {
-                        for (uint32_t i = wi + 1; i <= last_wi; i++) { w = W[i - k]; W[i] = w; }
-                    }
-                    op = opn;
-                    obuf = (op & 7) ? (w & ((1ull << (8 * (op & 7))) - 1)) : 0;
-                } else if (offset >= 8) {
-                    /* sources are always a full word below op -> already in LDS */
-                    const uint32_t mstart = op - offset;
-                    for (uint32_t j = 0; j < mlen; j++) PUT_BYTE(sout[mstart + j]);
-                } else {
-                    /* short offset: replicate the pattern from a register
-                     * (flush the partial word first so all bytes < op are
-                     * readable) */
-                    W[op >> 3] = obuf;
-                    uint64_t pat = 0;
-                    const uint32_t mstart = op - offset;
-                    for (uint32_t j = 0; j < offset; j++)
-                        pat |= (uint64_t)sout[mstart + j] << (8 * j);
-                    uint32_t pj = 0;
-                    for (uint32_t j = 0; j < mlen; j++) {
-                        PUT_BYTE(pat >> (8 * pj));
-                        pj++;
-                        if (pj == offset) pj = 0;
-                    }
-                }
-                continue;
-            }
-
-            /* generic path */
-            bs.pos++;                                   /* token */
             if (litlen == 15) {
                 uint32_t b;
                 do { b = bs_get(bs); litlen += b; } while (b == 255 && bs.pos < send);
             }
             if (bs.pos + litlen > send || op + litlen > dlen) { bad = true; break; }
-            {
-                uint32_t left = litlen;
-                while (left >= 8) { PUT8(bs_peek64(bs)); bs.pos += 8; left -= 8; }
-                if (left) {
-                    uint64_t v = bs_peek64(bs);
-                    bs.pos += left;
-                    for (uint32_t j = 0; j < left; j++) { PUT_BYTE(v); v >>= 8; }
-                }
-            }
+            for (uint32_t j = 0; j < litlen; j++) sout[op + j] = bs_get(bs);
+            op += litlen;
             if (bs.pos >= send) break;       /* last sequence: literals only */
 
             if (bs.pos + 2 > send) { bad = true; break; }
             uint32_t offset = (uint32_t)bs_get(bs);
             offset |= (uint32_t)bs_get(bs) << 8;
+            uint32_t mlen = token & 15;
             if (mlen == 15) {
                 uint32_t b;
                 do { b = bs_get(bs); mlen += b; } while (b == 255 && bs.pos < send);
             }
             mlen += 4;
             if (offset == 0 || offset > op || op + mlen > dlen) { bad = true; break; }
-            if (offset >= 8) {
-                const uint32_t mstart = op - offset;
-                for (uint32_t j = 0; j < mlen; j++) PUT_BYTE(sout[mstart + j]);
-            } else {
-                W[op >> 3] = obuf;
-                uint64_t pat = 0;
-                const uint32_t mstart = op - offset;
-                for (uint32_t j = 0; j < offset; j++)
-                    pat |= (uint64_t)sout[mstart + j] << (8 * j);
-                uint32_t pj = 0;
-                for (uint32_t j = 0; j < mlen; j++) {
-                    PUT_BYTE(pat >> (8 * pj));
-                    pj++;
-                    if (pj == offset) pj = 0;
-                }
-            }
+            /* uniform per-lane byte copy — control flow stays convergent
+             * across the wave (a branchier word-copy variant measured 2x the
+             * per-byte instructions from divergence; profiles/r01) */
+            const uint8_t *msrc = sout + op - offset;
+            for (uint32_t j = 0; j < mlen; j++) sout[op + j] = msrc[j];
+            op += mlen;
         }
-        if (op & 7) W[op >> 3] = obuf;       /* final partial word */
-        #undef PUT_BYTE
-        #undef PUT8
         if (bad || op != dlen) atomicOr(err, 4);
     }
     __syncthreads();
