@@ -1,0 +1,1090 @@
+/*
+ * cstripe_gpu.hip — MI355X (gfx950/CDNA4) device path of the columnar scan:
+ * HtoD staging of compressed chunk streams, LZ4 segment decode kernel, and
+ * the fused predicate-filter + partial-aggregate kernels.
+ *
+ * This replaces, with HIP kernels, the CPU loops under ColumnarReadNextRow:
+ *   - DecompressBuffer/LZ4_decompress_safe per chunk
+ *     (columnar_compression.c:165-198)           -> lz4_decode_kernel
+ *   - DeserializeBoolArray (columnar_reader.c:1506-1534): exists bitmaps are
+ *     NOT unpacked — kernels consume the packed bits directly (rank tables
+ *     precomputed host-side at stage)
+ *   - DeserializeDatumArray (columnar_reader.c:1542-1572): fixed-width value
+ *     streams are read as coalesced typed loads (layout guaranteed by
+ *     SerializeSingleDatum, columnar_writer.c:555-585)
+ *   - per-row qual eval (ExecQual under columnar_customscan.c:1907-1913)
+ *     -> predicate evaluation in-register, per 64-lane wavefront
+ *   - PG Agg transition + worker_partial_agg (aggregate_utils.c:501-607)
+ *     -> per-thread accumulate, wave shuffle reduce, LDS cross-wave reduce,
+ *        per-block partial, device-wide final reduce (no global atomics)
+ *
+ * All scan/filter/reduce — HBM-bandwidth bound; no MFMA (north star).
+ * Wavefront = 64 throughout; __ballot is 64-bit.
+ */
+#include "internal.h"
+#include "compress.h"
+
+#include <hip/hip_runtime.h>
+#include <cstring>
+#include <cstdio>
+#include <vector>
+
+#define WAVE 64
+#define AGG_BLOCK 256
+#define TILE_ROWS 4096
+#define MAX_PREDS 8
+#define MAX_AGGS 12
+#define MAX_PROJ 16
+
+/* ---------------- device-side descriptors ---------------- */
+
+struct SegDesc {
+    uint64_t src_off;       /* into d_data (compressed) */
+    uint64_t dst_off;       /* into d_scratch (decompressed) */
+    uint32_t comp_len;
+    uint32_t decomp_len;
+};
+
+struct ColLoc {
+    uint64_t val_off;       /* into d_scratch if (flags&1) else d_data */
+    uint64_t exists_off;    /* into d_data (packed bitmap, 8B-padded) */
+    uint32_t rank_off;      /* u32 word-rank table index into d_rank */
+    uint16_t flags;         /* 1 = values in scratch; 2 = dense (no nulls) */
+    uint8_t  type;          /* cstripe_type */
+    uint8_t  width;
+};
+
+struct GroupDesc {
+    uint32_t row_count;
+    uint32_t colbase;       /* index into d_colloc: colbase + proj_idx */
+};
+
+struct PredD {
+    uint16_t proj;
+    uint8_t  op;
+    uint8_t  is_float;
+    int64_t  ival;
+    double   fval;
+};
+
+struct AggD {
+    uint8_t kind;
+    uint8_t proj_a, proj_b, proj_c;
+    int64_t one;
+};
+
+struct AggParams {
+    uint32_t n_preds, n_aggs, n_proj, tiles_per_group, n_groups;
+    uint32_t pad[3];
+    PredD preds[MAX_PREDS];
+    AggD aggs[MAX_AGGS];
+};
+
+/* per-block / final accumulator cell (32 B) */
+struct AccCell {
+    int64_t lo;             /* i128 low  / i64 min-max */
+    int64_t hi;             /* i128 high */
+    double  f;              /* f64 sum / min / max */
+    int64_t cnt;            /* contributing rows */
+};
+
+struct cs_gpu_state {
+    int device = -1;
+    hipStream_t stream = nullptr;
+    uint8_t *d_data = nullptr;       /* staged compressed values + exists */
+    uint8_t *d_scratch = nullptr;    /* decompressed value streams */
+    uint32_t *d_rank = nullptr;
+    SegDesc *d_segs = nullptr;
+    GroupDesc *d_groups = nullptr;
+    ColLoc *d_colloc = nullptr;
+    AccCell *d_block = nullptr;
+    AccCell *d_final = nullptr;
+    int *d_error = nullptr;
+    hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr;
+
+    uint64_t data_bytes = 0;
+    uint64_t scratch_bytes = 0;
+    uint32_t n_segs = 0;
+    uint32_t max_seg_comp = 0;
+    uint32_t max_seg_dlen = 0;
+    bool segs_16aligned = true;
+    uint32_t n_groups = 0;
+    uint32_t n_proj = 0;
+    uint32_t max_blocks = 0;
+    int proj_of_col[64];             /* column index -> projected slot (-1) */
+    uint8_t proj_type[MAX_PROJ];
+    /* host-side mirror for next_batch (exists read from mmap) */
+    std::vector<uint64_t> scratch_off;  /* per (sel,proj): decomp offset */
+};
+
+/* ---------------- error helper ---------------- */
+
+#define HIP_TRY(x) do { hipError_t _e = (x); if (_e != hipSuccess) { \
+    cs_set_err("%s failed: %s", #x, hipGetErrorString(_e)); return CSTRIPE_ERR; } } while (0)
+
+extern "C" int cstripe_gpu_available(void)
+{
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n > 0 ? 1 : 0;
+}
+
+/* =====================================================================
+ * LZ4 block decode — one 64-lane wavefront per independently decodable
+ * segment. All lanes parse the sequence header redundantly (same-address
+ * loads broadcast); literal and match copies are lane-parallel. Overlapping
+ * matches (offset < length) resolve every byte to the pre-existing pattern
+ * window via modulo, so one parallel pass is race-free.
+ * Implements the LZ4 block format consumed by LZ4_decompress_safe
+ * (columnar_compression.c:183); any valid block decodes, segmented or not.
+ *
+ * Two variants:
+ *  - lz4_decode_lds_kernel: compressed segment is first staged into LDS with
+ *    coalesced u32 loads and the OUTPUT is decoded in LDS too, so the
+ *    byte-serial parse and all match copies run at LDS latency instead of
+ *    chained dependent HBM round trips; the finished segment is flushed to
+ *    global with coalesced 16 B stores. Needs in+out <= 64 KiB of LDS —
+ *    the writer's default 8 KiB segments give ~9 resident waves/CU.
+ *  - lz4_decode_kernel: the global-memory fallback for oversized segments
+ *    (e.g. single-block whole-chunk compatibility mode).
+ * ===================================================================== */
+
+template <bool IN_LDS>
+__device__ inline void lz4_body(const uint8_t *__restrict__ src, uint32_t slen,
+                                uint8_t *__restrict__ dst, uint32_t dlen,
+                                uint32_t lane, int *__restrict__ err)
+{
+    uint32_t ip = 0, op = 0;
+    while (ip < slen) {
+        const uint32_t token = src[ip++];
+        /* literals */
+        uint32_t litlen = token >> 4;
+        if (litlen == 15) {
+            uint32_t b;
+            do { b = src[ip++]; litlen += b; } while (b == 255 && ip < slen);
+        }
+        if (ip + litlen > slen || op + litlen > dlen) { if (lane == 0) atomicOr(err, 1); return; }
+        for (uint32_t j = lane; j < litlen; j += WAVE) dst[op + j] = src[ip + j];
+        ip += litlen;
+        op += litlen;
+        if (ip >= slen) break;           /* last sequence: literals only */
+
+        /* match */
+        if (ip + 2 > slen) { if (lane == 0) atomicOr(err, 1); return; }
+        uint32_t offset = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
+        ip += 2;
+        uint32_t mlen = token & 15;
+        if (mlen == 15) {
+            uint32_t b;
+            do { b = src[ip++]; mlen += b; } while (b == 255 && ip < slen);
+        }
+        mlen += 4;
+        if (offset == 0 || offset > op || op + mlen > dlen) { if (lane == 0) atomicOr(err, 1); return; }
+        const uint32_t mstart = op - offset;
+        if (offset >= mlen) {
+            for (uint32_t j = lane; j < mlen; j += WAVE) dst[op + j] = dst[mstart + j];
+        } else {
+            /* every byte's ultimate source lies in the pre-existing pattern
+             * window [op-offset, op) — single race-free parallel pass */
+            for (uint32_t j = lane; j < mlen; j += WAVE) dst[op + j] = dst[mstart + (j % offset)];
+        }
+        op += mlen;
+    }
+    if (op != dlen) { if (lane == 0) atomicOr(err, 2); }
+}
+
+__global__ __launch_bounds__(WAVE) void lz4_decode_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, int *__restrict__ err)
+{
+    const SegDesc s = segs[blockIdx.x];
+    lz4_body<false>(data + s.src_off, s.comp_len, scratch + s.dst_off,
+                    s.decomp_len, threadIdx.x, err);
+}
+
+/* =====================================================================
+ * Lane-parallel LZ4 decode — ONE LANE per micro-segment (writer default
+ * 512 B decompressed). Why: this data compresses to ~one sequence per
+ * 8-byte value (measured ~8 B/sequence), so a wave-cooperative decoder is
+ * serialized on the per-sequence parse chain and moves ~8 B per chain step.
+ * Here every wave advances 64 independent segment streams per step:
+ *  - input: per-lane 16-byte register window over the compressed stream,
+ *    refilled with aligned u64 loads (one dependent global load per 8
+ *    compressed bytes instead of per byte)
+ *  - output: per-lane LDS region (stride 528 = 16-byte aligned, 4-bank
+ *    skew per lane so equal-progress lanes hit different banks); match
+ *    copies are plain per-lane sequential byte moves (memmove-forward
+ *    semantics handles overlap for free)
+ *  - finish: block-cooperative 16 B coalesced flush LDS -> scratch
+ * ===================================================================== */
+
+struct ByteStream {
+    const uint8_t *base;    /* 8B-aligned */
+    uint64_t w0, w1, w2;    /* 24-byte window; w2 is ~2 rotations of prefetch */
+    uint32_t woff;          /* byte offset of w0 within base */
+    uint32_t pos;           /* current byte position (>= initial shift) */
+};
+
+__device__ inline void bs_init(ByteStream &b, const uint8_t *data, uint64_t off)
+{
+    uint64_t a = off & ~7ull;
+    b.base = data + a;
+    b.pos = (uint32_t)(off - a);
+    b.woff = 0;
+    b.w0 = *(const uint64_t *)(b.base);
+    b.w1 = *(const uint64_t *)(b.base + 8);
+    b.w2 = *(const uint64_t *)(b.base + 16);
+}
+
+__device__ inline void bs_norm(ByteStream &b)
+{
+    while (b.pos - b.woff >= 8) {
+        b.w0 = b.w1;
+        b.w1 = b.w2;
+        b.woff += 8;
+        b.w2 = *(const uint64_t *)(b.base + b.woff + 16);
+    }
+}
+
+__device__ inline uint8_t bs_get(ByteStream &b)
+{
+    bs_norm(b);
+    uint32_t rel = b.pos - b.woff;
+    b.pos++;
+    return (uint8_t)(b.w0 >> (8 * rel));
+}
+
+__global__ void lz4_decode_lane_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, uint32_t n_segs, uint32_t stride,
+    int *__restrict__ err)
+{
+    extern __shared__ uint8_t sout_all[];
+    const uint32_t tid = threadIdx.x;
+    const uint32_t first = blockIdx.x * blockDim.x;
+    const uint32_t gseg = first + tid;
+    uint8_t *sout = sout_all + (size_t)tid * stride;
+    /* per-segment (dst_off, decomp_len) cached in LDS for the flush loop */
+    uint64_t *sdst = (uint64_t *)(sout_all + (size_t)blockDim.x * stride);
+    uint32_t *sdlen = (uint32_t *)(sdst + blockDim.x);
+
+    if (gseg < n_segs) {
+        const SegDesc s = segs[gseg];
+        sdst[tid] = s.dst_off;
+        sdlen[tid] = s.decomp_len;
+        ByteStream bs;
+        bs_init(bs, data, s.src_off);
+        const uint32_t send = bs.pos + s.comp_len;   /* end position */
+        const uint32_t dlen = s.decomp_len;
+        uint32_t op = 0;
+        bool bad = false;
+        while (bs.pos < send) {
+            const uint32_t token = bs_get(bs);
+            uint32_t litlen = token >> 4;
+            if (litlen == 15) {
+                uint32_t b;
+                do { b = bs_get(bs); litlen += b; } while (b == 255 && bs.pos < send);
+            }
+            if (bs.pos + litlen > send || op + litlen > dlen) { bad = true; break; }
+            for (uint32_t j = 0; j < litlen; j++) sout[op + j] = bs_get(bs);
+            op += litlen;
+            if (bs.pos >= send) break;       /* last sequence: literals only */
+
+            if (bs.pos + 2 > send) { bad = true; break; }
+            uint32_t offset = (uint32_t)bs_get(bs);
+            offset |= (uint32_t)bs_get(bs) << 8;
+            uint32_t mlen = token & 15;
+            if (mlen == 15) {
+                uint32_t b;
+                do { b = bs_get(bs); mlen += b; } while (b == 255 && bs.pos < send);
+            }
+            mlen += 4;
+            if (offset == 0 || offset > op || op + mlen > dlen) { bad = true; break; }
+            /* uniform per-lane byte copy — control flow stays convergent
+             * across the wave (a branchier word-copy variant measured 2x the
+             * per-byte instructions from divergence; profiles/r01) */
+            const uint8_t *msrc = sout + op - offset;
+            for (uint32_t j = 0; j < mlen; j++) sout[op + j] = msrc[j];
+            op += mlen;
+        }
+        if (bad || op != dlen) atomicOr(err, 4);
+    }
+    __syncthreads();
+
+    /* cooperative coalesced flush: flat element index over the block's
+     * segments, stride/16 u128 slots per segment */
+    const uint32_t eps = stride >> 4;                 /* elements per segment */
+    const uint32_t in_block = min(blockDim.x, n_segs - first);
+    const uint32_t total = in_block * eps;
+    for (uint32_t f = tid; f < total; f += blockDim.x) {
+        const uint32_t sidx = f / eps;
+        const uint32_t boff = (f % eps) << 4;
+        const uint32_t dl = sdlen[sidx];
+        if (boff >= dl) continue;
+        uint8_t *dst = scratch + sdst[sidx] + boff;
+        const uint8_t *ls = sout_all + (size_t)sidx * stride + boff;
+        if (boff + 16 <= dl) {
+            *(uint4 *)dst = *(const uint4 *)ls;       /* dst 16B-aligned (writer) */
+        } else {
+            for (uint32_t j = 0; j < dl - boff; j++) dst[j] = ls[j];
+        }
+    }
+}
+
+__global__ __launch_bounds__(WAVE) void lz4_decode_lds_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, uint32_t in_cap, int *__restrict__ err)
+{
+    extern __shared__ uint8_t sbuf[];      /* [in_cap compressed][output] */
+    const SegDesc s = segs[blockIdx.x];
+    const uint32_t lane = threadIdx.x;
+
+    /* stage compressed bytes: aligned u32 loads from (src_off & ~3) */
+    const uint64_t base = s.src_off & ~3ull;
+    const uint32_t shift = (uint32_t)(s.src_off - base);
+    const uint32_t words = (s.comp_len + shift + 3) >> 2;
+    const uint32_t *gsrc = (const uint32_t *)(data + base);
+    uint32_t *lin = (uint32_t *)sbuf;
+    for (uint32_t j = lane; j < words; j += WAVE) lin[j] = gsrc[j];
+    __syncthreads();
+
+    uint8_t *out = sbuf + in_cap;
+    lz4_body<true>(sbuf + shift, s.comp_len, out, s.decomp_len, lane, err);
+    __syncthreads();
+
+    /* coalesced flush LDS -> global */
+    uint8_t *dst = scratch + s.dst_off;
+    const uint32_t dlen = s.decomp_len;
+    if (((uintptr_t)dst & 15) == 0) {
+        uint32_t vec = dlen >> 4;
+        uint4 *d4 = (uint4 *)dst;
+        const uint4 *o4 = (const uint4 *)out;
+        for (uint32_t j = lane; j < vec; j += WAVE) d4[j] = o4[j];
+        for (uint32_t j = (vec << 4) + lane; j < dlen; j += WAVE) dst[j] = out[j];
+    } else {
+        for (uint32_t j = lane; j < dlen; j += WAVE) dst[j] = out[j];
+    }
+}
+
+/* =====================================================================
+ * Fused filter + partial aggregate (generic, ungrouped)
+ * grid: n_groups * tiles_per_group blocks of AGG_BLOCK threads.
+ * Each thread strides rows of its tile; per-thread register accumulators;
+ * wave __shfl_down reduce; LDS cross-wave reduce; one AccCell per block.
+ * ===================================================================== */
+
+__device__ inline bool col_value(const uint8_t *__restrict__ data,
+                                 const uint8_t *__restrict__ scratch,
+                                 const uint32_t *__restrict__ rank,
+                                 const ColLoc &cl, uint32_t row,
+                                 int64_t &iv, double &fv)
+{
+    uint32_t idx = row;
+    if (!(cl.flags & 2)) {   /* sparse: null check + rank indirection */
+        const uint64_t *bm = (const uint64_t *)(data + cl.exists_off);
+        uint64_t w = bm[row >> 6];
+        uint64_t bit = 1ull << (row & 63);
+        if (!(w & bit)) return false;
+        idx = rank[cl.rank_off + (row >> 6)] + (uint32_t)__popcll(w & (bit - 1));
+    }
+    const uint8_t *base = (cl.flags & 1) ? scratch + cl.val_off : data + cl.val_off;
+    switch (cl.type) {
+        case CSTRIPE_I8:  iv = ((const int8_t  *)base)[idx]; fv = (double)iv; break;
+        case CSTRIPE_I16: iv = ((const int16_t *)base)[idx]; fv = (double)iv; break;
+        case CSTRIPE_I32: iv = ((const int32_t *)base)[idx]; fv = (double)iv; break;
+        case CSTRIPE_I64: iv = ((const int64_t *)base)[idx]; fv = (double)iv; break;
+        case CSTRIPE_F32: fv = ((const float  *)base)[idx]; iv = 0; break;
+        default:          fv = ((const double *)base)[idx]; iv = 0; break;
+    }
+    return true;
+}
+
+__device__ inline bool pred_eval(const PredD &p, int64_t iv, double fv)
+{
+    if (p.is_float) {
+        switch (p.op) {
+            case CSTRIPE_PRED_LT: return fv <  p.fval;
+            case CSTRIPE_PRED_LE: return fv <= p.fval;
+            case CSTRIPE_PRED_GT: return fv >  p.fval;
+            case CSTRIPE_PRED_GE: return fv >= p.fval;
+            case CSTRIPE_PRED_EQ: return fv == p.fval;
+            default:              return fv != p.fval;
+        }
+    }
+    switch (p.op) {
+        case CSTRIPE_PRED_LT: return iv <  p.ival;
+        case CSTRIPE_PRED_LE: return iv <= p.ival;
+        case CSTRIPE_PRED_GT: return iv >  p.ival;
+        case CSTRIPE_PRED_GE: return iv >= p.ival;
+        case CSTRIPE_PRED_EQ: return iv == p.ival;
+        default:              return iv != p.ival;
+    }
+}
+
+struct ThreadAcc {
+    int64_t lo, hi;   /* i128 sum (lo unsigned semantics) / i64 minmax in lo */
+    double  f;
+    int64_t cnt;
+};
+
+__device__ inline void acc_init(ThreadAcc &a, uint8_t kind)
+{
+    a.lo = 0; a.hi = 0; a.f = 0.0; a.cnt = 0;
+    switch (kind) {
+        case CSTRIPE_AGG_MIN_I64: a.lo = INT64_MAX; break;
+        case CSTRIPE_AGG_MAX_I64: a.lo = INT64_MIN; break;
+        case CSTRIPE_AGG_MIN_F64: a.f = INFINITY; break;
+        case CSTRIPE_AGG_MAX_F64: a.f = -INFINITY; break;
+        default: break;
+    }
+}
+
+__device__ inline void acc_add_i128(ThreadAcc &a, __int128 v)
+{
+    __int128 x = ((__int128)a.hi << 64) | (unsigned long long)a.lo;
+    x += v;
+    a.lo = (int64_t)(uint64_t)x;
+    a.hi = (int64_t)(x >> 64);
+}
+
+__device__ inline void acc_merge(ThreadAcc &a, const ThreadAcc &b, uint8_t kind)
+{
+    switch (kind) {
+        case CSTRIPE_AGG_MIN_I64: a.lo = min(a.lo, b.lo); break;
+        case CSTRIPE_AGG_MAX_I64: a.lo = max(a.lo, b.lo); break;
+        case CSTRIPE_AGG_MIN_F64: a.f = fmin(a.f, b.f); break;
+        case CSTRIPE_AGG_MAX_F64: a.f = fmax(a.f, b.f); break;
+        case CSTRIPE_AGG_SUM_F64: a.f += b.f; break;
+        default: {  /* counts + i128 sums */
+            unsigned long long lo = (unsigned long long)a.lo + (unsigned long long)b.lo;
+            int64_t carry = lo < (unsigned long long)a.lo ? 1 : 0;
+            a.lo = (int64_t)lo;
+            a.hi = a.hi + b.hi + carry;
+            break;
+        }
+    }
+    a.cnt += b.cnt;
+}
+
+__device__ inline void acc_row(ThreadAcc &a, const AggD &g,
+                               const uint8_t *__restrict__ data,
+                               const uint8_t *__restrict__ scratch,
+                               const uint32_t *__restrict__ rank,
+                               const ColLoc *__restrict__ cols, uint32_t row)
+{
+    int64_t iv; double fv;
+    switch (g.kind) {
+        case CSTRIPE_AGG_COUNT_STAR:
+            a.cnt++;
+            a.lo++;
+            break;
+        case CSTRIPE_AGG_COUNT_COL:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.cnt++; a.lo++; }
+            break;
+        case CSTRIPE_AGG_SUM_I64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) {
+                acc_add_i128(a, (__int128)iv);
+                a.cnt++;
+            }
+            break;
+        case CSTRIPE_AGG_SUM_F64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.f += fv; a.cnt++; }
+            break;
+        case CSTRIPE_AGG_MIN_I64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.lo = min(a.lo, iv); a.cnt++; }
+            break;
+        case CSTRIPE_AGG_MAX_I64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.lo = max(a.lo, iv); a.cnt++; }
+            break;
+        case CSTRIPE_AGG_MIN_F64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.f = fmin(a.f, fv); a.cnt++; }
+            break;
+        case CSTRIPE_AGG_MAX_F64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.f = fmax(a.f, fv); a.cnt++; }
+            break;
+        case CSTRIPE_AGG_SUM_PROD_I64: {
+            int64_t ib; double fb;
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
+                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb)) {
+                acc_add_i128(a, (__int128)iv * ib);
+                a.cnt++;
+            }
+            break;
+        }
+        case CSTRIPE_AGG_SUM_DISC_I64: {
+            int64_t ib; double fb;
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
+                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb)) {
+                acc_add_i128(a, (__int128)iv * (g.one - ib));
+                a.cnt++;
+            }
+            break;
+        }
+        case CSTRIPE_AGG_SUM_DISC_TAX_I64: {
+            int64_t ib, ic; double fb, fc;
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
+                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb) &&
+                col_value(data, scratch, rank, cols[g.proj_c], row, ic, fc)) {
+                acc_add_i128(a, (__int128)iv * (g.one - ib) * (g.one + ic));
+                a.cnt++;
+            }
+            break;
+        }
+    }
+}
+
+__device__ inline void wave_reduce(ThreadAcc &a, uint8_t kind)
+{
+    for (int d = WAVE / 2; d > 0; d >>= 1) {
+        ThreadAcc b;
+        b.lo  = __shfl_down((long long)a.lo, d, WAVE);
+        b.hi  = __shfl_down((long long)a.hi, d, WAVE);
+        b.f   = __shfl_down(a.f, d, WAVE);
+        b.cnt = __shfl_down((long long)a.cnt, d, WAVE);
+        acc_merge(a, b, kind);
+    }
+}
+
+__global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
+    const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
+    const uint32_t *__restrict__ rank, const GroupDesc *__restrict__ groups,
+    const ColLoc *__restrict__ colloc, AccCell *__restrict__ block_out,
+    const AggParams params)
+{
+    const uint32_t gid = blockIdx.x / params.tiles_per_group;
+    const uint32_t tile = blockIdx.x % params.tiles_per_group;
+    const GroupDesc g = groups[gid];
+    const ColLoc *cols = colloc + g.colbase;
+
+    uint32_t row_start = tile * TILE_ROWS;
+    uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
+
+    ThreadAcc acc[MAX_AGGS];
+    for (uint32_t a = 0; a < params.n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
+
+    for (uint32_t row = row_start + threadIdx.x; row < row_end; row += AGG_BLOCK) {
+        bool pass = true;
+        int last_proj = -1;
+        int64_t liv = 0; double lfv = 0; bool lok = false;
+        /* no short-circuit: loads stay control-independent so they issue
+         * back-to-back and pipeline instead of chaining load->wait->branch */
+        for (uint32_t p = 0; p < params.n_preds; p++) {
+            const PredD &pr = params.preds[p];
+            if ((int)pr.proj != last_proj) {   /* BETWEEN reuses the load */
+                lok = col_value(data, scratch, rank, cols[pr.proj], row, liv, lfv);
+                last_proj = (int)pr.proj;
+            }
+            pass = pass & (lok && pred_eval(pr, liv, lfv));
+        }
+        if (!pass) continue;
+        for (uint32_t a = 0; a < params.n_aggs; a++)
+            acc_row(acc[a], params.aggs[a], data, scratch, rank, cols, row);
+    }
+
+    /* wave reduce then cross-wave via LDS */
+    __shared__ ThreadAcc lds[AGG_BLOCK / WAVE][MAX_AGGS];
+    const uint32_t wid = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    for (uint32_t a = 0; a < params.n_aggs; a++) {
+        wave_reduce(acc[a], params.aggs[a].kind);
+        if (lane == 0) lds[wid][a] = acc[a];
+    }
+    __syncthreads();
+    if (wid == 0) {
+        for (uint32_t a = lane; a < params.n_aggs; a += WAVE) {
+            ThreadAcc r = lds[0][a];
+            for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
+                acc_merge(r, lds[w][a], params.aggs[a].kind);
+            AccCell c;
+            c.lo = r.lo; c.hi = r.hi; c.f = r.f; c.cnt = r.cnt;
+            block_out[(uint64_t)blockIdx.x * params.n_aggs + a] = c;
+        }
+    }
+}
+
+/* device-wide final reduce over block partials: one block, grid-stride */
+__global__ __launch_bounds__(AGG_BLOCK) void final_reduce_kernel(
+    const AccCell *__restrict__ block_in, uint32_t n_blocks,
+    AccCell *__restrict__ out, const AggParams params)
+{
+    __shared__ ThreadAcc lds[AGG_BLOCK / WAVE][MAX_AGGS];
+    const uint32_t wid = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    for (uint32_t a = 0; a < params.n_aggs; a++) {
+        ThreadAcc r;
+        acc_init(r, params.aggs[a].kind);
+        for (uint32_t b = threadIdx.x; b < n_blocks; b += AGG_BLOCK) {
+            const AccCell c = block_in[(uint64_t)b * params.n_aggs + a];
+            ThreadAcc t{c.lo, c.hi, c.f, c.cnt};
+            acc_merge(r, t, params.aggs[a].kind);
+        }
+        wave_reduce(r, params.aggs[a].kind);
+        if (lane == 0) lds[wid][a] = r;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            ThreadAcc f = lds[0][a];
+            for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
+                acc_merge(f, lds[w][a], params.aggs[a].kind);
+            AccCell c{f.lo, f.hi, f.f, f.cnt};
+            out[a] = c;
+        }
+        __syncthreads();
+    }
+}
+
+/* =====================================================================
+ * staging
+ * ===================================================================== */
+
+static uint64_t align_up(uint64_t x, uint64_t a) { return (x + a - 1) & ~(a - 1); }
+
+void csgpu_release(cstripe_scan *s)
+{
+    if (!s || !s->gpu) return;
+    cs_gpu_state *g = s->gpu;
+    if (g->d_data) hipFree(g->d_data);
+    if (g->d_scratch) hipFree(g->d_scratch);
+    if (g->d_rank) hipFree(g->d_rank);
+    if (g->d_segs) hipFree(g->d_segs);
+    if (g->d_groups) hipFree(g->d_groups);
+    if (g->d_colloc) hipFree(g->d_colloc);
+    if (g->d_block) hipFree(g->d_block);
+    if (g->d_final) hipFree(g->d_final);
+    if (g->d_error) hipFree(g->d_error);
+    if (g->ev0) hipEventDestroy(g->ev0);
+    if (g->ev1) hipEventDestroy(g->ev1);
+    if (g->ev2) hipEventDestroy(g->ev2);
+    if (g->stream) hipStreamDestroy(g->stream);
+    delete g;
+    s->gpu = nullptr;
+}
+
+uint64_t csgpu_staged_bytes(const cstripe_scan *s)
+{
+    return s->gpu ? s->gpu->data_bytes : 0;
+}
+
+int csgpu_stage(cstripe_scan *s, int device_id)
+{
+    if (!cstripe_gpu_available()) { cs_set_err("no HIP device visible — the cstripe GPU path requires an MI355X (no CPU fallback)"); return CSTRIPE_ERR_NOGPU; }
+    if (s->gpu) csgpu_release(s);
+
+    cstripe_reader *r = s->r;
+    auto *g = new cs_gpu_state();
+    s->gpu = g;
+    if (device_id >= 0) {
+        if (hipSetDevice(device_id) != hipSuccess) { cs_set_err("hipSetDevice(%d) failed", device_id); csgpu_release(s); return CSTRIPE_ERR; }
+        g->device = device_id;
+    } else {
+        hipGetDevice(&g->device);
+    }
+
+    /* projected column slots */
+    for (int i = 0; i < 64; i++) g->proj_of_col[i] = -1;
+    uint32_t n_proj = 0;
+    for (uint32_t c = 0; c < r->head.column_count; c++) {
+        if (s->cols_mask & (1ull << c)) {
+            if (n_proj >= MAX_PROJ) { cs_set_err("too many projected columns"); csgpu_release(s); return CSTRIPE_ERR_ARG; }
+            g->proj_of_col[c] = (int)n_proj;
+            g->proj_type[n_proj] = r->cols[c].type;
+            n_proj++;
+        }
+    }
+    if (n_proj == 0) { cs_set_err("empty projection"); csgpu_release(s); return CSTRIPE_ERR_ARG; }
+    g->n_proj = n_proj;
+    g->n_groups = (uint32_t)s->sel.size();
+
+    /* host pass 1: compute layout sizes */
+    uint64_t data_bytes = 0, scratch_bytes = 0, rank_words = 0;
+    uint32_t n_segs = 0;
+    bool zstd_host = false;
+    for (const auto &sc : s->sel) {
+        const cs_stripe_info &st = r->stripes[sc.stripe];
+        uint32_t rows = st.group_rows[sc.chunk];
+        for (uint32_t c = 0; c < r->head.column_count; c++) {
+            if (g->proj_of_col[c] < 0) continue;
+            const cs_skipnode &nd = st.nodes[c][sc.chunk];
+            data_bytes = align_up(data_bytes, 8) + align_up((rows + 7) / 8, 8); /* exists, 8B padded */
+            if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
+                data_bytes = align_up(data_bytes, 16) + align_up(nd.n.value_len, 16);
+                scratch_bytes = align_up(scratch_bytes, 16) + align_up(nd.n.decompressed_size, 16);
+                n_segs += nd.n.n_segs;
+            } else { /* NONE staged raw; ZSTD host-decoded at stage (documented fallback) */
+                if (nd.n.comp_type == CSTRIPE_COMP_ZSTD || nd.n.comp_type == CSTRIPE_COMP_PGLZ) zstd_host = true;
+                data_bytes = align_up(data_bytes, 16) + align_up(nd.n.decompressed_size, 16);
+            }
+            if (nd.n.n_present != nd.n.row_count)
+                rank_words += (rows + 63) / 64;
+        }
+    }
+    (void)zstd_host;
+
+    uint32_t tiles_pg = (r->head.chunk_row_limit + TILE_ROWS - 1) / TILE_ROWS;
+    if (tiles_pg == 0) tiles_pg = 1;
+    uint64_t max_blocks = (uint64_t)g->n_groups * tiles_pg;
+    g->max_blocks = (uint32_t)max_blocks;
+
+    HIP_TRY(hipStreamCreate(&g->stream));
+    HIP_TRY(hipEventCreate(&g->ev0));
+    HIP_TRY(hipEventCreate(&g->ev1));
+    HIP_TRY(hipEventCreate(&g->ev2));
+    data_bytes += 16;               /* register-window read slack */
+    if (data_bytes) HIP_TRY(hipMalloc(&g->d_data, data_bytes));
+    if (scratch_bytes) HIP_TRY(hipMalloc(&g->d_scratch, scratch_bytes));
+    if (rank_words) HIP_TRY(hipMalloc(&g->d_rank, rank_words * 4));
+    if (n_segs) HIP_TRY(hipMalloc(&g->d_segs, (uint64_t)n_segs * sizeof(SegDesc)));
+    HIP_TRY(hipMalloc(&g->d_groups, (uint64_t)g->n_groups * sizeof(GroupDesc)));
+    HIP_TRY(hipMalloc(&g->d_colloc, (uint64_t)g->n_groups * n_proj * sizeof(ColLoc)));
+    HIP_TRY(hipMalloc(&g->d_block, max_blocks * MAX_AGGS * sizeof(AccCell)));
+    HIP_TRY(hipMalloc(&g->d_final, MAX_AGGS * sizeof(AccCell)));
+    HIP_TRY(hipMalloc(&g->d_error, sizeof(int)));
+    g->data_bytes = data_bytes;
+    g->scratch_bytes = scratch_bytes;
+    g->n_segs = n_segs;
+
+    /* host pass 2: build staging buffer + descriptors */
+    std::vector<uint8_t> h_data(data_bytes, 0);
+    std::vector<uint32_t> h_rank;
+    h_rank.reserve(rank_words);
+    std::vector<SegDesc> h_segs;
+    h_segs.reserve(n_segs);
+    std::vector<GroupDesc> h_groups(g->n_groups);
+    std::vector<ColLoc> h_colloc((uint64_t)g->n_groups * n_proj);
+    g->scratch_off.assign((uint64_t)g->n_groups * n_proj, ~0ull);
+
+    uint64_t dpos = 0, spos = 0;
+    std::vector<uint8_t> tmp;   /* zstd host-decode scratch */
+    for (uint32_t gi = 0; gi < g->n_groups; gi++) {
+        const cs_selchunk &sc = s->sel[gi];
+        const cs_stripe_info &st = r->stripes[sc.stripe];
+        uint32_t rows = st.group_rows[sc.chunk];
+        h_groups[gi].row_count = rows;
+        h_groups[gi].colbase = gi * n_proj;
+        const uint8_t *stripe_base = r->map + st.meta.file_offset;
+        for (uint32_t c = 0; c < r->head.column_count; c++) {
+            int pj = g->proj_of_col[c];
+            if (pj < 0) continue;
+            const cs_skipnode &nd = st.nodes[c][sc.chunk];
+            ColLoc cl{};
+            cl.type = r->cols[c].type;
+            cl.width = (uint8_t)csf_type_width(cl.type);
+            bool dense = (nd.n.n_present == nd.n.row_count);
+
+            /* exists bitmap (8B padded) */
+            dpos = align_up(dpos, 8);
+            cl.exists_off = dpos;
+            memcpy(h_data.data() + dpos, stripe_base + nd.n.exists_off, nd.n.exists_len);
+            dpos += align_up((rows + 7) / 8, 8);
+
+            if (!dense) {
+                cl.rank_off = (uint32_t)h_rank.size();
+                uint32_t words = (rows + 63) / 64;
+                const uint8_t *eb = stripe_base + nd.n.exists_off;
+                uint32_t run = 0;
+                for (uint32_t w = 0; w < words; w++) {
+                    h_rank.push_back(run);
+                    uint64_t wv = 0;
+                    uint32_t nb = (uint32_t)nd.n.exists_len - w * 8 >= 8 ? 8 : (uint32_t)nd.n.exists_len - w * 8;
+                    memcpy(&wv, eb + w * 8, nb);
+                    run += (uint32_t)__builtin_popcountll(wv);
+                }
+            } else {
+                cl.flags |= 2;
+            }
+
+            if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
+                dpos = align_up(dpos, 16);
+                memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
+                spos = align_up(spos, 16);
+                cl.val_off = spos;
+                cl.flags |= 1;
+                g->scratch_off[(uint64_t)gi * n_proj + pj] = spos;
+                for (const csf_seg &sg : st.nodes[c][sc.chunk].segs) {
+                    SegDesc sd;
+                    sd.src_off = dpos + sg.comp_off;
+                    sd.dst_off = spos + sg.decomp_off;
+                    sd.comp_len = sg.comp_len;
+                    sd.decomp_len = sg.decomp_len;
+                    if (sg.comp_len > g->max_seg_comp) g->max_seg_comp = sg.comp_len;
+                    if (sg.decomp_len > g->max_seg_dlen) g->max_seg_dlen = sg.decomp_len;
+                    if (sg.decomp_off % 16 != 0) g->segs_16aligned = false;
+                    h_segs.push_back(sd);
+                }
+                dpos += align_up(nd.n.value_len, 16);
+                spos += align_up(nd.n.decompressed_size, 16);
+            } else if (nd.n.comp_type == CSTRIPE_COMP_NONE) {
+                dpos = align_up(dpos, 16);
+                cl.val_off = dpos;
+                memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
+                dpos += align_up(nd.n.value_len, 16);
+            } else if (nd.n.comp_type == CSTRIPE_COMP_ZSTD) {
+                /* host-decode at stage (documented fallback; GPU zstd is a
+                 * later-round item — SURVEY.md §8f(1)) */
+                dpos = align_up(dpos, 16);
+                cl.val_off = dpos;
+                tmp.resize(nd.n.decompressed_size);
+                uint64_t seg_base = 0;
+                (void)seg_base;
+                for (const csf_seg &sg : st.nodes[c][sc.chunk].segs) {
+                    size_t zr = ZSTD_decompress(tmp.data() + sg.decomp_off, sg.decomp_len,
+                                                stripe_base + nd.n.value_off + sg.comp_off, sg.comp_len);
+                    if (ZSTD_isError(zr) || zr != sg.decomp_len) { cs_set_err("zstd host decode failed"); csgpu_release(s); return CSTRIPE_ERR_FORMAT; }
+                }
+                memcpy(h_data.data() + dpos, tmp.data(), tmp.size());
+                dpos += align_up(nd.n.decompressed_size, 16);
+            } else {
+                cs_set_err("unsupported chunk compression %d", nd.n.comp_type);
+                csgpu_release(s);
+                return CSTRIPE_ERR_FORMAT;
+            }
+            h_colloc[(uint64_t)gi * n_proj + pj] = cl;
+        }
+    }
+
+    HIP_TRY(hipMemcpyAsync(g->d_data, h_data.data(), data_bytes, hipMemcpyHostToDevice, g->stream));
+    if (!h_rank.empty())
+        HIP_TRY(hipMemcpyAsync(g->d_rank, h_rank.data(), h_rank.size() * 4, hipMemcpyHostToDevice, g->stream));
+    if (!h_segs.empty())
+        HIP_TRY(hipMemcpyAsync(g->d_segs, h_segs.data(), h_segs.size() * sizeof(SegDesc), hipMemcpyHostToDevice, g->stream));
+    HIP_TRY(hipMemcpyAsync(g->d_groups, h_groups.data(), h_groups.size() * sizeof(GroupDesc), hipMemcpyHostToDevice, g->stream));
+    HIP_TRY(hipMemcpyAsync(g->d_colloc, h_colloc.data(), h_colloc.size() * sizeof(ColLoc), hipMemcpyHostToDevice, g->stream));
+    HIP_TRY(hipStreamSynchronize(g->stream));
+    return CSTRIPE_OK;
+}
+
+/* launch the decode grid: LDS-staged variant when every segment fits the
+ * 64 KiB budget (writer default 8 KiB segments -> ~9 waves/CU), else the
+ * global-memory fallback */
+static void launch_decode(cs_gpu_state *g)
+{
+    if (g->n_segs == 0) return;
+    /* lane-parallel path for micro-segments (one lane per segment).
+     * stride: 16 B-multiple, (stride/4)%64 != 0 so equal-progress lanes land
+     * on different banks; LDS/wave = 64*stride -> waves/CU 9 / 4 / 2. */
+    if (g->segs_16aligned && g->max_seg_dlen <= 1039) {
+        uint32_t block, stride;
+        if (g->max_seg_dlen <= 271)      { block = 256; stride = 272; }
+        else if (g->max_seg_dlen <= 527) { block = 256; stride = 528; }
+        else                             { block = 128; stride = 1040; }
+        uint32_t grid = (g->n_segs + block - 1) / block;
+        hipLaunchKernelGGL(lz4_decode_lane_kernel, dim3(grid), dim3(block),
+                           block * stride + block * 16, g->stream,
+                           g->d_data, g->d_scratch, g->d_segs, g->n_segs, stride,
+                           g->d_error);
+        return;
+    }
+    uint32_t in_cap = (g->max_seg_comp + 8 + 15) & ~15u;
+    uint32_t out_cap = (g->max_seg_dlen + 15) & ~15u;
+    if (in_cap + out_cap <= 64 * 1024) {
+        hipLaunchKernelGGL(lz4_decode_lds_kernel, dim3(g->n_segs), dim3(WAVE),
+                           in_cap + out_cap, g->stream,
+                           g->d_data, g->d_scratch, g->d_segs, in_cap, g->d_error);
+    } else {
+        hipLaunchKernelGGL(lz4_decode_kernel, dim3(g->n_segs), dim3(WAVE), 0, g->stream,
+                           g->d_data, g->d_scratch, g->d_segs, g->d_error);
+    }
+}
+
+/* =====================================================================
+ * scan_agg
+ * ===================================================================== */
+
+int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
+              const uint32_t *group_cols, uint32_t n_group_cols,
+              cstripe_group_result *gr, cstripe_partial *out)
+{
+    if (!s->gpu) { cs_set_err("scan not staged — call cstripe_gpu_stage first (GPU required; no CPU fallback)"); return CSTRIPE_ERR_NOGPU; }
+    cs_gpu_state *g = s->gpu;
+    cstripe_reader *r = s->r;
+    if (n_aggs > MAX_AGGS) { cs_set_err("too many aggs"); return CSTRIPE_ERR_ARG; }
+    if (s->preds.size() > MAX_PREDS) { cs_set_err("too many preds"); return CSTRIPE_ERR_ARG; }
+    if (n_group_cols > 0) { cs_set_err("grouped aggregation not implemented yet"); (void)group_cols; (void)gr; return CSTRIPE_ERR_ARG; }
+
+    AggParams p{};
+    p.n_preds = (uint32_t)s->preds.size();
+    p.n_aggs = n_aggs;
+    p.n_proj = g->n_proj;
+    p.tiles_per_group = (r->head.chunk_row_limit + TILE_ROWS - 1) / TILE_ROWS;
+    if (p.tiles_per_group == 0) p.tiles_per_group = 1;
+    p.n_groups = g->n_groups;
+    for (uint32_t i = 0; i < p.n_preds; i++) {
+        const cstripe_pred &q = s->preds[i];
+        p.preds[i].proj = (uint16_t)g->proj_of_col[q.column];
+        p.preds[i].op = (uint8_t)q.op;
+        uint8_t t = r->cols[q.column].type;
+        p.preds[i].is_float = (t == CSTRIPE_F32 || t == CSTRIPE_F64) ? 1 : 0;
+        p.preds[i].ival = q.ival;
+        p.preds[i].fval = q.fval;
+    }
+    auto proj_of = [&](int32_t col) -> int {
+        if (col < 0 || col >= (int32_t)r->head.column_count) return -1;
+        return g->proj_of_col[col];
+    };
+    for (uint32_t i = 0; i < n_aggs; i++) {
+        p.aggs[i].kind = (uint8_t)aggs[i].kind;
+        p.aggs[i].one = aggs[i].one;
+        int pa = proj_of(aggs[i].col_a), pb = proj_of(aggs[i].col_b), pc = proj_of(aggs[i].col_c);
+        if (aggs[i].kind != CSTRIPE_AGG_COUNT_STAR && pa < 0) { cs_set_err("agg %u: col_a not projected", i); return CSTRIPE_ERR_ARG; }
+        p.aggs[i].proj_a = (uint8_t)(pa < 0 ? 0 : pa);
+        p.aggs[i].proj_b = (uint8_t)(pb < 0 ? 0 : pb);
+        p.aggs[i].proj_c = (uint8_t)(pc < 0 ? 0 : pc);
+    }
+
+    uint32_t n_blocks = g->n_groups * p.tiles_per_group;
+    if (n_blocks == 0) {
+        /* empty selection — all-NULL partials (combine applies COUNT coalesce) */
+        for (uint32_t a = 0; a < n_aggs; a++) {
+            cstripe_partial z{};
+            z.is_null = 1;
+            if (aggs[a].kind == CSTRIPE_AGG_COUNT_STAR || aggs[a].kind == CSTRIPE_AGG_COUNT_COL) z.is_null = 0;
+            out[a] = z;
+        }
+        s->last_kernel_ms = s->last_decode_ms = s->last_agg_ms = 0;
+        return CSTRIPE_OK;
+    }
+
+    HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
+    HIP_TRY(hipEventRecord(g->ev0, g->stream));
+    launch_decode(g);
+    HIP_TRY(hipEventRecord(g->ev1, g->stream));
+    hipLaunchKernelGGL(filter_agg_kernel, dim3(n_blocks), dim3(AGG_BLOCK), 0, g->stream,
+                       g->d_data, g->d_scratch, g->d_rank, g->d_groups, g->d_colloc,
+                       g->d_block, p);
+    hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
+                       g->d_block, n_blocks, g->d_final, p);
+    HIP_TRY(hipEventRecord(g->ev2, g->stream));
+
+    AccCell h_final[MAX_AGGS];
+    int h_err = 0;
+    HIP_TRY(hipMemcpyAsync(h_final, g->d_final, n_aggs * sizeof(AccCell), hipMemcpyDeviceToHost, g->stream));
+    HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
+    HIP_TRY(hipStreamSynchronize(g->stream));
+    if (h_err) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+
+    float ms_decode = 0, ms_agg = 0;
+    hipEventElapsedTime(&ms_decode, g->ev0, g->ev1);
+    hipEventElapsedTime(&ms_agg, g->ev1, g->ev2);
+    s->last_decode_ms = ms_decode;
+    s->last_agg_ms = ms_agg;
+    s->last_kernel_ms = ms_decode + ms_agg;
+
+    for (uint32_t a = 0; a < n_aggs; a++) {
+        cstripe_partial o{};
+        const AccCell &c = h_final[a];
+        o.count = c.cnt;
+        o.is_null = (c.cnt == 0) ? 1 : 0;
+        if (o.is_null && aggs[a].kind != CSTRIPE_AGG_COUNT_STAR &&
+            aggs[a].kind != CSTRIPE_AGG_COUNT_COL) { out[a] = o; continue; }
+        switch (aggs[a].kind) {
+            case CSTRIPE_AGG_COUNT_STAR:
+            case CSTRIPE_AGG_COUNT_COL:
+                o.count = c.cnt;
+                o.i128_lo = c.cnt;
+                o.is_null = 0;
+                break;
+            case CSTRIPE_AGG_SUM_F64:
+            case CSTRIPE_AGG_MIN_F64:
+            case CSTRIPE_AGG_MAX_F64:
+                o.f64 = c.f;
+                break;
+            case CSTRIPE_AGG_MIN_I64:
+            case CSTRIPE_AGG_MAX_I64:
+                o.i128_lo = c.lo;
+                o.i128_hi = c.lo < 0 ? -1 : 0;
+                break;
+            default:
+                o.i128_lo = c.lo;
+                o.i128_hi = c.hi;
+                break;
+        }
+        out[a] = o;
+    }
+    return CSTRIPE_OK;
+}
+
+/* =====================================================================
+ * next_batch — parity/fallback access: GPU-decode the next surviving chunk
+ * group, copy decoded value streams back, expand to row-aligned on host
+ * using the exists bitmap (ReadChunkGroupNextRow contract,
+ * columnar_reader.c:868-901, batched).
+ * ===================================================================== */
+
+int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
+{
+    if (!s->gpu) { cs_set_err("scan not staged — call cstripe_gpu_stage first"); return CSTRIPE_ERR_NOGPU; }
+    cs_gpu_state *g = s->gpu;
+    cstripe_reader *r = s->r;
+    if (s->batch_pos >= s->sel.size()) return CSTRIPE_END;
+
+    /* make sure scratch holds decoded data (decode everything once per rewind) */
+    if (g->n_segs > 0 && s->batch_pos == 0) {
+        HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
+        launch_decode(g);
+        int h_err = 0;
+        HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
+        HIP_TRY(hipStreamSynchronize(g->stream));
+        if (h_err) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+    }
+
+    uint32_t gi = (uint32_t)s->batch_pos;
+    const cs_selchunk &sc = s->sel[gi];
+    const cs_stripe_info &st = r->stripes[sc.stripe];
+    uint32_t rows = st.group_rows[sc.chunk];
+    const uint8_t *stripe_base = r->map + st.meta.file_offset;
+
+    std::vector<uint8_t> packed;
+    for (uint32_t c = 0; c < r->head.column_count; c++) {
+        int pj = g->proj_of_col[c];
+        if (batch->col_values && batch->col_values[c] == nullptr) continue;
+        if (pj < 0) continue;
+        const cs_skipnode &nd = st.nodes[c][sc.chunk];
+        uint32_t width = csf_type_width(r->cols[c].type);
+        uint64_t soff = g->scratch_off[(uint64_t)gi * g->n_proj + pj];
+
+        packed.resize(nd.n.decompressed_size);
+        if (soff != ~0ull) {  /* LZ4-decoded on device */
+            HIP_TRY(hipMemcpyAsync(packed.data(), g->d_scratch + soff, nd.n.decompressed_size,
+                                   hipMemcpyDeviceToHost, g->stream));
+            HIP_TRY(hipStreamSynchronize(g->stream));
+        } else {
+            /* NONE / host-predecoded zstd: take staged raw from file side */
+            if (nd.n.comp_type == CSTRIPE_COMP_NONE) {
+                memcpy(packed.data(), stripe_base + nd.n.value_off, nd.n.decompressed_size);
+            } else { /* zstd: decode host-side as at stage */
+                for (const csf_seg &sg : st.nodes[c][sc.chunk].segs) {
+                    size_t zr = ZSTD_decompress(packed.data() + sg.decomp_off, sg.decomp_len,
+                                                stripe_base + nd.n.value_off + sg.comp_off, sg.comp_len);
+                    if (ZSTD_isError(zr) || zr != sg.decomp_len) { cs_set_err("zstd host decode failed"); return CSTRIPE_ERR_FORMAT; }
+                }
+            }
+        }
+
+        /* expand packed -> row-aligned using the exists bitmap */
+        uint8_t *dstv = (uint8_t *)batch->col_values[c];
+        uint8_t *dstn = batch->col_nulls ? batch->col_nulls[c] : nullptr;
+        const uint8_t *eb = stripe_base + nd.n.exists_off;
+        uint32_t vidx = 0;
+        if (nd.n.n_present == nd.n.row_count) {
+            memcpy(dstv, packed.data(), (size_t)rows * width);
+            if (dstn) memset(dstn, 0, rows);
+        } else {
+            for (uint32_t i = 0; i < rows; i++) {
+                bool present = (eb[i / 8] >> (i % 8)) & 1;
+                if (present) {
+                    memcpy(dstv + (size_t)i * width, packed.data() + (size_t)vidx * width, width);
+                    vidx++;
+                } else {
+                    memset(dstv + (size_t)i * width, 0, width);
+                }
+                if (dstn) dstn[i] = present ? 0 : 1;
+            }
+        }
+    }
+
+    batch->n_rows = rows;
+    uint64_t first = st.meta.first_row_number;
+    for (uint32_t k = 0; k < sc.chunk; k++) first += st.group_rows[k];
+    batch->first_row_number = first;
+    s->batch_pos++;
+    return CSTRIPE_OK;
+}
