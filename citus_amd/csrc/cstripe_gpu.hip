@@ -28,6 +28,7 @@
 #include <cstring>
 #include <cstdio>
 #include <vector>
+#include <algorithm>
 
 #define WAVE 64
 #define AGG_BLOCK 256
@@ -99,6 +100,11 @@ struct cs_gpu_state {
     ColLoc *d_colloc = nullptr;
     AccCell *d_block = nullptr;
     AccCell *d_final = nullptr;
+    uint16_t *d_gkeys = nullptr;     /* grouped: per-block key tables */
+    AccCell *d_gcells = nullptr;
+    uint16_t *d_gfkeys = nullptr;    /* grouped: final merged groups */
+    AccCell *d_gfcells = nullptr;
+    uint32_t *d_gn = nullptr;
     int *d_error = nullptr;
     hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr;
 
@@ -115,6 +121,7 @@ struct cs_gpu_state {
     uint8_t proj_type[MAX_PROJ];
     /* host-side mirror for next_batch (exists read from mmap) */
     std::vector<uint64_t> scratch_off;  /* per (sel,proj): decomp offset */
+    std::vector<ColLoc> colloc_host;
 };
 
 /* ---------------- error helper ---------------- */
@@ -602,6 +609,197 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     }
 }
 
+/* =====================================================================
+ * Grouped aggregation (TPC-H Q1 shape): GROUP BY 1-2 categorical i8
+ * columns, few distinct groups. The reference plan is worker HashAggregate
+ * over ColumnarScan + coordinator merge per group (multi_explain.out:80-82,
+ * multi_logical_optimizer.c:1807-1885). Here: per-WAVE LDS accumulator
+ * tables (no atomics); rows reduce wave-cooperatively one distinct key at
+ * a time (ballot + predicated identity + shuffle reduce); grid-stride
+ * blocks flush one compact table each; a final single-block kernel merges.
+ * ===================================================================== */
+
+#define GRP_SLOTS 32          /* distinct keys per wave table */
+#define GRP_GRID  1024        /* grid-stride blocks */
+
+struct GroupParams {
+    AggParams base;
+    uint32_t n_group_cols;
+    uint32_t gproj[CSTRIPE_MAX_GROUP_COLS];   /* projected slots of key cols */
+    uint32_t n_work;                          /* n_groups * tiles_per_group */
+};
+
+__global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
+    const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
+    const uint32_t *__restrict__ rank, const GroupDesc *__restrict__ groups,
+    const ColLoc *__restrict__ colloc, uint16_t *__restrict__ keys_out,
+    AccCell *__restrict__ cells_out, int *__restrict__ err,
+    const GroupParams gp)
+{
+    const AggParams &params = gp.base;
+    const uint32_t wid = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    const uint32_t n_waves = AGG_BLOCK / WAVE;
+
+    extern __shared__ uint8_t gsh[];
+    uint16_t *wkeys = (uint16_t *)gsh;                          /* [wave][S] */
+    ThreadAcc *wacc = (ThreadAcc *)(gsh + n_waves * GRP_SLOTS * sizeof(uint16_t));
+    /* wacc[((wave*S)+slot)*n_aggs + a] */
+
+    for (uint32_t i = threadIdx.x; i < n_waves * GRP_SLOTS; i += AGG_BLOCK)
+        wkeys[i] = 0xFFFF;
+    __syncthreads();
+
+    uint16_t *mykeys = wkeys + wid * GRP_SLOTS;
+    ThreadAcc *myacc = wacc + (size_t)wid * GRP_SLOTS * params.n_aggs;
+    uint32_t used = 0;       /* wave-uniform slot count (updated by lane 0 path) */
+
+    for (uint32_t work = blockIdx.x; work < gp.n_work; work += gridDim.x) {
+        const uint32_t gid = work / params.tiles_per_group;
+        const uint32_t tile = work % params.tiles_per_group;
+        const GroupDesc g = groups[gid];
+        const ColLoc *cols = colloc + g.colbase;
+        uint32_t row_start = tile * TILE_ROWS;
+        uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
+
+        for (uint32_t base_row = row_start; base_row < row_end; base_row += WAVE) {
+            uint32_t row = base_row + lane;
+            bool pass = row < row_end;
+            if (pass) {
+                int last_proj = -1;
+                int64_t liv = 0; double lfv = 0; bool lok = false;
+                for (uint32_t p = 0; p < params.n_preds; p++) {
+                    const PredD &pr = params.preds[p];
+                    if ((int)pr.proj != last_proj) {
+                        lok = col_value(data, scratch, rank, cols[pr.proj], row, liv, lfv);
+                        last_proj = (int)pr.proj;
+                    }
+                    pass = pass & (lok && pred_eval(pr, liv, lfv));
+                }
+            }
+            uint32_t key = 0;
+            if (pass) {
+                int64_t kv; double kf;
+                for (uint32_t gc = 0; gc < gp.n_group_cols; gc++) {
+                    col_value(data, scratch, rank, cols[gp.gproj[gc]], row, kv, kf);
+                    key |= ((uint32_t)kv & 0xFF) << (8 * gc);
+                }
+            }
+            /* wave-cooperative reduce, one distinct key per round */
+            uint64_t remaining = __ballot(pass);
+            while (remaining) {
+                int leader = __ffsll((unsigned long long)remaining) - 1;
+                uint32_t kk = (uint32_t)__shfl((int)key, leader, WAVE);
+                bool mine = pass && key == kk;
+                /* find/claim slot (wave-uniform) */
+                uint32_t slot = 0xFFFFFFFF;
+                for (uint32_t sidx = 0; sidx < used; sidx++)
+                    if (mykeys[sidx] == (uint16_t)kk) { slot = sidx; break; }
+                if (slot == 0xFFFFFFFF) {
+                    if (used >= GRP_SLOTS) {          /* overflow: flag + drop */
+                        if (lane == 0) atomicOr(err, 8);
+                        remaining &= ~__ballot(mine);
+                        continue;
+                    }
+                    slot = used++;
+                    if (lane == 0) mykeys[slot] = (uint16_t)kk;
+                    for (uint32_t a = 0; a < params.n_aggs; a++) {
+                        ThreadAcc z;
+                        acc_init(z, params.aggs[a].kind);
+                        if (lane == 0) myacc[slot * params.n_aggs + a] = z;
+                    }
+                }
+                for (uint32_t a = 0; a < params.n_aggs; a++) {
+                    ThreadAcc t;
+                    acc_init(t, params.aggs[a].kind);
+                    if (mine)
+                        acc_row(t, params.aggs[a], data, scratch, rank, cols, row);
+                    wave_reduce(t, params.aggs[a].kind);
+                    if (lane == 0) {
+                        ThreadAcc cur = myacc[slot * params.n_aggs + a];
+                        acc_merge(cur, t, params.aggs[a].kind);
+                        myacc[slot * params.n_aggs + a] = cur;
+                    }
+                }
+                remaining &= ~__ballot(mine);
+            }
+        }
+    }
+    __syncthreads();
+
+    /* merge the block's wave tables into one compact list; write to global.
+     * wave 0 lane 0 does it serially — tables are tiny. */
+    if (threadIdx.x == 0) {
+        uint16_t *bk = keys_out + (size_t)blockIdx.x * (n_waves * GRP_SLOTS);
+        AccCell *bc = cells_out + (size_t)blockIdx.x * (n_waves * GRP_SLOTS) * params.n_aggs;
+        uint32_t n = 0;
+        for (uint32_t w = 0; w < n_waves; w++) {
+            for (uint32_t sidx = 0; sidx < GRP_SLOTS; sidx++) {
+                uint16_t k = wkeys[w * GRP_SLOTS + sidx];
+                if (k == 0xFFFF) continue;
+                uint32_t at = n;
+                for (uint32_t j = 0; j < n; j++) if (bk[j] == k) { at = j; break; }
+                ThreadAcc *src = wacc + ((size_t)(w * GRP_SLOTS) + sidx) * params.n_aggs;
+                if (at == n) {
+                    bk[n] = k;
+                    for (uint32_t a = 0; a < params.n_aggs; a++) {
+                        AccCell c{src[a].lo, src[a].hi, src[a].f, src[a].cnt};
+                        bc[(size_t)n * params.n_aggs + a] = c;
+                    }
+                    n++;
+                } else {
+                    for (uint32_t a = 0; a < params.n_aggs; a++) {
+                        AccCell c = bc[(size_t)at * params.n_aggs + a];
+                        ThreadAcc cur{c.lo, c.hi, c.f, c.cnt};
+                        acc_merge(cur, src[a], params.aggs[a].kind);
+                        AccCell o{cur.lo, cur.hi, cur.f, cur.cnt};
+                        bc[(size_t)at * params.n_aggs + a] = o;
+                    }
+                }
+            }
+        }
+        for (uint32_t j = n; j < n_waves * GRP_SLOTS; j++) bk[j] = 0xFFFF;
+    }
+}
+
+/* single-block merge of per-block group tables -> final <=64 groups */
+__global__ __launch_bounds__(64) void grouped_final_kernel(
+    const uint16_t *__restrict__ keys_in, const AccCell *__restrict__ cells_in,
+    uint32_t n_blocks, uint32_t per_block, uint16_t *__restrict__ keys_out,
+    AccCell *__restrict__ cells_out, uint32_t *__restrict__ n_groups_out,
+    int *__restrict__ err, const GroupParams gp)
+{
+    if (threadIdx.x != 0) return;      /* serial: tables are tiny */
+    const AggParams &params = gp.base;
+    uint32_t n = 0;
+    for (uint32_t b = 0; b < n_blocks; b++) {
+        for (uint32_t j = 0; j < per_block; j++) {
+            uint16_t k = keys_in[(size_t)b * per_block + j];
+            if (k == 0xFFFF) continue;
+            uint32_t at = n;
+            for (uint32_t i = 0; i < n; i++) if (keys_out[i] == k) { at = i; break; }
+            const AccCell *src = cells_in + ((size_t)b * per_block + j) * params.n_aggs;
+            if (at == n) {
+                if (n >= CSTRIPE_MAX_GROUPS) { atomicOr(err, 16); return; }
+                keys_out[n] = k;
+                for (uint32_t a = 0; a < params.n_aggs; a++)
+                    cells_out[(size_t)n * params.n_aggs + a] = src[a];
+                n++;
+            } else {
+                for (uint32_t a = 0; a < params.n_aggs; a++) {
+                    AccCell c = cells_out[(size_t)at * params.n_aggs + a];
+                    ThreadAcc cur{c.lo, c.hi, c.f, c.cnt};
+                    ThreadAcc s{src[a].lo, src[a].hi, src[a].f, src[a].cnt};
+                    acc_merge(cur, s, params.aggs[a].kind);
+                    AccCell o{cur.lo, cur.hi, cur.f, cur.cnt};
+                    cells_out[(size_t)at * params.n_aggs + a] = o;
+                }
+            }
+        }
+    }
+    *n_groups_out = n;
+}
+
 /* device-wide final reduce over block partials: one block, grid-stride */
 __global__ __launch_bounds__(AGG_BLOCK) void final_reduce_kernel(
     const AccCell *__restrict__ block_in, uint32_t n_blocks,
@@ -650,6 +848,11 @@ void csgpu_release(cstripe_scan *s)
     if (g->d_colloc) hipFree(g->d_colloc);
     if (g->d_block) hipFree(g->d_block);
     if (g->d_final) hipFree(g->d_final);
+    if (g->d_gkeys) hipFree(g->d_gkeys);
+    if (g->d_gcells) hipFree(g->d_gcells);
+    if (g->d_gfkeys) hipFree(g->d_gfkeys);
+    if (g->d_gfcells) hipFree(g->d_gfcells);
+    if (g->d_gn) hipFree(g->d_gn);
     if (g->d_error) hipFree(g->d_error);
     if (g->ev0) hipEventDestroy(g->ev0);
     if (g->ev1) hipEventDestroy(g->ev1);
@@ -849,6 +1052,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     HIP_TRY(hipMemcpyAsync(g->d_groups, h_groups.data(), h_groups.size() * sizeof(GroupDesc), hipMemcpyHostToDevice, g->stream));
     HIP_TRY(hipMemcpyAsync(g->d_colloc, h_colloc.data(), h_colloc.size() * sizeof(ColLoc), hipMemcpyHostToDevice, g->stream));
     HIP_TRY(hipStreamSynchronize(g->stream));
+    g->colloc_host = h_colloc;
     return CSTRIPE_OK;
 }
 
@@ -898,7 +1102,6 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
     cstripe_reader *r = s->r;
     if (n_aggs > MAX_AGGS) { cs_set_err("too many aggs"); return CSTRIPE_ERR_ARG; }
     if (s->preds.size() > MAX_PREDS) { cs_set_err("too many preds"); return CSTRIPE_ERR_ARG; }
-    if (n_group_cols > 0) { cs_set_err("grouped aggregation not implemented yet"); (void)group_cols; (void)gr; return CSTRIPE_ERR_ARG; }
 
     AggParams p{};
     p.n_preds = (uint32_t)s->preds.size();
@@ -928,6 +1131,104 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         p.aggs[i].proj_a = (uint8_t)(pa < 0 ? 0 : pa);
         p.aggs[i].proj_b = (uint8_t)(pb < 0 ? 0 : pb);
         p.aggs[i].proj_c = (uint8_t)(pc < 0 ? 0 : pc);
+    }
+
+    if (n_group_cols > 0) {
+        GroupParams gp{};
+        gp.base = p;
+        gp.n_group_cols = n_group_cols;
+        for (uint32_t i = 0; i < n_group_cols; i++) {
+            int pj = proj_of((int32_t)group_cols[i]);
+            if (pj < 0) { cs_set_err("group col %u not projected", group_cols[i]); return CSTRIPE_ERR_ARG; }
+            if (r->cols[group_cols[i]].type != CSTRIPE_I8) { cs_set_err("group col %u must be I8", group_cols[i]); return CSTRIPE_ERR_ARG; }
+            gp.gproj[i] = (uint32_t)pj;
+            /* NULL group keys unsupported this round: require dense key cols */
+            for (uint32_t gi = 0; gi < g->n_groups; gi++)
+                if (!(g->colloc_host[(uint64_t)gi * g->n_proj + pj].flags & 2)) {
+                    cs_set_err("grouped aggregation requires non-NULL group key columns");
+                    return CSTRIPE_ERR_ARG;
+                }
+        }
+        gp.n_work = g->n_groups * p.tiles_per_group;
+        if (gp.n_work == 0) {
+            gr->n_groups = 0;
+            s->last_kernel_ms = s->last_decode_ms = s->last_agg_ms = 0;
+            return CSTRIPE_OK;
+        }
+        const uint32_t n_waves = AGG_BLOCK / WAVE;
+        const uint32_t per_block = n_waves * GRP_SLOTS;
+        uint32_t grid = gp.n_work < GRP_GRID ? gp.n_work : GRP_GRID;
+        if (!g->d_gkeys) {
+            HIP_TRY(hipMalloc(&g->d_gkeys, (uint64_t)GRP_GRID * per_block * 2));
+            HIP_TRY(hipMalloc(&g->d_gcells, (uint64_t)GRP_GRID * per_block * MAX_AGGS * sizeof(AccCell)));
+            HIP_TRY(hipMalloc(&g->d_gfkeys, CSTRIPE_MAX_GROUPS * 2));
+            HIP_TRY(hipMalloc(&g->d_gfcells, (uint64_t)CSTRIPE_MAX_GROUPS * MAX_AGGS * sizeof(AccCell)));
+            HIP_TRY(hipMalloc(&g->d_gn, 4));
+        }
+        uint32_t lds = n_waves * GRP_SLOTS * 2 + n_waves * GRP_SLOTS * n_aggs * (uint32_t)sizeof(ThreadAcc);
+        HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
+        HIP_TRY(hipEventRecord(g->ev0, g->stream));
+        launch_decode(g);
+        HIP_TRY(hipEventRecord(g->ev1, g->stream));
+        hipLaunchKernelGGL(grouped_agg_kernel, dim3(grid), dim3(AGG_BLOCK), lds, g->stream,
+                           g->d_data, g->d_scratch, g->d_rank, g->d_groups, g->d_colloc,
+                           g->d_gkeys, g->d_gcells, g->d_error, gp);
+        hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(64), 0, g->stream,
+                           g->d_gkeys, g->d_gcells, grid, per_block,
+                           g->d_gfkeys, g->d_gfcells, g->d_gn, g->d_error, gp);
+        HIP_TRY(hipEventRecord(g->ev2, g->stream));
+
+        uint16_t h_keys[CSTRIPE_MAX_GROUPS];
+        std::vector<AccCell> h_cells((size_t)CSTRIPE_MAX_GROUPS * n_aggs);
+        uint32_t h_n = 0;
+        int h_err = 0;
+        HIP_TRY(hipMemcpyAsync(h_keys, g->d_gfkeys, sizeof(h_keys), hipMemcpyDeviceToHost, g->stream));
+        HIP_TRY(hipMemcpyAsync(h_cells.data(), g->d_gfcells, h_cells.size() * sizeof(AccCell), hipMemcpyDeviceToHost, g->stream));
+        HIP_TRY(hipMemcpyAsync(&h_n, g->d_gn, 4, hipMemcpyDeviceToHost, g->stream));
+        HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
+        HIP_TRY(hipStreamSynchronize(g->stream));
+        if (h_err & 4) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+        if (h_err) { cs_set_err("too many distinct groups (device flag %d; caps: %d/wave, %d total)", h_err, GRP_SLOTS, CSTRIPE_MAX_GROUPS); return CSTRIPE_ERR; }
+
+        float ms_decode = 0, ms_agg = 0;
+        hipEventElapsedTime(&ms_decode, g->ev0, g->ev1);
+        hipEventElapsedTime(&ms_agg, g->ev1, g->ev2);
+        s->last_decode_ms = ms_decode;
+        s->last_agg_ms = ms_agg;
+        s->last_kernel_ms = ms_decode + ms_agg;
+
+        /* sort groups by key for deterministic output (oracle does the same) */
+        std::vector<uint32_t> order(h_n);
+        for (uint32_t i = 0; i < h_n; i++) order[i] = i;
+        std::sort(order.begin(), order.end(),
+                  [&](uint32_t a, uint32_t b) { return h_keys[a] < h_keys[b]; });
+        gr->n_groups = h_n;
+        for (uint32_t oi = 0; oi < h_n; oi++) {
+            uint32_t i = order[oi];
+            gr->keys[oi] = h_keys[i];
+            for (uint32_t a = 0; a < n_aggs; a++) {
+                const AccCell &c = h_cells[(size_t)i * n_aggs + a];
+                cstripe_partial o{};
+                o.count = c.cnt;
+                o.is_null = (c.cnt == 0) ? 1 : 0;
+                switch (aggs[a].kind) {
+                    case CSTRIPE_AGG_COUNT_STAR:
+                    case CSTRIPE_AGG_COUNT_COL:
+                        o.i128_lo = c.cnt; o.is_null = 0; break;
+                    case CSTRIPE_AGG_SUM_F64:
+                    case CSTRIPE_AGG_MIN_F64:
+                    case CSTRIPE_AGG_MAX_F64:
+                        if (!o.is_null) o.f64 = c.f; break;
+                    case CSTRIPE_AGG_MIN_I64:
+                    case CSTRIPE_AGG_MAX_I64:
+                        if (!o.is_null) { o.i128_lo = c.lo; o.i128_hi = c.lo < 0 ? -1 : 0; } break;
+                    default:
+                        if (!o.is_null) { o.i128_lo = c.lo; o.i128_hi = c.hi; } break;
+                }
+                out[(size_t)oi * n_aggs + a] = o;
+            }
+        }
+        return CSTRIPE_OK;
     }
 
     uint32_t n_blocks = g->n_groups * p.tiles_per_group;

@@ -226,3 +226,93 @@ def test_decode_kernel_variants(tmp_path, seg):
     op, ofilt, gp, gfilt = both(path, preds, aggs)
     assert ofilt == gfilt
     assert_parity(op, gp, aggs)
+
+
+def q1_aggs():
+    return [(ca.AGG_SUM_I64, 1), (ca.AGG_SUM_I64, 2),
+            (ca.AGG_SUM_DISC_I64, 2, 3, -1, 100),
+            (ca.AGG_SUM_DISC_TAX_I64, 2, 3, 4, 100),
+            (ca.AGG_COUNT_STAR, -1)]
+
+
+@pytest.mark.parametrize("variant", ["lz4", "none"])
+def test_q1_grouped_golden(golden_dir, expected, variant):
+    """TPC-H Q1 GROUP BY (returnflag, linestatus) on device vs the
+    reference's expected result table (multi_tpch_query1.out)."""
+    path = os.path.join(golden_dir, f"lineitem12k_{variant}.cs")
+    q1 = expected["q1"]
+    aggs = q1_aggs()
+    preds = [(5, ca.PRED_LE, q1["shipdate_le"])]
+    with ca.Reader(path) as r, \
+         r.scan(cols_mask=ca.agg_cols_mask(aggs) | (1 << 6) | (1 << 7),
+                preds=preds) as s:
+        s.stage()
+        res = s.agg_grouped(aggs, (6, 7))
+    rf = expected["flag_codes"]["returnflag"]
+    ls = expected["flag_codes"]["linestatus"]
+    assert len(res) == len(q1["groups"])
+    for key, vals in q1["groups"].items():
+        a, b = key.split(",")
+        got = res[(rf[a], ls[b])]
+        assert [got[0].i128, got[1].i128, got[2].i128, got[3].i128,
+                got[4].count] == vals
+    # and vs oracle partial-for-partial
+    with oracle.OracleTable(path) as t:
+        ores, _ = t.scan_agg(preds, aggs, group_cols=(6, 7))
+    for k, parts in ores.items():
+        for i in range(len(aggs)):
+            assert res[k][i].i128 == parts[i].i128
+            assert res[k][i].count == parts[i].count
+
+
+def test_grouped_random_many_groups(tmp_path):
+    """grouped parity on random data with ~40 distinct composite keys,
+    nullable measures, mixed agg kinds."""
+    n = 123_456
+    k0 = RNG.integers(0, 8, n).astype(np.int8)
+    k1 = RNG.integers(0, 5, n).astype(np.int8)
+    v = RNG.integers(-10**6, 10**6, n).astype(np.int64)
+    nv = (RNG.random(n) < 0.2).astype(np.uint8)
+    f = RNG.normal(size=n)
+    path = str(tmp_path / "grp.cs")
+    ca.write_table(path, [("k0", ca.I8, 0), ("k1", ca.I8, 0),
+                          ("v", ca.I64, 0), ("f", ca.F64, 0)],
+                   [k0, k1, v, f], nulls=[None, None, nv, None],
+                   compression=ca.COMP_LZ4, chunk_group_row_limit=3000)
+    aggs = [(ca.AGG_COUNT_STAR, -1), (ca.AGG_SUM_I64, 2),
+            (ca.AGG_MIN_I64, 2), (ca.AGG_MAX_I64, 2), (ca.AGG_SUM_F64, 3)]
+    preds = [(2, ca.PRED_GT, -900000)]
+    with ca.Reader(path) as r, \
+         r.scan(cols_mask=ca.agg_cols_mask(aggs) | 0b11, preds=preds) as s:
+        s.stage()
+        res = s.agg_grouped(aggs, (0, 1))
+    with oracle.OracleTable(path) as t:
+        ores, _ = t.scan_agg(preds, aggs, group_cols=(0, 1))
+    assert set(res.keys()) == set(ores.keys())
+    assert len(res) == 40
+    for k in ores:
+        for i, (kind, *_rest) in enumerate(aggs):
+            assert res[k][i].count == ores[k][i].count, (k, i)
+            assert res[k][i].is_null == ores[k][i].is_null
+            if kind == ca.AGG_SUM_F64:
+                if not ores[k][i].is_null and ores[k][i].f64 != 0:
+                    assert abs(res[k][i].f64 - ores[k][i].f64) <= 1e-6 * abs(ores[k][i].f64)
+            else:
+                assert res[k][i].i128 == ores[k][i].i128, (k, i)
+
+
+def test_grouped_null_keys_rejected(tmp_path):
+    n = 100
+    k = np.zeros(n, dtype=np.int8)
+    nk = np.zeros(n, dtype=np.uint8)
+    nk[5] = 1
+    v = np.arange(n, dtype=np.int64)
+    path = str(tmp_path / "nk.cs")
+    ca.write_table(path, [("k", ca.I8, 0), ("v", ca.I64, 0)], [k, v],
+                   nulls=[nk, None], compression=ca.COMP_LZ4)
+    aggs = [(ca.AGG_SUM_I64, 1)]
+    with ca.Reader(path) as r, \
+         r.scan(cols_mask=ca.agg_cols_mask(aggs) | 1) as s:
+        s.stage()
+        with pytest.raises(ca.CStripeError, match="non-NULL group key"):
+            s.agg_grouped(aggs, (0,))
