@@ -662,7 +662,8 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
         uint32_t row_start = tile * TILE_ROWS;
         uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
 
-        for (uint32_t base_row = row_start; base_row < row_end; base_row += WAVE) {
+        for (uint32_t base_row = row_start + wid * WAVE; base_row < row_end;
+             base_row += WAVE * n_waves) {
             uint32_t row = base_row + lane;
             bool pass = row < row_end;
             if (pass) {
