@@ -619,7 +619,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
  * blocks flush one compact table each; a final single-block kernel merges.
  * ===================================================================== */
 
-#define GRP_SLOTS 32          /* distinct keys per wave table */
+#define GRP_SLOTS CSTRIPE_MAX_GROUPS  /* distinct keys per wave table (a wave may see every group) */
 #define GRP_GRID  1024        /* grid-stride blocks */
 
 struct GroupParams {
