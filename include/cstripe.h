@@ -54,6 +54,10 @@ extern "C" {
 
 #define CSTRIPE_ABI_VERSION 1
 
+/* capacity limits of one scan */
+#define CSTRIPE_MAX_PREDS 8
+#define CSTRIPE_MAX_AGGS  12
+
 /* ---- error codes ---- */
 #define CSTRIPE_OK            0
 #define CSTRIPE_ERR          -1   /* generic; see cstripe_errmsg() */

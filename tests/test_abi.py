@@ -96,3 +96,21 @@ def test_combine_f64_and_count_col():
     out = ca.combine(aggs, parts)
     assert out[0].f64 == 1.25
     assert out[1].count == 3
+
+
+def test_host_c_demo_compiles_and_links(tmp_path):
+    """tools/q6_host.c — the plain-C consumer of the ABI — compiles with gcc
+    (C, not C++) and fails loudly at stage without a GPU."""
+    import subprocess
+    exe = str(tmp_path / "q6_host")
+    subprocess.check_call(
+        ["gcc", "-O2", "-std=c11", os.path.join(REPO, "tools", "q6_host.c"),
+         "-L" + os.path.join(REPO, "citus_amd"), "-lcstripe",
+         "-Wl,-rpath," + os.path.join(REPO, "citus_amd"),
+         "-I" + os.path.join(REPO, "include"), "-o", exe])
+    r = subprocess.run([exe, os.path.join(REPO, "tests", "golden",
+                                          "lineitem12k_lz4.cs")],
+                       capture_output=True, text=True)
+    assert "rows=12000" in r.stdout
+    if not ca.gpu_available():
+        assert r.returncode == 1 and "MI355X" in r.stderr
