@@ -9,6 +9,7 @@
  *             -Wl,-rpath,'$ORIGIN/../citus_amd' -Iinclude -o tools/q6_host
  * Usage:  tools/q6_host <stripe-file> [steps]
  */
+#define _POSIX_C_SOURCE 199309L
 #include <stdio.h>
 #include <stdlib.h>
 #include <time.h>
