@@ -227,7 +227,7 @@ __global__ __launch_bounds__(WAVE) void lz4_decode_kernel(
 
 struct ByteStream {
     const uint8_t *base;    /* 8B-aligned */
-    uint64_t w0, w1, w2;    /* 24-byte window; w2 is ~2 rotations of prefetch */
+    uint64_t w0, w1, w2, w3; /* 32-byte window; ~24 B of refill lookahead */
     uint32_t woff;          /* byte offset of w0 within base */
     uint32_t pos;           /* current byte position (>= initial shift) */
 };
@@ -241,6 +241,7 @@ __device__ inline void bs_init(ByteStream &b, const uint8_t *data, uint64_t off)
     b.w0 = *(const uint64_t *)(b.base);
     b.w1 = *(const uint64_t *)(b.base + 8);
     b.w2 = *(const uint64_t *)(b.base + 16);
+    b.w3 = *(const uint64_t *)(b.base + 24);
 }
 
 __device__ inline void bs_norm(ByteStream &b)
@@ -248,8 +249,9 @@ __device__ inline void bs_norm(ByteStream &b)
     while (b.pos - b.woff >= 8) {
         b.w0 = b.w1;
         b.w1 = b.w2;
+        b.w2 = b.w3;
         b.woff += 8;
-        b.w2 = *(const uint64_t *)(b.base + b.woff + 16);
+        b.w3 = *(const uint64_t *)(b.base + b.woff + 24);
     }
 }
 
@@ -552,12 +554,17 @@ __device__ inline void wave_reduce(ThreadAcc &a, uint8_t kind)
     }
 }
 
+template <int NPREDS, int NAGGS>
 __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
     const uint32_t *__restrict__ rank, const GroupDesc *__restrict__ groups,
     const ColLoc *__restrict__ colloc, AccCell *__restrict__ block_out,
     const AggParams params)
 {
+    /* compile-time trip counts (<0 = runtime) unroll the pred/agg loops so
+     * every load is straight-line and pipelines (cdna guide §5.4 trap c) */
+    const uint32_t n_preds = NPREDS >= 0 ? (uint32_t)NPREDS : params.n_preds;
+    const uint32_t n_aggs = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
     const uint32_t gid = blockIdx.x / params.tiles_per_group;
     const uint32_t tile = blockIdx.x % params.tiles_per_group;
     const GroupDesc g = groups[gid];
@@ -567,7 +574,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
 
     ThreadAcc acc[MAX_AGGS];
-    for (uint32_t a = 0; a < params.n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
+    for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
 
     for (uint32_t row = row_start + threadIdx.x; row < row_end; row += AGG_BLOCK) {
         bool pass = true;
@@ -575,7 +582,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
         int64_t liv = 0; double lfv = 0; bool lok = false;
         /* no short-circuit: loads stay control-independent so they issue
          * back-to-back and pipeline instead of chaining load->wait->branch */
-        for (uint32_t p = 0; p < params.n_preds; p++) {
+        for (uint32_t p = 0; p < n_preds; p++) {
             const PredD &pr = params.preds[p];
             if ((int)pr.proj != last_proj) {   /* BETWEEN reuses the load */
                 lok = col_value(data, scratch, rank, cols[pr.proj], row, liv, lfv);
@@ -584,7 +591,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
             pass = pass & (lok && pred_eval(pr, liv, lfv));
         }
         if (!pass) continue;
-        for (uint32_t a = 0; a < params.n_aggs; a++)
+        for (uint32_t a = 0; a < n_aggs; a++)
             acc_row(acc[a], params.aggs[a], data, scratch, rank, cols, row);
     }
 
@@ -592,19 +599,19 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     __shared__ ThreadAcc lds[AGG_BLOCK / WAVE][MAX_AGGS];
     const uint32_t wid = threadIdx.x / WAVE;
     const uint32_t lane = threadIdx.x % WAVE;
-    for (uint32_t a = 0; a < params.n_aggs; a++) {
+    for (uint32_t a = 0; a < n_aggs; a++) {
         wave_reduce(acc[a], params.aggs[a].kind);
         if (lane == 0) lds[wid][a] = acc[a];
     }
     __syncthreads();
     if (wid == 0) {
-        for (uint32_t a = lane; a < params.n_aggs; a += WAVE) {
+        for (uint32_t a = lane; a < n_aggs; a += WAVE) {
             ThreadAcc r = lds[0][a];
             for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
                 acc_merge(r, lds[w][a], params.aggs[a].kind);
             AccCell c;
             c.lo = r.lo; c.hi = r.hi; c.f = r.f; c.cnt = r.cnt;
-            block_out[(uint64_t)blockIdx.x * params.n_aggs + a] = c;
+            block_out[(uint64_t)blockIdx.x * n_aggs + a] = c;
         }
     }
 }
@@ -932,7 +939,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     HIP_TRY(hipEventCreate(&g->ev0));
     HIP_TRY(hipEventCreate(&g->ev1));
     HIP_TRY(hipEventCreate(&g->ev2));
-    data_bytes += 16;               /* register-window read slack */
+    data_bytes += 32;               /* register-window read slack */
     if (data_bytes) HIP_TRY(hipMalloc(&g->d_data, data_bytes));
     if (scratch_bytes) HIP_TRY(hipMalloc(&g->d_scratch, scratch_bytes));
     if (rank_words) HIP_TRY(hipMalloc(&g->d_rank, rank_words * 4));
@@ -1249,9 +1256,19 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
     HIP_TRY(hipEventRecord(g->ev0, g->stream));
     launch_decode(g);
     HIP_TRY(hipEventRecord(g->ev1, g->stream));
-    hipLaunchKernelGGL(filter_agg_kernel, dim3(n_blocks), dim3(AGG_BLOCK), 0, g->stream,
-                       g->d_data, g->d_scratch, g->d_rank, g->d_groups, g->d_colloc,
-                       g->d_block, p);
+    {
+        auto launch = [&](auto *kern) {
+            hipLaunchKernelGGL(kern, dim3(n_blocks), dim3(AGG_BLOCK), 0, g->stream,
+                               g->d_data, g->d_scratch, g->d_rank, g->d_groups,
+                               g->d_colloc, g->d_block, p);
+        };
+        /* hot shapes compiled with unrolled pred/agg loops */
+        if (p.n_preds == 5 && n_aggs == 2) launch(filter_agg_kernel<5, 2>);
+        else if (p.n_preds == 5 && n_aggs == 1) launch(filter_agg_kernel<5, 1>);
+        else if (p.n_preds == 1 && n_aggs == 1) launch(filter_agg_kernel<1, 1>);
+        else if (p.n_preds == 2 && n_aggs == 2) launch(filter_agg_kernel<2, 2>);
+        else launch(filter_agg_kernel<-1, -1>);
+    }
     hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
                        g->d_block, n_blocks, g->d_final, p);
     HIP_TRY(hipEventRecord(g->ev2, g->stream));
