@@ -311,9 +311,22 @@ __global__ void lz4_decode_lane_kernel(
             if (offset == 0 || offset > op || op + mlen > dlen) { bad = true; break; }
             /* uniform per-lane byte copy — control flow stays convergent
              * across the wave (a branchier word-copy variant measured 2x the
-             * per-byte instructions from divergence; profiles/r01) */
+             * per-byte instructions from divergence; profiles/r01).
+             * 4-wide manual unroll: the 4 loads are independent and issue
+             * together instead of load-wait-store per byte. Overlap safety:
+             * offset<4 would make loads of a quad depend on its own stores,
+             * so quads only when offset>=4; sources are then complete. */
             const uint8_t *msrc = sout + op - offset;
-            for (uint32_t j = 0; j < mlen; j++) sout[op + j] = msrc[j];
+            uint32_t j = 0;
+            if (offset >= 4) {
+                for (; j + 4 <= mlen; j += 4) {
+                    uint8_t b0 = msrc[j], b1 = msrc[j + 1];
+                    uint8_t b2 = msrc[j + 2], b3 = msrc[j + 3];
+                    sout[op + j] = b0; sout[op + j + 1] = b1;
+                    sout[op + j + 2] = b2; sout[op + j + 3] = b3;
+                }
+            }
+            for (; j < mlen; j++) sout[op + j] = msrc[j];
             op += mlen;
         }
         if (bad || op != dlen) atomicOr(err, 4);
