@@ -1065,12 +1065,6 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         }
     }
 
-    /* sort segments by compressed length (descending): a decode block's
-     * lanes get similar-sized work, so the block-level __syncthreads tail
-     * is short, and heavy blocks launch first */
-    std::sort(h_segs.begin(), h_segs.end(),
-              [](const SegDesc &a, const SegDesc &b) { return a.comp_len > b.comp_len; });
-
     HIP_TRY(hipMemcpyAsync(g->d_data, h_data.data(), data_bytes, hipMemcpyHostToDevice, g->stream));
     if (!h_rank.empty())
         HIP_TRY(hipMemcpyAsync(g->d_rank, h_rank.data(), h_rank.size() * 4, hipMemcpyHostToDevice, g->stream));
