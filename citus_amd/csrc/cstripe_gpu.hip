@@ -337,23 +337,27 @@ __global__ void lz4_decode_lane_kernel(
     }
     __syncthreads();
 
-    /* cooperative coalesced flush: flat element index over the block's
-     * segments, stride/4 u32 slots per segment (stride is 4-mod-8 so byte
-     * ops in the decode loop are bank-conflict-free; u32 keeps alignment) */
-    const uint32_t eps = stride >> 2;                 /* elements per segment */
+    /* cooperative coalesced flush: 16 B global stores (full coalescing)
+     * assembled from four 4 B LDS reads — the region stride is a 4-mod-8
+     * multiple so the decode loop's byte ops are bank-conflict-free, which
+     * makes LDS-side 16 B alignment unavailable */
+    const uint32_t eps = (stride + 15) >> 4;          /* 16B slots per segment */
     const uint32_t in_block = min(blockDim.x, n_segs - first);
     const uint32_t total = in_block * eps;
     for (uint32_t f = tid; f < total; f += blockDim.x) {
         const uint32_t sidx = f / eps;
-        const uint32_t boff = (f % eps) << 2;
+        const uint32_t boff = (f % eps) << 4;
         const uint32_t dl = sdlen[sidx];
         if (boff >= dl) continue;
         uint8_t *dst = scratch + sdst[sidx] + boff;
-        const uint8_t *ls = sout_all + (size_t)sidx * stride + boff;
-        if (boff + 4 <= dl) {
-            *(uint32_t *)dst = *(const uint32_t *)ls;
+        const uint32_t *ls = (const uint32_t *)(sout_all + (size_t)sidx * stride + boff);
+        if (boff + 16 <= dl) {
+            uint4 v;
+            v.x = ls[0]; v.y = ls[1]; v.z = ls[2]; v.w = ls[3];
+            *(uint4 *)dst = v;                        /* dst 16B-aligned (writer) */
         } else {
-            for (uint32_t j = 0; j < dl - boff; j++) dst[j] = ls[j];
+            const uint8_t *lb = (const uint8_t *)ls;
+            for (uint32_t j = 0; j < dl - boff; j++) dst[j] = lb[j];
         }
     }
 }
