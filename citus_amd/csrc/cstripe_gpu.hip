@@ -244,17 +244,14 @@ __device__ inline void bs_init(ByteStream &b, const uint8_t *data, uint64_t off)
     b.w3 = *(const uint64_t *)(b.base + 24);
 }
 
-/* rotate 16 bytes at a time with ONE 16B load — halves the (per-lane
- * divergent) refill events vs 8B rotations */
 __device__ inline void bs_norm(ByteStream &b)
 {
-    while (b.pos - b.woff >= 16) {
-        b.w0 = b.w2;
-        b.w1 = b.w3;
-        b.woff += 16;
-        const uint64_t *p = (const uint64_t *)(b.base + b.woff + 16);
-        b.w2 = p[0];
-        b.w3 = p[1];
+    while (b.pos - b.woff >= 8) {
+        b.w0 = b.w1;
+        b.w1 = b.w2;
+        b.w2 = b.w3;
+        b.woff += 8;
+        b.w3 = *(const uint64_t *)(b.base + b.woff + 24);
     }
 }
 
@@ -263,8 +260,7 @@ __device__ inline uint8_t bs_get(ByteStream &b)
     bs_norm(b);
     uint32_t rel = b.pos - b.woff;
     b.pos++;
-    uint64_t w = (rel & 8) ? b.w1 : b.w0;
-    return (uint8_t)(w >> (8 * (rel & 7)));
+    return (uint8_t)(b.w0 >> (8 * rel));
 }
 
 __global__ void lz4_decode_lane_kernel(
