@@ -27,7 +27,11 @@ import time
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
-ALG_BYTES_PER_ROW = 32.5   # 4 projected int64 cols + 4 exists bits
+# algorithmic bytes/row = decompressed projected column widths + exists bits,
+# counted once
+ALG_BYTES = {"q6": 32.5,       # 4 int64 + 4 exists bits
+             "q1": 42.875,     # 5 int64 + 2 i8 + 7 exists bits
+             "count": 8.125}   # 1 int64 + 1 exists bit
 HBM_PEAK_GBPS = 8000.0     # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
 
 
@@ -51,6 +55,21 @@ def q6_aggs(ca):
     return [(ca.AGG_SUM_PROD_I64, 2, 3), (ca.AGG_COUNT_STAR, -1)]
 
 
+def q1_preds(ca):
+    return [(5, ca.PRED_LE, 10471)]   # l_shipdate <= 1998-09-02
+
+
+def q1_aggs(ca):
+    return [(ca.AGG_SUM_I64, 1), (ca.AGG_SUM_I64, 2),
+            (ca.AGG_SUM_DISC_I64, 2, 3, -1, 100),
+            (ca.AGG_SUM_DISC_TAX_I64, 2, 3, 4, 100),
+            (ca.AGG_COUNT_STAR, -1)]
+
+
+def count_preds(ca):
+    return [(1, ca.PRED_LT, 2400)]    # l_quantity < 24 (config 1)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -59,6 +78,7 @@ def main():
     ap.add_argument("--rows", type=int, default=100_000_000,
                     help="rows per GPU (config 2: 100M)")
     ap.add_argument("--compression", default="lz4", choices=["lz4", "none", "zstd"])
+    ap.add_argument("--query", default="q6", choices=["q6", "q1", "count"])
     ap.add_argument("--seg-bytes", type=int, default=0,
                     help="LZ4 micro-segment size override (bytes)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
@@ -92,14 +112,29 @@ def main():
     gen_s = time.time() - t0
 
     reader = ca.Reader(shard)
-    aggs = q6_aggs(ca)
-    scan = reader.scan(cols_mask=ca.agg_cols_mask(aggs), preds=q6_preds(ca))
+    if args.query == "q6":
+        preds, aggs, group_cols = q6_preds(ca), q6_aggs(ca), None
+    elif args.query == "q1":
+        preds, aggs, group_cols = q1_preds(ca), q1_aggs(ca), (6, 7)
+    else:
+        preds, aggs, group_cols = count_preds(ca), [(ca.AGG_COUNT_STAR, -1)], None
+    mask = ca.agg_cols_mask(aggs)
+    if group_cols:
+        for c in group_cols:
+            mask |= 1 << c
+    scan = reader.scan(cols_mask=mask, preds=preds)
     t0 = time.time()
     scan.stage(local_rank if dist else -1)
     stage_s = time.time() - t0
     staged_gb = scan.staged_bytes / 1e9
 
     def step():
+        if group_cols:
+            res = scan.agg_grouped(aggs, group_cols)
+            if dist:
+                from citus_amd.dist import all_gather_combine_grouped
+                return all_gather_combine_grouped(aggs, res, device="cuda")
+            return {k: ca.combine(aggs, [parts]) for k, parts in res.items()}
         parts = scan.agg(aggs)
         if dist:
             return all_gather_combine(aggs, parts, device="cuda")
@@ -144,11 +179,12 @@ def main():
     value = total_rows * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    alg_bytes_per_row = ALG_BYTES[args.query]
     avg_decode = sum(decode_ms) / len(decode_ms)
     avg_agg = sum(agg_ms) / len(agg_ms)
     dominant = "lz4_decode_kernel" if avg_decode >= avg_agg else "filter_agg_kernel"
     dominant_ms = max(avg_decode, avg_agg)
-    alg_bytes = args.rows * ALG_BYTES_PER_ROW          # per launch (this rank)
+    alg_bytes = args.rows * alg_bytes_per_row          # per launch (this rank)
     achieved_gbps = alg_bytes / (dominant_ms / 1e3) / 1e9 if dominant_ms > 0 else None
 
     # PMC-measured HBM traffic for the dominant kernel, if a committed profile
@@ -178,7 +214,10 @@ def main():
                 ca.gen_lineitem(sample_path, sample_rows, seed=42, compression=comp)
         with oracle.OracleTable(sample_path) as t:
             t0 = time.time()
-            cpu_parts, _ = t.scan_agg(q6_preds(ca), aggs)
+            if group_cols:
+                cpu_res, _ = t.scan_agg(preds, aggs, group_cols=group_cols)
+            else:
+                cpu_parts, _ = t.scan_agg(preds, aggs)
             cpu_s = time.time() - t0
         cpu_baseline = {
             "value": sample_rows / cpu_s,
@@ -189,12 +228,19 @@ def main():
                       f"({cpu_s:.1f}s)",
         }
         if sample_path == shard:
-            parity = ("bit-exact" if (cpu_parts[0].i128 == result[0].i128 and
-                                      cpu_parts[1].count == result[1].count)
-                      else "MISMATCH")
+            if group_cols:
+                ok = set(cpu_res) == set(result) and all(
+                    cpu_res[k][a].i128 == result[k][a].i128 and
+                    cpu_res[k][a].count == result[k][a].count
+                    for k in cpu_res for a in range(len(aggs)))
+            else:
+                ok = all(cpu_parts[a].i128 == result[a].i128 and
+                         cpu_parts[a].count == result[a].count
+                         for a in range(len(aggs)))
+            parity = "bit-exact" if ok else "MISMATCH"
 
     out = {
-        "metric": "columnar_rows_per_s_q6",
+        "metric": f"columnar_rows_per_s_{args.query}",
         "value": value,
         "unit": "rows/s",
         "n_gpus": world,
@@ -207,8 +253,8 @@ def main():
         "dtype": "int64",
         "data": "synthetic",
         "config": {
-            "workload": f"q6_lineitem_{args.rows // 1_000_000}m_{args.compression}",
-            "query": "tpch_q6",
+            "workload": f"{args.query}_lineitem_{args.rows // 1_000_000}m_{args.compression}",
+            "query": f"tpch_{args.query}" if args.query != "count" else "count_star",
             "rows_per_gpu": args.rows,
             "compression": args.compression,
             "stripe_rows": 150000,
@@ -219,7 +265,9 @@ def main():
             "stage_s": round(stage_s, 2),
             "kernel_ms": {"decode": round(avg_decode, 3), "filter_agg": round(avg_agg, 3)},
             "parity_vs_oracle": parity,
-            "q6_revenue_scale4": result[0].i128 if world == 1 else None,
+            "q6_revenue_scale4": (result[0].i128 if world == 1 and args.query == "q6"
+                                  else None),
+            "n_groups": len(result) if group_cols else None,
         },
         "roofline": {
             "bound": "hbm",

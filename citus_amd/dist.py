@@ -48,3 +48,51 @@ def all_gather_combine(aggs, parts, device=None):
     td.all_gather(gathered, mine)
     parts_list = [tensor_to_partials(g, len(parts)) for g in gathered]
     return _combine(aggs, parts_list)
+
+
+def all_gather_combine_grouped(aggs, resdict, device=None, max_groups=64):
+    """Grouped variant of the coordinator merge: gather every rank's
+    (key -> partials) table in ONE collective (fixed 64-slot block) and
+    combine per distinct key locally. Ranks missing a key contribute NULL
+    partials (strict combine skips them — aggregate_utils.c:976-1000)."""
+    import ctypes as C
+    world = td.get_world_size()
+    if device is None:
+        device = "cuda" if td.get_backend() == "nccl" else "cpu"
+    n_aggs = len(aggs)
+
+    keys = sorted(resdict)
+    assert len(keys) <= max_groups
+    kbuf = torch.full((max_groups,), -1, dtype=torch.int32)
+    flat = (Partial * (max_groups * n_aggs))()
+    for g, k in enumerate(keys):
+        kbuf[g] = k[0] | (k[1] << 8)
+        for a in range(n_aggs):
+            flat[g * n_aggs + a] = resdict[k][a]
+    pbuf = torch.frombuffer(bytearray(C.string_at(flat, C.sizeof(flat))),
+                            dtype=torch.uint8)
+    mine = torch.cat([kbuf.view(torch.uint8).flatten(), pbuf]).to(device)
+    gathered = [torch.empty_like(mine) for _ in range(world)]
+    td.all_gather(gathered, mine)
+
+    # decode per rank
+    rank_tables = []
+    for g in gathered:
+        g = g.cpu()
+        ks = g[:max_groups * 4].view(torch.int32)
+        parts = (Partial * (max_groups * n_aggs)).from_buffer_copy(
+            bytes(g[max_groups * 4:].numpy().tobytes()))
+        table = {}
+        for i in range(max_groups):
+            kv = int(ks[i])
+            if kv < 0:
+                continue
+            table[(kv & 0xFF, kv >> 8)] = [parts[i * n_aggs + a] for a in range(n_aggs)]
+        rank_tables.append(table)
+
+    all_keys = sorted(set().union(*[t.keys() for t in rank_tables]))
+    out = {}
+    null_row = [Partial(is_null=1) for _ in range(n_aggs)]
+    for k in all_keys:
+        out[k] = _combine(aggs, [t.get(k, null_row) for t in rank_tables])
+    return out

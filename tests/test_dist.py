@@ -75,3 +75,63 @@ def test_two_rank_combine_matches_single_scan(tmp_path):
     assert line, out.stdout
     got = json.loads(line[0][len("RESULT "):])
     assert got == expect
+
+
+WORKER_GROUPED = r"""
+import json, os, sys
+sys.path.insert(0, os.environ["REPO"])
+import torch.distributed as td
+import citus_amd as ca
+import oracle
+from citus_amd.dist import all_gather_combine_grouped
+
+td.init_process_group("gloo")
+rank = td.get_rank()
+shard = os.path.join(os.environ["SHARD_DIR"], f"shard{rank}.cs")
+preds = [(5, ca.PRED_LE, 10471)]
+aggs = [(ca.AGG_SUM_I64, 1), (ca.AGG_SUM_DISC_I64, 2, 3, -1, 100),
+        (ca.AGG_COUNT_STAR, -1)]
+with oracle.OracleTable(shard) as t:
+    res, _ = t.scan_agg(preds, aggs, group_cols=(6, 7))
+combined = all_gather_combine_grouped(aggs, res, device="cpu")
+if rank == 0:
+    out = {f"{k[0]},{k[1]}": [p.as_dict() for p in v] for k, v in combined.items()}
+    print("RESULT " + json.dumps(out))
+td.destroy_process_group()
+"""
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_grouped_combine(tmp_path):
+    import citus_amd as ca
+    import oracle
+
+    ca.gen_lineitem(str(tmp_path / "shard0.cs"), 50_000, seed=42)
+    ca.gen_lineitem(str(tmp_path / "shard1.cs"), 50_000, seed=4242)
+    preds = [(5, ca.PRED_LE, 10471)]
+    aggs = [(ca.AGG_SUM_I64, 1), (ca.AGG_SUM_DISC_I64, 2, 3, -1, 100),
+            (ca.AGG_COUNT_STAR, -1)]
+    tables = []
+    for s in (0, 1):
+        with oracle.OracleTable(str(tmp_path / f"shard{s}.cs")) as t:
+            res, _ = t.scan_agg(preds, aggs, group_cols=(6, 7))
+            tables.append(res)
+    keys = sorted(set(tables[0]) | set(tables[1]))
+    null_row = [ca.Partial(is_null=1) for _ in aggs]
+    expect = {f"{k[0]},{k[1]}": [p.as_dict() for p in
+                                 ca.combine(aggs, [t.get(k, null_row) for t in tables])]
+              for k in keys}
+
+    env = dict(os.environ, REPO=REPO, SHARD_DIR=str(tmp_path),
+               MASTER_ADDR="127.0.0.1", MASTER_PORT="29531")
+    script = str(tmp_path / "workerg.py")
+    open(script, "w").write(WORKER_GROUPED)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29531", script],
+        env=env, capture_output=True, text=True, timeout=280)
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")]
+    got = json.loads(line[0][len("RESULT "):])
+    assert got == expect
