@@ -634,6 +634,83 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     }
 }
 
+
+/* per-row precomputed aggregate contribution: operands loaded ONCE per row,
+ * so the per-distinct-key reduce rounds touch registers only */
+struct PrepAcc {
+    int64_t lo, hi;    /* i128 contribution / i64 value */
+    double f;
+    bool valid;
+};
+
+__device__ inline void acc_prepare(PrepAcc &o, const AggD &g,
+                                   const uint8_t *__restrict__ data,
+                                   const uint8_t *__restrict__ scratch,
+                                   const uint32_t *__restrict__ rank,
+                                   const ColLoc *__restrict__ cols, uint32_t row)
+{
+    o.lo = 0; o.hi = 0; o.f = 0.0; o.valid = false;
+    int64_t iv, ib, ic; double fv, fb, fc;
+    switch (g.kind) {
+        case CSTRIPE_AGG_COUNT_STAR:
+            o.valid = true; o.lo = 1; break;
+        case CSTRIPE_AGG_COUNT_COL:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { o.valid = true; o.lo = 1; }
+            break;
+        case CSTRIPE_AGG_SUM_I64:
+        case CSTRIPE_AGG_MIN_I64:
+        case CSTRIPE_AGG_MAX_I64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) {
+                o.valid = true; o.lo = iv; o.hi = iv < 0 ? -1 : 0;
+            }
+            break;
+        case CSTRIPE_AGG_SUM_F64:
+        case CSTRIPE_AGG_MIN_F64:
+        case CSTRIPE_AGG_MAX_F64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { o.valid = true; o.f = fv; }
+            break;
+        case CSTRIPE_AGG_SUM_PROD_I64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
+                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb)) {
+                __int128 x = (__int128)iv * ib;
+                o.valid = true; o.lo = (int64_t)(uint64_t)x; o.hi = (int64_t)(x >> 64);
+            }
+            break;
+        case CSTRIPE_AGG_SUM_DISC_I64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
+                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb)) {
+                __int128 x = (__int128)iv * (g.one - ib);
+                o.valid = true; o.lo = (int64_t)(uint64_t)x; o.hi = (int64_t)(x >> 64);
+            }
+            break;
+        case CSTRIPE_AGG_SUM_DISC_TAX_I64:
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
+                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb) &&
+                col_value(data, scratch, rank, cols[g.proj_c], row, ic, fc)) {
+                __int128 x = (__int128)iv * (g.one - ib) * (g.one + ic);
+                o.valid = true; o.lo = (int64_t)(uint64_t)x; o.hi = (int64_t)(x >> 64);
+            }
+            break;
+    }
+}
+
+/* fold a prepared contribution into a ThreadAcc (registers only) */
+__device__ inline void acc_apply(ThreadAcc &a, uint8_t kind, const PrepAcc &p)
+{
+    if (!p.valid) return;
+    switch (kind) {
+        case CSTRIPE_AGG_MIN_I64: a.lo = min(a.lo, p.lo); break;
+        case CSTRIPE_AGG_MAX_I64: a.lo = max(a.lo, p.lo); break;
+        case CSTRIPE_AGG_MIN_F64: a.f = fmin(a.f, p.f); break;
+        case CSTRIPE_AGG_MAX_F64: a.f = fmax(a.f, p.f); break;
+        case CSTRIPE_AGG_SUM_F64: a.f += p.f; break;
+        default:
+            acc_add_i128(a, ((__int128)p.hi << 64) | (unsigned long long)p.lo);
+            break;
+    }
+    a.cnt++;
+}
+
 /* =====================================================================
  * Grouped aggregation (TPC-H Q1 shape): GROUP BY 1-2 categorical i8
  * columns, few distinct groups. The reference plan is worker HashAggregate
@@ -711,6 +788,13 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                     key |= ((uint32_t)kv & 0xFF) << (8 * gc);
                 }
             }
+            /* load agg operands ONCE per row */
+            PrepAcc prep[MAX_AGGS];
+            if (pass)
+                for (uint32_t a = 0; a < params.n_aggs; a++)
+                    acc_prepare(prep[a], params.aggs[a], data, scratch, rank, cols, row);
+            else
+                for (uint32_t a = 0; a < params.n_aggs; a++) prep[a].valid = false;
             /* wave-cooperative reduce, one distinct key per round */
             uint64_t remaining = __ballot(pass);
             while (remaining) {
@@ -739,7 +823,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                     ThreadAcc t;
                     acc_init(t, params.aggs[a].kind);
                     if (mine)
-                        acc_row(t, params.aggs[a], data, scratch, rank, cols, row);
+                        acc_apply(t, params.aggs[a].kind, prep[a]);
                     wave_reduce(t, params.aggs[a].kind);
                     if (lane == 0) {
                         ThreadAcc cur = myacc[slot * params.n_aggs + a];
