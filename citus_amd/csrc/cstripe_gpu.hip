@@ -731,6 +731,7 @@ struct GroupParams {
     uint32_t n_work;                          /* n_groups * tiles_per_group */
 };
 
+template <int NAGGS>
 __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
     const uint32_t *__restrict__ rank, const GroupDesc *__restrict__ groups,
@@ -739,6 +740,9 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     const GroupParams gp)
 {
     const AggParams &params = gp.base;
+    /* compile-time agg count unrolls the prep/reduce loops so the per-row
+     * PrepAcc array stays in registers instead of scratch */
+    const uint32_t n_aggs_ct = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
     const uint32_t wid = threadIdx.x / WAVE;
     const uint32_t lane = threadIdx.x % WAVE;
     const uint32_t n_waves = AGG_BLOCK / WAVE;
@@ -753,7 +757,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     __syncthreads();
 
     uint16_t *mykeys = wkeys + wid * GRP_SLOTS;
-    ThreadAcc *myacc = wacc + (size_t)wid * GRP_SLOTS * params.n_aggs;
+    ThreadAcc *myacc = wacc + (size_t)wid * GRP_SLOTS * n_aggs_ct;
     uint32_t used = 0;       /* wave-uniform slot count (updated by lane 0 path) */
 
     for (uint32_t work = blockIdx.x; work < gp.n_work; work += gridDim.x) {
@@ -789,12 +793,12 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                 }
             }
             /* load agg operands ONCE per row */
-            PrepAcc prep[MAX_AGGS];
+            PrepAcc prep[NAGGS >= 0 ? NAGGS : MAX_AGGS];
             if (pass)
-                for (uint32_t a = 0; a < params.n_aggs; a++)
+                for (uint32_t a = 0; a < n_aggs_ct; a++)
                     acc_prepare(prep[a], params.aggs[a], data, scratch, rank, cols, row);
             else
-                for (uint32_t a = 0; a < params.n_aggs; a++) prep[a].valid = false;
+                for (uint32_t a = 0; a < n_aggs_ct; a++) prep[a].valid = false;
             /* wave-cooperative reduce, one distinct key per round */
             uint64_t remaining = __ballot(pass);
             while (remaining) {
@@ -813,22 +817,22 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                     }
                     slot = used++;
                     if (lane == 0) mykeys[slot] = (uint16_t)kk;
-                    for (uint32_t a = 0; a < params.n_aggs; a++) {
+                    for (uint32_t a = 0; a < n_aggs_ct; a++) {
                         ThreadAcc z;
                         acc_init(z, params.aggs[a].kind);
-                        if (lane == 0) myacc[slot * params.n_aggs + a] = z;
+                        if (lane == 0) myacc[slot * n_aggs_ct + a] = z;
                     }
                 }
-                for (uint32_t a = 0; a < params.n_aggs; a++) {
+                for (uint32_t a = 0; a < n_aggs_ct; a++) {
                     ThreadAcc t;
                     acc_init(t, params.aggs[a].kind);
                     if (mine)
                         acc_apply(t, params.aggs[a].kind, prep[a]);
                     wave_reduce(t, params.aggs[a].kind);
                     if (lane == 0) {
-                        ThreadAcc cur = myacc[slot * params.n_aggs + a];
+                        ThreadAcc cur = myacc[slot * n_aggs_ct + a];
                         acc_merge(cur, t, params.aggs[a].kind);
-                        myacc[slot * params.n_aggs + a] = cur;
+                        myacc[slot * n_aggs_ct + a] = cur;
                     }
                 }
                 remaining &= ~__ballot(mine);
@@ -841,7 +845,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
      * wave 0 lane 0 does it serially — tables are tiny. */
     if (threadIdx.x == 0) {
         uint16_t *bk = keys_out + (size_t)blockIdx.x * (n_waves * GRP_SLOTS);
-        AccCell *bc = cells_out + (size_t)blockIdx.x * (n_waves * GRP_SLOTS) * params.n_aggs;
+        AccCell *bc = cells_out + (size_t)blockIdx.x * (n_waves * GRP_SLOTS) * n_aggs_ct;
         uint32_t n = 0;
         for (uint32_t w = 0; w < n_waves; w++) {
             for (uint32_t sidx = 0; sidx < GRP_SLOTS; sidx++) {
@@ -849,21 +853,21 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                 if (k == 0xFFFF) continue;
                 uint32_t at = n;
                 for (uint32_t j = 0; j < n; j++) if (bk[j] == k) { at = j; break; }
-                ThreadAcc *src = wacc + ((size_t)(w * GRP_SLOTS) + sidx) * params.n_aggs;
+                ThreadAcc *src = wacc + ((size_t)(w * GRP_SLOTS) + sidx) * n_aggs_ct;
                 if (at == n) {
                     bk[n] = k;
-                    for (uint32_t a = 0; a < params.n_aggs; a++) {
+                    for (uint32_t a = 0; a < n_aggs_ct; a++) {
                         AccCell c{src[a].lo, src[a].hi, src[a].f, src[a].cnt};
-                        bc[(size_t)n * params.n_aggs + a] = c;
+                        bc[(size_t)n * n_aggs_ct + a] = c;
                     }
                     n++;
                 } else {
-                    for (uint32_t a = 0; a < params.n_aggs; a++) {
-                        AccCell c = bc[(size_t)at * params.n_aggs + a];
+                    for (uint32_t a = 0; a < n_aggs_ct; a++) {
+                        AccCell c = bc[(size_t)at * n_aggs_ct + a];
                         ThreadAcc cur{c.lo, c.hi, c.f, c.cnt};
                         acc_merge(cur, src[a], params.aggs[a].kind);
                         AccCell o{cur.lo, cur.hi, cur.f, cur.cnt};
-                        bc[(size_t)at * params.n_aggs + a] = o;
+                        bc[(size_t)at * n_aggs_ct + a] = o;
                     }
                 }
             }
@@ -1280,9 +1284,17 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         HIP_TRY(hipEventRecord(g->ev0, g->stream));
         launch_decode(g);
         HIP_TRY(hipEventRecord(g->ev1, g->stream));
-        hipLaunchKernelGGL(grouped_agg_kernel, dim3(grid), dim3(AGG_BLOCK), lds, g->stream,
-                           g->d_data, g->d_scratch, g->d_rank, g->d_groups, g->d_colloc,
-                           g->d_gkeys, g->d_gcells, g->d_error, gp);
+        {
+            auto launchg = [&](auto *kern) {
+                hipLaunchKernelGGL(kern, dim3(grid), dim3(AGG_BLOCK), lds, g->stream,
+                                   g->d_data, g->d_scratch, g->d_rank, g->d_groups,
+                                   g->d_colloc, g->d_gkeys, g->d_gcells, g->d_error, gp);
+            };
+            if (n_aggs == 5) launchg(grouped_agg_kernel<5>);
+            else if (n_aggs == 1) launchg(grouped_agg_kernel<1>);
+            else if (n_aggs == 2) launchg(grouped_agg_kernel<2>);
+            else launchg(grouped_agg_kernel<-1>);
+        }
         hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(64), 0, g->stream,
                            g->d_gkeys, g->d_gcells, grid, per_block,
                            g->d_gfkeys, g->d_gfcells, g->d_gn, g->d_error, gp);
