@@ -792,50 +792,60 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                     key |= ((uint32_t)kv & 0xFF) << (8 * gc);
                 }
             }
-            /* load agg operands ONCE per row */
-            PrepAcc prep[NAGGS >= 0 ? NAGGS : MAX_AGGS];
-            if (pass)
-                for (uint32_t a = 0; a < n_aggs_ct; a++)
-                    acc_prepare(prep[a], params.aggs[a], data, scratch, rank, cols, row);
-            else
-                for (uint32_t a = 0; a < n_aggs_ct; a++) prep[a].valid = false;
-            /* wave-cooperative reduce, one distinct key per round */
-            uint64_t remaining = __ballot(pass);
-            while (remaining) {
-                int leader = __ffsll((unsigned long long)remaining) - 1;
-                uint32_t kk = (uint32_t)__shfl((int)key, leader, WAVE);
-                bool mine = pass && key == kk;
-                /* find/claim slot (wave-uniform) */
-                uint32_t slot = 0xFFFFFFFF;
-                for (uint32_t sidx = 0; sidx < used; sidx++)
-                    if (mykeys[sidx] == (uint16_t)kk) { slot = sidx; break; }
-                if (slot == 0xFFFFFFFF) {
-                    if (used >= GRP_SLOTS) {          /* overflow: flag + drop */
-                        if (lane == 0) atomicOr(err, 8);
-                        remaining &= ~__ballot(mine);
-                        continue;
+            /* round 0 (agg-independent): assign each passing lane its key's
+             * wave-table slot */
+            uint32_t myslot = 0xFFFFFFFF;
+            {
+                uint64_t remaining = __ballot(pass);
+                while (remaining) {
+                    int leader = __ffsll((unsigned long long)remaining) - 1;
+                    uint32_t kk = (uint32_t)__shfl((int)key, leader, WAVE);
+                    bool mine = pass && key == kk;
+                    uint32_t slot = 0xFFFFFFFF;
+                    for (uint32_t sidx = 0; sidx < used; sidx++)
+                        if (mykeys[sidx] == (uint16_t)kk) { slot = sidx; break; }
+                    if (slot == 0xFFFFFFFF) {
+                        if (used >= GRP_SLOTS) {      /* overflow: flag + drop */
+                            if (lane == 0) atomicOr(err, 8);
+                            remaining &= ~__ballot(mine);
+                            continue;
+                        }
+                        slot = used++;
+                        if (lane == 0) mykeys[slot] = (uint16_t)kk;
+                        for (uint32_t a = 0; a < n_aggs_ct; a++) {
+                            ThreadAcc z;
+                            acc_init(z, params.aggs[a].kind);
+                            if (lane == 0) myacc[slot * n_aggs_ct + a] = z;
+                        }
                     }
-                    slot = used++;
-                    if (lane == 0) mykeys[slot] = (uint16_t)kk;
-                    for (uint32_t a = 0; a < n_aggs_ct; a++) {
-                        ThreadAcc z;
-                        acc_init(z, params.aggs[a].kind);
-                        if (lane == 0) myacc[slot * n_aggs_ct + a] = z;
-                    }
+                    if (mine) myslot = slot;
+                    remaining &= ~__ballot(mine);
                 }
-                for (uint32_t a = 0; a < n_aggs_ct; a++) {
+            }
+            /* per agg (unrolled): ONE PrepAcc in registers, distinct-slot
+             * rounds of predicated wave reduce — nothing array-indexed by a
+             * survives across rounds, so no scratch */
+            for (uint32_t a = 0; a < n_aggs_ct; a++) {
+                PrepAcc p;
+                p.valid = false;
+                if (pass)
+                    acc_prepare(p, params.aggs[a], data, scratch, rank, cols, row);
+                uint64_t remaining = __ballot(pass && myslot != 0xFFFFFFFF);
+                while (remaining) {
+                    int leader = __ffsll((unsigned long long)remaining) - 1;
+                    uint32_t ss = (uint32_t)__shfl((int)myslot, leader, WAVE);
+                    bool mine = pass && myslot == ss;
                     ThreadAcc t;
                     acc_init(t, params.aggs[a].kind);
-                    if (mine)
-                        acc_apply(t, params.aggs[a].kind, prep[a]);
+                    if (mine) acc_apply(t, params.aggs[a].kind, p);
                     wave_reduce(t, params.aggs[a].kind);
                     if (lane == 0) {
-                        ThreadAcc cur = myacc[slot * n_aggs_ct + a];
+                        ThreadAcc cur = myacc[ss * n_aggs_ct + a];
                         acc_merge(cur, t, params.aggs[a].kind);
-                        myacc[slot * n_aggs_ct + a] = cur;
+                        myacc[ss * n_aggs_ct + a] = cur;
                     }
+                    remaining &= ~__ballot(mine);
                 }
-                remaining &= ~__ballot(mine);
             }
         }
     }
