@@ -731,6 +731,69 @@ struct GroupParams {
     uint32_t n_work;                          /* n_groups * tiles_per_group */
 };
 
+
+/* fold a prepared contribution into an LDS accumulator cell with atomics —
+ * per-lane, no wave-cooperative rounds. int128 sums stay EXACT: the carry of
+ * my own low-half add is detected from atomicAdd's return value and added to
+ * the high half (independent of interleaving). f64 sums become
+ * atomic-order-dependent (covered by the 1e-6 float tolerance). */
+__device__ inline void acc_apply_atomic(ThreadAcc *cell, uint8_t kind, const PrepAcc &p)
+{
+    if (!p.valid) return;
+    switch (kind) {
+        case CSTRIPE_AGG_MIN_I64:
+            atomicMin((long long *)&cell->lo, (long long)p.lo);
+            break;
+        case CSTRIPE_AGG_MAX_I64:
+            atomicMax((long long *)&cell->lo, (long long)p.lo);
+            break;
+        case CSTRIPE_AGG_MIN_F64: {
+            double v = p.f;
+            unsigned long long *addr = (unsigned long long *)&cell->f;
+            unsigned long long cur = *addr;
+            while (true) {
+                double c;
+                memcpy(&c, &cur, 8);
+                if (v >= c) break;
+                unsigned long long nv;
+                memcpy(&nv, &v, 8);
+                unsigned long long prev = atomicCAS(addr, cur, nv);
+                if (prev == cur) break;
+                cur = prev;
+            }
+            break;
+        }
+        case CSTRIPE_AGG_MAX_F64: {
+            double v = p.f;
+            unsigned long long *addr = (unsigned long long *)&cell->f;
+            unsigned long long cur = *addr;
+            while (true) {
+                double c;
+                memcpy(&c, &cur, 8);
+                if (v <= c) break;
+                unsigned long long nv;
+                memcpy(&nv, &v, 8);
+                unsigned long long prev = atomicCAS(addr, cur, nv);
+                if (prev == cur) break;
+                cur = prev;
+            }
+            break;
+        }
+        case CSTRIPE_AGG_SUM_F64:
+            atomicAdd((double *)&cell->f, p.f);
+            break;
+        default: {   /* counts + i128 sums */
+            unsigned long long old =
+                atomicAdd((unsigned long long *)&cell->lo, (unsigned long long)p.lo);
+            int64_t carry = (old + (unsigned long long)p.lo) < old ? 1 : 0;
+            int64_t hi = p.hi + carry;
+            if (hi) atomicAdd((unsigned long long *)&cell->hi, (unsigned long long)hi);
+            break;
+        }
+    }
+    atomicAdd((unsigned long long *)&cell->cnt, 1ull);
+}
+
 template <int NAGGS>
 __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
@@ -822,29 +885,14 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                     remaining &= ~__ballot(mine);
                 }
             }
-            /* per agg (unrolled): ONE PrepAcc in registers, distinct-slot
-             * rounds of predicated wave reduce — nothing array-indexed by a
-             * survives across rounds, so no scratch */
-            for (uint32_t a = 0; a < n_aggs_ct; a++) {
-                PrepAcc p;
-                p.valid = false;
-                if (pass)
+            /* per agg (unrolled): one prepared contribution per row folded
+             * into the wave's LDS table with per-lane atomics */
+            if (pass && myslot != 0xFFFFFFFF) {
+                for (uint32_t a = 0; a < n_aggs_ct; a++) {
+                    PrepAcc p;
                     acc_prepare(p, params.aggs[a], data, scratch, rank, cols, row);
-                uint64_t remaining = __ballot(pass && myslot != 0xFFFFFFFF);
-                while (remaining) {
-                    int leader = __ffsll((unsigned long long)remaining) - 1;
-                    uint32_t ss = (uint32_t)__shfl((int)myslot, leader, WAVE);
-                    bool mine = pass && myslot == ss;
-                    ThreadAcc t;
-                    acc_init(t, params.aggs[a].kind);
-                    if (mine) acc_apply(t, params.aggs[a].kind, p);
-                    wave_reduce(t, params.aggs[a].kind);
-                    if (lane == 0) {
-                        ThreadAcc cur = myacc[ss * n_aggs_ct + a];
-                        acc_merge(cur, t, params.aggs[a].kind);
-                        myacc[ss * n_aggs_ct + a] = cur;
-                    }
-                    remaining &= ~__ballot(mine);
+                    acc_apply_atomic(&myacc[myslot * n_aggs_ct + a],
+                                     params.aggs[a].kind, p);
                 }
             }
         }
