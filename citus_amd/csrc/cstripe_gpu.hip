@@ -934,42 +934,110 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     }
 }
 
-/* single-block merge of per-block group tables -> final <=64 groups */
-__global__ __launch_bounds__(64) void grouped_final_kernel(
+/* single-block merge of per-block group tables -> final <=64 groups.
+ * Parallel: 256 threads stride the (block, slot) entries; keys claim final
+ * slots via LDS CAS (linear probe); cells merge with the same atomic
+ * carry-exact machinery as the main kernel. */
+__device__ inline void cell_merge_atomic(ThreadAcc *dst, uint8_t kind, const AccCell &src)
+{
+    switch (kind) {
+        case CSTRIPE_AGG_MIN_I64:
+            atomicMin((long long *)&dst->lo, (long long)src.lo);
+            break;
+        case CSTRIPE_AGG_MAX_I64:
+            atomicMax((long long *)&dst->lo, (long long)src.lo);
+            break;
+        case CSTRIPE_AGG_MIN_F64: {
+            unsigned long long *addr = (unsigned long long *)&dst->f;
+            unsigned long long cur = *addr;
+            while (true) {
+                double c;
+                memcpy(&c, &cur, 8);
+                if (src.f >= c) break;
+                unsigned long long nv;
+                memcpy(&nv, &src.f, 8);
+                unsigned long long prev = atomicCAS(addr, cur, nv);
+                if (prev == cur) break;
+                cur = prev;
+            }
+            break;
+        }
+        case CSTRIPE_AGG_MAX_F64: {
+            unsigned long long *addr = (unsigned long long *)&dst->f;
+            unsigned long long cur = *addr;
+            while (true) {
+                double c;
+                memcpy(&c, &cur, 8);
+                if (src.f <= c) break;
+                unsigned long long nv;
+                memcpy(&nv, &src.f, 8);
+                unsigned long long prev = atomicCAS(addr, cur, nv);
+                if (prev == cur) break;
+                cur = prev;
+            }
+            break;
+        }
+        case CSTRIPE_AGG_SUM_F64:
+            atomicAdd((double *)&dst->f, src.f);
+            break;
+        default: {
+            unsigned long long old =
+                atomicAdd((unsigned long long *)&dst->lo, (unsigned long long)src.lo);
+            int64_t carry = (old + (unsigned long long)src.lo) < old ? 1 : 0;
+            int64_t hi = src.hi + carry;
+            if (hi) atomicAdd((unsigned long long *)&dst->hi, (unsigned long long)hi);
+            break;
+        }
+    }
+    atomicAdd((unsigned long long *)&dst->cnt, (unsigned long long)src.cnt);
+}
+
+__global__ __launch_bounds__(AGG_BLOCK) void grouped_final_kernel(
     const uint16_t *__restrict__ keys_in, const AccCell *__restrict__ cells_in,
     uint32_t n_blocks, uint32_t per_block, uint16_t *__restrict__ keys_out,
     AccCell *__restrict__ cells_out, uint32_t *__restrict__ n_groups_out,
     int *__restrict__ err, const GroupParams gp)
 {
-    if (threadIdx.x != 0) return;      /* serial: tables are tiny */
     const AggParams &params = gp.base;
-    uint32_t n = 0;
-    for (uint32_t b = 0; b < n_blocks; b++) {
-        for (uint32_t j = 0; j < per_block; j++) {
-            uint16_t k = keys_in[(size_t)b * per_block + j];
-            if (k == 0xFFFF) continue;
-            uint32_t at = n;
-            for (uint32_t i = 0; i < n; i++) if (keys_out[i] == k) { at = i; break; }
-            const AccCell *src = cells_in + ((size_t)b * per_block + j) * params.n_aggs;
-            if (at == n) {
-                if (n >= CSTRIPE_MAX_GROUPS) { atomicOr(err, 16); return; }
-                keys_out[n] = k;
-                for (uint32_t a = 0; a < params.n_aggs; a++)
-                    cells_out[(size_t)n * params.n_aggs + a] = src[a];
-                n++;
-            } else {
-                for (uint32_t a = 0; a < params.n_aggs; a++) {
-                    AccCell c = cells_out[(size_t)at * params.n_aggs + a];
-                    ThreadAcc cur{c.lo, c.hi, c.f, c.cnt};
-                    ThreadAcc s{src[a].lo, src[a].hi, src[a].f, src[a].cnt};
-                    acc_merge(cur, s, params.aggs[a].kind);
-                    AccCell o{cur.lo, cur.hi, cur.f, cur.cnt};
-                    cells_out[(size_t)at * params.n_aggs + a] = o;
-                }
-            }
-        }
+    __shared__ int skeys[CSTRIPE_MAX_GROUPS];
+    __shared__ ThreadAcc scells[CSTRIPE_MAX_GROUPS * MAX_AGGS];
+    for (uint32_t i = threadIdx.x; i < CSTRIPE_MAX_GROUPS; i += AGG_BLOCK) {
+        skeys[i] = -1;
+        for (uint32_t a = 0; a < params.n_aggs; a++)
+            acc_init(scells[i * MAX_AGGS + a], params.aggs[a].kind);
     }
-    *n_groups_out = n;
+    __syncthreads();
+
+    const uint32_t total = n_blocks * per_block;
+    for (uint32_t f = threadIdx.x; f < total; f += AGG_BLOCK) {
+        uint16_t k = keys_in[f];
+        if (k == 0xFFFF) continue;
+        int slot = -1;
+        for (uint32_t s = 0; s < CSTRIPE_MAX_GROUPS; s++) {
+            int old = atomicCAS(&skeys[s], -1, (int)k);
+            if (old == -1 || old == (int)k) { slot = (int)s; break; }
+        }
+        if (slot < 0) { atomicOr(err, 16); continue; }
+        const AccCell *src = cells_in + (size_t)f * params.n_aggs;
+        for (uint32_t a = 0; a < params.n_aggs; a++)
+            cell_merge_atomic(&scells[slot * MAX_AGGS + a], params.aggs[a].kind, src[a]);
+    }
+    __syncthreads();
+
+    if (threadIdx.x == 0) {
+        uint32_t n = 0;
+        for (uint32_t s = 0; s < CSTRIPE_MAX_GROUPS; s++) {
+            if (skeys[s] < 0) continue;
+            keys_out[n] = (uint16_t)skeys[s];
+            for (uint32_t a = 0; a < params.n_aggs; a++) {
+                const ThreadAcc &t = scells[s * MAX_AGGS + a];
+                AccCell c{t.lo, t.hi, t.f, t.cnt};
+                cells_out[(size_t)n * params.n_aggs + a] = c;
+            }
+            n++;
+        }
+        *n_groups_out = n;
+    }
 }
 
 /* device-wide final reduce over block partials: one block, grid-stride */
@@ -1353,7 +1421,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             else if (n_aggs == 2) launchg(grouped_agg_kernel<2>);
             else launchg(grouped_agg_kernel<-1>);
         }
-        hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(64), 0, g->stream,
+        hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
                            g->d_gkeys, g->d_gcells, grid, per_block,
                            g->d_gfkeys, g->d_gfcells, g->d_gn, g->d_error, gp);
         HIP_TRY(hipEventRecord(g->ev2, g->stream));
