@@ -822,6 +822,10 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     uint16_t *mykeys = wkeys + wid * GRP_SLOTS;
     ThreadAcc *myacc = wacc + (size_t)wid * GRP_SLOTS * n_aggs_ct;
     uint32_t used = 0;       /* wave-uniform slot count (updated by lane 0 path) */
+    /* per-lane 4-way key->slot register cache: hits skip the wave-ballot
+     * slot rounds entirely (few distinct groups => near-100% after warmup) */
+    uint32_t ck0 = ~0u, ck1 = ~0u, ck2 = ~0u, ck3 = ~0u;
+    uint32_t cs0 = 0, cs1 = 0, cs2 = 0, cs3 = 0;
 
     for (uint32_t work = blockIdx.x; work < gp.n_work; work += gridDim.x) {
         const uint32_t gid = work / params.tiles_per_group;
@@ -856,10 +860,17 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                 }
             }
             /* round 0 (agg-independent): assign each passing lane its key's
-             * wave-table slot */
+             * wave-table slot — register cache first, ballot rounds only
+             * for misses */
             uint32_t myslot = 0xFFFFFFFF;
+            if (pass) {
+                if (key == ck0) myslot = cs0;
+                else if (key == ck1) myslot = cs1;
+                else if (key == ck2) myslot = cs2;
+                else if (key == ck3) myslot = cs3;
+            }
             {
-                uint64_t remaining = __ballot(pass);
+                uint64_t remaining = __ballot(pass && myslot == 0xFFFFFFFF);
                 while (remaining) {
                     int leader = __ffsll((unsigned long long)remaining) - 1;
                     uint32_t kk = (uint32_t)__shfl((int)key, leader, WAVE);
@@ -881,7 +892,13 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                             if (lane == 0) myacc[slot * n_aggs_ct + a] = z;
                         }
                     }
-                    if (mine) myslot = slot;
+                    if (mine) {
+                        myslot = slot;
+                        ck3 = ck2; cs3 = cs2;
+                        ck2 = ck1; cs2 = cs1;
+                        ck1 = ck0; cs1 = cs0;
+                        ck0 = key; cs0 = slot;
+                    }
                     remaining &= ~__ballot(mine);
                 }
             }
