@@ -1097,24 +1097,26 @@ void csgpu_release(cstripe_scan *s)
 {
     if (!s || !s->gpu) return;
     cs_gpu_state *g = s->gpu;
-    if (g->d_data) hipFree(g->d_data);
-    if (g->d_scratch) hipFree(g->d_scratch);
-    if (g->d_rank) hipFree(g->d_rank);
-    if (g->d_segs) hipFree(g->d_segs);
-    if (g->d_groups) hipFree(g->d_groups);
-    if (g->d_colloc) hipFree(g->d_colloc);
-    if (g->d_block) hipFree(g->d_block);
-    if (g->d_final) hipFree(g->d_final);
-    if (g->d_gkeys) hipFree(g->d_gkeys);
-    if (g->d_gcells) hipFree(g->d_gcells);
-    if (g->d_gfkeys) hipFree(g->d_gfkeys);
-    if (g->d_gfcells) hipFree(g->d_gfcells);
-    if (g->d_gn) hipFree(g->d_gn);
-    if (g->d_error) hipFree(g->d_error);
-    if (g->ev0) hipEventDestroy(g->ev0);
-    if (g->ev1) hipEventDestroy(g->ev1);
-    if (g->ev2) hipEventDestroy(g->ev2);
-    if (g->stream) hipStreamDestroy(g->stream);
+    #define HIP_DROP(x) do { hipError_t _e = (x); (void)_e; } while (0)
+    if (g->d_data) HIP_DROP(hipFree(g->d_data));
+    if (g->d_scratch) HIP_DROP(hipFree(g->d_scratch));
+    if (g->d_rank) HIP_DROP(hipFree(g->d_rank));
+    if (g->d_segs) HIP_DROP(hipFree(g->d_segs));
+    if (g->d_groups) HIP_DROP(hipFree(g->d_groups));
+    if (g->d_colloc) HIP_DROP(hipFree(g->d_colloc));
+    if (g->d_block) HIP_DROP(hipFree(g->d_block));
+    if (g->d_final) HIP_DROP(hipFree(g->d_final));
+    if (g->d_gkeys) HIP_DROP(hipFree(g->d_gkeys));
+    if (g->d_gcells) HIP_DROP(hipFree(g->d_gcells));
+    if (g->d_gfkeys) HIP_DROP(hipFree(g->d_gfkeys));
+    if (g->d_gfcells) HIP_DROP(hipFree(g->d_gfcells));
+    if (g->d_gn) HIP_DROP(hipFree(g->d_gn));
+    if (g->d_error) HIP_DROP(hipFree(g->d_error));
+    if (g->ev0) HIP_DROP(hipEventDestroy(g->ev0));
+    if (g->ev1) HIP_DROP(hipEventDestroy(g->ev1));
+    if (g->ev2) HIP_DROP(hipEventDestroy(g->ev2));
+    if (g->stream) HIP_DROP(hipStreamDestroy(g->stream));
+    #undef HIP_DROP
     delete g;
     s->gpu = nullptr;
 }
@@ -1136,7 +1138,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         if (hipSetDevice(device_id) != hipSuccess) { cs_set_err("hipSetDevice(%d) failed", device_id); csgpu_release(s); return CSTRIPE_ERR; }
         g->device = device_id;
     } else {
-        hipGetDevice(&g->device);
+        (void)hipGetDevice(&g->device);
     }
 
     /* projected column slots */
@@ -1456,8 +1458,8 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         if (h_err) { cs_set_err("too many distinct groups (device flag %d; caps: %d/wave, %d total)", h_err, GRP_SLOTS, CSTRIPE_MAX_GROUPS); return CSTRIPE_ERR; }
 
         float ms_decode = 0, ms_agg = 0;
-        hipEventElapsedTime(&ms_decode, g->ev0, g->ev1);
-        hipEventElapsedTime(&ms_agg, g->ev1, g->ev2);
+        (void)hipEventElapsedTime(&ms_decode, g->ev0, g->ev1);
+        (void)hipEventElapsedTime(&ms_agg, g->ev1, g->ev2);
         s->last_decode_ms = ms_decode;
         s->last_agg_ms = ms_agg;
         s->last_kernel_ms = ms_decode + ms_agg;
@@ -1538,8 +1540,8 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
     if (h_err) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
 
     float ms_decode = 0, ms_agg = 0;
-    hipEventElapsedTime(&ms_decode, g->ev0, g->ev1);
-    hipEventElapsedTime(&ms_agg, g->ev1, g->ev2);
+    (void)hipEventElapsedTime(&ms_decode, g->ev0, g->ev1);
+    (void)hipEventElapsedTime(&ms_agg, g->ev1, g->ev2);
     s->last_decode_ms = ms_decode;
     s->last_agg_ms = ms_agg;
     s->last_kernel_ms = ms_decode + ms_agg;
