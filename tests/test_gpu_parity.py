@@ -316,3 +316,54 @@ def test_grouped_null_keys_rejected(tmp_path):
         s.stage()
         with pytest.raises(ca.CStripeError, match="non-NULL group key"):
             s.agg_grouped(aggs, (0,))
+
+
+def test_decode_torture_patterns(tmp_path):
+    """Adversarial LZ4 structures through the decode kernels: long RLE runs
+    (match len >> offset), incompressible random bytes (literal-heavy),
+    alternating patterns, tiny repeating dictionaries."""
+    n = 200_000
+    cases = {
+        "rle": np.zeros(n, dtype=np.int64),                       # one giant run
+        "rand": RNG.integers(-2**62, 2**62, n).astype(np.int64),  # incompressible
+        "alt": np.tile(np.array([7, -7], dtype=np.int64), n // 2),
+        "dict": RNG.integers(0, 3, n).astype(np.int64),           # tiny dictionary
+        "ramp": np.arange(n, dtype=np.int64) * 977,
+    }
+    for name, a in cases.items():
+        path = str(tmp_path / f"t_{name}.cs")
+        ca.write_table(path, [("a", ca.I64, 0)], [a], compression=ca.COMP_LZ4)
+        aggs = [(ca.AGG_SUM_I64, 0), (ca.AGG_MIN_I64, 0), (ca.AGG_MAX_I64, 0),
+                (ca.AGG_COUNT_STAR, -1)]
+        op, _, gp, _ = both(path, [], aggs)
+        assert_parity(op, gp, aggs)
+        # and byte-exact decode through next_batch for one chunk
+        with ca.Reader(path) as r, r.scan(cols_mask=1) as s:
+            s.stage()
+            v = np.zeros(10000, dtype=np.int64)
+            e = np.zeros(10000, dtype=np.uint8)
+            res = s.next_batch({0: v}, {0: e})
+            assert res is not None
+            nr, _ = res
+            np.testing.assert_array_equal(v[:nr], a[:nr], err_msg=name)
+
+
+def test_mixed_width_types_gpu(tmp_path):
+    """i8/i16/i32/f32 columns through decode + filter + aggregate."""
+    n = 60_000
+    c16 = RNG.integers(-30000, 30000, n).astype(np.int16)
+    c32 = RNG.integers(-2**30, 2**30, n).astype(np.int32)
+    f32 = RNG.normal(size=n).astype(np.float32)
+    c8 = RNG.integers(-100, 100, n).astype(np.int8)
+    path = str(tmp_path / "mix.cs")
+    ca.write_table(path, [("a", ca.I16, 0), ("b", ca.I32, 0),
+                          ("c", ca.F32, 0), ("d", ca.I8, 0)],
+                   [c16, c32, f32, c8], compression=ca.COMP_LZ4,
+                   chunk_group_row_limit=7000)
+    preds = [(0, ca.PRED_GT, -20000), (2, ca.PRED_LT, 1.5)]
+    aggs = [(ca.AGG_COUNT_STAR, -1), (ca.AGG_SUM_I64, 1),
+            (ca.AGG_MIN_I64, 3), (ca.AGG_MAX_I64, 0),
+            (ca.AGG_SUM_F64, 2), (ca.AGG_MIN_F64, 2)]
+    op, ofilt, gp, gfilt = both(path, preds, aggs)
+    assert ofilt == gfilt
+    assert_parity(op, gp, aggs)
