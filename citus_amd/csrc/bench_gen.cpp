@@ -16,6 +16,10 @@
 #include <cstring>
 #include <vector>
 
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
 static inline uint64_t xs64(uint64_t &x)
 {
     x ^= x << 13;
@@ -48,24 +52,37 @@ extern "C" int csbench_gen_lineitem(const char *path, uint64_t n_rows, uint64_t 
     cstripe_writer *w = cstripe_write_begin(path, cols, 8, &opts);
     if (!w) return CSTRIPE_ERR;
 
-    uint64_t x = seed ? seed : 42;
+    const uint64_t seed0 = seed ? seed : 42;
     const uint64_t BATCH = 1u << 20;
     std::vector<int64_t> c0(BATCH), c1(BATCH), c2(BATCH), c3(BATCH), c4(BATCH), c5(BATCH);
     std::vector<int8_t> c6(BATCH), c7(BATCH);
     uint64_t done = 0, orderkey = 1;
     while (done < n_rows) {
         uint64_t n = n_rows - done < BATCH ? n_rows - done : BATCH;
-        for (uint64_t i = 0; i < n; i++) {
-            c0[i] = (int64_t)orderkey++;
-            c1[i] = 100 + (int64_t)(xs64(x) % 4901);            /* quantity 1.00..50.00 */
-            c2[i] = 900 + (int64_t)(xs64(x) % 104101);          /* price 9.00..1050.00 */
-            c3[i] = (int64_t)(xs64(x) % 11);                    /* discount 0.00..0.10 */
-            c4[i] = (int64_t)(xs64(x) % 9);                     /* tax 0.00..0.08 */
-            c5[i] = 8035 + (int64_t)(xs64(x) % 2557);           /* shipdate 1992..1998 */
-            uint64_t rf = xs64(x) % 4;                          /* A/N 25/25, R 50 */
-            c6[i] = (int8_t)(rf == 0 ? 0 : (rf == 1 ? 1 : 2));
-            c7[i] = (int8_t)(xs64(x) % 2);
+        /* one independent xorshift64 stream per (column, batch) — columns
+         * generate in parallel, deterministically for a given seed */
+        const uint64_t b = done / BATCH;
+        #pragma omp parallel for schedule(static)
+        for (int col = 0; col < 7; col++) {
+            uint64_t x = seed0 * 0x9E3779B97F4A7C15ull + b * 0xBF58476D1CE4E5B9ull
+                         + (uint64_t)(col + 1) * 0x94D049BB133111EBull;
+            xs64(x); xs64(x);                       /* warm the stream */
+            switch (col) {
+                case 0: for (uint64_t i = 0; i < n; i++) c1[i] = 100 + (int64_t)(xs64(x) % 4901); break;
+                case 1: for (uint64_t i = 0; i < n; i++) c2[i] = 900 + (int64_t)(xs64(x) % 104101); break;
+                case 2: for (uint64_t i = 0; i < n; i++) c3[i] = (int64_t)(xs64(x) % 11); break;
+                case 3: for (uint64_t i = 0; i < n; i++) c4[i] = (int64_t)(xs64(x) % 9); break;
+                case 4: for (uint64_t i = 0; i < n; i++) c5[i] = 8035 + (int64_t)(xs64(x) % 2557); break;
+                case 5: for (uint64_t i = 0; i < n; i++) {
+                            uint64_t rf = xs64(x) % 4;
+                            c6[i] = (int8_t)(rf == 0 ? 0 : (rf == 1 ? 1 : 2));
+                        }
+                        break;
+                case 6: for (uint64_t i = 0; i < n; i++) c7[i] = (int8_t)(xs64(x) % 2); break;
+            }
         }
+        for (uint64_t i = 0; i < n; i++) c0[i] = (int64_t)(orderkey + i);
+        orderkey += n;
         const void *vals[8] = {c0.data(), c1.data(), c2.data(), c3.data(),
                                c4.data(), c5.data(), c6.data(), c7.data()};
         int rc = cstripe_write_rows(w, n, vals, nullptr);

@@ -19,6 +19,7 @@
  */
 #include "internal.h"
 #include "compress.h"
+#include "lz4_enc.h"
 
 #include <cstdarg>
 #include <cstdio>
@@ -286,6 +287,9 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     out.segs.clear();
     size_t off = 0;
     bool ok = true;
+    lz4e_state enc;                   /* one init per chunk, reused across
+                                       * segments (see lz4_enc.h) */
+    if (codec == CSTRIPE_COMP_LZ4) lz4e_init(&enc);
     while (off < n) {
         size_t len = std::min(per, n - off);
         if (n - off - len < 16) len = n - off;      /* absorb the tail */
@@ -297,7 +301,10 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
             int bound = LZ4_compressBound((int)len);
             size_t base = out.value_comp.size();
             out.value_comp.resize(base + (size_t)bound);
-            int csz = LZ4_compress_default((const char *)raw.data() + off,
+            int csz = lz4e_compress(&enc, raw.data() + off,
+                                    (int)len, out.value_comp.data() + base, bound);
+            if (csz <= 0)             /* capacity fallback: system liblz4 */
+                csz = LZ4_compress_default((const char *)raw.data() + off,
                                            (char *)out.value_comp.data() + base,
                                            (int)len, bound);
             if (csz <= 0) { ok = false; break; }
