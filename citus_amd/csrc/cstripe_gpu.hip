@@ -266,7 +266,7 @@ __device__ inline uint8_t bs_get(ByteStream &b)
 __global__ void lz4_decode_lane_kernel(
     const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
     const SegDesc *__restrict__ segs, uint32_t n_segs, uint32_t stride,
-    uint32_t in_cap, int *__restrict__ err)
+    int *__restrict__ err)
 {
     extern __shared__ uint8_t sout_all[];
     const uint32_t tid = threadIdx.x;
@@ -276,46 +276,13 @@ __global__ void lz4_decode_lane_kernel(
     /* per-segment (dst_off, decomp_len) cached in LDS for the flush loop */
     uint64_t *sdst = (uint64_t *)(sout_all + (size_t)blockDim.x * stride);
     uint32_t *sdlen = (uint32_t *)(sdst + blockDim.x);
-    /* per-wave compressed-input staging area */
-    uint8_t *sin_all = (uint8_t *)(sdlen + blockDim.x);
 
-    const uint32_t lane = tid % WAVE;
-    const uint32_t wid = tid / WAVE;
-
-    /* stage this wave's 64 compressed streams into LDS with ONE coalesced
-     * copy of their (contiguous) span — the per-lane register-window refills
-     * then read LDS instead of issuing TA-bound scattered gathers */
-    SegDesc s{};
-    uint64_t sbeg = ~0ull, send64 = 0;
     if (gseg < n_segs) {
-        s = segs[gseg];
+        const SegDesc s = segs[gseg];
         sdst[tid] = s.dst_off;
         sdlen[tid] = s.decomp_len;
-        sbeg = s.src_off;
-        send64 = s.src_off + s.comp_len;
-    }
-    for (int d = WAVE / 2; d > 0; d >>= 1) {
-        uint64_t o = (uint64_t)__shfl_down((long long)sbeg, d, WAVE);
-        uint64_t e = (uint64_t)__shfl_down((long long)send64, d, WAVE);
-        if (o < sbeg) sbeg = o;
-        if (e > send64) send64 = e;
-    }
-    sbeg = (uint64_t)__shfl((long long)sbeg, 0, WAVE) & ~7ull;
-    send64 = (uint64_t)__shfl((long long)send64, 0, WAVE);
-    uint8_t *win = sin_all + (size_t)wid * in_cap;
-    const bool lds_in = (send64 > sbeg) && (send64 - sbeg + 40 <= in_cap);
-    if (lds_in) {
-        const uint32_t words = (uint32_t)((send64 - sbeg + 40 + 3) >> 2);
-        const uint32_t *g = (const uint32_t *)(data + sbeg);
-        uint32_t *l = (uint32_t *)win;
-        for (uint32_t j = lane; j < words; j += WAVE) l[j] = g[j];
-        __builtin_amdgcn_s_waitcnt(0);   /* drain lgkm+vm: LDS writes visible wave-wide */
-    }
-
-    if (gseg < n_segs) {
         ByteStream bs;
-        if (lds_in) bs_init(bs, win, s.src_off - sbeg);
-        else bs_init(bs, data, s.src_off);
+        bs_init(bs, data, s.src_off);
         const uint32_t send = bs.pos + s.comp_len;   /* end position */
         const uint32_t dlen = s.decomp_len;
         uint32_t op = 0;
@@ -1363,17 +1330,10 @@ static void launch_decode(cs_gpu_state *g)
         else if (g->max_seg_dlen <= 531) { block = 256; stride = 532; }
         else                             { block = 128; stride = 1044; }
         uint32_t grid = (g->n_segs + block - 1) / block;
-        /* per-wave input staging: size for 64 worst-case segments + slack,
-         * bounded by the LDS budget (fallback to global reads per wave) */
-        uint32_t in_cap = 64 * ((g->max_seg_comp + 15) & ~15u) + 64;
-        uint32_t lds_fixed = block * stride + block * 16;
-        uint32_t waves = block / 64;
-        if (lds_fixed + waves * in_cap > 160 * 1024)
-            in_cap = ((160 * 1024 - lds_fixed) / waves) & ~7u;
         hipLaunchKernelGGL(lz4_decode_lane_kernel, dim3(grid), dim3(block),
-                           lds_fixed + waves * in_cap, g->stream,
+                           block * stride + block * 16, g->stream,
                            g->d_data, g->d_scratch, g->d_segs, g->n_segs, stride,
-                           in_cap, g->d_error);
+                           g->d_error);
         return;
     }
     uint32_t in_cap = (g->max_seg_comp + 8 + 15) & ~15u;
