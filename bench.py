@@ -81,6 +81,8 @@ def main():
     ap.add_argument("--query", default="q6", choices=["q6", "q1", "count"])
     ap.add_argument("--seg-bytes", type=int, default=0,
                     help="LZ4 micro-segment size override (bytes)")
+    ap.add_argument("--min-match", type=int, default=0,
+                    help="writer LZ4 min match length (>=4; GPU-decode knob)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -104,11 +106,12 @@ def main():
     # ---- setup (untimed): generate per-rank shard, open, prune, stage ----
     cache = os.environ.get("CSTRIPE_BENCH_DIR", "/tmp/cstripe_bench")
     os.makedirs(cache, exist_ok=True)
-    shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}_r{rank}.cs")
+    shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}"
+                                f"_m{args.min_match}_r{rank}.cs")
     t0 = time.time()
     if not os.path.exists(shard):
         ca.gen_lineitem(shard, args.rows, seed=42 + rank, compression=comp,
-                        seg_bytes=args.seg_bytes)
+                        seg_bytes=args.seg_bytes, min_match=args.min_match)
     gen_s = time.time() - t0
 
     reader = ca.Reader(shard)

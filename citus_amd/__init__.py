@@ -43,7 +43,8 @@ class Options(C.Structure):
                 ("chunk_group_row_limit", C.c_uint32),
                 ("compression", C.c_uint8), ("compression_level", C.c_int8),
                 ("lz4_seg_target_kb", C.c_uint16),
-                ("lz4_seg_target_bytes", C.c_uint32)]
+                ("lz4_seg_target_bytes", C.c_uint32),
+                ("lz4_min_match", C.c_uint8), ("_pad2", C.c_uint8 * 7)]
 
 
 class Pred(C.Structure):
@@ -124,6 +125,9 @@ _gpu_available = _sig("cstripe_gpu_available", C.c_int, [])
 _gen_lineitem = _sig("csbench_gen_lineitem", C.c_int,
                      [C.c_char_p, C.c_uint64, C.c_uint64, C.c_int, C.c_int, C.c_int,
                       C.c_uint64, C.c_uint32])
+_gen_lineitem_mm = _sig("csbench_gen_lineitem_mm", C.c_int,
+                        [C.c_char_p, C.c_uint64, C.c_uint64, C.c_int, C.c_int, C.c_int,
+                         C.c_uint64, C.c_uint32, C.c_int])
 
 
 def errmsg():
@@ -188,11 +192,11 @@ def write_table(path, defs, columns, nulls=None, **opt_kw):
 
 
 def gen_lineitem(path, n_rows, seed=42, compression=COMP_LZ4, level=3, seg_kb=0,
-                 stripe_rows=0, chunk_rows=0, seg_bytes=0):
+                 stripe_rows=0, chunk_rows=0, seg_bytes=0, min_match=0):
     if seg_bytes:
         seg_kb = -int(seg_bytes)      # negative seg_kb = bytes (csbench_gen_lineitem)
-    _check(_gen_lineitem(path.encode(), n_rows, seed, compression, level, seg_kb,
-                         stripe_rows, chunk_rows), "gen_lineitem")
+    _check(_gen_lineitem_mm(path.encode(), n_rows, seed, compression, level, seg_kb,
+                            stripe_rows, chunk_rows, min_match), "gen_lineitem")
 
 
 LINEITEM_COLS = {"l_orderkey": 0, "l_quantity": 1, "l_extendedprice": 2,

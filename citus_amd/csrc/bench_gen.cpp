@@ -20,6 +20,11 @@
 #include <omp.h>
 #endif
 
+extern "C" int csbench_gen_lineitem_mm(const char *path, uint64_t n_rows, uint64_t seed,
+                                       int compression, int level, int seg_kb,
+                                       uint64_t stripe_rows, uint32_t chunk_rows,
+                                       int min_match);
+
 static inline uint64_t xs64(uint64_t &x)
 {
     x ^= x << 13;
@@ -31,6 +36,15 @@ static inline uint64_t xs64(uint64_t &x)
 extern "C" int csbench_gen_lineitem(const char *path, uint64_t n_rows, uint64_t seed,
                                     int compression, int level, int seg_kb,
                                     uint64_t stripe_rows, uint32_t chunk_rows)
+{
+    return csbench_gen_lineitem_mm(path, n_rows, seed, compression, level, seg_kb,
+                                   stripe_rows, chunk_rows, 0);
+}
+
+extern "C" int csbench_gen_lineitem_mm(const char *path, uint64_t n_rows, uint64_t seed,
+                                       int compression, int level, int seg_kb,
+                                       uint64_t stripe_rows, uint32_t chunk_rows,
+                                       int min_match)
 {
     cstripe_coldef cols[8] = {};
     const char *names[8] = {"l_orderkey", "l_quantity", "l_extendedprice", "l_discount",
@@ -48,6 +62,7 @@ extern "C" int csbench_gen_lineitem(const char *path, uint64_t n_rows, uint64_t 
     opts.compression_level = (int8_t)level;
     if (seg_kb > 0) { opts.lz4_seg_target_kb = (uint16_t)seg_kb; opts.lz4_seg_target_bytes = 0; }
     else if (seg_kb < 0) { opts.lz4_seg_target_kb = 0; opts.lz4_seg_target_bytes = (uint32_t)(-seg_kb); }
+    if (min_match >= 4) opts.lz4_min_match = (uint8_t)min_match;
 
     cstripe_writer *w = cstripe_write_begin(path, cols, 8, &opts);
     if (!w) return CSTRIPE_ERR;

@@ -263,6 +263,14 @@ __device__ inline uint8_t bs_get(ByteStream &b)
     return (uint8_t)(b.w0 >> (8 * rel));
 }
 
+/* next 8 bytes at pos as a little-endian u64 (rel in [0,8) after bs_norm) */
+__device__ inline uint64_t bs_peek64(ByteStream &b)
+{
+    bs_norm(b);
+    uint32_t s = 8 * (b.pos - b.woff);
+    return s ? ((b.w0 >> s) | (b.w1 << (64 - s))) : b.w0;
+}
+
 __global__ void lz4_decode_lane_kernel(
     const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
     const SegDesc *__restrict__ segs, uint32_t n_segs, uint32_t stride,
@@ -295,8 +303,21 @@ __global__ void lz4_decode_lane_kernel(
                 do { b = bs_get(bs); litlen += b; } while (b == 255 && bs.pos < send);
             }
             if (bs.pos + litlen > send || op + litlen > dlen) { bad = true; break; }
-            for (uint32_t j = 0; j < litlen; j++) sout[op + j] = bs_get(bs);
-            op += litlen;
+            {   /* literal copy: 8 bytes per window peek, then byte tail */
+                uint32_t left = litlen;
+                while (left >= 8) {
+                    uint64_t v = bs_peek64(bs);
+                    bs.pos += 8;
+                    sout[op + 0] = (uint8_t)v;       sout[op + 1] = (uint8_t)(v >> 8);
+                    sout[op + 2] = (uint8_t)(v >> 16); sout[op + 3] = (uint8_t)(v >> 24);
+                    sout[op + 4] = (uint8_t)(v >> 32); sout[op + 5] = (uint8_t)(v >> 40);
+                    sout[op + 6] = (uint8_t)(v >> 48); sout[op + 7] = (uint8_t)(v >> 56);
+                    op += 8;
+                    left -= 8;
+                }
+                for (uint32_t j = 0; j < left; j++) sout[op + j] = bs_get(bs);
+                op += left;
+            }
             if (bs.pos >= send) break;       /* last sequence: literals only */
 
             if (bs.pos + 2 > send) { bad = true; break; }

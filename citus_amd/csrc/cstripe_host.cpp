@@ -61,6 +61,7 @@ extern "C" void cstripe_default_options(cstripe_options *o)
     o->compression_level = 3;
     o->lz4_seg_target_kb = 0;
     o->lz4_seg_target_bytes = 256;
+    o->lz4_min_match = 4;              /* standard greedy LZ4 parse */
 }
 
 static uint32_t type_width(uint8_t t) { return csf_type_width(t); }
@@ -301,8 +302,9 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
             int bound = LZ4_compressBound((int)len);
             size_t base = out.value_comp.size();
             out.value_comp.resize(base + (size_t)bound);
-            int csz = lz4e_compress(&enc, raw.data() + off,
-                                    (int)len, out.value_comp.data() + base, bound);
+            int csz = lz4e_compress_mm(&enc, raw.data() + off,
+                                       (int)len, out.value_comp.data() + base, bound,
+                                       w->opts.lz4_min_match >= 4 ? w->opts.lz4_min_match : 4);
             if (csz <= 0)             /* capacity fallback: system liblz4 */
                 csz = LZ4_compress_default((const char *)raw.data() + off,
                                            (char *)out.value_comp.data() + base,

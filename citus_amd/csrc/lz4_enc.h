@@ -45,9 +45,12 @@ static inline uint32_t lz4e_read32(const uint8_t *p)
     return v;
 }
 
-/* returns compressed size, or 0 if dst capacity insufficient */
-static inline int lz4e_compress(lz4e_state *st, const uint8_t *src, int slen,
-                                uint8_t *dst, int dcap)
+/* returns compressed size, or 0 if dst capacity insufficient.
+ * min_match (>= 4): matches shorter than this are skipped — fewer, longer
+ * sequences decode much faster on the lane-parallel GPU decoder at a small
+ * ratio cost (the output is standard LZ4 either way). */
+static inline int lz4e_compress_mm(lz4e_state *st, const uint8_t *src, int slen,
+                                   uint8_t *dst, int dcap, int min_match)
 {
     const uint32_t base = st->base;
     st->base += (uint32_t)slen + 1;
@@ -73,6 +76,7 @@ static inline int lz4e_compress(lz4e_state *st, const uint8_t *src, int slen,
                 int mlen = 4;
                 while (pos + mlen < matchlimit && src[mpos + mlen] == src[pos + mlen])
                     mlen++;
+                if (mlen < min_match) { pos++; continue; }
                 int litlen = pos - anchor;
                 /* emit: token + ext + literals + offset + ext */
                 uint8_t *tok = op++;
@@ -126,6 +130,12 @@ static inline int lz4e_compress(lz4e_state *st, const uint8_t *src, int slen,
         op += litlen;
     }
     return (int)(op - dst);
+}
+
+static inline int lz4e_compress(lz4e_state *st, const uint8_t *src, int slen,
+                                uint8_t *dst, int dcap)
+{
+    return lz4e_compress_mm(st, src, slen, dst, dcap, 4);
 }
 
 #endif

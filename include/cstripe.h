@@ -114,6 +114,12 @@ typedef struct cstripe_options {
     uint32_t    lz4_seg_target_bytes;  /* fine knob, overrides kb when nonzero;
                                         * default 256 B = one GPU lane per
                                         * segment (lane-parallel decode) */
+    uint8_t     lz4_min_match;         /* shortest match the writer's LZ4
+                                        * encoder emits (>=4; 0 => default).
+                                        * Larger = literal-heavier standard-LZ4
+                                        * streams that decode faster on GPU at
+                                        * a small compression-ratio cost */
+    uint8_t     _pad2[7];
 } cstripe_options;
 
 void cstripe_default_options(cstripe_options *opts);
