@@ -81,6 +81,8 @@ struct AggParams {
     AggD aggs[MAX_AGGS];
 };
 
+struct FusedTile;
+
 /* per-block / final accumulator cell (32 B) */
 struct AccCell {
     int64_t lo;             /* i128 low  / i64 min-max */
@@ -114,6 +116,9 @@ struct cs_gpu_state {
     uint32_t max_seg_comp = 0;
     uint32_t max_seg_dlen = 0;
     bool segs_16aligned = true;
+    bool fusable = true;             /* all proj cols dense i64, uniform 256B lz4 segs */
+    struct FusedTile *d_tiles = nullptr;
+    uint32_t n_tiles = 0;
     uint32_t n_groups = 0;
     uint32_t n_proj = 0;
     uint32_t max_blocks = 0;
@@ -271,24 +276,13 @@ __device__ inline uint64_t bs_peek64(ByteStream &b)
     return s ? ((b.w0 >> s) | (b.w1 << (64 - s))) : b.w0;
 }
 
-__global__ void lz4_decode_lane_kernel(
-    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
-    const SegDesc *__restrict__ segs, uint32_t n_segs, uint32_t stride,
-    int *__restrict__ err)
+/* decode ONE segment, per-lane serial, output to this lane's LDS region.
+ * Shared by the standalone decode kernel and the fused decode+agg kernel. */
+__device__ inline void lz4_lane_decode(const uint8_t *__restrict__ data,
+                                       const SegDesc &s, uint8_t *__restrict__ sout,
+                                       int *__restrict__ err)
 {
-    extern __shared__ uint8_t sout_all[];
-    const uint32_t tid = threadIdx.x;
-    const uint32_t first = blockIdx.x * blockDim.x;
-    const uint32_t gseg = first + tid;
-    uint8_t *sout = sout_all + (size_t)tid * stride;
-    /* per-segment (dst_off, decomp_len) cached in LDS for the flush loop */
-    uint64_t *sdst = (uint64_t *)(sout_all + (size_t)blockDim.x * stride);
-    uint32_t *sdlen = (uint32_t *)(sdst + blockDim.x);
-
-    if (gseg < n_segs) {
-        const SegDesc s = segs[gseg];
-        sdst[tid] = s.dst_off;
-        sdlen[tid] = s.decomp_len;
+    {
         ByteStream bs;
         bs_init(bs, data, s.src_off);
         const uint32_t send = bs.pos + s.comp_len;   /* end position */
@@ -351,6 +345,28 @@ __global__ void lz4_decode_lane_kernel(
             op += mlen;
         }
         if (bad || op != dlen) atomicOr(err, 4);
+    }
+}
+
+__global__ void lz4_decode_lane_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, uint32_t n_segs, uint32_t stride,
+    int *__restrict__ err)
+{
+    extern __shared__ uint8_t sout_all[];
+    const uint32_t tid = threadIdx.x;
+    const uint32_t first = blockIdx.x * blockDim.x;
+    const uint32_t gseg = first + tid;
+    uint8_t *sout = sout_all + (size_t)tid * stride;
+    /* per-segment (dst_off, decomp_len) cached in LDS for the flush loop */
+    uint64_t *sdst = (uint64_t *)(sout_all + (size_t)blockDim.x * stride);
+    uint32_t *sdlen = (uint32_t *)(sdst + blockDim.x);
+
+    if (gseg < n_segs) {
+        const SegDesc s = segs[gseg];
+        sdst[tid] = s.dst_off;
+        sdlen[tid] = s.decomp_len;
+        lz4_lane_decode(data, s, sout, err);
     }
     __syncthreads();
 
@@ -655,6 +671,114 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     }
 }
 
+
+/* =====================================================================
+ * FUSED decode -> filter -> partial aggregate: one block
+ * decodes a 2048-row tile of ALL projected columns into LDS (one lane per
+ * 256 B segment — 64 segments/column/tile, up to 4 i64 columns = 256 lanes)
+ * and aggregates straight from LDS. The decompressed stream never touches
+ * HBM: eliminates the scratch write + re-read of the two-kernel path.
+ * Eligible when every projected column is dense int64 in uniform 256 B LZ4
+ * segments (the writer default); otherwise the two-kernel path runs.
+ * ===================================================================== */
+
+#define FUSE_TILE_ROWS 2048
+#define FUSE_STRIDE 280          /* 8B-aligned per-lane region (val() u64 reads) */
+
+struct FusedTile {
+    uint32_t row_count;          /* rows in this tile (<= FUSE_TILE_ROWS) */
+    uint32_t seg_base[4];        /* d_segs index of tile's first segment per proj col */
+};
+
+template <int NPREDS, int NAGGS>
+__global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
+    const uint8_t *__restrict__ data, const SegDesc *__restrict__ segs,
+    const FusedTile *__restrict__ tiles, AccCell *__restrict__ block_out,
+    int *__restrict__ err, const AggParams params)
+{
+    const uint32_t n_preds = NPREDS >= 0 ? (uint32_t)NPREDS : params.n_preds;
+    const uint32_t n_aggs = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
+    extern __shared__ uint8_t lds[];                 /* 256 * FUSE_STRIDE */
+    const uint32_t tid = threadIdx.x;
+    const FusedTile t = tiles[blockIdx.x];
+
+    /* phase 1: decode my segment (lane = col*64 + seg_in_tile) */
+    const uint32_t c = tid >> 6;
+    const uint32_t sseg = tid & 63;
+    const uint32_t nsegs = (t.row_count * 8 + 255) >> 8;
+    if (c < params.n_proj && sseg < nsegs) {
+        const SegDesc sd = segs[t.seg_base[c] + sseg];
+        lz4_lane_decode(data, sd, lds + (size_t)tid * FUSE_STRIDE, err);
+    }
+    __syncthreads();
+
+    /* phase 2: filter + aggregate from LDS */
+    ThreadAcc acc[NAGGS >= 0 ? NAGGS : MAX_AGGS];
+    for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
+
+    auto val = [&](uint32_t proj, uint32_t row) -> int64_t {
+        const uint32_t region = proj * 64 + (row >> 5);
+        return *(const int64_t *)(lds + (size_t)region * FUSE_STRIDE + ((row & 31) << 3));
+    };
+
+    for (uint32_t row = tid; row < t.row_count; row += AGG_BLOCK) {
+        bool pass = true;
+        int last_proj = -1;
+        int64_t liv = 0;
+        for (uint32_t p = 0; p < n_preds; p++) {
+            const PredD &pr = params.preds[p];
+            if ((int)pr.proj != last_proj) { liv = val(pr.proj, row); last_proj = (int)pr.proj; }
+            pass = pass & pred_eval(pr, liv, 0.0);   /* cols are i64: is_float==0 */
+        }
+        if (!pass) continue;
+        for (uint32_t a = 0; a < n_aggs; a++) {
+            const AggD &g = params.aggs[a];
+            ThreadAcc &A = acc[a];
+            switch (g.kind) {
+                case CSTRIPE_AGG_COUNT_STAR:
+                case CSTRIPE_AGG_COUNT_COL:          /* dense: col always present */
+                    A.cnt++; A.lo++; break;
+                case CSTRIPE_AGG_SUM_I64:
+                    acc_add_i128(A, (__int128)val(g.proj_a, row)); A.cnt++; break;
+                case CSTRIPE_AGG_MIN_I64:
+                    A.lo = min(A.lo, val(g.proj_a, row)); A.cnt++; break;
+                case CSTRIPE_AGG_MAX_I64:
+                    A.lo = max(A.lo, val(g.proj_a, row)); A.cnt++; break;
+                case CSTRIPE_AGG_SUM_PROD_I64:
+                    acc_add_i128(A, (__int128)val(g.proj_a, row) * val(g.proj_b, row));
+                    A.cnt++; break;
+                case CSTRIPE_AGG_SUM_DISC_I64:
+                    acc_add_i128(A, (__int128)val(g.proj_a, row) * (g.one - val(g.proj_b, row)));
+                    A.cnt++; break;
+                case CSTRIPE_AGG_SUM_DISC_TAX_I64:
+                    acc_add_i128(A, (__int128)val(g.proj_a, row) * (g.one - val(g.proj_b, row))
+                                     * (g.one + val(g.proj_c, row)));
+                    A.cnt++; break;
+                default: break;                       /* host guards float kinds */
+            }
+        }
+    }
+
+    /* reduce: wave shuffle -> cross-wave LDS -> one AccCell per block */
+    __shared__ ThreadAcc lred[AGG_BLOCK / WAVE][MAX_AGGS];
+    const uint32_t wid = tid / WAVE;
+    const uint32_t lane = tid % WAVE;
+    for (uint32_t a = 0; a < n_aggs; a++) {
+        wave_reduce(acc[a], params.aggs[a].kind);
+        if (lane == 0) lred[wid][a] = acc[a];
+    }
+    __syncthreads();
+    if (wid == 0) {
+        for (uint32_t a = lane; a < n_aggs; a += WAVE) {
+            ThreadAcc r = lred[0][a];
+            for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
+                acc_merge(r, lred[w][a], params.aggs[a].kind);
+            AccCell cell;
+            cell.lo = r.lo; cell.hi = r.hi; cell.f = r.f; cell.cnt = r.cnt;
+            block_out[(uint64_t)blockIdx.x * n_aggs + a] = cell;
+        }
+    }
+}
 
 /* per-row precomputed aggregate contribution: operands loaded ONCE per row,
  * so the per-distinct-key reduce rounds touch registers only */
@@ -1132,6 +1256,7 @@ void csgpu_release(cstripe_scan *s)
     if (g->d_gfkeys) HIP_DROP(hipFree(g->d_gfkeys));
     if (g->d_gfcells) HIP_DROP(hipFree(g->d_gfcells));
     if (g->d_gn) HIP_DROP(hipFree(g->d_gn));
+    if (g->d_tiles) HIP_DROP(hipFree(g->d_tiles));
     if (g->d_error) HIP_DROP(hipFree(g->d_error));
     if (g->ev0) HIP_DROP(hipEventDestroy(g->ev0));
     if (g->ev1) HIP_DROP(hipEventDestroy(g->ev1));
@@ -1204,7 +1329,11 @@ int csgpu_stage(cstripe_scan *s, int device_id)
 
     uint32_t tiles_pg = (r->head.chunk_row_limit + TILE_ROWS - 1) / TILE_ROWS;
     if (tiles_pg == 0) tiles_pg = 1;
+    uint64_t n_tiles_max = 0;
+    for (const auto &sc : s->sel)
+        n_tiles_max += (r->stripes[sc.stripe].group_rows[sc.chunk] + 2047) / 2048;
     uint64_t max_blocks = (uint64_t)g->n_groups * tiles_pg;
+    if (n_tiles_max > max_blocks) max_blocks = n_tiles_max;
     g->max_blocks = (uint32_t)max_blocks;
 
     HIP_TRY(hipStreamCreate(&g->stream));
@@ -1231,6 +1360,8 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     h_rank.reserve(rank_words);
     std::vector<SegDesc> h_segs;
     h_segs.reserve(n_segs);
+    std::vector<uint32_t> seg_start((uint64_t)g->n_groups * n_proj, 0);
+    if (n_proj > 4) g->fusable = false;
     std::vector<GroupDesc> h_groups(g->n_groups);
     std::vector<ColLoc> h_colloc((uint64_t)g->n_groups * n_proj);
     g->scratch_off.assign((uint64_t)g->n_groups * n_proj, ~0ull);
@@ -1275,6 +1406,9 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 cl.flags |= 2;
             }
 
+            if (!(cl.flags & 2) || cl.type != CSTRIPE_I64 ||
+                nd.n.comp_type != CSTRIPE_COMP_LZ4)
+                g->fusable = false;
             if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
                 dpos = align_up(dpos, 16);
                 memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
@@ -1282,6 +1416,15 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 cl.val_off = spos;
                 cl.flags |= 1;
                 g->scratch_off[(uint64_t)gi * n_proj + pj] = spos;
+                seg_start[(uint64_t)gi * n_proj + pj] = (uint32_t)h_segs.size();
+                uint32_t segi = 0;
+                const auto &seglist = st.nodes[c][sc.chunk].segs;
+                for (const csf_seg &sg : seglist) {
+                    if (sg.decomp_off != segi * 256u ||
+                        (segi + 1 < seglist.size() && sg.decomp_len != 256))
+                        g->fusable = false;
+                    segi++;
+                }
                 for (const csf_seg &sg : st.nodes[c][sc.chunk].segs) {
                     SegDesc sd;
                     sd.src_off = dpos + sg.comp_off;
@@ -1331,6 +1474,28 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         HIP_TRY(hipMemcpyAsync(g->d_segs, h_segs.data(), h_segs.size() * sizeof(SegDesc), hipMemcpyHostToDevice, g->stream));
     HIP_TRY(hipMemcpyAsync(g->d_groups, h_groups.data(), h_groups.size() * sizeof(GroupDesc), hipMemcpyHostToDevice, g->stream));
     HIP_TRY(hipMemcpyAsync(g->d_colloc, h_colloc.data(), h_colloc.size() * sizeof(ColLoc), hipMemcpyHostToDevice, g->stream));
+    if (g->fusable && g->n_groups > 0) {
+        std::vector<FusedTile> h_tiles;
+        h_tiles.reserve(n_tiles_max);
+        for (uint32_t gi = 0; gi < g->n_groups; gi++) {
+            const cs_selchunk &sc = s->sel[gi];
+            uint32_t rows = r->stripes[sc.stripe].group_rows[sc.chunk];
+            uint32_t ntile = (rows + FUSE_TILE_ROWS - 1) / FUSE_TILE_ROWS;
+            for (uint32_t k = 0; k < ntile; k++) {
+                FusedTile ft{};
+                ft.row_count = (k + 1 < ntile) ? FUSE_TILE_ROWS
+                                               : rows - k * FUSE_TILE_ROWS;
+                for (uint32_t pj = 0; pj < n_proj; pj++)
+                    ft.seg_base[pj] = seg_start[(uint64_t)gi * n_proj + pj] + 64 * k;
+                h_tiles.push_back(ft);
+            }
+        }
+        g->n_tiles = (uint32_t)h_tiles.size();
+        HIP_TRY(hipMalloc(&g->d_tiles, h_tiles.size() * sizeof(FusedTile)));
+        HIP_TRY(hipMemcpyAsync(g->d_tiles, h_tiles.data(),
+                               h_tiles.size() * sizeof(FusedTile),
+                               hipMemcpyHostToDevice, g->stream));
+    }
     HIP_TRY(hipStreamSynchronize(g->stream));
     g->colloc_host = h_colloc;
     return CSTRIPE_OK;
@@ -1532,7 +1697,68 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         return CSTRIPE_OK;
     }
 
+    bool int_aggs = true;
+    for (uint32_t a = 0; a < n_aggs; a++)
+        if (aggs[a].kind == CSTRIPE_AGG_SUM_F64 || aggs[a].kind == CSTRIPE_AGG_MIN_F64 ||
+            aggs[a].kind == CSTRIPE_AGG_MAX_F64)
+            int_aggs = false;
+
     HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
+    if (g->fusable && int_aggs && g->d_tiles) {
+        /* fused decode+filter+aggregate: no scratch round trip */
+        HIP_TRY(hipEventRecord(g->ev0, g->stream));
+        auto launchf = [&](auto *kern) {
+            hipLaunchKernelGGL(kern, dim3(g->n_tiles), dim3(AGG_BLOCK),
+                               AGG_BLOCK * FUSE_STRIDE, g->stream,
+                               g->d_data, g->d_segs, g->d_tiles, g->d_block,
+                               g->d_error, p);
+        };
+        if (p.n_preds == 5 && n_aggs == 2) launchf(fused_agg_kernel<5, 2>);
+        else if (p.n_preds == 5 && n_aggs == 1) launchf(fused_agg_kernel<5, 1>);
+        else if (p.n_preds == 1 && n_aggs == 1) launchf(fused_agg_kernel<1, 1>);
+        else if (p.n_preds == 2 && n_aggs == 2) launchf(fused_agg_kernel<2, 2>);
+        else launchf(fused_agg_kernel<-1, -1>);
+        HIP_TRY(hipEventRecord(g->ev1, g->stream));
+        hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
+                           g->d_block, g->n_tiles, g->d_final, p);
+        HIP_TRY(hipEventRecord(g->ev2, g->stream));
+
+        AccCell h_final[MAX_AGGS];
+        int h_err = 0;
+        HIP_TRY(hipMemcpyAsync(h_final, g->d_final, n_aggs * sizeof(AccCell), hipMemcpyDeviceToHost, g->stream));
+        HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
+        HIP_TRY(hipStreamSynchronize(g->stream));
+        if (h_err) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+
+        float ms_fused = 0, ms_red = 0;
+        (void)hipEventElapsedTime(&ms_fused, g->ev0, g->ev1);
+        (void)hipEventElapsedTime(&ms_red, g->ev1, g->ev2);
+        s->last_decode_ms = ms_fused;     /* the fused kernel (decode dominates) */
+        s->last_agg_ms = ms_red;
+        s->last_kernel_ms = ms_fused + ms_red;
+
+        for (uint32_t a = 0; a < n_aggs; a++) {
+            cstripe_partial o{};
+            const AccCell &cc = h_final[a];
+            o.count = cc.cnt;
+            o.is_null = (cc.cnt == 0) ? 1 : 0;
+            if (o.is_null && aggs[a].kind != CSTRIPE_AGG_COUNT_STAR &&
+                aggs[a].kind != CSTRIPE_AGG_COUNT_COL) { out[a] = o; continue; }
+            switch (aggs[a].kind) {
+                case CSTRIPE_AGG_COUNT_STAR:
+                case CSTRIPE_AGG_COUNT_COL:
+                    o.i128_lo = cc.cnt; o.is_null = 0; break;
+                case CSTRIPE_AGG_MIN_I64:
+                case CSTRIPE_AGG_MAX_I64:
+                    o.i128_lo = cc.lo; o.i128_hi = cc.lo < 0 ? -1 : 0; break;
+                default:
+                    o.i128_lo = cc.lo; o.i128_hi = cc.hi; break;
+            }
+            out[a] = o;
+        }
+        return CSTRIPE_OK;
+    }
+
     HIP_TRY(hipEventRecord(g->ev0, g->stream));
     launch_decode(g);
     HIP_TRY(hipEventRecord(g->ev1, g->stream));
