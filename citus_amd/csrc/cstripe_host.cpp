@@ -279,10 +279,10 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
      * lets the GPU flush decoded segments with aligned 16 B stores) */
     (void)width;
     size_t n = raw.size();
-    size_t nseg = (n + target - 1) / target;
-    if (nseg < 1) nseg = 1;
-    size_t per = ((n / nseg) / 16) * 16;
-    if (per == 0) { per = n; nseg = 1; }
+    size_t per = (target / 16) * 16;    /* exact target (16B-aligned splits) —
+                                         * uniform segments are what the fused
+                                         * GPU kernel's tile math assumes */
+    if (per == 0) per = n;
 
     out.value_comp.clear();
     out.segs.clear();
