@@ -1420,8 +1420,11 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 uint32_t segi = 0;
                 const auto &seglist = st.nodes[c][sc.chunk].segs;
                 for (const csf_seg &sg : seglist) {
+                    /* non-final segments exactly 256 B; the final one must
+                     * still fit a FUSE_STRIDE LDS region */
                     if (sg.decomp_off != segi * 256u ||
-                        (segi + 1 < seglist.size() && sg.decomp_len != 256))
+                        (segi + 1 < seglist.size() ? sg.decomp_len != 256
+                                                   : sg.decomp_len > FUSE_STRIDE - 8))
                         g->fusable = false;
                     segi++;
                 }
@@ -1714,6 +1717,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
                                g->d_error, p);
         };
         if (p.n_preds == 5 && n_aggs == 2) launchf(fused_agg_kernel<5, 2>);
+        else if (p.n_preds == 5 && n_aggs == 4) launchf(fused_agg_kernel<5, 4>);
         else if (p.n_preds == 5 && n_aggs == 1) launchf(fused_agg_kernel<5, 1>);
         else if (p.n_preds == 1 && n_aggs == 1) launchf(fused_agg_kernel<1, 1>);
         else if (p.n_preds == 2 && n_aggs == 2) launchf(fused_agg_kernel<2, 2>);
