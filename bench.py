@@ -205,6 +205,7 @@ def main():
 
     # ---- CPU baseline (oracle restatement, rank 0, N=1 only) ----
     cpu_baseline = None
+    cpu_mt = None
     parity = None
     if world == 1 and not args.no_cpu_baseline:
         import oracle
@@ -222,6 +223,13 @@ def main():
             else:
                 cpu_parts, _ = t.scan_agg(preds, aggs)
             cpu_s = time.time() - t0
+            cpu_mt = None
+            if not group_cols:
+                t0 = time.time()
+                _mt_parts, mt_cores = t.scan_agg_mt(preds, aggs)
+                cpu_mt = {"value": sample_rows / (time.time() - t0),
+                          "unit": "rows/s", "cores": mt_cores, "kind": "port",
+                          "sample": f"all-core, same {sample_rows / 1e6:.0f}M-row pass"}
         cpu_baseline = {
             "value": sample_rows / cpu_s,
             "unit": "rows/s",
@@ -282,6 +290,7 @@ def main():
             "kernel": dominant,
         },
         "cpu_baseline": cpu_baseline,
+        "cpu_baseline_allcore": cpu_mt if (world == 1 and not args.no_cpu_baseline) else None,
     }
     print(json.dumps(out))
 

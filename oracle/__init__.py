@@ -29,6 +29,10 @@ _lib.oracle_column_count.argtypes = [C.c_void_p]
 _lib.oracle_read_chunk.restype = C.c_int
 _lib.oracle_read_chunk.argtypes = [C.c_void_p, C.c_uint32, C.c_uint32, C.c_uint32,
                                    C.c_void_p, C.c_void_p]
+_lib.oracle_scan_agg_mt.restype = C.c_int
+_lib.oracle_scan_agg_mt.argtypes = [C.c_void_p, C.POINTER(Pred), C.c_uint32,
+                                    C.POINTER(AggSpec), C.c_uint32,
+                                    C.POINTER(Partial), C.POINTER(C.c_int)]
 _lib.oracle_scan_agg.restype = C.c_int
 _lib.oracle_scan_agg.argtypes = [C.c_void_p, C.c_uint64, C.POINTER(Pred), C.c_uint32,
                                  C.POINTER(AggSpec), C.c_uint32,
@@ -92,3 +96,15 @@ class OracleTable:
                 res[key] = [out[g * len(aggs) + a] for a in range(len(aggs))]
             return res, filt.value
         return [out[a] for a in range(len(aggs))], filt.value
+
+    def scan_agg_mt(self, preds, aggs):
+        """all-core variant (OpenMP over chunk groups) -> (partials, cores)"""
+        parr = make_preds(preds)
+        aarr = make_aggs(aggs)
+        out = (Partial * len(aggs))()
+        cores = C.c_int(0)
+        rc = _lib.oracle_scan_agg_mt(self._h, parr, len(preds), aarr, len(aggs),
+                                     out, C.byref(cores))
+        if rc != 0:
+            raise RuntimeError(f"oracle_scan_agg_mt rc={rc}")
+        return [out[a] for a in range(len(aggs))], cores.value

@@ -107,3 +107,21 @@ def test_all_null_chunk_never_refuted(tmp_path):
         assert filtered == 0                 # no min/max -> never refuted
         assert parts[0].count == 0           # NULL fails the qual per-row
         assert parts[1].count == 0
+
+
+def test_mt_oracle_matches_serial(tmp_path):
+    """all-core oracle variant: identical integer results to the serial path."""
+    import citus_amd as ca2
+    path = str(tmp_path / "mt.cs")
+    ca2.gen_lineitem(path, 300_000)
+    preds = [(5, ca2.PRED_GE, 8766), (5, ca2.PRED_LT, 9131),
+             (3, ca2.PRED_GE, 5), (3, ca2.PRED_LE, 7), (1, ca2.PRED_LT, 2400)]
+    aggs = [(ca2.AGG_SUM_PROD_I64, 2, 3), (ca2.AGG_COUNT_STAR, -1),
+            (ca2.AGG_MIN_I64, 2), (ca2.AGG_MAX_I64, 2)]
+    with oracle.OracleTable(path) as t:
+        serial, _ = t.scan_agg(preds, aggs)
+        mt, cores = t.scan_agg_mt(preds, aggs)
+    assert cores >= 1
+    for a in range(len(aggs)):
+        assert mt[a].i128 == serial[a].i128
+        assert mt[a].count == serial[a].count
