@@ -682,10 +682,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
  * segments (the writer default); otherwise the two-kernel path runs.
  * ===================================================================== */
 
-#define FUSE_TILE_ROWS 1536     /* 48 x 256B segments per i64 column per tile:
-                                  * 4 cols = 192 decode lanes, 53.8 KB LDS ->
-                                  * 3 blocks/CU (vs 2 at 2048-row tiles) */
-#define FUSE_SEGS_PER_COL (FUSE_TILE_ROWS * 8 / 256)
+#define FUSE_TILE_ROWS 2048
 #define FUSE_STRIDE 280          /* 8B-aligned per-lane region (val() u64 reads) */
 
 struct FusedTile {
@@ -705,13 +702,13 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
     const uint32_t tid = threadIdx.x;
     const FusedTile t = tiles[blockIdx.x];
 
-    /* phase 1: decode my segment (lane = col*FUSE_SEGS_PER_COL + seg_in_tile) */
-    const uint32_t c = tid / FUSE_SEGS_PER_COL;
-    const uint32_t sseg = tid % FUSE_SEGS_PER_COL;
+    /* phase 1: decode my segment (lane = col*64 + seg_in_tile) */
+    const uint32_t c = tid >> 6;
+    const uint32_t sseg = tid & 63;
     const uint32_t nsegs = (t.row_count * 8 + 255) >> 8;
     if (c < params.n_proj && sseg < nsegs) {
         const SegDesc sd = segs[t.seg_base[c] + sseg];
-        lz4_lane_decode(data, sd, lds + ((size_t)c * FUSE_SEGS_PER_COL + sseg) * FUSE_STRIDE, err);
+        lz4_lane_decode(data, sd, lds + (size_t)tid * FUSE_STRIDE, err);
     }
     __syncthreads();
 
@@ -720,7 +717,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
     for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
 
     auto val = [&](uint32_t proj, uint32_t row) -> int64_t {
-        const uint32_t region = proj * FUSE_SEGS_PER_COL + (row >> 5);
+        const uint32_t region = proj * 64 + (row >> 5);
         return *(const int64_t *)(lds + (size_t)region * FUSE_STRIDE + ((row & 31) << 3));
     };
 
@@ -1492,8 +1489,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 ft.row_count = (k + 1 < ntile) ? FUSE_TILE_ROWS
                                                : rows - k * FUSE_TILE_ROWS;
                 for (uint32_t pj = 0; pj < n_proj; pj++)
-                    ft.seg_base[pj] = seg_start[(uint64_t)gi * n_proj + pj]
-                                      + FUSE_SEGS_PER_COL * k;
+                    ft.seg_base[pj] = seg_start[(uint64_t)gi * n_proj + pj] + 64 * k;
                 h_tiles.push_back(ft);
             }
         }
@@ -1714,10 +1710,9 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
     if (g->fusable && int_aggs && g->d_tiles) {
         /* fused decode+filter+aggregate: no scratch round trip */
         HIP_TRY(hipEventRecord(g->ev0, g->stream));
-        uint32_t fuse_lds = g->n_proj * FUSE_SEGS_PER_COL * FUSE_STRIDE;
         auto launchf = [&](auto *kern) {
             hipLaunchKernelGGL(kern, dim3(g->n_tiles), dim3(AGG_BLOCK),
-                               fuse_lds, g->stream,
+                               AGG_BLOCK * FUSE_STRIDE, g->stream,
                                g->d_data, g->d_segs, g->d_tiles, g->d_block,
                                g->d_error, p);
         };
