@@ -552,21 +552,114 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
     return rc;
 }
 
-#ifdef _OPENMP
-#include <omp.h>
-#endif
+#include <pthread.h>
+#include <unistd.h>
 
 /* All-core variant of the ungrouped scan (BASELINE.md protocol step 2:
- * "single-thread, and all-core (parallel over chunk groups)"). OpenMP over
- * the selected chunk-group list; per-thread accumulators merged with the
- * same strict-combine semantics. Integer results identical to the serial
- * oracle; f64 sums differ only in addition order (within the stated
- * tolerance). Returns the thread count used via *cores_out. */
+ * "single-thread, and all-core (parallel over chunk groups)"). Plain
+ * pthreads over the selected chunk-group list (NOT OpenMP: the product
+ * library already carries LLVM libomp, and loading gcc's libgomp into the
+ * same process is a known crash). Per-thread accumulators merged with the
+ * same strict-combine semantics; integer results identical to the serial
+ * oracle; f64 sums differ only in addition order (within tolerance). */
+
+typedef struct mt_ctx {
+    oracle_table *t;
+    const cstripe_pred *preds;
+    uint32_t n_preds;
+    const cstripe_agg_spec *aggs;
+    uint32_t n_aggs;
+    uint64_t mask;
+    const uint64_t *sel;
+    uint64_t nsel;
+    int nthreads;
+    int tid;
+    oacc *accs;            /* this thread's [n_aggs] */
+    int rc;
+} mt_ctx;
+
+static void *mt_worker(void *arg)
+{
+    mt_ctx *cx = arg;
+    oracle_table *t = cx->t;
+    uint32_t ncols = t->head.column_count;
+    ochunkcol *cc = calloc(ncols, sizeof(ochunkcol));
+    uint32_t chunk_cap = t->head.chunk_row_limit;
+    for (uint32_t c = 0; c < ncols; c++) {
+        if (!(cx->mask & (1ull << c))) continue;
+        cc[c].type = t->cols[c].type;
+        cc[c].width = (uint8_t)csf_type_width(cc[c].type);
+        cc[c].values = malloc((size_t)chunk_cap * cc[c].width);
+        cc[c].exists = malloc(chunk_cap);
+    }
+    for (uint64_t i = (uint64_t)cx->tid; i < cx->nsel; i += (uint64_t)cx->nthreads) {
+        uint32_t s = (uint32_t)cx->sel[i * 2], k = (uint32_t)cx->sel[i * 2 + 1];
+        const ostripe *st = &t->stripes[s];
+        uint32_t rows = st->group_rows[k];
+        int bad = 0;
+        for (uint32_t c = 0; c < ncols && !bad; c++) {
+            if (!(cx->mask & (1ull << c))) continue;
+            if (oracle_read_chunk(t, s, k, c, cc[c].values, cc[c].exists) != 0) bad = 1;
+        }
+        if (bad) { cx->rc = -2; continue; }
+        for (uint32_t row = 0; row < rows; row++) {
+            int pass = 1;
+            for (uint32_t p = 0; p < cx->n_preds && pass; p++) {
+                const cstripe_pred *pr = &cx->preds[p];
+                int64_t iv; double fv;
+                uint8_t ty = t->cols[pr->column].type;
+                if (!o_get(&cc[pr->column], row, &iv, &fv)) pass = 0;
+                else pass = o_pred_eval(pr, ty == CSTRIPE_F32 || ty == CSTRIPE_F64, iv, fv);
+            }
+            if (!pass) continue;
+            for (uint32_t a = 0; a < cx->n_aggs; a++) {
+                const cstripe_agg_spec *g = &cx->aggs[a];
+                oacc *ac = &cx->accs[a];
+                int64_t iv, ib, ic; double fv, fb, fc;
+                switch (g->kind) {
+                    case CSTRIPE_AGG_COUNT_STAR: ac->cnt++; break;
+                    case CSTRIPE_AGG_COUNT_COL:
+                        if (o_get(&cc[g->col_a], row, &iv, &fv)) ac->cnt++;
+                        break;
+                    case CSTRIPE_AGG_SUM_I64:
+                        if (o_get(&cc[g->col_a], row, &iv, &fv)) { ac->i128 += iv; ac->cnt++; }
+                        break;
+                    case CSTRIPE_AGG_SUM_F64:
+                        if (o_get(&cc[g->col_a], row, &iv, &fv)) { ac->f += fv; ac->cnt++; }
+                        break;
+                    case CSTRIPE_AGG_MIN_I64:
+                        if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (iv < ac->minmax_i) ac->minmax_i = iv; ac->cnt++; }
+                        break;
+                    case CSTRIPE_AGG_MAX_I64:
+                        if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (iv > ac->minmax_i) ac->minmax_i = iv; ac->cnt++; }
+                        break;
+                    case CSTRIPE_AGG_SUM_PROD_I64:
+                        if (o_get(&cc[g->col_a], row, &iv, &fv) &&
+                            o_get(&cc[g->col_b], row, &ib, &fb)) { ac->i128 += (__int128)iv * ib; ac->cnt++; }
+                        break;
+                    case CSTRIPE_AGG_SUM_DISC_I64:
+                        if (o_get(&cc[g->col_a], row, &iv, &fv) &&
+                            o_get(&cc[g->col_b], row, &ib, &fb)) { ac->i128 += (__int128)iv * (g->one - ib); ac->cnt++; }
+                        break;
+                    case CSTRIPE_AGG_SUM_DISC_TAX_I64:
+                        if (o_get(&cc[g->col_a], row, &iv, &fv) &&
+                            o_get(&cc[g->col_b], row, &ib, &fb) &&
+                            o_get(&cc[g->col_c], row, &ic, &fc)) { ac->i128 += (__int128)iv * (g->one - ib) * (g->one + ic); ac->cnt++; }
+                        break;
+                    default: cx->rc = -4; break;
+                }
+            }
+        }
+    }
+    for (uint32_t c = 0; c < ncols; c++) { free(cc[c].values); free(cc[c].exists); }
+    free(cc);
+    return NULL;
+}
+
 int oracle_scan_agg_mt(oracle_table *t, const cstripe_pred *preds, uint32_t n_preds,
                        const cstripe_agg_spec *aggs, uint32_t n_aggs,
                        cstripe_partial *out, int *cores_out)
 {
-    uint32_t ncols = t->head.column_count;
     uint64_t mask = 0;
     for (uint32_t i = 0; i < n_preds; i++) mask |= 1ull << preds[i].column;
     for (uint32_t i = 0; i < n_aggs; i++) {
@@ -575,7 +668,6 @@ int oracle_scan_agg_mt(oracle_table *t, const cstripe_pred *preds, uint32_t n_pr
         if (aggs[i].col_c >= 0) mask |= 1ull << aggs[i].col_c;
     }
 
-    /* selected chunk list (pruning identical to the serial path) */
     uint64_t nsel = 0, cap = 1024;
     uint64_t *sel = malloc(cap * 16);
     for (uint32_t s = 0; s < t->head.stripe_count; s++) {
@@ -596,112 +688,35 @@ int oracle_scan_agg_mt(oracle_table *t, const cstripe_pred *preds, uint32_t n_pr
         }
     }
 
-    int nthreads = 1;
-#ifdef _OPENMP
-    nthreads = omp_get_max_threads();
-#endif
+    long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+    int nthreads = ncpu > 1 ? (int)ncpu : 1;
+    if (nthreads > 64) nthreads = 64;
     oacc *tacc = calloc((size_t)nthreads * n_aggs, sizeof(oacc));
-    int64_t *tcnt_init = calloc(nthreads, 1);
-    (void)tcnt_init;
-    for (int th = 0; th < nthreads; th++)
+    mt_ctx *ctx = calloc(nthreads, sizeof(mt_ctx));
+    pthread_t *th = calloc(nthreads, sizeof(pthread_t));
+    for (int i = 0; i < nthreads; i++) {
         for (uint32_t a = 0; a < n_aggs; a++)
-            oacc_init(&tacc[(size_t)th * n_aggs + a], aggs[a].kind);
-    int rc = 0;
-
-#ifdef _OPENMP
-    #pragma omp parallel
-#endif
-    {
-        int th = 0;
-#ifdef _OPENMP
-        th = omp_get_thread_num();
-#endif
-        oacc *accs = &tacc[(size_t)th * n_aggs];
-        ochunkcol *cc = calloc(ncols, sizeof(ochunkcol));
-        uint32_t chunk_cap = t->head.chunk_row_limit;
-        for (uint32_t c = 0; c < ncols; c++) {
-            if (!(mask & (1ull << c))) continue;
-            cc[c].type = t->cols[c].type;
-            cc[c].width = (uint8_t)csf_type_width(cc[c].type);
-            cc[c].values = malloc((size_t)chunk_cap * cc[c].width);
-            cc[c].exists = malloc(chunk_cap);
-        }
-#ifdef _OPENMP
-        #pragma omp for schedule(dynamic, 8)
-#endif
-        for (int64_t i = 0; i < (int64_t)nsel; i++) {
-            uint32_t s = (uint32_t)sel[i * 2], k = (uint32_t)sel[i * 2 + 1];
-            const ostripe *st = &t->stripes[s];
-            uint32_t rows = st->group_rows[k];
-            int bad = 0;
-            for (uint32_t c = 0; c < ncols && !bad; c++) {
-                if (!(mask & (1ull << c))) continue;
-                if (oracle_read_chunk(t, s, k, c, cc[c].values, cc[c].exists) != 0) bad = 1;
-            }
-            if (bad) { rc = -2; continue; }
-            for (uint32_t row = 0; row < rows; row++) {
-                int pass = 1;
-                for (uint32_t p = 0; p < n_preds && pass; p++) {
-                    const cstripe_pred *pr = &preds[p];
-                    int64_t iv; double fv;
-                    uint8_t ty = t->cols[pr->column].type;
-                    if (!o_get(&cc[pr->column], row, &iv, &fv)) pass = 0;
-                    else pass = o_pred_eval(pr, ty == CSTRIPE_F32 || ty == CSTRIPE_F64, iv, fv);
-                }
-                if (!pass) continue;
-                for (uint32_t a = 0; a < n_aggs; a++) {
-                    const cstripe_agg_spec *g = &aggs[a];
-                    oacc *ac = &accs[a];
-                    int64_t iv, ib, ic; double fv, fb, fc;
-                    switch (g->kind) {
-                        case CSTRIPE_AGG_COUNT_STAR: ac->cnt++; break;
-                        case CSTRIPE_AGG_COUNT_COL:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv)) ac->cnt++;
-                            break;
-                        case CSTRIPE_AGG_SUM_I64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv)) { ac->i128 += iv; ac->cnt++; }
-                            break;
-                        case CSTRIPE_AGG_SUM_F64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv)) { ac->f += fv; ac->cnt++; }
-                            break;
-                        case CSTRIPE_AGG_MIN_I64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (iv < ac->minmax_i) ac->minmax_i = iv; ac->cnt++; }
-                            break;
-                        case CSTRIPE_AGG_MAX_I64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (iv > ac->minmax_i) ac->minmax_i = iv; ac->cnt++; }
-                            break;
-                        case CSTRIPE_AGG_SUM_PROD_I64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv) &&
-                                o_get(&cc[g->col_b], row, &ib, &fb)) { ac->i128 += (__int128)iv * ib; ac->cnt++; }
-                            break;
-                        case CSTRIPE_AGG_SUM_DISC_I64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv) &&
-                                o_get(&cc[g->col_b], row, &ib, &fb)) { ac->i128 += (__int128)iv * (g->one - ib); ac->cnt++; }
-                            break;
-                        case CSTRIPE_AGG_SUM_DISC_TAX_I64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv) &&
-                                o_get(&cc[g->col_b], row, &ib, &fb) &&
-                                o_get(&cc[g->col_c], row, &ic, &fc)) { ac->i128 += (__int128)iv * (g->one - ib) * (g->one + ic); ac->cnt++; }
-                            break;
-                        default: rc = -4; break;
-                    }
-                }
-            }
-        }
-        for (uint32_t c = 0; c < ncols; c++) { free(cc[c].values); free(cc[c].exists); }
-        free(cc);
+            oacc_init(&tacc[(size_t)i * n_aggs + a], aggs[a].kind);
+        ctx[i] = (mt_ctx){t, preds, n_preds, aggs, n_aggs, mask, sel, nsel,
+                          nthreads, i, &tacc[(size_t)i * n_aggs], 0};
+        if (i > 0) pthread_create(&th[i], NULL, mt_worker, &ctx[i]);
+    }
+    mt_worker(&ctx[0]);
+    int rc = ctx[0].rc;
+    for (int i = 1; i < nthreads; i++) {
+        pthread_join(th[i], NULL);
+        if (ctx[i].rc) rc = ctx[i].rc;
     }
 
-    /* merge thread accumulators (strict combine) */
     for (uint32_t a = 0; a < n_aggs; a++) {
         oacc m;
         oacc_init(&m, aggs[a].kind);
-        for (int th = 0; th < nthreads; th++) {
-            const oacc *x = &tacc[(size_t)th * n_aggs + a];
+        for (int i = 0; i < nthreads; i++) {
+            const oacc *x = &tacc[(size_t)i * n_aggs + a];
             if (x->cnt == 0) continue;
             m.i128 += x->i128;
             m.f += x->f;
-            if (x->minmax_i < m.minmax_i && (aggs[a].kind == CSTRIPE_AGG_MIN_I64)) m.minmax_i = x->minmax_i;
+            if (aggs[a].kind == CSTRIPE_AGG_MIN_I64 && x->minmax_i < m.minmax_i) m.minmax_i = x->minmax_i;
             if (aggs[a].kind == CSTRIPE_AGG_MAX_I64 && x->minmax_i > m.minmax_i) m.minmax_i = x->minmax_i;
             if (aggs[a].kind == CSTRIPE_AGG_MIN_F64 && x->minmax_f < m.minmax_f) m.minmax_f = x->minmax_f;
             if (aggs[a].kind == CSTRIPE_AGG_MAX_F64 && x->minmax_f > m.minmax_f) m.minmax_f = x->minmax_f;
@@ -710,7 +725,8 @@ int oracle_scan_agg_mt(oracle_table *t, const cstripe_pred *preds, uint32_t n_pr
         oacc_to_partial(&m, aggs[a].kind, &out[a]);
     }
     free(tacc);
-    free(tcnt_init);
+    free(ctx);
+    free(th);
     free(sel);
     if (cores_out) *cores_out = nthreads;
     return rc;
