@@ -185,7 +185,10 @@ def main():
     alg_bytes_per_row = ALG_BYTES[args.query]
     avg_decode = sum(decode_ms) / len(decode_ms)
     avg_agg = sum(agg_ms) / len(agg_ms)
-    dominant = "lz4_decode_kernel" if avg_decode >= avg_agg else "filter_agg_kernel"
+    if scan.last_fused:
+        dominant = "fused_agg_kernel"
+    else:
+        dominant = "lz4_decode_kernel" if avg_decode >= avg_agg else "filter_agg_kernel"
     dominant_ms = max(avg_decode, avg_agg)
     alg_bytes = args.rows * alg_bytes_per_row          # per launch (this rank)
     achieved_gbps = alg_bytes / (dominant_ms / 1e3) / 1e9 if dominant_ms > 0 else None

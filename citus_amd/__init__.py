@@ -115,6 +115,7 @@ _scan_agg_grouped = _sig("cstripe_scan_agg_grouped", C.c_int,
                           C.POINTER(Partial)])
 _next_batch = _sig("cstripe_scan_next_batch", C.c_int, [C.c_void_p, C.POINTER(Batch)])
 _rewind = _sig("cstripe_scan_rewind", C.c_int, [C.c_void_p])
+_last_fused = _sig("cstripe_scan_last_fused", C.c_int, [C.c_void_p])
 _last_kernel_ms = _sig("cstripe_scan_last_kernel_ms", C.c_double, [C.c_void_p])
 _last_decode_ms = _sig("cstripe_scan_last_decode_kernel_ms", C.c_double, [C.c_void_p])
 _last_agg_ms = _sig("cstripe_scan_last_agg_kernel_ms", C.c_double, [C.c_void_p])
@@ -305,6 +306,10 @@ class Scan:
     @property
     def staged_bytes(self):
         return _staged_bytes(self._h)
+
+    @property
+    def last_fused(self):
+        return bool(_last_fused(self._h))
 
     @property
     def last_kernel_ms(self):

@@ -1709,6 +1709,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
     HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
     if (g->fusable && int_aggs && g->d_tiles) {
         /* fused decode+filter+aggregate: no scratch round trip */
+        s->last_fused = 1;
         HIP_TRY(hipEventRecord(g->ev0, g->stream));
         auto launchf = [&](auto *kern) {
             hipLaunchKernelGGL(kern, dim3(g->n_tiles), dim3(AGG_BLOCK),
@@ -1763,6 +1764,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         return CSTRIPE_OK;
     }
 
+    s->last_fused = 0;
     HIP_TRY(hipEventRecord(g->ev0, g->stream));
     launch_decode(g);
     HIP_TRY(hipEventRecord(g->ev1, g->stream));

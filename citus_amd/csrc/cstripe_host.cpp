@@ -721,6 +721,7 @@ extern "C" int cstripe_scan_rewind(cstripe_scan *s)
     return CSTRIPE_OK;
 }
 
+extern "C" int cstripe_scan_last_fused(const cstripe_scan *s) { return s ? s->last_fused : 0; }
 extern "C" double cstripe_scan_last_kernel_ms(const cstripe_scan *s) { return s ? s->last_kernel_ms : 0; }
 extern "C" double cstripe_scan_last_decode_kernel_ms(const cstripe_scan *s) { return s ? s->last_decode_ms : 0; }
 extern "C" double cstripe_scan_last_agg_kernel_ms(const cstripe_scan *s) { return s ? s->last_agg_ms : 0; }

@@ -50,6 +50,7 @@ struct cstripe_scan {
     int64_t chunk_groups_filtered = 0;
     cs_gpu_state *gpu = nullptr;
     size_t batch_pos = 0;             /* next_batch cursor into sel */
+    int last_fused = 0;
     double last_kernel_ms = 0.0;
     double last_decode_ms = 0.0;
     double last_agg_ms = 0.0;

@@ -273,6 +273,10 @@ int cstripe_scan_agg_grouped(cstripe_scan *s, const cstripe_agg_spec *aggs,
 int cstripe_scan_next_batch(cstripe_scan *s, cstripe_batch *batch);
 int cstripe_scan_rewind(cstripe_scan *s);
 
+/* 1 when the last cstripe_scan_agg ran the fused decode+filter+aggregate
+ * kernel (no scratch round trip), 0 for the two-kernel path */
+int cstripe_scan_last_fused(const cstripe_scan *s);
+
 /* per-call timing of the last cstripe_scan_agg: kernel time (hipEvents, on the
  * scan's stream) and wall time inside the call, milliseconds. */
 double cstripe_scan_last_kernel_ms(const cstripe_scan *s);
