@@ -82,6 +82,7 @@ struct AggParams {
 };
 
 struct FusedTile;
+struct FusedTileG;
 
 /* per-block / final accumulator cell (32 B) */
 struct AccCell {
@@ -117,8 +118,14 @@ struct cs_gpu_state {
     uint32_t max_seg_dlen = 0;
     bool segs_16aligned = true;
     bool fusable = true;             /* all proj cols dense i64, uniform 256B lz4 segs */
+    bool fusable_mixed = true;       /* widths in {1,8}; enables fused grouped */
     struct FusedTile *d_tiles = nullptr;
     uint32_t n_tiles = 0;
+    struct FusedTileG *d_tiles2 = nullptr;
+    uint32_t n_tiles2 = 0;
+    uint16_t flane_base[8] = {0};
+    uint8_t fwidth[8] = {0};
+    uint16_t flanes_total = 0;
     uint32_t n_groups = 0;
     uint32_t n_proj = 0;
     uint32_t max_blocks = 0;
@@ -1096,6 +1103,163 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     }
 }
 
+/* =====================================================================
+ * FUSED GROUPED kernel (TPC-H Q1 shape): same no-scratch idea as
+ * fused_agg_kernel, generalized to mixed i64/i8 columns — tile = 1536 rows
+ * (1536*width is a multiple of 256 for both widths, so tile boundaries land
+ * on segment boundaries), lane ranges per column by prefix. Groups
+ * accumulate into ONE per-block 16-slot LDS table with the carry-exact
+ * atomics; blocks grid-stride over tiles and flush their table once.
+ * ===================================================================== */
+
+#define FUSEG_TILE_ROWS 1536
+#define FUSEG_SLOTS 16
+#define FUSEG_GRID 2048
+
+struct FusedTileG {
+    uint32_t row_count;
+    uint32_t seg_base[8];
+};
+
+struct FusedGParams {
+    AggParams base;
+    uint32_t n_group_cols;
+    uint32_t gproj[CSTRIPE_MAX_GROUP_COLS];
+    uint32_t n_tiles;
+    uint16_t lane_base[8];       /* per proj col: first LDS region */
+    uint8_t  width[8];           /* per proj col: value width (1 or 8) */
+    uint16_t lanes_total;
+};
+
+template <int NAGGS>
+__global__ __launch_bounds__(AGG_BLOCK) void fused_grouped_kernel(
+    const uint8_t *__restrict__ data, const SegDesc *__restrict__ segs,
+    const FusedTileG *__restrict__ tiles, uint16_t *__restrict__ keys_out,
+    AccCell *__restrict__ cells_out, int *__restrict__ err,
+    const FusedGParams gp)
+{
+    const AggParams &params = gp.base;
+    const uint32_t n_aggs_ct = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
+    const uint32_t tid = threadIdx.x;
+    extern __shared__ uint8_t lds[];
+    int *skeys = (int *)(lds + (size_t)gp.lanes_total * FUSE_STRIDE);
+    ThreadAcc *scells = (ThreadAcc *)(skeys + FUSEG_SLOTS);
+
+    for (uint32_t i = tid; i < FUSEG_SLOTS; i += AGG_BLOCK) {
+        skeys[i] = -1;
+        for (uint32_t a = 0; a < n_aggs_ct; a++)
+            acc_init(scells[i * n_aggs_ct + a], params.aggs[a].kind);
+    }
+    __syncthreads();
+
+    /* my (col, seg) assignment: first col whose lane range contains tid */
+    uint32_t mycol = 0xFFFFFFFF, myseg = 0;
+    for (uint32_t c2 = 0; c2 < params.n_proj; c2++) {
+        uint32_t base = gp.lane_base[c2];
+        uint32_t span = (uint32_t)FUSEG_TILE_ROWS * gp.width[c2] / 256;
+        if (tid >= base && tid < base + span) { mycol = c2; myseg = tid - base; }
+    }
+
+    auto val = [&](uint32_t proj, uint32_t row) -> int64_t {
+        const uint32_t byteoff = row * gp.width[proj];
+        const uint8_t *p = lds + ((size_t)gp.lane_base[proj] + (byteoff >> 8)) * FUSE_STRIDE
+                           + (byteoff & 255);
+        return gp.width[proj] == 8 ? *(const int64_t *)p : (int64_t)*(const int8_t *)p;
+    };
+
+    /* per-lane register key->slot cache */
+    uint32_t ck0 = ~0u, ck1 = ~0u;
+    uint32_t cs0 = 0, cs1 = 0;
+
+    for (uint32_t work = blockIdx.x; work < gp.n_tiles; work += gridDim.x) {
+        const FusedTileG t = tiles[work];
+
+        /* phase 1: decode */
+        if (mycol != 0xFFFFFFFF) {
+            const uint32_t nsegs = (t.row_count * gp.width[mycol] + 255) >> 8;
+            if (myseg < nsegs) {
+                const SegDesc sd = segs[t.seg_base[mycol] + myseg];
+                lz4_lane_decode(data, sd, lds + (size_t)tid * FUSE_STRIDE, err);
+            }
+        }
+        __syncthreads();
+
+        /* phase 2: filter + group + accumulate */
+        for (uint32_t row = tid; row < t.row_count; row += AGG_BLOCK) {
+            bool pass = true;
+            int last_proj = -1;
+            int64_t liv = 0;
+            for (uint32_t p = 0; p < params.n_preds; p++) {
+                const PredD &pr = params.preds[p];
+                if ((int)pr.proj != last_proj) { liv = val(pr.proj, row); last_proj = (int)pr.proj; }
+                pass = pass & pred_eval(pr, liv, 0.0);
+            }
+            if (!pass) continue;
+            uint32_t key = 0;
+            for (uint32_t gc = 0; gc < gp.n_group_cols; gc++)
+                key |= ((uint32_t)val(gp.gproj[gc], row) & 0xFF) << (8 * gc);
+            uint32_t slot = 0xFFFFFFFF;
+            if (key == ck0) slot = cs0;
+            else if (key == ck1) slot = cs1;
+            else {
+                for (uint32_t s2 = 0; s2 < FUSEG_SLOTS; s2++) {
+                    int old = atomicCAS(&skeys[s2], -1, (int)key);
+                    if (old == -1 || old == (int)key) { slot = s2; break; }
+                }
+                if (slot == 0xFFFFFFFF) { atomicOr(err, 8); continue; }
+                ck1 = ck0; cs1 = cs0;
+                ck0 = key; cs0 = slot;
+            }
+            for (uint32_t a = 0; a < n_aggs_ct; a++) {
+                const AggD &g = params.aggs[a];
+                PrepAcc p2;
+                p2.valid = true; p2.f = 0.0; p2.hi = 0;
+                switch (g.kind) {
+                    case CSTRIPE_AGG_COUNT_STAR:
+                    case CSTRIPE_AGG_COUNT_COL: p2.lo = 1; break;
+                    case CSTRIPE_AGG_SUM_I64:
+                    case CSTRIPE_AGG_MIN_I64:
+                    case CSTRIPE_AGG_MAX_I64: {
+                        int64_t v = val(g.proj_a, row);
+                        p2.lo = v; p2.hi = v < 0 ? -1 : 0; break;
+                    }
+                    case CSTRIPE_AGG_SUM_PROD_I64: {
+                        __int128 x = (__int128)val(g.proj_a, row) * val(g.proj_b, row);
+                        p2.lo = (int64_t)(uint64_t)x; p2.hi = (int64_t)(x >> 64); break;
+                    }
+                    case CSTRIPE_AGG_SUM_DISC_I64: {
+                        __int128 x = (__int128)val(g.proj_a, row) * (g.one - val(g.proj_b, row));
+                        p2.lo = (int64_t)(uint64_t)x; p2.hi = (int64_t)(x >> 64); break;
+                    }
+                    case CSTRIPE_AGG_SUM_DISC_TAX_I64: {
+                        __int128 x = (__int128)val(g.proj_a, row) * (g.one - val(g.proj_b, row))
+                                     * (g.one + val(g.proj_c, row));
+                        p2.lo = (int64_t)(uint64_t)x; p2.hi = (int64_t)(x >> 64); break;
+                    }
+                    default: p2.valid = false; break;
+                }
+                acc_apply_atomic(&scells[slot * n_aggs_ct + a], g.kind, p2);
+            }
+        }
+        __syncthreads();     /* LDS decode regions reused next tile */
+    }
+
+    /* flush the block table */
+    __syncthreads();
+    if (tid == 0) {
+        uint16_t *bk = keys_out + (size_t)blockIdx.x * FUSEG_SLOTS;
+        AccCell *bc = cells_out + (size_t)blockIdx.x * FUSEG_SLOTS * n_aggs_ct;
+        for (uint32_t s2 = 0; s2 < FUSEG_SLOTS; s2++) {
+            bk[s2] = skeys[s2] < 0 ? 0xFFFF : (uint16_t)skeys[s2];
+            for (uint32_t a = 0; a < n_aggs_ct; a++) {
+                const ThreadAcc &x = scells[s2 * n_aggs_ct + a];
+                AccCell cell{x.lo, x.hi, x.f, x.cnt};
+                bc[(size_t)s2 * n_aggs_ct + a] = cell;
+            }
+        }
+    }
+}
+
 /* single-block merge of per-block group tables -> final <=64 groups.
  * Parallel: 256 threads stride the (block, slot) entries; keys claim final
  * slots via LDS CAS (linear probe); cells merge with the same atomic
@@ -1257,6 +1421,7 @@ void csgpu_release(cstripe_scan *s)
     if (g->d_gfcells) HIP_DROP(hipFree(g->d_gfcells));
     if (g->d_gn) HIP_DROP(hipFree(g->d_gn));
     if (g->d_tiles) HIP_DROP(hipFree(g->d_tiles));
+    if (g->d_tiles2) HIP_DROP(hipFree(g->d_tiles2));
     if (g->d_error) HIP_DROP(hipFree(g->d_error));
     if (g->ev0) HIP_DROP(hipEventDestroy(g->ev0));
     if (g->ev1) HIP_DROP(hipEventDestroy(g->ev1));
@@ -1409,6 +1574,9 @@ int csgpu_stage(cstripe_scan *s, int device_id)
             if (!(cl.flags & 2) || cl.type != CSTRIPE_I64 ||
                 nd.n.comp_type != CSTRIPE_COMP_LZ4)
                 g->fusable = false;
+            if (!(cl.flags & 2) || (cl.width != 8 && cl.width != 1) ||
+                nd.n.comp_type != CSTRIPE_COMP_LZ4)
+                g->fusable_mixed = false;
             if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
                 dpos = align_up(dpos, 16);
                 memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
@@ -1424,8 +1592,10 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                      * still fit a FUSE_STRIDE LDS region */
                     if (sg.decomp_off != segi * 256u ||
                         (segi + 1 < seglist.size() ? sg.decomp_len != 256
-                                                   : sg.decomp_len > FUSE_STRIDE - 8))
+                                                   : sg.decomp_len > FUSE_STRIDE - 8)) {
                         g->fusable = false;
+                        g->fusable_mixed = false;
+                    }
                     segi++;
                 }
                 for (const csf_seg &sg : st.nodes[c][sc.chunk].segs) {
@@ -1498,6 +1668,42 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         HIP_TRY(hipMemcpyAsync(g->d_tiles, h_tiles.data(),
                                h_tiles.size() * sizeof(FusedTile),
                                hipMemcpyHostToDevice, g->stream));
+    }
+    /* mixed-width (fused grouped) tiles: lane ranges by prefix */
+    if (g->fusable_mixed && g->n_groups > 0 && n_proj <= 8) {
+        uint32_t lb = 0;
+        for (uint32_t pj = 0; pj < n_proj; pj++) {
+            /* width from any group's colloc (uniform per column) */
+            uint8_t w = h_colloc[pj].width;   /* uniform per column */
+            g->fwidth[pj] = w;
+            g->flane_base[pj] = (uint16_t)lb;
+            lb += (uint32_t)FUSEG_TILE_ROWS * w / 256;
+        }
+        if (lb <= AGG_BLOCK) {
+            g->flanes_total = (uint16_t)lb;
+            std::vector<FusedTileG> h_tiles2;
+            for (uint32_t gi = 0; gi < g->n_groups; gi++) {
+                const cs_selchunk &sc = s->sel[gi];
+                uint32_t rows = r->stripes[sc.stripe].group_rows[sc.chunk];
+                uint32_t ntile = (rows + FUSEG_TILE_ROWS - 1) / FUSEG_TILE_ROWS;
+                for (uint32_t k = 0; k < ntile; k++) {
+                    FusedTileG ft{};
+                    ft.row_count = (k + 1 < ntile) ? FUSEG_TILE_ROWS
+                                                   : rows - k * FUSEG_TILE_ROWS;
+                    for (uint32_t pj = 0; pj < n_proj; pj++)
+                        ft.seg_base[pj] = seg_start[(uint64_t)gi * n_proj + pj]
+                                          + k * ((uint32_t)FUSEG_TILE_ROWS * g->fwidth[pj] / 256);
+                    h_tiles2.push_back(ft);
+                }
+            }
+            g->n_tiles2 = (uint32_t)h_tiles2.size();
+            HIP_TRY(hipMalloc(&g->d_tiles2, h_tiles2.size() * sizeof(FusedTileG)));
+            HIP_TRY(hipMemcpyAsync(g->d_tiles2, h_tiles2.data(),
+                                   h_tiles2.size() * sizeof(FusedTileG),
+                                   hipMemcpyHostToDevice, g->stream));
+        } else {
+            g->fusable_mixed = false;
+        }
     }
     HIP_TRY(hipStreamSynchronize(g->stream));
     g->colloc_host = h_colloc;
@@ -1613,6 +1819,91 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             HIP_TRY(hipMalloc(&g->d_gfcells, (uint64_t)CSTRIPE_MAX_GROUPS * MAX_AGGS * sizeof(AccCell)));
             HIP_TRY(hipMalloc(&g->d_gn, 4));
         }
+        bool int_aggs_g = true;
+        for (uint32_t a = 0; a < n_aggs; a++)
+            if (aggs[a].kind == CSTRIPE_AGG_SUM_F64 || aggs[a].kind == CSTRIPE_AGG_MIN_F64 ||
+                aggs[a].kind == CSTRIPE_AGG_MAX_F64)
+                int_aggs_g = false;
+        if (g->fusable_mixed && int_aggs_g && g->d_tiles2) {
+            /* fused grouped: decode tile to LDS + atomic group table; falls
+             * back below when >16 distinct groups (device flag 8) */
+            FusedGParams fgp{};
+            fgp.base = p;
+            fgp.n_group_cols = n_group_cols;
+            for (uint32_t i = 0; i < n_group_cols; i++) fgp.gproj[i] = gp.gproj[i];
+            fgp.n_tiles = g->n_tiles2;
+            for (uint32_t i = 0; i < 8; i++) { fgp.lane_base[i] = g->flane_base[i]; fgp.width[i] = g->fwidth[i]; }
+            fgp.lanes_total = g->flanes_total;
+            uint32_t fgrid = g->n_tiles2 < FUSEG_GRID ? g->n_tiles2 : FUSEG_GRID;
+            uint32_t flds = (uint32_t)fgp.lanes_total * FUSE_STRIDE + FUSEG_SLOTS * 4
+                            + FUSEG_SLOTS * n_aggs * (uint32_t)sizeof(ThreadAcc);
+            HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
+            HIP_TRY(hipEventRecord(g->ev0, g->stream));
+            auto launchfg = [&](auto *kern) {
+                hipLaunchKernelGGL(kern, dim3(fgrid), dim3(AGG_BLOCK), flds, g->stream,
+                                   g->d_data, g->d_segs, g->d_tiles2, g->d_gkeys,
+                                   g->d_gcells, g->d_error, fgp);
+            };
+            if (n_aggs == 5) launchfg(fused_grouped_kernel<5>);
+            else if (n_aggs == 1) launchfg(fused_grouped_kernel<1>);
+            else if (n_aggs == 2) launchfg(fused_grouped_kernel<2>);
+            else launchfg(fused_grouped_kernel<-1>);
+            HIP_TRY(hipEventRecord(g->ev1, g->stream));
+            hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
+                               g->d_gkeys, g->d_gcells, fgrid, FUSEG_SLOTS,
+                               g->d_gfkeys, g->d_gfcells, g->d_gn, g->d_error, gp);
+            HIP_TRY(hipEventRecord(g->ev2, g->stream));
+            int h_err = 0;
+            HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
+            HIP_TRY(hipStreamSynchronize(g->stream));
+            if (h_err & 4) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+            if (h_err == 0) {
+                s->last_fused = 1;
+                uint16_t h_keys[CSTRIPE_MAX_GROUPS];
+                std::vector<AccCell> h_cells((size_t)CSTRIPE_MAX_GROUPS * n_aggs);
+                uint32_t h_n = 0;
+                HIP_TRY(hipMemcpyAsync(h_keys, g->d_gfkeys, sizeof(h_keys), hipMemcpyDeviceToHost, g->stream));
+                HIP_TRY(hipMemcpyAsync(h_cells.data(), g->d_gfcells, h_cells.size() * sizeof(AccCell), hipMemcpyDeviceToHost, g->stream));
+                HIP_TRY(hipMemcpyAsync(&h_n, g->d_gn, 4, hipMemcpyDeviceToHost, g->stream));
+                HIP_TRY(hipStreamSynchronize(g->stream));
+                float ms_f = 0, ms_r = 0;
+                (void)hipEventElapsedTime(&ms_f, g->ev0, g->ev1);
+                (void)hipEventElapsedTime(&ms_r, g->ev1, g->ev2);
+                s->last_decode_ms = ms_f;
+                s->last_agg_ms = ms_r;
+                s->last_kernel_ms = ms_f + ms_r;
+                std::vector<uint32_t> order(h_n);
+                for (uint32_t i = 0; i < h_n; i++) order[i] = i;
+                std::sort(order.begin(), order.end(),
+                          [&](uint32_t a, uint32_t b) { return h_keys[a] < h_keys[b]; });
+                gr->n_groups = h_n;
+                for (uint32_t oi = 0; oi < h_n; oi++) {
+                    uint32_t i = order[oi];
+                    gr->keys[oi] = h_keys[i];
+                    for (uint32_t a = 0; a < n_aggs; a++) {
+                        const AccCell &cc2 = h_cells[(size_t)i * n_aggs + a];
+                        cstripe_partial o{};
+                        o.count = cc2.cnt;
+                        o.is_null = (cc2.cnt == 0) ? 1 : 0;
+                        switch (aggs[a].kind) {
+                            case CSTRIPE_AGG_COUNT_STAR:
+                            case CSTRIPE_AGG_COUNT_COL:
+                                o.i128_lo = cc2.cnt; o.is_null = 0; break;
+                            case CSTRIPE_AGG_MIN_I64:
+                            case CSTRIPE_AGG_MAX_I64:
+                                if (!o.is_null) { o.i128_lo = cc2.lo; o.i128_hi = cc2.lo < 0 ? -1 : 0; }
+                                break;
+                            default:
+                                if (!o.is_null) { o.i128_lo = cc2.lo; o.i128_hi = cc2.hi; }
+                                break;
+                        }
+                        out[(size_t)oi * n_aggs + a] = o;
+                    }
+                }
+                return CSTRIPE_OK;
+            }
+        }
+
         uint32_t lds = n_waves * GRP_SLOTS * 2 + n_waves * GRP_SLOTS * n_aggs * (uint32_t)sizeof(ThreadAcc);
         HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
         HIP_TRY(hipEventRecord(g->ev0, g->stream));
