@@ -407,3 +407,55 @@ def test_shard_split_combine_property(tmp_path):
     for i in range(len(aggs)):
         assert combined[i].i128 == ocombined[i].i128
         assert combined[i].count == ocombined[i].count
+
+
+def test_fuzz_differential():
+    """Randomized differential fuzz: random schemas / sizes / null patterns /
+    codecs / predicates / aggregates — GPU vs oracle, 24 cases."""
+    TYPES = [(ca.I8, np.int8), (ca.I16, np.int16), (ca.I32, np.int32),
+             (ca.I64, np.int64), (ca.F32, np.float32), (ca.F64, np.float64)]
+    INT_AGGS = [ca.AGG_SUM_I64, ca.AGG_MIN_I64, ca.AGG_MAX_I64, ca.AGG_COUNT_COL]
+    FLT_AGGS = [ca.AGG_SUM_F64, ca.AGG_MIN_F64, ca.AGG_MAX_F64, ca.AGG_COUNT_COL]
+    import tempfile
+    for seed in range(24):
+        rng = np.random.default_rng(1000 + seed)
+        ncols = int(rng.integers(1, 6))
+        kinds = [TYPES[int(rng.integers(0, 6))] for _ in range(ncols)]
+        n = int(rng.integers(1, 60001))
+        chunk = int(rng.integers(1, 11)) * 1000
+        stripe = [10000, 50000, 150000][int(rng.integers(0, 3))]
+        comp = [ca.COMP_NONE, ca.COMP_LZ4, ca.COMP_LZ4, ca.COMP_ZSTD][int(rng.integers(0, 4))]
+        cols, nulls, defs = [], [], []
+        for i, (t, dt) in enumerate(kinds):
+            if dt in (np.float32, np.float64):
+                a = rng.normal(size=n).astype(dt)
+            else:
+                info = np.iinfo(dt)
+                a = rng.integers(info.min // 2, info.max // 2, n).astype(dt)
+            cols.append(a)
+            p = float(rng.random() * 0.5) if rng.random() < 0.4 else 0.0
+            nulls.append((rng.random(n) < p).astype(np.uint8) if p else None)
+            defs.append((f"c{i}", t, 0))
+        with tempfile.TemporaryDirectory() as td:
+            path = os.path.join(td, "f.cs")
+            ca.write_table(path, defs, cols, nulls=nulls, compression=comp,
+                           stripe_row_limit=stripe, chunk_group_row_limit=chunk)
+            preds = []
+            for _ in range(int(rng.integers(0, 4))):
+                ci = int(rng.integers(0, ncols))
+                op = int(rng.integers(0, 6))
+                t, dt = kinds[ci]
+                if dt in (np.float32, np.float64):
+                    preds.append((ci, op, float(rng.normal())))
+                else:
+                    preds.append((ci, op, int(rng.integers(-100, 100))))
+            aggs = [(ca.AGG_COUNT_STAR, -1)]
+            for ci, (t, dt) in enumerate(kinds):
+                pool = FLT_AGGS if dt in (np.float32, np.float64) else INT_AGGS
+                aggs.append((pool[int(rng.integers(0, len(pool)))], ci))
+            op_, ofilt, gp_, gfilt = both(path, preds, aggs)
+            assert ofilt == gfilt, f"seed {seed}"
+            try:
+                assert_parity(op_, gp_, aggs)
+            except AssertionError as e:
+                raise AssertionError(f"fuzz seed {seed}: {e}")
