@@ -340,6 +340,18 @@ __device__ inline void lz4_lane_decode(const uint8_t *__restrict__ data,
              * so quads only when offset>=4; sources are then complete. */
             const uint8_t *msrc = sout + op - offset;
             uint32_t j = 0;
+            if (offset >= 8) {
+                for (; j + 8 <= mlen; j += 8) {
+                    uint8_t b0 = msrc[j], b1 = msrc[j + 1];
+                    uint8_t b2 = msrc[j + 2], b3 = msrc[j + 3];
+                    uint8_t b4 = msrc[j + 4], b5 = msrc[j + 5];
+                    uint8_t b6 = msrc[j + 6], b7 = msrc[j + 7];
+                    sout[op + j] = b0; sout[op + j + 1] = b1;
+                    sout[op + j + 2] = b2; sout[op + j + 3] = b3;
+                    sout[op + j + 4] = b4; sout[op + j + 5] = b5;
+                    sout[op + j + 6] = b6; sout[op + j + 7] = b7;
+                }
+            }
             if (offset >= 4) {
                 for (; j + 4 <= mlen; j += 4) {
                     uint8_t b0 = msrc[j], b1 = msrc[j + 1];
