@@ -858,22 +858,6 @@ __device__ inline void acc_prepare(PrepAcc &o, const AggD &g,
     }
 }
 
-/* fold a prepared contribution into a ThreadAcc (registers only) */
-__device__ inline void acc_apply(ThreadAcc &a, uint8_t kind, const PrepAcc &p)
-{
-    if (!p.valid) return;
-    switch (kind) {
-        case CSTRIPE_AGG_MIN_I64: a.lo = min(a.lo, p.lo); break;
-        case CSTRIPE_AGG_MAX_I64: a.lo = max(a.lo, p.lo); break;
-        case CSTRIPE_AGG_MIN_F64: a.f = fmin(a.f, p.f); break;
-        case CSTRIPE_AGG_MAX_F64: a.f = fmax(a.f, p.f); break;
-        case CSTRIPE_AGG_SUM_F64: a.f += p.f; break;
-        default:
-            acc_add_i128(a, ((__int128)p.hi << 64) | (unsigned long long)p.lo);
-            break;
-    }
-    a.cnt++;
-}
 
 /* =====================================================================
  * Grouped aggregation (TPC-H Q1 shape): GROUP BY 1-2 categorical i8

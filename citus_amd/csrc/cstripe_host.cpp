@@ -767,12 +767,6 @@ static bool agg_is_count(uint32_t k)
     return k == CSTRIPE_AGG_COUNT_STAR || k == CSTRIPE_AGG_COUNT_COL;
 }
 
-static bool agg_is_i128_sum(uint32_t k)
-{
-    return k == CSTRIPE_AGG_SUM_I64 || k == CSTRIPE_AGG_SUM_PROD_I64 ||
-           k == CSTRIPE_AGG_SUM_DISC_I64 || k == CSTRIPE_AGG_SUM_DISC_TAX_I64;
-}
-
 extern "C" int cagg_combine(const cstripe_agg_spec *aggs, uint32_t n_aggs,
                             const cstripe_partial *parts, uint32_t n_parts,
                             cstripe_partial *out)
@@ -835,7 +829,6 @@ extern "C" int cagg_combine(const cstripe_agg_spec *aggs, uint32_t n_aggs,
             acc.is_null = 0;
             acc.count = 0;
         }
-        (void)agg_is_i128_sum;
         out[a] = acc;
     }
     return CSTRIPE_OK;
