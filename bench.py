@@ -201,7 +201,8 @@ def main():
         profs = sorted(glob.glob(os.path.join(REPO, "profiles", "traffic_r*.json")))
         if profs:
             tj = json.load(open(profs[-1]))
-            if tj.get("workload_rows") == args.rows and tj.get("kernel") == dominant:
+            if (tj.get("workload_rows") == args.rows and tj.get("kernel") == dominant
+                    and tj.get("workload") == f"{args.query}_lineitem_{args.rows // 1_000_000}m_{args.compression}"):
                 traffic = tj.get("hbm_bytes_per_launch")
     except Exception:
         traffic = None
