@@ -83,6 +83,9 @@ def main():
                     help="LZ4 micro-segment size override (bytes)")
     ap.add_argument("--min-match", type=int, default=0,
                     help="writer LZ4 min match length (>=4; GPU-decode knob)")
+    ap.add_argument("--shards", type=int, default=1,
+                    help="shard files per GPU scanned as one table "
+                         "(config 3: 4 shards/GPU x 8 GPUs = 32 shards)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -106,12 +109,25 @@ def main():
     # ---- setup (untimed): generate per-rank shard, open, prune, stage ----
     cache = os.environ.get("CSTRIPE_BENCH_DIR", "/tmp/cstripe_bench")
     os.makedirs(cache, exist_ok=True)
-    shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}"
-                                f"_m{args.min_match}_r{rank}.cs")
-    t0 = time.time()
-    if not os.path.exists(shard):
-        ca.gen_lineitem(shard, args.rows, seed=42 + rank, compression=comp,
-                        seg_bytes=args.seg_bytes, min_match=args.min_match)
+    if args.shards > 1:
+        shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}"
+                                    f"_m{args.min_match}_n{args.shards}_r{rank}")
+        os.makedirs(shard, exist_ok=True)
+        t0 = time.time()
+        per = args.rows // args.shards
+        for i in range(args.shards):
+            p = os.path.join(shard, f"shard{i:02d}.cs")
+            if not os.path.exists(p):
+                ca.gen_lineitem(p, per + (args.rows % args.shards if i == 0 else 0),
+                                seed=42 + rank * args.shards + i, compression=comp,
+                                seg_bytes=args.seg_bytes, min_match=args.min_match)
+    else:
+        shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}"
+                                    f"_m{args.min_match}_r{rank}.cs")
+        t0 = time.time()
+        if not os.path.exists(shard):
+            ca.gen_lineitem(shard, args.rows, seed=42 + rank, compression=comp,
+                            seg_bytes=args.seg_bytes, min_match=args.min_match)
     gen_s = time.time() - t0
 
     reader = ca.Reader(shard)
@@ -274,7 +290,7 @@ def main():
             "compression": args.compression,
             "stripe_rows": 150000,
             "chunk_group_rows": 10000,
-            "parallelism": f"shard-dp{world}",
+            "parallelism": f"shard-dp{world}x{args.shards}",
             "staged_gb_compressed": round(staged_gb, 3),
             "gen_s": round(gen_s, 1),
             "stage_s": round(stage_s, 2),

@@ -1535,7 +1535,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         uint32_t rows = st.group_rows[sc.chunk];
         h_groups[gi].row_count = rows;
         h_groups[gi].colbase = gi * n_proj;
-        const uint8_t *stripe_base = r->map + st.meta.file_offset;
+        const uint8_t *stripe_base = r->stripe_base(st);
         for (uint32_t c = 0; c < r->head.column_count; c++) {
             int pj = g->proj_of_col[c];
             if (pj < 0) continue;
@@ -2148,7 +2148,7 @@ int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
     const cs_selchunk &sc = s->sel[gi];
     const cs_stripe_info &st = r->stripes[sc.stripe];
     uint32_t rows = st.group_rows[sc.chunk];
-    const uint8_t *stripe_base = r->map + st.meta.file_offset;
+    const uint8_t *stripe_base = r->stripe_base(st);
 
     std::vector<uint8_t> packed;
     for (uint32_t c = 0; c < r->head.column_count; c++) {

@@ -30,6 +30,8 @@
 #include <fcntl.h>
 #include <sys/mman.h>
 #include <sys/stat.h>
+#include <dirent.h>
+#include <string>
 #include <unistd.h>
 
 #ifdef _OPENMP
@@ -542,33 +544,53 @@ extern "C" void cstripe_write_abort(cstripe_writer *w)
 
 /* ============================ reader ============================ */
 
-static int footer_read(cstripe_reader *r)
+/* parse one shard file's footer, appending its stripes (tagged file_idx) */
+static int footer_read_one(cstripe_reader *r, uint32_t file_idx)
 {
-    if (r->map_size < CSF_HEADER_SIZE + 16) { cs_set_err("file too small"); return CSTRIPE_ERR_FORMAT; }
-    if (memcmp(r->map, CSF_MAGIC, 8) != 0) { cs_set_err("bad header magic"); return CSTRIPE_ERR_FORMAT; }
-    if (memcmp(r->map + r->map_size - 8, CSF_FOOT_MAGIC, 8) != 0) { cs_set_err("bad footer magic"); return CSTRIPE_ERR_FORMAT; }
+    const cs_file &fl = r->files[file_idx];
+    const uint8_t *map = fl.map;
+    size_t map_size = fl.map_size;
+    if (map_size < CSF_HEADER_SIZE + 16) { cs_set_err("file too small"); return CSTRIPE_ERR_FORMAT; }
+    if (memcmp(map, CSF_MAGIC, 8) != 0) { cs_set_err("bad header magic"); return CSTRIPE_ERR_FORMAT; }
+    if (memcmp(map + map_size - 8, CSF_FOOT_MAGIC, 8) != 0) { cs_set_err("bad footer magic"); return CSTRIPE_ERR_FORMAT; }
     uint64_t foff;
-    memcpy(&foff, r->map + r->map_size - 16, 8);
-    if (foff + 16 > r->map_size) { cs_set_err("bad footer offset"); return CSTRIPE_ERR_FORMAT; }
+    memcpy(&foff, map + map_size - 16, 8);
+    if (foff + 16 > map_size) { cs_set_err("bad footer offset"); return CSTRIPE_ERR_FORMAT; }
 
-    const uint8_t *p = r->map + foff;
-    const uint8_t *end = r->map + r->map_size - 16;
+    const uint8_t *p = map + foff;
+    const uint8_t *end = map + map_size - 16;
     auto need = [&](size_t n) -> bool { return (size_t)(end - p) >= n; };
 
+    csf_footer_head head;
     if (!need(sizeof(csf_footer_head))) { cs_set_err("truncated footer"); return CSTRIPE_ERR_FORMAT; }
-    memcpy(&r->head, p, sizeof(r->head));
-    p += sizeof(r->head);
-    if (r->head.version != CSF_VERSION) { cs_set_err("bad version %u", r->head.version); return CSTRIPE_ERR_FORMAT; }
-    if (r->head.column_count == 0 || r->head.column_count > 64) { cs_set_err("bad column count"); return CSTRIPE_ERR_FORMAT; }
+    memcpy(&head, p, sizeof(head));
+    p += sizeof(head);
+    if (head.version != CSF_VERSION) { cs_set_err("bad version %u", head.version); return CSTRIPE_ERR_FORMAT; }
+    if (head.column_count == 0 || head.column_count > 64) { cs_set_err("bad column count"); return CSTRIPE_ERR_FORMAT; }
 
-    r->cols.resize(r->head.column_count);
-    for (auto &c : r->cols) {
+    std::vector<csf_coldef> cols(head.column_count);
+    for (auto &c : cols) {
         if (!need(sizeof(c))) { cs_set_err("truncated coldefs"); return CSTRIPE_ERR_FORMAT; }
         memcpy(&c, p, sizeof(c));
         p += sizeof(c);
     }
-    r->stripes.resize(r->head.stripe_count);
-    for (auto &s : r->stripes) {
+    if (file_idx == 0) {
+        r->head = head;
+        r->cols = cols;
+    } else {
+        /* shard schema must match (same physical types/widths) */
+        if (head.column_count != r->head.column_count) { cs_set_err("shard schema mismatch"); return CSTRIPE_ERR_FORMAT; }
+        for (uint32_t i = 0; i < head.column_count; i++)
+            if (cols[i].type != r->cols[i].type) { cs_set_err("shard schema mismatch (col %u)", i); return CSTRIPE_ERR_FORMAT; }
+        r->head.total_rows += head.total_rows;
+        r->head.stripe_count += head.stripe_count;
+    }
+
+    size_t base = r->stripes.size();
+    r->stripes.resize(base + head.stripe_count);
+    for (size_t si = base; si < r->stripes.size(); si++) {
+        auto &s = r->stripes[si];
+        s.file_idx = file_idx;
         if (!need(sizeof(s.meta))) { cs_set_err("truncated stripe meta"); return CSTRIPE_ERR_FORMAT; }
         memcpy(&s.meta, p, sizeof(s.meta));
         p += sizeof(s.meta);
@@ -577,7 +599,7 @@ static int footer_read(cstripe_reader *r)
         if (!need(s.group_rows.size() * 4)) { cs_set_err("truncated group rows"); return CSTRIPE_ERR_FORMAT; }
         memcpy(s.group_rows.data(), p, s.group_rows.size() * 4);
         p += s.group_rows.size() * 4;
-        s.nodes.resize(r->head.column_count);
+        s.nodes.resize(head.column_count);
         for (auto &cn : s.nodes) {
             cn.resize(s.meta.chunk_count);
             for (auto &nd : cn) {
@@ -595,27 +617,60 @@ static int footer_read(cstripe_reader *r)
     return CSTRIPE_OK;
 }
 
-extern "C" cstripe_reader *cstripe_open(const char *path)
+static int map_one(cstripe_reader *r, const char *path)
 {
     int fd = open(path, O_RDONLY);
-    if (fd < 0) { cs_set_err("open %s failed", path); return nullptr; }
+    if (fd < 0) { cs_set_err("open %s failed", path); return CSTRIPE_ERR_IO; }
     struct stat st;
-    if (fstat(fd, &st) != 0 || st.st_size <= 0) { cs_set_err("stat failed"); close(fd); return nullptr; }
+    if (fstat(fd, &st) != 0 || st.st_size <= 0) { cs_set_err("stat %s failed", path); close(fd); return CSTRIPE_ERR_IO; }
     void *m = mmap(nullptr, (size_t)st.st_size, PROT_READ, MAP_PRIVATE, fd, 0);
-    if (m == MAP_FAILED) { cs_set_err("mmap failed"); close(fd); return nullptr; }
+    if (m == MAP_FAILED) { cs_set_err("mmap %s failed", path); close(fd); return CSTRIPE_ERR_IO; }
+    cs_file fl;
+    fl.fd = fd;
+    fl.map = (const uint8_t *)m;
+    fl.map_size = (size_t)st.st_size;
+    r->files.push_back(fl);
+    return CSTRIPE_OK;
+}
+
+/* path may be one stripe file, or a DIRECTORY of shard files (*.cs, sorted
+ * by name) scanned as one table — the static shard-group-per-GPU mapping of
+ * SURVEY §8e (the reference's shard placements, adaptive_executor.c) */
+extern "C" cstripe_reader *cstripe_open(const char *path)
+{
     auto *r = new cstripe_reader();
-    r->fd = fd;
-    r->map = (const uint8_t *)m;
-    r->map_size = (size_t)st.st_size;
-    if (footer_read(r) != CSTRIPE_OK) { cstripe_close(r); return nullptr; }
+    struct stat st;
+    if (stat(path, &st) != 0) { cs_set_err("stat %s failed", path); delete r; return nullptr; }
+    if (S_ISDIR(st.st_mode)) {
+        DIR *d = opendir(path);
+        if (!d) { cs_set_err("opendir %s failed", path); delete r; return nullptr; }
+        std::vector<std::string> names;
+        struct dirent *de;
+        while ((de = readdir(d)) != nullptr) {
+            std::string n = de->d_name;
+            if (n.size() > 3 && n.substr(n.size() - 3) == ".cs")
+                names.push_back(std::string(path) + "/" + n);
+        }
+        closedir(d);
+        std::sort(names.begin(), names.end());
+        if (names.empty()) { cs_set_err("no *.cs shard files in %s", path); delete r; return nullptr; }
+        for (auto &n : names)
+            if (map_one(r, n.c_str()) != CSTRIPE_OK) { cstripe_close(r); return nullptr; }
+    } else {
+        if (map_one(r, path) != CSTRIPE_OK) { cstripe_close(r); return nullptr; }
+    }
+    for (uint32_t i = 0; i < r->files.size(); i++)
+        if (footer_read_one(r, i) != CSTRIPE_OK) { cstripe_close(r); return nullptr; }
     return r;
 }
 
 extern "C" void cstripe_close(cstripe_reader *r)
 {
     if (!r) return;
-    if (r->map) munmap((void *)r->map, r->map_size);
-    if (r->fd >= 0) close(r->fd);
+    for (auto &fl : r->files) {
+        if (fl.map) munmap((void *)fl.map, fl.map_size);
+        if (fl.fd >= 0) close(fl.fd);
+    }
     delete r;
 }
 

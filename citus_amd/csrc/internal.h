@@ -22,17 +22,30 @@ struct cs_skipnode {
 
 struct cs_stripe_info {
     csf_stripe_meta meta;
+    uint32_t file_idx = 0;                       /* which mapped shard file */
     std::vector<uint32_t> group_rows;            /* [chunk] */
     std::vector<std::vector<cs_skipnode>> nodes; /* [col][chunk] */
 };
 
-struct cstripe_reader {
+/* one mapped shard file */
+struct cs_file {
     int fd = -1;
     const uint8_t *map = nullptr;
     size_t map_size = 0;
+};
+
+/* a table = one stripe file OR a directory of shard files (the static
+ * shard-group mapping of SURVEY §8e: N shards scanned as one table, one
+ * partial out) */
+struct cstripe_reader {
+    std::vector<cs_file> files;
     csf_footer_head head{};
     std::vector<csf_coldef> cols;
     std::vector<cs_stripe_info> stripes;
+    const uint8_t *stripe_base(const cs_stripe_info &st) const
+    {
+        return files[st.file_idx].map + st.meta.file_offset;
+    }
 };
 
 struct cs_selchunk {
