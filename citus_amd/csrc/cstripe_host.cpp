@@ -839,6 +839,7 @@ extern "C" int cagg_combine(const cstripe_agg_spec *aggs, uint32_t n_aggs,
                 case CSTRIPE_AGG_COUNT_STAR:
                 case CSTRIPE_AGG_COUNT_COL:
                     acc.count += in.count;
+                    acc.i128_lo = acc.count;     /* keep the value mirror consistent */
                     break;
                 case CSTRIPE_AGG_SUM_I64:
                 case CSTRIPE_AGG_SUM_PROD_I64:
