@@ -178,3 +178,31 @@ def test_exists_bitpacking_matches_reference_layout(tmp_path):
         if not nl[i]:
             expect[i // 8] |= 1 << (i % 8)
     assert bits == bytes(expect)
+
+
+def test_corrupted_footer_fails_cleanly(tmp_path):
+    """Truncated/corrupted files: clean error, no crash (the ABI's ereport
+    analog)."""
+    import citus_amd as ca2
+    path = str(tmp_path / "c.cs")
+    ca2.gen_lineitem(path, 10_000)
+    raw = bytearray(open(path, "rb").read())
+    # corrupt footer magic
+    bad1 = str(tmp_path / "bad1.cs")
+    b = bytearray(raw)
+    b[-4:] = b"XXXX"
+    open(bad1, "wb").write(bytes(b))
+    with pytest.raises(ca2.CStripeError):
+        ca2.Reader(bad1)
+    # truncate mid-file
+    bad2 = str(tmp_path / "bad2.cs")
+    open(bad2, "wb").write(bytes(raw[:len(raw) // 3]))
+    with pytest.raises(ca2.CStripeError):
+        ca2.Reader(bad2)
+    # absurd footer offset
+    bad3 = str(tmp_path / "bad3.cs")
+    b = bytearray(raw)
+    b[-16:-8] = (2**60).to_bytes(8, "little")
+    open(bad3, "wb").write(bytes(b))
+    with pytest.raises(ca2.CStripeError):
+        ca2.Reader(bad3)

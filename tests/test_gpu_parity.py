@@ -530,3 +530,46 @@ def test_fuzz_grouped_differential():
                     assert res[k][i].count == ores[k][i].count, (seed, k, i)
                     assert res[k][i].is_null == ores[k][i].is_null, (seed, k, i)
                     assert res[k][i].i128 == ores[k][i].i128, (seed, k, i)
+
+
+def test_corrupted_stream_flags_error(tmp_path):
+    """Flip bytes inside a compressed value stream: the decode kernel must
+    flag the malformed block (or, at worst, decode to the declared size) —
+    never crash or hang."""
+    path = str(tmp_path / "c.cs")
+    n = 50_000
+    ca.write_table(path, [("a", ca.I64, 0)],
+                   [RNG.integers(0, 1000, n).astype(np.int64)],
+                   compression=ca.COMP_LZ4)
+    raw = bytearray(open(path, "rb").read())
+    # smash a 64-byte span in the middle of the data region
+    mid = len(raw) // 3
+    for i in range(64):
+        raw[mid + i] ^= 0xA5
+    bad = str(tmp_path / "bad.cs")
+    open(bad, "wb").write(bytes(raw))
+    try:
+        with ca.Reader(bad) as r, r.scan(cols_mask=1) as s:
+            s.stage()
+            try:
+                s.agg([(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_STAR, -1)])
+                # rare: the flips may land outside selected streams or still
+                # form a valid block — surviving without a crash is the bar
+            except ca.CStripeError as e:
+                assert "decode" in str(e).lower() or "format" in str(e).lower()
+    except ca.CStripeError:
+        pass   # footer-region corruption: clean open/stage failure also fine
+
+
+def test_stage_cycle_no_leak(tmp_path):
+    """Repeated stage/end cycles: device allocations are released (a leak
+    would OOM long before 120 iterations of ~60 MB)."""
+    path = str(tmp_path / "l.cs")
+    ca.gen_lineitem(path, 2_000_000)
+    aggs = [(ca.AGG_COUNT_STAR, -1)]
+    with ca.Reader(path) as r:
+        for i in range(120):
+            with r.scan(cols_mask=1 << 1, preds=[(1, ca.PRED_LT, 2400)]) as s:
+                s.stage()
+                parts = s.agg(aggs)
+                assert parts[0].count > 0
