@@ -482,16 +482,30 @@ __device__ inline bool col_value(const uint8_t *__restrict__ data,
     return true;
 }
 
+/* PG float ordering (float.c float8_cmp_internal): NaN sorts greater than
+ * every non-NaN and equal to itself — predicate + MIN/MAX semantics must
+ * match the reference's float8 operators, not IEEE compares */
+__device__ inline int f64cmp_pg(double a, double b)
+{
+    if (a > b) return 1;
+    if (a < b) return -1;
+    if (a == b) return 0;
+    const bool na = (a != a), nb = (b != b);
+    if (na && nb) return 0;
+    return na ? 1 : -1;
+}
+
 __device__ inline bool pred_eval(const PredD &p, int64_t iv, double fv)
 {
     if (p.is_float) {
+        const int c = f64cmp_pg(fv, p.fval);
         switch (p.op) {
-            case CSTRIPE_PRED_LT: return fv <  p.fval;
-            case CSTRIPE_PRED_LE: return fv <= p.fval;
-            case CSTRIPE_PRED_GT: return fv >  p.fval;
-            case CSTRIPE_PRED_GE: return fv >= p.fval;
-            case CSTRIPE_PRED_EQ: return fv == p.fval;
-            default:              return fv != p.fval;
+            case CSTRIPE_PRED_LT: return c <  0;
+            case CSTRIPE_PRED_LE: return c <= 0;
+            case CSTRIPE_PRED_GT: return c >  0;
+            case CSTRIPE_PRED_GE: return c >= 0;
+            case CSTRIPE_PRED_EQ: return c == 0;
+            default:              return c != 0;
         }
     }
     switch (p.op) {
@@ -516,7 +530,9 @@ __device__ inline void acc_init(ThreadAcc &a, uint8_t kind)
     switch (kind) {
         case CSTRIPE_AGG_MIN_I64: a.lo = INT64_MAX; break;
         case CSTRIPE_AGG_MAX_I64: a.lo = INT64_MIN; break;
-        case CSTRIPE_AGG_MIN_F64: a.f = INFINITY; break;
+        /* PG order: NaN is the greatest float, so NaN is MIN's identity
+         * (min(NaN, x) = x; min over only-NaN rows correctly stays NaN) */
+        case CSTRIPE_AGG_MIN_F64: a.f = NAN; break;
         case CSTRIPE_AGG_MAX_F64: a.f = -INFINITY; break;
         default: break;
     }
@@ -535,8 +551,8 @@ __device__ inline void acc_merge(ThreadAcc &a, const ThreadAcc &b, uint8_t kind)
     switch (kind) {
         case CSTRIPE_AGG_MIN_I64: a.lo = min(a.lo, b.lo); break;
         case CSTRIPE_AGG_MAX_I64: a.lo = max(a.lo, b.lo); break;
-        case CSTRIPE_AGG_MIN_F64: a.f = fmin(a.f, b.f); break;
-        case CSTRIPE_AGG_MAX_F64: a.f = fmax(a.f, b.f); break;
+        case CSTRIPE_AGG_MIN_F64: if (f64cmp_pg(b.f, a.f) < 0) a.f = b.f; break;
+        case CSTRIPE_AGG_MAX_F64: if (f64cmp_pg(b.f, a.f) > 0) a.f = b.f; break;
         case CSTRIPE_AGG_SUM_F64: a.f += b.f; break;
         default: {  /* counts + i128 sums */
             unsigned long long lo = (unsigned long long)a.lo + (unsigned long long)b.lo;
@@ -580,10 +596,16 @@ __device__ inline void acc_row(ThreadAcc &a, const AggD &g,
             if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.lo = max(a.lo, iv); a.cnt++; }
             break;
         case CSTRIPE_AGG_MIN_F64:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.f = fmin(a.f, fv); a.cnt++; }
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) {
+                if (f64cmp_pg(fv, a.f) < 0) a.f = fv;
+                a.cnt++;
+            }
             break;
         case CSTRIPE_AGG_MAX_F64:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.f = fmax(a.f, fv); a.cnt++; }
+            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) {
+                if (f64cmp_pg(fv, a.f) > 0) a.f = fv;
+                a.cnt++;
+            }
             break;
         case CSTRIPE_AGG_SUM_PROD_I64: {
             int64_t ib; double fb;
@@ -902,7 +924,7 @@ __device__ inline void acc_apply_atomic(ThreadAcc *cell, uint8_t kind, const Pre
             while (true) {
                 double c;
                 memcpy(&c, &cur, 8);
-                if (v >= c) break;
+                if (f64cmp_pg(v, c) >= 0) break;   /* PG order (NaN high) */
                 unsigned long long nv;
                 memcpy(&nv, &v, 8);
                 unsigned long long prev = atomicCAS(addr, cur, nv);
@@ -918,7 +940,7 @@ __device__ inline void acc_apply_atomic(ThreadAcc *cell, uint8_t kind, const Pre
             while (true) {
                 double c;
                 memcpy(&c, &cur, 8);
-                if (v <= c) break;
+                if (f64cmp_pg(v, c) <= 0) break;   /* PG order (NaN high) */
                 unsigned long long nv;
                 memcpy(&nv, &v, 8);
                 unsigned long long prev = atomicCAS(addr, cur, nv);
@@ -1275,7 +1297,7 @@ __device__ inline void cell_merge_atomic(ThreadAcc *dst, uint8_t kind, const Acc
             while (true) {
                 double c;
                 memcpy(&c, &cur, 8);
-                if (src.f >= c) break;
+                if (f64cmp_pg(src.f, c) >= 0) break;   /* PG order (NaN high) */
                 unsigned long long nv;
                 memcpy(&nv, &src.f, 8);
                 unsigned long long prev = atomicCAS(addr, cur, nv);
@@ -1290,7 +1312,7 @@ __device__ inline void cell_merge_atomic(ThreadAcc *dst, uint8_t kind, const Acc
             while (true) {
                 double c;
                 memcpy(&c, &cur, 8);
-                if (src.f <= c) break;
+                if (f64cmp_pg(src.f, c) <= 0) break;   /* PG order (NaN high) */
                 unsigned long long nv;
                 memcpy(&nv, &src.f, 8);
                 unsigned long long prev = atomicCAS(addr, cur, nv);
@@ -1584,11 +1606,14 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 uint32_t segi = 0;
                 const auto &seglist = st.nodes[c][sc.chunk].segs;
                 for (const csf_seg &sg : seglist) {
-                    /* non-final segments exactly 256 B; the final one must
-                     * still fit a FUSE_STRIDE LDS region */
+                    /* non-final segments exactly 256 B, final <= 256 B: the
+                     * fused kernels map row -> lane region as byteoff>>8 and
+                     * count tile segments as ceil(bytes/256), so an absorbed
+                     * 257-271 B final segment would read the wrong region
+                     * and overrun the segment list (round-1 advisor) */
                     if (sg.decomp_off != segi * 256u ||
                         (segi + 1 < seglist.size() ? sg.decomp_len != 256
-                                                   : sg.decomp_len > FUSE_STRIDE - 8)) {
+                                                   : sg.decomp_len > 256)) {
                         g->fusable = false;
                         g->fusable_mixed = false;
                     }
@@ -1709,9 +1734,9 @@ int csgpu_stage(cstripe_scan *s, int device_id)
 /* launch the decode grid: LDS-staged variant when every segment fits the
  * 64 KiB budget (writer default 8 KiB segments -> ~9 waves/CU), else the
  * global-memory fallback */
-static void launch_decode(cs_gpu_state *g)
+static int launch_decode(cs_gpu_state *g)
 {
-    if (g->n_segs == 0) return;
+    if (g->n_segs == 0) return CSTRIPE_OK;
     /* lane-parallel path for micro-segments (one lane per segment).
      * stride: 16 B-multiple, (stride/4)%64 != 0 so equal-progress lanes land
      * on different banks; LDS/wave = 64*stride -> waves/CU 9 / 4 / 2. */
@@ -1725,7 +1750,8 @@ static void launch_decode(cs_gpu_state *g)
                            block * stride + block * 16, g->stream,
                            g->d_data, g->d_scratch, g->d_segs, g->n_segs, stride,
                            g->d_error);
-        return;
+        HIP_TRY(hipGetLastError());
+        return CSTRIPE_OK;
     }
     uint32_t in_cap = (g->max_seg_comp + 8 + 15) & ~15u;
     uint32_t out_cap = (g->max_seg_dlen + 15) & ~15u;
@@ -1737,6 +1763,10 @@ static void launch_decode(cs_gpu_state *g)
         hipLaunchKernelGGL(lz4_decode_kernel, dim3(g->n_segs), dim3(WAVE), 0, g->stream,
                            g->d_data, g->d_scratch, g->d_segs, g->d_error);
     }
+    /* a failed launch would leave d_scratch stale while later memcpys
+       succeed -> silent garbage; surface it as an error (round-1 advisor) */
+    HIP_TRY(hipGetLastError());
+    return CSTRIPE_OK;
 }
 
 /* =====================================================================
@@ -1844,10 +1874,12 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             else if (n_aggs == 1) launchfg(fused_grouped_kernel<1>);
             else if (n_aggs == 2) launchfg(fused_grouped_kernel<2>);
             else launchfg(fused_grouped_kernel<-1>);
+            HIP_TRY(hipGetLastError());
             HIP_TRY(hipEventRecord(g->ev1, g->stream));
             hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
                                g->d_gkeys, g->d_gcells, fgrid, FUSEG_SLOTS,
                                g->d_gfkeys, g->d_gfcells, g->d_gn, g->d_error, gp);
+        HIP_TRY(hipGetLastError());
             HIP_TRY(hipEventRecord(g->ev2, g->stream));
             int h_err = 0;
             HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
@@ -1903,7 +1935,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         uint32_t lds = n_waves * GRP_SLOTS * 2 + n_waves * GRP_SLOTS * n_aggs * (uint32_t)sizeof(ThreadAcc);
         HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
         HIP_TRY(hipEventRecord(g->ev0, g->stream));
-        launch_decode(g);
+        { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
         HIP_TRY(hipEventRecord(g->ev1, g->stream));
         {
             auto launchg = [&](auto *kern) {
@@ -1915,10 +1947,12 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             else if (n_aggs == 1) launchg(grouped_agg_kernel<1>);
             else if (n_aggs == 2) launchg(grouped_agg_kernel<2>);
             else launchg(grouped_agg_kernel<-1>);
+            HIP_TRY(hipGetLastError());
         }
         hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
                            g->d_gkeys, g->d_gcells, grid, per_block,
                            g->d_gfkeys, g->d_gfcells, g->d_gn, g->d_error, gp);
+        HIP_TRY(hipGetLastError());
         HIP_TRY(hipEventRecord(g->ev2, g->stream));
 
         uint16_t h_keys[CSTRIPE_MAX_GROUPS];
@@ -2010,9 +2044,11 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         else if (p.n_preds == 1 && n_aggs == 1) launchf(fused_agg_kernel<1, 1>);
         else if (p.n_preds == 2 && n_aggs == 2) launchf(fused_agg_kernel<2, 2>);
         else launchf(fused_agg_kernel<-1, -1>);
+        HIP_TRY(hipGetLastError());
         HIP_TRY(hipEventRecord(g->ev1, g->stream));
         hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
                            g->d_block, g->n_tiles, g->d_final, p);
+        HIP_TRY(hipGetLastError());
         HIP_TRY(hipEventRecord(g->ev2, g->stream));
 
         AccCell h_final[MAX_AGGS];
@@ -2053,7 +2089,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
 
     s->last_fused = 0;
     HIP_TRY(hipEventRecord(g->ev0, g->stream));
-    launch_decode(g);
+    { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
     HIP_TRY(hipEventRecord(g->ev1, g->stream));
     {
         auto launch = [&](auto *kern) {
@@ -2067,9 +2103,11 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         else if (p.n_preds == 1 && n_aggs == 1) launch(filter_agg_kernel<1, 1>);
         else if (p.n_preds == 2 && n_aggs == 2) launch(filter_agg_kernel<2, 2>);
         else launch(filter_agg_kernel<-1, -1>);
+        HIP_TRY(hipGetLastError());
     }
     hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
                        g->d_block, n_blocks, g->d_final, p);
+        HIP_TRY(hipGetLastError());
     HIP_TRY(hipEventRecord(g->ev2, g->stream));
 
     AccCell h_final[MAX_AGGS];
@@ -2137,7 +2175,7 @@ int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
     /* make sure scratch holds decoded data (decode everything once per rewind) */
     if (g->n_segs > 0 && s->batch_pos == 0) {
         HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
-        launch_decode(g);
+        { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
         int h_err = 0;
         HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
         HIP_TRY(hipStreamSynchronize(g->stream));

@@ -68,6 +68,20 @@ extern "C" void cstripe_default_options(cstripe_options *o)
 
 static uint32_t type_width(uint8_t t) { return csf_type_width(t); }
 
+/* PG float ordering (backend/utils/adt/float.c float8_cmp_internal):
+ * NaN sorts greater than every non-NaN and equal to itself. Used for
+ * writer min/max skip nodes and for chunk refutation so float pruning
+ * matches the reference's operator semantics (round-1 advisor finding). */
+static inline int f64_cmp_pg(double a, double b)
+{
+    if (a > b) return 1;
+    if (a < b) return -1;
+    if (a == b) return 0;
+    const bool na = (a != a), nb = (b != b);
+    if (na && nb) return 0;
+    return na ? 1 : -1;
+}
+
 /* ============================ writer ============================ */
 
 namespace {
@@ -155,26 +169,31 @@ static void update_minmax(ColCur &c, uint8_t type, const uint8_t *vals, size_t n
         }
         case CSTRIPE_F32: {
             const float *v = (const float *)vals;
-            float mn = v[0], mx = v[0];
-            for (size_t i = 1; i < n; i++) { if (v[i] < mn) mn = v[i]; if (v[i] > mx) mx = v[i]; }
-            double dmn = mn, dmx = mx;
-            if (!c.has_min_max) { memcpy(&c.min_i, &dmn, 8); memcpy(&c.max_i, &dmx, 8); c.has_min_max = true; }
+            double mn = v[0], mx = v[0];
+            for (size_t i = 1; i < n; i++) {
+                if (f64_cmp_pg(v[i], mn) < 0) mn = v[i];
+                if (f64_cmp_pg(v[i], mx) > 0) mx = v[i];
+            }
+            if (!c.has_min_max) { memcpy(&c.min_i, &mn, 8); memcpy(&c.max_i, &mx, 8); c.has_min_max = true; }
             else {
                 double omn, omx; memcpy(&omn, &c.min_i, 8); memcpy(&omx, &c.max_i, 8);
-                if (dmn < omn) memcpy(&c.min_i, &dmn, 8);
-                if (dmx > omx) memcpy(&c.max_i, &dmx, 8);
+                if (f64_cmp_pg(mn, omn) < 0) memcpy(&c.min_i, &mn, 8);
+                if (f64_cmp_pg(mx, omx) > 0) memcpy(&c.max_i, &mx, 8);
             }
             break;
         }
         case CSTRIPE_F64: {
             const double *v = (const double *)vals;
             double mn = v[0], mx = v[0];
-            for (size_t i = 1; i < n; i++) { if (v[i] < mn) mn = v[i]; if (v[i] > mx) mx = v[i]; }
+            for (size_t i = 1; i < n; i++) {
+                if (f64_cmp_pg(v[i], mn) < 0) mn = v[i];
+                if (f64_cmp_pg(v[i], mx) > 0) mx = v[i];
+            }
             if (!c.has_min_max) { memcpy(&c.min_i, &mn, 8); memcpy(&c.max_i, &mx, 8); c.has_min_max = true; }
             else {
                 double omn, omx; memcpy(&omn, &c.min_i, 8); memcpy(&omx, &c.max_i, 8);
-                if (mn < omn) memcpy(&c.min_i, &mn, 8);
-                if (mx > omx) memcpy(&c.max_i, &mx, 8);
+                if (f64_cmp_pg(mn, omn) < 0) memcpy(&c.min_i, &mn, 8);
+                if (f64_cmp_pg(mx, omx) > 0) memcpy(&c.max_i, &mx, 8);
             }
             break;
         }
@@ -255,8 +274,9 @@ static void finalize_chunk(cstripe_writer *w, uint32_t rows_in_chunk)
 
 /* compress one pending chunk's value stream into out (segments); mirrors
  * SerializeChunkData/CompressBuffer: lz4/zstd per segment, raw when the codec
- * is NONE or fails */
-static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, ChunkOut &out)
+ * is NONE or fails. Returns false only on a hard format-capacity error
+ * (> CSF segment cap), never on incompressible data. */
+static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, ChunkOut &out)
 {
     out.node = pc.node;
     out.exists_packed = pc.exists_packed;
@@ -274,7 +294,7 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
         out.node.comp_level = 0;
         out.node.n_segs = 1;
         out.node.value_len = raw.size();
-        return;
+        return true;
     }
 
     /* segment boundaries 16 B-aligned (pure byte split of the raw stream;
@@ -295,7 +315,13 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     if (codec == CSTRIPE_COMP_LZ4) lz4e_init(&enc);
     while (off < n) {
         size_t len = std::min(per, n - off);
-        if (n - off - len < 16) len = n - off;      /* absorb the tail */
+        /* absorb a tiny (<16 B) tail only in large-segment mode: the GPU
+         * fused/lane kernels assume non-final segments are EXACTLY `per`
+         * decompressed bytes and the final one is <= per, so for the
+         * 256 B lane-parallel default a 257-271 B absorbed final segment
+         * would be read from the wrong lane region (round-1 advisor
+         * finding); a standalone <16 B final segment decodes fine */
+        if (per > 256 && n - off - len < 16) len = n - off;
         csf_seg s{};
         s.decomp_off = (uint32_t)off;
         s.decomp_len = (uint32_t)len;
@@ -341,8 +367,18 @@ static void compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
         out.node.comp_type = codec;
         out.node.comp_level = w->opts.compression_level;
     }
+    /* reader cap is 4096 segments and n_segs is uint16 — refuse to emit a
+     * chunk the reader would reject (or, after wraparound, misparse) instead
+     * of silently truncating (round-1 advisor finding) */
+    if (out.segs.size() > 4096) {
+        cs_set_err("chunk needs %zu segments (> 4096 cap): raise "
+                   "lz4_seg_target_bytes or lower chunk_group_row_limit",
+                   out.segs.size());
+        return false;
+    }
     out.node.n_segs = (uint16_t)out.segs.size();
     out.node.value_len = out.value_comp.size();
+    return true;
 }
 
 /* FlushStripe (columnar_writer.c:391-516): compress pending chunks, lay out
@@ -362,9 +398,18 @@ static int flush_stripe(cstripe_writer *w)
 
     /* compress all pending chunks (parallel; independent per column-chunk) */
     std::vector<ChunkOut> outs(w->pending.size());
+    bool comp_ok = true;
     #pragma omp parallel for schedule(dynamic)
     for (long i = 0; i < (long)w->pending.size(); i++)
-        compress_chunk(w, w->pending[i], outs[i]);
+        if (!compress_chunk(w, w->pending[i], outs[i]))
+            comp_ok = false;
+    if (!comp_ok) {
+        /* compress_chunk set the detailed message on its OMP worker thread's
+         * buffer; restate on the caller's thread where errmsg() reads */
+        cs_set_err("chunk exceeds the 4096-segment cap: raise "
+                   "lz4_seg_target_bytes or lower chunk_group_row_limit");
+        return CSTRIPE_ERR_FORMAT;
+    }
 
     /* outs indexed by pending order: chunk-major (all columns of chunk 0, then
      * chunk 1, ...). Reorder access as [col][chunk]. */
@@ -582,6 +627,14 @@ static int footer_read_one(cstripe_reader *r, uint32_t file_idx)
         if (head.column_count != r->head.column_count) { cs_set_err("shard schema mismatch"); return CSTRIPE_ERR_FORMAT; }
         for (uint32_t i = 0; i < head.column_count; i++)
             if (cols[i].type != r->cols[i].type) { cs_set_err("shard schema mismatch (col %u)", i); return CSTRIPE_ERR_FORMAT; }
+        /* scan tiling derives tiles_per_group from file 0's chunk_row_limit;
+         * a shard with a larger limit would have rows past that bound
+         * silently skipped (round-1 advisor finding) — reject mismatches */
+        if (head.chunk_row_limit != r->head.chunk_row_limit) {
+            cs_set_err("shard chunk_group_row_limit mismatch (%u vs %u)",
+                       head.chunk_row_limit, r->head.chunk_row_limit);
+            return CSTRIPE_ERR_FORMAT;
+        }
         r->head.total_rows += head.total_rows;
         r->head.stripe_count += head.stripe_count;
     }
@@ -700,13 +753,17 @@ static bool pred_refutes(const cstripe_pred &p, uint8_t type, int64_t min_i, int
         memcpy(&mn, &min_i, 8);
         memcpy(&mx, &max_i, 8);
         double c = p.fval;
+        /* PG float ordering throughout (NaN high, NaN == NaN) — the
+         * reference goes through predicate_refuted_by with float8 operator
+         * semantics, so e.g. a chunk whose only rows above c are NaN must
+         * NOT be pruned for `col > c` (its max IS NaN under this order) */
         switch (p.op) {
-            case CSTRIPE_PRED_LT: return mn >= c;
-            case CSTRIPE_PRED_LE: return mn > c;
-            case CSTRIPE_PRED_GT: return mx <= c;
-            case CSTRIPE_PRED_GE: return mx < c;
-            case CSTRIPE_PRED_EQ: return c < mn || c > mx;
-            case CSTRIPE_PRED_NE: return mn == c && mx == c;
+            case CSTRIPE_PRED_LT: return f64_cmp_pg(mn, c) >= 0;
+            case CSTRIPE_PRED_LE: return f64_cmp_pg(mn, c) > 0;
+            case CSTRIPE_PRED_GT: return f64_cmp_pg(mx, c) <= 0;
+            case CSTRIPE_PRED_GE: return f64_cmp_pg(mx, c) < 0;
+            case CSTRIPE_PRED_EQ: return f64_cmp_pg(c, mn) < 0 || f64_cmp_pg(c, mx) > 0;
+            case CSTRIPE_PRED_NE: return f64_cmp_pg(mn, c) == 0 && f64_cmp_pg(mx, c) == 0;
         }
         return false;
     }
@@ -868,11 +925,11 @@ extern "C" int cagg_combine(const cstripe_agg_spec *aggs, uint32_t n_aggs,
                     acc.count += in.count;
                     break;
                 case CSTRIPE_AGG_MIN_F64:
-                    acc.f64 = std::min(acc.f64, in.f64);
+                    if (f64_cmp_pg(in.f64, acc.f64) < 0) acc.f64 = in.f64;
                     acc.count += in.count;
                     break;
                 case CSTRIPE_AGG_MAX_F64:
-                    acc.f64 = std::max(acc.f64, in.f64);
+                    if (f64_cmp_pg(in.f64, acc.f64) > 0) acc.f64 = in.f64;
                     acc.count += in.count;
                     break;
                 default:

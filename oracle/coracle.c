@@ -244,6 +244,20 @@ int oracle_read_chunk(const oracle_table *t, uint32_t stripe, uint32_t chunk,
 
 /* ---------------- pruning (SelectedChunkMask) ---------------- */
 
+/* PG float ordering (float.c float8_cmp_internal, restated): NaN sorts
+ * greater than every non-NaN and equal to itself. */
+static int o_f64_cmp_pg(double a, double b)
+{
+    if (a > b) return 1;
+    if (a < b) return -1;
+    if (a == b) return 0;
+    {
+        const int na = (a != a), nb = (b != b);
+        if (na && nb) return 0;
+        return na ? 1 : -1;
+    }
+}
+
 static int o_pred_refutes(const cstripe_pred *p, uint8_t type, int64_t min_i, int64_t max_i)
 {
     if (type == CSTRIPE_F32 || type == CSTRIPE_F64) {
@@ -251,13 +265,15 @@ static int o_pred_refutes(const cstripe_pred *p, uint8_t type, int64_t min_i, in
         memcpy(&mn, &min_i, 8);
         memcpy(&mx, &max_i, 8);
         double c = p->fval;
+        /* PG float ordering (NaN high): matches predicate_refuted_by with
+         * float8 operator semantics */
         switch (p->op) {
-            case CSTRIPE_PRED_LT: return mn >= c;
-            case CSTRIPE_PRED_LE: return mn > c;
-            case CSTRIPE_PRED_GT: return mx <= c;
-            case CSTRIPE_PRED_GE: return mx < c;
-            case CSTRIPE_PRED_EQ: return c < mn || c > mx;
-            case CSTRIPE_PRED_NE: return mn == c && mx == c;
+            case CSTRIPE_PRED_LT: return o_f64_cmp_pg(mn, c) >= 0;
+            case CSTRIPE_PRED_LE: return o_f64_cmp_pg(mn, c) > 0;
+            case CSTRIPE_PRED_GT: return o_f64_cmp_pg(mx, c) <= 0;
+            case CSTRIPE_PRED_GE: return o_f64_cmp_pg(mx, c) < 0;
+            case CSTRIPE_PRED_EQ: return o_f64_cmp_pg(c, mn) < 0 || o_f64_cmp_pg(c, mx) > 0;
+            case CSTRIPE_PRED_NE: return o_f64_cmp_pg(mn, c) == 0 && o_f64_cmp_pg(mx, c) == 0;
         }
         return 0;
     }
@@ -288,7 +304,7 @@ static void oacc_init(oacc *a, uint32_t kind)
     memset(a, 0, sizeof(*a));
     if (kind == CSTRIPE_AGG_MIN_I64) a->minmax_i = INT64_MAX;
     if (kind == CSTRIPE_AGG_MAX_I64) a->minmax_i = INT64_MIN;
-    if (kind == CSTRIPE_AGG_MIN_F64) a->minmax_f = 1.0 / 0.0;
+    if (kind == CSTRIPE_AGG_MIN_F64) a->minmax_f = 0.0 / 0.0;   /* NaN: PG MIN identity (NaN high) */
     if (kind == CSTRIPE_AGG_MAX_F64) a->minmax_f = -1.0 / 0.0;
 }
 
@@ -351,13 +367,14 @@ static int o_get(const ochunkcol *cc, uint32_t row, int64_t *iv, double *fv)
 static int o_pred_eval(const cstripe_pred *p, uint8_t is_float, int64_t iv, double fv)
 {
     if (is_float) {
+        const int c = o_f64_cmp_pg(fv, p->fval);   /* PG order: NaN high, NaN==NaN */
         switch (p->op) {
-            case CSTRIPE_PRED_LT: return fv <  p->fval;
-            case CSTRIPE_PRED_LE: return fv <= p->fval;
-            case CSTRIPE_PRED_GT: return fv >  p->fval;
-            case CSTRIPE_PRED_GE: return fv >= p->fval;
-            case CSTRIPE_PRED_EQ: return fv == p->fval;
-            default:              return fv != p->fval;
+            case CSTRIPE_PRED_LT: return c <  0;
+            case CSTRIPE_PRED_LE: return c <= 0;
+            case CSTRIPE_PRED_GT: return c >  0;
+            case CSTRIPE_PRED_GE: return c >= 0;
+            case CSTRIPE_PRED_EQ: return c == 0;
+            default:              return c != 0;
         }
     }
     switch (p->op) {
@@ -484,10 +501,10 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
                             if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (iv > ac->minmax_i) ac->minmax_i = iv; ac->cnt++; }
                             break;
                         case CSTRIPE_AGG_MIN_F64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (fv < ac->minmax_f) ac->minmax_f = fv; ac->cnt++; }
+                            if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (o_f64_cmp_pg(fv, ac->minmax_f) < 0) ac->minmax_f = fv; ac->cnt++; }
                             break;
                         case CSTRIPE_AGG_MAX_F64:
-                            if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (fv > ac->minmax_f) ac->minmax_f = fv; ac->cnt++; }
+                            if (o_get(&cc[g->col_a], row, &iv, &fv)) { if (o_f64_cmp_pg(fv, ac->minmax_f) > 0) ac->minmax_f = fv; ac->cnt++; }
                             break;
                         case CSTRIPE_AGG_SUM_PROD_I64:
                             if (o_get(&cc[g->col_a], row, &iv, &fv) &&
@@ -718,8 +735,8 @@ int oracle_scan_agg_mt(oracle_table *t, const cstripe_pred *preds, uint32_t n_pr
             m.f += x->f;
             if (aggs[a].kind == CSTRIPE_AGG_MIN_I64 && x->minmax_i < m.minmax_i) m.minmax_i = x->minmax_i;
             if (aggs[a].kind == CSTRIPE_AGG_MAX_I64 && x->minmax_i > m.minmax_i) m.minmax_i = x->minmax_i;
-            if (aggs[a].kind == CSTRIPE_AGG_MIN_F64 && x->minmax_f < m.minmax_f) m.minmax_f = x->minmax_f;
-            if (aggs[a].kind == CSTRIPE_AGG_MAX_F64 && x->minmax_f > m.minmax_f) m.minmax_f = x->minmax_f;
+            if (aggs[a].kind == CSTRIPE_AGG_MIN_F64 && o_f64_cmp_pg(x->minmax_f, m.minmax_f) < 0) m.minmax_f = x->minmax_f;
+            if (aggs[a].kind == CSTRIPE_AGG_MAX_F64 && o_f64_cmp_pg(x->minmax_f, m.minmax_f) > 0) m.minmax_f = x->minmax_f;
             m.cnt += x->cnt;
         }
         oacc_to_partial(&m, aggs[a].kind, &out[a]);

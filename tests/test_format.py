@@ -206,3 +206,79 @@ def test_corrupted_footer_fails_cleanly(tmp_path):
     open(bad3, "wb").write(bytes(b))
     with pytest.raises(ca2.CStripeError):
         ca2.Reader(bad3)
+
+
+def test_tail_segment_not_absorbed(tmp_path):
+    """Round-1 advisor (high): a dense i64 chunk with rows % 32 == 1 used to
+    absorb the 8 B tail into the final 256 B segment (257-271 B), which the
+    fused kernels then read from the wrong lane region. The writer now emits
+    the tiny tail as its own segment; the stream must still round-trip."""
+    for n in (321, 10001, 9985):
+        a = (np.arange(n, dtype=np.int64) * 7) % 5000
+        path = str(tmp_path / f"tail{n}.cs")
+        ca.write_table(path, [("a", ca.I64, 0)], [a], compression=ca.COMP_LZ4)
+        with oracle.OracleTable(path) as t:
+            parts, _ = t.scan_agg([], [(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_STAR, -1)])
+            assert parts[0].i128 == int(a.sum())
+            assert parts[1].count == n
+
+
+def test_n_segs_over_cap_rejected(tmp_path):
+    """Round-1 advisor (medium): a chunk needing > 4096 segments used to be
+    silently truncated to uint16 / misparsed; the writer must now refuse."""
+    n = 10000  # 80 KB i64 chunk at 16 B/segment = 5000 segments > 4096
+    a = RNG.integers(0, 1 << 40, n).astype(np.int64)
+    path = str(tmp_path / "overcap.cs")
+    with pytest.raises(ca.CStripeError, match="4096"):
+        ca.write_table(path, [("a", ca.I64, 0)], [a], compression=ca.COMP_LZ4,
+                       lz4_seg_target_bytes=16)
+
+
+def test_shard_chunk_row_limit_mismatch_rejected(tmp_path):
+    """Round-1 advisor (medium): shard files with a larger chunk_group_row_limit
+    than file 0's had rows silently dropped; the directory open must reject."""
+    d = tmp_path / "shards"
+    d.mkdir()
+    a = np.arange(5000, dtype=np.int64)
+    ca.write_table(str(d / "a.cs"), [("a", ca.I64, 0)], [a],
+                   compression=ca.COMP_LZ4, chunk_group_row_limit=1000)
+    ca.write_table(str(d / "b.cs"), [("a", ca.I64, 0)], [a],
+                   compression=ca.COMP_LZ4, chunk_group_row_limit=2000)
+    r = ca._open(str(d).encode())
+    assert not r, "mismatched chunk_row_limit shard dir must fail to open"
+    assert "chunk_group_row_limit" in ca.errmsg()
+
+
+def test_nan_pruning_not_refuted(tmp_path):
+    """Round-1 advisor (low): PG float ordering treats NaN as greater than
+    every value, so a chunk whose only rows matching `col > c` are NaN must
+    NOT be pruned, and `col > c` must match NaN rows (float8_gt semantics)."""
+    n = 1000
+    a = np.full(n, 5.0)
+    a[::10] = np.nan          # chunk min/max: finite 5.0 .. NaN (PG order)
+    path = str(tmp_path / "nan.cs")
+    ca.write_table(path, [("a", ca.F64, 0)], [a], compression=ca.COMP_LZ4,
+                   chunk_group_row_limit=500)
+    with oracle.OracleTable(path) as t:
+        # col > 7: only NaN rows qualify; chunk must survive and count 100
+        parts, filtered = t.scan_agg([(0, ca.PRED_GT, 7.0)],
+                                     [(ca.AGG_COUNT_STAR, -1)])
+        assert filtered == 0
+        assert parts[0].count == n // 10
+        # col < 7: NaN rows excluded (NaN sorts high)
+        parts, _ = t.scan_agg([(0, ca.PRED_LT, 7.0)], [(ca.AGG_COUNT_STAR, -1)])
+        assert parts[0].count == n - n // 10
+        # MAX over a NaN-containing column is NaN; MIN is the finite value
+        parts, _ = t.scan_agg([], [(ca.AGG_MAX_F64, 0), (ca.AGG_MIN_F64, 0)])
+        assert np.isnan(parts[0].f64)
+        assert parts[1].f64 == 5.0
+
+
+def test_all_nan_min_is_nan(tmp_path):
+    a = np.full(64, np.nan)
+    path = str(tmp_path / "allnan.cs")
+    ca.write_table(path, [("a", ca.F64, 0)], [a], compression=ca.COMP_LZ4)
+    with oracle.OracleTable(path) as t:
+        parts, _ = t.scan_agg([], [(ca.AGG_MIN_F64, 0), (ca.AGG_MAX_F64, 0)])
+        assert np.isnan(parts[0].f64) and np.isnan(parts[1].f64)
+        assert parts[0].count == 64
