@@ -44,7 +44,8 @@ class Options(C.Structure):
                 ("compression", C.c_uint8), ("compression_level", C.c_int8),
                 ("lz4_seg_target_kb", C.c_uint16),
                 ("lz4_seg_target_bytes", C.c_uint32),
-                ("lz4_min_match", C.c_uint8), ("_pad2", C.c_uint8 * 7)]
+                ("lz4_min_match", C.c_uint8), ("canonical", C.c_uint8),
+                ("_pad2", C.c_uint8 * 6)]
 
 
 class Pred(C.Structure):

@@ -50,9 +50,17 @@ struct ColLoc {
     uint64_t val_off;       /* into d_scratch if (flags&1) else d_data */
     uint64_t exists_off;    /* into d_data (packed bitmap, 8B-padded) */
     uint32_t rank_off;      /* u32 word-rank table index into d_rank */
-    uint16_t flags;         /* 1 = values in scratch; 2 = dense (no nulls) */
+    uint16_t flags;         /* 1 = values in scratch; 2 = dense (no nulls);
+                             * 4 = canonical stream (closed-form access) */
     uint8_t  type;          /* cstripe_type */
     uint8_t  width;
+    /* canonical-stream chunks (flags & 4): value bytes are read closed-form
+     * straight from the staged COMPRESSED stream at val_off (format.h) */
+    int64_t  hval;          /* P: shared high bytes, already in place;
+                             * CONST: the value; LIT: literal-run header len */
+    uint8_t  mode;          /* CSF_SEGMODE_* */
+    uint8_t  L;             /* P(L): varying low bytes per value */
+    uint8_t  pad[6];
 };
 
 struct GroupDesc {
@@ -108,6 +116,8 @@ struct cs_gpu_state {
     uint16_t *d_gfkeys = nullptr;    /* grouped: final merged groups */
     AccCell *d_gfcells = nullptr;
     uint32_t *d_gn = nullptr;
+    uint8_t *d_tmp = nullptr;        /* next_batch: canonical-chunk decode buf */
+    uint64_t tmp_bytes = 0;
     int *d_error = nullptr;
     hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr;
 
@@ -471,6 +481,36 @@ __device__ inline bool col_value(const uint8_t *__restrict__ data,
         idx = rank[cl.rank_off + (row >> 6)] + (uint32_t)__popcll(w & (bit - 1));
     }
     const uint8_t *base = (cl.flags & 1) ? scratch + cl.val_off : data + cl.val_off;
+    if (cl.flags & 4) {
+        /* canonical stream: value bytes at a CLOSED-FORM position in the
+         * compressed stream — no LZ4 sequence parse, no scratch (format.h).
+         * One unaligned dwordx2 load per value; consecutive rows sit
+         * (L+3) bytes apart, so a wave's loads coalesce into a few lines. */
+        if (cl.mode == CSF_SEGMODE_LIT) {
+            base += (uint64_t)cl.hval;           /* skip literal-run header */
+        } else {
+            uint64_t raw;
+            if (cl.mode == CSF_SEGMODE_CONST) {
+                raw = (uint64_t)cl.hval;
+            } else {                             /* P(L) */
+                const uint32_t Lx = cl.L;
+                uint32_t pos = idx * (Lx + 3u) + (6u - Lx);
+                pos = idx == 0 ? 1u : (idx == 1 ? 9u : pos);
+                uint64_t lo;
+                __builtin_memcpy(&lo, base + pos, 8);
+                const uint64_t m = (~0ull) >> ((8u - Lx) * 8u);
+                raw = (lo & m) | (uint64_t)cl.hval;
+            }
+            if (cl.type == CSTRIPE_F64) {
+                __builtin_memcpy(&fv, &raw, 8);
+                iv = 0;
+            } else {
+                iv = (int64_t)raw;
+                fv = (double)iv;
+            }
+            return true;
+        }
+    }
     switch (cl.type) {
         case CSTRIPE_I8:  iv = ((const int8_t  *)base)[idx]; fv = (double)iv; break;
         case CSTRIPE_I16: iv = ((const int16_t *)base)[idx]; fv = (double)iv; break;
@@ -1414,6 +1454,37 @@ __global__ __launch_bounds__(AGG_BLOCK) void final_reduce_kernel(
     }
 }
 
+/* closed-form decode of ONE canonical segment to a dense value buffer —
+ * used by the batch/parity path (next_batch) for canonical chunks, which
+ * the agg kernels never materialize. One thread per value; coalesced
+ * stores; the "decode" is just the canonical position formula (format.h). */
+__global__ void canon_decode_kernel(const uint8_t *__restrict__ src,
+                                    uint8_t *__restrict__ dst,
+                                    uint32_t n_values, uint32_t mode,
+                                    uint32_t L, uint32_t width, uint64_t hval)
+{
+    const uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n_values) return;
+    if (mode == CSF_SEGMODE_LIT) {
+        const uint8_t *s = src + hval + (uint64_t)i * width;
+        uint8_t *d = dst + (uint64_t)i * width;
+        for (uint32_t j = 0; j < width; j++) d[j] = s[j];
+        return;
+    }
+    uint64_t v;
+    if (mode == CSF_SEGMODE_CONST) {
+        v = hval;
+    } else {
+        uint32_t pos = i * (L + 3u) + (6u - L);
+        pos = i == 0 ? 1u : (i == 1 ? 9u : pos);
+        uint64_t lo;
+        __builtin_memcpy(&lo, src + pos, 8);
+        const uint64_t m = (~0ull) >> ((8u - L) * 8u);
+        v = (lo & m) | hval;
+    }
+    *(uint64_t *)(dst + (uint64_t)i * 8) = v;
+}
+
 /* =====================================================================
  * staging
  * ===================================================================== */
@@ -1438,6 +1509,7 @@ void csgpu_release(cstripe_scan *s)
     if (g->d_gfkeys) HIP_DROP(hipFree(g->d_gfkeys));
     if (g->d_gfcells) HIP_DROP(hipFree(g->d_gfcells));
     if (g->d_gn) HIP_DROP(hipFree(g->d_gn));
+    if (g->d_tmp) HIP_DROP(hipFree(g->d_tmp));
     if (g->d_tiles) HIP_DROP(hipFree(g->d_tiles));
     if (g->d_tiles2) HIP_DROP(hipFree(g->d_tiles2));
     if (g->d_error) HIP_DROP(hipFree(g->d_error));
@@ -1496,7 +1568,14 @@ int csgpu_stage(cstripe_scan *s, int device_id)
             if (g->proj_of_col[c] < 0) continue;
             const cs_skipnode &nd = st.nodes[c][sc.chunk];
             data_bytes = align_up(data_bytes, 8) + align_up((rows + 7) / 8, 8); /* exists, 8B padded */
-            if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
+            const bool canon = nd.n.comp_type == CSTRIPE_COMP_LZ4 &&
+                               nd.n.n_segs == 1 &&
+                               nd.seg_modes[0] != CSF_SEGMODE_GENERIC;
+            if (canon) {
+                /* canonical stream: values are read closed-form from the
+                 * compressed bytes — no scratch, no decode segments */
+                data_bytes = align_up(data_bytes, 16) + align_up(nd.n.value_len, 16);
+            } else if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
                 data_bytes = align_up(data_bytes, 16) + align_up(nd.n.value_len, 16);
                 scratch_bytes = align_up(scratch_bytes, 16) + align_up(nd.n.decompressed_size, 16);
                 n_segs += nd.n.n_segs;
@@ -1589,13 +1668,39 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 cl.flags |= 2;
             }
 
-            if (!(cl.flags & 2) || cl.type != CSTRIPE_I64 ||
+            const bool canon = nd.n.comp_type == CSTRIPE_COMP_LZ4 &&
+                               nd.n.n_segs == 1 &&
+                               nd.seg_modes[0] != CSF_SEGMODE_GENERIC;
+            if (canon || !(cl.flags & 2) || cl.type != CSTRIPE_I64 ||
                 nd.n.comp_type != CSTRIPE_COMP_LZ4)
                 g->fusable = false;
-            if (!(cl.flags & 2) || (cl.width != 8 && cl.width != 1) ||
+            if (canon || !(cl.flags & 2) || (cl.width != 8 && cl.width != 1) ||
                 nd.n.comp_type != CSTRIPE_COMP_LZ4)
                 g->fusable_mixed = false;
-            if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
+            if (canon) {
+                /* stage the compressed stream; kernels read values straight
+                 * out of it (col_value canonical path) */
+                dpos = align_up(dpos, 16);
+                cl.val_off = dpos;
+                memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
+                dpos += align_up(nd.n.value_len, 16);
+                cl.flags |= 4;
+                cl.mode = nd.seg_modes[0];
+                const uint8_t *strm = stripe_base + nd.n.value_off;
+                if (cl.mode == CSF_SEGMODE_LIT) {
+                    cl.hval = (int64_t)csf_canon_lit_hdr(nd.segs[0].decomp_len);
+                } else {
+                    uint64_t v0;
+                    memcpy(&v0, strm + 1, 8);    /* v0 full, S0's first literal */
+                    if (cl.mode == CSF_SEGMODE_CONST) {
+                        cl.hval = (int64_t)v0;
+                    } else {                      /* P(L) */
+                        cl.L = (uint8_t)(cl.mode & 0xF);
+                        const uint64_t m = (~0ull) >> ((8u - cl.L) * 8u);
+                        cl.hval = (int64_t)(v0 & ~m);
+                    }
+                }
+            } else if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
                 dpos = align_up(dpos, 16);
                 memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
                 spos = align_up(spos, 16);
@@ -2198,7 +2303,26 @@ int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
         uint64_t soff = g->scratch_off[(uint64_t)gi * g->n_proj + pj];
 
         packed.resize(nd.n.decompressed_size);
-        if (soff != ~0ull) {  /* LZ4-decoded on device */
+        const ColLoc &cl = g->colloc_host[(uint64_t)gi * g->n_proj + pj];
+        if (cl.flags & 4) {
+            /* canonical chunk: closed-form device decode into the temp
+             * buffer, then copy back (the agg path never materializes it) */
+            if (g->tmp_bytes < nd.n.decompressed_size) {
+                if (g->d_tmp) HIP_TRY(hipFree(g->d_tmp));
+                g->tmp_bytes = align_up(nd.n.decompressed_size, 4096);
+                HIP_TRY(hipMalloc(&g->d_tmp, g->tmp_bytes));
+            }
+            const uint32_t nv = (uint32_t)(nd.n.decompressed_size / width);
+            const uint32_t grid2 = (nv + 255) / 256;
+            hipLaunchKernelGGL(canon_decode_kernel, dim3(grid2), dim3(256), 0, g->stream,
+                               g->d_data + cl.val_off, g->d_tmp, nv,
+                               (uint32_t)cl.mode, (uint32_t)cl.L, width,
+                               (uint64_t)cl.hval);
+            HIP_TRY(hipGetLastError());
+            HIP_TRY(hipMemcpyAsync(packed.data(), g->d_tmp, nd.n.decompressed_size,
+                                   hipMemcpyDeviceToHost, g->stream));
+            HIP_TRY(hipStreamSynchronize(g->stream));
+        } else if (soff != ~0ull) {  /* LZ4-decoded on device */
             HIP_TRY(hipMemcpyAsync(packed.data(), g->d_scratch + soff, nd.n.decompressed_size,
                                    hipMemcpyDeviceToHost, g->stream));
             HIP_TRY(hipStreamSynchronize(g->stream));

@@ -64,6 +64,7 @@ extern "C" void cstripe_default_options(cstripe_options *o)
     o->lz4_seg_target_kb = 0;
     o->lz4_seg_target_bytes = 256;
     o->lz4_min_match = 4;              /* standard greedy LZ4 parse */
+    o->canonical = 1;                  /* closed-form canonical parses on */
 }
 
 static uint32_t type_width(uint8_t t) { return csf_type_width(t); }
@@ -223,6 +224,17 @@ extern "C" cstripe_writer *cstripe_write_begin(const char *path, const cstripe_c
     if (opts) w->opts = *opts; else cstripe_default_options(&w->opts);
     if (w->opts.lz4_seg_target_bytes == 0 && w->opts.lz4_seg_target_kb == 0)
         w->opts.lz4_seg_target_bytes = 256;
+    /* segment decomp_len carries the stream-shape mode in bits 24-31, so a
+     * single-segment chunk's raw stream must fit 24 bits (default 10000-row
+     * chunks are 80 KB; 1M rows x 8 B = 8 MB still fits with margin) */
+    if ((uint64_t)w->opts.chunk_group_row_limit * 8 >= (1u << 24) ||
+        w->opts.chunk_group_row_limit == 0) {
+        cs_set_err("chunk_group_row_limit %u out of range (1 .. 2097151)",
+                   w->opts.chunk_group_row_limit);
+        fclose(f);
+        delete w;
+        return nullptr;
+    }
     for (uint32_t i = 0; i < n_cols; i++) {
         csf_coldef d{};
         memcpy(d.name, cols[i].name, sizeof(d.name));
@@ -297,6 +309,53 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
         return true;
     }
 
+    /* ---- canonical parses (closed-form GPU access; see format.h) ----
+     * The emitted stream is a standard LZ4 block either way; canonical just
+     * picks a parse whose value positions are computable, killing the GPU's
+     * per-sequence parse chain (round-1's instruction wall). */
+    if (codec == CSTRIPE_COMP_LZ4 && w->opts.canonical && width == 8 &&
+        raw.size() >= 24 && raw.size() < (1u << 24)) {
+        const size_t nv = raw.size() / 8;
+        uint64_t v0, x = 0;
+        memcpy(&v0, raw.data(), 8);
+        const uint8_t *rp = raw.data();
+        for (size_t i = 1; i < nv; i++) {
+            uint64_t vi;
+            memcpy(&vi, rp + i * 8, 8);
+            x |= vi ^ v0;
+        }
+        int mode = -1, csz = 0;
+        size_t cap = raw.size() + raw.size() / 8 + 64;
+        out.value_comp.resize(cap);
+        if (x == 0) {
+            csz = lz4e_canon_const(rp, (int)nv, out.value_comp.data(), (int)cap);
+            if (csz > 0) mode = CSF_SEGMODE_CONST;
+        } else {
+            int L = (63 - __builtin_clzll(x)) / 8 + 1;
+            if (L <= 4) {
+                csz = lz4e_canon_p(rp, (int)nv, L, out.value_comp.data(), (int)cap);
+                if (csz > 0) mode = (int)(CSF_SEGMODE_P_BASE | (uint32_t)L);
+            }
+            /* L > 4: high bytes vary — no canonical parse; the greedy path
+             * below runs, and its incompressible->NONE fallback (reference
+             * CompressBuffer semantics) already yields closed-form access */
+        }
+        if (mode >= 0 && (size_t)csz >= raw.size())
+            mode = -1;   /* did not shrink -> raw NONE below (reference rule) */
+        if (mode >= 0) {
+            out.value_comp.resize((size_t)csz);
+            csf_seg s{0, (uint32_t)csz, 0,
+                      (uint32_t)raw.size() | ((uint32_t)mode << 24)};
+            out.segs.push_back(s);
+            out.node.comp_type = CSTRIPE_COMP_LZ4;
+            out.node.comp_level = 0;
+            out.node.n_segs = 1;
+            out.node.value_len = (uint64_t)csz;
+            return true;
+        }
+        out.value_comp.clear();
+    }
+
     /* segment boundaries 16 B-aligned (pure byte split of the raw stream;
      * lets the GPU flush decoded segments with aligned 16 B stores) */
     (void)width;
@@ -356,6 +415,11 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
         off += len;
     }
 
+    /* incompressible -> keep raw, like the reference (CompressBuffer returns
+     * false when compression does not shrink, columnar_compression.c:88-94;
+     * raw NONE chunks also get closed-form GPU access for free) */
+    if (ok && out.value_comp.size() >= raw.size())
+        ok = false;
     if (!ok) {  /* CompressBuffer returned false -> keep uncompressed */
         out.value_comp = raw;
         out.segs.clear();
@@ -664,6 +728,13 @@ static int footer_read_one(cstripe_reader *r, uint32_t file_idx)
                 if (!need(nd.segs.size() * sizeof(csf_seg))) { cs_set_err("truncated segs"); return CSTRIPE_ERR_FORMAT; }
                 memcpy(nd.segs.data(), p, nd.segs.size() * sizeof(csf_seg));
                 p += nd.segs.size() * sizeof(csf_seg);
+                /* split the packed decomp_len: 24-bit length + stream-shape
+                 * mode (format.h); v1 files carry mode 0 = generic */
+                nd.seg_modes.resize(nd.segs.size());
+                for (size_t si2 = 0; si2 < nd.segs.size(); si2++) {
+                    nd.seg_modes[si2] = csf_seg_mode(&nd.segs[si2]);
+                    nd.segs[si2].decomp_len &= CSF_SEG_DLEN_MASK;
+                }
             }
         }
     }

@@ -92,15 +92,57 @@ typedef struct csf_skipnode {
     uint16_t reserved2;
 } csf_skipnode;
 
-/* one independently decodable compressed segment of a chunk's value stream */
+/* one independently decodable compressed segment of a chunk's value stream.
+ *
+ * decomp_len bits 24-31 carry the segment's STREAM-SHAPE MODE, a redundant
+ * hint the writer records about the standard-LZ4 stream it emitted (the
+ * stream itself is always decodable by LZ4_decompress_safe, mode or not;
+ * round-1 v1 files have 0 there since segments were < 16 MiB):
+ *   0     generic         — arbitrary LZ4 block (greedy parse); sequence
+ *                           parse required
+ *   0x10|L canonical P(L) — width-8 values whose high (8-L) bytes are all
+ *                           identical across the segment, emitted as the
+ *                           periodic parse [lit v0+v1.low | off-8 match] then
+ *                           [lit v(i).low | off-8 match]* then [lit v(n-1)];
+ *                           every value's bytes sit at a CLOSED-FORM stream
+ *                           position, so a GPU lane reads values straight out
+ *                           of the compressed stream with no parsing
+ *   0x20  constant        — all values equal v0: [lit v0 | off-8 match 8(n-2)]
+ *                           [lit v(n-1)]; decode = broadcast v0
+ *   0x30  literal         — one literal run; value i at hdr + i*width
+ * Modes P/constant require 1 <= L <= 4 (match >= 4 bytes) and n >= 3 values. */
 typedef struct csf_seg {
     uint32_t comp_off;             /* relative to chunk's value_off */
     uint32_t comp_len;
     uint32_t decomp_off;           /* relative to chunk's decompressed stream */
-    uint32_t decomp_len;
+    uint32_t decomp_len;           /* bits 0-23 length; bits 24-31 mode */
 } csf_seg;
 
 #pragma pack(pop)
+
+#define CSF_SEGMODE_GENERIC 0x00u
+#define CSF_SEGMODE_P_BASE  0x10u  /* 0x10|L, L in 1..4 */
+#define CSF_SEGMODE_CONST   0x20u
+#define CSF_SEGMODE_LIT     0x30u
+#define CSF_SEG_DLEN_MASK   0x00FFFFFFu
+
+static inline uint32_t csf_seg_dlen(const csf_seg *s) { return s->decomp_len & CSF_SEG_DLEN_MASK; }
+static inline uint8_t  csf_seg_mode(const csf_seg *s) { return (uint8_t)(s->decomp_len >> 24); }
+
+/* stream byte offset of value j's low-L bytes inside a canonical P(L)
+ * segment (see layout above): j=0 -> 1, j=1 -> 9, else j*(L+3) + (6-L) */
+static inline uint32_t csf_canon_p_pos(uint32_t j, uint32_t L)
+{
+    if (j == 0) return 1;
+    if (j == 1) return 9;
+    return j * (L + 3) + (6 - L);
+}
+
+/* literal-run header size for a mode-LIT segment of len decompressed bytes */
+static inline uint32_t csf_canon_lit_hdr(uint32_t len)
+{
+    return len < 15 ? 1 : 2 + (len - 15) / 255;
+}
 
 /* footer layout, starting at footer_offset:
  *   csf_footer_head

@@ -17,7 +17,8 @@ void cs_set_err(const char *fmt, ...);
 
 struct cs_skipnode {
     csf_skipnode n;
-    std::vector<csf_seg> segs;
+    std::vector<csf_seg> segs;       /* decomp_len MASKED at parse (24-bit) */
+    std::vector<uint8_t> seg_modes;  /* CSF_SEGMODE_* per segment */
 };
 
 struct cs_stripe_info {

@@ -138,4 +138,92 @@ static inline int lz4e_compress(lz4e_state *st, const uint8_t *src, int slen,
     return lz4e_compress_mm(st, src, slen, dst, dcap, 4);
 }
 
+/* =====================================================================
+ * Canonical emitters — alternative PARSES of the same input that are still
+ * standard LZ4 blocks (decodable by LZ4_decompress_safe, pinned in
+ * tests/test_format.py) but whose byte layout is closed-form, so a GPU
+ * lane can read any value straight out of the compressed stream without
+ * walking the sequence chain. See format.h CSF_SEGMODE_* for the layout.
+ * ===================================================================== */
+
+/* canonical P(L): n width-8 values, all sharing their high (8-L) bytes
+ * (1 <= L <= 4, n >= 3). Emits:
+ *   S0:       [token lit=8+L match=8-L] v0(8B) v1.low(L) [off=8]
+ *   S1..Sn-3: [token lit=L   match=8-L] v(i+1).low(L)    [off=8]
+ *   final:    [token lit=8] v(n-1)(8B)
+ * (last match starts 16-L >= 12 bytes before end; trailing 8 literals
+ * satisfy the 5-literal end rule — LZ4 block format constraints) */
+static inline int lz4e_canon_p(const uint8_t *src, int n, int L,
+                               uint8_t *dst, int dcap)
+{
+    if (n < 3 || L < 1 || L > 4) return 0;
+    const int need = (11 + L) + (n - 3) * (L + 3) + 9;
+    if (need > dcap) return 0;
+    uint8_t *op = dst;
+    const uint8_t mtok = (uint8_t)(8 - L - 4);
+    *op++ = (uint8_t)(((8 + L) << 4) | mtok);
+    memcpy(op, src, 8 + (size_t)L);          /* v0 full + v1 low */
+    op += 8 + L;
+    *op++ = 8; *op++ = 0;                    /* offset 8 */
+    for (int i = 1; i <= n - 3; i++) {
+        *op++ = (uint8_t)((L << 4) | mtok);
+        memcpy(op, src + (size_t)(i + 1) * 8, (size_t)L);
+        op += L;
+        *op++ = 8; *op++ = 0;
+    }
+    *op++ = (uint8_t)(8 << 4);               /* final: 8 literals */
+    memcpy(op, src + (size_t)(n - 1) * 8, 8);
+    op += 8;
+    return (int)(op - dst);
+}
+
+/* canonical constant: n equal width-8 values (n >= 3):
+ *   S0: [token lit=8 match=ext] v0(8B) [off=8] [ext...]   (match 8(n-2) B)
+ *   final: [token lit=8] v0(8B) */
+static inline int lz4e_canon_const(const uint8_t *src, int n,
+                                   uint8_t *dst, int dcap)
+{
+    if (n < 3) return 0;
+    int mlen = 8 * (n - 2);                  /* bytes the match reproduces */
+    int mex = mlen - 4;
+    uint8_t *op = dst;
+    uint8_t *oend = dst + dcap;
+    if (op + 11 > oend) return 0;
+    *op++ = (uint8_t)((8 << 4) | (mex >= 15 ? 15 : mex));
+    memcpy(op, src, 8);
+    op += 8;
+    *op++ = 8; *op++ = 0;
+    if (mex >= 15) {
+        mex -= 15;
+        while (mex >= 255) { if (op >= oend) return 0; *op++ = 255; mex -= 255; }
+        if (op >= oend) return 0;
+        *op++ = (uint8_t)mex;
+    }
+    if (op + 9 > oend) return 0;
+    *op++ = (uint8_t)(8 << 4);
+    memcpy(op, src, 8);
+    op += 8;
+    return (int)(op - dst);
+}
+
+/* canonical literal run: the whole stream as one literal sequence
+ * (value i at csf_canon_lit_hdr(slen) + i*width) */
+static inline int lz4e_canon_lit(const uint8_t *src, int slen,
+                                 uint8_t *dst, int dcap)
+{
+    uint8_t *op = dst;
+    int hdr = slen < 15 ? 1 : 2 + (slen - 15) / 255;
+    if (hdr + slen > dcap) return 0;
+    if (slen < 15) {
+        *op++ = (uint8_t)(slen << 4);
+    } else {
+        *op++ = (uint8_t)(15 << 4);
+        int l = slen - 15;
+        while (l >= 255) { *op++ = 255; l -= 255; }
+        *op++ = (uint8_t)l;
+    }
+    memcpy(op, src, (size_t)slen);
+    return hdr + slen;
+}
+
 #endif

@@ -119,7 +119,17 @@ typedef struct cstripe_options {
                                         * Larger = literal-heavier standard-LZ4
                                         * streams that decode faster on GPU at
                                         * a small compression-ratio cost */
-    uint8_t     _pad2[7];
+    uint8_t     canonical;             /* 1 (default): width-8 chunks whose
+                                        * values share their high bytes are
+                                        * written as a CANONICAL periodic LZ4
+                                        * parse — still a standard block that
+                                        * LZ4_decompress_safe decodes — whose
+                                        * value positions are closed-form, so
+                                        * GPU kernels read values straight
+                                        * from the compressed stream with no
+                                        * sequence parsing. 0: greedy parse
+                                        * everywhere (round-1 behaviour). */
+    uint8_t     _pad2[6];
 } cstripe_options;
 
 void cstripe_default_options(cstripe_options *opts);

@@ -124,6 +124,11 @@ static int o_parse_footer(oracle_table *t)
                 nd->segs = calloc(nd->n.n_segs, sizeof(csf_seg));
                 memcpy(nd->segs, p, (size_t)nd->n.n_segs * sizeof(csf_seg));
                 p += (size_t)nd->n.n_segs * sizeof(csf_seg);
+                /* decomp_len bits 24-31 are a stream-shape hint for the GPU
+                 * (format.h); the oracle decodes every mode through the same
+                 * generic LZ4/zstd call, so just mask the length */
+                for (uint16_t si = 0; si < nd->n.n_segs; si++)
+                    nd->segs[si].decomp_len &= CSF_SEG_DLEN_MASK;
             }
         }
     }

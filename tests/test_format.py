@@ -227,11 +227,11 @@ def test_n_segs_over_cap_rejected(tmp_path):
     """Round-1 advisor (medium): a chunk needing > 4096 segments used to be
     silently truncated to uint16 / misparsed; the writer must now refuse."""
     n = 10000  # 80 KB i64 chunk at 16 B/segment = 5000 segments > 4096
-    a = RNG.integers(0, 1 << 40, n).astype(np.int64)
+    a = RNG.integers(0, 127, n).astype(np.int64)   # compressible (stays LZ4)
     path = str(tmp_path / "overcap.cs")
     with pytest.raises(ca.CStripeError, match="4096"):
         ca.write_table(path, [("a", ca.I64, 0)], [a], compression=ca.COMP_LZ4,
-                       lz4_seg_target_bytes=16)
+                       lz4_seg_target_bytes=16, canonical=0)
 
 
 def test_shard_chunk_row_limit_mismatch_rejected(tmp_path):
@@ -282,3 +282,134 @@ def test_all_nan_min_is_nan(tmp_path):
         parts, _ = t.scan_agg([], [(ca.AGG_MIN_F64, 0), (ca.AGG_MAX_F64, 0)])
         assert np.isnan(parts[0].f64) and np.isnan(parts[1].f64)
         assert parts[0].count == 64
+
+
+# ---------- canonical stream-shape modes (round 2; format.h) ----------
+
+import futil
+
+
+def _liblz4():
+    lz4 = C.CDLL("liblz4.so.1")
+    lz4.LZ4_decompress_safe.restype = C.c_int
+    lz4.LZ4_decompress_safe.argtypes = [C.c_char_p, C.c_char_p, C.c_int, C.c_int]
+    return lz4
+
+
+def _decode_sys(lz4, comp, dlen):
+    out = C.create_string_buffer(dlen)
+    r = lz4.LZ4_decompress_safe(bytes(comp), out, len(comp), dlen)
+    assert r == dlen, f"system LZ4_decompress_safe returned {r}, want {dlen}"
+    return out.raw
+
+
+@pytest.mark.parametrize("case", [
+    ("L1", lambda n: np.arange(n, dtype=np.int64) % 200),
+    ("L2", lambda n: 1000 + (np.arange(n, dtype=np.int64) * 37) % 50000),
+    ("L3", lambda n: (np.arange(n, dtype=np.int64) * 104729) % (1 << 23)),
+    ("L4", lambda n: (np.arange(n, dtype=np.int64) * 2654435761) % (1 << 31)),
+    ("neg", lambda n: -((np.arange(n, dtype=np.int64) * 13) % 100) - 1),
+    ("const", lambda n: np.full(n, -123456789, dtype=np.int64)),
+    # f64 whose low mantissa bytes vary under a fixed exponent: the bit
+    # patterns share their high bytes, exactly the canonical-P shape
+    ("f64", lambda n: (np.uint64(0x3FF0000000000000)
+                       | (np.arange(n, dtype=np.uint64) % 251)).view(np.float64)),
+])
+@pytest.mark.parametrize("n", [3, 4, 37, 9999, 10000])
+def test_canonical_streams_system_decodable(tmp_path, case, n):
+    """The canonical parses (P(L)/const) are STANDARD LZ4 blocks: the system
+    LZ4_decompress_safe — the exact call the reference's DecompressBuffer
+    makes (columnar_compression.c:183) — must reproduce the raw value stream
+    byte-for-byte, and the footer must carry the expected mode tag."""
+    name, gen = case
+    a = np.ascontiguousarray(gen(n))
+    typ = ca.F64 if a.dtype == np.float64 else ca.I64
+    path = str(tmp_path / f"c_{name}_{n}.cs")
+    ca.write_table(path, [("a", typ, 0)], [a], compression=ca.COMP_LZ4)
+    foot = futil.read_footer(path)
+    node = foot["stripes"][0]["nodes"][0][0]
+    if n <= 4:
+        # tiny chunks may store raw (no shrink; reference CompressBuffer
+        # rule) or pick the greedy parse; only content parity matters
+        with oracle.OracleTable(path) as t:
+            kind = ca.AGG_SUM_F64 if typ == ca.F64 else ca.AGG_SUM_I64
+            parts, _ = t.scan_agg([], [(kind, 0)])
+            if typ == ca.F64:
+                assert abs(parts[0].f64 - float(a.sum())) < 1e-9
+            else:
+                assert parts[0].i128 == int(a.sum())
+        return
+    assert node["comp_type"] == ca.COMP_LZ4
+    assert node["n_segs"] == 1
+    seg = node["segs"][0]
+    if name == "const":
+        assert seg["mode"] == futil.SEGMODE_CONST
+    else:
+        assert futil.SEGMODE_P_BASE < seg["mode"] <= futil.SEGMODE_P_BASE + 4, \
+            f"expected P mode, got {seg['mode']:#x}"
+    assert seg["decomp_len"] == node["decompressed_size"]
+    comp = futil.chunk_stream(path, node)
+    raw = _decode_sys(_liblz4(), comp, seg["decomp_len"])
+    assert raw == a.tobytes()
+    # compression must actually compress for these shapes
+    assert len(comp) < len(raw)
+
+
+def test_canonical_mode_choice(tmp_path):
+    """Writer mode decisions: high-entropy 8B values (L>4) fall back to the
+    greedy parse (or raw when incompressible, reference CompressBuffer
+    semantics); sub-8B widths stay greedy; canonical=0 disables tagging."""
+    n = 10000
+    rnd = RNG.integers(-2**62, 2**62, n).astype(np.int64)      # incompressible
+    lowcard = RNG.integers(0, 10, n).astype(np.int64)          # canonical P(1)
+    flags = RNG.integers(0, 3, n).astype(np.int8)              # i8 -> greedy
+    path = str(tmp_path / "mix.cs")
+    ca.write_table(path, [("r", ca.I64, 0), ("l", ca.I64, 0), ("f", ca.I8, 0)],
+                   [rnd, lowcard, flags], compression=ca.COMP_LZ4)
+    foot = futil.read_footer(path)
+    nodes = foot["stripes"][0]["nodes"]
+    assert nodes[0][0]["comp_type"] == ca.COMP_NONE             # incompressible
+    assert nodes[1][0]["segs"][0]["mode"] == futil.SEGMODE_P_BASE + 1
+    assert nodes[2][0]["comp_type"] == ca.COMP_LZ4
+    assert all(s["mode"] == futil.SEGMODE_GENERIC for s in nodes[2][0]["segs"])
+
+    path2 = str(tmp_path / "off.cs")
+    ca.write_table(path2, [("l", ca.I64, 0)], [lowcard],
+                   compression=ca.COMP_LZ4, canonical=0)
+    foot2 = futil.read_footer(path2)
+    assert all(s["mode"] == futil.SEGMODE_GENERIC
+               for s in foot2["stripes"][0]["nodes"][0][0]["segs"])
+
+
+def test_canonical_sparse_roundtrip(tmp_path):
+    """Canonical addressing is by VALUE index, so NULL-bearing chunks work:
+    present values round-trip, null slots read as null (oracle reader)."""
+    n = 10000
+    a = (np.arange(n, dtype=np.int64) * 7) % 3000
+    nulls = (np.arange(n) % 5 == 0).astype(np.uint8)
+    path = str(tmp_path / "sp.cs")
+    ca.write_table(path, [("a", ca.I64, 0)], [a], nulls=[nulls],
+                   compression=ca.COMP_LZ4)
+    foot = futil.read_footer(path)
+    assert foot["stripes"][0]["nodes"][0][0]["segs"][0]["mode"] > 0
+    with oracle.OracleTable(path) as t:
+        parts, _ = t.scan_agg([], [(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_COL, 0)])
+        mask = nulls == 0
+        assert parts[0].i128 == int(a[mask].sum())
+        assert parts[1].count == int(mask.sum())
+
+
+def test_canonical_orderkey_boundary(tmp_path):
+    """A sorted key crossing a 2^16 boundary inside one chunk needs L=3 —
+    the per-chunk XOR-based L choice must cover it exactly."""
+    a = np.arange(60000, 70001, dtype=np.int64)    # crosses 65536
+    path = str(tmp_path / "ok.cs")
+    ca.write_table(path, [("a", ca.I64, 0)], [a], compression=ca.COMP_LZ4)
+    foot = futil.read_footer(path)
+    first = foot["stripes"][0]["nodes"][0][0]
+    assert first["segs"][0]["mode"] == futil.SEGMODE_P_BASE + 3  # crosses 2^16
+    with oracle.OracleTable(path) as t:
+        parts, _ = t.scan_agg([(0, ca.PRED_GE, 65530)], [(ca.AGG_COUNT_STAR, -1),
+                                                         (ca.AGG_SUM_I64, 0)])
+        assert parts[0].count == 70001 - 65530
+        assert parts[1].i128 == int(np.arange(65530, 70001, dtype=np.int64).sum())
