@@ -83,6 +83,9 @@ def main():
                     help="LZ4 micro-segment size override (bytes)")
     ap.add_argument("--min-match", type=int, default=0,
                     help="writer LZ4 min match length (>=4; GPU-decode knob)")
+    ap.add_argument("--canonical", type=int, default=1,
+                    help="writer canonical stream modes (closed-form GPU "
+                         "access); 0 = greedy parse (round-1 behaviour)")
     ap.add_argument("--shards", type=int, default=1,
                     help="shard files per GPU scanned as one table "
                          "(config 3: 4 shards/GPU x 8 GPUs = 32 shards)")
@@ -111,7 +114,7 @@ def main():
     os.makedirs(cache, exist_ok=True)
     if args.shards > 1:
         shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}"
-                                    f"_m{args.min_match}_n{args.shards}_r{rank}")
+                                    f"_m{args.min_match}_c{args.canonical}_n{args.shards}_r{rank}")
         os.makedirs(shard, exist_ok=True)
         t0 = time.time()
         per = args.rows // args.shards
@@ -120,14 +123,16 @@ def main():
             if not os.path.exists(p):
                 ca.gen_lineitem(p, per + (args.rows % args.shards if i == 0 else 0),
                                 seed=42 + rank * args.shards + i, compression=comp,
-                                seg_bytes=args.seg_bytes, min_match=args.min_match)
+                                seg_bytes=args.seg_bytes, min_match=args.min_match,
+                                canonical=args.canonical)
     else:
         shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}"
-                                    f"_m{args.min_match}_r{rank}.cs")
+                                    f"_m{args.min_match}_c{args.canonical}_r{rank}.cs")
         t0 = time.time()
         if not os.path.exists(shard):
             ca.gen_lineitem(shard, args.rows, seed=42 + rank, compression=comp,
-                            seg_bytes=args.seg_bytes, min_match=args.min_match)
+                            seg_bytes=args.seg_bytes, min_match=args.min_match,
+                            canonical=args.canonical)
     gen_s = time.time() - t0
 
     reader = ca.Reader(shard)

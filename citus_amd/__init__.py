@@ -127,9 +127,9 @@ _gpu_available = _sig("cstripe_gpu_available", C.c_int, [])
 _gen_lineitem = _sig("csbench_gen_lineitem", C.c_int,
                      [C.c_char_p, C.c_uint64, C.c_uint64, C.c_int, C.c_int, C.c_int,
                       C.c_uint64, C.c_uint32])
-_gen_lineitem_mm = _sig("csbench_gen_lineitem_mm", C.c_int,
-                        [C.c_char_p, C.c_uint64, C.c_uint64, C.c_int, C.c_int, C.c_int,
-                         C.c_uint64, C.c_uint32, C.c_int])
+_gen_lineitem2 = _sig("csbench_gen_lineitem2", C.c_int,
+                      [C.c_char_p, C.c_uint64, C.c_uint64, C.c_int, C.c_int, C.c_int,
+                       C.c_uint64, C.c_uint32, C.c_int, C.c_int])
 
 
 def errmsg():
@@ -194,11 +194,13 @@ def write_table(path, defs, columns, nulls=None, **opt_kw):
 
 
 def gen_lineitem(path, n_rows, seed=42, compression=COMP_LZ4, level=3, seg_kb=0,
-                 stripe_rows=0, chunk_rows=0, seg_bytes=0, min_match=0):
+                 stripe_rows=0, chunk_rows=0, seg_bytes=0, min_match=0,
+                 canonical=1):
     if seg_bytes:
         seg_kb = -int(seg_bytes)      # negative seg_kb = bytes (csbench_gen_lineitem)
-    _check(_gen_lineitem_mm(path.encode(), n_rows, seed, compression, level, seg_kb,
-                            stripe_rows, chunk_rows, min_match), "gen_lineitem")
+    _check(_gen_lineitem2(path.encode(), n_rows, seed, compression, level, seg_kb,
+                          stripe_rows, chunk_rows, min_match, canonical),
+           "gen_lineitem")
 
 
 LINEITEM_COLS = {"l_orderkey": 0, "l_quantity": 1, "l_extendedprice": 2,

@@ -24,6 +24,10 @@ extern "C" int csbench_gen_lineitem_mm(const char *path, uint64_t n_rows, uint64
                                        int compression, int level, int seg_kb,
                                        uint64_t stripe_rows, uint32_t chunk_rows,
                                        int min_match);
+extern "C" int csbench_gen_lineitem2(const char *path, uint64_t n_rows, uint64_t seed,
+                                     int compression, int level, int seg_kb,
+                                     uint64_t stripe_rows, uint32_t chunk_rows,
+                                     int min_match, int canonical);
 
 static inline uint64_t xs64(uint64_t &x)
 {
@@ -46,6 +50,15 @@ extern "C" int csbench_gen_lineitem_mm(const char *path, uint64_t n_rows, uint64
                                        uint64_t stripe_rows, uint32_t chunk_rows,
                                        int min_match)
 {
+    return csbench_gen_lineitem2(path, n_rows, seed, compression, level, seg_kb,
+                                 stripe_rows, chunk_rows, min_match, 1);
+}
+
+extern "C" int csbench_gen_lineitem2(const char *path, uint64_t n_rows, uint64_t seed,
+                                     int compression, int level, int seg_kb,
+                                     uint64_t stripe_rows, uint32_t chunk_rows,
+                                     int min_match, int canonical)
+{
     cstripe_coldef cols[8] = {};
     const char *names[8] = {"l_orderkey", "l_quantity", "l_extendedprice", "l_discount",
                             "l_tax", "l_shipdate", "l_returnflag", "l_linestatus"};
@@ -63,6 +76,7 @@ extern "C" int csbench_gen_lineitem_mm(const char *path, uint64_t n_rows, uint64
     if (seg_kb > 0) { opts.lz4_seg_target_kb = (uint16_t)seg_kb; opts.lz4_seg_target_bytes = 0; }
     else if (seg_kb < 0) { opts.lz4_seg_target_kb = 0; opts.lz4_seg_target_bytes = (uint32_t)(-seg_kb); }
     if (min_match >= 4) opts.lz4_min_match = (uint8_t)min_match;
+    opts.canonical = canonical ? 1 : 0;
 
     cstripe_writer *w = cstripe_write_begin(path, cols, 8, &opts);
     if (!w) return CSTRIPE_ERR;

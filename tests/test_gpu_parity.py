@@ -210,14 +210,17 @@ def test_q6_large_synthetic(tmp_path):
     assert_parity(op, gp, aggs_q6)
 
 
-@pytest.mark.parametrize("seg", [("bytes", 0), ("kb", 8), ("kb", 1024)])
+@pytest.mark.parametrize("seg", [("bytes", 0), ("kb", 8), ("kb", 1024), ("canon", 0)])
 def test_decode_kernel_variants(tmp_path, seg):
-    """All three decode kernels give identical, oracle-exact results:
-    lane-parallel (default 256B micro-segments), wave-cooperative LDS
-    (8KB segments), global fallback (single whole-chunk block)."""
+    """All decode paths give identical, oracle-exact results: lane-parallel
+    (greedy 256B micro-segments), wave-cooperative LDS (8KB segments), global
+    fallback (single whole-chunk block), and the canonical closed-form path
+    (default writer)."""
     kind, val = seg
-    path = str(tmp_path / f"v{val}.cs")
+    path = str(tmp_path / f"v{kind}{val}.cs")
     kw = {"seg_kb": val} if kind == "kb" else {}
+    if kind != "canon":
+        kw["canonical"] = 0          # force the greedy parse + decode kernels
     ca.gen_lineitem(path, 300_000, **kw)
     preds = [(5, ca.PRED_GE, 8766), (5, ca.PRED_LT, 9131),
              (3, ca.PRED_GE, 5), (3, ca.PRED_LE, 7), (1, ca.PRED_LT, 2400)]
@@ -425,6 +428,7 @@ def test_fuzz_differential():
         chunk = int(rng.integers(1, 11)) * 1000
         stripe = [10000, 50000, 150000][int(rng.integers(0, 3))]
         comp = [ca.COMP_NONE, ca.COMP_LZ4, ca.COMP_LZ4, ca.COMP_ZSTD][int(rng.integers(0, 4))]
+        canon = int(rng.integers(0, 2))   # both writer parses fuzzed
         cols, nulls, defs = [], [], []
         for i, (t, dt) in enumerate(kinds):
             if dt in (np.float32, np.float64):
@@ -439,7 +443,8 @@ def test_fuzz_differential():
         with tempfile.TemporaryDirectory() as td:
             path = os.path.join(td, "f.cs")
             ca.write_table(path, defs, cols, nulls=nulls, compression=comp,
-                           stripe_row_limit=stripe, chunk_group_row_limit=chunk)
+                           stripe_row_limit=stripe, chunk_group_row_limit=chunk,
+                           canonical=canon)
             preds = []
             for _ in range(int(rng.integers(0, 4))):
                 ci = int(rng.integers(0, ncols))
@@ -509,7 +514,8 @@ def test_fuzz_grouped_differential():
         with tempfile.TemporaryDirectory() as td:
             path = os.path.join(td, "g.cs")
             ca.write_table(path, defs, cols, nulls=nulls, compression=comp,
-                           chunk_group_row_limit=int(rng.integers(1, 11)) * 1000)
+                           chunk_group_row_limit=int(rng.integers(1, 11)) * 1000,
+                           canonical=int(rng.integers(0, 2)))
             preds = []
             if rng.random() < 0.6:
                 preds.append((nk, int(rng.integers(0, 4)), int(rng.integers(-10**5, 10**5))))
