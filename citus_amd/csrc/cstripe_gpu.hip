@@ -709,7 +709,8 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     uint32_t row_start = tile * TILE_ROWS;
     uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
 
-    ThreadAcc acc[MAX_AGGS];
+    ThreadAcc acc[NAGGS >= 0 ? NAGGS : MAX_AGGS];
+    #pragma unroll
     for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
 
     for (uint32_t row = row_start + threadIdx.x; row < row_end; row += AGG_BLOCK) {
@@ -718,6 +719,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
         int64_t liv = 0; double lfv = 0; bool lok = false;
         /* no short-circuit: loads stay control-independent so they issue
          * back-to-back and pipeline instead of chaining load->wait->branch */
+        #pragma unroll
         for (uint32_t p = 0; p < n_preds; p++) {
             const PredD &pr = params.preds[p];
             if ((int)pr.proj != last_proj) {   /* BETWEEN reuses the load */
@@ -727,6 +729,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
             pass = pass & (lok && pred_eval(pr, liv, lfv));
         }
         if (!pass) continue;
+        #pragma unroll
         for (uint32_t a = 0; a < n_aggs; a++)
             acc_row(acc[a], params.aggs[a], data, scratch, rank, cols, row);
     }
@@ -735,6 +738,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     __shared__ ThreadAcc lds[AGG_BLOCK / WAVE][MAX_AGGS];
     const uint32_t wid = threadIdx.x / WAVE;
     const uint32_t lane = threadIdx.x % WAVE;
+    #pragma unroll
     for (uint32_t a = 0; a < n_aggs; a++) {
         wave_reduce(acc[a], params.aggs[a].kind);
         if (lane == 0) lds[wid][a] = acc[a];
@@ -795,6 +799,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
 
     /* phase 2: filter + aggregate from LDS */
     ThreadAcc acc[NAGGS >= 0 ? NAGGS : MAX_AGGS];
+    #pragma unroll
     for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
 
     auto val = [&](uint32_t proj, uint32_t row) -> int64_t {
@@ -812,6 +817,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
             pass = pass & pred_eval(pr, liv, 0.0);   /* cols are i64: is_float==0 */
         }
         if (!pass) continue;
+        #pragma unroll
         for (uint32_t a = 0; a < n_aggs; a++) {
             const AggD &g = params.aggs[a];
             ThreadAcc &A = acc[a];
@@ -844,6 +850,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
     __shared__ ThreadAcc lred[AGG_BLOCK / WAVE][MAX_AGGS];
     const uint32_t wid = tid / WAVE;
     const uint32_t lane = tid % WAVE;
+    #pragma unroll
     for (uint32_t a = 0; a < n_aggs; a++) {
         wave_reduce(acc[a], params.aggs[a].kind);
         if (lane == 0) lred[wid][a] = acc[a];
