@@ -129,6 +129,7 @@ struct cs_gpu_state {
     bool segs_16aligned = true;
     bool fusable = true;             /* all proj cols dense i64, uniform 256B lz4 segs */
     bool fusable_mixed = true;       /* widths in {1,8}; enables fused grouped */
+    bool all_dense = true;           /* no NULLs anywhere -> pair_agg_kernel */
     struct FusedTile *d_tiles = nullptr;
     uint32_t n_tiles = 0;
     struct FusedTileG *d_tiles2 = nullptr;
@@ -609,48 +610,66 @@ __device__ inline void acc_row(ThreadAcc &a, const AggD &g,
                                const uint8_t *__restrict__ data,
                                const uint8_t *__restrict__ scratch,
                                const uint32_t *__restrict__ rank,
-                               const ColLoc *__restrict__ cols, uint32_t row)
+                               const ColLoc *__restrict__ cols, uint32_t row,
+                               bool pass)
 {
+    /* predicated accumulate: operand loads issue unconditionally (so row
+     * iterations pipeline with no divergent skip), the fold is masked by
+     * `pass` — the row loop in filter_agg_kernel has no continue */
     int64_t iv; double fv;
     switch (g.kind) {
         case CSTRIPE_AGG_COUNT_STAR:
-            a.cnt++;
-            a.lo++;
+            if (pass) { a.cnt++; a.lo++; }
             break;
-        case CSTRIPE_AGG_COUNT_COL:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.cnt++; a.lo++; }
+        case CSTRIPE_AGG_COUNT_COL: {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            if (ok & pass) { a.cnt++; a.lo++; }
             break;
-        case CSTRIPE_AGG_SUM_I64:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) {
+        }
+        case CSTRIPE_AGG_SUM_I64: {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            if (ok & pass) {
                 acc_add_i128(a, (__int128)iv);
                 a.cnt++;
             }
             break;
-        case CSTRIPE_AGG_SUM_F64:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.f += fv; a.cnt++; }
+        }
+        case CSTRIPE_AGG_SUM_F64: {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            if (ok & pass) { a.f += fv; a.cnt++; }
             break;
-        case CSTRIPE_AGG_MIN_I64:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.lo = min(a.lo, iv); a.cnt++; }
+        }
+        case CSTRIPE_AGG_MIN_I64: {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            if (ok & pass) { a.lo = min(a.lo, iv); a.cnt++; }
             break;
-        case CSTRIPE_AGG_MAX_I64:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) { a.lo = max(a.lo, iv); a.cnt++; }
+        }
+        case CSTRIPE_AGG_MAX_I64: {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            if (ok & pass) { a.lo = max(a.lo, iv); a.cnt++; }
             break;
-        case CSTRIPE_AGG_MIN_F64:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) {
+        }
+        case CSTRIPE_AGG_MIN_F64: {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            if (ok & pass) {
                 if (f64cmp_pg(fv, a.f) < 0) a.f = fv;
                 a.cnt++;
             }
             break;
-        case CSTRIPE_AGG_MAX_F64:
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv)) {
+        }
+        case CSTRIPE_AGG_MAX_F64: {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            if (ok & pass) {
                 if (f64cmp_pg(fv, a.f) > 0) a.f = fv;
                 a.cnt++;
             }
             break;
+        }
         case CSTRIPE_AGG_SUM_PROD_I64: {
             int64_t ib; double fb;
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
-                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb)) {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            ok = ok & col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb);
+            if (ok & pass) {
                 acc_add_i128(a, (__int128)iv * ib);
                 a.cnt++;
             }
@@ -658,8 +677,9 @@ __device__ inline void acc_row(ThreadAcc &a, const AggD &g,
         }
         case CSTRIPE_AGG_SUM_DISC_I64: {
             int64_t ib; double fb;
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
-                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb)) {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            ok = ok & col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb);
+            if (ok & pass) {
                 acc_add_i128(a, (__int128)iv * (g.one - ib));
                 a.cnt++;
             }
@@ -667,9 +687,10 @@ __device__ inline void acc_row(ThreadAcc &a, const AggD &g,
         }
         case CSTRIPE_AGG_SUM_DISC_TAX_I64: {
             int64_t ib, ic; double fb, fc;
-            if (col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv) &&
-                col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb) &&
-                col_value(data, scratch, rank, cols[g.proj_c], row, ic, fc)) {
+            bool ok = col_value(data, scratch, rank, cols[g.proj_a], row, iv, fv);
+            ok = ok & col_value(data, scratch, rank, cols[g.proj_b], row, ib, fb);
+            ok = ok & col_value(data, scratch, rank, cols[g.proj_c], row, ic, fc);
+            if (ok & pass) {
                 acc_add_i128(a, (__int128)iv * (g.one - ib) * (g.one + ic));
                 a.cnt++;
             }
@@ -731,7 +752,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
         if (!pass) continue;
         #pragma unroll
         for (uint32_t a = 0; a < n_aggs; a++)
-            acc_row(acc[a], params.aggs[a], data, scratch, rank, cols, row);
+            acc_row(acc[a], params.aggs[a], data, scratch, rank, cols, row, true);
     }
 
     /* wave reduce then cross-wave via LDS */
@@ -864,6 +885,220 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
             AccCell cell;
             cell.lo = r.lo; cell.hi = r.hi; cell.f = r.f; cell.cnt = r.cnt;
             block_out[(uint64_t)blockIdx.x * n_aggs + a] = cell;
+        }
+    }
+}
+
+/* =====================================================================
+ * Paired-row scan kernel for ALL-DENSE chunks (canonical, raw NONE, or
+ * scratch-decoded value streams): each thread evaluates TWO adjacent rows
+ * per iteration, and each column's pair of values comes from ONE wide
+ * unaligned load (16 B covers both canonical P(L) positions since the
+ * stride L+3 <= 7) — halving load transactions, the binding resource of
+ * this HBM-side scan. A wave-uniform ballot skips the aggregate operand
+ * loads when no lane's rows pass, without divergence inside the wave.
+ * ===================================================================== */
+
+__device__ inline void col_pair(const uint8_t *__restrict__ data,
+                                const uint8_t *__restrict__ scratch,
+                                const ColLoc &cl, uint32_t row,
+                                int64_t &iv0, int64_t &iv1,
+                                double &fv0, double &fv1)
+{
+    const uint8_t *base = (cl.flags & 1) ? scratch + cl.val_off : data + cl.val_off;
+    if ((cl.flags & 4) && cl.mode != CSF_SEGMODE_LIT) {
+        uint64_t r0, r1;
+        if (cl.mode == CSF_SEGMODE_CONST) {
+            r0 = r1 = (uint64_t)cl.hval;
+        } else {                             /* P(L) */
+            const uint32_t Lx = cl.L, step = Lx + 3u;
+            const uint64_t m = (~0ull) >> ((8u - Lx) * 8u);
+            if (row < 2) {                   /* j=0 -> 1, j=1 -> 9 specials */
+                uint64_t a0, a1;
+                __builtin_memcpy(&a0, base + (row == 0 ? 1u : 9u), 8);
+                __builtin_memcpy(&a1, base + (row == 0 ? 9u : 2u * step + 6u - Lx), 8);
+                r0 = (a0 & m) | (uint64_t)cl.hval;
+                r1 = (a1 & m) | (uint64_t)cl.hval;
+            } else {
+                const uint32_t pos = row * step + (6u - Lx);
+                uint64_t lo01, lo23;
+                __builtin_memcpy(&lo01, base + pos, 8);
+                __builtin_memcpy(&lo23, base + pos + 8, 8);
+                r0 = (lo01 & m) | (uint64_t)cl.hval;
+                const uint32_t sh = 8u * step;       /* 32..56 */
+                uint64_t v1raw = (lo01 >> sh) | (lo23 << (64u - sh));
+                r1 = (v1raw & m) | (uint64_t)cl.hval;
+            }
+        }
+        if (cl.type == CSTRIPE_F64) {
+            __builtin_memcpy(&fv0, &r0, 8);
+            __builtin_memcpy(&fv1, &r1, 8);
+            iv0 = iv1 = 0;
+        } else {
+            iv0 = (int64_t)r0; fv0 = (double)iv0;
+            iv1 = (int64_t)r1; fv1 = (double)iv1;
+        }
+        return;
+    }
+    if (cl.flags & 4) base += (uint64_t)cl.hval;     /* LIT header */
+    switch (cl.type) {                                /* row is even */
+        case CSTRIPE_I8: {
+            const int8_t *p = (const int8_t *)base + row;
+            iv0 = p[0]; iv1 = p[1]; fv0 = (double)iv0; fv1 = (double)iv1; break;
+        }
+        case CSTRIPE_I16: {
+            const int16_t *p = (const int16_t *)base + row;
+            iv0 = p[0]; iv1 = p[1]; fv0 = (double)iv0; fv1 = (double)iv1; break;
+        }
+        case CSTRIPE_I32: {
+            const int32_t *p = (const int32_t *)base + row;
+            iv0 = p[0]; iv1 = p[1]; fv0 = (double)iv0; fv1 = (double)iv1; break;
+        }
+        case CSTRIPE_I64: {
+            const int64_t *p = (const int64_t *)base + row;
+            iv0 = p[0]; iv1 = p[1]; fv0 = (double)iv0; fv1 = (double)iv1; break;
+        }
+        case CSTRIPE_F32: {
+            const float *p = (const float *)base + row;
+            fv0 = p[0]; fv1 = p[1]; iv0 = iv1 = 0; break;
+        }
+        default: {
+            const double *p = (const double *)base + row;
+            fv0 = p[0]; fv1 = p[1]; iv0 = iv1 = 0; break;
+        }
+    }
+}
+
+/* fold one row-pair into acc for one agg (dense: no null checks) */
+__device__ inline void acc_pair(ThreadAcc &a, const AggD &g,
+                                const uint8_t *__restrict__ data,
+                                const uint8_t *__restrict__ scratch,
+                                const ColLoc *__restrict__ cols, uint32_t row,
+                                bool p0, bool p1)
+{
+    int64_t a0, a1, b0, b1, c0, c1;
+    double fa0, fa1, fb0, fb1, fc0, fc1;
+    switch (g.kind) {
+        case CSTRIPE_AGG_COUNT_STAR:
+        case CSTRIPE_AGG_COUNT_COL:            /* dense: col always present */
+            a.cnt += p0 + p1;
+            a.lo += p0 + p1;
+            return;
+        case CSTRIPE_AGG_SUM_I64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            if (p0) acc_add_i128(a, (__int128)a0);
+            if (p1) acc_add_i128(a, (__int128)a1);
+            break;
+        case CSTRIPE_AGG_SUM_F64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            if (p0) a.f += fa0;
+            if (p1) a.f += fa1;
+            break;
+        case CSTRIPE_AGG_MIN_I64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            if (p0) a.lo = min(a.lo, a0);
+            if (p1) a.lo = min(a.lo, a1);
+            break;
+        case CSTRIPE_AGG_MAX_I64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            if (p0) a.lo = max(a.lo, a0);
+            if (p1) a.lo = max(a.lo, a1);
+            break;
+        case CSTRIPE_AGG_MIN_F64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            if (p0 && f64cmp_pg(fa0, a.f) < 0) a.f = fa0;
+            if (p1 && f64cmp_pg(fa1, a.f) < 0) a.f = fa1;
+            break;
+        case CSTRIPE_AGG_MAX_F64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            if (p0 && f64cmp_pg(fa0, a.f) > 0) a.f = fa0;
+            if (p1 && f64cmp_pg(fa1, a.f) > 0) a.f = fa1;
+            break;
+        case CSTRIPE_AGG_SUM_PROD_I64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            col_pair(data, scratch, cols[g.proj_b], row, b0, b1, fb0, fb1);
+            if (p0) acc_add_i128(a, (__int128)a0 * b0);
+            if (p1) acc_add_i128(a, (__int128)a1 * b1);
+            break;
+        case CSTRIPE_AGG_SUM_DISC_I64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            col_pair(data, scratch, cols[g.proj_b], row, b0, b1, fb0, fb1);
+            if (p0) acc_add_i128(a, (__int128)a0 * (g.one - b0));
+            if (p1) acc_add_i128(a, (__int128)a1 * (g.one - b1));
+            break;
+        case CSTRIPE_AGG_SUM_DISC_TAX_I64:
+            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
+            col_pair(data, scratch, cols[g.proj_b], row, b0, b1, fb0, fb1);
+            col_pair(data, scratch, cols[g.proj_c], row, c0, c1, fc0, fc1);
+            if (p0) acc_add_i128(a, (__int128)a0 * (g.one - b0) * (g.one + c0));
+            if (p1) acc_add_i128(a, (__int128)a1 * (g.one - b1) * (g.one + c1));
+            break;
+        default:
+            break;
+    }
+    a.cnt += p0 + p1;
+}
+
+template <int NPREDS, int NAGGS>
+__global__ __launch_bounds__(AGG_BLOCK) void pair_agg_kernel(
+    const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
+    const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
+    AccCell *__restrict__ block_out, const AggParams params)
+{
+    const uint32_t n_preds = NPREDS >= 0 ? (uint32_t)NPREDS : params.n_preds;
+    const uint32_t n_aggs = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
+    const uint32_t gid = blockIdx.x / params.tiles_per_group;
+    const uint32_t tile = blockIdx.x % params.tiles_per_group;
+    const GroupDesc g = groups[gid];
+    const ColLoc *cols = colloc + g.colbase;
+
+    const uint32_t row_start = tile * TILE_ROWS;
+    const uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
+
+    ThreadAcc acc[NAGGS >= 0 ? NAGGS : MAX_AGGS];
+    #pragma unroll
+    for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
+
+    for (uint32_t row = row_start + 2 * threadIdx.x; row < row_end;
+         row += 2 * AGG_BLOCK) {
+        bool p0 = true, p1 = row + 1 < row_end;
+        int last_proj = -1;
+        int64_t v0 = 0, v1 = 0; double f0 = 0, f1 = 0;
+        #pragma unroll
+        for (uint32_t p = 0; p < n_preds; p++) {
+            const PredD &pr = params.preds[p];
+            if ((int)pr.proj != last_proj) {
+                col_pair(data, scratch, cols[pr.proj], row, v0, v1, f0, f1);
+                last_proj = (int)pr.proj;
+            }
+            p0 = p0 & pred_eval(pr, v0, f0);
+            p1 = p1 & pred_eval(pr, v1, f1);
+        }
+        /* wave-uniform skip: agg operand loads only issue when some lane
+         * has a passing row (no intra-wave divergence) */
+        if (__ballot(p0 | p1) == 0) continue;
+        #pragma unroll
+        for (uint32_t a = 0; a < n_aggs; a++)
+            acc_pair(acc[a], params.aggs[a], data, scratch, cols, row, p0, p1);
+    }
+
+    __shared__ ThreadAcc lds[AGG_BLOCK / WAVE][MAX_AGGS];
+    const uint32_t wid = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    #pragma unroll
+    for (uint32_t a = 0; a < n_aggs; a++) {
+        wave_reduce(acc[a], params.aggs[a].kind);
+        if (lane == 0) lds[wid][a] = acc[a];
+    }
+    __syncthreads();
+    if (wid == 0) {
+        for (uint32_t a = lane; a < n_aggs; a += WAVE) {
+            ThreadAcc r = lds[0][a];
+            for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
+                acc_merge(r, lds[w][a], params.aggs[a].kind);
+            AccCell c;
+            c.lo = r.lo; c.hi = r.hi; c.f = r.f; c.cnt = r.cnt;
+            block_out[(uint64_t)blockIdx.x * n_aggs + a] = c;
         }
     }
 }
@@ -1674,6 +1909,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
             } else {
                 cl.flags |= 2;
             }
+            if (!(cl.flags & 2)) g->all_dense = false;
 
             const bool canon = nd.n.comp_type == CSTRIPE_COMP_LZ4 &&
                                nd.n.n_segs == 1 &&
@@ -2203,7 +2439,22 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
     HIP_TRY(hipEventRecord(g->ev0, g->stream));
     { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
     HIP_TRY(hipEventRecord(g->ev1, g->stream));
-    {
+    if (g->all_dense) {
+        /* all-dense: paired-row kernel — one wide load per column covers
+         * two rows (canonical closed-form or decoded/raw typed arrays) */
+        auto launchp = [&](auto *kern) {
+            hipLaunchKernelGGL(kern, dim3(n_blocks), dim3(AGG_BLOCK), 0, g->stream,
+                               g->d_data, g->d_scratch, g->d_groups,
+                               g->d_colloc, g->d_block, p);
+        };
+        if (p.n_preds == 5 && n_aggs == 2) launchp(pair_agg_kernel<5, 2>);
+        else if (p.n_preds == 5 && n_aggs == 1) launchp(pair_agg_kernel<5, 1>);
+        else if (p.n_preds == 1 && n_aggs == 1) launchp(pair_agg_kernel<1, 1>);
+        else if (p.n_preds == 2 && n_aggs == 2) launchp(pair_agg_kernel<2, 2>);
+        else if (p.n_preds == 5 && n_aggs == 4) launchp(pair_agg_kernel<5, 4>);
+        else launchp(pair_agg_kernel<-1, -1>);
+        HIP_TRY(hipGetLastError());
+    } else {
         auto launch = [&](auto *kern) {
             hipLaunchKernelGGL(kern, dim3(n_blocks), dim3(AGG_BLOCK), 0, g->stream,
                                g->d_data, g->d_scratch, g->d_rank, g->d_groups,
