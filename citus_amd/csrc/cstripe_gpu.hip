@@ -1103,6 +1103,258 @@ __global__ __launch_bounds__(AGG_BLOCK) void pair_agg_kernel(
     }
 }
 
+/* ---- quad-row variant: 4 adjacent rows per thread, a 32 B window per
+ * column (covers 4 canonical P(L) values for every L <= 4) -> twice the
+ * loads in flight per wave vs the pair kernel; the scan is latency-bound
+ * (SQ_WAIT_ANY ~82% of wave cycles, profiles/), so MLP is the lever. ---- */
+
+struct Quad {
+    int64_t v[4];
+    double f[4];
+};
+
+__device__ inline void col_quad(const uint8_t *__restrict__ data,
+                                const uint8_t *__restrict__ scratch,
+                                const ColLoc &cl, uint32_t row, Quad &q)
+{
+    const uint8_t *base = (cl.flags & 1) ? scratch + cl.val_off : data + cl.val_off;
+    if ((cl.flags & 4) && cl.mode != CSF_SEGMODE_LIT) {
+        uint64_t r[4];
+        if (cl.mode == CSF_SEGMODE_CONST) {
+            r[0] = r[1] = r[2] = r[3] = (uint64_t)cl.hval;
+        } else {                              /* P(L) */
+            const uint32_t Lx = cl.L, step = Lx + 3u;
+            const uint64_t m = (~0ull) >> ((8u - Lx) * 8u);
+            if (row < 2) {                    /* j=0 -> 1, j=1 -> 9 specials */
+                #pragma unroll
+                for (uint32_t k = 0; k < 4; k++) {
+                    const uint32_t j = row + k;
+                    uint32_t pos = j * step + (6u - Lx);
+                    pos = j == 0 ? 1u : (j == 1 ? 9u : pos);
+                    uint64_t a;
+                    __builtin_memcpy(&a, base + pos, 8);
+                    r[k] = (a & m) | (uint64_t)cl.hval;
+                }
+            } else {
+                const uint32_t pos = row * step + (6u - Lx);
+                uint64_t w0, w1, w2, w3;      /* 32 B window, 4 loads issue together */
+                __builtin_memcpy(&w0, base + pos, 8);
+                __builtin_memcpy(&w1, base + pos + 8, 8);
+                __builtin_memcpy(&w2, base + pos + 16, 8);
+                __builtin_memcpy(&w3, base + pos + 24, 8);
+                r[0] = (w0 & m) | (uint64_t)cl.hval;
+                #pragma unroll
+                for (uint32_t k = 1; k < 4; k++) {
+                    const uint32_t d = k * step;          /* 4..21 */
+                    const uint32_t wi = d >> 3, sh = (d & 7u) * 8u;
+                    uint64_t lo = wi == 0 ? w0 : (wi == 1 ? w1 : w2);
+                    uint64_t hi = wi == 0 ? w1 : (wi == 1 ? w2 : w3);
+                    uint64_t raw = sh ? ((lo >> sh) | (hi << (64u - sh))) : lo;
+                    r[k] = (raw & m) | (uint64_t)cl.hval;
+                }
+            }
+        }
+        if (cl.type == CSTRIPE_F64) {
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++) {
+                __builtin_memcpy(&q.f[k], &r[k], 8);
+                q.v[k] = 0;
+            }
+        } else {
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++) {
+                q.v[k] = (int64_t)r[k];
+                q.f[k] = (double)q.v[k];
+            }
+        }
+        return;
+    }
+    if (cl.flags & 4) base += (uint64_t)cl.hval;     /* LIT header */
+    switch (cl.type) {                                /* row % 4 == 0 */
+        case CSTRIPE_I8: {
+            const int8_t *p = (const int8_t *)base + row;
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++) { q.v[k] = p[k]; q.f[k] = (double)q.v[k]; }
+            break;
+        }
+        case CSTRIPE_I16: {
+            const int16_t *p = (const int16_t *)base + row;
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++) { q.v[k] = p[k]; q.f[k] = (double)q.v[k]; }
+            break;
+        }
+        case CSTRIPE_I32: {
+            const int32_t *p = (const int32_t *)base + row;
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++) { q.v[k] = p[k]; q.f[k] = (double)q.v[k]; }
+            break;
+        }
+        case CSTRIPE_I64: {
+            const int64_t *p = (const int64_t *)base + row;
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++) { q.v[k] = p[k]; q.f[k] = (double)q.v[k]; }
+            break;
+        }
+        case CSTRIPE_F32: {
+            const float *p = (const float *)base + row;
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++) { q.f[k] = p[k]; q.v[k] = 0; }
+            break;
+        }
+        default: {
+            const double *p = (const double *)base + row;
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++) { q.f[k] = p[k]; q.v[k] = 0; }
+            break;
+        }
+    }
+}
+
+__device__ inline void acc_quad(ThreadAcc &a, const AggD &g,
+                                const uint8_t *__restrict__ data,
+                                const uint8_t *__restrict__ scratch,
+                                const ColLoc *__restrict__ cols, uint32_t row,
+                                const bool pv[4])
+{
+    Quad qa, qb, qc;
+    switch (g.kind) {
+        case CSTRIPE_AGG_COUNT_STAR:
+        case CSTRIPE_AGG_COUNT_COL: {
+            const int64_t c = (int)pv[0] + (int)pv[1] + (int)pv[2] + (int)pv[3];
+            a.cnt += c;
+            a.lo += c;
+            return;
+        }
+        case CSTRIPE_AGG_SUM_I64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k]) acc_add_i128(a, (__int128)qa.v[k]);
+            break;
+        case CSTRIPE_AGG_SUM_F64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k]) a.f += qa.f[k];
+            break;
+        case CSTRIPE_AGG_MIN_I64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k]) a.lo = min(a.lo, qa.v[k]);
+            break;
+        case CSTRIPE_AGG_MAX_I64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k]) a.lo = max(a.lo, qa.v[k]);
+            break;
+        case CSTRIPE_AGG_MIN_F64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k] && f64cmp_pg(qa.f[k], a.f) < 0) a.f = qa.f[k];
+            break;
+        case CSTRIPE_AGG_MAX_F64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k] && f64cmp_pg(qa.f[k], a.f) > 0) a.f = qa.f[k];
+            break;
+        case CSTRIPE_AGG_SUM_PROD_I64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_quad(data, scratch, cols[g.proj_b], row, qb);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k]) acc_add_i128(a, (__int128)qa.v[k] * qb.v[k]);
+            break;
+        case CSTRIPE_AGG_SUM_DISC_I64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_quad(data, scratch, cols[g.proj_b], row, qb);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k]) acc_add_i128(a, (__int128)qa.v[k] * (g.one - qb.v[k]));
+            break;
+        case CSTRIPE_AGG_SUM_DISC_TAX_I64:
+            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_quad(data, scratch, cols[g.proj_b], row, qb);
+            col_quad(data, scratch, cols[g.proj_c], row, qc);
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                if (pv[k]) acc_add_i128(a, (__int128)qa.v[k] * (g.one - qb.v[k])
+                                           * (g.one + qc.v[k]));
+            break;
+        default:
+            break;
+    }
+    a.cnt += (int)pv[0] + (int)pv[1] + (int)pv[2] + (int)pv[3];
+}
+
+template <int NPREDS, int NAGGS>
+__global__ __launch_bounds__(AGG_BLOCK) void quad_agg_kernel(
+    const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
+    const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
+    AccCell *__restrict__ block_out, const AggParams params)
+{
+    const uint32_t n_preds = NPREDS >= 0 ? (uint32_t)NPREDS : params.n_preds;
+    const uint32_t n_aggs = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
+    const uint32_t gid = blockIdx.x / params.tiles_per_group;
+    const uint32_t tile = blockIdx.x % params.tiles_per_group;
+    const GroupDesc g = groups[gid];
+    const ColLoc *cols = colloc + g.colbase;
+
+    const uint32_t row_start = tile * TILE_ROWS;
+    const uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
+
+    ThreadAcc acc[NAGGS >= 0 ? NAGGS : MAX_AGGS];
+    #pragma unroll
+    for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
+
+    for (uint32_t row = row_start + 4 * threadIdx.x; row < row_end;
+         row += 4 * AGG_BLOCK) {
+        bool pv[4];
+        #pragma unroll
+        for (uint32_t k = 0; k < 4; k++) pv[k] = row + k < row_end;
+        int last_proj = -1;
+        Quad q{};
+        #pragma unroll
+        for (uint32_t p = 0; p < n_preds; p++) {
+            const PredD &pr = params.preds[p];
+            if ((int)pr.proj != last_proj) {
+                col_quad(data, scratch, cols[pr.proj], row, q);
+                last_proj = (int)pr.proj;
+            }
+            #pragma unroll
+            for (uint32_t k = 0; k < 4; k++)
+                pv[k] = pv[k] & pred_eval(pr, q.v[k], q.f[k]);
+        }
+        if (__ballot(pv[0] | pv[1] | pv[2] | pv[3]) == 0) continue;
+        #pragma unroll
+        for (uint32_t a = 0; a < n_aggs; a++)
+            acc_quad(acc[a], params.aggs[a], data, scratch, cols, row, pv);
+    }
+
+    __shared__ ThreadAcc lds[AGG_BLOCK / WAVE][MAX_AGGS];
+    const uint32_t wid = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    #pragma unroll
+    for (uint32_t a = 0; a < n_aggs; a++) {
+        wave_reduce(acc[a], params.aggs[a].kind);
+        if (lane == 0) lds[wid][a] = acc[a];
+    }
+    __syncthreads();
+    if (wid == 0) {
+        for (uint32_t a = lane; a < n_aggs; a += WAVE) {
+            ThreadAcc r = lds[0][a];
+            for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
+                acc_merge(r, lds[w][a], params.aggs[a].kind);
+            AccCell c;
+            c.lo = r.lo; c.hi = r.hi; c.f = r.f; c.cnt = r.cnt;
+            block_out[(uint64_t)blockIdx.x * n_aggs + a] = c;
+        }
+    }
+}
+
 /* per-row precomputed aggregate contribution: operands loaded ONCE per row,
  * so the per-distinct-key reduce rounds touch registers only */
 struct PrepAcc {
@@ -2447,12 +2699,12 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
                                g->d_data, g->d_scratch, g->d_groups,
                                g->d_colloc, g->d_block, p);
         };
-        if (p.n_preds == 5 && n_aggs == 2) launchp(pair_agg_kernel<5, 2>);
-        else if (p.n_preds == 5 && n_aggs == 1) launchp(pair_agg_kernel<5, 1>);
-        else if (p.n_preds == 1 && n_aggs == 1) launchp(pair_agg_kernel<1, 1>);
-        else if (p.n_preds == 2 && n_aggs == 2) launchp(pair_agg_kernel<2, 2>);
-        else if (p.n_preds == 5 && n_aggs == 4) launchp(pair_agg_kernel<5, 4>);
-        else launchp(pair_agg_kernel<-1, -1>);
+        if (p.n_preds == 5 && n_aggs == 2) launchp(quad_agg_kernel<5, 2>);
+        else if (p.n_preds == 5 && n_aggs == 1) launchp(quad_agg_kernel<5, 1>);
+        else if (p.n_preds == 1 && n_aggs == 1) launchp(quad_agg_kernel<1, 1>);
+        else if (p.n_preds == 2 && n_aggs == 2) launchp(quad_agg_kernel<2, 2>);
+        else if (p.n_preds == 5 && n_aggs == 4) launchp(quad_agg_kernel<5, 4>);
+        else launchp(quad_agg_kernel<-1, -1>);
         HIP_TRY(hipGetLastError());
     } else {
         auto launch = [&](auto *kern) {
