@@ -889,245 +889,68 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
     }
 }
 
-/* =====================================================================
- * Paired-row scan kernel for ALL-DENSE chunks (canonical, raw NONE, or
- * scratch-decoded value streams): each thread evaluates TWO adjacent rows
- * per iteration, and each column's pair of values comes from ONE wide
- * unaligned load (16 B covers both canonical P(L) positions since the
- * stride L+3 <= 7) — halving load transactions, the binding resource of
- * this HBM-side scan. A wave-uniform ballot skips the aggregate operand
- * loads when no lane's rows pass, without divergence inside the wave.
- * ===================================================================== */
+/* ---- R-row variant (R=8): one (R-1)*step+8 <= 57 B window per column.
+ * step in {4..7} is wave-uniform per column, so a 4-way uniform branch gives
+ * each extraction COMPILE-TIME shift amounts (no dynamic register indexing,
+ * no scratch). Doubles in-flight loads again over the quad kernel. ---- */
 
-__device__ inline void col_pair(const uint8_t *__restrict__ data,
-                                const uint8_t *__restrict__ scratch,
-                                const ColLoc &cl, uint32_t row,
-                                int64_t &iv0, int64_t &iv1,
-                                double &fv0, double &fv1)
-{
-    const uint8_t *base = (cl.flags & 1) ? scratch + cl.val_off : data + cl.val_off;
-    if ((cl.flags & 4) && cl.mode != CSF_SEGMODE_LIT) {
-        uint64_t r0, r1;
-        if (cl.mode == CSF_SEGMODE_CONST) {
-            r0 = r1 = (uint64_t)cl.hval;
-        } else {                             /* P(L) */
-            const uint32_t Lx = cl.L, step = Lx + 3u;
-            const uint64_t m = (~0ull) >> ((8u - Lx) * 8u);
-            if (row < 2) {                   /* j=0 -> 1, j=1 -> 9 specials */
-                uint64_t a0, a1;
-                __builtin_memcpy(&a0, base + (row == 0 ? 1u : 9u), 8);
-                __builtin_memcpy(&a1, base + (row == 0 ? 9u : 2u * step + 6u - Lx), 8);
-                r0 = (a0 & m) | (uint64_t)cl.hval;
-                r1 = (a1 & m) | (uint64_t)cl.hval;
-            } else {
-                const uint32_t pos = row * step + (6u - Lx);
-                uint64_t lo01, lo23;
-                __builtin_memcpy(&lo01, base + pos, 8);
-                __builtin_memcpy(&lo23, base + pos + 8, 8);
-                r0 = (lo01 & m) | (uint64_t)cl.hval;
-                const uint32_t sh = 8u * step;       /* 32..56 */
-                uint64_t v1raw = (lo01 >> sh) | (lo23 << (64u - sh));
-                r1 = (v1raw & m) | (uint64_t)cl.hval;
-            }
-        }
-        if (cl.type == CSTRIPE_F64) {
-            __builtin_memcpy(&fv0, &r0, 8);
-            __builtin_memcpy(&fv1, &r1, 8);
-            iv0 = iv1 = 0;
-        } else {
-            iv0 = (int64_t)r0; fv0 = (double)iv0;
-            iv1 = (int64_t)r1; fv1 = (double)iv1;
-        }
-        return;
+/* raw per-row values: int columns store the integer, float columns store
+ * the double's BIT PATTERN (converted only at use) — halves the register
+ * footprint of a staged row group, which is what caps occupancy here */
+template <int R>
+struct ValsR {
+    int64_t v[R];
+    __device__ inline double fget(int k) const {
+        double d;
+        __builtin_memcpy(&d, &v[k], 8);
+        return d;
     }
-    if (cl.flags & 4) base += (uint64_t)cl.hval;     /* LIT header */
-    switch (cl.type) {                                /* row is even */
-        case CSTRIPE_I8: {
-            const int8_t *p = (const int8_t *)base + row;
-            iv0 = p[0]; iv1 = p[1]; fv0 = (double)iv0; fv1 = (double)iv1; break;
-        }
-        case CSTRIPE_I16: {
-            const int16_t *p = (const int16_t *)base + row;
-            iv0 = p[0]; iv1 = p[1]; fv0 = (double)iv0; fv1 = (double)iv1; break;
-        }
-        case CSTRIPE_I32: {
-            const int32_t *p = (const int32_t *)base + row;
-            iv0 = p[0]; iv1 = p[1]; fv0 = (double)iv0; fv1 = (double)iv1; break;
-        }
-        case CSTRIPE_I64: {
-            const int64_t *p = (const int64_t *)base + row;
-            iv0 = p[0]; iv1 = p[1]; fv0 = (double)iv0; fv1 = (double)iv1; break;
-        }
-        case CSTRIPE_F32: {
-            const float *p = (const float *)base + row;
-            fv0 = p[0]; fv1 = p[1]; iv0 = iv1 = 0; break;
-        }
-        default: {
-            const double *p = (const double *)base + row;
-            fv0 = p[0]; fv1 = p[1]; iv0 = iv1 = 0; break;
-        }
-    }
-}
-
-/* fold one row-pair into acc for one agg (dense: no null checks) */
-__device__ inline void acc_pair(ThreadAcc &a, const AggD &g,
-                                const uint8_t *__restrict__ data,
-                                const uint8_t *__restrict__ scratch,
-                                const ColLoc *__restrict__ cols, uint32_t row,
-                                bool p0, bool p1)
-{
-    int64_t a0, a1, b0, b1, c0, c1;
-    double fa0, fa1, fb0, fb1, fc0, fc1;
-    switch (g.kind) {
-        case CSTRIPE_AGG_COUNT_STAR:
-        case CSTRIPE_AGG_COUNT_COL:            /* dense: col always present */
-            a.cnt += p0 + p1;
-            a.lo += p0 + p1;
-            return;
-        case CSTRIPE_AGG_SUM_I64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            if (p0) acc_add_i128(a, (__int128)a0);
-            if (p1) acc_add_i128(a, (__int128)a1);
-            break;
-        case CSTRIPE_AGG_SUM_F64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            if (p0) a.f += fa0;
-            if (p1) a.f += fa1;
-            break;
-        case CSTRIPE_AGG_MIN_I64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            if (p0) a.lo = min(a.lo, a0);
-            if (p1) a.lo = min(a.lo, a1);
-            break;
-        case CSTRIPE_AGG_MAX_I64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            if (p0) a.lo = max(a.lo, a0);
-            if (p1) a.lo = max(a.lo, a1);
-            break;
-        case CSTRIPE_AGG_MIN_F64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            if (p0 && f64cmp_pg(fa0, a.f) < 0) a.f = fa0;
-            if (p1 && f64cmp_pg(fa1, a.f) < 0) a.f = fa1;
-            break;
-        case CSTRIPE_AGG_MAX_F64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            if (p0 && f64cmp_pg(fa0, a.f) > 0) a.f = fa0;
-            if (p1 && f64cmp_pg(fa1, a.f) > 0) a.f = fa1;
-            break;
-        case CSTRIPE_AGG_SUM_PROD_I64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            col_pair(data, scratch, cols[g.proj_b], row, b0, b1, fb0, fb1);
-            if (p0) acc_add_i128(a, (__int128)a0 * b0);
-            if (p1) acc_add_i128(a, (__int128)a1 * b1);
-            break;
-        case CSTRIPE_AGG_SUM_DISC_I64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            col_pair(data, scratch, cols[g.proj_b], row, b0, b1, fb0, fb1);
-            if (p0) acc_add_i128(a, (__int128)a0 * (g.one - b0));
-            if (p1) acc_add_i128(a, (__int128)a1 * (g.one - b1));
-            break;
-        case CSTRIPE_AGG_SUM_DISC_TAX_I64:
-            col_pair(data, scratch, cols[g.proj_a], row, a0, a1, fa0, fa1);
-            col_pair(data, scratch, cols[g.proj_b], row, b0, b1, fb0, fb1);
-            col_pair(data, scratch, cols[g.proj_c], row, c0, c1, fc0, fc1);
-            if (p0) acc_add_i128(a, (__int128)a0 * (g.one - b0) * (g.one + c0));
-            if (p1) acc_add_i128(a, (__int128)a1 * (g.one - b1) * (g.one + c1));
-            break;
-        default:
-            break;
-    }
-    a.cnt += p0 + p1;
-}
-
-template <int NPREDS, int NAGGS>
-__global__ __launch_bounds__(AGG_BLOCK) void pair_agg_kernel(
-    const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
-    const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
-    AccCell *__restrict__ block_out, const AggParams params)
-{
-    const uint32_t n_preds = NPREDS >= 0 ? (uint32_t)NPREDS : params.n_preds;
-    const uint32_t n_aggs = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
-    const uint32_t gid = blockIdx.x / params.tiles_per_group;
-    const uint32_t tile = blockIdx.x % params.tiles_per_group;
-    const GroupDesc g = groups[gid];
-    const ColLoc *cols = colloc + g.colbase;
-
-    const uint32_t row_start = tile * TILE_ROWS;
-    const uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
-
-    ThreadAcc acc[NAGGS >= 0 ? NAGGS : MAX_AGGS];
-    #pragma unroll
-    for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
-
-    for (uint32_t row = row_start + 2 * threadIdx.x; row < row_end;
-         row += 2 * AGG_BLOCK) {
-        bool p0 = true, p1 = row + 1 < row_end;
-        int last_proj = -1;
-        int64_t v0 = 0, v1 = 0; double f0 = 0, f1 = 0;
-        #pragma unroll
-        for (uint32_t p = 0; p < n_preds; p++) {
-            const PredD &pr = params.preds[p];
-            if ((int)pr.proj != last_proj) {
-                col_pair(data, scratch, cols[pr.proj], row, v0, v1, f0, f1);
-                last_proj = (int)pr.proj;
-            }
-            p0 = p0 & pred_eval(pr, v0, f0);
-            p1 = p1 & pred_eval(pr, v1, f1);
-        }
-        /* wave-uniform skip: agg operand loads only issue when some lane
-         * has a passing row (no intra-wave divergence) */
-        if (__ballot(p0 | p1) == 0) continue;
-        #pragma unroll
-        for (uint32_t a = 0; a < n_aggs; a++)
-            acc_pair(acc[a], params.aggs[a], data, scratch, cols, row, p0, p1);
-    }
-
-    __shared__ ThreadAcc lds[AGG_BLOCK / WAVE][MAX_AGGS];
-    const uint32_t wid = threadIdx.x / WAVE;
-    const uint32_t lane = threadIdx.x % WAVE;
-    #pragma unroll
-    for (uint32_t a = 0; a < n_aggs; a++) {
-        wave_reduce(acc[a], params.aggs[a].kind);
-        if (lane == 0) lds[wid][a] = acc[a];
-    }
-    __syncthreads();
-    if (wid == 0) {
-        for (uint32_t a = lane; a < n_aggs; a += WAVE) {
-            ThreadAcc r = lds[0][a];
-            for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
-                acc_merge(r, lds[w][a], params.aggs[a].kind);
-            AccCell c;
-            c.lo = r.lo; c.hi = r.hi; c.f = r.f; c.cnt = r.cnt;
-            block_out[(uint64_t)blockIdx.x * n_aggs + a] = c;
-        }
-    }
-}
-
-/* ---- quad-row variant: 4 adjacent rows per thread, a 32 B window per
- * column (covers 4 canonical P(L) values for every L <= 4) -> twice the
- * loads in flight per wave vs the pair kernel; the scan is latency-bound
- * (SQ_WAIT_ANY ~82% of wave cycles, profiles/), so MLP is the lever. ---- */
-
-struct Quad {
-    int64_t v[4];
-    double f[4];
 };
 
-__device__ inline void col_quad(const uint8_t *__restrict__ data,
-                                const uint8_t *__restrict__ scratch,
-                                const ColLoc &cl, uint32_t row, Quad &q)
+template <int R, uint32_t STEP>
+__device__ inline void canon_extract(const uint8_t *__restrict__ sp,
+                                     uint32_t pos, uint64_t m, uint64_t hval,
+                                     uint64_t (&r)[R])
+{
+    /* window loaded as unaligned 16 B dwordx4 quads: half the vmem
+     * instructions of per-u64 loads on this load-issue-bound scan */
+    constexpr uint32_t NW = ((R - 1) * STEP + 8 + 7) / 8;
+    constexpr uint32_t NQ = (NW + 1) / 2;
+    uint64_t w[NQ * 2];
+    #pragma unroll
+    for (uint32_t i = 0; i < NQ; i++) {
+        struct { uint64_t a, b; } t;
+        __builtin_memcpy(&t, sp + pos + 16 * i, 16);
+        w[2 * i] = t.a;
+        w[2 * i + 1] = t.b;
+    }
+    #pragma unroll
+    for (uint32_t k = 0; k < R; k++) {
+        const uint32_t d = k * STEP;
+        const uint32_t wi = d >> 3, sh = (d & 7u) * 8u;
+        uint64_t raw = sh ? ((w[wi] >> sh) | (w[wi + 1] << (64u - sh)))
+                          : w[wi];
+        r[k] = (raw & m) | hval;
+    }
+}
+
+template <int R>
+__device__ inline void col_multi(const uint8_t *__restrict__ data,
+                                 const uint8_t *__restrict__ scratch,
+                                 const ColLoc &cl, uint32_t row, ValsR<R> &q)
 {
     const uint8_t *base = (cl.flags & 1) ? scratch + cl.val_off : data + cl.val_off;
     if ((cl.flags & 4) && cl.mode != CSF_SEGMODE_LIT) {
-        uint64_t r[4];
+        uint64_t r[R];
         if (cl.mode == CSF_SEGMODE_CONST) {
-            r[0] = r[1] = r[2] = r[3] = (uint64_t)cl.hval;
+            #pragma unroll
+            for (int k = 0; k < R; k++) r[k] = (uint64_t)cl.hval;
         } else {                              /* P(L) */
             const uint32_t Lx = cl.L, step = Lx + 3u;
             const uint64_t m = (~0ull) >> ((8u - Lx) * 8u);
             if (row < 2) {                    /* j=0 -> 1, j=1 -> 9 specials */
                 #pragma unroll
-                for (uint32_t k = 0; k < 4; k++) {
+                for (int k = 0; k < R; k++) {
                     const uint32_t j = row + k;
                     uint32_t pos = j * step + (6u - Lx);
                     pos = j == 0 ? 1u : (j == 1 ? 9u : pos);
@@ -1137,161 +960,157 @@ __device__ inline void col_quad(const uint8_t *__restrict__ data,
                 }
             } else {
                 const uint32_t pos = row * step + (6u - Lx);
-                uint64_t w0, w1, w2, w3;      /* 32 B window, 4 loads issue together */
-                __builtin_memcpy(&w0, base + pos, 8);
-                __builtin_memcpy(&w1, base + pos + 8, 8);
-                __builtin_memcpy(&w2, base + pos + 16, 8);
-                __builtin_memcpy(&w3, base + pos + 24, 8);
-                r[0] = (w0 & m) | (uint64_t)cl.hval;
-                #pragma unroll
-                for (uint32_t k = 1; k < 4; k++) {
-                    const uint32_t d = k * step;          /* 4..21 */
-                    const uint32_t wi = d >> 3, sh = (d & 7u) * 8u;
-                    uint64_t lo = wi == 0 ? w0 : (wi == 1 ? w1 : w2);
-                    uint64_t hi = wi == 0 ? w1 : (wi == 1 ? w2 : w3);
-                    uint64_t raw = sh ? ((lo >> sh) | (hi << (64u - sh))) : lo;
-                    r[k] = (raw & m) | (uint64_t)cl.hval;
+                /* wave-uniform branch on step -> compile-time shifts */
+                switch (step) {
+                    case 4: canon_extract<R, 4>(base, pos, m, (uint64_t)cl.hval, r); break;
+                    case 5: canon_extract<R, 5>(base, pos, m, (uint64_t)cl.hval, r); break;
+                    case 6: canon_extract<R, 6>(base, pos, m, (uint64_t)cl.hval, r); break;
+                    default: canon_extract<R, 7>(base, pos, m, (uint64_t)cl.hval, r); break;
                 }
             }
         }
-        if (cl.type == CSTRIPE_F64) {
-            #pragma unroll
-            for (uint32_t k = 0; k < 4; k++) {
-                __builtin_memcpy(&q.f[k], &r[k], 8);
-                q.v[k] = 0;
-            }
-        } else {
-            #pragma unroll
-            for (uint32_t k = 0; k < 4; k++) {
-                q.v[k] = (int64_t)r[k];
-                q.f[k] = (double)q.v[k];
-            }
-        }
+        #pragma unroll
+        for (int k = 0; k < R; k++) q.v[k] = (int64_t)r[k];   /* F64: bits */
         return;
     }
     if (cl.flags & 4) base += (uint64_t)cl.hval;     /* LIT header */
-    switch (cl.type) {                                /* row % 4 == 0 */
+    switch (cl.type) {                                /* row % R == 0 */
         case CSTRIPE_I8: {
             const int8_t *p = (const int8_t *)base + row;
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++) { q.v[k] = p[k]; q.f[k] = (double)q.v[k]; }
+            for (int k = 0; k < R; k++) q.v[k] = p[k];
             break;
         }
         case CSTRIPE_I16: {
             const int16_t *p = (const int16_t *)base + row;
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++) { q.v[k] = p[k]; q.f[k] = (double)q.v[k]; }
+            for (int k = 0; k < R; k++) q.v[k] = p[k];
             break;
         }
         case CSTRIPE_I32: {
             const int32_t *p = (const int32_t *)base + row;
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++) { q.v[k] = p[k]; q.f[k] = (double)q.v[k]; }
+            for (int k = 0; k < R; k++) q.v[k] = p[k];
             break;
         }
         case CSTRIPE_I64: {
-            const int64_t *p = (const int64_t *)base + row;
+            const uint8_t *p = base + (size_t)row * 8;
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++) { q.v[k] = p[k]; q.f[k] = (double)q.v[k]; }
+            for (int k = 0; k < R; k += 2) {
+                struct { int64_t a, b; } t;
+                __builtin_memcpy(&t, p + 16 * (k / 2), 16);
+                q.v[k] = t.a;
+                if (k + 1 < R) q.v[k + 1] = t.b;
+            }
             break;
         }
         case CSTRIPE_F32: {
             const float *p = (const float *)base + row;
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++) { q.f[k] = p[k]; q.v[k] = 0; }
+            for (int k = 0; k < R; k++) {
+                const double d = p[k];
+                __builtin_memcpy(&q.v[k], &d, 8);     /* widen, store bits */
+            }
             break;
         }
-        default: {
-            const double *p = (const double *)base + row;
+        default: {                                    /* F64 bits */
+            const uint8_t *p = base + (size_t)row * 8;
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++) { q.f[k] = p[k]; q.v[k] = 0; }
+            for (int k = 0; k < R; k += 2) {
+                struct { int64_t a, b; } t;
+                __builtin_memcpy(&t, p + 16 * (k / 2), 16);
+                q.v[k] = t.a;
+                if (k + 1 < R) q.v[k + 1] = t.b;
+            }
             break;
         }
     }
 }
 
-__device__ inline void acc_quad(ThreadAcc &a, const AggD &g,
-                                const uint8_t *__restrict__ data,
-                                const uint8_t *__restrict__ scratch,
-                                const ColLoc *__restrict__ cols, uint32_t row,
-                                const bool pv[4])
+template <int R>
+__device__ inline void acc_multi(ThreadAcc &a, const AggD &g,
+                                 const uint8_t *__restrict__ data,
+                                 const uint8_t *__restrict__ scratch,
+                                 const ColLoc *__restrict__ cols, uint32_t row,
+                                 const bool (&pv)[R])
 {
-    Quad qa, qb, qc;
+    ValsR<R> qa, qb, qc;
+    int64_t c = 0;
+    #pragma unroll
+    for (int k = 0; k < R; k++) c += (int)pv[k];
     switch (g.kind) {
         case CSTRIPE_AGG_COUNT_STAR:
-        case CSTRIPE_AGG_COUNT_COL: {
-            const int64_t c = (int)pv[0] + (int)pv[1] + (int)pv[2] + (int)pv[3];
+        case CSTRIPE_AGG_COUNT_COL:
             a.cnt += c;
             a.lo += c;
             return;
-        }
         case CSTRIPE_AGG_SUM_I64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
+            for (int k = 0; k < R; k++)
                 if (pv[k]) acc_add_i128(a, (__int128)qa.v[k]);
             break;
         case CSTRIPE_AGG_SUM_F64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
-                if (pv[k]) a.f += qa.f[k];
+            for (int k = 0; k < R; k++)
+                if (pv[k]) a.f += qa.fget(k);
             break;
         case CSTRIPE_AGG_MIN_I64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
+            for (int k = 0; k < R; k++)
                 if (pv[k]) a.lo = min(a.lo, qa.v[k]);
             break;
         case CSTRIPE_AGG_MAX_I64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
+            for (int k = 0; k < R; k++)
                 if (pv[k]) a.lo = max(a.lo, qa.v[k]);
             break;
         case CSTRIPE_AGG_MIN_F64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
-                if (pv[k] && f64cmp_pg(qa.f[k], a.f) < 0) a.f = qa.f[k];
+            for (int k = 0; k < R; k++)
+                if (pv[k] && f64cmp_pg(qa.fget(k), a.f) < 0) a.f = qa.fget(k);
             break;
         case CSTRIPE_AGG_MAX_F64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
-                if (pv[k] && f64cmp_pg(qa.f[k], a.f) > 0) a.f = qa.f[k];
+            for (int k = 0; k < R; k++)
+                if (pv[k] && f64cmp_pg(qa.fget(k), a.f) > 0) a.f = qa.fget(k);
             break;
         case CSTRIPE_AGG_SUM_PROD_I64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
-            col_quad(data, scratch, cols[g.proj_b], row, qb);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_b], row, qb);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
+            for (int k = 0; k < R; k++)
                 if (pv[k]) acc_add_i128(a, (__int128)qa.v[k] * qb.v[k]);
             break;
         case CSTRIPE_AGG_SUM_DISC_I64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
-            col_quad(data, scratch, cols[g.proj_b], row, qb);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_b], row, qb);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
+            for (int k = 0; k < R; k++)
                 if (pv[k]) acc_add_i128(a, (__int128)qa.v[k] * (g.one - qb.v[k]));
             break;
         case CSTRIPE_AGG_SUM_DISC_TAX_I64:
-            col_quad(data, scratch, cols[g.proj_a], row, qa);
-            col_quad(data, scratch, cols[g.proj_b], row, qb);
-            col_quad(data, scratch, cols[g.proj_c], row, qc);
+            col_multi<R>(data, scratch, cols[g.proj_a], row, qa);
+            col_multi<R>(data, scratch, cols[g.proj_b], row, qb);
+            col_multi<R>(data, scratch, cols[g.proj_c], row, qc);
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
+            for (int k = 0; k < R; k++)
                 if (pv[k]) acc_add_i128(a, (__int128)qa.v[k] * (g.one - qb.v[k])
                                            * (g.one + qc.v[k]));
             break;
         default:
             break;
     }
-    a.cnt += (int)pv[0] + (int)pv[1] + (int)pv[2] + (int)pv[3];
+    a.cnt += c;
 }
 
-template <int NPREDS, int NAGGS>
-__global__ __launch_bounds__(AGG_BLOCK) void quad_agg_kernel(
+template <int NPREDS, int NAGGS, int R, int MINW = 1>
+__global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_agg_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
     const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
     AccCell *__restrict__ block_out, const AggParams params)
@@ -1310,28 +1129,31 @@ __global__ __launch_bounds__(AGG_BLOCK) void quad_agg_kernel(
     #pragma unroll
     for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
 
-    for (uint32_t row = row_start + 4 * threadIdx.x; row < row_end;
-         row += 4 * AGG_BLOCK) {
-        bool pv[4];
+    for (uint32_t row = row_start + R * threadIdx.x; row < row_end;
+         row += R * AGG_BLOCK) {
+        bool pv[R];
         #pragma unroll
-        for (uint32_t k = 0; k < 4; k++) pv[k] = row + k < row_end;
+        for (int k = 0; k < R; k++) pv[k] = row + k < row_end;
         int last_proj = -1;
-        Quad q{};
+        ValsR<R> q{};
         #pragma unroll
         for (uint32_t p = 0; p < n_preds; p++) {
             const PredD &pr = params.preds[p];
             if ((int)pr.proj != last_proj) {
-                col_quad(data, scratch, cols[pr.proj], row, q);
+                col_multi<R>(data, scratch, cols[pr.proj], row, q);
                 last_proj = (int)pr.proj;
             }
             #pragma unroll
-            for (uint32_t k = 0; k < 4; k++)
-                pv[k] = pv[k] & pred_eval(pr, q.v[k], q.f[k]);
+            for (int k = 0; k < R; k++)
+                pv[k] = pv[k] & pred_eval(pr, q.v[k], q.fget(k));
         }
-        if (__ballot(pv[0] | pv[1] | pv[2] | pv[3]) == 0) continue;
+        bool any = false;
+        #pragma unroll
+        for (int k = 0; k < R; k++) any |= pv[k];
+        if (__ballot(any) == 0) continue;
         #pragma unroll
         for (uint32_t a = 0; a < n_aggs; a++)
-            acc_quad(acc[a], params.aggs[a], data, scratch, cols, row, pv);
+            acc_multi<R>(acc[a], params.aggs[a], data, scratch, cols, row, pv);
     }
 
     __shared__ ThreadAcc lds[AGG_BLOCK / WAVE][MAX_AGGS];
@@ -2699,12 +2521,33 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
                                g->d_data, g->d_scratch, g->d_groups,
                                g->d_colloc, g->d_block, p);
         };
-        if (p.n_preds == 5 && n_aggs == 2) launchp(quad_agg_kernel<5, 2>);
-        else if (p.n_preds == 5 && n_aggs == 1) launchp(quad_agg_kernel<5, 1>);
-        else if (p.n_preds == 1 && n_aggs == 1) launchp(quad_agg_kernel<1, 1>);
-        else if (p.n_preds == 2 && n_aggs == 2) launchp(quad_agg_kernel<2, 2>);
-        else if (p.n_preds == 5 && n_aggs == 4) launchp(quad_agg_kernel<5, 4>);
-        else launchp(quad_agg_kernel<-1, -1>);
+        /* variant knob for on-hardware A/B (default = R8, natural occupancy) */
+        static const int kvar = [] {
+            const char *e = getenv("CSTRIPE_KERNEL_VARIANT");
+            return e ? atoi(e) : 1;     /* measured best: R8 + 8 waves/SIMD */
+        }();
+        if (kvar == 1) {        /* forced 8 waves/SIMD (64 VGPRs, some spill) */
+            if (p.n_preds == 5 && n_aggs == 2) launchp(multi_agg_kernel<5, 2, 8, 8>);
+            else if (p.n_preds == 5 && n_aggs == 1) launchp(multi_agg_kernel<5, 1, 8, 8>);
+            else if (p.n_preds == 1 && n_aggs == 1) launchp(multi_agg_kernel<1, 1, 8, 8>);
+            else if (p.n_preds == 2 && n_aggs == 2) launchp(multi_agg_kernel<2, 2, 8, 8>);
+            else if (p.n_preds == 5 && n_aggs == 4) launchp(multi_agg_kernel<5, 4, 8, 8>);
+            else launchp(multi_agg_kernel<-1, -1, 4>);
+        } else if (kvar == 2) { /* R=4 */
+            if (p.n_preds == 5 && n_aggs == 2) launchp(multi_agg_kernel<5, 2, 4>);
+            else if (p.n_preds == 5 && n_aggs == 1) launchp(multi_agg_kernel<5, 1, 4>);
+            else if (p.n_preds == 1 && n_aggs == 1) launchp(multi_agg_kernel<1, 1, 4>);
+            else if (p.n_preds == 2 && n_aggs == 2) launchp(multi_agg_kernel<2, 2, 4>);
+            else if (p.n_preds == 5 && n_aggs == 4) launchp(multi_agg_kernel<5, 4, 4>);
+            else launchp(multi_agg_kernel<-1, -1, 4>);
+        } else {
+            if (p.n_preds == 5 && n_aggs == 2) launchp(multi_agg_kernel<5, 2, 8>);
+            else if (p.n_preds == 5 && n_aggs == 1) launchp(multi_agg_kernel<5, 1, 8>);
+            else if (p.n_preds == 1 && n_aggs == 1) launchp(multi_agg_kernel<1, 1, 8>);
+            else if (p.n_preds == 2 && n_aggs == 2) launchp(multi_agg_kernel<2, 2, 8>);
+            else if (p.n_preds == 5 && n_aggs == 4) launchp(multi_agg_kernel<5, 4, 8>);
+            else launchp(multi_agg_kernel<-1, -1, 4>);
+        }
         HIP_TRY(hipGetLastError());
     } else {
         auto launch = [&](auto *kern) {
