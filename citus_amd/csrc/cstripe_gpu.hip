@@ -912,23 +912,19 @@ __device__ inline void canon_extract(const uint8_t *__restrict__ sp,
                                      uint32_t pos, uint64_t m, uint64_t hval,
                                      uint64_t (&r)[R])
 {
-    /* window loaded as unaligned 16 B dwordx4 quads: half the vmem
-     * instructions of per-u64 loads on this load-issue-bound scan */
+    /* per-u64 window loads (measured faster than 16 B dwordx4 quads here:
+     * the unaligned quads span more lines per instruction and raised the
+     * kernel's spill pressure for no latency win) */
     constexpr uint32_t NW = ((R - 1) * STEP + 8 + 7) / 8;
-    constexpr uint32_t NQ = (NW + 1) / 2;
-    uint64_t w[NQ * 2];
+    uint64_t w[NW];
     #pragma unroll
-    for (uint32_t i = 0; i < NQ; i++) {
-        struct { uint64_t a, b; } t;
-        __builtin_memcpy(&t, sp + pos + 16 * i, 16);
-        w[2 * i] = t.a;
-        w[2 * i + 1] = t.b;
-    }
+    for (uint32_t i = 0; i < NW; i++)
+        __builtin_memcpy(&w[i], sp + pos + 8 * i, 8);
     #pragma unroll
     for (uint32_t k = 0; k < R; k++) {
         const uint32_t d = k * STEP;
         const uint32_t wi = d >> 3, sh = (d & 7u) * 8u;
-        uint64_t raw = sh ? ((w[wi] >> sh) | (w[wi + 1] << (64u - sh)))
+        uint64_t raw = sh ? ((w[wi] >> sh) | (w[wi + 1 < NW ? wi + 1 : wi] << (64u - sh)))
                           : w[wi];
         r[k] = (raw & m) | hval;
     }
