@@ -73,10 +73,12 @@ def count_preds(ca):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=10)
-    ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--rows", type=int, default=100_000_000,
-                    help="rows per GPU (config 2: 100M)")
+    ap.add_argument("--steps", type=int, default=0,
+                    help="timed steps (0 = auto: ~10 s of steady-state steps)")
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--rows", type=int, default=1_000_000_000,
+                    help="rows per GPU (default 1B: the north-star single-GPU "
+                         "config; config 2 = 100M)")
     ap.add_argument("--compression", default="lz4", choices=["lz4", "none", "zstd"])
     ap.add_argument("--query", default="q6", choices=["q6", "q1", "count"])
     ap.add_argument("--seg-bytes", type=int, default=0,
@@ -86,9 +88,10 @@ def main():
     ap.add_argument("--canonical", type=int, default=1,
                     help="writer canonical stream modes (closed-form GPU "
                          "access); 0 = greedy parse (round-1 behaviour)")
-    ap.add_argument("--shards", type=int, default=1,
-                    help="shard files per GPU scanned as one table "
-                         "(config 3: 4 shards/GPU x 8 GPUs = 32 shards)")
+    ap.add_argument("--shards", type=int, default=0,
+                    help="shard files per GPU scanned as one table (config 3: "
+                         "4 shards/GPU x 8 GPUs = 32 shards). 0 = auto: 16 for "
+                         ">=100M rows (parallel generation), else 1")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -109,28 +112,32 @@ def main():
 
     comp = {"lz4": ca.COMP_LZ4, "none": ca.COMP_NONE, "zstd": ca.COMP_ZSTD}[args.compression]
 
-    # ---- setup (untimed): generate per-rank shard, open, prune, stage ----
+    # ---- setup (untimed): generate per-rank shards, open, prune, stage ----
+    if args.shards == 0:
+        args.shards = 16 if args.rows >= 100_000_000 else 1
+    base_seed = 42 + rank * args.shards
     cache = os.environ.get("CSTRIPE_BENCH_DIR", "/tmp/cstripe_bench")
     os.makedirs(cache, exist_ok=True)
+    seg_kb = -args.seg_bytes if args.seg_bytes else 0
     if args.shards > 1:
         shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}"
                                     f"_m{args.min_match}_c{args.canonical}_n{args.shards}_r{rank}")
-        os.makedirs(shard, exist_ok=True)
         t0 = time.time()
-        per = args.rows // args.shards
-        for i in range(args.shards):
-            p = os.path.join(shard, f"shard{i:02d}.cs")
-            if not os.path.exists(p):
-                ca.gen_lineitem(p, per + (args.rows % args.shards if i == 0 else 0),
-                                seed=42 + rank * args.shards + i, compression=comp,
-                                seg_bytes=args.seg_bytes, min_match=args.min_match,
-                                canonical=args.canonical)
+        if not os.path.exists(os.path.join(shard, f"shard{args.shards - 1:02d}.cs")):
+            # parallel generation, one writer thread per shard; when several
+            # ranks generate on one host, split the cores between them
+            ncpu = os.cpu_count() or 8
+            ca.gen_lineitem_shards(shard, args.rows, args.shards,
+                                   base_seed=base_seed, compression=comp,
+                                   seg_kb=seg_kb, min_match=args.min_match,
+                                   canonical=args.canonical,
+                                   threads=max(2, ncpu // max(1, world)))
     else:
         shard = os.path.join(cache, f"li_{args.rows}_{args.compression}_s{args.seg_bytes}"
                                     f"_m{args.min_match}_c{args.canonical}_r{rank}.cs")
         t0 = time.time()
         if not os.path.exists(shard):
-            ca.gen_lineitem(shard, args.rows, seed=42 + rank, compression=comp,
+            ca.gen_lineitem(shard, args.rows, seed=base_seed, compression=comp,
                             seg_bytes=args.seg_bytes, min_match=args.min_match,
                             canonical=args.canonical)
     gen_s = time.time() - t0
@@ -164,9 +171,15 @@ def main():
             return all_gather_combine(aggs, parts, device="cuda")
         return ca.combine(aggs, [parts])
 
-    # ---- warmup ----
-    for _ in range(args.warmup):
+    # ---- warmup (also calibrates auto step count) ----
+    t0 = time.time()
+    for _ in range(max(1, args.warmup)):
         result = step()
+    per_step_est = (time.time() - t0) / max(1, args.warmup)
+    if args.steps == 0:
+        # ~10 s of timed steps: long enough for the driver-side activity
+        # sampler to observe the GPU busy, bounded for tiny workloads
+        args.steps = max(10, min(20000, int(10.0 / max(per_step_est, 1e-4))))
 
     decode_ms = []
     agg_ms = []
@@ -208,8 +221,11 @@ def main():
     avg_agg = sum(agg_ms) / len(agg_ms)
     if scan.last_fused:
         dominant = "fused_agg_kernel"
+    elif avg_decode >= avg_agg:
+        dominant = "lz4_decode_kernel"
     else:
-        dominant = "lz4_decode_kernel" if avg_decode >= avg_agg else "filter_agg_kernel"
+        # all-dense scans (canonical writer default) run the multi-row kernel
+        dominant = "multi_agg_kernel"
     dominant_ms = max(avg_decode, avg_agg)
     alg_bytes = args.rows * alg_bytes_per_row          # per launch (this rank)
     achieved_gbps = alg_bytes / (dominant_ms / 1e3) / 1e9 if dominant_ms > 0 else None
@@ -228,25 +244,60 @@ def main():
     except Exception:
         traffic = None
 
+    # ---- in-run parity pin: generator-exact expected values (any size) ----
+    # csbench_expected_q6/_q1 recompute the query straight from the seeded
+    # generator streams with int128 on host CPU (bench_gen.cpp) — an
+    # independent end-to-end check of writer -> stage -> decode -> filter ->
+    # aggregate -> combine at the FULL benchmark size.
+    parity = None
+    if world == 1:
+        shard_rows = [args.rows // args.shards + (args.rows % args.shards if i == 0 else 0)
+                      for i in range(args.shards)]
+        if args.query == "q6":
+            exp_rev, exp_cnt = 0, 0
+            for i, nrows in enumerate(shard_rows):
+                r_, c_ = ca.expected_q6(nrows, base_seed + i)
+                exp_rev += r_
+                exp_cnt += c_
+            parity = ("bit-exact" if (result[0].i128 == exp_rev and
+                                      result[1].count == exp_cnt)
+                      else "MISMATCH")
+        elif args.query == "q1":
+            exp = {}
+            for i, nrows in enumerate(shard_rows):
+                for k, v in ca.expected_q1(nrows, base_seed + i).items():
+                    cur = exp.setdefault(k, [0, 0, 0, 0, 0])
+                    for j in range(5):
+                        cur[j] += v[j]
+            exp = {k: v for k, v in exp.items() if v[4] > 0}
+            ok = set(exp) == set(result)
+            if ok:
+                for k, v in exp.items():
+                    got = result[k]
+                    ok = ok and [got[0].i128, got[1].i128, got[2].i128,
+                                 got[3].i128, got[4].count] == v
+            parity = "bit-exact" if ok else "MISMATCH"
+
     # ---- CPU baseline (oracle restatement, rank 0, N=1 only) ----
     cpu_baseline = None
     cpu_mt = None
-    parity = None
     if world == 1 and not args.no_cpu_baseline:
         import oracle
-        sample_rows = args.rows
-        sample_path = shard
-        if args.rows > 150_000_000:
-            sample_rows = 100_000_000
-            sample_path = os.path.join(cache, f"li_{sample_rows}_{args.compression}_r0.cs")
-            if not os.path.exists(sample_path):
-                ca.gen_lineitem(sample_path, sample_rows, seed=42, compression=comp)
+        # bounded sample: ONE shard file of the same dataset (~10-30 s CPU)
+        if args.shards > 1:
+            sample_path = os.path.join(shard, "shard01.cs")
+            sample_rows = args.rows // args.shards
+            sample_desc = f"one {sample_rows / 1e6:.0f}M-row shard of the same dataset"
+        else:
+            sample_path = shard
+            sample_rows = args.rows
+            sample_desc = f"full {sample_rows / 1e6:.0f}M-row pass"
         with oracle.OracleTable(sample_path) as t:
             t0 = time.time()
             if group_cols:
-                cpu_res, _ = t.scan_agg(preds, aggs, group_cols=group_cols)
+                t.scan_agg(preds, aggs, group_cols=group_cols)
             else:
-                cpu_parts, _ = t.scan_agg(preds, aggs)
+                t.scan_agg(preds, aggs)
             cpu_s = time.time() - t0
             cpu_mt = None
             if not group_cols:
@@ -254,26 +305,14 @@ def main():
                 _mt_parts, mt_cores = t.scan_agg_mt(preds, aggs)
                 cpu_mt = {"value": sample_rows / (time.time() - t0),
                           "unit": "rows/s", "cores": mt_cores, "kind": "port",
-                          "sample": f"all-core, same {sample_rows / 1e6:.0f}M-row pass"}
+                          "sample": f"all-core, {sample_desc}"}
         cpu_baseline = {
             "value": sample_rows / cpu_s,
             "unit": "rows/s",
             "cores": 1,
             "kind": "port",
-            "sample": f"full {sample_rows / 1e6:.0f}M-row Q6 pass, single thread "
-                      f"({cpu_s:.1f}s)",
+            "sample": f"{sample_desc}, single thread ({cpu_s:.1f}s)",
         }
-        if sample_path == shard:
-            if group_cols:
-                ok = set(cpu_res) == set(result) and all(
-                    cpu_res[k][a].i128 == result[k][a].i128 and
-                    cpu_res[k][a].count == result[k][a].count
-                    for k in cpu_res for a in range(len(aggs)))
-            else:
-                ok = all(cpu_parts[a].i128 == result[a].i128 and
-                         cpu_parts[a].count == result[a].count
-                         for a in range(len(aggs)))
-            parity = "bit-exact" if ok else "MISMATCH"
 
     out = {
         "metric": f"columnar_rows_per_s_{args.query}",

@@ -130,6 +130,15 @@ _gen_lineitem = _sig("csbench_gen_lineitem", C.c_int,
 _gen_lineitem2 = _sig("csbench_gen_lineitem2", C.c_int,
                       [C.c_char_p, C.c_uint64, C.c_uint64, C.c_int, C.c_int, C.c_int,
                        C.c_uint64, C.c_uint32, C.c_int, C.c_int])
+_gen_shards = _sig("csbench_gen_lineitem_shards", C.c_int,
+                   [C.c_char_p, C.c_uint64, C.c_uint32, C.c_uint64, C.c_int, C.c_int,
+                    C.c_int, C.c_uint64, C.c_uint32, C.c_int, C.c_int, C.c_int])
+_expected_q6 = _sig("csbench_expected_q6", C.c_int,
+                    [C.c_uint64, C.c_uint64, C.POINTER(C.c_int64),
+                     C.POINTER(C.c_int64), C.POINTER(C.c_int64)])
+_expected_q1 = _sig("csbench_expected_q1", C.c_int,
+                    [C.c_uint64, C.c_uint64, C.POINTER(C.c_int64),
+                     C.POINTER(C.c_int64), C.POINTER(C.c_int64)])
 
 
 def errmsg():
@@ -191,6 +200,43 @@ def write_table(path, defs, columns, nulls=None, **opt_kw):
             nl[i] = a.ctypes.data_as(C.c_void_p).value if a is not None else None
     _check(_write_rows(w, n, vals, nl), "write_rows")
     _check(_write_end(w), "write_end")
+
+
+def gen_lineitem_shards(dirpath, total_rows, n_shards, base_seed=42,
+                        compression=COMP_LZ4, level=3, seg_kb=0, stripe_rows=0,
+                        chunk_rows=0, min_match=0, canonical=1, threads=0):
+    """Parallel sharded generation: n_shards files shardNN.cs under dirpath
+    (the config-3 shard-directory shape), seeds base_seed + i."""
+    os.makedirs(dirpath, exist_ok=True)
+    _check(_gen_shards(dirpath.encode(), total_rows, n_shards, base_seed,
+                       compression, level, seg_kb, stripe_rows, chunk_rows,
+                       min_match, canonical, threads), "gen_lineitem_shards")
+
+
+def expected_q6(n_rows, seed=42):
+    """Exact Q6 (revenue int128 at scale 4, count) recomputed from the
+    generator streams on host CPU — the bench's in-run parity pin."""
+    lo, hi, cnt = C.c_int64(), C.c_int64(), C.c_int64()
+    _check(_expected_q6(n_rows, seed, C.byref(lo), C.byref(hi), C.byref(cnt)),
+           "expected_q6")
+    return (hi.value << 64) | (lo.value & ((1 << 64) - 1)), cnt.value
+
+
+def expected_q1(n_rows, seed=42):
+    """Exact Q1 per-(returnflag, linestatus) sums/counts from the generator.
+    Returns {(rf, ls): [sum_qty, sum_price, sum_disc_price, sum_charge, count]}."""
+    lo = (C.c_int64 * 24)()
+    hi = (C.c_int64 * 24)()
+    cnt = (C.c_int64 * 6)()
+    _check(_expected_q1(n_rows, seed, lo, hi, cnt), "expected_q1")
+    out = {}
+    for rf in range(3):
+        for ls in range(2):
+            g = rf * 2 + ls
+            sums = [(hi[g * 4 + k] << 64) | (lo[g * 4 + k] & ((1 << 64) - 1))
+                    for k in range(4)]
+            out[(rf, ls)] = sums + [cnt[g]]
+    return out
 
 
 def gen_lineitem(path, n_rows, seed=42, compression=COMP_LZ4, level=3, seg_kb=0,

@@ -111,6 +111,7 @@ struct cs_gpu_state {
     ColLoc *d_colloc = nullptr;
     AccCell *d_block = nullptr;
     AccCell *d_final = nullptr;
+    AccCell *d_final2 = nullptr;     /* wide-pass outputs of the final reduce */
     uint16_t *d_gkeys = nullptr;     /* grouped: per-block key tables */
     AccCell *d_gcells = nullptr;
     uint16_t *d_gfkeys = nullptr;    /* grouped: final merged groups */
@@ -1736,7 +1737,10 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_final_kernel(
     }
 }
 
-/* device-wide final reduce over block partials: one block, grid-stride */
+/* device-wide reduce over block partials; each launched block reduces a
+ * grid-strided slice into out[blockIdx * n_aggs + a]. Large grids (1B-row
+ * scans emit ~300k block partials) run this twice: wide pass then a
+ * single-block pass over the wide pass's outputs. */
 __global__ __launch_bounds__(AGG_BLOCK) void final_reduce_kernel(
     const AccCell *__restrict__ block_in, uint32_t n_blocks,
     AccCell *__restrict__ out, const AggParams params)
@@ -1747,8 +1751,9 @@ __global__ __launch_bounds__(AGG_BLOCK) void final_reduce_kernel(
     for (uint32_t a = 0; a < params.n_aggs; a++) {
         ThreadAcc r;
         acc_init(r, params.aggs[a].kind);
-        for (uint32_t b = threadIdx.x; b < n_blocks; b += AGG_BLOCK) {
-            const AccCell c = block_in[(uint64_t)b * params.n_aggs + a];
+        for (uint64_t b = (uint64_t)blockIdx.x * AGG_BLOCK + threadIdx.x;
+             b < n_blocks; b += (uint64_t)gridDim.x * AGG_BLOCK) {
+            const AccCell c = block_in[b * params.n_aggs + a];
             ThreadAcc t{c.lo, c.hi, c.f, c.cnt};
             acc_merge(r, t, params.aggs[a].kind);
         }
@@ -1760,10 +1765,32 @@ __global__ __launch_bounds__(AGG_BLOCK) void final_reduce_kernel(
             for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
                 acc_merge(f, lds[w][a], params.aggs[a].kind);
             AccCell c{f.lo, f.hi, f.f, f.cnt};
-            out[a] = c;
+            out[(uint64_t)blockIdx.x * params.n_aggs + a] = c;
         }
         __syncthreads();
     }
+}
+
+#define FINAL_WIDE_GRID 240
+
+/* launch the (possibly two-stage) final reduce: d_block[n_blocks] -> d_final */
+static int launch_final_reduce(cs_gpu_state *g, uint32_t n_blocks,
+                               const AggParams &p)
+{
+    if (n_blocks > 8192) {
+        hipLaunchKernelGGL(final_reduce_kernel, dim3(FINAL_WIDE_GRID),
+                           dim3(AGG_BLOCK), 0, g->stream,
+                           g->d_block, n_blocks, g->d_final2, p);
+        HIP_TRY(hipGetLastError());
+        hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0,
+                           g->stream, g->d_final2, FINAL_WIDE_GRID, g->d_final, p);
+        HIP_TRY(hipGetLastError());
+    } else {
+        hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0,
+                           g->stream, g->d_block, n_blocks, g->d_final, p);
+        HIP_TRY(hipGetLastError());
+    }
+    return CSTRIPE_OK;
 }
 
 /* closed-form decode of ONE canonical segment to a dense value buffer —
@@ -1816,6 +1843,7 @@ void csgpu_release(cstripe_scan *s)
     if (g->d_colloc) HIP_DROP(hipFree(g->d_colloc));
     if (g->d_block) HIP_DROP(hipFree(g->d_block));
     if (g->d_final) HIP_DROP(hipFree(g->d_final));
+    if (g->d_final2) HIP_DROP(hipFree(g->d_final2));
     if (g->d_gkeys) HIP_DROP(hipFree(g->d_gkeys));
     if (g->d_gcells) HIP_DROP(hipFree(g->d_gcells));
     if (g->d_gfkeys) HIP_DROP(hipFree(g->d_gfkeys));
@@ -1923,6 +1951,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     HIP_TRY(hipMalloc(&g->d_colloc, (uint64_t)g->n_groups * n_proj * sizeof(ColLoc)));
     HIP_TRY(hipMalloc(&g->d_block, max_blocks * MAX_AGGS * sizeof(AccCell)));
     HIP_TRY(hipMalloc(&g->d_final, MAX_AGGS * sizeof(AccCell)));
+    HIP_TRY(hipMalloc(&g->d_final2, (uint64_t)240 * MAX_AGGS * sizeof(AccCell)));
     HIP_TRY(hipMalloc(&g->d_error, sizeof(int)));
     g->data_bytes = data_bytes;
     g->scratch_bytes = scratch_bytes;
@@ -2464,9 +2493,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         else launchf(fused_agg_kernel<-1, -1>);
         HIP_TRY(hipGetLastError());
         HIP_TRY(hipEventRecord(g->ev1, g->stream));
-        hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
-                           g->d_block, g->n_tiles, g->d_final, p);
-        HIP_TRY(hipGetLastError());
+        { int _rc = launch_final_reduce(g, g->n_tiles, p); if (_rc != CSTRIPE_OK) return _rc; }
         HIP_TRY(hipEventRecord(g->ev2, g->stream));
 
         AccCell h_final[MAX_AGGS];
@@ -2559,9 +2586,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         else launch(filter_agg_kernel<-1, -1>);
         HIP_TRY(hipGetLastError());
     }
-    hipLaunchKernelGGL(final_reduce_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
-                       g->d_block, n_blocks, g->d_final, p);
-        HIP_TRY(hipGetLastError());
+    { int _rc = launch_final_reduce(g, n_blocks, p); if (_rc != CSTRIPE_OK) return _rc; }
     HIP_TRY(hipEventRecord(g->ev2, g->stream));
 
     AccCell h_final[MAX_AGGS];
