@@ -114,9 +114,13 @@ def main():
 
     # ---- setup (untimed): generate per-rank shards, open, prune, stage ----
     if args.shards == 0:
-        args.shards = 16 if args.rows >= 100_000_000 else 1
+        args.shards = 32 if args.rows >= 100_000_000 else 1
     base_seed = 42 + rank * args.shards
-    cache = os.environ.get("CSTRIPE_BENCH_DIR", "/tmp/cstripe_bench")
+    # default to tmpfs: the 1B dataset is ~21 GB/rank and the 8-GPU scale run
+    # needs 8 of them — /dev/shm is host-RAM-sized, container /tmp often isn't
+    default_cache = ("/dev/shm/cstripe_bench" if os.path.isdir("/dev/shm")
+                     else "/tmp/cstripe_bench")
+    cache = os.environ.get("CSTRIPE_BENCH_DIR", default_cache)
     os.makedirs(cache, exist_ok=True)
     seg_kb = -args.seg_bytes if args.seg_bytes else 0
     if args.shards > 1:
@@ -177,9 +181,9 @@ def main():
         result = step()
     per_step_est = (time.time() - t0) / max(1, args.warmup)
     if args.steps == 0:
-        # ~10 s of timed steps: long enough for the driver-side activity
+        # ~20 s of timed steps: long enough for the driver-side activity
         # sampler to observe the GPU busy, bounded for tiny workloads
-        args.steps = max(10, min(20000, int(10.0 / max(per_step_est, 1e-4))))
+        args.steps = max(10, min(40000, int(20.0 / max(per_step_est, 1e-4))))
 
     decode_ms = []
     agg_ms = []
