@@ -224,12 +224,12 @@ def main():
     avg_decode = sum(decode_ms) / len(decode_ms)
     avg_agg = sum(agg_ms) / len(agg_ms)
     if scan.last_fused:
-        dominant = "fused_agg_kernel"
+        dominant = "fused_agg_kernel" if not group_cols else "fused_grouped_kernel"
     elif avg_decode >= avg_agg:
-        dominant = "lz4_decode_kernel"
+        dominant = "lz4_decode_lane_kernel"
     else:
-        # all-dense scans (canonical writer default) run the multi-row kernel
-        dominant = "multi_agg_kernel"
+        # all-dense scans (canonical writer default) run the multi-row kernels
+        dominant = "multi_grouped_kernel" if group_cols else "multi_agg_kernel"
     dominant_ms = max(avg_decode, avg_agg)
     alg_bytes = args.rows * alg_bytes_per_row          # per launch (this rank)
     achieved_gbps = alg_bytes / (dominant_ms / 1e3) / 1e9 if dominant_ms > 0 else None

@@ -1317,6 +1317,192 @@ __device__ inline void acc_apply_atomic(ThreadAcc *cell, uint8_t kind, const Pre
     atomicAdd((unsigned long long *)&cell->cnt, 1ull);
 }
 
+/* =====================================================================
+ * Multi-row GROUPED kernel for ALL-DENSE chunks (TPC-H Q1 shape): the same
+ * R-row windowed loads as multi_agg_kernel (the scan is load-latency
+ * bound), folding into a per-WAVE 16-slot LDS group table with the
+ * carry-exact atomics. >16 distinct keys set device flag 8 and the host
+ * re-runs the 64-group fallback (grouped_agg_kernel).
+ * ===================================================================== */
+
+#define MGRP_SLOTS 16
+
+template <int NAGGS, int R>
+__global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
+    const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
+    const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
+    uint16_t *__restrict__ keys_out, AccCell *__restrict__ cells_out,
+    int *__restrict__ err, const GroupParams gp)
+{
+    const AggParams &params = gp.base;
+    const uint32_t n_aggs = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
+    const uint32_t wid = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    const uint32_t n_waves = AGG_BLOCK / WAVE;
+
+    __shared__ int wkeys[AGG_BLOCK / WAVE][MGRP_SLOTS];
+    extern __shared__ uint8_t mg_lds[];
+    ThreadAcc *wacc = (ThreadAcc *)mg_lds;   /* [wave][slot][agg] */
+
+    for (uint32_t i = threadIdx.x; i < n_waves * MGRP_SLOTS; i += AGG_BLOCK) {
+        wkeys[i / MGRP_SLOTS][i % MGRP_SLOTS] = -1;
+        for (uint32_t a = 0; a < n_aggs; a++)
+            acc_init(wacc[i * n_aggs + a], params.aggs[a].kind);
+    }
+    __syncthreads();
+
+    ThreadAcc *myacc = wacc + (size_t)wid * MGRP_SLOTS * n_aggs;
+    int *mykeys = wkeys[wid];
+    uint32_t ck0 = ~0u, ck1 = ~0u, cs0 = 0, cs1 = 0;   /* key->slot cache */
+
+    /* grid-stride over (chunk, tile) work items: tables persist across
+     * items and flush once, keeping per-block output buffers small */
+    for (uint32_t work = blockIdx.x; work < gp.n_work; work += gridDim.x) {
+    const uint32_t gid = work / params.tiles_per_group;
+    const uint32_t tile = work % params.tiles_per_group;
+    const GroupDesc g = groups[gid];
+    const ColLoc *cols = colloc + g.colbase;
+    const uint32_t row_start = tile * TILE_ROWS;
+    const uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
+
+    for (uint32_t row = row_start + R * threadIdx.x; row < row_end;
+         row += R * AGG_BLOCK) {
+        bool pv[R];
+        #pragma unroll
+        for (int k = 0; k < R; k++) pv[k] = row + k < row_end;
+        int last_proj = -1;
+        ValsR<R> q{};
+        #pragma unroll
+        for (uint32_t pp = 0; pp < params.n_preds; pp++) {
+            const PredD &pr = params.preds[pp];
+            if ((int)pr.proj != last_proj) {
+                col_multi<R>(data, scratch, cols[pr.proj], row, q);
+                last_proj = (int)pr.proj;
+            }
+            #pragma unroll
+            for (int k = 0; k < R; k++)
+                pv[k] = pv[k] & pred_eval(pr, q.v[k], q.fget(k));
+        }
+        bool any = false;
+        #pragma unroll
+        for (int k = 0; k < R; k++) any |= pv[k];
+        if (__ballot(any) == 0) continue;
+
+        /* group keys (k0 | k1<<8) for the R rows */
+        ValsR<R> k0, k1;
+        col_multi<R>(data, scratch, cols[gp.gproj[0]], row, k0);
+        if (gp.n_group_cols > 1)
+            col_multi<R>(data, scratch, cols[gp.gproj[1]], row, k1);
+        uint32_t key[R];
+        #pragma unroll
+        for (int k = 0; k < R; k++) {
+            key[k] = (uint32_t)k0.v[k] & 0xFF;
+            if (gp.n_group_cols > 1) key[k] |= ((uint32_t)k1.v[k] & 0xFF) << 8;
+        }
+        /* slot per row: register cache, else LDS claim (CAS linear probe) */
+        uint32_t slot[R];
+        #pragma unroll
+        for (int k = 0; k < R; k++) {
+            slot[k] = 0xFFFFFFFF;
+            if (!pv[k]) continue;
+            if (key[k] == ck0) slot[k] = cs0;
+            else if (key[k] == ck1) slot[k] = cs1;
+            else {
+                for (uint32_t s2 = 0; s2 < MGRP_SLOTS; s2++) {
+                    int old = atomicCAS(&mykeys[s2], -1, (int)key[k]);
+                    if (old == -1 || old == (int)key[k]) { slot[k] = s2; break; }
+                }
+                if (slot[k] == 0xFFFFFFFF) { atomicOr(err, 8); pv[k] = false; continue; }
+                ck1 = ck0; cs1 = cs0;
+                ck0 = key[k]; cs0 = slot[k];
+            }
+        }
+
+        /* operand loads once per agg column, contributions folded per row */
+        #pragma unroll
+        for (uint32_t a = 0; a < n_aggs; a++) {
+            const AggD &ag = params.aggs[a];
+            const uint8_t kind = ag.kind;
+            ValsR<R> qa, qb, qc;
+            if (kind != CSTRIPE_AGG_COUNT_STAR && kind != CSTRIPE_AGG_COUNT_COL) {
+                col_multi<R>(data, scratch, cols[ag.proj_a], row, qa);
+                if (kind >= CSTRIPE_AGG_SUM_PROD_I64)
+                    col_multi<R>(data, scratch, cols[ag.proj_b], row, qb);
+                if (kind == CSTRIPE_AGG_SUM_DISC_TAX_I64)
+                    col_multi<R>(data, scratch, cols[ag.proj_c], row, qc);
+            }
+            #pragma unroll
+            for (int k = 0; k < R; k++) {
+                if (!pv[k]) continue;
+                PrepAcc pc;
+                pc.valid = true; pc.f = 0.0; pc.lo = 0; pc.hi = 0;
+                switch (kind) {
+                    case CSTRIPE_AGG_COUNT_STAR:
+                    case CSTRIPE_AGG_COUNT_COL: pc.lo = 1; break;
+                    case CSTRIPE_AGG_SUM_I64:
+                    case CSTRIPE_AGG_MIN_I64:
+                    case CSTRIPE_AGG_MAX_I64:
+                        pc.lo = qa.v[k]; pc.hi = qa.v[k] < 0 ? -1 : 0; break;
+                    case CSTRIPE_AGG_SUM_F64:
+                    case CSTRIPE_AGG_MIN_F64:
+                    case CSTRIPE_AGG_MAX_F64:
+                        pc.f = qa.fget(k); break;
+                    case CSTRIPE_AGG_SUM_PROD_I64: {
+                        __int128 x = (__int128)qa.v[k] * qb.v[k];
+                        pc.lo = (int64_t)(uint64_t)x; pc.hi = (int64_t)(x >> 64); break;
+                    }
+                    case CSTRIPE_AGG_SUM_DISC_I64: {
+                        __int128 x = (__int128)qa.v[k] * (ag.one - qb.v[k]);
+                        pc.lo = (int64_t)(uint64_t)x; pc.hi = (int64_t)(x >> 64); break;
+                    }
+                    case CSTRIPE_AGG_SUM_DISC_TAX_I64: {
+                        __int128 x = (__int128)qa.v[k] * (ag.one - qb.v[k])
+                                     * (ag.one + qc.v[k]);
+                        pc.lo = (int64_t)(uint64_t)x; pc.hi = (int64_t)(x >> 64); break;
+                    }
+                    default: pc.valid = false; break;
+                }
+                acc_apply_atomic(&myacc[slot[k] * n_aggs + a], kind, pc);
+            }
+        }
+    }
+    }   /* work loop */
+    __syncthreads();
+
+    /* merge the block's wave tables into one compact list (as grouped_agg) */
+    if (threadIdx.x == 0) {
+        uint16_t *bk = keys_out + (size_t)blockIdx.x * (n_waves * MGRP_SLOTS);
+        AccCell *bc = cells_out + (size_t)blockIdx.x * (n_waves * MGRP_SLOTS) * n_aggs;
+        uint32_t n = 0;
+        for (uint32_t w = 0; w < n_waves; w++) {
+            for (uint32_t s2 = 0; s2 < MGRP_SLOTS; s2++) {
+                int kk = wkeys[w][s2];
+                if (kk < 0) continue;
+                uint32_t at = n;
+                for (uint32_t j = 0; j < n; j++) if (bk[j] == (uint16_t)kk) { at = j; break; }
+                ThreadAcc *src = wacc + ((size_t)(w * MGRP_SLOTS) + s2) * n_aggs;
+                if (at == n) {
+                    bk[n] = (uint16_t)kk;
+                    for (uint32_t a = 0; a < n_aggs; a++) {
+                        AccCell c{src[a].lo, src[a].hi, src[a].f, src[a].cnt};
+                        bc[(size_t)n * n_aggs + a] = c;
+                    }
+                    n++;
+                } else {
+                    for (uint32_t a = 0; a < n_aggs; a++) {
+                        AccCell c = bc[(size_t)at * n_aggs + a];
+                        ThreadAcc cur{c.lo, c.hi, c.f, c.cnt};
+                        acc_merge(cur, src[a], params.aggs[a].kind);
+                        AccCell o{cur.lo, cur.hi, cur.f, cur.cnt};
+                        bc[(size_t)at * n_aggs + a] = o;
+                    }
+                }
+            }
+        }
+        for (uint32_t j = n; j < n_waves * MGRP_SLOTS; j++) bk[j] = 0xFFFF;
+    }
+}
+
 template <int NAGGS>
 __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
@@ -2297,6 +2483,87 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             if (aggs[a].kind == CSTRIPE_AGG_SUM_F64 || aggs[a].kind == CSTRIPE_AGG_MIN_F64 ||
                 aggs[a].kind == CSTRIPE_AGG_MAX_F64)
                 int_aggs_g = false;
+        if (g->all_dense) {
+            /* multi-row grouped kernel (windowed loads + per-wave 16-slot
+             * LDS tables); falls back below on >16 distinct keys (flag 8) */
+            const uint32_t mg_per_block = n_waves * MGRP_SLOTS;
+            uint32_t mgrid = gp.n_work < GRP_GRID ? gp.n_work : GRP_GRID;
+            uint32_t mlds = n_waves * MGRP_SLOTS * n_aggs * (uint32_t)sizeof(ThreadAcc);
+            HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
+            HIP_TRY(hipEventRecord(g->ev0, g->stream));
+            { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
+            HIP_TRY(hipEventRecord(g->ev1, g->stream));
+            auto launchmg = [&](auto *kern) {
+                hipLaunchKernelGGL(kern, dim3(mgrid), dim3(AGG_BLOCK), mlds, g->stream,
+                                   g->d_data, g->d_scratch, g->d_groups, g->d_colloc,
+                                   g->d_gkeys, g->d_gcells, g->d_error, gp);
+            };
+            if (n_aggs == 5) launchmg(multi_grouped_kernel<5, 8>);
+            else if (n_aggs == 1) launchmg(multi_grouped_kernel<1, 8>);
+            else if (n_aggs == 2) launchmg(multi_grouped_kernel<2, 8>);
+            else if (n_aggs == 4) launchmg(multi_grouped_kernel<4, 8>);
+            else launchmg(multi_grouped_kernel<-1, 4>);
+            HIP_TRY(hipGetLastError());
+            hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
+                               g->d_gkeys, g->d_gcells, mgrid, mg_per_block,
+                               g->d_gfkeys, g->d_gfcells, g->d_gn, g->d_error, gp);
+            HIP_TRY(hipGetLastError());
+            HIP_TRY(hipEventRecord(g->ev2, g->stream));
+            int h_err = 0;
+            HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
+            HIP_TRY(hipStreamSynchronize(g->stream));
+            if (h_err & 4) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+            if (h_err == 0) {
+                s->last_fused = 0;   /* i8 key columns decode to scratch */
+                uint16_t h_keys[CSTRIPE_MAX_GROUPS];
+                std::vector<AccCell> h_cells((size_t)CSTRIPE_MAX_GROUPS * n_aggs);
+                uint32_t h_n = 0;
+                HIP_TRY(hipMemcpyAsync(h_keys, g->d_gfkeys, sizeof(h_keys), hipMemcpyDeviceToHost, g->stream));
+                HIP_TRY(hipMemcpyAsync(h_cells.data(), g->d_gfcells, h_cells.size() * sizeof(AccCell), hipMemcpyDeviceToHost, g->stream));
+                HIP_TRY(hipMemcpyAsync(&h_n, g->d_gn, 4, hipMemcpyDeviceToHost, g->stream));
+                HIP_TRY(hipStreamSynchronize(g->stream));
+                float ms_d = 0, ms_k = 0;
+                (void)hipEventElapsedTime(&ms_d, g->ev0, g->ev1);
+                (void)hipEventElapsedTime(&ms_k, g->ev1, g->ev2);
+                s->last_decode_ms = ms_d;
+                s->last_agg_ms = ms_k;
+                s->last_kernel_ms = ms_d + ms_k;
+                std::vector<uint32_t> order(h_n);
+                for (uint32_t i = 0; i < h_n; i++) order[i] = i;
+                std::sort(order.begin(), order.end(),
+                          [&](uint32_t a, uint32_t b) { return h_keys[a] < h_keys[b]; });
+                gr->n_groups = h_n;
+                for (uint32_t oi = 0; oi < h_n; oi++) {
+                    uint32_t i = order[oi];
+                    gr->keys[oi] = h_keys[i];
+                    for (uint32_t a = 0; a < n_aggs; a++) {
+                        const AccCell &cc2 = h_cells[(size_t)i * n_aggs + a];
+                        cstripe_partial o{};
+                        o.count = cc2.cnt;
+                        o.is_null = (cc2.cnt == 0) ? 1 : 0;
+                        switch (aggs[a].kind) {
+                            case CSTRIPE_AGG_COUNT_STAR:
+                            case CSTRIPE_AGG_COUNT_COL:
+                                o.i128_lo = cc2.cnt; o.is_null = 0; break;
+                            case CSTRIPE_AGG_SUM_F64:
+                            case CSTRIPE_AGG_MIN_F64:
+                            case CSTRIPE_AGG_MAX_F64:
+                                if (!o.is_null) o.f64 = cc2.f; break;
+                            case CSTRIPE_AGG_MIN_I64:
+                            case CSTRIPE_AGG_MAX_I64:
+                                if (!o.is_null) { o.i128_lo = cc2.lo; o.i128_hi = cc2.lo < 0 ? -1 : 0; }
+                                break;
+                            default:
+                                if (!o.is_null) { o.i128_lo = cc2.lo; o.i128_hi = cc2.hi; }
+                                break;
+                        }
+                        out[(size_t)oi * n_aggs + a] = o;
+                    }
+                }
+                return CSTRIPE_OK;
+            }
+            /* flag 8: >16 distinct keys — fall through to the 64-group path */
+        }
         if (g->fusable_mixed && int_aggs_g && g->d_tiles2) {
             /* fused grouped: decode tile to LDS + atomic group table; falls
              * back below when >16 distinct groups (device flag 8) */
