@@ -104,11 +104,16 @@ def main():
     dist = world > 1
 
     import torch
+    rccl_comm = None
     if dist:
         import torch.distributed as td
-        from citus_amd.dist import all_gather_combine
+        from citus_amd.dist import make_rccl_comm
         torch.cuda.set_device(local_rank)
         td.init_process_group("nccl")
+        # the combine data path runs through the C ABI (cagg_combine_rccl,
+        # librccl); torch.distributed only bootstraps the unique id and
+        # handles the timing all-reduce
+        rccl_comm = make_rccl_comm(local_rank)
 
     comp = {"lz4": ca.COMP_LZ4, "none": ca.COMP_NONE, "zstd": ca.COMP_ZSTD}[args.compression]
 
@@ -167,12 +172,12 @@ def main():
         if group_cols:
             res = scan.agg_grouped(aggs, group_cols)
             if dist:
-                from citus_amd.dist import all_gather_combine_grouped
-                return all_gather_combine_grouped(aggs, res, device="cuda")
+                from citus_amd.dist import all_gather_combine_grouped_rccl
+                return all_gather_combine_grouped_rccl(rccl_comm, aggs, res)
             return {k: ca.combine(aggs, [parts]) for k, parts in res.items()}
         parts = scan.agg(aggs)
         if dist:
-            return all_gather_combine(aggs, parts, device="cuda")
+            return rccl_comm.combine(aggs, parts)
         return ca.combine(aggs, [parts])
 
     # ---- warmup (also calibrates auto step count) ----
@@ -213,6 +218,7 @@ def main():
         scan.end()
         reader.close()
         if dist:
+            rccl_comm.close()
             td.destroy_process_group()
         return
 
@@ -365,6 +371,7 @@ def main():
     scan.end()
     reader.close()
     if dist:
+        rccl_comm.close()
         td.destroy_process_group()
 
 

@@ -124,6 +124,15 @@ _combine = _sig("cagg_combine", C.c_int,
                 [C.POINTER(AggSpec), C.c_uint32, C.POINTER(Partial), C.c_uint32,
                  C.POINTER(Partial)])
 _gpu_available = _sig("cstripe_gpu_available", C.c_int, [])
+_comm_uid = _sig("cagg_comm_unique_id", C.c_int, [C.c_uint8 * 128])
+_comm_init = _sig("cagg_comm_init", C.c_int,
+                  [C.POINTER(C.c_void_p), C.c_int, C.c_int, C.c_uint8 * 128, C.c_int])
+_comm_destroy = _sig("cagg_comm_destroy", None, [C.c_void_p])
+_comm_allgather = _sig("cagg_allgather", C.c_int,
+                       [C.c_void_p, C.c_void_p, C.c_uint64, C.c_void_p])
+_combine_rccl = _sig("cagg_combine_rccl", C.c_int,
+                     [C.c_void_p, C.POINTER(AggSpec), C.c_uint32,
+                      C.POINTER(Partial), C.POINTER(Partial)])
 _gen_lineitem = _sig("csbench_gen_lineitem", C.c_int,
                      [C.c_char_p, C.c_uint64, C.c_uint64, C.c_int, C.c_int, C.c_int,
                       C.c_uint64, C.c_uint32])
@@ -417,6 +426,52 @@ class Scan:
 
     def rewind(self):
         _check(_rewind(self._h), "rewind")
+
+
+class RcclComm:
+    """C-ABI RCCL communicator (cagg_comm): the data-path collective of the
+    combine step runs in host C over librccl; the launcher only distributes
+    the 128-byte unique id (ncclCommInitRank's own bootstrap contract)."""
+
+    def __init__(self, n_ranks, rank, uid_bytes, device=-1):
+        uid = (C.c_uint8 * 128)(*uid_bytes)
+        h = C.c_void_p()
+        _check(_comm_init(C.byref(h), n_ranks, rank, uid, device), "cagg_comm_init")
+        self._h = h
+        self.rank = rank
+        self.n_ranks = n_ranks
+
+    @staticmethod
+    def unique_id():
+        uid = (C.c_uint8 * 128)()
+        _check(_comm_uid(uid), "cagg_comm_unique_id")
+        return bytes(uid)
+
+    def close(self):
+        if self._h:
+            _comm_destroy(self._h)
+            self._h = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+    def allgather_bytes(self, data):
+        src = (C.c_uint8 * len(data)).from_buffer_copy(data)
+        dst = (C.c_uint8 * (len(data) * self.n_ranks))()
+        _check(_comm_allgather(self._h, src, len(data), dst), "cagg_allgather")
+        return bytes(dst)
+
+    def combine(self, aggs, parts):
+        """cagg_combine_rccl: gather + strict merge, in C, over RCCL."""
+        a = make_aggs(aggs)
+        local = (Partial * len(parts))(*parts)
+        out = (Partial * len(parts))()
+        _check(_combine_rccl(self._h, a, len(aggs), local, out),
+               "cagg_combine_rccl")
+        return list(out)
 
 
 def combine(aggs, parts_list):

@@ -20,7 +20,65 @@ import ctypes as C
 import torch
 import torch.distributed as td
 
-from . import Partial, combine as _combine
+from . import Partial, RcclComm, combine as _combine
+
+
+def make_rccl_comm(device=-1):
+    """Create the C-ABI cagg_comm for this process group: rank 0 makes the
+    128-byte unique id, torch.distributed only BROADCASTS it (control plane);
+    the data-path collective then runs in host C over librccl (the north
+    star's 'host code stays C' contract for the combine hop)."""
+    rank = td.get_rank()
+    world = td.get_world_size()
+    if rank == 0:
+        uid = list(RcclComm.unique_id())
+    else:
+        uid = [0] * 128
+    box = [bytes(uid)]
+    td.broadcast_object_list(box, src=0)
+    return RcclComm(world, rank, box[0], device=device)
+
+
+def all_gather_combine_rccl(comm, aggs, parts):
+    """The combine surface over the C ABI: cagg_combine_rccl (ncclAllGather
+    of the partial block + strict local merge, both in C)."""
+    return comm.combine(aggs, parts)
+
+
+def all_gather_combine_grouped_rccl(comm, aggs, resdict, max_groups=64):
+    """Grouped combine over the C-ABI collective: fixed 64-slot
+    (keys, partials) block per rank through cagg_allgather, merged locally
+    with cagg_combine per distinct key."""
+    n_aggs = len(aggs)
+    keys = sorted(resdict)
+    assert len(keys) <= max_groups
+    import struct
+    kbuf = bytearray(struct.pack(f"<{max_groups}i", *([-1] * max_groups)))
+    for g, k in enumerate(keys):
+        struct.pack_into("<i", kbuf, g * 4, k[0] | (k[1] << 8))
+    flat = (Partial * (max_groups * n_aggs))()
+    for g, k in enumerate(keys):
+        for a in range(n_aggs):
+            flat[g * n_aggs + a] = resdict[k][a]
+    blob = bytes(kbuf) + C.string_at(flat, C.sizeof(flat))
+    allblob = comm.allgather_bytes(blob)
+    bs = len(blob)
+    rank_tables = []
+    for r in range(comm.n_ranks):
+        chunk = allblob[r * bs:(r + 1) * bs]
+        ks = struct.unpack(f"<{max_groups}i", chunk[:max_groups * 4])
+        parts = (Partial * (max_groups * n_aggs)).from_buffer_copy(chunk[max_groups * 4:])
+        table = {}
+        for i in range(max_groups):
+            if ks[i] < 0:
+                continue
+            table[(ks[i] & 0xFF, ks[i] >> 8)] = [parts[i * n_aggs + a]
+                                                 for a in range(n_aggs)]
+        rank_tables.append(table)
+    all_keys = sorted(set().union(*[t.keys() for t in rank_tables]))
+    null_row = [Partial(is_null=1) for _ in range(n_aggs)]
+    return {k: _combine(aggs, [t.get(k, null_row) for t in rank_tables])
+            for k in all_keys}
 
 _PARTIAL_BYTES = C.sizeof(Partial)
 

@@ -302,6 +302,35 @@ int cagg_combine(const cstripe_agg_spec *aggs, uint32_t n_aggs,
                  const cstripe_partial *parts, uint32_t n_parts,
                  cstripe_partial *out);
 
+/* =================== RCCL combine (device collective) ===================
+ * The north star replaces the coordinator merge with a collective over
+ * xGMI. The data-path hop is ncclAllGather of each rank's partial block
+ * (RCCL over xGMI), followed by the same strict cagg_combine merge locally
+ * on every rank: int128 carries cannot ride a sum collective, and gathering
+ * is byte-faithful to the coordinator receiving one partial row per shard
+ * (aggregate_utils.c:820-1021; adaptive_executor.c:775-882). Host code stays
+ * C; the launcher only has to distribute the 128-byte unique id (the same
+ * bootstrap contract as ncclCommInitRank). */
+typedef struct cagg_comm cagg_comm;
+#define CAGG_UNIQUE_ID_BYTES 128
+
+int cagg_comm_unique_id(uint8_t id[CAGG_UNIQUE_ID_BYTES]);   /* rank 0 makes */
+int cagg_comm_init(cagg_comm **out, int n_ranks, int rank,
+                   const uint8_t id[CAGG_UNIQUE_ID_BYTES], int device);
+void cagg_comm_destroy(cagg_comm *c);
+int cagg_comm_rank(const cagg_comm *c);
+int cagg_comm_size(const cagg_comm *c);
+
+/* all-gather equal-size byte blocks from every rank (device-staged
+ * internally; dst holds n_ranks * bytes, rank-major) */
+int cagg_allgather(cagg_comm *c, const void *src, uint64_t bytes, void *dst);
+
+/* the combine surface over RCCL: gather every rank's partial block, then
+ * cagg_combine locally; every rank returns the final merged row */
+int cagg_combine_rccl(cagg_comm *c, const cstripe_agg_spec *aggs,
+                      uint32_t n_aggs, const cstripe_partial *local,
+                      cstripe_partial *out);
+
 /* =================== misc =================== */
 const char *cstripe_errmsg(void);
 int cstripe_gpu_available(void);     /* 1 if a HIP device is visible */

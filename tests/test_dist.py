@@ -135,3 +135,41 @@ def test_two_rank_grouped_combine(tmp_path):
     line = [l for l in out.stdout.splitlines() if l.startswith("RESULT ")]
     got = json.loads(line[0][len("RESULT "):])
     assert got == expect
+
+
+@pytest.mark.gpu
+def test_rccl_combine_single_rank():
+    """The C-ABI RCCL combine path executed on hardware: comm init, the
+    ncclAllGather byte round trip, and cagg_combine_rccl semantics at
+    world size 1 (the 8-GPU run is the driver's; 2-rank semantics are
+    covered by the gloo tests above)."""
+    import citus_amd as ca
+    assert ca.gpu_available()
+    uid = ca.RcclComm.unique_id()
+    with ca.RcclComm(1, 0, uid, device=0) as comm:
+        blob = b"cstripe-rccl-roundtrip" * 11
+        assert comm.allgather_bytes(blob) == blob
+        aggs = [(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_STAR, -1), (ca.AGG_MIN_I64, 0)]
+        parts = [ca.Partial(i128_lo=-5, i128_hi=-1, count=3),
+                 ca.Partial(i128_lo=3, count=3),
+                 ca.Partial(i128_lo=-7, i128_hi=-1, count=3)]
+        out = comm.combine(aggs, parts)
+        assert out[0].i128 == -5
+        assert out[1].count == 3
+        assert out[2].i128 == -7
+
+
+@pytest.mark.gpu
+def test_rccl_grouped_combine_single_rank():
+    import citus_amd as ca
+    from citus_amd import dist as cdist
+    uid = ca.RcclComm.unique_id()
+    with ca.RcclComm(1, 0, uid, device=0) as comm:
+        aggs = [(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_STAR, -1)]
+        res = {(0, 1): [ca.Partial(i128_lo=10, count=2), ca.Partial(i128_lo=2, count=2)],
+               (2, 0): [ca.Partial(i128_lo=-4, i128_hi=-1, count=1),
+                        ca.Partial(i128_lo=1, count=1)]}
+        out = cdist.all_gather_combine_grouped_rccl(comm, aggs, res)
+        assert set(out) == {(0, 1), (2, 0)}
+        assert out[(0, 1)][0].i128 == 10
+        assert out[(2, 0)][0].i128 == -4
