@@ -50,7 +50,8 @@ class Options(C.Structure):
 
 class Pred(C.Structure):
     _fields_ = [("column", C.c_uint32), ("op", C.c_uint32),
-                ("ival", C.c_int64), ("fval", C.c_double)]
+                ("ival", C.c_int64), ("fval", C.c_double),
+                ("or_group", C.c_uint32), ("_pad", C.c_uint32)]
 
 
 class AggSpec(C.Structure):
@@ -303,6 +304,9 @@ class Reader:
 
 
 def make_preds(preds):
+    """preds: (column, op, value) or (column, op, value, or_group) tuples.
+    A nonzero or_group groups predicates into a disjunction (CNF; see
+    include/cstripe.h)."""
     arr = (Pred * max(1, len(preds)))()
     for i, p in enumerate(preds):
         arr[i].column = p[0]
@@ -314,6 +318,7 @@ def make_preds(preds):
         else:
             arr[i].ival = int(v)
             arr[i].fval = 0.0
+        arr[i].or_group = p[3] if len(p) > 3 else 0
     return arr
 
 

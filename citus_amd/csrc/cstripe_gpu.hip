@@ -33,7 +33,7 @@
 #define WAVE 64
 #define AGG_BLOCK 256
 #define TILE_ROWS 4096
-#define MAX_PREDS 8
+#define MAX_PREDS CSTRIPE_MAX_PREDS
 #define MAX_AGGS 12
 #define MAX_PROJ 16
 
@@ -72,6 +72,9 @@ struct PredD {
     uint16_t proj;
     uint8_t  op;
     uint8_t  is_float;
+    uint8_t  gend;       /* 1 = last member of its OR group (groups are
+                          * contiguous; pass &= OR of the group's members) */
+    uint8_t  pad[7];
     int64_t  ival;
     double   fval;
 };
@@ -736,11 +739,12 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
     for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
 
     for (uint32_t row = row_start + threadIdx.x; row < row_end; row += AGG_BLOCK) {
-        bool pass = true;
+        bool pass = true, gv = false;
         int last_proj = -1;
         int64_t liv = 0; double lfv = 0; bool lok = false;
         /* no short-circuit: loads stay control-independent so they issue
-         * back-to-back and pipeline instead of chaining load->wait->branch */
+         * back-to-back and pipeline instead of chaining load->wait->branch.
+         * contiguous OR groups accumulate into gv, folded at group end */
         #pragma unroll
         for (uint32_t p = 0; p < n_preds; p++) {
             const PredD &pr = params.preds[p];
@@ -748,7 +752,8 @@ __global__ __launch_bounds__(AGG_BLOCK) void filter_agg_kernel(
                 lok = col_value(data, scratch, rank, cols[pr.proj], row, liv, lfv);
                 last_proj = (int)pr.proj;
             }
-            pass = pass & (lok && pred_eval(pr, liv, lfv));
+            gv = gv | (lok && pred_eval(pr, liv, lfv));
+            if (pr.gend) { pass = pass & gv; gv = false; }
         }
         if (!pass) continue;
         #pragma unroll
@@ -830,13 +835,14 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_agg_kernel(
     };
 
     for (uint32_t row = tid; row < t.row_count; row += AGG_BLOCK) {
-        bool pass = true;
+        bool pass = true, gv = false;
         int last_proj = -1;
         int64_t liv = 0;
         for (uint32_t p = 0; p < n_preds; p++) {
             const PredD &pr = params.preds[p];
             if ((int)pr.proj != last_proj) { liv = val(pr.proj, row); last_proj = (int)pr.proj; }
-            pass = pass & pred_eval(pr, liv, 0.0);   /* cols are i64: is_float==0 */
+            gv = gv | pred_eval(pr, liv, 0.0);       /* cols are i64: is_float==0 */
+            if (pr.gend) { pass = pass & gv; gv = false; }
         }
         if (!pass) continue;
         #pragma unroll
@@ -1128,9 +1134,9 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_agg_kernel(
 
     for (uint32_t row = row_start + R * threadIdx.x; row < row_end;
          row += R * AGG_BLOCK) {
-        bool pv[R];
+        bool pv[R], gv[R];
         #pragma unroll
-        for (int k = 0; k < R; k++) pv[k] = row + k < row_end;
+        for (int k = 0; k < R; k++) { pv[k] = row + k < row_end; gv[k] = false; }
         int last_proj = -1;
         ValsR<R> q{};
         #pragma unroll
@@ -1142,7 +1148,11 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_agg_kernel(
             }
             #pragma unroll
             for (int k = 0; k < R; k++)
-                pv[k] = pv[k] & pred_eval(pr, q.v[k], q.fget(k));
+                gv[k] = gv[k] | pred_eval(pr, q.v[k], q.fget(k));
+            if (pr.gend) {
+                #pragma unroll
+                for (int k = 0; k < R; k++) { pv[k] = pv[k] & gv[k]; gv[k] = false; }
+            }
         }
         bool any = false;
         #pragma unroll
@@ -1367,9 +1377,9 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
 
     for (uint32_t row = row_start + R * threadIdx.x; row < row_end;
          row += R * AGG_BLOCK) {
-        bool pv[R];
+        bool pv[R], gv[R];
         #pragma unroll
-        for (int k = 0; k < R; k++) pv[k] = row + k < row_end;
+        for (int k = 0; k < R; k++) { pv[k] = row + k < row_end; gv[k] = false; }
         int last_proj = -1;
         ValsR<R> q{};
         #pragma unroll
@@ -1381,7 +1391,11 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
             }
             #pragma unroll
             for (int k = 0; k < R; k++)
-                pv[k] = pv[k] & pred_eval(pr, q.v[k], q.fget(k));
+                gv[k] = gv[k] | pred_eval(pr, q.v[k], q.fget(k));
+            if (pr.gend) {
+                #pragma unroll
+                for (int k = 0; k < R; k++) { pv[k] = pv[k] & gv[k]; gv[k] = false; }
+            }
         }
         bool any = false;
         #pragma unroll
@@ -1551,13 +1565,15 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
             if (pass) {
                 int last_proj = -1;
                 int64_t liv = 0; double lfv = 0; bool lok = false;
+                bool gv = false;
                 for (uint32_t p = 0; p < params.n_preds; p++) {
                     const PredD &pr = params.preds[p];
                     if ((int)pr.proj != last_proj) {
                         lok = col_value(data, scratch, rank, cols[pr.proj], row, liv, lfv);
                         last_proj = (int)pr.proj;
                     }
-                    pass = pass & (lok && pred_eval(pr, liv, lfv));
+                    gv = gv | (lok && pred_eval(pr, liv, lfv));
+                    if (pr.gend) { pass = pass & gv; gv = false; }
                 }
             }
             uint32_t key = 0;
@@ -1743,13 +1759,14 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_grouped_kernel(
 
         /* phase 2: filter + group + accumulate */
         for (uint32_t row = tid; row < t.row_count; row += AGG_BLOCK) {
-            bool pass = true;
+            bool pass = true, gv = false;
             int last_proj = -1;
             int64_t liv = 0;
             for (uint32_t p = 0; p < params.n_preds; p++) {
                 const PredD &pr = params.preds[p];
                 if ((int)pr.proj != last_proj) { liv = val(pr.proj, row); last_proj = (int)pr.proj; }
-                pass = pass & pred_eval(pr, liv, 0.0);
+                gv = gv | pred_eval(pr, liv, 0.0);
+                if (pr.gend) { pass = pass & gv; gv = false; }
             }
             if (!pass) continue;
             uint32_t key = 0;
@@ -2429,6 +2446,8 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         p.preds[i].op = (uint8_t)q.op;
         uint8_t t = r->cols[q.column].type;
         p.preds[i].is_float = (t == CSTRIPE_F32 || t == CSTRIPE_F64) ? 1 : 0;
+        p.preds[i].gend = (i + 1 == p.n_preds ||
+                           s->preds[i + 1].or_group != q.or_group) ? 1 : 0;
         p.preds[i].ival = q.ival;
         p.preds[i].fval = q.fval;
     }

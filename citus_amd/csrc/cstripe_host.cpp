@@ -857,26 +857,49 @@ extern "C" cstripe_scan *cstripe_scan_begin(cstripe_reader *r, uint64_t cols_mas
     auto *s = new cstripe_scan();
     s->r = r;
     s->cols_mask = cols_mask;
+    if (n_preds > CSTRIPE_MAX_PREDS) { cs_set_err("too many predicates (max %d)", CSTRIPE_MAX_PREDS); delete s; return nullptr; }
     for (uint32_t i = 0; i < n_preds; i++) {
         if (preds[i].column >= r->cols.size()) { cs_set_err("pred column out of range"); delete s; return nullptr; }
         s->preds.push_back(preds[i]);
+        /* normalize: every predicate belongs to exactly one OR group;
+         * standalone conjuncts each get a private id */
+        if (s->preds.back().or_group == 0)
+            s->preds.back().or_group = 0x40000000u + i;
         s->cols_mask |= 1ull << preds[i].column;   /* pred columns must be read */
     }
+    /* contiguous groups (stable: original order kept within a group) */
+    std::stable_sort(s->preds.begin(), s->preds.end(),
+                     [](const cstripe_pred &a, const cstripe_pred &b) {
+                         return a.or_group < b.or_group;
+                     });
     if (s->cols_mask == 0)
         s->cols_mask = 1;   /* pure count(*): still scan one column's chunks */
     for (uint32_t si = 0; si < r->stripes.size(); si++) {
         const cs_stripe_info &st = r->stripes[si];
         for (uint32_t k = 0; k < st.meta.chunk_count; k++) {
+            /* chunk removed iff some OR group is WHOLLY refuted — the CNF
+             * equivalent of predicate_refuted_by over the reference's
+             * AND/OR clause tree (SelectedChunkMask,
+             * columnar_reader.c:1132-1187; OR recursion
+             * columnar_customscan.c:770-829). Count a removed chunk once. */
             bool selected = true;
-            for (const auto &p : s->preds) {
-                const csf_skipnode &nd = st.nodes[p.column][k].n;
-                /* all-NULL chunks have no min/max and are never refuted
-                 * (columnar_reader.c:1160-1166) */
-                if (!nd.has_min_max) continue;
-                if (pred_refutes(p, r->cols[p.column].type, nd.min_i, nd.max_i)) {
-                    selected = false;
-                    break;      /* count once (:1178-1182) */
+            size_t i = 0;
+            while (i < s->preds.size() && selected) {
+                size_t j = i;
+                bool group_refuted = true;
+                while (j < s->preds.size() &&
+                       s->preds[j].or_group == s->preds[i].or_group) {
+                    const cstripe_pred &p = s->preds[j];
+                    const csf_skipnode &nd = st.nodes[p.column][k].n;
+                    /* all-NULL chunks have no min/max and are never refuted
+                     * (columnar_reader.c:1160-1166) */
+                    if (!nd.has_min_max ||
+                        !pred_refutes(p, r->cols[p.column].type, nd.min_i, nd.max_i))
+                        group_refuted = false;
+                    j++;
                 }
+                if (group_refuted) selected = false;
+                i = j;
             }
             if (selected) s->sel.push_back({si, k});
             else s->chunk_groups_filtered++;

@@ -52,10 +52,10 @@
 extern "C" {
 #endif
 
-#define CSTRIPE_ABI_VERSION 1
+#define CSTRIPE_ABI_VERSION 2
 
 /* capacity limits of one scan */
-#define CSTRIPE_MAX_PREDS 8
+#define CSTRIPE_MAX_PREDS 16
 #define CSTRIPE_MAX_AGGS  12
 
 /* ---- error codes ---- */
@@ -135,10 +135,22 @@ typedef struct cstripe_options {
 void cstripe_default_options(cstripe_options *opts);
 
 /* ---- predicates ----
- * Same family the reference pushes down: `Var op pseudo-const`, AND of those
- * (ExtractPushdownClause, columnar_customscan.c:712-952). Multiple predicates
- * in an array are implicitly ANDed. Values are given in the column's physical
- * representation (i64 for integer/fixed-point/date columns, f64 for floats). */
+ * The family the reference pushes down: `Var op pseudo-const` atoms combined
+ * with AND and OR (ExtractPushdownClause, columnar_customscan.c:712-952,
+ * OR recursion :770-829). The ABI carries the filter in CNF: predicates
+ * sharing a nonzero or_group form one DISJUNCTION; the groups (and all
+ * or_group==0 predicates, each its own group) are ANDed. Any AND/OR tree of
+ * atoms distributes into this form — e.g. the reference test's
+ * `(a>1000 AND a<10000) OR (a>20000 AND a<50000)` becomes the four groups
+ * (a>1000 | a>20000)(a>1000 | a<50000)(a<10000 | a>20000)(a<10000 | a<50000)
+ * — and chunk refutation over the CNF is equivalent to the reference's
+ * predicate_refuted_by on the original tree (a group prunes a chunk only if
+ * EVERY member refutes its [min,max]). Trees that exceed CSTRIPE_MAX_PREDS
+ * after distribution are not pushdownable through this ABI, mirroring the
+ * reference leaving such clauses to ExecQual. Values are given in the
+ * column's physical representation (i64 for integer/fixed-point/date
+ * columns, f64 for floats). NULL operands fail the atom (SQL semantics at
+ * the top-level filter). */
 typedef enum cstripe_predop {
     CSTRIPE_PRED_LT = 0, CSTRIPE_PRED_LE, CSTRIPE_PRED_GT, CSTRIPE_PRED_GE,
     CSTRIPE_PRED_EQ, CSTRIPE_PRED_NE,
@@ -149,6 +161,8 @@ typedef struct cstripe_pred {
     uint32_t    op;        /* cstripe_predop */
     int64_t     ival;      /* comparison constant for integer columns */
     double      fval;      /* comparison constant for float columns */
+    uint32_t    or_group;  /* 0 = standalone conjunct; same nonzero id = OR'd */
+    uint32_t    _pad;
 } cstripe_pred;
 
 /* ---- aggregates ----

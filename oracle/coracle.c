@@ -392,6 +392,74 @@ static int o_pred_eval(const cstripe_pred *p, uint8_t is_float, int64_t iv, doub
     }
 }
 
+/* ---- OR-group (CNF) predicate helpers: predicates sharing a nonzero
+ * or_group are a disjunction; groups are ANDed (ExtractPushdownClause's
+ * AND/OR recursion, columnar_customscan.c:770-829, distributed to CNF at
+ * the ABI). Standalone predicates get private groups; groups made
+ * contiguous by a stable sort. ---- */
+static void o_norm_preds(cstripe_pred *dst, const cstripe_pred *src, uint32_t n)
+{
+    for (uint32_t i = 0; i < n; i++) {
+        dst[i] = src[i];
+        if (dst[i].or_group == 0) dst[i].or_group = 0x40000000u + i;
+    }
+    /* insertion sort (n <= 16), stable */
+    for (uint32_t i = 1; i < n; i++) {
+        cstripe_pred key = dst[i];
+        int32_t j = (int32_t)i - 1;
+        while (j >= 0 && dst[j].or_group > key.or_group) {
+            dst[j + 1] = dst[j];
+            j--;
+        }
+        dst[j + 1] = key;
+    }
+}
+
+/* chunk survives unless some OR group is WHOLLY refuted by min/max */
+static int o_chunk_selected(const oracle_table *t, const ostripe *st, uint32_t k,
+                            const cstripe_pred *preds, uint32_t n_preds)
+{
+    uint32_t i = 0;
+    while (i < n_preds) {
+        uint32_t j = i;
+        int group_refuted = 1;
+        while (j < n_preds && preds[j].or_group == preds[i].or_group) {
+            const onode *nd = &st->nodes[(size_t)preds[j].column * st->meta.chunk_count + k];
+            if (!nd->n.has_min_max ||
+                !o_pred_refutes(&preds[j], t->cols[preds[j].column].type,
+                                nd->n.min_i, nd->n.max_i))
+                group_refuted = 0;
+            j++;
+        }
+        if (group_refuted) return 0;
+        i = j;
+    }
+    return 1;
+}
+
+/* row filter: AND over groups of OR over members; NULL operand fails the
+ * atom (SQL three-valued logic collapses to keep/drop at the top filter) */
+static int o_row_pass(const oracle_table *t, const ochunkcol *cc,
+                      const cstripe_pred *preds, uint32_t n_preds, uint32_t row)
+{
+    uint32_t i = 0;
+    while (i < n_preds) {
+        uint32_t j = i;
+        int gv = 0;
+        while (j < n_preds && preds[j].or_group == preds[i].or_group) {
+            const cstripe_pred *pr = &preds[j];
+            int64_t iv; double fv;
+            uint8_t ty = t->cols[pr->column].type;
+            if (o_get(&cc[pr->column], row, &iv, &fv))
+                gv |= o_pred_eval(pr, ty == CSTRIPE_F32 || ty == CSTRIPE_F64, iv, fv);
+            j++;
+        }
+        if (!gv) return 0;
+        i = j;
+    }
+    return 1;
+}
+
 /* The whole-scan oracle: stripe loop -> pruning -> chunk decode -> per-row
  * qual eval -> agg transition. Mirrors the call stack of SURVEY.md §3.1.
  * group_cols: optional (n_group_cols may be 0). For grouped mode, out must
@@ -408,6 +476,10 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
     uint32_t ncols = t->head.column_count;
     uint64_t mask = cols_mask;
     for (uint32_t i = 0; i < n_preds; i++) mask |= 1ull << preds[i].column;
+    cstripe_pred npreds[CSTRIPE_MAX_PREDS];
+    if (n_preds > CSTRIPE_MAX_PREDS) return -5;
+    o_norm_preds(npreds, preds, n_preds);
+    preds = npreds;
     for (uint32_t i = 0; i < n_aggs; i++) {
         if (aggs[i].col_a >= 0) mask |= 1ull << aggs[i].col_a;
         if (aggs[i].col_b >= 0) mask |= 1ull << aggs[i].col_b;
@@ -435,15 +507,8 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
     for (uint32_t s = 0; s < t->head.stripe_count && rc == 0; s++) {
         const ostripe *st = &t->stripes[s];
         for (uint32_t k = 0; k < st->meta.chunk_count && rc == 0; k++) {
-            /* SelectedChunkMask */
-            int selected = 1;
-            for (uint32_t p = 0; p < n_preds; p++) {
-                const onode *nd = &st->nodes[(size_t)preds[p].column * st->meta.chunk_count + k];
-                if (!nd->n.has_min_max) continue;
-                if (o_pred_refutes(&preds[p], t->cols[preds[p].column].type,
-                                   nd->n.min_i, nd->n.max_i)) { selected = 0; break; }
-            }
-            if (!selected) { filtered++; continue; }
+            /* SelectedChunkMask (OR groups: see o_chunk_selected) */
+            if (!o_chunk_selected(t, st, k, preds, n_preds)) { filtered++; continue; }
 
             uint32_t rows = st->group_rows[k];
             for (uint32_t c = 0; c < ncols; c++) {
@@ -453,15 +518,7 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
             if (rc) break;
 
             for (uint32_t row = 0; row < rows; row++) {
-                int pass = 1;
-                for (uint32_t p = 0; p < n_preds && pass; p++) {
-                    const cstripe_pred *pr = &preds[p];
-                    int64_t iv; double fv;
-                    uint8_t ty = t->cols[pr->column].type;
-                    if (!o_get(&cc[pr->column], row, &iv, &fv)) pass = 0;
-                    else pass = o_pred_eval(pr, ty == CSTRIPE_F32 || ty == CSTRIPE_F64, iv, fv);
-                }
-                if (!pass) continue;
+                if (!o_row_pass(t, cc, preds, n_preds, row)) continue;
 
                 oacc *grp = accs;
                 if (n_group_cols) {
@@ -684,6 +741,10 @@ int oracle_scan_agg_mt(oracle_table *t, const cstripe_pred *preds, uint32_t n_pr
 {
     uint64_t mask = 0;
     for (uint32_t i = 0; i < n_preds; i++) mask |= 1ull << preds[i].column;
+    cstripe_pred npreds[CSTRIPE_MAX_PREDS];
+    if (n_preds > CSTRIPE_MAX_PREDS) return -5;
+    o_norm_preds(npreds, preds, n_preds);
+    preds = npreds;
     for (uint32_t i = 0; i < n_aggs; i++) {
         if (aggs[i].col_a >= 0) mask |= 1ull << aggs[i].col_a;
         if (aggs[i].col_b >= 0) mask |= 1ull << aggs[i].col_b;
@@ -695,14 +756,7 @@ int oracle_scan_agg_mt(oracle_table *t, const cstripe_pred *preds, uint32_t n_pr
     for (uint32_t s = 0; s < t->head.stripe_count; s++) {
         const ostripe *st = &t->stripes[s];
         for (uint32_t k = 0; k < st->meta.chunk_count; k++) {
-            int selected = 1;
-            for (uint32_t p = 0; p < n_preds; p++) {
-                const onode *nd = &st->nodes[(size_t)preds[p].column * st->meta.chunk_count + k];
-                if (!nd->n.has_min_max) continue;
-                if (o_pred_refutes(&preds[p], t->cols[preds[p].column].type,
-                                   nd->n.min_i, nd->n.max_i)) { selected = 0; break; }
-            }
-            if (!selected) continue;
+            if (!o_chunk_selected(t, st, k, preds, n_preds)) continue;
             if (nsel == cap) { cap *= 2; sel = realloc(sel, cap * 16); }
             sel[nsel * 2] = s;
             sel[nsel * 2 + 1] = k;

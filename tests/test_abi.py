@@ -26,7 +26,7 @@ def test_all_header_symbols_exported():
 
 
 def test_abi_version():
-    assert C.CDLL(os.path.join(REPO, "citus_amd", "libcstripe.so")).cstripe_abi_version() == 1
+    assert C.CDLL(os.path.join(REPO, "citus_amd", "libcstripe.so")).cstripe_abi_version() == 2
 
 
 def test_gpu_calls_fail_loudly_without_device(tmp_path):

@@ -626,3 +626,54 @@ def test_nan_semantics_parity(tmp_path):
     # col > huge: only NaN rows pass; count must be exactly the NaN count
     op, _, gp, _ = both(path, [(0, ca.PRED_GT, 1e12)], [(ca.AGG_COUNT_STAR, -1)])
     assert gp[0].count == op[0].count == int(np.isnan(a).sum())
+
+
+def test_or_pushdown_golden_gpu(tmp_path):
+    """OR pushdown on device vs the reference's expected chunk-filter vectors
+    (expected/columnar_chunk_filtering.out:837-930) and the oracle."""
+    a = np.arange(1, 200001, dtype=np.int64)
+    path = str(tmp_path / "pushdown_test.cs")
+    ca.write_table(path, [("a", ca.I64, 0)], [a], compression=ca.COMP_LZ4,
+                   stripe_row_limit=2000, chunk_group_row_limit=1000)
+    cases = [
+        ([(0, ca.PRED_EQ, 204356, 1), (0, ca.PRED_EQ, 104356, 1),
+          (0, ca.PRED_EQ, 76556, 1)], 2, 198, 180912),
+        ([(0, ca.PRED_EQ, 194356, 1), (0, ca.PRED_EQ, 104356, 1),
+          (0, ca.PRED_EQ, 76556, 1)], 3, 197, 375268),
+        ([(0, ca.PRED_GT, 1000, 1), (0, ca.PRED_GT, 20000, 1),
+          (0, ca.PRED_GT, 1000, 2), (0, ca.PRED_LT, 50000, 2),
+          (0, ca.PRED_LT, 10000, 3), (0, ca.PRED_GT, 20000, 3),
+          (0, ca.PRED_LT, 10000, 4), (0, ca.PRED_LT, 50000, 4)],
+         38998, 161, 1099459500),
+    ]
+    aggs = [(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_STAR, -1)]
+    for preds, exp_rows, exp_filtered, exp_sum in cases:
+        op, ofilt, gp, gfilt = both(path, preds, aggs)
+        assert gfilt == ofilt == exp_filtered
+        assert gp[1].count == exp_rows
+        assert gp[0].i128 == exp_sum
+        assert_parity(op, gp, aggs)
+
+
+def test_or_pushdown_fuzz_gpu(tmp_path):
+    """randomized CNF filters (mixed OR groups + standalone conjuncts,
+    canonical and greedy writers) — GPU vs oracle."""
+    rng = np.random.default_rng(99)
+    for trial in range(8):
+        n = int(rng.integers(5000, 60001))
+        a = rng.integers(0, 5000, n).astype(np.int64)
+        b = rng.integers(-1000, 1000, n).astype(np.int64)
+        path = str(tmp_path / f"orf{trial}.cs")
+        ca.write_table(path, [("a", ca.I64, 0), ("b", ca.I64, 0)], [a, b],
+                       compression=ca.COMP_LZ4,
+                       chunk_group_row_limit=int(rng.integers(1, 8)) * 1000,
+                       canonical=int(rng.integers(0, 2)))
+        preds = []
+        for _ in range(int(rng.integers(1, 7))):
+            preds.append((int(rng.integers(0, 2)), int(rng.integers(0, 6)),
+                          int(rng.integers(-1200, 5200)),
+                          int(rng.integers(0, 4))))
+        aggs = [(ca.AGG_COUNT_STAR, -1), (ca.AGG_SUM_I64, 0), (ca.AGG_MIN_I64, 1)]
+        op, ofilt, gp, gfilt = both(path, preds, aggs)
+        assert ofilt == gfilt, (trial, preds)
+        assert_parity(op, gp, aggs)

@@ -125,3 +125,69 @@ def test_mt_oracle_matches_serial(tmp_path):
     for a in range(len(aggs)):
         assert mt[a].i128 == serial[a].i128
         assert mt[a].count == serial[a].count
+
+
+def _pushdown_table(tmp_path):
+    """the reference's pushdown_test: a = 1..200000, stripe 2000, chunk 1000
+    (columnar_chunk_filtering.out:826-833)"""
+    a = np.arange(1, 200001, dtype=np.int64)
+    path = str(tmp_path / "pushdown_test.cs")
+    ca.write_table(path, [("a", ca.I64, 0)], [a], compression=ca.COMP_LZ4,
+                   stripe_row_limit=2000, chunk_group_row_limit=1000)
+    return path
+
+
+def test_or_pushdown_golden(tmp_path):
+    """OR chunk filtering against the reference's own expected output
+    (expected/columnar_chunk_filtering.out:837-930): row counts, chunk
+    groups removed, and sums for three OR shapes."""
+    path = _pushdown_table(tmp_path)
+    with oracle.OracleTable(path) as t:
+        # a = 204356 OR a = 104356 OR a = 76556 -> 2 rows, 198 chunks removed
+        preds = [(0, ca.PRED_EQ, 204356, 1), (0, ca.PRED_EQ, 104356, 1),
+                 (0, ca.PRED_EQ, 76556, 1)]
+        parts, filtered = t.scan_agg(preds, [(ca.AGG_SUM_I64, 0),
+                                             (ca.AGG_COUNT_STAR, -1)])
+        assert parts[1].count == 2
+        assert filtered == 198
+        assert parts[0].i128 == 180912
+
+        # a = 194356 OR a = 104356 OR a = 76556 -> 3 rows, 197 removed
+        preds = [(0, ca.PRED_EQ, 194356, 1), (0, ca.PRED_EQ, 104356, 1),
+                 (0, ca.PRED_EQ, 76556, 1)]
+        parts, filtered = t.scan_agg(preds, [(ca.AGG_SUM_I64, 0),
+                                             (ca.AGG_COUNT_STAR, -1)])
+        assert parts[1].count == 3
+        assert filtered == 197
+        assert parts[0].i128 == 375268
+
+        # (a>1000 AND a<10000) OR (a>20000 AND a<50000) -> 38998 rows,
+        # 161 chunks removed, sum 1099459500. CNF distribution:
+        # (a>1000|a>20000)(a>1000|a<50000)(a<10000|a>20000)(a<10000|a<50000)
+        preds = [(0, ca.PRED_GT, 1000, 1), (0, ca.PRED_GT, 20000, 1),
+                 (0, ca.PRED_GT, 1000, 2), (0, ca.PRED_LT, 50000, 2),
+                 (0, ca.PRED_LT, 10000, 3), (0, ca.PRED_GT, 20000, 3),
+                 (0, ca.PRED_LT, 10000, 4), (0, ca.PRED_LT, 50000, 4)]
+        parts, filtered = t.scan_agg(preds, [(ca.AGG_SUM_I64, 0),
+                                             (ca.AGG_COUNT_STAR, -1)])
+        assert parts[1].count == 38998
+        assert filtered == 161
+        assert parts[0].i128 == 1099459500
+
+
+def test_or_group_null_semantics(tmp_path):
+    """A NULL operand fails its atom but the OR can still pass via another
+    arm (SQL: NULL OR TRUE = TRUE)."""
+    n = 100
+    a = np.arange(n, dtype=np.int64)
+    b = np.arange(n, dtype=np.int64)
+    na = np.zeros(n, dtype=np.uint8)
+    na[:50] = 1   # a NULL for rows 0..49
+    path = str(tmp_path / "nl.cs")
+    ca.write_table(path, [("a", ca.I64, 0), ("b", ca.I64, 0)], [a, b],
+                   nulls=[na, None], compression=ca.COMP_LZ4)
+    with oracle.OracleTable(path) as t:
+        # a < 10 OR b < 20: rows 0..19 pass (0..9 via b since a is NULL)
+        preds = [(0, ca.PRED_LT, 10, 1), (1, ca.PRED_LT, 20, 1)]
+        parts, _ = t.scan_agg(preds, [(ca.AGG_COUNT_STAR, -1)])
+        assert parts[0].count == 20
