@@ -80,7 +80,26 @@ class Batch(C.Structure):
 
 
 class GroupResult(C.Structure):
-    _fields_ = [("n_groups", C.c_uint32), ("keys", C.c_uint16 * 64)]
+    _fields_ = [("n_groups", C.c_uint32), ("keys", C.c_uint32 * 64)]
+
+
+GROUP_KEY_NULL = 0x100          # 9-bit key encoding: bit 8 = NULL key
+
+
+def decode_group_key(kv):
+    """(k0, k1) from the packed 9-bit-per-column encoding; NULL keys (their
+    own group, like the reference HashAggregate) decode to None."""
+    e0 = kv & 0x1FF
+    e1 = (kv >> 9) & 0x1FF
+    return (None if e0 & GROUP_KEY_NULL else e0,
+            None if e1 & GROUP_KEY_NULL else e1)
+
+
+def encode_group_key(k):
+    e0 = GROUP_KEY_NULL if k[0] is None else k[0]
+    k1 = k[1] if len(k) > 1 else 0
+    e1 = GROUP_KEY_NULL if k1 is None else k1
+    return e0 | (e1 << 9)
 
 
 def _sig(name, res, args):
@@ -405,7 +424,7 @@ class Scan:
                                  C.byref(gr), out), "scan_agg_grouped")
         res = {}
         for g in range(gr.n_groups):
-            key = (gr.keys[g] & 0xFF, gr.keys[g] >> 8)
+            key = decode_group_key(gr.keys[g])
             res[key] = [out[g * len(aggs) + a] for a in range(len(aggs))]
         return res
 

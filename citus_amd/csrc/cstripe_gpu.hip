@@ -115,9 +115,9 @@ struct cs_gpu_state {
     AccCell *d_block = nullptr;
     AccCell *d_final = nullptr;
     AccCell *d_final2 = nullptr;     /* wide-pass outputs of the final reduce */
-    uint16_t *d_gkeys = nullptr;     /* grouped: per-block key tables */
+    uint32_t *d_gkeys = nullptr;     /* grouped: per-block key tables */
     AccCell *d_gcells = nullptr;
-    uint16_t *d_gfkeys = nullptr;    /* grouped: final merged groups */
+    uint32_t *d_gfkeys = nullptr;    /* grouped: final merged groups */
     AccCell *d_gfcells = nullptr;
     uint32_t *d_gn = nullptr;
     uint8_t *d_tmp = nullptr;        /* next_batch: canonical-chunk decode buf */
@@ -1341,7 +1341,7 @@ template <int NAGGS, int R>
 __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
     const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
-    uint16_t *__restrict__ keys_out, AccCell *__restrict__ cells_out,
+    uint32_t *__restrict__ keys_out, AccCell *__restrict__ cells_out,
     int *__restrict__ err, const GroupParams gp)
 {
     const AggParams &params = gp.base;
@@ -1411,7 +1411,7 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
         #pragma unroll
         for (int k = 0; k < R; k++) {
             key[k] = (uint32_t)k0.v[k] & 0xFF;
-            if (gp.n_group_cols > 1) key[k] |= ((uint32_t)k1.v[k] & 0xFF) << 8;
+            if (gp.n_group_cols > 1) key[k] |= ((uint32_t)k1.v[k] & 0xFF) << 9;
         }
         /* slot per row: register cache, else LDS claim (CAS linear probe) */
         uint32_t slot[R];
@@ -1485,7 +1485,7 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
 
     /* merge the block's wave tables into one compact list (as grouped_agg) */
     if (threadIdx.x == 0) {
-        uint16_t *bk = keys_out + (size_t)blockIdx.x * (n_waves * MGRP_SLOTS);
+        uint32_t *bk = keys_out + (size_t)blockIdx.x * (n_waves * MGRP_SLOTS);
         AccCell *bc = cells_out + (size_t)blockIdx.x * (n_waves * MGRP_SLOTS) * n_aggs;
         uint32_t n = 0;
         for (uint32_t w = 0; w < n_waves; w++) {
@@ -1493,10 +1493,10 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
                 int kk = wkeys[w][s2];
                 if (kk < 0) continue;
                 uint32_t at = n;
-                for (uint32_t j = 0; j < n; j++) if (bk[j] == (uint16_t)kk) { at = j; break; }
+                for (uint32_t j = 0; j < n; j++) if (bk[j] == (uint32_t)kk) { at = j; break; }
                 ThreadAcc *src = wacc + ((size_t)(w * MGRP_SLOTS) + s2) * n_aggs;
                 if (at == n) {
-                    bk[n] = (uint16_t)kk;
+                    bk[n] = (uint32_t)kk;
                     for (uint32_t a = 0; a < n_aggs; a++) {
                         AccCell c{src[a].lo, src[a].hi, src[a].f, src[a].cnt};
                         bc[(size_t)n * n_aggs + a] = c;
@@ -1513,7 +1513,7 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
                 }
             }
         }
-        for (uint32_t j = n; j < n_waves * MGRP_SLOTS; j++) bk[j] = 0xFFFF;
+        for (uint32_t j = n; j < n_waves * MGRP_SLOTS; j++) bk[j] = 0xFFFFFFFFu;
     }
 }
 
@@ -1521,7 +1521,7 @@ template <int NAGGS>
 __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
     const uint32_t *__restrict__ rank, const GroupDesc *__restrict__ groups,
-    const ColLoc *__restrict__ colloc, uint16_t *__restrict__ keys_out,
+    const ColLoc *__restrict__ colloc, uint32_t *__restrict__ keys_out,
     AccCell *__restrict__ cells_out, int *__restrict__ err,
     const GroupParams gp)
 {
@@ -1534,15 +1534,15 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     const uint32_t n_waves = AGG_BLOCK / WAVE;
 
     extern __shared__ uint8_t gsh[];
-    uint16_t *wkeys = (uint16_t *)gsh;                          /* [wave][S] */
-    ThreadAcc *wacc = (ThreadAcc *)(gsh + n_waves * GRP_SLOTS * sizeof(uint16_t));
+    uint32_t *wkeys = (uint32_t *)gsh;                          /* [wave][S] */
+    ThreadAcc *wacc = (ThreadAcc *)(gsh + n_waves * GRP_SLOTS * sizeof(uint32_t));
     /* wacc[((wave*S)+slot)*n_aggs + a] */
 
     for (uint32_t i = threadIdx.x; i < n_waves * GRP_SLOTS; i += AGG_BLOCK)
-        wkeys[i] = 0xFFFF;
+        wkeys[i] = 0xFFFFFFFFu;
     __syncthreads();
 
-    uint16_t *mykeys = wkeys + wid * GRP_SLOTS;
+    uint32_t *mykeys = wkeys + wid * GRP_SLOTS;
     ThreadAcc *myacc = wacc + (size_t)wid * GRP_SLOTS * n_aggs_ct;
     uint32_t used = 0;       /* wave-uniform slot count (updated by lane 0 path) */
     /* per-lane 4-way key->slot register cache: hits skip the wave-ballot
@@ -1580,8 +1580,12 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
             if (pass) {
                 int64_t kv; double kf;
                 for (uint32_t gc = 0; gc < gp.n_group_cols; gc++) {
-                    col_value(data, scratch, rank, cols[gp.gproj[gc]], row, kv, kf);
-                    key |= ((uint32_t)kv & 0xFF) << (8 * gc);
+                    /* NULL keys form their own group (reference
+                     * HashAggregate treats NULLs as equal): 9-bit encoding,
+                     * bit 8 = null */
+                    bool ok = col_value(data, scratch, rank, cols[gp.gproj[gc]], row, kv, kf);
+                    uint32_t enc = ok ? ((uint32_t)kv & 0xFF) : CSTRIPE_GROUP_KEY_NULL;
+                    key |= enc << (9 * gc);
                 }
             }
             /* round 0 (agg-independent): assign each passing lane its key's
@@ -1602,7 +1606,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                     bool mine = pass && key == kk;
                     uint32_t slot = 0xFFFFFFFF;
                     for (uint32_t sidx = 0; sidx < used; sidx++)
-                        if (mykeys[sidx] == (uint16_t)kk) { slot = sidx; break; }
+                        if (mykeys[sidx] == kk) { slot = sidx; break; }
                     if (slot == 0xFFFFFFFF) {
                         if (used >= GRP_SLOTS) {      /* overflow: flag + drop */
                             if (lane == 0) atomicOr(err, 8);
@@ -1610,7 +1614,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                             continue;
                         }
                         slot = used++;
-                        if (lane == 0) mykeys[slot] = (uint16_t)kk;
+                        if (lane == 0) mykeys[slot] = kk;
                         for (uint32_t a = 0; a < n_aggs_ct; a++) {
                             ThreadAcc z;
                             acc_init(z, params.aggs[a].kind);
@@ -1644,13 +1648,13 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
     /* merge the block's wave tables into one compact list; write to global.
      * wave 0 lane 0 does it serially — tables are tiny. */
     if (threadIdx.x == 0) {
-        uint16_t *bk = keys_out + (size_t)blockIdx.x * (n_waves * GRP_SLOTS);
+        uint32_t *bk = keys_out + (size_t)blockIdx.x * (n_waves * GRP_SLOTS);
         AccCell *bc = cells_out + (size_t)blockIdx.x * (n_waves * GRP_SLOTS) * n_aggs_ct;
         uint32_t n = 0;
         for (uint32_t w = 0; w < n_waves; w++) {
             for (uint32_t sidx = 0; sidx < GRP_SLOTS; sidx++) {
-                uint16_t k = wkeys[w * GRP_SLOTS + sidx];
-                if (k == 0xFFFF) continue;
+                uint32_t k = wkeys[w * GRP_SLOTS + sidx];
+                if (k == 0xFFFFFFFFu) continue;
                 uint32_t at = n;
                 for (uint32_t j = 0; j < n; j++) if (bk[j] == k) { at = j; break; }
                 ThreadAcc *src = wacc + ((size_t)(w * GRP_SLOTS) + sidx) * n_aggs_ct;
@@ -1672,7 +1676,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                 }
             }
         }
-        for (uint32_t j = n; j < n_waves * GRP_SLOTS; j++) bk[j] = 0xFFFF;
+        for (uint32_t j = n; j < n_waves * GRP_SLOTS; j++) bk[j] = 0xFFFFFFFFu;
     }
 }
 
@@ -1707,7 +1711,7 @@ struct FusedGParams {
 template <int NAGGS>
 __global__ __launch_bounds__(AGG_BLOCK) void fused_grouped_kernel(
     const uint8_t *__restrict__ data, const SegDesc *__restrict__ segs,
-    const FusedTileG *__restrict__ tiles, uint16_t *__restrict__ keys_out,
+    const FusedTileG *__restrict__ tiles, uint32_t *__restrict__ keys_out,
     AccCell *__restrict__ cells_out, int *__restrict__ err,
     const FusedGParams gp)
 {
@@ -1771,7 +1775,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_grouped_kernel(
             if (!pass) continue;
             uint32_t key = 0;
             for (uint32_t gc = 0; gc < gp.n_group_cols; gc++)
-                key |= ((uint32_t)val(gp.gproj[gc], row) & 0xFF) << (8 * gc);
+                key |= ((uint32_t)val(gp.gproj[gc], row) & 0xFF) << (9 * gc);
             uint32_t slot = 0xFFFFFFFF;
             if (key == ck0) slot = cs0;
             else if (key == ck1) slot = cs1;
@@ -1821,10 +1825,10 @@ __global__ __launch_bounds__(AGG_BLOCK) void fused_grouped_kernel(
     /* flush the block table */
     __syncthreads();
     if (tid == 0) {
-        uint16_t *bk = keys_out + (size_t)blockIdx.x * FUSEG_SLOTS;
+        uint32_t *bk = keys_out + (size_t)blockIdx.x * FUSEG_SLOTS;
         AccCell *bc = cells_out + (size_t)blockIdx.x * FUSEG_SLOTS * n_aggs_ct;
         for (uint32_t s2 = 0; s2 < FUSEG_SLOTS; s2++) {
-            bk[s2] = skeys[s2] < 0 ? 0xFFFF : (uint16_t)skeys[s2];
+            bk[s2] = skeys[s2] < 0 ? 0xFFFFFFFFu : (uint32_t)skeys[s2];
             for (uint32_t a = 0; a < n_aggs_ct; a++) {
                 const ThreadAcc &x = scells[s2 * n_aggs_ct + a];
                 AccCell cell{x.lo, x.hi, x.f, x.cnt};
@@ -1893,8 +1897,8 @@ __device__ inline void cell_merge_atomic(ThreadAcc *dst, uint8_t kind, const Acc
 }
 
 __global__ __launch_bounds__(AGG_BLOCK) void grouped_final_kernel(
-    const uint16_t *__restrict__ keys_in, const AccCell *__restrict__ cells_in,
-    uint32_t n_blocks, uint32_t per_block, uint16_t *__restrict__ keys_out,
+    const uint32_t *__restrict__ keys_in, const AccCell *__restrict__ cells_in,
+    uint32_t n_blocks, uint32_t per_block, uint32_t *__restrict__ keys_out,
     AccCell *__restrict__ cells_out, uint32_t *__restrict__ n_groups_out,
     int *__restrict__ err, const GroupParams gp)
 {
@@ -1910,8 +1914,8 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_final_kernel(
 
     const uint32_t total = n_blocks * per_block;
     for (uint32_t f = threadIdx.x; f < total; f += AGG_BLOCK) {
-        uint16_t k = keys_in[f];
-        if (k == 0xFFFF) continue;
+        uint32_t k = keys_in[f];
+        if (k == 0xFFFFFFFFu) continue;
         int slot = -1;
         for (uint32_t s = 0; s < CSTRIPE_MAX_GROUPS; s++) {
             int old = atomicCAS(&skeys[s], -1, (int)k);
@@ -1928,7 +1932,7 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_final_kernel(
         uint32_t n = 0;
         for (uint32_t s = 0; s < CSTRIPE_MAX_GROUPS; s++) {
             if (skeys[s] < 0) continue;
-            keys_out[n] = (uint16_t)skeys[s];
+            keys_out[n] = (uint32_t)skeys[s];
             for (uint32_t a = 0; a < params.n_aggs; a++) {
                 const ThreadAcc &t = scells[s * MAX_AGGS + a];
                 AccCell c{t.lo, t.hi, t.f, t.cnt};
@@ -2474,12 +2478,6 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             if (pj < 0) { cs_set_err("group col %u not projected", group_cols[i]); return CSTRIPE_ERR_ARG; }
             if (r->cols[group_cols[i]].type != CSTRIPE_I8) { cs_set_err("group col %u must be I8", group_cols[i]); return CSTRIPE_ERR_ARG; }
             gp.gproj[i] = (uint32_t)pj;
-            /* NULL group keys unsupported this round: require dense key cols */
-            for (uint32_t gi = 0; gi < g->n_groups; gi++)
-                if (!(g->colloc_host[(uint64_t)gi * g->n_proj + pj].flags & 2)) {
-                    cs_set_err("grouped aggregation requires non-NULL group key columns");
-                    return CSTRIPE_ERR_ARG;
-                }
         }
         gp.n_work = g->n_groups * p.tiles_per_group;
         if (gp.n_work == 0) {
@@ -2491,9 +2489,9 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         const uint32_t per_block = n_waves * GRP_SLOTS;
         uint32_t grid = gp.n_work < GRP_GRID ? gp.n_work : GRP_GRID;
         if (!g->d_gkeys) {
-            HIP_TRY(hipMalloc(&g->d_gkeys, (uint64_t)GRP_GRID * per_block * 2));
+            HIP_TRY(hipMalloc(&g->d_gkeys, (uint64_t)GRP_GRID * per_block * 4));
             HIP_TRY(hipMalloc(&g->d_gcells, (uint64_t)GRP_GRID * per_block * MAX_AGGS * sizeof(AccCell)));
-            HIP_TRY(hipMalloc(&g->d_gfkeys, CSTRIPE_MAX_GROUPS * 2));
+            HIP_TRY(hipMalloc(&g->d_gfkeys, CSTRIPE_MAX_GROUPS * 4));
             HIP_TRY(hipMalloc(&g->d_gfcells, (uint64_t)CSTRIPE_MAX_GROUPS * MAX_AGGS * sizeof(AccCell)));
             HIP_TRY(hipMalloc(&g->d_gn, 4));
         }
@@ -2534,7 +2532,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             if (h_err & 4) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
             if (h_err == 0) {
                 s->last_fused = 0;   /* i8 key columns decode to scratch */
-                uint16_t h_keys[CSTRIPE_MAX_GROUPS];
+                uint32_t h_keys[CSTRIPE_MAX_GROUPS];
                 std::vector<AccCell> h_cells((size_t)CSTRIPE_MAX_GROUPS * n_aggs);
                 uint32_t h_n = 0;
                 HIP_TRY(hipMemcpyAsync(h_keys, g->d_gfkeys, sizeof(h_keys), hipMemcpyDeviceToHost, g->stream));
@@ -2620,7 +2618,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             if (h_err & 4) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
             if (h_err == 0) {
                 s->last_fused = 1;
-                uint16_t h_keys[CSTRIPE_MAX_GROUPS];
+                uint32_t h_keys[CSTRIPE_MAX_GROUPS];
                 std::vector<AccCell> h_cells((size_t)CSTRIPE_MAX_GROUPS * n_aggs);
                 uint32_t h_n = 0;
                 HIP_TRY(hipMemcpyAsync(h_keys, g->d_gfkeys, sizeof(h_keys), hipMemcpyDeviceToHost, g->stream));
@@ -2665,7 +2663,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             }
         }
 
-        uint32_t lds = n_waves * GRP_SLOTS * 2 + n_waves * GRP_SLOTS * n_aggs * (uint32_t)sizeof(ThreadAcc);
+        uint32_t lds = n_waves * GRP_SLOTS * 4 + n_waves * GRP_SLOTS * n_aggs * (uint32_t)sizeof(ThreadAcc);
         HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
         HIP_TRY(hipEventRecord(g->ev0, g->stream));
         { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
@@ -2688,7 +2686,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         HIP_TRY(hipGetLastError());
         HIP_TRY(hipEventRecord(g->ev2, g->stream));
 
-        uint16_t h_keys[CSTRIPE_MAX_GROUPS];
+        uint32_t h_keys[CSTRIPE_MAX_GROUPS];
         std::vector<AccCell> h_cells((size_t)CSTRIPE_MAX_GROUPS * n_aggs);
         uint32_t h_n = 0;
         int h_err = 0;

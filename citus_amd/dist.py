@@ -49,13 +49,15 @@ def all_gather_combine_grouped_rccl(comm, aggs, resdict, max_groups=64):
     """Grouped combine over the C-ABI collective: fixed 64-slot
     (keys, partials) block per rank through cagg_allgather, merged locally
     with cagg_combine per distinct key."""
+    from . import encode_group_key
     n_aggs = len(aggs)
-    keys = sorted(resdict)
+    keys = sorted(resdict, key=encode_group_key)
     assert len(keys) <= max_groups
     import struct
     kbuf = bytearray(struct.pack(f"<{max_groups}i", *([-1] * max_groups)))
+    from . import encode_group_key
     for g, k in enumerate(keys):
-        struct.pack_into("<i", kbuf, g * 4, k[0] | (k[1] << 8))
+        struct.pack_into("<i", kbuf, g * 4, encode_group_key(k))
     flat = (Partial * (max_groups * n_aggs))()
     for g, k in enumerate(keys):
         for a in range(n_aggs):
@@ -72,10 +74,12 @@ def all_gather_combine_grouped_rccl(comm, aggs, resdict, max_groups=64):
         for i in range(max_groups):
             if ks[i] < 0:
                 continue
-            table[(ks[i] & 0xFF, ks[i] >> 8)] = [parts[i * n_aggs + a]
-                                                 for a in range(n_aggs)]
+            from . import decode_group_key
+            table[decode_group_key(ks[i])] = [parts[i * n_aggs + a]
+                                              for a in range(n_aggs)]
         rank_tables.append(table)
-    all_keys = sorted(set().union(*[t.keys() for t in rank_tables]))
+    all_keys = sorted(set().union(*[t.keys() for t in rank_tables]),
+                      key=encode_group_key)
     null_row = [Partial(is_null=1) for _ in range(n_aggs)]
     return {k: _combine(aggs, [t.get(k, null_row) for t in rank_tables])
             for k in all_keys}
@@ -119,12 +123,14 @@ def all_gather_combine_grouped(aggs, resdict, device=None, max_groups=64):
         device = "cuda" if td.get_backend() == "nccl" else "cpu"
     n_aggs = len(aggs)
 
-    keys = sorted(resdict)
+    from . import encode_group_key
+    keys = sorted(resdict, key=encode_group_key)
     assert len(keys) <= max_groups
     kbuf = torch.full((max_groups,), -1, dtype=torch.int32)
     flat = (Partial * (max_groups * n_aggs))()
+    from . import encode_group_key
     for g, k in enumerate(keys):
-        kbuf[g] = k[0] | (k[1] << 8)
+        kbuf[g] = encode_group_key(k)
         for a in range(n_aggs):
             flat[g * n_aggs + a] = resdict[k][a]
     pbuf = torch.frombuffer(bytearray(C.string_at(flat, C.sizeof(flat))),
@@ -145,10 +151,12 @@ def all_gather_combine_grouped(aggs, resdict, device=None, max_groups=64):
             kv = int(ks[i])
             if kv < 0:
                 continue
-            table[(kv & 0xFF, kv >> 8)] = [parts[i * n_aggs + a] for a in range(n_aggs)]
+            from . import decode_group_key
+            table[decode_group_key(kv)] = [parts[i * n_aggs + a] for a in range(n_aggs)]
         rank_tables.append(table)
 
-    all_keys = sorted(set().union(*[t.keys() for t in rank_tables]))
+    from . import encode_group_key as _enc
+    all_keys = sorted(set().union(*[t.keys() for t in rank_tables]), key=_enc)
     out = {}
     null_row = [Partial(is_null=1) for _ in range(n_aggs)]
     for k in all_keys:

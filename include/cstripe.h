@@ -204,13 +204,17 @@ typedef struct cstripe_partial {
     uint8_t     _pad[7];
 } cstripe_partial;
 
-/* ---- group-by (Q1 shape): up to 2 categorical u8/i8 key columns ---- */
+/* ---- group-by (Q1 shape): up to 2 categorical u8/i8 key columns ----
+ * NULL keys form their own group, like the reference's HashAggregate
+ * (grouping treats NULLs as equal). Key encoding: 9 bits per key column —
+ * enc = (value & 0xFF) | (is_null << 8) — packed enc0 | enc1 << 9. */
 #define CSTRIPE_MAX_GROUP_COLS 2
 #define CSTRIPE_MAX_GROUPS     64
+#define CSTRIPE_GROUP_KEY_NULL 0x100u
 
 typedef struct cstripe_group_result {
     uint32_t    n_groups;
-    uint16_t    keys[CSTRIPE_MAX_GROUPS];        /* key0 | key1<<8 */
+    uint32_t    keys[CSTRIPE_MAX_GROUPS];        /* enc0 | enc1<<9 */
     /* partials[g*n_aggs + a] for group g, agg a — caller-provided, size
      * CSTRIPE_MAX_GROUPS * n_aggs */
 } cstripe_group_result;

@@ -37,7 +37,7 @@ _lib.oracle_scan_agg.restype = C.c_int
 _lib.oracle_scan_agg.argtypes = [C.c_void_p, C.c_uint64, C.POINTER(Pred), C.c_uint32,
                                  C.POINTER(AggSpec), C.c_uint32,
                                  C.POINTER(C.c_uint32), C.c_uint32,
-                                 C.POINTER(Partial), C.POINTER(C.c_uint16),
+                                 C.POINTER(Partial), C.POINTER(C.c_uint32),
                                  C.POINTER(C.c_uint32), C.POINTER(C.c_int64)]
 
 
@@ -82,7 +82,7 @@ class OracleTable:
         gc = (C.c_uint32 * max(1, ngc))(*group_cols) if ngc else None
         cap = 64 if ngc else 1
         out = (Partial * (cap * len(aggs)))()
-        keys = (C.c_uint16 * 64)()
+        keys = (C.c_uint32 * 64)()
         n_groups = C.c_uint32(0)
         filt = C.c_int64(0)
         rc = _lib.oracle_scan_agg(self._h, 0, parr, len(preds), aarr, len(aggs),
@@ -92,7 +92,8 @@ class OracleTable:
         if ngc:
             res = {}
             for g in range(n_groups.value):
-                key = (keys[g] & 0xFF, keys[g] >> 8)
+                from citus_amd import decode_group_key
+                key = decode_group_key(keys[g])
                 res[key] = [out[g * len(aggs) + a] for a in range(len(aggs))]
             return res, filt.value
         return [out[a] for a in range(len(aggs))], filt.value

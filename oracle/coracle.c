@@ -470,7 +470,7 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
                     const cstripe_pred *preds, uint32_t n_preds,
                     const cstripe_agg_spec *aggs, uint32_t n_aggs,
                     const uint32_t *group_cols, uint32_t n_group_cols,
-                    cstripe_partial *out, uint16_t *keys_out,
+                    cstripe_partial *out, uint32_t *keys_out,
                     uint32_t *n_groups_out, int64_t *chunk_groups_filtered)
 {
     uint32_t ncols = t->head.column_count;
@@ -488,7 +488,7 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
     for (uint32_t i = 0; i < n_group_cols; i++) mask |= 1ull << group_cols[i];
 
     uint32_t n_groups = 0;
-    uint16_t gkeys[CSTRIPE_MAX_GROUPS];
+    uint32_t gkeys[CSTRIPE_MAX_GROUPS];
     oacc *accs = calloc((size_t)(n_group_cols ? CSTRIPE_MAX_GROUPS : 1) * n_aggs, sizeof(oacc));
     for (uint32_t a = 0; a < n_aggs; a++) oacc_init(&accs[a], aggs[a].kind);
 
@@ -522,11 +522,17 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
 
                 oacc *grp = accs;
                 if (n_group_cols) {
-                    uint16_t key = 0;
+                    uint32_t key = 0;
                     int64_t iv; double fv;
                     for (uint32_t gcn = 0; gcn < n_group_cols; gcn++) {
-                        if (!o_get(&cc[group_cols[gcn]], row, &iv, &fv)) { iv = 0xFF; }
-                        key |= (uint16_t)((iv & 0xFF) << (8 * gcn));
+                        /* NULL keys form their own group (HashAggregate
+                         * groups NULLs together); 9-bit enc, bit 8 = null */
+                        uint32_t enc;
+                        if (o_get(&cc[group_cols[gcn]], row, &iv, &fv))
+                            enc = (uint32_t)(iv & 0xFF);
+                        else
+                            enc = CSTRIPE_GROUP_KEY_NULL;
+                        key |= enc << (9 * gcn);
                     }
                     uint32_t gi = n_groups;
                     for (uint32_t g = 0; g < n_groups; g++) if (gkeys[g] == key) { gi = g; break; }

@@ -304,21 +304,34 @@ def test_grouped_random_many_groups(tmp_path):
                 assert res[k][i].i128 == ores[k][i].i128, (k, i)
 
 
-def test_grouped_null_keys_rejected(tmp_path):
-    n = 100
-    k = np.zeros(n, dtype=np.int8)
-    nk = np.zeros(n, dtype=np.uint8)
-    nk[5] = 1
+def test_grouped_null_keys(tmp_path):
+    """NULL group keys form their own group, like the reference's
+    HashAggregate (grouping treats NULLs as equal) — round-1 limit removed."""
+    n = 12_345
+    k = (np.arange(n) % 3).astype(np.int8)
+    nk = (np.arange(n) % 7 == 0).astype(np.uint8)     # NULL key rows
     v = np.arange(n, dtype=np.int64)
     path = str(tmp_path / "nk.cs")
     ca.write_table(path, [("k", ca.I8, 0), ("v", ca.I64, 0)], [k, v],
-                   nulls=[nk, None], compression=ca.COMP_LZ4)
-    aggs = [(ca.AGG_SUM_I64, 1)]
+                   nulls=[nk, None], compression=ca.COMP_LZ4,
+                   chunk_group_row_limit=2000)
+    aggs = [(ca.AGG_SUM_I64, 1), (ca.AGG_COUNT_STAR, -1)]
     with ca.Reader(path) as r, \
          r.scan(cols_mask=ca.agg_cols_mask(aggs) | 1) as s:
         s.stage()
-        with pytest.raises(ca.CStripeError, match="non-NULL group key"):
-            s.agg_grouped(aggs, (0,))
+        res = s.agg_grouped(aggs, (0,))
+    with oracle.OracleTable(path) as t:
+        ores, _ = t.scan_agg([], aggs, group_cols=(0,))
+    assert set(res) == set(ores)
+    assert any(key[0] is None for key in res)
+    for key in ores:
+        assert res[key][0].i128 == ores[key][0].i128
+        assert res[key][1].count == ores[key][1].count
+    # independent expectation for the NULL group
+    nmask = nk == 1
+    nkey = (None, 0)
+    assert res[nkey][0].i128 == int(v[nmask].sum())
+    assert res[nkey][1].count == int(nmask.sum())
 
 
 def test_decode_torture_patterns(tmp_path):
