@@ -23,6 +23,7 @@
  */
 #include "internal.h"
 #include "compress.h"
+#include "pglz.h"
 
 #include <hip/hip_runtime.h>
 #include <cstring>
@@ -2291,18 +2292,33 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 cl.val_off = dpos;
                 memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
                 dpos += align_up(nd.n.value_len, 16);
-            } else if (nd.n.comp_type == CSTRIPE_COMP_ZSTD) {
-                /* host-decode at stage (documented fallback; GPU zstd is a
-                 * later-round item — SURVEY.md §8f(1)) */
+            } else if (nd.n.comp_type == CSTRIPE_COMP_ZSTD ||
+                       nd.n.comp_type == CSTRIPE_COMP_PGLZ) {
+                /* host-decode at stage (zstd: documented fallback, GPU zstd
+                 * is a later item — SURVEY.md §8f(1); pglz is host-only by
+                 * design: reference-migrated tables read correctly) */
                 dpos = align_up(dpos, 16);
                 cl.val_off = dpos;
                 tmp.resize(nd.n.decompressed_size);
-                uint64_t seg_base = 0;
-                (void)seg_base;
                 for (const csf_seg &sg : st.nodes[c][sc.chunk].segs) {
-                    size_t zr = ZSTD_decompress(tmp.data() + sg.decomp_off, sg.decomp_len,
-                                                stripe_base + nd.n.value_off + sg.comp_off, sg.comp_len);
-                    if (ZSTD_isError(zr) || zr != sg.decomp_len) { cs_set_err("zstd host decode failed"); csgpu_release(s); return CSTRIPE_ERR_FORMAT; }
+                    if (nd.n.comp_type == CSTRIPE_COMP_ZSTD) {
+                        size_t zr = ZSTD_decompress(tmp.data() + sg.decomp_off, sg.decomp_len,
+                                                    stripe_base + nd.n.value_off + sg.comp_off, sg.comp_len);
+                        if (ZSTD_isError(zr) || zr != sg.decomp_len) { cs_set_err("zstd host decode failed"); csgpu_release(s); return CSTRIPE_ERR_FORMAT; }
+                    } else {
+                        const uint8_t *pb = stripe_base + nd.n.value_off + sg.comp_off;
+                        if (sg.comp_len < CSPGLZ_HDRSZ ||
+                            cspglz_varsize(pb) != sg.comp_len ||
+                            cspglz_rawsize(pb) != (int32_t)sg.decomp_len ||
+                            cspglz_decompress(pb + CSPGLZ_HDRSZ,
+                                              (int32_t)(sg.comp_len - CSPGLZ_HDRSZ),
+                                              tmp.data() + sg.decomp_off,
+                                              (int32_t)sg.decomp_len) < 0) {
+                            cs_set_err("pglz host decode failed");
+                            csgpu_release(s);
+                            return CSTRIPE_ERR_FORMAT;
+                        }
+                    }
                 }
                 memcpy(h_data.data() + dpos, tmp.data(), tmp.size());
                 dpos += align_up(nd.n.decompressed_size, 16);
@@ -2988,11 +3004,23 @@ int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
             /* NONE / host-predecoded zstd: take staged raw from file side */
             if (nd.n.comp_type == CSTRIPE_COMP_NONE) {
                 memcpy(packed.data(), stripe_base + nd.n.value_off, nd.n.decompressed_size);
-            } else { /* zstd: decode host-side as at stage */
+            } else { /* zstd / pglz: decode host-side as at stage */
                 for (const csf_seg &sg : st.nodes[c][sc.chunk].segs) {
-                    size_t zr = ZSTD_decompress(packed.data() + sg.decomp_off, sg.decomp_len,
-                                                stripe_base + nd.n.value_off + sg.comp_off, sg.comp_len);
-                    if (ZSTD_isError(zr) || zr != sg.decomp_len) { cs_set_err("zstd host decode failed"); return CSTRIPE_ERR_FORMAT; }
+                    if (nd.n.comp_type == CSTRIPE_COMP_ZSTD) {
+                        size_t zr = ZSTD_decompress(packed.data() + sg.decomp_off, sg.decomp_len,
+                                                    stripe_base + nd.n.value_off + sg.comp_off, sg.comp_len);
+                        if (ZSTD_isError(zr) || zr != sg.decomp_len) { cs_set_err("zstd host decode failed"); return CSTRIPE_ERR_FORMAT; }
+                    } else {
+                        const uint8_t *pb = stripe_base + nd.n.value_off + sg.comp_off;
+                        if (sg.comp_len < CSPGLZ_HDRSZ ||
+                            cspglz_decompress(pb + CSPGLZ_HDRSZ,
+                                              (int32_t)(sg.comp_len - CSPGLZ_HDRSZ),
+                                              packed.data() + sg.decomp_off,
+                                              (int32_t)sg.decomp_len) < 0) {
+                            cs_set_err("pglz host decode failed");
+                            return CSTRIPE_ERR_FORMAT;
+                        }
+                    }
                 }
             }
         }

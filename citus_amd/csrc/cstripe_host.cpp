@@ -20,6 +20,7 @@
 #include "internal.h"
 #include "compress.h"
 #include "lz4_enc.h"
+#include "pglz.h"
 
 #include <cstdarg>
 #include <cstdio>
@@ -298,6 +299,30 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
                                                  : (size_t)w->opts.lz4_seg_target_kb * 1024;
     uint32_t width = type_width(w->cols[pc.col].type);
 
+    if (codec == CSTRIPE_COMP_PGLZ && !raw.empty()) {
+        /* reference PG_LZ layout: ColumnarCompressHeader (varlena len +
+         * rawsize) then the pglz stream, one whole-chunk buffer
+         * (columnar_compression.c:122-151) */
+        out.value_comp.resize(raw.size() + CSPGLZ_HDRSZ);
+        int32_t csz = cspglz_compress(raw.data(), (int32_t)raw.size(),
+                                      out.value_comp.data() + CSPGLZ_HDRSZ,
+                                      (int32_t)raw.size());
+        if (csz > 0) {
+            out.value_comp.resize((size_t)csz + CSPGLZ_HDRSZ);
+            cspglz_set_header(out.value_comp.data(),
+                              (uint32_t)out.value_comp.size(),
+                              (int32_t)raw.size());
+            csf_seg s{0, (uint32_t)out.value_comp.size(), 0, (uint32_t)raw.size()};
+            out.segs.push_back(s);
+            out.node.comp_type = CSTRIPE_COMP_PGLZ;
+            out.node.comp_level = 0;
+            out.node.n_segs = 1;
+            out.node.value_len = out.value_comp.size();
+            return true;
+        }
+        out.value_comp.clear();
+        codec = CSTRIPE_COMP_NONE;      /* did not shrink -> raw */
+    }
     if (raw.empty() || codec == CSTRIPE_COMP_NONE) {
         out.value_comp = raw;
         csf_seg s{0, (uint32_t)raw.size(), 0, (uint32_t)raw.size()};

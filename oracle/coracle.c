@@ -64,6 +64,7 @@
  * libcstripe) */
 #include "../include/cstripe.h"
 #include "../citus_amd/csrc/format.h"
+#include "../citus_amd/csrc/pglz.h"
 
 /* system codec prototypes (stable ABIs; reference links the same functions) */
 int LZ4_decompress_safe(const char *src, char *dst, int compressedSize, int dstCapacity);
@@ -200,6 +201,19 @@ static const uint8_t *o_decompress(const oracle_table *t, const ostripe *st,
             size_t r = ZSTD_decompress(out + sg->decomp_off, sg->decomp_len,
                                        src + sg->comp_off, sg->comp_len);
             if (ZSTD_isError(r) || r != sg->decomp_len) { free(out); *needs_free = 0; return NULL; }
+        } else if (nd->n.comp_type == CSTRIPE_COMP_PGLZ) {
+            /* ColumnarCompressHeader (varlena + rawsize) then the pglz
+             * stream (columnar_compression.c:230-262) */
+            const uint8_t *pb = src + sg->comp_off;
+            if (sg->comp_len < CSPGLZ_HDRSZ ||
+                cspglz_varsize(pb) != sg->comp_len ||
+                cspglz_rawsize(pb) != (int32_t)sg->decomp_len ||
+                cspglz_decompress(pb + CSPGLZ_HDRSZ,
+                                  (int32_t)(sg->comp_len - CSPGLZ_HDRSZ),
+                                  out + sg->decomp_off,
+                                  (int32_t)sg->decomp_len) < 0) {
+                free(out); *needs_free = 0; return NULL;
+            }
         } else {
             free(out); *needs_free = 0; return NULL;
         }

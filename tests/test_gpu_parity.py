@@ -64,7 +64,8 @@ def test_q6_golden_fixture(golden_dir, expected, variant):
     assert_parity(op, gp, aggs)
 
 
-@pytest.mark.parametrize("comp", [ca.COMP_NONE, ca.COMP_LZ4, ca.COMP_ZSTD])
+@pytest.mark.parametrize("comp", [ca.COMP_NONE, ca.COMP_LZ4, ca.COMP_ZSTD,
+                                  ca.COMP_PGLZ])
 @pytest.mark.parametrize("n", [1, 999, 10000, 123457])
 def test_agg_kinds_random(tmp_path, comp, n):
     a = RNG.integers(-10**6, 10**6, n).astype(np.int64)
