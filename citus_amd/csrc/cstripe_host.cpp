@@ -21,6 +21,7 @@
 #include "compress.h"
 #include "lz4_enc.h"
 #include "pglz.h"
+#include "zstd_r.h"
 
 #include <cstdarg>
 #include <cstdio>
@@ -388,6 +389,8 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     size_t per = (target / 16) * 16;    /* exact target (16B-aligned splits) —
                                          * uniform segments are what the fused
                                          * GPU kernel's tile math assumes */
+    if (codec == CSTRIPE_COMP_ZSTD && per < 1024)
+        per = 1024;                     /* amortize the ~14 B frame overhead */
     if (per == 0) per = n;
 
     out.value_comp.clear();
@@ -425,14 +428,40 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
             out.value_comp.resize(base + (size_t)csz);
             s.comp_len = (uint32_t)csz;
         } else if (codec == CSTRIPE_COMP_ZSTD) {
-            size_t bound = ZSTD_compressBound(len);
+            /* restricted zstd frames (zstd_r.h): standard RFC 8878 frames
+             * the GPU lane decoder handles (raw literals + predefined-FSE
+             * sequences); libzstd-decodable, pinned in tests. Falls back to
+             * a raw-block frame when the parse does not shrink a segment. */
+            static zr_ctables zct;
+            static bool zct_init = false;
+            #pragma omp critical(zr_ct)
+            {
+                if (!zct_init) { zr_build_ctables(&zct); zct_init = true; }
+            }
+            size_t bound = (size_t)len + len / 2 + 64;
             size_t base = out.value_comp.size();
             out.value_comp.resize(base + bound);
-            size_t csz = ZSTD_compress(out.value_comp.data() + base, bound,
-                                       raw.data() + off, len, w->opts.compression_level);
-            if (ZSTD_isError(csz)) { ok = false; break; }
-            out.value_comp.resize(base + csz);
+            int csz = zr_compress(raw.data() + off, (int)len,
+                                  out.value_comp.data() + base, (int)bound, &zct);
+            if (csz <= 0) {
+                /* raw-block frame: magic+fhd+fcs+bh then the bytes */
+                uint8_t *d = out.value_comp.data() + base;
+                int op2 = 0;
+                d[op2++] = 0x28; d[op2++] = 0xB5; d[op2++] = 0x2F; d[op2++] = 0xFD;
+                if (len <= 255) { d[op2++] = 0x20; d[op2++] = (uint8_t)len; }
+                else {
+                    d[op2++] = 0x60;
+                    uint32_t f2 = (uint32_t)len - 256;
+                    d[op2++] = (uint8_t)f2; d[op2++] = (uint8_t)(f2 >> 8);
+                }
+                uint32_t bh = 1u | (0u << 1) | ((uint32_t)len << 3);
+                d[op2++] = (uint8_t)bh; d[op2++] = (uint8_t)(bh >> 8); d[op2++] = (uint8_t)(bh >> 16);
+                memcpy(d + op2, raw.data() + off, len);
+                csz = op2 + (int)len;
+            }
+            out.value_comp.resize(base + (size_t)csz);
             s.comp_len = (uint32_t)csz;
+            s.decomp_len |= CSF_SEGMODE_ZR << 24;
         } else {
             ok = false; break;
         }

@@ -124,6 +124,10 @@ typedef struct csf_seg {
 #define CSF_SEGMODE_P_BASE  0x10u  /* 0x10|L, L in 1..4 */
 #define CSF_SEGMODE_CONST   0x20u
 #define CSF_SEGMODE_LIT     0x30u
+#define CSF_SEGMODE_ZR      0x50u  /* zstd: RESTRICTED frame this writer
+                                    * emitted (raw literals + predefined-FSE
+                                    * sequences, zstd_r.h) — device-decodable;
+                                    * untagged zstd frames host-decode */
 #define CSF_SEG_DLEN_MASK   0x00FFFFFFu
 
 static inline uint32_t csf_seg_dlen(const csf_seg *s) { return s->decomp_len & CSF_SEG_DLEN_MASK; }

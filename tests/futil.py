@@ -13,6 +13,7 @@ SEGMODE_GENERIC = 0x00
 SEGMODE_P_BASE = 0x10
 SEGMODE_CONST = 0x20
 SEGMODE_LIT = 0x30
+SEGMODE_ZR = 0x50
 
 
 def read_footer(path):

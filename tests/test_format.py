@@ -467,3 +467,45 @@ def test_pglz_stream_fuzz():
             with oracle.OracleTable(pth) as t:
                 v, e = read_all(t, 0, n, np.int8, 10000, stripe_rows=150000)
                 np.testing.assert_array_equal(v.view(np.uint8), data)
+
+
+def test_zstd_restricted_system_decodable(tmp_path):
+    """The writer's zstd chunks are RESTRICTED frames (raw literals +
+    predefined-FSE sequences, zstd_r.h) tagged CSF_SEGMODE_ZR for the GPU
+    lane decoder — and they are STANDARD zstd: the system ZSTD_decompress
+    (the call the reference's DecompressBuffer makes,
+    columnar_compression.c:207) must reproduce the raw stream exactly."""
+    n = 25000
+    cases = {
+        "li": (np.arange(n, dtype=np.int64) * 7) % 5000,
+        "runs": np.repeat(np.arange(n // 50, dtype=np.int64), 50),
+        "rand": RNG.integers(-2**62, 2**62, n).astype(np.int64),
+    }
+    zstd = C.CDLL("libzstd.so.1")
+    zstd.ZSTD_decompress.restype = C.c_size_t
+    zstd.ZSTD_decompress.argtypes = [C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
+    zstd.ZSTD_isError.restype = C.c_uint
+    zstd.ZSTD_isError.argtypes = [C.c_size_t]
+    for name, a in cases.items():
+        path = str(tmp_path / f"zr_{name}.cs")
+        ca.write_table(path, [("a", ca.I64, 0)], [np.ascontiguousarray(a)],
+                       compression=ca.COMP_ZSTD)
+        foot = futil.read_footer(path)
+        node = foot["stripes"][0]["nodes"][0][0]
+        if node["comp_type"] != ca.COMP_ZSTD:
+            assert name == "rand"      # incompressible -> raw NONE (ref rule)
+            continue
+        assert all(s["mode"] == futil.SEGMODE_ZR for s in node["segs"]), name
+        comp = futil.chunk_stream(path, node)
+        out = bytearray(node["decompressed_size"])
+        pos = 0
+        for seg in node["segs"]:
+            frame = comp[seg["comp_off"]:seg["comp_off"] + seg["comp_len"]]
+            buf = C.create_string_buffer(seg["decomp_len"])
+            r = zstd.ZSTD_decompress(buf, seg["decomp_len"], bytes(frame), len(frame))
+            assert not zstd.ZSTD_isError(r) and r == seg["decomp_len"], \
+                f"{name}: libzstd rejected the restricted frame"
+            out[seg["decomp_off"]:seg["decomp_off"] + seg["decomp_len"]] = buf.raw
+            pos += seg["decomp_len"]
+        nrows0 = node["decompressed_size"] // 8   # first chunk only
+        assert bytes(out) == a[:nrows0].tobytes(), name
