@@ -149,6 +149,12 @@ extern "C" int csbench_gen_lineitem_shards(const char *dir, uint64_t total_rows,
     std::atomic<int> rc{CSTRIPE_OK};
     const uint64_t per = total_rows / n_shards;
     auto worker = [&]() {
+#ifdef _OPENMP
+        /* shard workers ARE the parallelism: without this every worker
+         * spawns its own full-width OMP team inside flush_stripe's
+         * parallel-for -> thousands of threads, ~10x slower generation */
+        omp_set_num_threads(1);
+#endif
         for (;;) {
             uint32_t i = next.fetch_add(1);
             if (i >= n_shards || rc.load() != CSTRIPE_OK) return;

@@ -432,12 +432,12 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
              * the GPU lane decoder handles (raw literals + predefined-FSE
              * sequences); libzstd-decodable, pinned in tests. Falls back to
              * a raw-block frame when the parse does not shrink a segment. */
-            static zr_ctables zct;
-            static bool zct_init = false;
-            #pragma omp critical(zr_ct)
-            {
-                if (!zct_init) { zr_build_ctables(&zct); zct_init = true; }
-            }
+            /* magic static: thread-safe one-time build (C++11) */
+            static const zr_ctables zct = [] {
+                zr_ctables t;
+                zr_build_ctables(&t);
+                return t;
+            }();
             size_t bound = (size_t)len + len / 2 + 64;
             size_t base = out.value_comp.size();
             out.value_comp.resize(base + bound);
