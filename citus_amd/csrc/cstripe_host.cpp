@@ -389,8 +389,8 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     size_t per = (target / 16) * 16;    /* exact target (16B-aligned splits) —
                                          * uniform segments are what the fused
                                          * GPU kernel's tile math assumes */
-    if (codec == CSTRIPE_COMP_ZSTD && per < 1024)
-        per = 1024;                     /* amortize the ~14 B frame overhead */
+    if (codec == CSTRIPE_COMP_ZSTD && per < 512)
+        per = 512;                      /* ~14 B frame overhead vs lane count */
     if (per == 0) per = n;
 
     out.value_comp.clear();

@@ -473,17 +473,24 @@ static inline int zr_compress(const uint8_t *src, int slen,
 
 /* ---------------- shared host/device decoder (restricted frames) -------- */
 
-/* backward bit reader over [src, src+len) */
+/* backward bit reader over [src, src+len) with a cached 64-bit window —
+ * the cursor only moves DOWN, so one unaligned 8-byte load serves several
+ * reads (the naive per-read byte loop was the device decoder's bottleneck:
+ * ~5 dependent global loads per bit-read) */
 typedef struct {
     const uint8_t *src;
     int32_t len;
     int32_t bitpos;       /* bits remaining below the cursor */
+    int32_t wbase;        /* byte index of the window's LSB (-1 = empty) */
+    uint64_t w;
 } zr_br;
 
 ZR_HOSTDEV static inline int zr_br_init(zr_br *b, const uint8_t *src, int len)
 {
     b->src = src;
     b->len = len;
+    b->wbase = -1;
+    b->w = 0;
     int last = len - 1;
     while (last >= 0 && src[last] == 0) last--;
     if (last < 0) return -1;
@@ -499,11 +506,26 @@ ZR_HOSTDEV static inline uint32_t zr_br_read(zr_br *b, int nbits)
     b->bitpos -= nbits;
     const int32_t bp = b->bitpos < 0 ? 0 : b->bitpos;
     const int32_t byte = bp >> 3;
-    uint64_t w = 0;
-    for (int i = 0; i < 5; i++)
-        if (byte + i < b->len)
-            w |= (uint64_t)b->src[byte + i] << (8 * i);
-    return (uint32_t)(w >> (bp & 7)) & ((1u << nbits) - 1u);
+    if (b->wbase < 0 || byte < b->wbase ||
+        (bp - b->wbase * 8) + nbits > 64) {
+        int32_t base = byte - 3;          /* room for a few more reads below */
+        if (base > b->len - 8) base = b->len - 8;
+        if (base < 0) base = 0;
+        if (b->len >= 8) {
+            uint64_t w;
+            __builtin_memcpy(&w, b->src + base, 8);
+            b->w = w;
+        } else {
+            uint64_t w = 0;
+            for (int i = 0; i < b->len; i++)
+                w |= (uint64_t)b->src[i] << (8 * i);
+            b->w = w;
+            base = 0;
+        }
+        b->wbase = base;
+    }
+    const int32_t sh = bp - b->wbase * 8;
+    return (uint32_t)(b->w >> sh) & ((1u << nbits) - 1u);
 }
 
 /* decode ONE restricted frame; returns decompressed size or -1 */
