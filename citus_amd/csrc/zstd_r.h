@@ -605,13 +605,40 @@ ZR_HOSTDEV static inline int zr_decode_frame(const uint8_t *src, int slen,
         if (of_val <= 3) return -1;       /* repeat offsets not emitted */
         const uint32_t off = of_val - 3;
         if (lp + (int)ll > lit_size || op + (int)(ll + ml) > content) return -1;
-        for (uint32_t i = 0; i < ll; i++) dst[op + (int)i] = lits[lp + (int)i];
+        {   /* word-wise literal copy (unaligned 8B ok on gfx950 and host) */
+            uint32_t i = 0;
+            for (; i + 8 <= ll; i += 8) {
+                uint64_t v;
+                __builtin_memcpy(&v, lits + lp + (int)i, 8);
+                __builtin_memcpy(dst + op + (int)i, &v, 8);
+            }
+            for (; i < ll; i++) dst[op + (int)i] = lits[lp + (int)i];
+        }
         op += (int)ll; lp += (int)ll;
         if (off > (uint32_t)op) return -1;
-        for (uint32_t i = 0; i < ml; i++, op++) dst[op] = dst[op - (int)off];
+        {
+            uint32_t i = 0;
+            if (off >= 8) {               /* no overlap within a word */
+                for (; i + 8 <= ml; i += 8) {
+                    uint64_t v;
+                    __builtin_memcpy(&v, dst + op + (int)i - (int)off, 8);
+                    __builtin_memcpy(dst + op + (int)i, &v, 8);
+                }
+            }
+            for (; i < ml; i++) dst[op + (int)i] = dst[op + (int)i - (int)off];
+            op += (int)ml;
+        }
     }
     const int rem = lit_size - lp;
     if (op + rem != content) return -1;
-    for (int i = 0; i < rem; i++) dst[op + i] = lits[lp + i];
+    {
+        int i = 0;
+        for (; i + 8 <= rem; i += 8) {
+            uint64_t v;
+            __builtin_memcpy(&v, lits + lp + i, 8);
+            __builtin_memcpy(dst + op + i, &v, 8);
+        }
+        for (; i < rem; i++) dst[op + i] = lits[lp + i];
+    }
     return content;
 }
