@@ -273,10 +273,15 @@ def main():
                                       result[1].count == exp_cnt)
                       else "MISMATCH")
         elif args.query == "q1":
+            # generator flags are real char(1) varlena columns: group keys
+            # are the payload bytes; expected_q1 indexes map to those chars
+            RF = {0: ord("A"), 1: ord("N"), 2: ord("R")}
+            LS = {0: ord("O"), 1: ord("F")}
             exp = {}
             for i, nrows in enumerate(shard_rows):
                 for k, v in ca.expected_q1(nrows, base_seed + i).items():
-                    cur = exp.setdefault(k, [0, 0, 0, 0, 0])
+                    kk = (RF[k[0]], LS[k[1]])
+                    cur = exp.setdefault(kk, [0, 0, 0, 0, 0])
                     for j in range(5):
                         cur[j] += v[j]
             exp = {k: v for k, v in exp.items() if v[4] > 0}

@@ -23,7 +23,7 @@ if not os.path.exists(_LIB_PATH):
 _lib = C.CDLL(_LIB_PATH)
 
 # ---- enums (include/cstripe.h) ----
-I8, I16, I32, I64, F32, F64 = 1, 2, 3, 4, 5, 6
+I8, I16, I32, I64, F32, F64, TEXT = 1, 2, 3, 4, 5, 6, 7
 COMP_NONE, COMP_PGLZ, COMP_LZ4, COMP_ZSTD = 0, 1, 2, 3
 PRED_LT, PRED_LE, PRED_GT, PRED_GE, PRED_EQ, PRED_NE = range(6)
 (AGG_COUNT_STAR, AGG_COUNT_COL, AGG_SUM_I64, AGG_SUM_F64,
@@ -195,6 +195,29 @@ def default_options(**kw):
     for k, v in kw.items():
         setattr(o, k, v)
     return o
+
+
+def text_slot(sval):
+    """4-byte short-varlena slot (the reference's stored datum bytes) for a
+    text value of <= 3 payload bytes: hdr = ((len+1)<<1)|1, payload, zero pad.
+    Returns the slot as a little-endian u32 (use for TEXT predicates)."""
+    b = sval.encode() if isinstance(sval, str) else bytes(sval)
+    assert 1 <= len(b) <= 3
+    slot = bytes([((len(b) + 1) << 1) | 1]) + b + b"\0" * (3 - len(b))
+    return int.from_bytes(slot, "little")
+
+
+def text_slots(values):
+    """numpy uint32 array of varlena slots for a sequence of short strings"""
+    import numpy as np
+    return np.array([text_slot(v) for v in values], dtype=np.uint32)
+
+
+def slot_text(slot):
+    """decode a u32 varlena slot back to its string"""
+    b = int(slot).to_bytes(4, "little")
+    n = (b[0] >> 1) - 1
+    return b[1:1 + n].decode()
 
 
 def make_coldefs(defs):

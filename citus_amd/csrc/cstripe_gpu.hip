@@ -527,6 +527,7 @@ __device__ inline bool col_value(const uint8_t *__restrict__ data,
         case CSTRIPE_I32: iv = ((const int32_t *)base)[idx]; fv = (double)iv; break;
         case CSTRIPE_I64: iv = ((const int64_t *)base)[idx]; fv = (double)iv; break;
         case CSTRIPE_F32: fv = ((const float  *)base)[idx]; iv = 0; break;
+        case CSTRIPE_TEXT: iv = (int64_t)((const uint32_t *)base)[idx]; fv = 0; break;
         default:          fv = ((const double *)base)[idx]; iv = 0; break;
     }
     return true;
@@ -1021,6 +1022,12 @@ __device__ inline void col_multi(const uint8_t *__restrict__ data,
             }
             break;
         }
+        case CSTRIPE_TEXT: {
+            const uint32_t *p = (const uint32_t *)base + row;
+            #pragma unroll
+            for (int k = 0; k < R; k++) q.v[k] = (int64_t)p[k];
+            break;
+        }
         default: {                                    /* F64 bits */
             const uint8_t *p = base + (size_t)row * 8;
             #pragma unroll
@@ -1412,11 +1419,20 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
         col_multi<R>(data, scratch, cols[gp.gproj[0]], row, k0);
         if (gp.n_group_cols > 1)
             col_multi<R>(data, scratch, cols[gp.gproj[1]], row, k1);
+        const bool t0 = cols[gp.gproj[0]].type == CSTRIPE_TEXT;
+        const bool t1 = gp.n_group_cols > 1 &&
+                        cols[gp.gproj[1]].type == CSTRIPE_TEXT;
         uint32_t key[R];
         #pragma unroll
         for (int k = 0; k < R; k++) {
-            key[k] = (uint32_t)k0.v[k] & 0xFF;
-            if (gp.n_group_cols > 1) key[k] |= ((uint32_t)k1.v[k] & 0xFF) << 9;
+            uint32_t e0 = (uint32_t)k0.v[k];
+            if (t0) e0 >>= 8;                 /* TEXT: first payload byte */
+            key[k] = e0 & 0xFF;
+            if (gp.n_group_cols > 1) {
+                uint32_t e1 = (uint32_t)k1.v[k];
+                if (t1) e1 >>= 8;
+                key[k] |= (e1 & 0xFF) << 9;
+            }
         }
         /* slot per row: register cache, else LDS claim (CAS linear probe) */
         uint32_t slot[R];
@@ -1587,9 +1603,13 @@ __global__ __launch_bounds__(AGG_BLOCK) void grouped_agg_kernel(
                 for (uint32_t gc = 0; gc < gp.n_group_cols; gc++) {
                     /* NULL keys form their own group (reference
                      * HashAggregate treats NULLs as equal): 9-bit encoding,
-                     * bit 8 = null */
-                    bool ok = col_value(data, scratch, rank, cols[gp.gproj[gc]], row, kv, kf);
-                    uint32_t enc = ok ? ((uint32_t)kv & 0xFF) : CSTRIPE_GROUP_KEY_NULL;
+                     * bit 8 = null. TEXT keys group by the first payload
+                     * byte of the varlena slot (char(1) semantics). */
+                    const ColLoc &kc = cols[gp.gproj[gc]];
+                    bool ok = col_value(data, scratch, rank, kc, row, kv, kf);
+                    uint32_t raw = (uint32_t)kv;
+                    if (kc.type == CSTRIPE_TEXT) raw >>= 8;
+                    uint32_t enc = ok ? (raw & 0xFF) : CSTRIPE_GROUP_KEY_NULL;
                     key |= enc << (9 * gc);
                 }
             }
@@ -2571,7 +2591,8 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
         for (uint32_t i = 0; i < n_group_cols; i++) {
             int pj = proj_of((int32_t)group_cols[i]);
             if (pj < 0) { cs_set_err("group col %u not projected", group_cols[i]); return CSTRIPE_ERR_ARG; }
-            if (r->cols[group_cols[i]].type != CSTRIPE_I8) { cs_set_err("group col %u must be I8", group_cols[i]); return CSTRIPE_ERR_ARG; }
+            if (r->cols[group_cols[i]].type != CSTRIPE_I8 &&
+                r->cols[group_cols[i]].type != CSTRIPE_TEXT) { cs_set_err("group col %u must be I8 or TEXT", group_cols[i]); return CSTRIPE_ERR_ARG; }
             gp.gproj[i] = (uint32_t)pj;
         }
         gp.n_work = g->n_groups * p.tiles_per_group;

@@ -618,6 +618,21 @@ extern "C" int cstripe_write_rows(cstripe_writer *w, uint64_t n_rows,
             uint32_t width = type_width(w->cols[c].type);
             const uint8_t *src = (const uint8_t *)values[c] + done * width;
             const uint8_t *nl = (nulls && nulls[c]) ? nulls[c] + done : nullptr;
+            if (w->cols[c].type == CSTRIPE_TEXT) {
+                /* slots must be well-formed short varlena (hdr odd, total
+                 * length 2..4 incl. header: payload <= 3 bytes) */
+                for (uint64_t i = 0; i < span; i++) {
+                    if (nl && nl[i]) continue;
+                    const uint8_t hdr = src[i * 4];
+                    const uint32_t tot = hdr >> 1;
+                    if (!(hdr & 1) || tot < 2 || tot > 4) {
+                        cs_set_err("text slot row %llu: not a short varlena "
+                                   "with payload <= 3 bytes (hdr 0x%02x)",
+                                   (unsigned long long)(done + i), hdr);
+                        return CSTRIPE_ERR_ARG;
+                    }
+                }
+            }
             size_t base = cc.exists.size();
             cc.exists.resize(base + span);
             if (!nl) {

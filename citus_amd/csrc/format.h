@@ -169,6 +169,7 @@ static inline uint32_t csf_type_width(uint8_t t)
         case 4: return 8;          /* I64 */
         case 5: return 4;          /* F32 */
         case 6: return 8;          /* F64 */
+        case 7: return 4;          /* TEXT: short-varlena slot (hdr+<=3B) */
         default: return 0;
     }
 }

@@ -80,6 +80,17 @@ typedef enum cstripe_type {
     CSTRIPE_I64 = 4,   /* also NUMERIC(15,2) as fixed-point int64 (scale in schema) */
     CSTRIPE_F32 = 5,
     CSTRIPE_F64 = 6,
+    /* short varlena text (char(1)/bpchar up to 3 payload bytes): each value
+     * is stored EXACTLY as the reference serializes a short-form varlena
+     * datum — 1-byte header ((total_len << 1) | 1), payload, padded to the
+     * 4-byte att_align_nominal boundary (SerializeSingleDatum,
+     * columnar_writer.c:555-585; read back by fetch_att/att_addlength_datum,
+     * columnar_reader.c:1542-1572). With payload <= 3 every slot is exactly
+     * 4 bytes, so the device reads the stream as fixed-stride u32 slots.
+     * The ABI passes/returns those raw 4-byte slots; predicates compare
+     * whole slots (ival = the constant's slot); GROUP BY keys are the first
+     * payload byte (char(1) semantics). */
+    CSTRIPE_TEXT = 7,
 } cstripe_type;
 
 /* compression codec per column chunk — columnar_compression.c:62-270.

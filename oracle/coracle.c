@@ -378,6 +378,7 @@ static int o_get(const ochunkcol *cc, uint32_t row, int64_t *iv, double *fv)
         case CSTRIPE_I32: { int32_t v; memcpy(&v, p, 4); *iv = v; *fv = (double)v; break; }
         case CSTRIPE_I64: { int64_t v; memcpy(&v, p, 8); *iv = v; *fv = (double)v; break; }
         case CSTRIPE_F32: { float v; memcpy(&v, p, 4); *fv = v; *iv = 0; break; }
+        case CSTRIPE_TEXT: { uint32_t v; memcpy(&v, p, 4); *iv = (int64_t)v; *fv = 0; break; }
         default:          { double v; memcpy(&v, p, 8); *fv = v; *iv = 0; break; }
     }
     return 1;
@@ -540,12 +541,17 @@ int oracle_scan_agg(oracle_table *t, uint64_t cols_mask,
                     int64_t iv; double fv;
                     for (uint32_t gcn = 0; gcn < n_group_cols; gcn++) {
                         /* NULL keys form their own group (HashAggregate
-                         * groups NULLs together); 9-bit enc, bit 8 = null */
+                         * groups NULLs together); 9-bit enc, bit 8 = null.
+                         * TEXT: first payload byte of the varlena slot. */
                         uint32_t enc;
-                        if (o_get(&cc[group_cols[gcn]], row, &iv, &fv))
-                            enc = (uint32_t)(iv & 0xFF);
-                        else
+                        if (o_get(&cc[group_cols[gcn]], row, &iv, &fv)) {
+                            uint32_t raw = (uint32_t)iv;
+                            if (cc[group_cols[gcn]].type == CSTRIPE_TEXT)
+                                raw >>= 8;
+                            enc = raw & 0xFF;
+                        } else {
                             enc = CSTRIPE_GROUP_KEY_NULL;
+                        }
                         key |= enc << (9 * gcn);
                     }
                     uint32_t gi = n_groups;

@@ -509,3 +509,57 @@ def test_zstd_restricted_system_decodable(tmp_path):
             pos += seg["decomp_len"]
         nrows0 = node["decompressed_size"] // 8   # first chunk only
         assert bytes(out) == a[:nrows0].tobytes(), name
+
+
+def test_text_varlena_layout(tmp_path):
+    """TEXT columns store EXACTLY the reference's short-varlena serialization
+    (SerializeSingleDatum: 1-byte header (total<<1)|1, payload, zero pad to
+    the 4-byte att_align_nominal boundary) — byte-level pin on the stream."""
+    vals = ["A", "N", "R", "xy", "abc", "Z"]
+    slots = ca.text_slots(vals)
+    path = str(tmp_path / "t.cs")
+    ca.write_table(path, [("f", ca.TEXT, 0)], [slots], compression=ca.COMP_NONE)
+    foot = futil.read_footer(path)
+    node = foot["stripes"][0]["nodes"][0][0]
+    stream = futil.chunk_stream(path, node)
+    expect = b""
+    for v in vals:
+        b = v.encode()
+        expect += bytes([((len(b) + 1) << 1) | 1]) + b + b"\0" * (3 - len(b))
+    assert stream == expect
+    # decode helpers round-trip
+    assert [ca.slot_text(sl) for sl in slots] == vals
+
+
+def test_text_group_and_pred(tmp_path):
+    n = 9000
+    flags = np.array([["A", "N", "R"][i % 3] for i in range(n)])
+    slots = ca.text_slots(flags)
+    v = np.arange(n, dtype=np.int64)
+    nulls = (np.arange(n) % 11 == 0).astype(np.uint8)
+    path = str(tmp_path / "g.cs")
+    ca.write_table(path, [("f", ca.TEXT, 0), ("v", ca.I64, 0)], [slots, v],
+                   nulls=[nulls, None], compression=ca.COMP_LZ4,
+                   chunk_group_row_limit=2000)
+    with oracle.OracleTable(path) as t:
+        # group by the char(1) column; NULL flags form their own group
+        res, _ = t.scan_agg([], [(ca.AGG_SUM_I64, 1), (ca.AGG_COUNT_STAR, -1)],
+                            group_cols=(0,))
+        for ch in "ANR":
+            mask = (flags == ch) & (nulls == 0)
+            key = (ord(ch), 0)
+            assert res[key][0].i128 == int(v[mask].sum())
+            assert res[key][1].count == int(mask.sum())
+        nmask = nulls == 1
+        assert res[(None, 0)][1].count == int(nmask.sum())
+        # predicate: whole-slot equality (f = 'R')
+        parts, _ = t.scan_agg([(0, ca.PRED_EQ, ca.text_slot("R"))],
+                              [(ca.AGG_COUNT_STAR, -1)])
+        assert parts[0].count == int(((flags == "R") & (nulls == 0)).sum())
+
+
+def test_text_bad_slot_rejected(tmp_path):
+    import numpy as np
+    bad = np.array([0x00000004], dtype=np.uint32)   # even header: not varlena
+    with pytest.raises(ca.CStripeError, match="short varlena"):
+        ca.write_table(str(tmp_path / "b.cs"), [("f", ca.TEXT, 0)], [bad])
