@@ -30,7 +30,7 @@ sys.path.insert(0, REPO)
 # algorithmic bytes/row = decompressed projected column widths + exists bits,
 # counted once
 ALG_BYTES = {"q6": 32.5,       # 4 int64 + 4 exists bits
-             "q1": 42.875,     # 5 int64 + 2 i8 + 7 exists bits
+             "q1": 48.875,     # 5 int64 + 2 varlena char(1) slots + 7 exists bits
              "count": 8.125}   # 1 int64 + 1 exists bit
 HBM_PEAK_GBPS = 8000.0     # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
 
