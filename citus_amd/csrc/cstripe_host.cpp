@@ -392,6 +392,11 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     if (codec == CSTRIPE_COMP_ZSTD && per < 512)
         per = 512;                      /* ~14 B frame overhead vs lane count */
     if (per == 0) per = n;
+    /* caps: segment decomp_len carries the mode byte in bits 24-31 (8 MB is
+     * comfortably under 2^24), and a restricted-zstd RAW-block fallback is
+     * bounded by the 21-bit block size (1 MB) */
+    if (per > (1u << 23)) per = 1u << 23;
+    if (codec == CSTRIPE_COMP_ZSTD && per > (1u << 20)) per = 1u << 20;
 
     out.value_comp.clear();
     out.segs.clear();
@@ -933,7 +938,7 @@ extern "C" cstripe_scan *cstripe_scan_begin(cstripe_reader *r, uint64_t cols_mas
         /* normalize: every predicate belongs to exactly one OR group;
          * standalone conjuncts each get a private id */
         if (s->preds.back().or_group == 0)
-            s->preds.back().or_group = 0x40000000u + i;
+            s->preds.back().or_group = 0xFFFFFF00u + i;   /* private ids */
         s->cols_mask |= 1ull << preds[i].column;   /* pred columns must be read */
     }
     /* contiguous groups (stable: original order kept within a group) */

@@ -416,7 +416,7 @@ static void o_norm_preds(cstripe_pred *dst, const cstripe_pred *src, uint32_t n)
 {
     for (uint32_t i = 0; i < n; i++) {
         dst[i] = src[i];
-        if (dst[i].or_group == 0) dst[i].or_group = 0x40000000u + i;
+        if (dst[i].or_group == 0) dst[i].or_group = 0xFFFFFF00u + i;
     }
     /* insertion sort (n <= 16), stable */
     for (uint32_t i = 1; i < n; i++) {
