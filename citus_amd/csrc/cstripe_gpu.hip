@@ -360,7 +360,26 @@ __device__ inline void lz4_lane_decode(const uint8_t *__restrict__ data,
              * so quads only when offset>=4; sources are then complete. */
             const uint8_t *msrc = sout + op - offset;
             uint32_t j = 0;
-            if (offset >= 8) {
+            if (((offset | op) & 3) == 0) {
+                /* word-granular stream (e.g. the 4-byte varlena slots of
+                 * text columns): source, destination and offset are all
+                 * 4-aligned, so copies run as aligned LDS words — the
+                 * u32-splat case (offset 4: one repeated slot) is the
+                 * dominant shape of low-cardinality flag columns */
+                uint32_t *d32 = (uint32_t *)(sout + op);
+                if (offset == 4) {
+                    const uint32_t v = *(const uint32_t *)msrc;
+                    for (; j + 4 <= mlen; j += 4) d32[j >> 2] = v;
+                } else {
+                    const uint32_t *s32 = (const uint32_t *)msrc;
+                    const uint32_t woff = offset >> 2;
+                    for (; j + 4 <= mlen; j += 4) {
+                        /* overlap-safe: reads trail writes by woff words */
+                        d32[j >> 2] = (j >> 2) < woff ? s32[j >> 2]
+                                                      : d32[(j >> 2) - woff];
+                    }
+                }
+            } else if (offset >= 8) {
                 for (; j + 8 <= mlen; j += 8) {
                     uint8_t b0 = msrc[j], b1 = msrc[j + 1];
                     uint8_t b2 = msrc[j + 2], b3 = msrc[j + 3];
@@ -371,8 +390,7 @@ __device__ inline void lz4_lane_decode(const uint8_t *__restrict__ data,
                     sout[op + j + 4] = b4; sout[op + j + 5] = b5;
                     sout[op + j + 6] = b6; sout[op + j + 7] = b7;
                 }
-            }
-            if (offset >= 4) {
+            } else if (offset >= 4) {
                 for (; j + 4 <= mlen; j += 4) {
                     uint8_t b0 = msrc[j], b1 = msrc[j + 1];
                     uint8_t b2 = msrc[j + 2], b3 = msrc[j + 3];
