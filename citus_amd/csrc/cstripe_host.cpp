@@ -422,9 +422,13 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
             int bound = LZ4_compressBound((int)len);
             size_t base = out.value_comp.size();
             out.value_comp.resize(base + (size_t)bound);
-            int csz = lz4e_compress_mm(&enc, raw.data() + off,
-                                       (int)len, out.value_comp.data() + base, bound,
-                                       w->opts.lz4_min_match >= 4 ? w->opts.lz4_min_match : 4);
+            /* width-granular matches for 4-byte slot streams (TEXT/I32/F32):
+             * the GPU lane decoder's aligned-word path then covers every copy */
+            const int malign = (width == 4) ? 4 : 1;
+            int csz = lz4e_compress_mm_a(&enc, raw.data() + off,
+                                         (int)len, out.value_comp.data() + base, bound,
+                                         w->opts.lz4_min_match >= 4 ? w->opts.lz4_min_match : 4,
+                                         malign);
             if (csz <= 0)             /* capacity fallback: system liblz4 */
                 csz = LZ4_compress_default((const char *)raw.data() + off,
                                            (char *)out.value_comp.data() + base,

@@ -48,9 +48,14 @@ static inline uint32_t lz4e_read32(const uint8_t *p)
 /* returns compressed size, or 0 if dst capacity insufficient.
  * min_match (>= 4): matches shorter than this are skipped — fewer, longer
  * sequences decode much faster on the lane-parallel GPU decoder at a small
- * ratio cost (the output is standard LZ4 either way). */
-static inline int lz4e_compress_mm(lz4e_state *st, const uint8_t *src, int slen,
-                                   uint8_t *dst, int dcap, int min_match)
+ * ratio cost (the output is standard LZ4 either way).
+ * align (1, 2 or 4): restrict match positions/offsets/lengths to multiples
+ * of the column's value width, so every copy in the stream is word-granular
+ * — the GPU lane decoder then runs its aligned-word path throughout (the
+ * dominant win for 4-byte varlena slot streams). Standard LZ4 either way. */
+static inline int lz4e_compress_mm_a(lz4e_state *st, const uint8_t *src, int slen,
+                                     uint8_t *dst, int dcap, int min_match,
+                                     int align)
 {
     const uint32_t base = st->base;
     st->base += (uint32_t)slen + 1;
@@ -65,18 +70,21 @@ static inline int lz4e_compress_mm(lz4e_state *st, const uint8_t *src, int slen,
         const int matchlimit = slen - 5;         /* matches may extend to here */
         int pos = 0;
         while (pos <= mflimit) {
+            if (align > 1 && (pos % align) != 0) { pos++; continue; }
             uint32_t cur32 = lz4e_read32(src + pos);
             uint32_t h = lz4e_hash(cur32);
             uint32_t cand = st->table[h];
             st->table[h] = base + (uint32_t)pos;
             int mpos = (int)(cand - base);       /* may be negative/garbage */
             if (cand >= base && mpos < pos && pos - mpos <= 65535 &&
+                (align <= 1 || ((pos - mpos) % align) == 0) &&
                 lz4e_read32(src + mpos) == cur32) {
                 /* extend match */
                 int mlen = 4;
                 while (pos + mlen < matchlimit && src[mpos + mlen] == src[pos + mlen])
                     mlen++;
-                if (mlen < min_match) { pos++; continue; }
+                if (align > 1) mlen -= mlen % align;
+                if (mlen < min_match || mlen < 4) { pos++; continue; }
                 int litlen = pos - anchor;
                 /* emit: token + ext + literals + offset + ext */
                 uint8_t *tok = op++;
@@ -130,6 +138,12 @@ static inline int lz4e_compress_mm(lz4e_state *st, const uint8_t *src, int slen,
         op += litlen;
     }
     return (int)(op - dst);
+}
+
+static inline int lz4e_compress_mm(lz4e_state *st, const uint8_t *src, int slen,
+                                   uint8_t *dst, int dcap, int min_match)
+{
+    return lz4e_compress_mm_a(st, src, slen, dst, dcap, min_match, 1);
 }
 
 static inline int lz4e_compress(lz4e_state *st, const uint8_t *src, int slen,
