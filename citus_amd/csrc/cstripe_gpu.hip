@@ -1367,6 +1367,54 @@ __device__ inline void acc_apply_atomic(ThreadAcc *cell, uint8_t kind, const Pre
 
 #define MGRP_SLOTS 16
 
+/* value-only LDS fold: multi_grouped tracks ONE per-slot row counter (all
+ * operands are dense there, so every agg's count equals the group's row
+ * count) — halves the per-row LDS atomic traffic vs acc_apply_atomic */
+__device__ inline void acc_apply_value_atomic(ThreadAcc *cell, uint8_t kind,
+                                              const PrepAcc &p)
+{
+    switch (kind) {
+        case CSTRIPE_AGG_COUNT_STAR:
+        case CSTRIPE_AGG_COUNT_COL:
+            return;                         /* derived from the slot counter */
+        case CSTRIPE_AGG_MIN_I64:
+            atomicMin((long long *)&cell->lo, (long long)p.lo);
+            return;
+        case CSTRIPE_AGG_MAX_I64:
+            atomicMax((long long *)&cell->lo, (long long)p.lo);
+            return;
+        case CSTRIPE_AGG_SUM_F64:
+            atomicAdd((double *)&cell->f, p.f);
+            return;
+        case CSTRIPE_AGG_MIN_F64:
+        case CSTRIPE_AGG_MAX_F64: {
+            unsigned long long *addr = (unsigned long long *)&cell->f;
+            unsigned long long cur = *addr;
+            const bool want_min = kind == CSTRIPE_AGG_MIN_F64;
+            while (true) {
+                double c;
+                memcpy(&c, &cur, 8);
+                const int cmp = f64cmp_pg(p.f, c);
+                if (want_min ? cmp >= 0 : cmp <= 0) break;
+                unsigned long long nv;
+                memcpy(&nv, &p.f, 8);
+                unsigned long long prev = atomicCAS(addr, cur, nv);
+                if (prev == cur) break;
+                cur = prev;
+            }
+            return;
+        }
+        default: {                          /* i128 sums, carry-exact */
+            unsigned long long old =
+                atomicAdd((unsigned long long *)&cell->lo, (unsigned long long)p.lo);
+            int64_t carry = (old + (unsigned long long)p.lo) < old ? 1 : 0;
+            int64_t hi = p.hi + carry;
+            if (hi) atomicAdd((unsigned long long *)&cell->hi, (unsigned long long)hi);
+            return;
+        }
+    }
+}
+
 template <int NAGGS, int R>
 __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
@@ -1381,11 +1429,13 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
     const uint32_t n_waves = AGG_BLOCK / WAVE;
 
     __shared__ int wkeys[AGG_BLOCK / WAVE][MGRP_SLOTS];
+    __shared__ unsigned long long wcnt[AGG_BLOCK / WAVE][MGRP_SLOTS];
     extern __shared__ uint8_t mg_lds[];
     ThreadAcc *wacc = (ThreadAcc *)mg_lds;   /* [wave][slot][agg] */
 
     for (uint32_t i = threadIdx.x; i < n_waves * MGRP_SLOTS; i += AGG_BLOCK) {
         wkeys[i / MGRP_SLOTS][i % MGRP_SLOTS] = -1;
+        wcnt[i / MGRP_SLOTS][i % MGRP_SLOTS] = 0;
         for (uint32_t a = 0; a < n_aggs; a++)
             acc_init(wacc[i * n_aggs + a], params.aggs[a].kind);
     }
@@ -1393,6 +1443,7 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
 
     ThreadAcc *myacc = wacc + (size_t)wid * MGRP_SLOTS * n_aggs;
     int *mykeys = wkeys[wid];
+    unsigned long long *mycnt = wcnt[wid];
     uint32_t ck0 = ~0u, ck1 = ~0u, cs0 = 0, cs1 = 0;   /* key->slot cache */
 
     /* grid-stride over (chunk, tile) work items: tables persist across
@@ -1471,6 +1522,11 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
             }
         }
 
+        /* one row counter per slot (dense: every agg's count equals it) */
+        #pragma unroll
+        for (int k = 0; k < R; k++)
+            if (pv[k]) atomicAdd(&mycnt[slot[k]], 1ull);
+
         /* operand loads once per agg column, contributions folded per row */
         #pragma unroll
         for (uint32_t a = 0; a < n_aggs; a++) {
@@ -1515,7 +1571,7 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
                     }
                     default: pc.valid = false; break;
                 }
-                acc_apply_atomic(&myacc[slot[k] * n_aggs + a], kind, pc);
+                acc_apply_value_atomic(&myacc[slot[k] * n_aggs + a], kind, pc);
             }
         }
     }
@@ -1534,6 +1590,13 @@ __global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
                 uint32_t at = n;
                 for (uint32_t j = 0; j < n; j++) if (bk[j] == (uint32_t)kk) { at = j; break; }
                 ThreadAcc *src = wacc + ((size_t)(w * MGRP_SLOTS) + s2) * n_aggs;
+                const int64_t rowcnt = (int64_t)wcnt[w][s2];
+                for (uint32_t a = 0; a < n_aggs; a++) {
+                    src[a].cnt = rowcnt;
+                    const uint8_t kd = params.aggs[a].kind;
+                    if (kd == CSTRIPE_AGG_COUNT_STAR || kd == CSTRIPE_AGG_COUNT_COL)
+                        src[a].lo = rowcnt;
+                }
                 if (at == n) {
                     bk[n] = (uint32_t)kk;
                     for (uint32_t a = 0; a < n_aggs; a++) {
