@@ -1415,8 +1415,8 @@ __device__ inline void acc_apply_value_atomic(ThreadAcc *cell, uint8_t kind,
     }
 }
 
-template <int NAGGS, int R>
-__global__ __launch_bounds__(AGG_BLOCK, 4) void multi_grouped_kernel(
+template <int NAGGS, int R, int MINW = 4>
+__global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_grouped_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
     const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
     uint32_t *__restrict__ keys_out, AccCell *__restrict__ cells_out,
@@ -2712,11 +2712,23 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
                                    g->d_data, g->d_scratch, g->d_groups, g->d_colloc,
                                    g->d_gkeys, g->d_gcells, g->d_error, gp);
             };
-            if (n_aggs == 5) launchmg(multi_grouped_kernel<5, 8>);
-            else if (n_aggs == 1) launchmg(multi_grouped_kernel<1, 8>);
-            else if (n_aggs == 2) launchmg(multi_grouped_kernel<2, 8>);
-            else if (n_aggs == 4) launchmg(multi_grouped_kernel<4, 8>);
-            else launchmg(multi_grouped_kernel<-1, 4>);
+            static const int gvar = [] {
+                const char *e = getenv("CSTRIPE_GROUPED_VARIANT");
+                return e ? atoi(e) : 0;
+            }();
+            if (gvar == 1) {       /* forced 6 waves/SIMD A/B variant */
+                if (n_aggs == 5) launchmg(multi_grouped_kernel<5, 8, 6>);
+                else if (n_aggs == 1) launchmg(multi_grouped_kernel<1, 8, 6>);
+                else if (n_aggs == 2) launchmg(multi_grouped_kernel<2, 8, 6>);
+                else if (n_aggs == 4) launchmg(multi_grouped_kernel<4, 8, 6>);
+                else launchmg(multi_grouped_kernel<-1, 4>);
+            } else {
+                if (n_aggs == 5) launchmg(multi_grouped_kernel<5, 8>);
+                else if (n_aggs == 1) launchmg(multi_grouped_kernel<1, 8>);
+                else if (n_aggs == 2) launchmg(multi_grouped_kernel<2, 8>);
+                else if (n_aggs == 4) launchmg(multi_grouped_kernel<4, 8>);
+                else launchmg(multi_grouped_kernel<-1, 4>);
+            }
             HIP_TRY(hipGetLastError());
             hipLaunchKernelGGL(grouped_final_kernel, dim3(1), dim3(AGG_BLOCK), 0, g->stream,
                                g->d_gkeys, g->d_gcells, mgrid, mg_per_block,
