@@ -139,6 +139,7 @@ struct cs_gpu_state {
     bool fusable = true;             /* all proj cols dense i64, uniform 256B lz4 segs */
     bool fusable_mixed = true;       /* widths in {1,8}; enables fused grouped */
     bool all_dense = true;           /* no NULLs anywhere -> pair_agg_kernel */
+    bool all_canonP = true;          /* every col canonical P/CONST -> lds kernel */
     struct FusedTile *d_tiles = nullptr;
     uint32_t n_tiles = 0;
     struct FusedTileG *d_tiles2 = nullptr;
@@ -1214,6 +1215,176 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_agg_kernel(
     }
 }
 
+/* =====================================================================
+ * LDS-staged scan (experimental variant): the R-row kernel is latency
+ * bound (SQ_WAIT_ANY ~72% of wave cycles even at 8 waves/SIMD). Here each
+ * block BULK-LOADS its tile's canonical predicate streams into LDS with
+ * coalesced dwordx4 loads (deep MLP, few waits), then rows extract values
+ * at LDS latency. Aggregate operand columns stay in global memory behind
+ * the wave-uniform ballot (rarely touched at Q6 selectivity).
+ * Eligible when every projected column is dense canonical P/CONST.
+ * ===================================================================== */
+
+#define LDSK_TILE 2048
+
+template <int NPREDS, int NAGGS>
+__global__ __launch_bounds__(AGG_BLOCK) void lds_agg_kernel(
+    const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
+    const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
+    AccCell *__restrict__ block_out, uint32_t tiles2_per_group,
+    const AggParams params)
+{
+    const uint32_t n_preds = NPREDS >= 0 ? (uint32_t)NPREDS : params.n_preds;
+    const uint32_t n_aggs = NAGGS >= 0 ? (uint32_t)NAGGS : params.n_aggs;
+    const uint32_t gid = blockIdx.x / tiles2_per_group;
+    const uint32_t tile = blockIdx.x % tiles2_per_group;
+    const GroupDesc g = groups[gid];
+    const ColLoc *cols = colloc + g.colbase;
+
+    const uint32_t row_start = tile * LDSK_TILE;
+    if (row_start >= g.row_count) {     /* empty tile: zero partials */
+        if (threadIdx.x == 0)
+            for (uint32_t a = 0; a < n_aggs; a++) {
+                AccCell c{0, 0, 0.0, 0};
+                ThreadAcc z;
+                acc_init(z, params.aggs[a].kind);
+                c.lo = z.lo; c.hi = z.hi; c.f = z.f; c.cnt = 0;
+                block_out[(uint64_t)blockIdx.x * n_aggs + a] = c;
+            }
+        return;
+    }
+    const uint32_t row_end = min(row_start + LDSK_TILE, g.row_count);
+
+    extern __shared__ uint8_t sld[];
+
+    /* distinct predicate columns (wave-uniform bookkeeping) */
+    uint32_t pcols[4];
+    uint32_t npc = 0;
+    for (uint32_t p = 0; p < n_preds; p++) {
+        const uint32_t pj = params.preds[p].proj;
+        bool seen = false;
+        for (uint32_t i = 0; i < npc; i++) seen |= pcols[i] == pj;
+        if (!seen && npc < 4) pcols[npc++] = pj;
+    }
+
+    /* phase A: stage each pred column's stream window (uniform control) */
+    uint32_t roff[4], rdelta[4], rstep[4], rstart_pos[4];
+    uint64_t rmask[4], rhval[4];
+    uint8_t rmode[4];
+    uint32_t lds_used = 0;
+    for (uint32_t i = 0; i < npc; i++) {
+        const ColLoc cl = cols[pcols[i]];
+        rmode[i] = cl.mode;
+        rhval[i] = (uint64_t)cl.hval;
+        if (cl.mode == CSF_SEGMODE_CONST) {
+            roff[i] = 0; rdelta[i] = 0; rstep[i] = 0; rstart_pos[i] = 0;
+            rmask[i] = 0;
+            continue;
+        }
+        const uint32_t Lx = cl.L, step = Lx + 3u;
+        rstep[i] = step;
+        rmask[i] = (~0ull) >> ((8u - Lx) * 8u);
+        const uint32_t p0 = tile == 0 ? 0u
+                            : row_start * step + (6u - Lx);
+        rstart_pos[i] = p0;
+        const uint64_t src0 = cl.val_off + p0;
+        const uint64_t asrc = src0 & ~15ull;
+        rdelta[i] = (uint32_t)(src0 - asrc);
+        const uint32_t last_pos = (row_end - 1) * step + (6u - Lx);
+        const uint32_t nbytes = (uint32_t)(cl.val_off + last_pos + 8 - asrc + 15) & ~15u;
+        roff[i] = lds_used;
+        /* cooperative 16B copy */
+        const uint4 *gsrc = (const uint4 *)(data + asrc);
+        uint4 *ldst = (uint4 *)(sld + lds_used);
+        for (uint32_t o = threadIdx.x; o < (nbytes >> 4); o += AGG_BLOCK)
+            ldst[o] = gsrc[o];
+        lds_used += nbytes;
+    }
+    __syncthreads();
+
+    ThreadAcc acc[NAGGS >= 0 ? NAGGS : MAX_AGGS];
+    #pragma unroll
+    for (uint32_t a = 0; a < n_aggs; a++) acc_init(acc[a], params.aggs[a].kind);
+
+    /* extraction from LDS: value j of pred-col slot i */
+    auto lval = [&](uint32_t i, uint32_t j) -> int64_t {
+        if (rmode[i] == CSF_SEGMODE_CONST) return (int64_t)rhval[i];
+        uint32_t pos = j * rstep[i] + (9u - rstep[i]);
+        pos = j == 0 ? 1u : (j == 1 ? 9u : pos);
+        const uint32_t rel = rdelta[i] + pos - rstart_pos[i];
+        const uint32_t *l32 = (const uint32_t *)(sld + roff[i]);
+        const uint32_t wi = rel >> 2, sh = (rel & 3u) * 8u;
+        uint64_t lo = ((uint64_t)l32[wi + 1] << 32) | l32[wi];
+        if (sh) lo = (lo >> sh) | ((uint64_t)l32[wi + 2] << (64u - sh));
+        return (int64_t)((lo & rmask[i]) | rhval[i]);
+    };
+
+    constexpr int R = LDSK_TILE / AGG_BLOCK;      /* 8 rows per thread */
+    const uint32_t my0 = row_start + threadIdx.x * R;
+    bool pv[R], gv[R];
+    #pragma unroll
+    for (int k = 0; k < R; k++) {
+        pv[k] = my0 + k < row_end;
+        gv[k] = false;
+    }
+    int64_t v[R];
+    int last_proj = -1;
+    if (my0 < row_end) {
+        #pragma unroll
+        for (uint32_t p = 0; p < n_preds; p++) {
+            const PredD &pr = params.preds[p];
+            if ((int)pr.proj != last_proj) {
+                uint32_t slot = 0;
+                for (uint32_t i = 0; i < npc; i++)
+                    if (pcols[i] == pr.proj) slot = i;
+                #pragma unroll
+                for (int k = 0; k < R; k++) v[k] = lval(slot, my0 + k);
+                last_proj = (int)pr.proj;
+            }
+            #pragma unroll
+            for (int k = 0; k < R; k++)
+                gv[k] = gv[k] | pred_eval(pr, v[k], 0.0);   /* canon P: ints */
+            if (pr.gend) {
+                #pragma unroll
+                for (int k = 0; k < R; k++) { pv[k] = pv[k] & gv[k]; gv[k] = false; }
+            }
+        }
+    } else {
+        #pragma unroll
+        for (int k = 0; k < R; k++) pv[k] = false;
+    }
+    bool any = false;
+    #pragma unroll
+    for (int k = 0; k < R; k++) any |= pv[k];
+    if (__ballot(any) != 0) {
+        /* lanes past row_end load a safe in-bounds base; pv masks them */
+        const uint32_t safe0 = my0 < row_end ? my0 : row_start;
+        #pragma unroll
+        for (uint32_t a = 0; a < n_aggs; a++)
+            acc_multi<R>(acc[a], params.aggs[a], data, scratch, cols, safe0, pv);
+    }
+
+    __shared__ ThreadAcc lred2[AGG_BLOCK / WAVE][MAX_AGGS];
+    const uint32_t wid = threadIdx.x / WAVE;
+    const uint32_t lane = threadIdx.x % WAVE;
+    #pragma unroll
+    for (uint32_t a = 0; a < n_aggs; a++) {
+        wave_reduce(acc[a], params.aggs[a].kind);
+        if (lane == 0) lred2[wid][a] = acc[a];
+    }
+    __syncthreads();
+    if (wid == 0) {
+        for (uint32_t a = lane; a < n_aggs; a += WAVE) {
+            ThreadAcc r = lred2[0][a];
+            for (uint32_t w = 1; w < AGG_BLOCK / WAVE; w++)
+                acc_merge(r, lred2[w][a], params.aggs[a].kind);
+            AccCell c;
+            c.lo = r.lo; c.hi = r.hi; c.f = r.f; c.cnt = r.cnt;
+            block_out[(uint64_t)blockIdx.x * n_aggs + a] = c;
+        }
+    }
+}
+
 /* per-row precomputed aggregate contribution: operands loaded ONCE per row,
  * so the per-distinct-key reduce rounds touch registers only */
 struct PrepAcc {
@@ -2282,13 +2453,20 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         n_tiles_max += (r->stripes[sc.stripe].group_rows[sc.chunk] + 2047) / 2048;
     uint64_t max_blocks = (uint64_t)g->n_groups * tiles_pg;
     if (n_tiles_max > max_blocks) max_blocks = n_tiles_max;
+    {   /* lds_agg_kernel tiles every chunk at the full 2048-row pitch */
+        uint64_t b2 = (uint64_t)g->n_groups *
+                      ((r->head.chunk_row_limit + 2047) / 2048);
+        if (b2 > max_blocks) max_blocks = b2;
+    }
     g->max_blocks = (uint32_t)max_blocks;
 
     HIP_TRY(hipStreamCreate(&g->stream));
     HIP_TRY(hipEventCreate(&g->ev0));
     HIP_TRY(hipEventCreate(&g->ev1));
     HIP_TRY(hipEventCreate(&g->ev2));
-    data_bytes += 48;               /* register-window read slack */
+    data_bytes += 128;              /* window read slack: R-row extraction
+                                     * windows reach up to ~60 B past the
+                                     * last region's stream end */
     if (data_bytes) HIP_TRY(hipMalloc(&g->d_data, data_bytes));
     if (scratch_bytes) HIP_TRY(hipMalloc(&g->d_scratch, scratch_bytes));
     if (rank_words) HIP_TRY(hipMalloc(&g->d_rank, rank_words * 4));
@@ -2362,6 +2540,15 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 cl.flags |= 2;
             }
             if (!(cl.flags & 2)) g->all_dense = false;
+            {
+                const cs_skipnode &nd2 = st.nodes[c][sc.chunk];
+                const bool cp = nd2.n.comp_type == CSTRIPE_COMP_LZ4 &&
+                                nd2.n.n_segs == 1 &&
+                                (nd2.seg_modes[0] == CSF_SEGMODE_CONST ||
+                                 (nd2.seg_modes[0] > CSF_SEGMODE_P_BASE &&
+                                  nd2.seg_modes[0] <= CSF_SEGMODE_P_BASE + 4));
+                if (!cp || !(cl.flags & 2)) g->all_canonP = false;
+            }
 
             const bool canon = nd.n.comp_type == CSTRIPE_COMP_LZ4 &&
                                nd.n.n_segs == 1 &&
@@ -3042,6 +3229,47 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             const char *e = getenv("CSTRIPE_KERNEL_VARIANT");
             return e ? atoi(e) : 1;     /* measured best: R8 + 8 waves/SIMD */
         }();
+        if (kvar == 3 && g->all_canonP) {
+            /* experimental LDS-staged scan: bulk-load pred streams to LDS */
+            uint32_t tiles2 = (r->head.chunk_row_limit + LDSK_TILE - 1) / LDSK_TILE;
+            if (tiles2 == 0) tiles2 = 1;
+            const uint32_t nb2 = g->n_groups * tiles2;
+            uint32_t lds_need = 0;
+            bool lds_ok = true;
+            {
+                bool seen[MAX_PROJ] = {};
+                for (uint32_t i = 0; i < p.n_preds; i++) {
+                    const uint32_t pj = p.preds[i].proj;
+                    if (seen[pj]) continue;
+                    seen[pj] = true;
+                    uint32_t maxstep = 0;
+                    for (uint32_t gi = 0; gi < g->n_groups; gi++) {
+                        const ColLoc &cl = g->colloc_host[(uint64_t)gi * g->n_proj + pj];
+                        if (cl.mode != CSF_SEGMODE_CONST)
+                            maxstep = max(maxstep, (uint32_t)cl.L + 3u);
+                    }
+                    lds_need += LDSK_TILE * maxstep + 64;
+                }
+                if (lds_need > 100 * 1024 || p.n_preds == 0) lds_ok = false;
+            }
+            if (lds_ok) {
+                auto launchl = [&](auto *kern) {
+                    hipLaunchKernelGGL(kern, dim3(nb2), dim3(AGG_BLOCK), lds_need,
+                                       g->stream, g->d_data, g->d_scratch,
+                                       g->d_groups, g->d_colloc, g->d_block,
+                                       tiles2, p);
+                };
+                if (p.n_preds == 5 && n_aggs == 2) launchl(lds_agg_kernel<5, 2>);
+                else if (p.n_preds == 5 && n_aggs == 1) launchl(lds_agg_kernel<5, 1>);
+                else if (p.n_preds == 1 && n_aggs == 1) launchl(lds_agg_kernel<1, 1>);
+                else launchl(lds_agg_kernel<-1, -1>);
+                HIP_TRY(hipGetLastError());
+                HIP_TRY(hipEventRecord(g->ev1, g->stream));
+                { int _rc = launch_final_reduce(g, nb2, p); if (_rc != CSTRIPE_OK) return _rc; }
+                HIP_TRY(hipEventRecord(g->ev2, g->stream));
+                goto collect_ungrouped;
+            }
+        }
         if (kvar == 1) {        /* forced 8 waves/SIMD (64 VGPRs, some spill) */
             if (p.n_preds == 5 && n_aggs == 2) launchp(multi_agg_kernel<5, 2, 8, 8>);
             else if (p.n_preds == 5 && n_aggs == 1) launchp(multi_agg_kernel<5, 1, 8, 8>);
@@ -3082,6 +3310,8 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
     { int _rc = launch_final_reduce(g, n_blocks, p); if (_rc != CSTRIPE_OK) return _rc; }
     HIP_TRY(hipEventRecord(g->ev2, g->stream));
 
+collect_ungrouped:
+    ;
     AccCell h_final[MAX_AGGS];
     int h_err = 0;
     HIP_TRY(hipMemcpyAsync(h_final, g->d_final, n_aggs * sizeof(AccCell), hipMemcpyDeviceToHost, g->stream));
