@@ -127,6 +127,9 @@ struct cs_gpu_state {
     SegDesc *d_zsegs = nullptr;      /* restricted-zstd segments (zstd_r.h) */
     uint32_t n_zsegs = 0;
     zr_dtables *d_zrtab = nullptr;   /* predefined FSE decode tables */
+    uint32_t *d_segstart = nullptr;  /* per (group, proj): first d_segs index */
+    uint64_t greedy256_mask = 0;     /* proj cols stored as uniform 256B lz4 */
+    std::vector<uint8_t> col_scratch;/* per proj: any chunk lands in scratch */
     int *d_error = nullptr;
     hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr;
 
@@ -1463,6 +1466,11 @@ struct GroupParams {
     uint32_t n_group_cols;
     uint32_t gproj[CSTRIPE_MAX_GROUP_COLS];   /* projected slots of key cols */
     uint32_t n_work;                          /* n_groups * tiles_per_group */
+    /* multi_grouped fused key decode: key cols stored as uniform 256 B
+     * greedy-LZ4 segments are decoded into LDS inside the kernel (half-tile
+     * granularity) instead of a scratch round trip */
+    uint32_t kdec_mask;                       /* bit i: key col i from LDS */
+    uint32_t kwidth[CSTRIPE_MAX_GROUP_COLS];  /* value width of key col i */
 };
 
 
@@ -1590,6 +1598,7 @@ template <int NAGGS, int R, int MINW = 4>
 __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_grouped_kernel(
     const uint8_t *__restrict__ data, const uint8_t *__restrict__ scratch,
     const GroupDesc *__restrict__ groups, const ColLoc *__restrict__ colloc,
+    const SegDesc *__restrict__ segs, const uint32_t *__restrict__ segstart,
     uint32_t *__restrict__ keys_out, AccCell *__restrict__ cells_out,
     int *__restrict__ err, const GroupParams gp)
 {
@@ -1598,11 +1607,26 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_grouped_kernel(
     const uint32_t wid = threadIdx.x / WAVE;
     const uint32_t lane = threadIdx.x % WAVE;
     const uint32_t n_waves = AGG_BLOCK / WAVE;
+    constexpr uint32_t HALF = (uint32_t)R * AGG_BLOCK;   /* rows per pass */
 
     __shared__ int wkeys[AGG_BLOCK / WAVE][MGRP_SLOTS];
     __shared__ unsigned long long wcnt[AGG_BLOCK / WAVE][MGRP_SLOTS];
     extern __shared__ uint8_t mg_lds[];
     ThreadAcc *wacc = (ThreadAcc *)mg_lds;   /* [wave][slot][agg] */
+    /* fused key-decode regions follow the accumulator tables: per decoded
+     * key col, ceil(HALF*width/256) lanes x 280 B (the lane-decode stride) */
+    uint32_t kbase[CSTRIPE_MAX_GROUP_COLS] = {0, 0};
+    uint32_t ksegs_half[CSTRIPE_MAX_GROUP_COLS] = {0, 0};
+    {
+        uint32_t off = (uint32_t)(n_waves * MGRP_SLOTS * n_aggs * sizeof(ThreadAcc));
+        off = (off + 15u) & ~15u;
+        for (uint32_t i = 0; i < gp.n_group_cols; i++) {
+            if (!(gp.kdec_mask & (1u << i))) continue;
+            kbase[i] = off;
+            ksegs_half[i] = (HALF * gp.kwidth[i] + 255u) / 256u;
+            off += ksegs_half[i] * 280u;
+        }
+    }
 
     for (uint32_t i = threadIdx.x; i < n_waves * MGRP_SLOTS; i += AGG_BLOCK) {
         wkeys[i / MGRP_SLOTS][i % MGRP_SLOTS] = -1;
@@ -1627,18 +1651,46 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_grouped_kernel(
     const uint32_t row_start = tile * TILE_ROWS;
     const uint32_t row_end = min(row_start + TILE_ROWS, g.row_count);
 
-    for (uint32_t row = row_start + R * threadIdx.x; row < row_end;
-         row += R * AGG_BLOCK) {
+    for (uint32_t half_start = row_start; half_start < row_end;
+         half_start += HALF) {
+        const uint32_t half_end = min(half_start + HALF, row_end);
+
+        /* phase 1: decode this half's key-column segments into LDS (uniform
+         * 256 B greedy-LZ4 segments, one lane each) */
+        if (gp.kdec_mask) {
+            __syncthreads();           /* LDS regions reused across halves */
+            for (uint32_t i = 0; i < gp.n_group_cols; i++) {
+                if (!(gp.kdec_mask & (1u << i))) continue;
+                const uint32_t base_lane = i * 64u;
+                const uint32_t rows_half = half_end - half_start;
+                const uint32_t nseg = (rows_half * gp.kwidth[i] + 255u) / 256u;
+                const uint32_t k = threadIdx.x - base_lane;
+                if (threadIdx.x >= base_lane && k < nseg) {
+                    const uint32_t pj = gp.gproj[i];
+                    const uint32_t seg0 = segstart[(uint64_t)gid * params.n_proj + pj]
+                                          + (half_start * gp.kwidth[i]) / 256u;
+                    const SegDesc sd = segs[seg0 + k];
+                    lz4_lane_decode(data, sd,
+                                    mg_lds + kbase[i] + (size_t)k * 280u, err);
+                }
+            }
+            __syncthreads();
+        }
+
+        /* NOTE: no `continue` below this point — every thread must reach
+         * the next half's __syncthreads (the key-decode LDS is reused) */
+        const uint32_t my0 = half_start + R * threadIdx.x;
         bool pv[R], gv[R];
         #pragma unroll
-        for (int k = 0; k < R; k++) { pv[k] = row + k < row_end; gv[k] = false; }
+        for (int k = 0; k < R; k++) { pv[k] = my0 + k < half_end; gv[k] = false; }
         int last_proj = -1;
         ValsR<R> q{};
         #pragma unroll
         for (uint32_t pp = 0; pp < params.n_preds; pp++) {
             const PredD &pr = params.preds[pp];
             if ((int)pr.proj != last_proj) {
-                col_multi<R>(data, scratch, cols[pr.proj], row, q);
+                col_multi<R>(data, scratch, cols[pr.proj],
+                             my0 < half_end ? my0 : half_start, q);
                 last_proj = (int)pr.proj;
             }
             #pragma unroll
@@ -1652,26 +1704,41 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_grouped_kernel(
         bool any = false;
         #pragma unroll
         for (int k = 0; k < R; k++) any |= pv[k];
-        if (__ballot(any) == 0) continue;
+        if (__ballot(any) != 0) {
 
-        /* group keys (k0 | k1<<8) for the R rows */
-        ValsR<R> k0, k1;
-        col_multi<R>(data, scratch, cols[gp.gproj[0]], row, k0);
-        if (gp.n_group_cols > 1)
-            col_multi<R>(data, scratch, cols[gp.gproj[1]], row, k1);
-        const bool t0 = cols[gp.gproj[0]].type == CSTRIPE_TEXT;
-        const bool t1 = gp.n_group_cols > 1 &&
-                        cols[gp.gproj[1]].type == CSTRIPE_TEXT;
+        /* group keys: LDS-decoded cols read at half-local offsets, others
+         * through col_multi; 9-bit enc (bit 8 = NULL never set: dense) */
         uint32_t key[R];
-        #pragma unroll
-        for (int k = 0; k < R; k++) {
-            uint32_t e0 = (uint32_t)k0.v[k];
-            if (t0) e0 >>= 8;                 /* TEXT: first payload byte */
-            key[k] = e0 & 0xFF;
-            if (gp.n_group_cols > 1) {
-                uint32_t e1 = (uint32_t)k1.v[k];
-                if (t1) e1 >>= 8;
-                key[k] |= (e1 & 0xFF) << 9;
+        {
+            const uint32_t safe0 = my0 < half_end ? my0 : half_start;
+            #pragma unroll
+            for (int k = 0; k < R; k++) key[k] = 0;
+            for (uint32_t i = 0; i < gp.n_group_cols; i++) {
+                const uint32_t pj = gp.gproj[i];
+                const bool is_text = cols[pj].type == CSTRIPE_TEXT;
+                if (gp.kdec_mask & (1u << i)) {
+                    const uint32_t kw = gp.kwidth[i];
+                    #pragma unroll
+                    for (int k = 0; k < R; k++) {
+                        const uint32_t r = (safe0 + k) - half_start;
+                        const uint32_t bo = r * kw;
+                        const uint8_t *pp2 = mg_lds + kbase[i]
+                                             + (bo >> 8) * 280u + (bo & 255u);
+                        uint32_t raw = kw == 4 ? *(const uint32_t *)pp2
+                                               : (uint32_t)*pp2;
+                        if (is_text) raw >>= 8;
+                        key[k] |= (raw & 0xFF) << (9 * i);
+                    }
+                } else {
+                    ValsR<R> kq;
+                    col_multi<R>(data, scratch, cols[pj], safe0, kq);
+                    #pragma unroll
+                    for (int k = 0; k < R; k++) {
+                        uint32_t raw = (uint32_t)kq.v[k];
+                        if (is_text) raw >>= 8;
+                        key[k] |= (raw & 0xFF) << (9 * i);
+                    }
+                }
             }
         }
         /* slot per row: register cache, else LDS claim (CAS linear probe) */
@@ -1699,17 +1766,18 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_grouped_kernel(
             if (pv[k]) atomicAdd(&mycnt[slot[k]], 1ull);
 
         /* operand loads once per agg column, contributions folded per row */
+        const uint32_t safe0 = my0 < half_end ? my0 : half_start;
         #pragma unroll
         for (uint32_t a = 0; a < n_aggs; a++) {
             const AggD &ag = params.aggs[a];
             const uint8_t kind = ag.kind;
             ValsR<R> qa, qb, qc;
             if (kind != CSTRIPE_AGG_COUNT_STAR && kind != CSTRIPE_AGG_COUNT_COL) {
-                col_multi<R>(data, scratch, cols[ag.proj_a], row, qa);
+                col_multi<R>(data, scratch, cols[ag.proj_a], safe0, qa);
                 if (kind >= CSTRIPE_AGG_SUM_PROD_I64)
-                    col_multi<R>(data, scratch, cols[ag.proj_b], row, qb);
+                    col_multi<R>(data, scratch, cols[ag.proj_b], safe0, qb);
                 if (kind == CSTRIPE_AGG_SUM_DISC_TAX_I64)
-                    col_multi<R>(data, scratch, cols[ag.proj_c], row, qc);
+                    col_multi<R>(data, scratch, cols[ag.proj_c], safe0, qc);
             }
             #pragma unroll
             for (int k = 0; k < R; k++) {
@@ -1745,7 +1813,8 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_grouped_kernel(
                 acc_apply_value_atomic(&myacc[slot[k] * n_aggs + a], kind, pc);
             }
         }
-    }
+        }   /* wave has passing rows */
+    }   /* half loop */
     }   /* work loop */
     __syncthreads();
 
@@ -2357,6 +2426,7 @@ void csgpu_release(cstripe_scan *s)
     if (g->d_tmp) HIP_DROP(hipFree(g->d_tmp));
     if (g->d_zsegs) HIP_DROP(hipFree(g->d_zsegs));
     if (g->d_zrtab) HIP_DROP(hipFree(g->d_zrtab));
+    if (g->d_segstart) HIP_DROP(hipFree(g->d_segstart));
     if (g->d_tiles) HIP_DROP(hipFree(g->d_tiles));
     if (g->d_tiles2) HIP_DROP(hipFree(g->d_tiles2));
     if (g->d_error) HIP_DROP(hipFree(g->d_error));
@@ -2495,6 +2565,9 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     std::vector<SegDesc> h_zsegs;
     h_zsegs.reserve(n_zsegs);
     std::vector<uint32_t> seg_start((uint64_t)g->n_groups * n_proj, 0);
+    std::vector<uint8_t> col_g256(n_proj, 1);   /* stays 1 if every chunk is
+                                                 * dense uniform-256B lz4 */
+    g->col_scratch.assign(n_proj, 0);
     if (n_proj > 4) g->fusable = false;
     std::vector<GroupDesc> h_groups(g->n_groups);
     std::vector<ColLoc> h_colloc((uint64_t)g->n_groups * n_proj);
@@ -2564,6 +2637,8 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 for (uint8_t m : nd.seg_modes)
                     if (m != CSF_SEGMODE_ZR) { zr_dev = false; break; }
             if (zr_dev) {
+                col_g256[pj] = 0;
+                g->col_scratch[pj] = 1;
                 dpos = align_up(dpos, 16);
                 memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
                 spos = align_up(spos, 16);
@@ -2581,6 +2656,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 dpos += align_up(nd.n.value_len, 16);
                 spos += align_up(nd.n.decompressed_size, 16);
             } else if (canon) {
+                col_g256[pj] = 0;
                 /* stage the compressed stream; kernels read values straight
                  * out of it (col_value canonical path) */
                 dpos = align_up(dpos, 16);
@@ -2611,6 +2687,8 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 cl.flags |= 1;
                 g->scratch_off[(uint64_t)gi * n_proj + pj] = spos;
                 seg_start[(uint64_t)gi * n_proj + pj] = (uint32_t)h_segs.size();
+                g->col_scratch[pj] = 1;
+                if (!(cl.flags & 2)) col_g256[pj] = 0;
                 uint32_t segi = 0;
                 const auto &seglist = st.nodes[c][sc.chunk].segs;
                 for (const csf_seg &sg : seglist) {
@@ -2624,6 +2702,7 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                                                    : sg.decomp_len > 256)) {
                         g->fusable = false;
                         g->fusable_mixed = false;
+                        col_g256[pj] = 0;
                     }
                     segi++;
                 }
@@ -2641,12 +2720,14 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 dpos += align_up(nd.n.value_len, 16);
                 spos += align_up(nd.n.decompressed_size, 16);
             } else if (nd.n.comp_type == CSTRIPE_COMP_NONE) {
+                col_g256[pj] = 0;
                 dpos = align_up(dpos, 16);
                 cl.val_off = dpos;
                 memcpy(h_data.data() + dpos, stripe_base + nd.n.value_off, nd.n.value_len);
                 dpos += align_up(nd.n.value_len, 16);
             } else if (nd.n.comp_type == CSTRIPE_COMP_ZSTD ||
                        nd.n.comp_type == CSTRIPE_COMP_PGLZ) {
+                col_g256[pj] = 0;
                 /* host-decode at stage (zstd: documented fallback, GPU zstd
                  * is a later item — SURVEY.md §8f(1); pglz is host-only by
                  * design: reference-migrated tables read correctly) */
@@ -2755,6 +2836,15 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         } else {
             g->fusable_mixed = false;
         }
+    }
+    g->greedy256_mask = 0;
+    for (uint32_t pj = 0; pj < n_proj; pj++)
+        if (col_g256[pj]) g->greedy256_mask |= 1ull << pj;
+    if (g->n_groups > 0 && n_proj > 0) {
+        HIP_TRY(hipMalloc(&g->d_segstart, seg_start.size() * 4));
+        HIP_TRY(hipMemcpyAsync(g->d_segstart, seg_start.data(),
+                               seg_start.size() * 4, hipMemcpyHostToDevice,
+                               g->stream));
     }
     HIP_TRY(hipStreamSynchronize(g->stream));
     g->colloc_host = h_colloc;
@@ -2886,17 +2976,43 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
                 int_aggs_g = false;
         if (g->all_dense) {
             /* multi-row grouped kernel (windowed loads + per-wave 16-slot
-             * LDS tables); falls back below on >16 distinct keys (flag 8) */
+             * LDS tables); falls back below on >16 distinct keys (flag 8).
+             * Key columns stored as uniform 256 B greedy-LZ4 segments are
+             * decoded INSIDE the kernel (half-tile LDS staging) — when they
+             * are the only scratch consumers, the standalone decode kernel
+             * and its scratch round trip are skipped entirely. */
             const uint32_t mg_per_block = n_waves * MGRP_SLOTS;
             uint32_t mgrid = gp.n_work < GRP_GRID ? gp.n_work : GRP_GRID;
-            uint32_t mlds = n_waves * MGRP_SLOTS * n_aggs * (uint32_t)sizeof(ThreadAcc);
+            gp.kdec_mask = 0;
+            uint32_t extra_lds = 0;
+            for (uint32_t i = 0; i < n_group_cols; i++) {
+                const uint32_t pj = gp.gproj[i];
+                const uint32_t w = g->colloc_host[pj].width;
+                gp.kwidth[i] = w;
+                if ((g->greedy256_mask >> pj) & 1) {
+                    gp.kdec_mask |= 1u << i;
+                    extra_lds += ((2048u * w + 255u) / 256u) * 280u;
+                }
+            }
+            uint32_t mlds = (n_waves * MGRP_SLOTS * n_aggs *
+                             (uint32_t)sizeof(ThreadAcc) + 15u) & ~15u;
+            mlds += extra_lds;
+            bool need_decode = g->n_zsegs > 0;
+            for (uint32_t pj = 0; pj < g->n_proj && !need_decode; pj++) {
+                if (!g->col_scratch[pj]) continue;
+                bool fused_key = false;
+                for (uint32_t i = 0; i < n_group_cols; i++)
+                    fused_key |= gp.gproj[i] == pj && (gp.kdec_mask & (1u << i));
+                if (!fused_key) need_decode = true;
+            }
             HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
             HIP_TRY(hipEventRecord(g->ev0, g->stream));
-            { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
+            if (need_decode) { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
             HIP_TRY(hipEventRecord(g->ev1, g->stream));
             auto launchmg = [&](auto *kern) {
                 hipLaunchKernelGGL(kern, dim3(mgrid), dim3(AGG_BLOCK), mlds, g->stream,
                                    g->d_data, g->d_scratch, g->d_groups, g->d_colloc,
+                                   g->d_segs, g->d_segstart,
                                    g->d_gkeys, g->d_gcells, g->d_error, gp);
             };
             static const int gvar = [] {
