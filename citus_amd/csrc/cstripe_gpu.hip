@@ -2985,11 +2985,19 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             uint32_t mgrid = gp.n_work < GRP_GRID ? gp.n_work : GRP_GRID;
             gp.kdec_mask = 0;
             uint32_t extra_lds = 0;
+            /* fused in-kernel key decode measured SLOWER than the separate
+             * massively-parallel decode kernel (idle lanes during the
+             * half-tile decode phase + LDS-driven occupancy loss): keep it
+             * behind the variant knob for future work */
+            static const int gvar2 = [] {
+                const char *e = getenv("CSTRIPE_GROUPED_VARIANT");
+                return e ? atoi(e) : 0;
+            }();
             for (uint32_t i = 0; i < n_group_cols; i++) {
                 const uint32_t pj = gp.gproj[i];
                 const uint32_t w = g->colloc_host[pj].width;
                 gp.kwidth[i] = w;
-                if ((g->greedy256_mask >> pj) & 1) {
+                if (gvar2 == 2 && ((g->greedy256_mask >> pj) & 1)) {
                     gp.kdec_mask |= 1u << i;
                     extra_lds += ((2048u * w + 255u) / 256u) * 280u;
                 }
