@@ -135,6 +135,9 @@ _scan_agg_grouped = _sig("cstripe_scan_agg_grouped", C.c_int,
                           C.POINTER(C.c_uint32), C.c_uint32, C.POINTER(GroupResult),
                           C.POINTER(Partial)])
 _next_batch = _sig("cstripe_scan_next_batch", C.c_int, [C.c_void_p, C.POINTER(Batch)])
+_read_row = _sig("cstripe_read_row", C.c_int,
+                 [C.c_void_p, C.c_uint64, C.POINTER(C.c_void_p),
+                  C.POINTER(C.c_uint8)])
 _rewind = _sig("cstripe_scan_rewind", C.c_int, [C.c_void_p])
 _last_fused = _sig("cstripe_scan_last_fused", C.c_int, [C.c_void_p])
 _last_kernel_ms = _sig("cstripe_scan_last_kernel_ms", C.c_double, [C.c_void_p])
@@ -392,6 +395,13 @@ class Scan:
         if not self._h:
             raise CStripeError("scan_begin: " + errmsg())
         self.reader = reader
+        self._reader = reader
+        # pred columns are implicitly projected (scan_begin does the same)
+        self._mask = cols_mask
+        for p in preds:
+            self._mask |= 1 << p[0]
+        if self._mask == 0:
+            self._mask = 1
 
     def end(self):
         if self._h:
@@ -470,6 +480,30 @@ class Scan:
             return None
         _check(rc, "next_batch")
         return b.n_rows, b.first_row_number
+
+    def read_row(self, row_number):
+        """Random access (ColumnarReadRowByRowNumber): {col: value-or-None}
+        over the projected columns, or None when the row does not exist.
+        Requires a predicate-free scan."""
+        import numpy as np
+        r = self._reader
+        ncols = r.column_count
+        DT = {1: np.int8, 2: np.int16, 3: np.int32, 4: np.int64,
+              5: np.float32, 6: np.float64, 7: np.uint32}
+        bufs, vp = {}, (C.c_void_p * ncols)()
+        nulls = (C.c_uint8 * ncols)()
+        for c in range(ncols):
+            if not (self._mask >> c) & 1:
+                continue
+            a = np.zeros(1, dtype=DT[r.column_def(c)[1]])
+            bufs[c] = a
+            vp[c] = a.ctypes.data_as(C.c_void_p).value
+        rc = _read_row(self._h, row_number, vp, nulls)
+        if rc == END:
+            return None
+        _check(rc, "read_row")
+        return {c: (None if nulls[c] else bufs[c][0].item())
+                for c in bufs}
 
     def rewind(self):
         _check(_rewind(self._h), "rewind")

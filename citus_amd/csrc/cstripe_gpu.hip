@@ -124,6 +124,7 @@ struct cs_gpu_state {
     uint32_t *d_gn = nullptr;
     uint8_t *d_tmp = nullptr;        /* next_batch: canonical-chunk decode buf */
     uint64_t tmp_bytes = 0;
+    bool batch_decoded = false;      /* scratch holds decoded streams */
     SegDesc *d_zsegs = nullptr;      /* restricted-zstd segments (zstd_r.h) */
     uint32_t n_zsegs = 0;
     zr_dtables *d_zrtab = nullptr;   /* predefined FSE decode tables */
@@ -3491,24 +3492,24 @@ collect_ungrouped:
  * columnar_reader.c:868-901, batched).
  * ===================================================================== */
 
-int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
+int csgpu_fetch_batch(cstripe_scan *s, uint32_t gi, cstripe_batch *batch)
 {
     if (!s->gpu) { cs_set_err("scan not staged — call cstripe_gpu_stage first"); return CSTRIPE_ERR_NOGPU; }
     cs_gpu_state *g = s->gpu;
     cstripe_reader *r = s->r;
-    if (s->batch_pos >= s->sel.size()) return CSTRIPE_END;
+    if (gi >= s->sel.size()) return CSTRIPE_END;
 
-    /* make sure scratch holds decoded data (decode everything once per rewind) */
-    if ((g->n_segs > 0 || g->n_zsegs > 0) && s->batch_pos == 0) {
+    /* make sure scratch holds decoded data (decode everything once) */
+    if ((g->n_segs > 0 || g->n_zsegs > 0) && !g->batch_decoded) {
         HIP_TRY(hipMemsetAsync(g->d_error, 0, sizeof(int), g->stream));
         { int _rc = launch_decode(g); if (_rc != CSTRIPE_OK) return _rc; }
         int h_err = 0;
         HIP_TRY(hipMemcpyAsync(&h_err, g->d_error, sizeof(int), hipMemcpyDeviceToHost, g->stream));
         HIP_TRY(hipStreamSynchronize(g->stream));
         if (h_err) { cs_set_err("decode error on device (flag %d)", h_err); return CSTRIPE_ERR_FORMAT; }
+        g->batch_decoded = true;
     }
 
-    uint32_t gi = (uint32_t)s->batch_pos;
     const cs_selchunk &sc = s->sel[gi];
     const cs_stripe_info &st = r->stripes[sc.stripe];
     uint32_t rows = st.group_rows[sc.chunk];
@@ -3598,6 +3599,12 @@ int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
     uint64_t first = st.meta.first_row_number;
     for (uint32_t k = 0; k < sc.chunk; k++) first += st.group_rows[k];
     batch->first_row_number = first;
-    s->batch_pos++;
     return CSTRIPE_OK;
+}
+
+int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
+{
+    int rc = csgpu_fetch_batch(s, (uint32_t)s->batch_pos, batch);
+    if (rc == CSTRIPE_OK) s->batch_pos++;
+    return rc;
 }

@@ -1038,6 +1038,75 @@ extern "C" int cstripe_scan_agg_grouped(cstripe_scan *s, const cstripe_agg_spec 
     return csgpu_agg(s, aggs, n_aggs, group_cols, n_group_cols, gr, out);
 }
 
+extern "C" int cstripe_read_row(cstripe_scan *s, uint64_t row_number,
+                                void **col_values, uint8_t *col_nulls)
+{
+    if (!s || !col_values) { cs_set_err("read_row: bad args"); return CSTRIPE_ERR_ARG; }
+    if (!s->preds.empty()) {
+        cs_set_err("read_row: random access requires a predicate-free scan "
+                   "(the reference's ColumnarReadRowByRowNumber reads with "
+                   "empty clause lists)");
+        return CSTRIPE_ERR_ARG;
+    }
+    cstripe_reader *r = s->r;
+    /* locate the containing selected chunk (scan order = global row order;
+     * with no predicates every chunk is selected) */
+    if (s->rr_gi < 0 || row_number < s->rr_first ||
+        row_number >= s->rr_first + s->rr_rows) {
+        uint64_t row_base = 0;
+        int64_t gi = -1;
+        uint64_t hit_base = 0;
+        uint64_t idx = 0;
+        for (const auto &sc : s->sel) {
+            const cs_stripe_info &st = r->stripes[sc.stripe];
+            const uint64_t rows = st.group_rows[sc.chunk];
+            if (row_number < row_base + rows) {
+                gi = (int64_t)idx;
+                hit_base = row_base;
+                break;
+            }
+            row_base += rows;
+            idx++;
+        }
+        if (gi < 0) return CSTRIPE_END;         /* no such row */
+        const uint32_t n_cols = r->head.column_count;
+        const uint32_t cap = r->head.chunk_row_limit;
+        if (s->rr_vals.empty()) {
+            s->rr_vals.resize(n_cols);
+            s->rr_nulls.resize(n_cols);
+            for (uint32_t c = 0; c < n_cols; c++) {
+                if (!(s->cols_mask & (1ull << c))) continue;
+                s->rr_vals[c].resize((size_t)cap * type_width(r->cols[c].type));
+                s->rr_nulls[c].resize(cap);
+            }
+        }
+        std::vector<void *> vp(n_cols, nullptr);
+        std::vector<uint8_t *> np(n_cols, nullptr);
+        for (uint32_t c = 0; c < n_cols; c++) {
+            if (!(s->cols_mask & (1ull << c))) continue;
+            vp[c] = s->rr_vals[c].data();
+            np[c] = s->rr_nulls[c].data();
+        }
+        cstripe_batch b{};
+        b.col_values = vp.data();
+        b.col_nulls = np.data();
+        int rc = csgpu_fetch_batch(s, (uint32_t)gi, &b);
+        if (rc != CSTRIPE_OK) return rc;
+        s->rr_gi = gi;
+        s->rr_rows = b.n_rows;
+        s->rr_first = hit_base;
+    }
+    const uint64_t off = row_number - s->rr_first;
+    for (uint32_t c = 0; c < r->head.column_count; c++) {
+        if (!col_values[c]) continue;
+        if (!(s->cols_mask & (1ull << c))) { cs_set_err("read_row: column %u not projected", c); return CSTRIPE_ERR_ARG; }
+        const uint32_t w = type_width(r->cols[c].type);
+        memcpy(col_values[c], s->rr_vals[c].data() + off * w, w);
+        if (col_nulls) col_nulls[c] = s->rr_nulls[c][off];
+    }
+    return CSTRIPE_OK;
+}
+
 extern "C" int cstripe_scan_next_batch(cstripe_scan *s, cstripe_batch *batch)
 {
     if (!s || !batch) return CSTRIPE_ERR_ARG;

@@ -68,6 +68,13 @@ struct cstripe_scan {
     double last_kernel_ms = 0.0;
     double last_decode_ms = 0.0;
     double last_agg_ms = 0.0;
+    /* random-access read cache (the reference keeps the current stripe
+     * open across ColumnarReadRowByRowNumber calls the same way) */
+    int64_t rr_gi = -1;
+    std::vector<std::vector<uint8_t>> rr_vals;   /* per col, row-aligned */
+    std::vector<std::vector<uint8_t>> rr_nulls;
+    uint32_t rr_rows = 0;
+    uint64_t rr_first = 0;
 };
 
 /* implemented in cstripe_gpu.hip */
@@ -78,5 +85,6 @@ int  csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
                const uint32_t *group_cols, uint32_t n_group_cols,
                cstripe_group_result *gr, cstripe_partial *out);
 int  csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch);
+int  csgpu_fetch_batch(cstripe_scan *s, uint32_t gi, cstripe_batch *batch);
 
 #endif

@@ -304,6 +304,20 @@ int cstripe_scan_agg_grouped(cstripe_scan *s, const cstripe_agg_spec *aggs,
                              uint32_t n_group_cols, cstripe_group_result *gr,
                              cstripe_partial *partials_out);
 
+/* Random-access read — ColumnarReadRowByRowNumber
+ * (columnar_reader.c:386-441): reads the row with the given row number
+ * (0-based scan-order position over the file/shard directory) into the
+ * caller's per-column buffers: col_values[c] points at one value slot of
+ * column c's physical width (unprojected columns may be NULL), col_nulls[c]
+ * gets 0/1. Returns CSTRIPE_OK, CSTRIPE_END when no such row exists (the
+ * reference returns false), or an error. Requires a scan begun WITHOUT
+ * predicates — the reference's random-access path passes empty clause
+ * lists the same way (:423-424). The containing chunk is device-decoded
+ * and cached across consecutive calls, like the reference keeping the
+ * current stripe open. */
+int cstripe_read_row(cstripe_scan *s, uint64_t row_number,
+                     void **col_values, uint8_t *col_nulls);
+
 /* Batch access (parity path): returns CSTRIPE_OK with a filled batch, or
  * CSTRIPE_END when exhausted. GPU-decodes chunk by chunk and copies back;
  * residual (non-pruned-chunk) predicate filtering is NOT applied — caller

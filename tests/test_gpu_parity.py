@@ -743,3 +743,33 @@ def test_text_pred_gpu(tmp_path):
     assert_parity(op, gp, aggs)
     m = flags == "N"
     assert gp[0].i128 == int(v[m].sum())
+
+
+def test_read_row_random_access(tmp_path):
+    """cstripe_read_row (ColumnarReadRowByRowNumber, columnar_reader.c:
+    386-441): point lookups across chunk/stripe boundaries, nulls, TEXT
+    slots, and the no-such-row / predicate-rejection contracts."""
+    n = 25_000
+    a = (np.arange(n, dtype=np.int64) * 13) % 100_000
+    f = ca.text_slots([["A", "N", "R"][i % 3] for i in range(n)])
+    na = (np.arange(n) % 9 == 0).astype(np.uint8)
+    path = str(tmp_path / "rr.cs")
+    ca.write_table(path, [("a", ca.I64, 0), ("f", ca.TEXT, 0)], [a, f],
+                   nulls=[na, None], compression=ca.COMP_LZ4,
+                   stripe_row_limit=8000, chunk_group_row_limit=1000)
+    with ca.Reader(path) as r, r.scan(cols_mask=0b11) as s:
+        s.stage()
+        for row in (0, 1, 999, 1000, 7999, 8000, 13333, n - 1):
+            got = s.read_row(row)
+            if na[row]:
+                assert got[0] is None
+            else:
+                assert got[0] == int(a[row]), row
+            assert got[1] == int(f[row]), row
+        assert s.read_row(n) is None            # no such row -> END/None
+        assert s.read_row(2) is not None        # cache revisit
+    with ca.Reader(path) as r, r.scan(cols_mask=0b11,
+                                      preds=[(0, ca.PRED_GT, 10)]) as s:
+        s.stage()
+        with pytest.raises(ca.CStripeError, match="predicate-free"):
+            s.read_row(5)
