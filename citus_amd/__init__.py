@@ -116,6 +116,8 @@ _write_begin = _sig("cstripe_write_begin", C.c_void_p,
 _write_rows = _sig("cstripe_write_rows", C.c_int,
                    [C.c_void_p, C.c_uint64, C.POINTER(C.c_void_p), C.POINTER(C.c_void_p)])
 _write_end = _sig("cstripe_write_end", C.c_int, [C.c_void_p])
+_write_rows_device = _sig("cstripe_write_rows_device", C.c_int,
+                          [C.c_void_p, C.c_uint64, C.POINTER(C.c_void_p)])
 _open = _sig("cstripe_open", C.c_void_p, [C.c_char_p])
 _close = _sig("cstripe_close", None, [C.c_void_p])
 _row_count = _sig("cstripe_row_count", C.c_uint64, [C.c_void_p])
@@ -254,6 +256,23 @@ def write_table(path, defs, columns, nulls=None, **opt_kw):
         for i, a in enumerate(nulls):
             nl[i] = a.ctypes.data_as(C.c_void_p).value if a is not None else None
     _check(_write_rows(w, n, vals, nl), "write_rows")
+    _check(_write_end(w), "write_end")
+
+
+def write_table_device(path, defs, dev_ptrs, n_rows, **opt_kw):
+    """Device-side write (cstripe_write_rows_device): dev_ptrs are raw HBM
+    addresses of column arrays (e.g. torch tensor.data_ptr()); full chunks
+    compress on the GPU, only compressed bytes cross PCIe."""
+    opts = default_options(**opt_kw)
+    cols = make_coldefs(defs)
+    w = _write_begin(path.encode(), cols, len(defs), C.byref(opts))
+    if not w:
+        raise CStripeError("write_begin: " + errmsg())
+    vals = (C.c_void_p * len(defs))(*[int(p) for p in dev_ptrs])
+    rc = _write_rows_device(w, n_rows, vals)
+    if rc != OK:
+        _write_end(w)
+        raise CStripeError(f"write_rows_device failed (rc={rc}): {errmsg()}")
     _check(_write_end(w), "write_end")
 
 

@@ -25,6 +25,7 @@
 #include "compress.h"
 #include "pglz.h"
 #include "zstd_r.h"
+#include "lz4_enc.h"
 
 #include <hip/hip_runtime.h>
 #include <cstring>
@@ -3607,4 +3608,293 @@ int csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch)
     int rc = csgpu_fetch_batch(s, (uint32_t)s->batch_pos, batch);
     if (rc == CSTRIPE_OK) s->batch_pos++;
     return rc;
+}
+
+/* =====================================================================
+ * Device write path: compress one chunk's columns ON the
+ * GPU. Stats kernel reduces min/max and the OR-of-XOR signature that
+ * decides the canonical mode; the emit kernel then writes the canonical
+ * P(L)/CONST stream with one lane per value (every byte position is
+ * closed-form — the same property the read path exploits). Columns that
+ * do not fit a canonical parse are copied back raw and compressed by the
+ * host writer path (greedy LZ4 / zstd / pglz), so any data works.
+ * Replaces, for HBM-resident data, the host serializer of
+ * SerializeChunkData/CompressBuffer (columnar_writer.c:592-654,
+ * columnar_compression.c:62-158); only compressed bytes cross PCIe.
+ * ===================================================================== */
+
+struct WStats {
+    int64_t mn, mx;            /* int: values; float: f64 bit patterns */
+    uint64_t xsig;             /* OR of (v ^ v0) over the chunk (width 8) */
+};
+
+template <typename T>
+__global__ __launch_bounds__(AGG_BLOCK) void wstats_kernel(
+    const T *__restrict__ v, uint32_t n, int is_float, WStats *__restrict__ out)
+{
+    const uint32_t tid = blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t xs = 0;
+    uint64_t v0 = 0;
+    if (n) __builtin_memcpy(&v0, &v[0], sizeof(T) < 8 ? sizeof(T) : 8);
+    double fmn = 0, fmx = 0;
+    bool seen = false;
+    int64_t imn = INT64_MAX, imx = INT64_MIN;
+    for (uint32_t i = tid; i < n; i += gridDim.x * blockDim.x) {
+        if (is_float) {
+            double d;
+            if (sizeof(T) == 4) { float f; __builtin_memcpy(&f, &v[i], 4); d = f; }
+            else __builtin_memcpy(&d, &v[i], 8);
+            if (!seen) { fmn = fmx = d; seen = true; }
+            else {
+                if (f64cmp_pg(d, fmn) < 0) fmn = d;
+                if (f64cmp_pg(d, fmx) > 0) fmx = d;
+            }
+        } else {
+            const int64_t x = (int64_t)v[i];
+            imn = min(imn, x);
+            imx = max(imx, x);
+        }
+        if (sizeof(T) == 8) {               /* canonical signature: BIT xor */
+            uint64_t bits;
+            __builtin_memcpy(&bits, &v[i], 8);
+            xs |= bits ^ v0;
+        }
+    }
+    /* wave + LDS + global-atomic reduce */
+    __shared__ int64_t smn[AGG_BLOCK / WAVE], smx[AGG_BLOCK / WAVE];
+    __shared__ uint64_t sxs[AGG_BLOCK / WAVE];
+    __shared__ int sseen[AGG_BLOCK / WAVE];
+    for (int d = WAVE / 2; d > 0; d >>= 1) {
+        const int64_t omn = __shfl_down((long long)(is_float ? (int64_t)__double_as_longlong(fmn) : imn), d, WAVE);
+        const int64_t omx = __shfl_down((long long)(is_float ? (int64_t)__double_as_longlong(fmx) : imx), d, WAVE);
+        const uint64_t oxs = (uint64_t)__shfl_down((long long)xs, d, WAVE);
+        const int osn = __shfl_down((int)seen, d, WAVE);
+        if (is_float) {
+            double a, b;
+            a = __longlong_as_double((long long)omn);
+            b = __longlong_as_double((long long)omx);
+            if (osn) {
+                if (!seen) { fmn = a; fmx = b; seen = true; }
+                else {
+                    if (f64cmp_pg(a, fmn) < 0) fmn = a;
+                    if (f64cmp_pg(b, fmx) > 0) fmx = b;
+                }
+            }
+        } else {
+            imn = min(imn, omn);
+            imx = max(imx, omx);
+        }
+        xs |= oxs;
+    }
+    const uint32_t wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+    if (lane == 0) {
+        smn[wid] = is_float ? (int64_t)__double_as_longlong(fmn) : imn;
+        smx[wid] = is_float ? (int64_t)__double_as_longlong(fmx) : imx;
+        sxs[wid] = xs;
+        sseen[wid] = seen;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int64_t bmn = smn[0], bmx = smx[0];
+        uint64_t bxs = sxs[0];
+        int bseen = sseen[0];
+        for (uint32_t w2 = 1; w2 < AGG_BLOCK / WAVE; w2++) {
+            if (is_float) {
+                if (sseen[w2]) {
+                    double a = __longlong_as_double((long long)smn[w2]);
+                    double b = __longlong_as_double((long long)smx[w2]);
+                    if (!bseen) { bmn = smn[w2]; bmx = smx[w2]; bseen = 1; }
+                    else {
+                        if (f64cmp_pg(a, __longlong_as_double((long long)bmn)) < 0) bmn = smn[w2];
+                        if (f64cmp_pg(b, __longlong_as_double((long long)bmx)) > 0) bmx = smx[w2];
+                    }
+                }
+            } else {
+                bmn = min(bmn, smn[w2]);
+                bmx = max(bmx, smx[w2]);
+            }
+            bxs |= sxs[w2];
+        }
+        /* per-block results merged with CAS loops on the single out cell */
+        if (is_float) {
+            unsigned long long *amn = (unsigned long long *)&out->mn;
+            unsigned long long cur = *amn;
+            while (bseen) {
+                double c = __longlong_as_double((long long)cur);
+                double nv = __longlong_as_double((long long)bmn);
+                if (f64cmp_pg(nv, c) >= 0) break;
+                unsigned long long prev = atomicCAS(amn, cur, (unsigned long long)bmn);
+                if (prev == cur) break;
+                cur = prev;
+            }
+            unsigned long long *amx = (unsigned long long *)&out->mx;
+            cur = *amx;
+            while (bseen) {
+                double c = __longlong_as_double((long long)cur);
+                double nv = __longlong_as_double((long long)bmx);
+                if (f64cmp_pg(nv, c) <= 0) break;
+                unsigned long long prev = atomicCAS(amx, cur, (unsigned long long)bmx);
+                if (prev == cur) break;
+                cur = prev;
+            }
+        } else {
+            atomicMin((long long *)&out->mn, (long long)bmn);
+            atomicMax((long long *)&out->mx, (long long)bmx);
+        }
+        atomicOr((unsigned long long *)&out->xsig, (unsigned long long)bxs);
+    }
+}
+
+/* emit the canonical P(L) stream for one width-8 column chunk: lane j
+ * writes its value's literal bytes plus the token/offset of the sequence
+ * it opens — every byte position is the closed form of format.h */
+__global__ __launch_bounds__(AGG_BLOCK) void wemit_p_kernel(
+    const uint64_t *__restrict__ v, uint32_t n, uint32_t L,
+    uint8_t *__restrict__ out)
+{
+    const uint32_t j = blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= n) return;
+    const uint32_t step = L + 3u;
+    const uint8_t mtok = (uint8_t)(8u - L - 4u);
+    const uint64_t val = v[j];
+    if (j == 0) {
+        out[0] = (uint8_t)(((8u + L) << 4) | mtok);
+        #pragma unroll
+        for (int b = 0; b < 8; b++) out[1 + b] = (uint8_t)(val >> (8 * b));
+        /* S0's offset closes after v1's low bytes (lane 1 writes those) */
+        out[9 + L] = 8;
+        out[10 + L] = 0;
+        return;
+    }
+    if (j == n - 1) {                       /* final: token + full literals */
+        const uint32_t fpos = (11u + L) + (n - 3u) * step;
+        out[fpos] = (uint8_t)(8u << 4);
+        #pragma unroll
+        for (int b = 0; b < 8; b++) out[fpos + 1 + b] = (uint8_t)(val >> (8 * b));
+        return;
+    }
+    /* v_j low bytes: j==1 lives in S0's literals at 9; j>=2 at pos(j) with
+     * its sequence header [token at pos-1] and trailing offset */
+    if (j == 1) {
+        for (uint32_t b = 0; b < L; b++) out[9 + b] = (uint8_t)(val >> (8 * b));
+        return;
+    }
+    const uint32_t pos = j * step + (6u - L);
+    out[pos - 1] = (uint8_t)((L << 4) | mtok);
+    for (uint32_t b = 0; b < L; b++) out[pos + b] = (uint8_t)(val >> (8 * b));
+    out[pos + L] = 8;                       /* every S_i carries its offset */
+    out[pos + L + 1] = 0;
+}
+
+int csgpu_compress_chunk(const void *const *dev_vals, const uint8_t *types,
+                         uint32_t n_cols, uint32_t rows, uint64_t row_offset,
+                         std::vector<cs_dev_chunk_col> &out)
+{
+    if (!cstripe_gpu_available()) { cs_set_err("device write path requires a visible MI355X"); return CSTRIPE_ERR_NOGPU; }
+    out.assign(n_cols, {});
+    hipStream_t st;
+    HIP_TRY(hipStreamCreate(&st));
+    WStats *d_stats = nullptr;
+    uint8_t *d_out = nullptr;
+    size_t d_out_cap = 0;
+    auto cleanup = [&]() {
+        if (d_stats) (void)hipFree(d_stats);
+        if (d_out) (void)hipFree(d_out);
+        (void)hipStreamDestroy(st);
+    };
+    #define WTRY(x) do { hipError_t _e = (x); if (_e != hipSuccess) { \
+        cs_set_err("%s failed: %s", #x, hipGetErrorString(_e)); cleanup(); return CSTRIPE_ERR; } } while (0)
+    WTRY(hipMalloc(&d_stats, sizeof(WStats)));
+
+    for (uint32_t c = 0; c < n_cols; c++) {
+        const uint32_t width = csf_type_width(types[c]);
+        const bool is_float = types[c] == CSTRIPE_F32 || types[c] == CSTRIPE_F64;
+        const uint8_t *dv = (const uint8_t *)dev_vals[c] + row_offset * width;
+        cs_dev_chunk_col &oc = out[c];
+        oc.has_min_max = rows > 0;
+
+        WStats init;
+        if (is_float) {
+            const double dnan = 0.0 / 0.0, dninf = -1.0 / 0.0;
+            memcpy(&init.mn, &dnan, 8);     /* PG MIN identity: NaN high */
+            memcpy(&init.mx, &dninf, 8);
+        } else {
+            init.mn = INT64_MAX;
+            init.mx = INT64_MIN;
+        }
+        init.xsig = 0;
+        WTRY(hipMemcpyAsync(d_stats, &init, sizeof(init), hipMemcpyHostToDevice, st));
+        const uint32_t grid = rows ? min((rows + AGG_BLOCK - 1) / AGG_BLOCK, 1024u) : 1;
+        if (rows) {
+            switch (width) {
+                case 1: hipLaunchKernelGGL(wstats_kernel<int8_t>, dim3(grid), dim3(AGG_BLOCK), 0, st, (const int8_t *)dv, rows, is_float, d_stats); break;
+                case 2: hipLaunchKernelGGL(wstats_kernel<int16_t>, dim3(grid), dim3(AGG_BLOCK), 0, st, (const int16_t *)dv, rows, is_float, d_stats); break;
+                case 4: hipLaunchKernelGGL(wstats_kernel<int32_t>, dim3(grid), dim3(AGG_BLOCK), 0, st, (const int32_t *)dv, rows, is_float, d_stats); break;
+                default: hipLaunchKernelGGL(wstats_kernel<int64_t>, dim3(grid), dim3(AGG_BLOCK), 0, st, (const int64_t *)dv, rows, is_float, d_stats); break;
+            }
+            WTRY(hipGetLastError());
+        }
+        WStats hs;
+        WTRY(hipMemcpyAsync(&hs, d_stats, sizeof(hs), hipMemcpyDeviceToHost, st));
+        WTRY(hipStreamSynchronize(st));
+        oc.min_i = hs.mn;
+        oc.max_i = hs.mx;
+        /* TEXT columns carry no skip-node min/max (host path does the same) */
+        if (types[c] == CSTRIPE_TEXT) oc.has_min_max = false;
+
+        const uint64_t raw_bytes = (uint64_t)rows * width;
+        bool emitted = false;
+        if (width == 8 && rows >= 3 && hs.xsig != 0) {
+            int hb = 63;
+            while (hb > 0 && !((hs.xsig >> hb) & 1)) hb--;
+            const uint32_t L = (uint32_t)(hb / 8 + 1);
+            if (L <= 4) {
+                const uint64_t csz = (11ull + L) + (uint64_t)(rows - 3) * (L + 3) + 9;
+                if (csz < raw_bytes) {
+                    if (d_out_cap < csz) {
+                        if (d_out) WTRY(hipFree(d_out));
+                        d_out_cap = (csz + 4095) & ~4095ull;
+                        WTRY(hipMalloc(&d_out, d_out_cap));
+                    }
+                    hipLaunchKernelGGL(wemit_p_kernel,
+                                       dim3((rows + AGG_BLOCK - 1) / AGG_BLOCK),
+                                       dim3(AGG_BLOCK), 0, st,
+                                       (const uint64_t *)dv, rows, L, d_out);
+                    WTRY(hipGetLastError());
+                    oc.data.resize(csz);
+                    WTRY(hipMemcpyAsync(oc.data.data(), d_out, csz, hipMemcpyDeviceToHost, st));
+                    WTRY(hipStreamSynchronize(st));
+                    oc.mode = (uint8_t)(CSF_SEGMODE_P_BASE | L);
+                    oc.canonical = true;
+                    emitted = true;
+                }
+            }
+        } else if (width == 8 && rows >= 3 && hs.xsig == 0) {
+            /* constant chunk: emit on host from v0 (8 bytes over PCIe) */
+            uint64_t v0;
+            WTRY(hipMemcpyAsync(&v0, dv, 8, hipMemcpyDeviceToHost, st));
+            WTRY(hipStreamSynchronize(st));
+            oc.data.resize(64 + rows / 16);
+            int csz = lz4e_canon_const((const uint8_t *)&v0, (int)rows,
+                                       oc.data.data(), (int)oc.data.size());
+            if (csz > 0 && (uint64_t)csz < raw_bytes) {
+                oc.data.resize((size_t)csz);
+                oc.mode = CSF_SEGMODE_CONST;
+                oc.canonical = true;
+                emitted = true;
+            }
+        }
+        if (!emitted) {
+            /* raw copy-back: the host writer path compresses it */
+            oc.data.resize(raw_bytes);
+            if (raw_bytes) {
+                WTRY(hipMemcpyAsync(oc.data.data(), dv, raw_bytes, hipMemcpyDeviceToHost, st));
+                WTRY(hipStreamSynchronize(st));
+            }
+            oc.canonical = false;
+        }
+    }
+    #undef WTRY
+    cleanup();
+    return CSTRIPE_OK;
 }

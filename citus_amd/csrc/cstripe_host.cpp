@@ -23,6 +23,8 @@
 #include "pglz.h"
 #include "zstd_r.h"
 
+#include <hip/hip_runtime.h>
+
 #include <cstdarg>
 #include <cstdio>
 #include <cstring>
@@ -110,6 +112,10 @@ struct PendingChunk {                     /* raw chunk awaiting compression at f
     std::vector<uint8_t> raw_values;
     std::vector<uint8_t> exists_packed;
     csf_skipnode node;                    /* min/max/row_count/n_present/decomp filled */
+    /* device write path: stream already compressed on the GPU */
+    bool has_ready = false;
+    std::vector<uint8_t> ready_comp;
+    uint8_t ready_mode = 0;
 };
 
 } /* namespace */
@@ -294,6 +300,18 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
 {
     out.node = pc.node;
     out.exists_packed = pc.exists_packed;
+    if (pc.has_ready) {                   /* GPU-compressed canonical stream */
+        out.value_comp = pc.ready_comp;
+        csf_seg s{0, (uint32_t)pc.ready_comp.size(), 0,
+                  (uint32_t)pc.node.decompressed_size |
+                  ((uint32_t)pc.ready_mode << 24)};
+        out.segs.push_back(s);
+        out.node.comp_type = CSTRIPE_COMP_LZ4;
+        out.node.comp_level = 0;
+        out.node.n_segs = 1;
+        out.node.value_len = pc.ready_comp.size();
+        return true;
+    }
     const std::vector<uint8_t> &raw = pc.raw_values;
     uint8_t codec = w->opts.compression;
     size_t target = w->opts.lz4_seg_target_bytes ? w->opts.lz4_seg_target_bytes
@@ -604,6 +622,92 @@ static int flush_stripe(cstripe_writer *w)
     w->chunk_rows.clear();
     w->stripe_first_row += w->stripe_rows;
     w->stripe_rows = 0;
+    return CSTRIPE_OK;
+}
+
+/* device-side append: full chunks compress on the GPU (csgpu_compress_chunk,
+ * cstripe_gpu.hip); partial spans round-trip through the host accumulation
+ * path to keep chunk/stripe boundary semantics identical */
+extern "C" int cstripe_write_rows_device(cstripe_writer *w, uint64_t n_rows,
+                                         const void *const *dev_values)
+{
+    if (!w || !dev_values) { cs_set_err("write_rows_device: bad args"); return CSTRIPE_ERR_ARG; }
+    const uint32_t n_cols = (uint32_t)w->cols.size();
+    const uint32_t chunk_limit = w->opts.chunk_group_row_limit;
+    const uint64_t stripe_limit = w->opts.stripe_row_limit;
+    std::vector<uint8_t> types(n_cols);
+    for (uint32_t c = 0; c < n_cols; c++) types[c] = w->cols[c].type;
+
+    uint64_t done = 0;
+    while (done < n_rows) {
+        const uint32_t in_chunk = (uint32_t)(w->stripe_rows % chunk_limit);
+        const uint64_t to_chunk = chunk_limit - in_chunk;
+        const uint64_t to_stripe = stripe_limit - w->stripe_rows;
+        const uint64_t span = std::min({n_rows - done, to_chunk, to_stripe});
+
+        const bool full_chunk = in_chunk == 0 && span == chunk_limit &&
+                                w->opts.compression == CSTRIPE_COMP_LZ4 &&
+                                w->opts.canonical;
+        if (full_chunk) {
+            std::vector<const void *> vp(n_cols);
+            for (uint32_t c = 0; c < n_cols; c++) vp[c] = dev_values[c];
+            std::vector<cs_dev_chunk_col> cc;
+            int rc = csgpu_compress_chunk(vp.data(), types.data(), n_cols,
+                                          (uint32_t)span, done, cc);
+            if (rc != CSTRIPE_OK) return rc;
+            const uint32_t chunk_index = (uint32_t)w->chunk_rows.size();
+            std::vector<uint8_t> ones((span + 7) / 8, 0xFF);
+            if (span % 8) ones.back() = (uint8_t)(0xFF >> (8 - span % 8));
+            for (uint32_t c = 0; c < n_cols; c++) {
+                PendingChunk pc;
+                pc.col = c;
+                pc.chunk = chunk_index;
+                memset(&pc.node, 0, sizeof(pc.node));
+                pc.node.min_i = cc[c].min_i;
+                pc.node.max_i = cc[c].max_i;
+                pc.node.has_min_max = cc[c].has_min_max ? 1 : 0;
+                pc.node.row_count = span;
+                pc.node.n_present = (uint32_t)span;
+                pc.node.decompressed_size = span * type_width(types[c]);
+                pc.exists_packed = ones;
+                if (cc[c].canonical) {
+                    pc.has_ready = true;
+                    pc.ready_comp = std::move(cc[c].data);
+                    pc.ready_mode = cc[c].mode;
+                } else {
+                    pc.raw_values = std::move(cc[c].data);
+                }
+                w->pending.push_back(std::move(pc));
+            }
+            w->chunk_rows.push_back((uint32_t)span);
+            w->stripe_rows += (uint32_t)span;
+            w->total_rows += span;
+        } else {
+            /* partial span / non-canonical codec: host accumulation path */
+            std::vector<std::vector<uint8_t>> host(n_cols);
+            std::vector<const void *> hp(n_cols);
+            for (uint32_t c = 0; c < n_cols; c++) {
+                const uint32_t width = type_width(types[c]);
+                host[c].resize(span * width);
+                if (hipMemcpy(host[c].data(),
+                              (const uint8_t *)dev_values[c] + done * width,
+                              span * width, hipMemcpyDeviceToHost) != hipSuccess) {
+                    cs_set_err("write_rows_device: D2H copy failed");
+                    return CSTRIPE_ERR;
+                }
+                hp[c] = host[c].data();
+            }
+            int rc = cstripe_write_rows(w, span, hp.data(), nullptr);
+            if (rc != CSTRIPE_OK) return rc;
+            done += span;
+            continue;
+        }
+        done += span;
+        if (w->stripe_rows >= stripe_limit) {
+            int rc = flush_stripe(w);
+            if (rc != CSTRIPE_OK) return rc;
+        }
+    }
     return CSTRIPE_OK;
 }
 

@@ -87,4 +87,17 @@ int  csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
 int  csgpu_next_batch(cstripe_scan *s, cstripe_batch *batch);
 int  csgpu_fetch_batch(cstripe_scan *s, uint32_t gi, cstripe_batch *batch);
 
+/* device write path: one chunk's columns compressed on the GPU (canonical
+ * parses where the data fits, raw copy-back otherwise) */
+struct cs_dev_chunk_col {
+    std::vector<uint8_t> data;   /* canonical LZ4 stream, or raw bytes */
+    uint8_t mode = 0;            /* CSF_SEGMODE_* when canonical, else 0 */
+    int64_t min_i = 0, max_i = 0;
+    bool canonical = false;
+    bool has_min_max = false;
+};
+int csgpu_compress_chunk(const void *const *dev_vals, const uint8_t *types,
+                         uint32_t n_cols, uint32_t rows, uint64_t row_offset,
+                         std::vector<cs_dev_chunk_col> &out);
+
 #endif

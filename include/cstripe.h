@@ -263,6 +263,14 @@ cstripe_writer *cstripe_write_begin(const char *path, const cstripe_coldef *cols
 int cstripe_write_rows(cstripe_writer *w, uint64_t n_rows,
                        const void *const *values, const uint8_t *const *nulls);
 int cstripe_write_end(cstripe_writer *w);     /* flush + footer + close; frees w */
+/* device-side append: values[c] are DEVICE pointers to
+ * HBM-resident column arrays (query results / ETL). Full chunks are
+ * compressed ON the GPU (canonical parses where the data fits; see
+ * format.h) so only compressed bytes cross PCIe; other shapes copy back
+ * and take the host compressors. Dense rows only — NULL-bearing batches
+ * go through cstripe_write_rows. Requires a visible MI355X. */
+int cstripe_write_rows_device(cstripe_writer *w, uint64_t n_rows,
+                              const void *const *dev_values);
 void cstripe_write_abort(cstripe_writer *w);
 
 /* =================== reader / scan =================== */
