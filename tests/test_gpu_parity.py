@@ -12,6 +12,7 @@ import numpy as np
 import pytest
 
 import citus_amd as ca
+import futil
 import oracle
 
 from conftest import q6_preds
