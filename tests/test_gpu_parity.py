@@ -14,6 +14,7 @@ import pytest
 import citus_amd as ca
 import futil
 import oracle
+from test_format import read_all
 
 from conftest import q6_preds
 
@@ -826,8 +827,9 @@ def test_device_write_matches_host_writer(tmp_path):
     const = torch.full((n,), 42, dtype=torch.int64)
     pd = str(tmp_path / "d.cs")
     ph = str(tmp_path / "h.cs")
+    dvals, dconst = vals.cuda(), const.cuda()   # keep refs: data_ptr lifetime
     ca.write_table_device(pd, [("v", ca.I64, 0), ("c", ca.I64, 0)],
-                          [vals.cuda().data_ptr(), const.cuda().data_ptr()], n)
+                          [dvals.data_ptr(), dconst.data_ptr()], n)
     torch.cuda.synchronize()
     ca.write_table(ph, [("v", ca.I64, 0), ("c", ca.I64, 0)],
                    [vals.numpy(), const.numpy()])
