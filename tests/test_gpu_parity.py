@@ -844,3 +844,33 @@ def test_device_write_matches_host_writer(tmp_path):
             assert parts[0].i128 == int(vals.sum())
             assert parts[1].i128 == int(vals.min())
             assert parts[2].i128 == 42
+
+
+def test_device_write_edge_shapes(tmp_path):
+    """device write: spans that do not align to chunks (host round trip),
+    tiny tables, chunk/stripe boundary crossings, canonical=0."""
+    import torch
+    for n in (1, 2, 7, 9_999, 10_001, 150_001):
+        t = ((torch.arange(n, dtype=torch.int64) * 13) % 300).cuda()
+        path = str(tmp_path / f"e{n}.cs")
+        ca.write_table_device(path, [("v", ca.I64, 0)], [t.data_ptr()], n)
+        torch.cuda.synchronize()
+        with oracle.OracleTable(path) as ot:
+            assert ot.row_count == n
+            parts, _ = ot.scan_agg([], [(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_STAR, -1)])
+            assert parts[0].i128 == int(t.sum().item())
+            assert parts[1].count == n
+        del t
+    # canonical off: everything goes through the host greedy path
+    n = 25_000
+    t = ((torch.arange(n, dtype=torch.int64) * 7) % 50).cuda()
+    path = str(tmp_path / "nc.cs")
+    ca.write_table_device(path, [("v", ca.I64, 0)], [t.data_ptr()], n,
+                          canonical=0)
+    torch.cuda.synchronize()
+    foot = futil.read_footer(path)
+    assert all(sg["mode"] == futil.SEGMODE_GENERIC
+               for sg in foot["stripes"][0]["nodes"][0][0]["segs"])
+    with oracle.OracleTable(path) as ot:
+        parts, _ = ot.scan_agg([], [(ca.AGG_SUM_I64, 0)])
+        assert parts[0].i128 == int(t.sum().item())
