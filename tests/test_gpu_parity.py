@@ -535,6 +535,9 @@ def test_fuzz_grouped_differential():
             preds = []
             if rng.random() < 0.6:
                 preds.append((nk, int(rng.integers(0, 4)), int(rng.integers(-10**5, 10**5))))
+            if rng.random() < 0.5:        # OR group over the measure column
+                preds.append((nk, ca.PRED_LT, int(rng.integers(-10**5, 0)), 7))
+                preds.append((nk, ca.PRED_GT, int(rng.integers(0, 10**5)), 7))
             aggs = [(ca.AGG_COUNT_STAR, -1), (ca.AGG_SUM_I64, nk),
                     (ca.AGG_MIN_I64, nk), (ca.AGG_MAX_I64, nk)]
             gcols = tuple(range(nk))
