@@ -19,13 +19,16 @@ __global__ void read_kernel(const ulonglong2 *__restrict__ p, size_t n,
 
 int main()
 {
-    const size_t bytes = 24ull << 30;              // 24 GiB
+    const size_t bytes = 8ull << 30;               // 8 GiB
     const size_t n = bytes / 16;
-    ulonglong2 *d;
-    unsigned long long *sink;
-    hipMalloc(&d, bytes);
-    hipMalloc(&sink, 8);
-    hipMemset(d, 0x5A, bytes);
+    ulonglong2 *d = nullptr;
+    unsigned long long *sink = nullptr;
+    if (hipMalloc(&d, bytes) != hipSuccess ||
+        hipMalloc(&sink, 8) != hipSuccess ||
+        hipMemset(d, 0x5A, bytes) != hipSuccess) {
+        fprintf(stderr, "membw: allocation failed\n");
+        return 1;
+    }
     hipEvent_t e0, e1;
     hipEventCreate(&e0);
     hipEventCreate(&e1);
