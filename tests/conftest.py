@@ -33,3 +33,13 @@ def q6_preds(ca, exp):
     q6 = exp["q6"]
     return [(5, ca.PRED_GE, q6["shipdate_ge"]), (5, ca.PRED_LT, q6["shipdate_lt"]),
             (3, ca.PRED_GE, 5), (3, ca.PRED_LE, 7), (1, ca.PRED_LT, 2400)]
+
+# torch and the system ROCm runtime disagree when libcstripe (ctypes ->
+# system libamdhip64) initializes HIP first in a process: torch.cuda then
+# enumerates 0 devices (torch bundles its own HIP runtime). Initializing
+# torch's view FIRST makes both stacks coexist; harmless on CPU-only boxes.
+try:
+    import torch
+    torch.cuda.is_available()
+except Exception:
+    pass
