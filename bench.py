@@ -104,6 +104,9 @@ def main():
     dist = world > 1
 
     import torch
+    # initialize torch's device view BEFORE libcstripe touches HIP (the two
+    # ROCm runtime instances coexist only in that order; tests/conftest.py)
+    torch.cuda.is_available()
     rccl_comm = None
     if dist:
         import torch.distributed as td
