@@ -3029,11 +3029,11 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
                 const char *e = getenv("CSTRIPE_GROUPED_VARIANT");
                 return e ? atoi(e) : 0;
             }();
-            if (gvar == 1) {       /* forced 6 waves/SIMD A/B variant */
-                if (n_aggs == 5) launchmg(multi_grouped_kernel<5, 8, 6>);
-                else if (n_aggs == 1) launchmg(multi_grouped_kernel<1, 8, 6>);
-                else if (n_aggs == 2) launchmg(multi_grouped_kernel<2, 8, 6>);
-                else if (n_aggs == 4) launchmg(multi_grouped_kernel<4, 8, 6>);
+            if (gvar == 1) {       /* forced 5 waves/SIMD A/B variant */
+                if (n_aggs == 5) launchmg(multi_grouped_kernel<5, 8, 5>);
+                else if (n_aggs == 1) launchmg(multi_grouped_kernel<1, 8, 5>);
+                else if (n_aggs == 2) launchmg(multi_grouped_kernel<2, 8, 5>);
+                else if (n_aggs == 4) launchmg(multi_grouped_kernel<4, 8, 5>);
                 else launchmg(multi_grouped_kernel<-1, 4>);
             } else {
                 if (n_aggs == 5) launchmg(multi_grouped_kernel<5, 8>);
