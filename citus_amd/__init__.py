@@ -20,6 +20,22 @@ if not os.path.exists(_LIB_PATH):
         "libcstripe.so not built — run `make -C citus_amd/csrc` "
         "(or __graft_entry__.build())")
 
+# Two HIP runtime instances in one process (torch's bundled copy + the
+# system /opt/rocm one) break device enumeration for whichever initializes
+# second — in either direction on some boxes. Both ship the same soname
+# (libamdhip64.so.7), so preloading torch's copy globally makes
+# libcstripe's NEEDED entry resolve to the SAME mapped runtime: one HIP
+# instance shared by both stacks. Plain-C ABI consumers (no torch in the
+# process) are unaffected and keep the system runtime.
+try:
+    import torch as _torch
+    _torch_hip = os.path.join(os.path.dirname(_torch.__file__), "lib",
+                              "libamdhip64.so")
+    if os.path.exists(_torch_hip):
+        C.CDLL(_torch_hip, mode=C.RTLD_GLOBAL)
+except Exception:
+    pass
+
 _lib = C.CDLL(_LIB_PATH)
 
 # ---- enums (include/cstripe.h) ----
