@@ -1,4 +1,6 @@
-import numpy as np, sys, time
+"""Perf probe: gen N-row lineitem, stage, run Q6 scan_agg 3x and print
+per-pass kernel ms (used for on-hardware A/B iterations this round)."""
+import sys, time
 sys.path.insert(0, "/root/repo")
 import citus_amd as ca
 
