@@ -476,10 +476,15 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
                 int op2 = 0;
                 d[op2++] = 0x28; d[op2++] = 0xB5; d[op2++] = 0x2F; d[op2++] = 0xFD;
                 if (len <= 255) { d[op2++] = 0x20; d[op2++] = (uint8_t)len; }
-                else {
+                else if (len <= 65535 + 256) {
                     d[op2++] = 0x60;
                     uint32_t f2 = (uint32_t)len - 256;
                     d[op2++] = (uint8_t)f2; d[op2++] = (uint8_t)(f2 >> 8);
+                } else {
+                    d[op2++] = 0xA0;
+                    uint32_t f2 = (uint32_t)len;
+                    d[op2++] = (uint8_t)f2; d[op2++] = (uint8_t)(f2 >> 8);
+                    d[op2++] = (uint8_t)(f2 >> 16); d[op2++] = (uint8_t)(f2 >> 24);
                 }
                 uint32_t bh = 1u | (0u << 1) | ((uint32_t)len << 3);
                 d[op2++] = (uint8_t)bh; d[op2++] = (uint8_t)(bh >> 8); d[op2++] = (uint8_t)(bh >> 16);

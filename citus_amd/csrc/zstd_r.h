@@ -323,16 +323,24 @@ static inline int zr_compress(const uint8_t *src, int slen,
     if (slen <= 0 || slen > (1 << 20)) return 0;
     int op = 0;
     /* frame header: magic + FHD (single segment) + content size */
-    if (op + 7 > dcap) return 0;
+    if (op + 9 > dcap) return 0;
     dst[op++] = 0x28; dst[op++] = 0xB5; dst[op++] = 0x2F; dst[op++] = 0xFD;
     if (slen <= 255) {
         dst[op++] = 0x20;                 /* FHD: single_segment, FCS 1 byte */
         dst[op++] = (uint8_t)slen;
-    } else {
-        dst[op++] = 0x60;                 /* FHD: single_segment, FCS 2 bytes */
+    } else if (slen <= 65535 + 256) {
+        dst[op++] = 0x60;                 /* FHD: single_segment, FCS 2 bytes
+                                           * (stores size - 256; caps 65791) */
         const uint32_t f = (uint32_t)slen - 256;
         dst[op++] = (uint8_t)f;
         dst[op++] = (uint8_t)(f >> 8);
+    } else {
+        dst[op++] = 0xA0;                 /* FHD: single_segment, FCS 4 bytes */
+        const uint32_t f = (uint32_t)slen;
+        dst[op++] = (uint8_t)f;
+        dst[op++] = (uint8_t)(f >> 8);
+        dst[op++] = (uint8_t)(f >> 16);
+        dst[op++] = (uint8_t)(f >> 24);
     }
     const int bh_pos = op;                /* 3-byte block header backpatched */
     op += 3;
@@ -541,10 +549,20 @@ ZR_HOSTDEV static inline int zr_decode_frame(const uint8_t *src, int slen,
         return -1;
     ip = 4;
     const uint8_t fhd = src[ip++];
-    if (!(fhd == 0x20 || fhd == 0x60)) return -1;    /* restricted frames only */
+    if (!(fhd == 0x20 || fhd == 0x60 || fhd == 0xA0)) return -1;  /* restricted */
     int content;
-    if (fhd == 0x20) content = src[ip++];
-    else { content = 256 + src[ip] + (src[ip + 1] << 8); ip += 2; }
+    if (fhd == 0x20) {
+        content = src[ip++];
+    } else if (fhd == 0x60) {
+        content = 256 + src[ip] + (src[ip + 1] << 8);
+        ip += 2;
+    } else {
+        uint32_t f = 0;
+        for (int i = 0; i < 4; i++) f |= (uint32_t)src[ip + i] << (8 * i);
+        if (f > (1u << 23)) return -1;
+        content = (int)f;
+        ip += 4;
+    }
     if (content > dcap) return -1;
     /* block header */
     const uint32_t bh = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8) |

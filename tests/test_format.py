@@ -563,3 +563,23 @@ def test_text_bad_slot_rejected(tmp_path):
     bad = np.array([0x00000004], dtype=np.uint32)   # even header: not varlena
     with pytest.raises(ca.CStripeError, match="short varlena"):
         ca.write_table(str(tmp_path / "b.cs"), [("f", ca.TEXT, 0)], [bad])
+
+
+def test_zstd_large_segment_frames(tmp_path):
+    """restricted-zstd frames above the 2-byte frame-content-size limit
+    (65 791 B) must use the 4-byte FCS form — large zstd segment targets
+    round-trip through libzstd exactly."""
+    n = 40_000                       # 320 KB chunk at 256 KB segment target
+    a = (np.arange(n, dtype=np.int64) * 17) % 911
+    path = str(tmp_path / "bigseg.cs")
+    ca.write_table(path, [("a", ca.I64, 0)], [a], compression=ca.COMP_ZSTD,
+                   lz4_seg_target_kb=256, chunk_group_row_limit=40_000,
+                   stripe_row_limit=40_000)
+    foot = futil.read_footer(path)
+    node = foot["stripes"][0]["nodes"][0][0]
+    assert node["comp_type"] == ca.COMP_ZSTD
+    assert any(s2["decomp_len"] > 65_791 for s2 in node["segs"])
+    with oracle.OracleTable(path) as t:     # oracle decodes via libzstd
+        parts, _ = t.scan_agg([], [(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_STAR, -1)])
+        assert parts[0].i128 == int(a.sum())
+        assert parts[1].count == n
