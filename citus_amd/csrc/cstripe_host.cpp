@@ -414,7 +414,8 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
      * comfortably under 2^24), and a restricted-zstd RAW-block fallback is
      * bounded by the 21-bit block size (1 MB) */
     if (per > (1u << 23)) per = 1u << 23;
-    if (codec == CSTRIPE_COMP_ZSTD && per > (1u << 20)) per = 1u << 20;
+    if (codec == CSTRIPE_COMP_ZSTD && per > (96u << 10))
+        per = 96u << 10;               /* zstd block content cap is 128 KB */
 
     out.value_comp.clear();
     out.segs.clear();

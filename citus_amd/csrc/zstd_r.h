@@ -320,7 +320,10 @@ typedef struct {
 static inline int zr_compress(const uint8_t *src, int slen,
                               uint8_t *dst, int dcap, const zr_ctables *ct)
 {
-    if (slen <= 0 || slen > (1 << 20)) return 0;
+    /* zstd blocks carry at most 128 KB of regenerated content; this
+     * encoder emits ONE block per frame, so segments are capped below
+     * that (the writer splits chunks accordingly) */
+    if (slen <= 0 || slen > (110 << 10)) return 0;
     int op = 0;
     /* frame header: magic + FHD (single segment) + content size */
     if (op + 9 > dcap) return 0;
@@ -368,6 +371,10 @@ static inline int zr_compress(const uint8_t *src, int slen,
                     int ml = 4;
                     while (pos + ml < slen - 1 && src[cand + ml] == src[pos + ml])
                         ml++;
+                    if (ml > 131070) ml = 131070;   /* ML code ceiling */
+                    if (pos - anchor > 131000)
+                        return 0;   /* LL code ceiling: fall back to a raw
+                                     * frame rather than an unencodable run */
                     seqs[nseq].ll = (uint32_t)(pos - anchor);
                     seqs[nseq].off = (uint32_t)(pos - cand);
                     seqs[nseq].ml = (uint32_t)ml;
