@@ -192,6 +192,12 @@ def main():
         # ~20 s of timed steps: long enough for the driver-side activity
         # sampler to observe the GPU busy, bounded for tiny workloads
         args.steps = max(10, min(40000, int(20.0 / max(per_step_est, 1e-4))))
+        if dist:
+            # every rank MUST run the same step count (the per-step combine
+            # is a collective): take the max of the per-rank calibrations
+            t = torch.tensor([args.steps], device="cuda")
+            td.all_reduce(t, op=td.ReduceOp.MAX)
+            args.steps = int(t.item())
 
     decode_ms = []
     agg_ms = []

@@ -2767,6 +2767,16 @@ int csgpu_stage(cstripe_scan *s, int device_id)
         }
     }
 
+    /* invariant: the layout pass and the fill pass must agree exactly —
+     * any drift between them is silent corruption (heap overflow on the
+     * host copy or device reads past staged regions) */
+    if (align_up(dpos, 16) > data_bytes || align_up(spos, 16) > scratch_bytes + 15) {
+        cs_set_err("staging layout drift: filled %llu/%llu data, %llu/%llu scratch",
+                   (unsigned long long)dpos, (unsigned long long)data_bytes,
+                   (unsigned long long)spos, (unsigned long long)scratch_bytes);
+        csgpu_release(s);
+        return CSTRIPE_ERR;
+    }
     HIP_TRY(hipMemcpyAsync(g->d_data, h_data.data(), data_bytes, hipMemcpyHostToDevice, g->stream));
     if (!h_rank.empty())
         HIP_TRY(hipMemcpyAsync(g->d_rank, h_rank.data(), h_rank.size() * 4, hipMemcpyHostToDevice, g->stream));
