@@ -3413,6 +3413,11 @@ int csgpu_agg(cstripe_scan *s, const cstripe_agg_spec *aggs, uint32_t n_aggs,
             else if (p.n_preds == 2 && n_aggs == 2) launchp(multi_agg_kernel<2, 2, 8, 8>);
             else if (p.n_preds == 5 && n_aggs == 4) launchp(multi_agg_kernel<5, 4, 8, 8>);
             else launchp(multi_agg_kernel<-1, -1, 4>);
+        } else if (kvar == 4) { /* R8 + 7 waves/SIMD */
+            if (p.n_preds == 5 && n_aggs == 2) launchp(multi_agg_kernel<5, 2, 8, 7>);
+            else if (p.n_preds == 5 && n_aggs == 1) launchp(multi_agg_kernel<5, 1, 8, 7>);
+            else if (p.n_preds == 1 && n_aggs == 1) launchp(multi_agg_kernel<1, 1, 8, 7>);
+            else launchp(multi_agg_kernel<-1, -1, 4>);
         } else if (kvar == 2) { /* R=4 */
             if (p.n_preds == 5 && n_aggs == 2) launchp(multi_agg_kernel<5, 2, 4>);
             else if (p.n_preds == 5 && n_aggs == 1) launchp(multi_agg_kernel<5, 1, 4>);
