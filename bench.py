@@ -269,13 +269,19 @@ def main():
     # independent end-to-end check of writer -> stage -> decode -> filter ->
     # aggregate -> combine at the FULL benchmark size.
     parity = None
-    if world == 1:
+    if rank == 0:
+        # expected values cover EVERY rank's shards (seeds are deterministic:
+        # rank r uses base 42 + r*shards), so the pin also holds for the
+        # multi-GPU combined result
         shard_rows = [args.rows // args.shards + (args.rows % args.shards if i == 0 else 0)
                       for i in range(args.shards)]
+        all_seeds = [42 + r * args.shards + i
+                     for r in range(world) for i in range(args.shards)]
+        all_rows = shard_rows * world
         if args.query == "q6":
             exp_rev, exp_cnt = 0, 0
-            for i, nrows in enumerate(shard_rows):
-                r_, c_ = ca.expected_q6(nrows, base_seed + i)
+            for nrows, sd in zip(all_rows, all_seeds):
+                r_, c_ = ca.expected_q6(nrows, sd)
                 exp_rev += r_
                 exp_cnt += c_
             parity = ("bit-exact" if (result[0].i128 == exp_rev and
@@ -287,8 +293,8 @@ def main():
             RF = {0: ord("A"), 1: ord("N"), 2: ord("R")}
             LS = {0: ord("O"), 1: ord("F")}
             exp = {}
-            for i, nrows in enumerate(shard_rows):
-                for k, v in ca.expected_q1(nrows, base_seed + i).items():
+            for nrows, sd in zip(all_rows, all_seeds):
+                for k, v in ca.expected_q1(nrows, sd).items():
                     kk = (RF[k[0]], LS[k[1]])
                     cur = exp.setdefault(kk, [0, 0, 0, 0, 0])
                     for j in range(5):
