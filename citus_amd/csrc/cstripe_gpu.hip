@@ -128,6 +128,8 @@ struct cs_gpu_state {
     bool batch_decoded = false;      /* scratch holds decoded streams */
     SegDesc *d_zsegs = nullptr;      /* restricted-zstd segments (zstd_r.h) */
     uint32_t n_zsegs = 0;
+    uint32_t max_zseg_comp = 0;      /* largest zstd frame (LDS-variant gate) */
+    uint32_t max_zseg_dlen = 0;      /* largest zstd frame decompressed */
     zr_dtables *d_zrtab = nullptr;   /* predefined FSE decode tables */
     uint32_t *d_segstart = nullptr;  /* per (group, proj): first d_segs index */
     uint64_t greedy256_mask = 0;     /* proj cols stored as uniform 256B lz4 */
@@ -2400,6 +2402,332 @@ __global__ void zr_decode_kernel(const uint8_t *__restrict__ data,
     if (r != (int)s.decomp_len) atomicOr(err, 32);
 }
 
+/* ---- device-tuned restricted-frame decoder -----------------------------
+ * Same algorithm as zstd_r.h zr_decode_frame (which stays the host decoder
+ * and the spec reference — parity is pinned lane-for-lane by the GPU
+ * tests), restated for a 4-BYTE-ALIGNED LDS-resident frame. The reason it
+ * exists: ds_read/ds_write require natural alignment, so every unaligned
+ * 8-byte memcpy in the shared decoder compiles to EIGHT ds_read_u8 ops on
+ * the serial FSE chain. Here every LDS access is an aligned dword:
+ *  - bit-window refill = two ds_read_b32 (frame base is 4-aligned and the
+ *    LDS stride leaves >= 8 B slack, so the aligned window may run past
+ *    comp_len into the lane's own slack — those junk bits sit above the
+ *    marker bit and are never returned);
+ *  - FSE cell fetch = one ds_read_b32 (cells are 4 B);
+ *  - literal reads = funnel over aligned dwords.
+ * Global-memory accesses (decoded output, match copies) keep unaligned
+ * 8-byte memcpys — gfx950 global/flat loads are alignment-tolerant and
+ * already compile to dwordx2. */
+struct zr_brd {
+    const uint8_t *src;    /* 4-aligned LDS frame base */
+    int32_t bitpos;
+    int32_t wbase;         /* 4-aligned byte index of window LSB (-1 empty) */
+    uint64_t w;
+};
+
+__device__ inline uint32_t zr_lds_u32(const uint8_t *p)   /* p 4-aligned */
+{
+    return *(const uint32_t *)__builtin_assume_aligned(p, 4);
+}
+
+__device__ inline int zr_brd_init(zr_brd *b, const uint8_t *src, int len)
+{
+    b->src = src;
+    b->wbase = -1;
+    b->w = 0;
+    int last = len - 1;
+    while (last >= 0 && src[last] == 0) last--;
+    if (last < 0) return -1;
+    int hb = 7;
+    while (!(src[last] & (1 << hb))) hb--;
+    b->bitpos = last * 8 + hb;
+    return 0;
+}
+
+__device__ inline uint32_t zr_brd_read(zr_brd *b, int nbits)
+{
+    if (nbits == 0) return 0;
+    b->bitpos -= nbits;
+    const int32_t bp = b->bitpos < 0 ? 0 : b->bitpos;
+    const int32_t byte = bp >> 3;
+    if (b->wbase < 0 || byte < b->wbase || (bp - b->wbase * 8) + nbits > 64) {
+        int32_t base = (byte - 2) & ~3;     /* byte-base in [2,5] -> any
+                                             * <=16-bit read fits the window */
+        if (base < 0) base = 0;
+        b->w = (uint64_t)zr_lds_u32(b->src + base) |
+               ((uint64_t)zr_lds_u32(b->src + base + 4) << 32);
+        b->wbase = base;
+    }
+    const int32_t sh = bp - b->wbase * 8;
+    return (uint32_t)(b->w >> sh) & ((1u << nbits) - 1u);
+}
+
+/* one 4-byte FSE cell as a single dword: sym | nbits<<8 | base<<16 */
+__device__ inline uint32_t zr_cell_d(const zr_dcell *t, uint32_t s)
+{
+    return *(const uint32_t *)__builtin_assume_aligned(t + s, 4);
+}
+
+/* copy n bytes from the LDS frame (arbitrary offset so, frame base
+ * 4-aligned with slack) to global dst: aligned-dword funnel loads,
+ * unaligned 8-byte global stores */
+__device__ inline void zr_lds_copy(uint8_t *dst, const uint8_t *src4,
+                                   uint32_t so, uint32_t n)
+{
+    const uint32_t mis = so & 3u;
+    const uint8_t *al = src4 + (so & ~3u);
+    uint32_t i = 0;
+    if (n >= 8) {
+        const uint32_t sh = mis * 8u;
+        uint32_t w0 = zr_lds_u32(al);
+        for (; i + 8 <= n; i += 8) {
+            uint32_t w1 = zr_lds_u32(al + i + 4);
+            uint32_t w2 = zr_lds_u32(al + i + 8);
+            uint64_t lo = (uint64_t)w0 | ((uint64_t)w1 << 32);
+            uint64_t v = sh ? (lo >> sh) | ((uint64_t)w2 << (64 - sh)) : lo;
+            __builtin_memcpy(dst + i, &v, 8);
+            w0 = w2;
+        }
+    }
+    for (; i < n; i++) dst[i] = src4[so + i];
+}
+
+__device__ inline int zr_decode_frame_a4(const uint8_t *src, int slen,
+                                         uint8_t *dst, int dcap,
+                                         const zr_dcell *llt,
+                                         const zr_dcell *mlt,
+                                         const zr_dcell *oft)
+{
+    int ip = 0;
+    if (slen < 7) return -1;
+    if (!(src[0] == 0x28 && src[1] == 0xB5 && src[2] == 0x2F && src[3] == 0xFD))
+        return -1;
+    ip = 4;
+    const uint8_t fhd = src[ip++];
+    if (!(fhd == 0x20 || fhd == 0x60 || fhd == 0xA0)) return -1;
+    int content;
+    if (fhd == 0x20) {
+        content = src[ip++];
+    } else if (fhd == 0x60) {
+        content = 256 + src[ip] + (src[ip + 1] << 8);
+        ip += 2;
+    } else {
+        uint32_t f = 0;
+        for (int i = 0; i < 4; i++) f |= (uint32_t)src[ip + i] << (8 * i);
+        if (f > (1u << 23)) return -1;
+        content = (int)f;
+        ip += 4;
+    }
+    if (content > dcap) return -1;
+    const uint32_t bh = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8) |
+                        ((uint32_t)src[ip + 2] << 16);
+    ip += 3;
+    const int btype = (int)((bh >> 1) & 3);
+    const int bsize = (int)(bh >> 3);
+    if (ip + bsize > slen) return -1;
+    if (btype == 0) {                     /* raw block */
+        if (bsize != content) return -1;
+        zr_lds_copy(dst, src, (uint32_t)ip, (uint32_t)content);
+        return content;
+    }
+    if (btype != 2) return -1;
+    const int bend = ip + bsize;
+
+    const uint8_t lh = src[ip];           /* literals: RAW only */
+    if ((lh & 3) != 0) return -1;
+    int lit_size;
+    if (!(lh & 0x04)) { lit_size = lh >> 3; ip += 1; }
+    else if (!(lh & 0x08)) { lit_size = (lh >> 4) | ((int)src[ip + 1] << 4); ip += 2; }
+    else { lit_size = (lh >> 4) | ((int)src[ip + 1] << 4) | ((int)src[ip + 2] << 12); ip += 3; }
+    const uint32_t lit_off = (uint32_t)ip;
+    ip += lit_size;
+    if (ip > bend) return -1;
+
+    int nseq = src[ip++];
+    if (nseq >= 128) {
+        if (nseq == 255) { nseq = src[ip] + (src[ip + 1] << 8) + 0x7F00; ip += 2; }
+        else { nseq = ((nseq - 128) << 8) + src[ip]; ip += 1; }
+    }
+    int op = 0, lp = 0;
+    if (nseq == 0) {
+        if (lit_size != content) return -1;
+        zr_lds_copy(dst, src, lit_off, (uint32_t)content);
+        return content;
+    }
+    if (src[ip++] != 0x00) return -1;
+
+    zr_brd br;
+    if (zr_brd_init(&br, src + ip, bend - ip) < 0) return -1;
+    uint32_t sll = zr_brd_read(&br, ZR_LL_ACCLOG);
+    uint32_t sof = zr_brd_read(&br, ZR_OF_ACCLOG);
+    uint32_t sml = zr_brd_read(&br, ZR_ML_ACCLOG);
+
+    for (int n = 0; n < nseq; n++) {
+        const uint32_t cll = zr_cell_d(llt, sll);
+        const uint32_t cml = zr_cell_d(mlt, sml);
+        const uint32_t cof = zr_cell_d(oft, sof);
+        const uint32_t of_sym = cof & 0xFF, ml_sym = cml & 0xFF, ll_sym = cll & 0xFF;
+        const uint32_t of_val = (1u << of_sym) + zr_brd_read(&br, (int)of_sym);
+        uint32_t ml = ZR_ML_BASE[ml_sym] + zr_brd_read(&br, ZR_ML_BITS[ml_sym]);
+        uint32_t ll = ZR_LL_BASE[ll_sym] + zr_brd_read(&br, ZR_LL_BITS[ll_sym]);
+        if (n + 1 < nseq) {
+            sll = (cll >> 16) + zr_brd_read(&br, (int)((cll >> 8) & 0xFF));
+            sml = (cml >> 16) + zr_brd_read(&br, (int)((cml >> 8) & 0xFF));
+            sof = (cof >> 16) + zr_brd_read(&br, (int)((cof >> 8) & 0xFF));
+        }
+        if (of_val <= 3) return -1;
+        const uint32_t off = of_val - 3;
+        if (lp + (int)ll > lit_size || op + (int)(ll + ml) > content) return -1;
+        zr_lds_copy(dst + op, src, lit_off + (uint32_t)lp, ll);
+        op += (int)ll; lp += (int)ll;
+        if (off > (uint32_t)op) return -1;
+        {   /* match: global->global, unaligned 8B memcpys are fine there */
+            uint32_t i = 0;
+            if (off >= 8) {
+                for (; i + 8 <= ml; i += 8) {
+                    uint64_t v;
+                    __builtin_memcpy(&v, dst + op + (int)i - (int)off, 8);
+                    __builtin_memcpy(dst + op + (int)i, &v, 8);
+                }
+            }
+            for (; i < ml; i++) dst[op + (int)i] = dst[op + (int)i - (int)off];
+            op += (int)ml;
+        }
+    }
+    const int rem = lit_size - lp;
+    if (op + rem != content) return -1;
+    zr_lds_copy(dst + op, src, lit_off + (uint32_t)lp, (uint32_t)rem);
+    return content;
+}
+
+/* LDS-staged variant: the global kernel's wall is the serial FSE decode's
+ * dependent byte loads — each lane walks its own frame, the block's frame
+ * working set (~128 x 512 B) blows past L1, so every window refill / literal
+ * read is an uncoalesced L2-or-HBM round trip. Here the block first copies
+ * its 128 frames into LDS with coalesced dword loads (and the 640 B spec
+ * FSE tables once), then each lane decodes ITS frame entirely out of LDS
+ * via the aligned-dword decoder above; only the decoded output goes to
+ * global. Chosen by launch_decode when the largest frame fits the LDS
+ * budget (writer default 512 B zstd segments -> ~72 KB/block, 2 blocks/CU). */
+#define ZR_LDS_BLOCK 128
+__global__ __launch_bounds__(ZR_LDS_BLOCK) void zr_decode_lds_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, uint32_t n, uint32_t stride,
+    const zr_dtables *__restrict__ dt, int *__restrict__ err)
+{
+    extern __shared__ uint8_t zbuf[];   /* [ZR_LDS_BLOCK*stride] | tables */
+    zr_dcell *tll = (zr_dcell *)(zbuf + (size_t)ZR_LDS_BLOCK * stride);
+    zr_dcell *tml = tll + (1 << ZR_LL_ACCLOG);
+    zr_dcell *tof = tml + (1 << ZR_ML_ACCLOG);
+    __shared__ uint64_t s_src[ZR_LDS_BLOCK];
+    __shared__ uint32_t s_len[ZR_LDS_BLOCK];
+    {   /* spec tables: 640 B of 4 B cells, cooperative copy */
+        const uint32_t *ts = (const uint32_t *)dt;
+        uint32_t *td = (uint32_t *)tll;
+        for (uint32_t j = threadIdx.x; j < sizeof(zr_dtables) / 4; j += ZR_LDS_BLOCK)
+            td[j] = ts[j];
+    }
+    const uint32_t base_i = blockIdx.x * ZR_LDS_BLOCK;
+    const uint32_t i = base_i + threadIdx.x;
+    if (i < n) { s_src[threadIdx.x] = segs[i].src_off; s_len[threadIdx.x] = segs[i].comp_len; }
+    else s_len[threadIdx.x] = 0;
+    __syncthreads();
+    const uint32_t nf = n - base_i < ZR_LDS_BLOCK ? n - base_i : ZR_LDS_BLOCK;
+    for (uint32_t f = 0; f < nf; f++) {
+        /* coalesced staging: unaligned global dword loads (hardware-
+         * tolerated), aligned LDS dword writes */
+        const uint8_t *src = data + s_src[f];
+        uint32_t *d4 = (uint32_t *)(zbuf + (size_t)f * stride);
+        const uint32_t nw = (s_len[f] + 3) >> 2;
+        for (uint32_t j = threadIdx.x; j < nw; j += ZR_LDS_BLOCK) {
+            uint32_t v;
+            __builtin_memcpy(&v, src + 4 * j, 4);
+            d4[j] = v;
+        }
+    }
+    __syncthreads();
+    if (i >= n) return;
+    const SegDesc s = segs[i];
+    const int r = zr_decode_frame_a4(zbuf + (size_t)threadIdx.x * stride, (int)s.comp_len,
+                                     scratch + s.dst_off, (int)s.decomp_len,
+                                     tll, tml, tof);
+    if (r != (int)s.decomp_len) atomicOr(err, 32);
+}
+
+/* LDS-in + LDS-out variant: PMC on the kernel above shows 86% of wave
+ * cycles parked on s_waitcnt — with the frames LDS-resident, the residual
+ * latency chain is the OUTPUT path: every match copy is a load-after-store
+ * round trip through global L2 (stores bypass L1 on CDNA), serialized by
+ * vmcnt waits 8 bytes at a time. Here each lane decodes into its own LDS
+ * output slot, so match copies stay in LDS, and the block flushes decoded
+ * slots to global with coalesced 16 B stores at the end. Costs ~2x the LDS
+ * (1 block/CU at the 512 B writer default) but removes the global round
+ * trips from the serial chain. */
+__global__ __launch_bounds__(ZR_LDS_BLOCK) void zr_decode_lds2_kernel(
+    const uint8_t *__restrict__ data, uint8_t *__restrict__ scratch,
+    const SegDesc *__restrict__ segs, uint32_t n, uint32_t stride,
+    uint32_t ostride, const zr_dtables *__restrict__ dt, int *__restrict__ err)
+{
+    extern __shared__ uint8_t zbuf[];   /* [in frames][out slots][tables] */
+    uint8_t *obuf = zbuf + (size_t)ZR_LDS_BLOCK * stride;
+    zr_dcell *tll = (zr_dcell *)(obuf + (size_t)ZR_LDS_BLOCK * ostride);
+    zr_dcell *tml = tll + (1 << ZR_LL_ACCLOG);
+    zr_dcell *tof = tml + (1 << ZR_ML_ACCLOG);
+    __shared__ uint64_t s_src[ZR_LDS_BLOCK];
+    __shared__ uint64_t s_dst[ZR_LDS_BLOCK];
+    __shared__ uint32_t s_len[ZR_LDS_BLOCK];
+    __shared__ uint32_t s_dlen[ZR_LDS_BLOCK];
+    {
+        const uint32_t *ts = (const uint32_t *)dt;
+        uint32_t *td = (uint32_t *)tll;
+        for (uint32_t j = threadIdx.x; j < sizeof(zr_dtables) / 4; j += ZR_LDS_BLOCK)
+            td[j] = ts[j];
+    }
+    const uint32_t base_i = blockIdx.x * ZR_LDS_BLOCK;
+    const uint32_t i = base_i + threadIdx.x;
+    if (i < n) {
+        const SegDesc s = segs[i];
+        s_src[threadIdx.x] = s.src_off;
+        s_dst[threadIdx.x] = s.dst_off;
+        s_len[threadIdx.x] = s.comp_len;
+        s_dlen[threadIdx.x] = s.decomp_len;
+    } else { s_len[threadIdx.x] = 0; s_dlen[threadIdx.x] = 0; }
+    __syncthreads();
+    const uint32_t nf = n - base_i < ZR_LDS_BLOCK ? n - base_i : ZR_LDS_BLOCK;
+    for (uint32_t f = 0; f < nf; f++) {
+        const uint8_t *src = data + s_src[f];
+        uint32_t *d4 = (uint32_t *)(zbuf + (size_t)f * stride);
+        const uint32_t nw = (s_len[f] + 3) >> 2;
+        for (uint32_t j = threadIdx.x; j < nw; j += ZR_LDS_BLOCK) {
+            uint32_t v;
+            __builtin_memcpy(&v, src + 4 * j, 4);
+            d4[j] = v;
+        }
+    }
+    __syncthreads();
+    if (i < n) {
+        const int r = zr_decode_frame_a4(zbuf + (size_t)threadIdx.x * stride,
+                                         (int)s_len[threadIdx.x],
+                                         obuf + (size_t)threadIdx.x * ostride,
+                                         (int)s_dlen[threadIdx.x],
+                                         tll, tml, tof);
+        if (r != (int)s_dlen[threadIdx.x]) atomicOr(err, 32);
+    }
+    __syncthreads();
+    for (uint32_t f = 0; f < nf; f++) {     /* coalesced flush, exact dlen */
+        uint8_t *dst = scratch + s_dst[f];
+        const uint8_t *out = obuf + (size_t)f * ostride;   /* 4-aligned slot */
+        const uint32_t dlen = s_dlen[f];
+        const uint32_t vec = dlen >> 2;
+        for (uint32_t j = threadIdx.x; j < vec; j += ZR_LDS_BLOCK) {
+            const uint32_t v = zr_lds_u32(out + 4 * j);
+            __builtin_memcpy(dst + 4 * j, &v, 4);   /* global: unaligned ok */
+        }
+        for (uint32_t j = (vec << 2) + threadIdx.x; j < dlen; j += ZR_LDS_BLOCK)
+            dst[j] = out[j];
+    }
+}
+
 /* =====================================================================
  * staging
  * ===================================================================== */
@@ -2653,6 +2981,8 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                     sd.dst_off = spos + sg.decomp_off;
                     sd.comp_len = sg.comp_len;
                     sd.decomp_len = sg.decomp_len;
+                    if (sg.comp_len > g->max_zseg_comp) g->max_zseg_comp = sg.comp_len;
+                    if (sd.decomp_len > g->max_zseg_dlen) g->max_zseg_dlen = sd.decomp_len;
                     h_zsegs.push_back(sd);
                 }
                 dpos += align_up(nd.n.value_len, 16);
@@ -2783,6 +3113,15 @@ int csgpu_stage(cstripe_scan *s, int device_id)
     if (!h_segs.empty())
         HIP_TRY(hipMemcpyAsync(g->d_segs, h_segs.data(), h_segs.size() * sizeof(SegDesc), hipMemcpyHostToDevice, g->stream));
     if (!h_zsegs.empty()) {
+        /* sort frames by compressed length so a wave's 64 lanes decode
+         * similar-cost frames: the serial FSE decode runs in lockstep per
+         * wave (and per block up to the flush barrier), so one expensive
+         * frame otherwise gates 127 cheap ones (frames are independent —
+         * dst_off travels with the descriptor) */
+        std::stable_sort(h_zsegs.begin(), h_zsegs.end(),
+                         [](const SegDesc &a, const SegDesc &b) {
+                             return a.comp_len > b.comp_len;
+                         });
         HIP_TRY(hipMemcpyAsync(g->d_zsegs, h_zsegs.data(), h_zsegs.size() * sizeof(SegDesc), hipMemcpyHostToDevice, g->stream));
         static zr_dtables h_zrtab;
         static bool zrtab_init = false;
@@ -2869,10 +3208,40 @@ int csgpu_stage(cstripe_scan *s, int device_id)
 static int launch_decode(cs_gpu_state *g)
 {
     if (g->n_zsegs > 0) {
-        const uint32_t grid = (g->n_zsegs + 255) / 256;
-        hipLaunchKernelGGL(zr_decode_kernel, dim3(grid), dim3(256), 0, g->stream,
-                           g->d_data, g->d_scratch, g->d_zsegs, g->n_zsegs,
-                           g->d_zrtab, g->d_error);
+        /* CSTRIPE_ZR_VARIANT: 0 = force global kernel (A/B knob), default =
+         * LDS-staged when the largest frame fits the budget */
+        const char *zv = getenv("CSTRIPE_ZR_VARIANT");
+        const int force = zv ? atoi(zv) : -1;
+        /* stride: largest frame + 8 B slack (the aligned bit window and the
+         * funnel loads may read a few bytes past comp_len, into the lane's
+         * own slack), 16-aligned; tweaked off bank-aligned lane spacing */
+        /* strides in ODD dword counts: lane l's frame starts at l*stride, so
+         * equal-progress lanes hit bank (l*stride/4 + c) mod 64 — an odd
+         * dword stride makes those 64 distinct banks (even residues conflict
+         * 2-32 way; same rule as the lz4 lane kernel's 276/532/1044) */
+        uint32_t stride = ((g->max_zseg_comp + 8 + 3) & ~3u) + 4;
+        if (((stride / 4) & 1) == 0) stride += 4;
+        uint32_t ostride = ((g->max_zseg_dlen + 3) & ~3u) + 4;
+        if (((ostride / 4) & 1) == 0) ostride += 4;
+        const uint32_t grid = (g->n_zsegs + ZR_LDS_BLOCK - 1) / ZR_LDS_BLOCK;
+        const size_t lds1 = (size_t)ZR_LDS_BLOCK * stride + sizeof(zr_dtables);
+        const size_t lds2 = lds1 + (size_t)ZR_LDS_BLOCK * ostride;
+        if ((force == -1 || force >= 2) && lds2 <= 152 * 1024) {
+            hipLaunchKernelGGL(zr_decode_lds2_kernel, dim3(grid), dim3(ZR_LDS_BLOCK),
+                               lds2, g->stream,
+                               g->d_data, g->d_scratch, g->d_zsegs, g->n_zsegs,
+                               stride, ostride, g->d_zrtab, g->d_error);
+        } else if (force != 0 && lds1 <= 152 * 1024) {
+            hipLaunchKernelGGL(zr_decode_lds_kernel, dim3(grid), dim3(ZR_LDS_BLOCK),
+                               lds1, g->stream,
+                               g->d_data, g->d_scratch, g->d_zsegs, g->n_zsegs,
+                               stride, g->d_zrtab, g->d_error);
+        } else {
+            const uint32_t grid = (g->n_zsegs + 255) / 256;
+            hipLaunchKernelGGL(zr_decode_kernel, dim3(grid), dim3(256), 0, g->stream,
+                               g->d_data, g->d_scratch, g->d_zsegs, g->n_zsegs,
+                               g->d_zrtab, g->d_error);
+        }
         HIP_TRY(hipGetLastError());
     }
     if (g->n_segs == 0) return CSTRIPE_OK;

@@ -407,8 +407,11 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     size_t per = (target / 16) * 16;    /* exact target (16B-aligned splits) —
                                          * uniform segments are what the fused
                                          * GPU kernel's tile math assumes */
-    if (codec == CSTRIPE_COMP_ZSTD && per < 512)
-        per = 512;                      /* ~14 B frame overhead vs lane count */
+    if (codec == CSTRIPE_COMP_ZSTD && per < 256)
+        per = 256;                      /* ~13 B frame overhead (5%) is worth
+                                         * the GPU decoder occupancy: LDS
+                                         * bytes/frame bound waves/CU, and
+                                         * 256 B frames double them vs 512 */
     if (per == 0) per = n;
     /* caps: segment decomp_len carries the mode byte in bits 24-31 (8 MB is
      * comfortably under 2^24), and a restricted-zstd RAW-block fallback is
