@@ -502,6 +502,18 @@ __global__ __launch_bounds__(WAVE) void lz4_decode_lds_kernel(
  * wave __shfl_down reduce; LDS cross-wave reduce; one AccCell per block.
  * ===================================================================== */
 
+/* canonical-stream byte position of value j's low-L bytes, by mode family:
+ * lz4 P(L) interleaves them with match tokens (csf_canon_p_pos); the
+ * restricted-zstd P(L) frame holds them densely in its literal section
+ * (csf_canon_zrp_pos) */
+__device__ inline uint32_t canon_pos(uint8_t mode, uint32_t L, uint32_t j)
+{
+    if ((mode & 0xF0u) == CSF_SEGMODE_ZRP_BASE)
+        return j == 0 ? 15u : 23u + (j - 1u) * L;
+    const uint32_t pos = j * (L + 3u) + (6u - L);
+    return j == 0 ? 1u : (j == 1 ? 9u : pos);
+}
+
 __device__ inline bool col_value(const uint8_t *__restrict__ data,
                                  const uint8_t *__restrict__ scratch,
                                  const uint32_t *__restrict__ rank,
@@ -526,12 +538,11 @@ __device__ inline bool col_value(const uint8_t *__restrict__ data,
             base += (uint64_t)cl.hval;           /* skip literal-run header */
         } else {
             uint64_t raw;
-            if (cl.mode == CSF_SEGMODE_CONST) {
+            if (cl.mode == CSF_SEGMODE_CONST || cl.mode == CSF_SEGMODE_ZR_CONST) {
                 raw = (uint64_t)cl.hval;
-            } else {                             /* P(L) */
+            } else {                             /* P(L), lz4 or zstd layout */
                 const uint32_t Lx = cl.L;
-                uint32_t pos = idx * (Lx + 3u) + (6u - Lx);
-                pos = idx == 0 ? 1u : (idx == 1 ? 9u : pos);
+                const uint32_t pos = canon_pos(cl.mode, Lx, idx);
                 uint64_t lo;
                 __builtin_memcpy(&lo, base + pos, 8);
                 const uint64_t m = (~0ull) >> ((8u - Lx) * 8u);
@@ -977,10 +988,31 @@ __device__ inline void col_multi(const uint8_t *__restrict__ data,
     const uint8_t *base = (cl.flags & 1) ? scratch + cl.val_off : data + cl.val_off;
     if ((cl.flags & 4) && cl.mode != CSF_SEGMODE_LIT) {
         uint64_t r[R];
-        if (cl.mode == CSF_SEGMODE_CONST) {
+        if (cl.mode == CSF_SEGMODE_CONST || cl.mode == CSF_SEGMODE_ZR_CONST) {
             #pragma unroll
             for (int k = 0; k < R; k++) r[k] = (uint64_t)cl.hval;
-        } else {                              /* P(L) */
+        } else if ((cl.mode & 0xF0u) == CSF_SEGMODE_ZRP_BASE) {
+            /* zstd P(L): dense low-L bytes in the literal section, byte
+             * stride L (canon_pos); only j=0 is special */
+            const uint32_t Lx = cl.L;
+            const uint64_t m = (~0ull) >> ((8u - Lx) * 8u);
+            if (row == 0) {
+                #pragma unroll
+                for (int k = 0; k < R; k++) {
+                    uint64_t a;
+                    __builtin_memcpy(&a, base + canon_pos(cl.mode, Lx, row + k), 8);
+                    r[k] = (a & m) | (uint64_t)cl.hval;
+                }
+            } else {
+                const uint32_t pos = 23u + (row - 1u) * Lx;
+                switch (Lx) {                 /* wave-uniform */
+                    case 1: canon_extract<R, 1>(base, pos, m, (uint64_t)cl.hval, r); break;
+                    case 2: canon_extract<R, 2>(base, pos, m, (uint64_t)cl.hval, r); break;
+                    case 3: canon_extract<R, 3>(base, pos, m, (uint64_t)cl.hval, r); break;
+                    default: canon_extract<R, 4>(base, pos, m, (uint64_t)cl.hval, r); break;
+                }
+            }
+        } else {                              /* lz4 P(L) */
             const uint32_t Lx = cl.L, step = Lx + 3u;
             const uint64_t m = (~0ull) >> ((8u - Lx) * 8u);
             if (row < 2) {                    /* j=0 -> 1, j=1 -> 9 specials */
@@ -2368,11 +2400,10 @@ __global__ void canon_decode_kernel(const uint8_t *__restrict__ src,
         return;
     }
     uint64_t v;
-    if (mode == CSF_SEGMODE_CONST) {
+    if (mode == CSF_SEGMODE_CONST || mode == CSF_SEGMODE_ZR_CONST) {
         v = hval;
     } else {
-        uint32_t pos = i * (L + 3u) + (6u - L);
-        pos = i == 0 ? 1u : (i == 1 ? 9u : pos);
+        const uint32_t pos = canon_pos((uint8_t)mode, L, i);
         uint64_t lo;
         __builtin_memcpy(&lo, src + pos, 8);
         const uint64_t m = (~0ull) >> ((8u - L) * 8u);
@@ -2734,6 +2765,20 @@ __global__ __launch_bounds__(ZR_LDS_BLOCK) void zr_decode_lds2_kernel(
 
 static uint64_t align_up(uint64_t x, uint64_t a) { return (x + a - 1) & ~(a - 1); }
 
+/* single-segment chunk with a CLOSED-FORM canonical stream (lz4 P/CONST/LIT
+ * or restricted-zstd P/CONST, format.h): staged compressed, read in place */
+static inline bool cs_node_canon(const cs_skipnode &nd)
+{
+    if (nd.n.n_segs != 1) return false;
+    const uint8_t m = nd.seg_modes[0];
+    if (nd.n.comp_type == CSTRIPE_COMP_LZ4)
+        return m != CSF_SEGMODE_GENERIC;
+    if (nd.n.comp_type == CSTRIPE_COMP_ZSTD)
+        return m == CSF_SEGMODE_ZR_CONST ||
+               (m > CSF_SEGMODE_ZRP_BASE && m <= CSF_SEGMODE_ZRP_BASE + 4);
+    return false;
+}
+
 void csgpu_release(cstripe_scan *s)
 {
     if (!s || !s->gpu) return;
@@ -2815,10 +2860,8 @@ int csgpu_stage(cstripe_scan *s, int device_id)
             if (g->proj_of_col[c] < 0) continue;
             const cs_skipnode &nd = st.nodes[c][sc.chunk];
             data_bytes = align_up(data_bytes, 8) + align_up((rows + 7) / 8, 8); /* exists, 8B padded */
-            const bool canon = nd.n.comp_type == CSTRIPE_COMP_LZ4 &&
-                               nd.n.n_segs == 1 &&
-                               nd.seg_modes[0] != CSF_SEGMODE_GENERIC;
-            bool zr_dev = nd.n.comp_type == CSTRIPE_COMP_ZSTD;
+            const bool canon = cs_node_canon(nd);
+            bool zr_dev = !canon && nd.n.comp_type == CSTRIPE_COMP_ZSTD;
             if (zr_dev)
                 for (uint8_t m : nd.seg_modes)
                     if (m != CSF_SEGMODE_ZR) { zr_dev = false; break; }
@@ -2953,16 +2996,14 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 if (!cp || !(cl.flags & 2)) g->all_canonP = false;
             }
 
-            const bool canon = nd.n.comp_type == CSTRIPE_COMP_LZ4 &&
-                               nd.n.n_segs == 1 &&
-                               nd.seg_modes[0] != CSF_SEGMODE_GENERIC;
+            const bool canon = cs_node_canon(nd);
             if (canon || !(cl.flags & 2) || cl.type != CSTRIPE_I64 ||
                 nd.n.comp_type != CSTRIPE_COMP_LZ4)
                 g->fusable = false;
             if (canon || !(cl.flags & 2) || (cl.width != 8 && cl.width != 1) ||
                 nd.n.comp_type != CSTRIPE_COMP_LZ4)
                 g->fusable_mixed = false;
-            bool zr_dev = nd.n.comp_type == CSTRIPE_COMP_ZSTD;
+            bool zr_dev = !canon && nd.n.comp_type == CSTRIPE_COMP_ZSTD;
             if (zr_dev)
                 for (uint8_t m : nd.seg_modes)
                     if (m != CSF_SEGMODE_ZR) { zr_dev = false; break; }
@@ -3001,9 +3042,14 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 if (cl.mode == CSF_SEGMODE_LIT) {
                     cl.hval = (int64_t)csf_canon_lit_hdr(nd.segs[0].decomp_len);
                 } else {
+                    /* v0 full: first literal byte — offset 1 in a canonical
+                     * LZ4 block (token first), 15 in a canonical zstd frame
+                     * (fixed header, format.h) */
+                    const bool zrm = cl.mode >= CSF_SEGMODE_ZRP_BASE;
                     uint64_t v0;
-                    memcpy(&v0, strm + 1, 8);    /* v0 full, S0's first literal */
-                    if (cl.mode == CSF_SEGMODE_CONST) {
+                    memcpy(&v0, strm + (zrm ? 15 : 1), 8);
+                    if (cl.mode == CSF_SEGMODE_CONST ||
+                        cl.mode == CSF_SEGMODE_ZR_CONST) {
                         cl.hval = (int64_t)v0;
                     } else {                      /* P(L) */
                         cl.L = (uint8_t)(cl.mode & 0xF);

@@ -354,11 +354,14 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
     }
 
     /* ---- canonical parses (closed-form GPU access; see format.h) ----
-     * The emitted stream is a standard LZ4 block either way; canonical just
-     * picks a parse whose value positions are computable, killing the GPU's
-     * per-sequence parse chain (round-1's instruction wall). */
-    if (codec == CSTRIPE_COMP_LZ4 && w->opts.canonical && width == 8 &&
-        raw.size() >= 24 && raw.size() < (1u << 24)) {
+     * The emitted stream is a standard LZ4 block / zstd frame either way;
+     * canonical just picks a parse whose value positions are computable,
+     * killing the GPU's per-sequence/FSE parse chain (round-1's instruction
+     * wall). */
+    if ((codec == CSTRIPE_COMP_LZ4 || codec == CSTRIPE_COMP_ZSTD) &&
+        w->opts.canonical && width == 8 &&
+        raw.size() >= 24 && raw.size() < (1u << 24) &&
+        (codec == CSTRIPE_COMP_LZ4 || raw.size() < (127u << 10))) {
         const size_t nv = raw.size() / 8;
         uint64_t v0, x = 0;
         memcpy(&v0, raw.data(), 8);
@@ -368,17 +371,32 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
             memcpy(&vi, rp + i * 8, 8);
             x |= vi ^ v0;
         }
+        static const zr_ctables zct_c = [] {
+            zr_ctables t;
+            zr_build_ctables(&t);
+            return t;
+        }();
         int mode = -1, csz = 0;
         size_t cap = raw.size() + raw.size() / 8 + 64;
         out.value_comp.resize(cap);
         if (x == 0) {
-            csz = lz4e_canon_const(rp, (int)nv, out.value_comp.data(), (int)cap);
-            if (csz > 0) mode = CSF_SEGMODE_CONST;
+            if (codec == CSTRIPE_COMP_LZ4) {
+                csz = lz4e_canon_const(rp, (int)nv, out.value_comp.data(), (int)cap);
+                if (csz > 0) mode = CSF_SEGMODE_CONST;
+            } else {
+                csz = zr_canon_const(rp, (int)nv, out.value_comp.data(), (int)cap, &zct_c);
+                if (csz > 0) mode = CSF_SEGMODE_ZR_CONST;
+            }
         } else {
             int L = (63 - __builtin_clzll(x)) / 8 + 1;
             if (L <= 4) {
-                csz = lz4e_canon_p(rp, (int)nv, L, out.value_comp.data(), (int)cap);
-                if (csz > 0) mode = (int)(CSF_SEGMODE_P_BASE | (uint32_t)L);
+                if (codec == CSTRIPE_COMP_LZ4) {
+                    csz = lz4e_canon_p(rp, (int)nv, L, out.value_comp.data(), (int)cap);
+                    if (csz > 0) mode = (int)(CSF_SEGMODE_P_BASE | (uint32_t)L);
+                } else {
+                    csz = zr_canon_p(rp, (int)nv, L, out.value_comp.data(), (int)cap, &zct_c);
+                    if (csz > 0) mode = (int)(CSF_SEGMODE_ZRP_BASE | (uint32_t)L);
+                }
             }
             /* L > 4: high bytes vary — no canonical parse; the greedy path
              * below runs, and its incompressible->NONE fallback (reference
@@ -391,7 +409,7 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
             csf_seg s{0, (uint32_t)csz, 0,
                       (uint32_t)raw.size() | ((uint32_t)mode << 24)};
             out.segs.push_back(s);
-            out.node.comp_type = CSTRIPE_COMP_LZ4;
+            out.node.comp_type = codec;
             out.node.comp_level = 0;
             out.node.n_segs = 1;
             out.node.value_len = (uint64_t)csz;

@@ -128,6 +128,13 @@ typedef struct csf_seg {
                                     * emitted (raw literals + predefined-FSE
                                     * sequences, zstd_r.h) — device-decodable;
                                     * untagged zstd frames host-decode */
+#define CSF_SEGMODE_ZRP_BASE 0x60u /* 0x60|L: CANONICAL restricted-zstd P(L)
+                                    * frame (zr_canon_p): value 0's bytes at
+                                    * stream offset 15, value j>=1's low-L
+                                    * bytes at 23+(j-1)*L — closed-form GPU
+                                    * access, still ZSTD_decompress-decodable */
+#define CSF_SEGMODE_ZR_CONST 0x70u /* canonical restricted-zstd constant
+                                    * (zr_canon_const): v0 at offset 15 */
 #define CSF_SEG_DLEN_MASK   0x00FFFFFFu
 
 static inline uint32_t csf_seg_dlen(const csf_seg *s) { return s->decomp_len & CSF_SEG_DLEN_MASK; }
@@ -140,6 +147,14 @@ static inline uint32_t csf_canon_p_pos(uint32_t j, uint32_t L)
     if (j == 0) return 1;
     if (j == 1) return 9;
     return j * (L + 3) + (6 - L);
+}
+
+/* stream byte offset of value j's low-L bytes inside a canonical
+ * restricted-zstd P(L) frame (zr_canon_p layout): j=0 -> 15 (v0 stored
+ * whole; its low-L bytes start there), else 23 + (j-1)*L */
+static inline uint32_t csf_canon_zrp_pos(uint32_t j, uint32_t L)
+{
+    return j == 0 ? 15u : 23u + (j - 1u) * L;
 }
 
 /* literal-run header size for a mode-LIT segment of len decompressed bytes */

@@ -301,8 +301,6 @@ static inline int zr_cstate_flush(zr_cstate *st, const zr_ctable *ct, zr_bw *bw)
     return zr_bw_add(bw, st->value, ct->acclog);
 }
 
-#endif /* CSTRIPE_ZSTD_R_H */
-
 /* ---------------- greedy matcher (zstd min-match 3) ---------------- */
 
 #define ZR_MAX_SEQS 4096
@@ -391,7 +389,7 @@ static inline int zr_compress(const uint8_t *src, int slen,
         lit_total += trailing;
     }
 
-    int payload = 0;
+
     if (nseq == 0) {
         /* nothing found: raw block */
         const uint32_t bh = 1u | (0u << 1) | ((uint32_t)slen << 3);
@@ -667,3 +665,122 @@ ZR_HOSTDEV static inline int zr_decode_frame(const uint8_t *src, int slen,
     }
     return content;
 }
+
+/* ---------------- canonical frame emitters (closed-form GPU access) ----
+ * The zstd analogue of lz4_enc.h's canonical parses: a VALID restricted
+ * frame (standard RFC 8878, raw literals + predefined-FSE sequences —
+ * ZSTD_decompress-decodable, pinned in tests) whose literal section lays
+ * the values out at CLOSED-FORM positions, so the GPU reads values straight
+ * from the compressed stream and never walks the FSE chain (the per-lane
+ * serial decode is the instruction wall, round-1 VERDICT #1 for LZ4).
+ * Fixed header shape: magic(4) + FHD 0xA0(1) + FCS(4) + block header(3) +
+ * 3-byte literals header = 15 bytes to the first literal byte.
+ *   P(L):  literals = v0 (8 B) + low-L bytes of v1..v_{n-1};
+ *          sequences = S0 (ll=8+L, ml=8-L, off=8) then n-2 x (ll=L,
+ *          ml=8-L, off=8); value 0's bytes at 15, value j>=1's low-L bytes
+ *          at 23 + (j-1)*L (csf_canon_zrp_pos)
+ *   CONST: literals = v0 (8 B); one sequence (ll=8, ml=8(n-1), off=8)
+ * Requires 1 <= L <= 4, n >= 3, 8n < 127 KB (zstd block content cap). */
+#define ZR_CANON_LIT0 15
+
+static inline int zr_canon_hdr(uint8_t *dst, int dcap, uint32_t content,
+                               uint32_t lit_size)
+{
+    if (dcap < ZR_CANON_LIT0) return -1;
+    int op = 0;
+    dst[op++] = 0x28; dst[op++] = 0xB5; dst[op++] = 0x2F; dst[op++] = 0xFD;
+    dst[op++] = 0xA0;                      /* single_segment, FCS 4 bytes */
+    dst[op++] = (uint8_t)content; dst[op++] = (uint8_t)(content >> 8);
+    dst[op++] = (uint8_t)(content >> 16); dst[op++] = (uint8_t)(content >> 24);
+    op += 3;                               /* block header backpatched */
+    dst[op++] = (uint8_t)(0x0C | ((lit_size & 0x0F) << 4));   /* RAW, 3-byte */
+    dst[op++] = (uint8_t)(lit_size >> 4);
+    dst[op++] = (uint8_t)(lit_size >> 12);
+    return op;                             /* == ZR_CANON_LIT0 */
+}
+
+static inline int zr_canon_close(uint8_t *dst, int op)   /* backpatch bh */
+{
+    const uint32_t bsize = (uint32_t)(op - 9 - 3);
+    if (bsize > 0x1FFFFF) return 0;
+    const uint32_t bh = 1u | (2u << 1) | (bsize << 3);     /* last, compressed */
+    dst[9] = (uint8_t)bh; dst[10] = (uint8_t)(bh >> 8); dst[11] = (uint8_t)(bh >> 16);
+    return op;
+}
+
+/* canonical P(L): n width-8 values sharing their high (8-L) bytes */
+static inline int zr_canon_p(const uint8_t *src, int n, int L,
+                             uint8_t *dst, int dcap, const zr_ctables *ct)
+{
+    if (n < 3 || L < 1 || L > 4 || (int64_t)n * 8 >= (127 << 10)) return 0;
+    const uint32_t content = (uint32_t)n * 8;
+    const uint32_t lit_size = 8 + (uint32_t)(n - 1) * (uint32_t)L;
+    int op = zr_canon_hdr(dst, dcap, content, lit_size);
+    if (op < 0 || op + (int)lit_size + 4 > dcap) return 0;
+    memcpy(dst + op, src, 8);              /* v0 full */
+    op += 8;
+    for (int j = 1; j < n; j++) {
+        memcpy(dst + op, src + (size_t)j * 8, (size_t)L);
+        op += L;
+    }
+    const int nseq = n - 1;
+    if (nseq < 128) dst[op++] = (uint8_t)nseq;
+    else { dst[op++] = (uint8_t)((nseq >> 8) + 128); dst[op++] = (uint8_t)nseq; }
+    dst[op++] = 0x00;                      /* predefined modes */
+    const uint8_t llc_first = (uint8_t)(8 + L), llc_rest = (uint8_t)L;
+    const uint8_t mlc = (uint8_t)(8 - L - 3);
+    const uint8_t ofc = 3;                 /* off 8 -> value 11: 3 extra bits */
+    zr_bw bw;
+    zr_bw_init(&bw, dst + op, dcap - op);
+    zr_cstate sml, sof, sll;
+    zr_cstate_init(&sml, &ct->ml, mlc);
+    zr_cstate_init(&sof, &ct->of, ofc);
+    zr_cstate_init(&sll, &ct->ll, llc_rest);            /* last seq (n>=3) */
+    if (zr_bw_add(&bw, 3, 3) < 0) return 0;             /* last seq OF extra
+                                                         * (LL/ML codes here
+                                                         * carry 0 bits) */
+    for (int i = nseq - 2; i >= 0; i--) {
+        if (zr_cstate_encode(&sof, &ct->of, ofc, &bw) < 0) return 0;
+        if (zr_cstate_encode(&sml, &ct->ml, mlc, &bw) < 0) return 0;
+        if (zr_cstate_encode(&sll, &ct->ll, i == 0 ? llc_first : llc_rest, &bw) < 0) return 0;
+        if (zr_bw_add(&bw, 3, 3) < 0) return 0;
+    }
+    if (zr_cstate_flush(&sml, &ct->ml, &bw) < 0) return 0;
+    if (zr_cstate_flush(&sof, &ct->of, &bw) < 0) return 0;
+    if (zr_cstate_flush(&sll, &ct->ll, &bw) < 0) return 0;
+    const int bs = zr_bw_close(&bw);
+    if (bs < 0) return 0;
+    return zr_canon_close(dst, op + bs);
+}
+
+/* canonical constant: n equal width-8 values */
+static inline int zr_canon_const(const uint8_t *src, int n,
+                                 uint8_t *dst, int dcap, const zr_ctables *ct)
+{
+    if (n < 3 || (int64_t)n * 8 >= (127 << 10)) return 0;
+    const uint32_t content = (uint32_t)n * 8;
+    int op = zr_canon_hdr(dst, dcap, content, 8);
+    if (op < 0 || op + 40 > dcap) return 0;
+    memcpy(dst + op, src, 8);              /* v0 */
+    op += 8;
+    dst[op++] = 1;                         /* nseq */
+    dst[op++] = 0x00;
+    const uint32_t ml = (uint32_t)(n - 1) * 8;
+    const uint8_t llc = 8, mlc = zr_ml_code(ml), ofc = 3;
+    zr_bw bw;
+    zr_bw_init(&bw, dst + op, dcap - op);
+    zr_cstate sml, sof, sll;
+    zr_cstate_init(&sml, &ct->ml, mlc);
+    zr_cstate_init(&sof, &ct->of, ofc);
+    zr_cstate_init(&sll, &ct->ll, llc);
+    if (zr_bw_add(&bw, ml - ZR_ML_BASE[mlc], ZR_ML_BITS[mlc]) < 0) return 0;
+    if (zr_bw_add(&bw, 3, 3) < 0) return 0;
+    if (zr_cstate_flush(&sml, &ct->ml, &bw) < 0) return 0;
+    if (zr_cstate_flush(&sof, &ct->of, &bw) < 0) return 0;
+    if (zr_cstate_flush(&sll, &ct->ll, &bw) < 0) return 0;
+    const int bs = zr_bw_close(&bw);
+    if (bs < 0) return 0;
+    return zr_canon_close(dst, op + bs);
+}
+
+#endif /* CSTRIPE_ZSTD_R_H */

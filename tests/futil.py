@@ -14,6 +14,13 @@ SEGMODE_P_BASE = 0x10
 SEGMODE_CONST = 0x20
 SEGMODE_LIT = 0x30
 SEGMODE_ZR = 0x50
+SEGMODE_ZRP_BASE = 0x60          # 0x60|L: canonical restricted-zstd P(L)
+SEGMODE_ZR_CONST = 0x70
+
+
+def is_zr_mode(m):
+    """any restricted-zstd stream shape (generic frames or canonical)"""
+    return m == SEGMODE_ZR or m == SEGMODE_ZR_CONST or 0x61 <= m <= 0x64
 
 
 def read_footer(path):

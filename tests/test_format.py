@@ -495,7 +495,10 @@ def test_zstd_restricted_system_decodable(tmp_path):
         if node["comp_type"] != ca.COMP_ZSTD:
             assert name == "rand"      # incompressible -> raw NONE (ref rule)
             continue
-        assert all(s["mode"] == futil.SEGMODE_ZR for s in node["segs"]), name
+        # P-eligible columns now get a single-segment CANONICAL frame
+        # (0x60|L / 0x70); anything else carries generic ZR frames — every
+        # one of them must still be plain zstd to the system decoder below
+        assert all(futil.is_zr_mode(s["mode"]) for s in node["segs"]), name
         comp = futil.chunk_stream(path, node)
         out = bytearray(node["decompressed_size"])
         pos = 0
@@ -583,3 +586,49 @@ def test_zstd_large_segment_frames(tmp_path):
         parts, _ = t.scan_agg([], [(ca.AGG_SUM_I64, 0), (ca.AGG_COUNT_STAR, -1)])
         assert parts[0].i128 == int(a.sum())
         assert parts[1].count == n
+
+
+def test_zstd_canonical_layout(tmp_path):
+    """Canonical restricted-zstd frames (zr_canon_p / zr_canon_const): value
+    bytes sit at the CLOSED-FORM literal-section positions format.h documents
+    (csf_canon_zrp_pos: value 0 at 15, value j>=1 low-L at 23+(j-1)*L), one
+    segment, mode 0x60|L / 0x70 — and every frame is still plain zstd to the
+    system ZSTD_decompress (the reference's DecompressBuffer call,
+    columnar_compression.c:207)."""
+    n = 25000
+    zstd = C.CDLL("libzstd.so.1")
+    zstd.ZSTD_decompress.restype = C.c_size_t
+    zstd.ZSTD_decompress.argtypes = [C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
+    zstd.ZSTD_isError.restype = C.c_uint
+    zstd.ZSTD_isError.argtypes = [C.c_size_t]
+    cases = [
+        ("p2", (np.arange(n, dtype=np.int64) * 7) % 5000, 0x62, 2),
+        ("p1", np.asarray((np.arange(n) * 31) % 200, dtype=np.int64), 0x61, 1),
+        ("const", np.full(n, 0x1234, dtype=np.int64), 0x70, 0),
+    ]
+    for name, a, want_mode, L in cases:
+        path = str(tmp_path / f"zc_{name}.cs")
+        ca.write_table(path, [("a", ca.I64, 0)], [np.ascontiguousarray(a)],
+                       compression=ca.COMP_ZSTD)
+        foot = futil.read_footer(path)
+        node = foot["stripes"][0]["nodes"][0][0]
+        assert node["comp_type"] == ca.COMP_ZSTD, name
+        assert node["n_segs"] == 1 and node["segs"][0]["mode"] == want_mode, name
+        comp = futil.chunk_stream(path, node)
+        rows0 = node["decompressed_size"] // 8
+        # closed-form position pin against the raw stream bytes
+        assert comp[15:23] == a[:1].tobytes(), name
+        if L:
+            for j in (1, 2, 777, rows0 - 1):
+                pos = 23 + (j - 1) * L
+                assert comp[pos:pos + L] == a[j:j + 1].tobytes()[:L], (name, j)
+        # libzstd reproduces the raw stream from the canonical frame
+        buf = C.create_string_buffer(node["decompressed_size"])
+        r = zstd.ZSTD_decompress(buf, node["decompressed_size"],
+                                 bytes(comp), len(comp))
+        assert not zstd.ZSTD_isError(r) and r == node["decompressed_size"], name
+        assert buf.raw == a[:rows0].tobytes(), name
+        # and the oracle reads the same values end to end
+        with oracle.OracleTable(path) as t:
+            v, e = read_all(t, 0, n, np.int64, 10000, stripe_rows=150000)
+            np.testing.assert_array_equal(v, a)

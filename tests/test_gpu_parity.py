@@ -877,3 +877,37 @@ def test_device_write_edge_shapes(tmp_path):
     with oracle.OracleTable(path) as ot:
         parts, _ = ot.scan_agg([], [(ca.AGG_SUM_I64, 0)])
         assert parts[0].i128 == int(t.sum().item())
+
+
+def test_zstd_canonical_vs_generic_parity(tmp_path):
+    """Canonical restricted-zstd frames (closed-form GPU reads, no FSE walk)
+    must aggregate identically to the same data written with canonical=0
+    (generic ZR frames, zr_decode kernel) and to the oracle."""
+    n = 60000
+    qty = ((RNG.integers(1, 51, n)) * 100).astype(np.int64)        # P(2)
+    price = RNG.integers(90000, 200000, n).astype(np.int64)        # P(3)
+    disc = RNG.integers(0, 11, n).astype(np.int64)                 # P(1)
+    const = np.full(n, 7, dtype=np.int64)                          # CONST
+    defs = [("q", ca.I64, 0), ("p", ca.I64, 0), ("d", ca.I64, 0),
+            ("c", ca.I64, 0)]
+    cols = [qty, price, disc, const]
+    preds = [(0, ca.PRED_LT, 2400), (2, ca.PRED_GE, 1), (2, ca.PRED_LE, 6)]
+    aggs = [(ca.AGG_SUM_PROD_I64, 1, 2), (ca.AGG_COUNT_STAR, -1),
+            (ca.AGG_MIN_I64, 1), (ca.AGG_MAX_I64, 3)]
+    results = []
+    for canon in (1, 0):
+        path = str(tmp_path / f"zc{canon}.cs")
+        ca.write_table(path, defs, cols, compression=ca.COMP_ZSTD,
+                       canonical=canon)
+        foot = futil.read_footer(path)
+        modes = {s["mode"] for nd_col in foot["stripes"][0]["nodes"]
+                 for s in nd_col[0]["segs"]}
+        if canon:
+            assert any(m >= futil.SEGMODE_ZRP_BASE for m in modes), modes
+        else:
+            assert all(m == futil.SEGMODE_ZR for m in modes), modes
+        op, ofilt, gp, gfilt = both(path, preds, aggs)
+        assert ofilt == gfilt
+        assert_parity(op, gp, aggs)
+        results.append([(p.i128, p.count, p.is_null) for p in gp])
+    assert results[0] == results[1]
