@@ -536,6 +536,24 @@ __device__ inline bool col_value(const uint8_t *__restrict__ data,
          * (L+3) bytes apart, so a wave's loads coalesce into a few lines. */
         if (cl.mode == CSF_SEGMODE_LIT) {
             base += (uint64_t)cl.hval;           /* skip literal-run header */
+        } else if ((cl.mode & 0xF8u) == CSF_SEGMODE_ZR4B_BASE) {
+            /* width-4 single-varying-byte / const slots (char(1) flags) */
+            uint32_t slot = (uint32_t)cl.hval;
+            if (cl.mode != CSF_SEGMODE_ZR4_CONST) {
+                const uint32_t kk = cl.L;
+                const uint32_t pos = 15u + (idx == 0 ? kk : idx + 3u + kk);
+                slot |= (uint32_t)base[pos] << (8u * kk);
+            }
+            if (cl.type == CSTRIPE_F32) {
+                float ff;
+                __builtin_memcpy(&ff, &slot, 4);
+                fv = ff;
+                iv = 0;
+            } else {
+                iv = cl.type == CSTRIPE_TEXT ? (int64_t)slot : (int64_t)(int32_t)slot;
+                fv = cl.type == CSTRIPE_TEXT ? 0.0 : (double)iv;
+            }
+            return true;
         } else {
             uint64_t raw;
             if (cl.mode == CSF_SEGMODE_CONST || cl.mode == CSF_SEGMODE_ZR_CONST) {
@@ -987,6 +1005,48 @@ __device__ inline void col_multi(const uint8_t *__restrict__ data,
 {
     const uint8_t *base = (cl.flags & 1) ? scratch + cl.val_off : data + cl.val_off;
     if ((cl.flags & 4) && cl.mode != CSF_SEGMODE_LIT) {
+        if ((cl.mode & 0xF8u) == CSF_SEGMODE_ZR4B_BASE) {
+            /* width-4 single-varying-byte / const slots (char(1) flags):
+             * ONE u64 window load covers 8 rows' varying bytes (stride 1) */
+            const uint32_t kk = cl.L;
+            uint32_t slots[R];
+            if (cl.mode == CSF_SEGMODE_ZR4_CONST) {
+                #pragma unroll
+                for (int k = 0; k < R; k++) slots[k] = (uint32_t)cl.hval;
+            } else if (row == 0) {
+                #pragma unroll
+                for (int k = 0; k < R; k++) {
+                    const uint32_t j = row + (uint32_t)k;
+                    const uint32_t pos = 15u + (j == 0 ? kk : j + 3u + kk);
+                    slots[k] = (uint32_t)cl.hval |
+                               ((uint32_t)base[pos] << (8u * kk));
+                }
+            } else {
+                uint64_t w[(R + 7) / 8];
+                #pragma unroll
+                for (int i = 0; i < (R + 7) / 8; i++)
+                    __builtin_memcpy(&w[i], base + 15u + row + 3u + kk + 8u * i, 8);
+                #pragma unroll
+                for (int k = 0; k < R; k++)
+                    slots[k] = (uint32_t)cl.hval |
+                               ((uint32_t)((w[k >> 3] >> (8 * (k & 7))) & 0xFFu)
+                                << (8u * kk));
+            }
+            #pragma unroll
+            for (int k = 0; k < R; k++) {
+                if (cl.type == CSTRIPE_F32) {
+                    float ff;
+                    __builtin_memcpy(&ff, &slots[k], 4);
+                    const double d = ff;
+                    __builtin_memcpy(&q.v[k], &d, 8);
+                } else if (cl.type == CSTRIPE_TEXT) {
+                    q.v[k] = (int64_t)slots[k];
+                } else {
+                    q.v[k] = (int64_t)(int32_t)slots[k];
+                }
+            }
+            return;
+        }
         uint64_t r[R];
         if (cl.mode == CSF_SEGMODE_CONST || cl.mode == CSF_SEGMODE_ZR_CONST) {
             #pragma unroll
@@ -2399,6 +2459,15 @@ __global__ void canon_decode_kernel(const uint8_t *__restrict__ src,
         for (uint32_t j = 0; j < width; j++) d[j] = s[j];
         return;
     }
+    if ((mode & 0xF8u) == CSF_SEGMODE_ZR4B_BASE) {      /* width-4 slots */
+        uint32_t slot = (uint32_t)hval;
+        if (mode != CSF_SEGMODE_ZR4_CONST) {
+            const uint32_t pos = 15u + (i == 0 ? L : i + 3u + L);
+            slot |= (uint32_t)src[pos] << (8u * L);
+        }
+        *(uint32_t *)(dst + (uint64_t)i * 4) = slot;
+        return;
+    }
     uint64_t v;
     if (mode == CSF_SEGMODE_CONST || mode == CSF_SEGMODE_ZR_CONST) {
         v = hval;
@@ -2775,7 +2844,8 @@ static inline bool cs_node_canon(const cs_skipnode &nd)
         return m != CSF_SEGMODE_GENERIC;
     if (nd.n.comp_type == CSTRIPE_COMP_ZSTD)
         return m == CSF_SEGMODE_ZR_CONST ||
-               (m > CSF_SEGMODE_ZRP_BASE && m <= CSF_SEGMODE_ZRP_BASE + 4);
+               (m > CSF_SEGMODE_ZRP_BASE && m <= CSF_SEGMODE_ZRP_BASE + 4) ||
+               (m >= CSF_SEGMODE_ZR4B_BASE && m <= CSF_SEGMODE_ZR4_CONST);
     return false;
 }
 
@@ -3041,6 +3111,15 @@ int csgpu_stage(cstripe_scan *s, int device_id)
                 const uint8_t *strm = stripe_base + nd.n.value_off;
                 if (cl.mode == CSF_SEGMODE_LIT) {
                     cl.hval = (int64_t)csf_canon_lit_hdr(nd.segs[0].decomp_len);
+                } else if ((cl.mode & 0xF8u) == CSF_SEGMODE_ZR4B_BASE) {
+                    uint32_t s0;
+                    memcpy(&s0, strm + 15, 4);       /* slot0 at lit offset 0 */
+                    if (cl.mode == CSF_SEGMODE_ZR4_CONST) {
+                        cl.hval = (int64_t)s0;
+                    } else {
+                        cl.L = (uint8_t)(cl.mode & 0x3);
+                        cl.hval = (int64_t)(s0 & ~(0xFFu << (8u * cl.L)));
+                    }
                 } else {
                     /* v0 full: first literal byte — offset 1 in a canonical
                      * LZ4 block (token first), 15 in a canonical zstd frame

@@ -415,6 +415,59 @@ static bool compress_chunk(const cstripe_writer *w, const PendingChunk &pc, Chun
             out.node.value_len = (uint64_t)csz;
             return true;
         }
+    }
+
+    /* width-4 canonical zstd: slots that differ in at most ONE byte (the
+     * char(1) short-varlena case: [0x05 ch 00 00] varies only in byte 1) —
+     * expressible in zstd (min match 3) but not LZ4 (min match 4); the GPU
+     * reads one byte per row closed-form (csf_canon_zr4b_pos) */
+    if (codec == CSTRIPE_COMP_ZSTD && w->opts.canonical && width == 4 &&
+        raw.size() >= 12 && raw.size() < (127u << 10)) {
+        const size_t nv = raw.size() / 4;
+        uint32_t v0, x = 0;
+        memcpy(&v0, raw.data(), 4);
+        const uint8_t *rp = raw.data();
+        for (size_t i = 1; i < nv; i++) {
+            uint32_t vi;
+            memcpy(&vi, rp + i * 4, 4);
+            x |= vi ^ v0;
+        }
+        static const zr_ctables zct4 = [] {
+            zr_ctables t;
+            zr_build_ctables(&t);
+            return t;
+        }();
+        int mode = -1, csz = 0, k = -1;
+        if (x == 0) {
+            k = 4;                           /* const form */
+        } else {
+            for (int b = 0; b < 4; b++)
+                if ((x & ~(0xFFu << (8 * b))) == 0) { k = b; break; }
+        }
+        if (k >= 0) {
+            size_t cap = raw.size() + 64;
+            out.value_comp.resize(cap);
+            if (k == 4) {
+                csz = zr_canon_const4(rp, (int)nv, out.value_comp.data(), (int)cap, &zct4);
+                if (csz > 0) mode = CSF_SEGMODE_ZR4_CONST;
+            } else {
+                csz = zr_canon_b4(rp, (int)nv, k, out.value_comp.data(), (int)cap, &zct4);
+                if (csz > 0) mode = (int)(CSF_SEGMODE_ZR4B_BASE | (uint32_t)k);
+            }
+        }
+        if (mode >= 0 && (size_t)csz >= raw.size())
+            mode = -1;
+        if (mode >= 0) {
+            out.value_comp.resize((size_t)csz);
+            csf_seg s{0, (uint32_t)csz, 0,
+                      (uint32_t)raw.size() | ((uint32_t)mode << 24)};
+            out.segs.push_back(s);
+            out.node.comp_type = codec;
+            out.node.comp_level = 0;
+            out.node.n_segs = 1;
+            out.node.value_len = (uint64_t)csz;
+            return true;
+        }
         out.value_comp.clear();
     }
 

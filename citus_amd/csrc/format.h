@@ -135,6 +135,14 @@ typedef struct csf_seg {
                                     * access, still ZSTD_decompress-decodable */
 #define CSF_SEGMODE_ZR_CONST 0x70u /* canonical restricted-zstd constant
                                     * (zr_canon_const): v0 at offset 15 */
+#define CSF_SEGMODE_ZR4B_BASE 0x80u /* 0x80|k: width-4 slots differing only
+                                     * in byte k (char(1) flag columns):
+                                     * varying byte of row j at stream offset
+                                     * 15 + (j==0 ? k : j+3+k); slot =
+                                     * (slot0 & ~(0xFF<<8k)) | (b<<8k)
+                                     * (zr_canon_b4) */
+#define CSF_SEGMODE_ZR4_CONST 0x84u /* width-4 all-equal (zr_canon_const4):
+                                     * slot0 at offset 15 */
 #define CSF_SEG_DLEN_MASK   0x00FFFFFFu
 
 static inline uint32_t csf_seg_dlen(const csf_seg *s) { return s->decomp_len & CSF_SEG_DLEN_MASK; }
@@ -155,6 +163,13 @@ static inline uint32_t csf_canon_p_pos(uint32_t j, uint32_t L)
 static inline uint32_t csf_canon_zrp_pos(uint32_t j, uint32_t L)
 {
     return j == 0 ? 15u : 23u + (j - 1u) * L;
+}
+
+/* stream byte offset of row j's varying byte inside a canonical width-4
+ * single-varying-byte frame (zr_canon_b4 layout) */
+static inline uint32_t csf_canon_zr4b_pos(uint32_t j, uint32_t k)
+{
+    return 15u + (j == 0 ? k : j + 3u + k);
 }
 
 /* literal-run header size for a mode-LIT segment of len decompressed bytes */

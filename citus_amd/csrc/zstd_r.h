@@ -783,4 +783,92 @@ static inline int zr_canon_const(const uint8_t *src, int n,
     return zr_canon_close(dst, op + bs);
 }
 
+/* ---------------- width-4 canonical frames (char(1) flag columns) -------
+ * A width-4 slot stream (short-varlena char(1): [0x05 ch 00 00]) where ALL
+ * slots share three of their four bytes and differ only in byte k. zstd's
+ * 3-byte minimum match makes a closed-form parse possible that LZ4 (min
+ * match 4) cannot express: per row, ONE literal byte (the varying byte)
+ * and one ml=3 off=4 match reproducing the 3 shared bytes that separate
+ * consecutive varying bytes in the stream.
+ *   literals = [slot0 (4 B) + pre1 (k B) + b_1] + [b_2 .. b_{n-2}] +
+ *              [b_{n-1} + post_{n-1} (3-k B)]          (lit_size = n + 6)
+ *   sequences = S0 (ll=5+k, ml=3, off=4) then n-3 x (ll=1, ml=3, off=4)
+ *               — nseq = n-2 — then 4-k trailing literals
+ * Varying byte of row j at lit position (j==0 ? k : j+3+k); the slot is
+ * hval | (b_j << 8k) with hval = slot0 & ~(0xFF << 8k). Also the all-equal
+ * width-4 case as one big ml=4(n-1) off=4 match (ZR4 CONST). */
+
+static inline int zr_canon_b4(const uint8_t *src, int n, int k,
+                              uint8_t *dst, int dcap, const zr_ctables *ct)
+{
+    if (n < 3 || k < 0 || k > 3 || (int64_t)n * 4 >= (127 << 10)) return 0;
+    const uint32_t content = (uint32_t)n * 4;
+    const uint32_t lit_size = (uint32_t)n + 6;
+    int op = zr_canon_hdr(dst, dcap, content, lit_size);
+    if (op < 0 || op + (int)lit_size + 4 > dcap) return 0;
+    memcpy(dst + op, src, 4 + (size_t)k);    /* slot0 + pre of slot1 */
+    op += 4 + k;
+    dst[op++] = src[4 + (size_t)k];          /* b_1 */
+    for (int j = 2; j <= n - 2; j++)
+        dst[op++] = src[(size_t)j * 4 + k];
+    dst[op++] = src[(size_t)(n - 1) * 4 + k];
+    memcpy(dst + op, src + (size_t)(n - 1) * 4 + k + 1, (size_t)(3 - k));
+    op += 3 - k;
+    const int nseq = n - 2;
+    if (nseq < 128) dst[op++] = (uint8_t)nseq;
+    else { dst[op++] = (uint8_t)((nseq >> 8) + 128); dst[op++] = (uint8_t)nseq; }
+    dst[op++] = 0x00;
+    const uint8_t llc_first = (uint8_t)(5 + k), llc_rest = 1;
+    const uint8_t mlc = 0;                   /* ml 3 */
+    const uint8_t ofc = 2;                   /* off 4 -> value 7: 2 extra bits */
+    zr_bw bw;
+    zr_bw_init(&bw, dst + op, dcap - op);
+    zr_cstate sml, sof, sll;
+    zr_cstate_init(&sml, &ct->ml, mlc);
+    zr_cstate_init(&sof, &ct->of, ofc);
+    zr_cstate_init(&sll, &ct->ll, nseq == 1 ? llc_first : llc_rest);
+    if (zr_bw_add(&bw, 3, 2) < 0) return 0;  /* last seq OF extra (7-4) */
+    for (int i = nseq - 2; i >= 0; i--) {
+        if (zr_cstate_encode(&sof, &ct->of, ofc, &bw) < 0) return 0;
+        if (zr_cstate_encode(&sml, &ct->ml, mlc, &bw) < 0) return 0;
+        if (zr_cstate_encode(&sll, &ct->ll, i == 0 ? llc_first : llc_rest, &bw) < 0) return 0;
+        if (zr_bw_add(&bw, 3, 2) < 0) return 0;
+    }
+    if (zr_cstate_flush(&sml, &ct->ml, &bw) < 0) return 0;
+    if (zr_cstate_flush(&sof, &ct->of, &bw) < 0) return 0;
+    if (zr_cstate_flush(&sll, &ct->ll, &bw) < 0) return 0;
+    const int bs = zr_bw_close(&bw);
+    if (bs < 0) return 0;
+    return zr_canon_close(dst, op + bs);
+}
+
+static inline int zr_canon_const4(const uint8_t *src, int n,
+                                  uint8_t *dst, int dcap, const zr_ctables *ct)
+{
+    if (n < 3 || (int64_t)n * 4 >= (127 << 10)) return 0;
+    const uint32_t content = (uint32_t)n * 4;
+    int op = zr_canon_hdr(dst, dcap, content, 4);
+    if (op < 0 || op + 40 > dcap) return 0;
+    memcpy(dst + op, src, 4);
+    op += 4;
+    dst[op++] = 1;
+    dst[op++] = 0x00;
+    const uint32_t ml = (uint32_t)(n - 1) * 4;
+    const uint8_t llc = 4, mlc = zr_ml_code(ml), ofc = 2;
+    zr_bw bw;
+    zr_bw_init(&bw, dst + op, dcap - op);
+    zr_cstate sml, sof, sll;
+    zr_cstate_init(&sml, &ct->ml, mlc);
+    zr_cstate_init(&sof, &ct->of, ofc);
+    zr_cstate_init(&sll, &ct->ll, llc);
+    if (zr_bw_add(&bw, ml - ZR_ML_BASE[mlc], ZR_ML_BITS[mlc]) < 0) return 0;
+    if (zr_bw_add(&bw, 3, 2) < 0) return 0;
+    if (zr_cstate_flush(&sml, &ct->ml, &bw) < 0) return 0;
+    if (zr_cstate_flush(&sof, &ct->of, &bw) < 0) return 0;
+    if (zr_cstate_flush(&sll, &ct->ll, &bw) < 0) return 0;
+    const int bs = zr_bw_close(&bw);
+    if (bs < 0) return 0;
+    return zr_canon_close(dst, op + bs);
+}
+
 #endif /* CSTRIPE_ZSTD_R_H */

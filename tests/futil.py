@@ -16,11 +16,14 @@ SEGMODE_LIT = 0x30
 SEGMODE_ZR = 0x50
 SEGMODE_ZRP_BASE = 0x60          # 0x60|L: canonical restricted-zstd P(L)
 SEGMODE_ZR_CONST = 0x70
+SEGMODE_ZR4B_BASE = 0x80         # 0x80|k: width-4 single-varying-byte slots
+SEGMODE_ZR4_CONST = 0x84
 
 
 def is_zr_mode(m):
     """any restricted-zstd stream shape (generic frames or canonical)"""
-    return m == SEGMODE_ZR or m == SEGMODE_ZR_CONST or 0x61 <= m <= 0x64
+    return (m == SEGMODE_ZR or m == SEGMODE_ZR_CONST or 0x61 <= m <= 0x64 or
+            0x80 <= m <= 0x84)
 
 
 def read_footer(path):

@@ -632,3 +632,54 @@ def test_zstd_canonical_layout(tmp_path):
         with oracle.OracleTable(path) as t:
             v, e = read_all(t, 0, n, np.int64, 10000, stripe_rows=150000)
             np.testing.assert_array_equal(v, a)
+
+
+def test_zstd_canonical_width4_flags(tmp_path):
+    """Width-4 canonical zstd (zr_canon_b4): char(1) short-varlena slots
+    [0x05 ch 00 00] vary only in byte 1 — per row ONE literal byte at the
+    closed-form position csf_canon_zr4b_pos documents (15 + (j==0 ? k :
+    j+3+k)), mode 0x80|k; all-equal slots -> 0x84. Both still plain zstd
+    to ZSTD_decompress. (LZ4's 4-byte minimum match cannot express this
+    parse — zstd's is 3.)"""
+    n = 25000
+    zstd = C.CDLL("libzstd.so.1")
+    zstd.ZSTD_decompress.restype = C.c_size_t
+    zstd.ZSTD_decompress.argtypes = [C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
+    zstd.ZSTD_isError.restype = C.c_uint
+    zstd.ZSTD_isError.argtypes = [C.c_size_t]
+    flags = np.array([ord("ANR"[i % 3]) for i in range(n)], dtype=np.uint32)
+    slots_var = (0x05 | (flags << 8)).astype(np.uint32)        # byte 1 varies
+    slots_const = np.full(n, 0x05 | (ord("X") << 8), dtype=np.uint32)
+    lowbyte = np.arange(n, dtype=np.uint32) % 200              # byte 0 varies
+    for name, slots, want_mode, k in [("flags", slots_var, 0x81, 1),
+                                      ("const", slots_const, 0x84, None),
+                                      ("i32low", lowbyte, 0x80, 0)]:
+        typ = ca.TEXT if name != "i32low" else ca.I32
+        col = slots.view(np.int32)
+        path = str(tmp_path / f"z4_{name}.cs")
+        ca.write_table(path, [("f", typ, 0)], [np.ascontiguousarray(col)],
+                       compression=ca.COMP_ZSTD)
+        foot = futil.read_footer(path)
+        node = foot["stripes"][0]["nodes"][0][0]
+        assert node["comp_type"] == ca.COMP_ZSTD, name
+        assert node["n_segs"] == 1 and node["segs"][0]["mode"] == want_mode, \
+            (name, hex(node["segs"][0]["mode"]))
+        comp = futil.chunk_stream(path, node)
+        rows0 = node["decompressed_size"] // 4
+        raw0 = slots[:rows0].tobytes()
+        # closed-form position pin
+        assert comp[15:19] == raw0[:4], name
+        if k is not None:
+            for j in (0, 1, 2, 777, rows0 - 1):
+                pos = 15 + (k if j == 0 else j + 3 + k)
+                assert comp[pos] == (slots[j] >> (8 * k)) & 0xFF, (name, j)
+        # system-zstd round trip
+        buf = C.create_string_buffer(node["decompressed_size"])
+        r = zstd.ZSTD_decompress(buf, node["decompressed_size"],
+                                 bytes(comp), len(comp))
+        assert not zstd.ZSTD_isError(r) and r == node["decompressed_size"], name
+        assert buf.raw == raw0, name
+        # oracle end-to-end
+        with oracle.OracleTable(path) as t:
+            v, e = read_all(t, 0, n, np.int32, 10000, stripe_rows=150000)
+            np.testing.assert_array_equal(v.view(np.uint32), slots)

@@ -911,3 +911,41 @@ def test_zstd_canonical_vs_generic_parity(tmp_path):
         assert_parity(op, gp, aggs)
         results.append([(p.i128, p.count, p.is_null) for p in gp])
     assert results[0] == results[1]
+
+
+def test_grouped_zstd_width4_canonical_flags(tmp_path):
+    """Q1-shaped GROUP BY over char(1) short-varlena flag columns stored as
+    width-4 canonical zstd frames (zr_canon_b4: one closed-form byte per
+    row — no decode kernel): grouped partials must match the oracle and the
+    same data written as lz4 (lane-decoded) bit-for-bit."""
+    n = 50000
+    rf = ca.text_slots(["A", "N", "R"][i % 3] for i in range(n))
+    lst = ca.text_slots(["O", "F"][i % 2] for i in range(n))
+    qty = ((RNG.integers(1, 51, n)) * 100).astype(np.int64)
+    defs = [("rf", ca.TEXT, 0), ("ls", ca.TEXT, 0), ("q", ca.I64, 0)]
+    cols = [rf.view(np.int32), lst.view(np.int32), qty]
+    aggs = [(ca.AGG_SUM_I64, 2), (ca.AGG_COUNT_STAR, -1),
+            (ca.AGG_MIN_I64, 2), (ca.AGG_MAX_I64, 2)]
+    outs = {}
+    for comp, name in [(ca.COMP_ZSTD, "zstd"), (ca.COMP_LZ4, "lz4")]:
+        path = str(tmp_path / f"g4_{name}.cs")
+        ca.write_table(path, defs, cols, compression=comp)
+        if comp == ca.COMP_ZSTD:
+            foot = futil.read_footer(path)
+            m0 = foot["stripes"][0]["nodes"][0][0]["segs"][0]["mode"]
+            assert m0 == futil.SEGMODE_ZR4B_BASE | 1, hex(m0)
+        with oracle.OracleTable(path) as t:
+            og, _ = t.scan_agg([], aggs, group_cols=(0, 1))
+        with ca.Reader(path) as r, \
+             r.scan(cols_mask=0b111, preds=[]) as s:
+            s.stage()
+            gg = s.agg_grouped(aggs, (0, 1))
+        okeys = sorted(og.keys())
+        gkeys = sorted(gg.keys())
+        assert okeys == gkeys and len(okeys) == 6, name
+        for k in okeys:
+            for i in range(len(aggs)):
+                assert og[k][i].i128 == gg[k][i].i128, (name, k, i)
+                assert og[k][i].count == gg[k][i].count, (name, k, i)
+        outs[name] = {k: [(p.i128, p.count) for p in v] for k, v in gg.items()}
+    assert outs["zstd"] == outs["lz4"]
