@@ -683,3 +683,59 @@ def test_zstd_canonical_width4_flags(tmp_path):
         with oracle.OracleTable(path) as t:
             v, e = read_all(t, 0, n, np.int32, 10000, stripe_rows=150000)
             np.testing.assert_array_equal(v.view(np.uint32), slots)
+
+
+def test_zstd_canonical_boundary_sizes(tmp_path):
+    """Canonical-zstd emitters across size boundaries (nseq 1-byte/2-byte
+    forms, minimum n, chunk tails): every emitted frame must reproduce the
+    raw stream through the system libzstd, and the oracle must read the
+    values back exactly."""
+    zstd = C.CDLL("libzstd.so.1")
+    zstd.ZSTD_decompress.restype = C.c_size_t
+    zstd.ZSTD_decompress.argtypes = [C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
+    zstd.ZSTD_isError.restype = C.c_uint
+    zstd.ZSTD_isError.argtypes = [C.c_size_t]
+    rng = np.random.default_rng(11)
+    sizes = [3, 4, 5, 127, 128, 129, 130, 131, 999, 10000, 10003]
+    for n in sizes:
+        for case in ("p1", "p4", "c8", "b4", "c4"):
+            if case in ("p1", "p4", "c8"):
+                if case == "p1":
+                    a = rng.integers(0, 200, n).astype(np.int64)
+                elif case == "p4":
+                    a = rng.integers(0, 2**31, n).astype(np.int64)
+                else:
+                    a = np.full(n, -12345, dtype=np.int64)
+                defs, cols, dt, w = [("a", ca.I64, 0)], [a], np.int64, 8
+            else:
+                if case == "b4":
+                    ch = rng.integers(65, 68, n).astype(np.uint32)
+                    a = (0x05 | (ch << 8)).view(np.int32)
+                else:
+                    a = np.full(n, 0x05 | (70 << 8), dtype=np.uint32).view(np.int32)
+                defs, cols, dt, w = [("a", ca.TEXT, 0)], [a], np.int32, 4
+            path = str(tmp_path / f"b_{case}_{n}.cs")
+            ca.write_table(path, defs, cols, compression=ca.COMP_ZSTD,
+                           chunk_group_row_limit=10000)
+            foot = futil.read_footer(path)
+            raw = np.asarray(cols[0])
+            pos = 0
+            for st in foot["stripes"]:
+                for ci, nd in enumerate(st["nodes"][0]):
+                    rows = nd["decompressed_size"] // w
+                    if nd["comp_type"] == ca.COMP_ZSTD:
+                        comp = futil.chunk_stream(path, nd)
+                        out = bytearray(nd["decompressed_size"])
+                        for sg in nd["segs"]:
+                            fr = comp[sg["comp_off"]:sg["comp_off"] + sg["comp_len"]]
+                            buf = C.create_string_buffer(sg["decomp_len"])
+                            r = zstd.ZSTD_decompress(buf, sg["decomp_len"],
+                                                     bytes(fr), len(fr))
+                            assert not zstd.ZSTD_isError(r) and r == sg["decomp_len"], \
+                                (case, n, hex(sg["mode"]))
+                            out[sg["decomp_off"]:sg["decomp_off"] + sg["decomp_len"]] = buf.raw
+                        assert bytes(out) == raw[pos:pos + rows].tobytes(), (case, n)
+                    pos += rows
+            with oracle.OracleTable(path) as t:
+                v, e = read_all(t, 0, n, dt, 10000, stripe_rows=150000)
+                np.testing.assert_array_equal(v, raw[:n].astype(dt) if dt != np.int32 else raw[:n])

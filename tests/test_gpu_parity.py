@@ -949,3 +949,42 @@ def test_grouped_zstd_width4_canonical_flags(tmp_path):
                 assert og[k][i].count == gg[k][i].count, (name, k, i)
         outs[name] = {k: [(p.i128, p.count) for p in v] for k, v in gg.items()}
     assert outs["zstd"] == outs["lz4"]
+
+
+def test_batch_and_read_row_zstd_canonical(tmp_path):
+    """canon_decode_kernel over CANONICAL zstd chunks (width-8 P and the
+    width-4 single-varying-byte family): next_batch row-aligned output and
+    random-access read_row must equal the oracle's view."""
+    n = 26000
+    q = ((RNG.integers(1, 51, n)) * 100).astype(np.int64)          # zstd P(2)
+    fl = ca.text_slots(["A", "N", "R"][i % 3] for i in range(n))   # zr4b k=1
+    path = str(tmp_path / "zb.cs")
+    ca.write_table(path, [("q", ca.I64, 0), ("f", ca.TEXT, 0)],
+                   [q, fl.view(np.int32)], compression=ca.COMP_ZSTD)
+    foot = futil.read_footer(path)
+    assert foot["stripes"][0]["nodes"][0][0]["segs"][0]["mode"] == 0x62
+    assert foot["stripes"][0]["nodes"][1][0]["segs"][0]["mode"] == 0x81
+    with ca.Reader(path) as r, r.scan(cols_mask=0b11) as s:
+        s.stage()
+        vq = np.zeros(10000, dtype=np.int64)
+        vf = np.zeros(10000, dtype=np.int32)
+        eq_ = np.zeros(10000, dtype=np.uint8)
+        ef = np.zeros(10000, dtype=np.uint8)
+        seen = 0
+        while True:
+            res = s.next_batch({0: vq, 1: vf}, {0: eq_, 1: ef})
+            if res is None:
+                break
+            nr, first = res
+            np.testing.assert_array_equal(vq[:nr], q[seen:seen + nr])
+            np.testing.assert_array_equal(vf[:nr].view(np.uint32),
+                                          fl[seen:seen + nr])
+            assert eq_[:nr].all() and ef[:nr].all()
+            seen += nr
+        assert seen == n
+    with ca.Reader(path) as r, r.scan(cols_mask=0b11) as s:
+        s.stage()
+        for rn in (0, 1, 9999, 10000, 10001, 12345, n - 1):
+            vals = s.read_row(rn)
+            assert vals[0] == q[rn], rn
+            assert (vals[1] & 0xFFFFFFFF) == int(fl[rn]), rn
