@@ -979,7 +979,7 @@ def test_batch_and_read_row_zstd_canonical(tmp_path):
             np.testing.assert_array_equal(vq[:nr], q[seen:seen + nr])
             np.testing.assert_array_equal(vf[:nr].view(np.uint32),
                                           fl[seen:seen + nr])
-            assert eq_[:nr].all() and ef[:nr].all()
+            assert not eq_[:nr].any() and not ef[:nr].any()   # 1 = NULL
             seen += nr
         assert seen == n
     with ca.Reader(path) as r, r.scan(cols_mask=0b11) as s:
