@@ -1282,6 +1282,13 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_agg_kernel(
             if (pr.gend) {
                 #pragma unroll
                 for (int k = 0; k < R; k++) { pv[k] = pv[k] & gv[k]; gv[k] = false; }
+                /* early wave exit: if every lane's window is dead after
+                 * this AND-group, skip the remaining pred columns' loads
+                 * and evals (wave-uniform ballot branch) */
+                bool alive = false;
+                #pragma unroll
+                for (int k = 0; k < R; k++) alive |= pv[k];
+                if (__ballot(alive) == 0) break;
             }
         }
         bool any = false;
@@ -1795,6 +1802,11 @@ __global__ __launch_bounds__(AGG_BLOCK, MINW) void multi_grouped_kernel(
             if (pr.gend) {
                 #pragma unroll
                 for (int k = 0; k < R; k++) { pv[k] = pv[k] & gv[k]; gv[k] = false; }
+                /* early wave exit after a dead AND-group (see multi_agg) */
+                bool alive = false;
+                #pragma unroll
+                for (int k = 0; k < R; k++) alive |= pv[k];
+                if (__ballot(alive) == 0) break;
             }
         }
         bool any = false;
