@@ -2841,6 +2841,147 @@ __global__ __launch_bounds__(ZR_LDS_BLOCK) void zr_decode_lds2_kernel(
 }
 
 /* =====================================================================
+ * device chunk-group pruning (SelectedChunkMask on GPU — SURVEY §8f3)
+ * One thread per chunk group evaluates the same CNF refutation as the
+ * host loop in cstripe_scan_begin (reference: SelectedChunkMask,
+ * columnar_reader.c:1132-1187; OR recursion columnar_customscan.c:770-829),
+ * including PG float ordering and the all-NULL-chunk rule. The host only
+ * GATHERS the skip entries (a copy pass, no comparisons); every compare
+ * runs on device.
+ * ===================================================================== */
+
+struct PruneMM { int64_t mn, mx; uint64_t has; };
+
+struct PrunePred {
+    int64_t ival;
+    double fval;
+    uint8_t op, is_float, slot, gend;
+};
+
+struct PruneParams {
+    uint32_t n_preds;
+    PrunePred preds[CSTRIPE_MAX_PREDS];
+};
+
+__global__ void chunk_prune_kernel(const PruneMM *__restrict__ mm,
+                                   uint32_t n_chunks,
+                                   uint8_t *__restrict__ selected,
+                                   const PruneParams pp)
+{
+    const uint32_t c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= n_chunks) return;
+    bool sel = true;
+    bool group_refuted = true;
+    for (uint32_t i = 0; i < pp.n_preds; i++) {
+        const PrunePred &p = pp.preds[i];
+        const PruneMM e = mm[(size_t)p.slot * n_chunks + c];
+        bool ref = false;
+        if (e.has) {          /* all-NULL chunks are never refuted */
+            if (p.is_float) {
+                double mn, mx;
+                __builtin_memcpy(&mn, &e.mn, 8);
+                __builtin_memcpy(&mx, &e.mx, 8);
+                const double cv = p.fval;
+                switch (p.op) {     /* PG float ordering (NaN high) */
+                    case CSTRIPE_PRED_LT: ref = f64cmp_pg(mn, cv) >= 0; break;
+                    case CSTRIPE_PRED_LE: ref = f64cmp_pg(mn, cv) > 0; break;
+                    case CSTRIPE_PRED_GT: ref = f64cmp_pg(mx, cv) <= 0; break;
+                    case CSTRIPE_PRED_GE: ref = f64cmp_pg(mx, cv) < 0; break;
+                    case CSTRIPE_PRED_EQ: ref = f64cmp_pg(cv, mn) < 0 ||
+                                                f64cmp_pg(cv, mx) > 0; break;
+                    default:              ref = f64cmp_pg(mn, cv) == 0 &&
+                                                f64cmp_pg(mx, cv) == 0; break;
+                }
+            } else {
+                const int64_t cv = p.ival;
+                switch (p.op) {
+                    case CSTRIPE_PRED_LT: ref = e.mn >= cv; break;
+                    case CSTRIPE_PRED_LE: ref = e.mn > cv; break;
+                    case CSTRIPE_PRED_GT: ref = e.mx <= cv; break;
+                    case CSTRIPE_PRED_GE: ref = e.mx < cv; break;
+                    case CSTRIPE_PRED_EQ: ref = cv < e.mn || cv > e.mx; break;
+                    default:              ref = e.mn == cv && e.mx == cv; break;
+                }
+            }
+        }
+        group_refuted &= ref;
+        if (p.gend) {
+            if (group_refuted) { sel = false; break; }
+            group_refuted = true;
+        }
+    }
+    selected[c] = sel ? 1 : 0;
+}
+
+int csgpu_prune(cstripe_reader *r, const std::vector<cstripe_pred> &preds,
+                std::vector<uint8_t> &selected)
+{
+    if (!cstripe_gpu_available()) return CSTRIPE_ERR_NOGPU;
+    if (preds.empty() || preds.size() > CSTRIPE_MAX_PREDS) return CSTRIPE_ERR_ARG;
+
+    /* distinct predicate columns -> slots */
+    PruneParams pp{};
+    pp.n_preds = (uint32_t)preds.size();
+    uint32_t slot_col[CSTRIPE_MAX_PREDS];
+    uint32_t n_slots = 0;
+    for (uint32_t i = 0; i < pp.n_preds; i++) {
+        const cstripe_pred &p = preds[i];
+        uint32_t sl = n_slots;
+        for (uint32_t j = 0; j < n_slots; j++)
+            if (slot_col[j] == p.column) { sl = j; break; }
+        if (sl == n_slots) slot_col[n_slots++] = p.column;
+        const uint8_t ty = r->cols[p.column].type;
+        pp.preds[i].ival = p.ival;
+        pp.preds[i].fval = p.fval;
+        pp.preds[i].op = p.op;
+        pp.preds[i].is_float = ty == CSTRIPE_F32 || ty == CSTRIPE_F64;
+        pp.preds[i].slot = (uint8_t)sl;
+        pp.preds[i].gend = i + 1 == pp.n_preds ||
+                           preds[i + 1].or_group != preds[i].or_group;
+    }
+
+    uint64_t n_chunks = 0;
+    for (const auto &st : r->stripes) n_chunks += st.meta.chunk_count;
+    if (n_chunks == 0) { selected.clear(); return CSTRIPE_OK; }
+
+    /* gather skip entries per slot, global chunk order */
+    std::vector<PruneMM> h_mm((size_t)n_slots * n_chunks);
+    uint64_t ci = 0;
+    for (const auto &st : r->stripes) {
+        for (uint32_t k = 0; k < st.meta.chunk_count; k++, ci++) {
+            for (uint32_t sl = 0; sl < n_slots; sl++) {
+                const csf_skipnode &nd = st.nodes[slot_col[sl]][k].n;
+                PruneMM &e = h_mm[(size_t)sl * n_chunks + ci];
+                e.mn = nd.min_i;
+                e.mx = nd.max_i;
+                e.has = nd.has_min_max;
+            }
+        }
+    }
+
+    PruneMM *d_mm = nullptr;
+    uint8_t *d_sel = nullptr;
+    #define PR_TRY(x) do { hipError_t _e = (x); if (_e != hipSuccess) { \
+        cs_set_err("prune: %s", hipGetErrorString(_e)); \
+        if (d_mm) (void)hipFree(d_mm); if (d_sel) (void)hipFree(d_sel); \
+        return CSTRIPE_ERR; } } while (0)
+    PR_TRY(hipMalloc(&d_mm, h_mm.size() * sizeof(PruneMM)));
+    PR_TRY(hipMalloc(&d_sel, n_chunks));
+    PR_TRY(hipMemcpy(d_mm, h_mm.data(), h_mm.size() * sizeof(PruneMM),
+                     hipMemcpyHostToDevice));
+    const uint32_t grid = (uint32_t)((n_chunks + 255) / 256);
+    hipLaunchKernelGGL(chunk_prune_kernel, dim3(grid), dim3(256), 0, 0,
+                       d_mm, (uint32_t)n_chunks, d_sel, pp);
+    PR_TRY(hipGetLastError());
+    selected.resize(n_chunks);
+    PR_TRY(hipMemcpy(selected.data(), d_sel, n_chunks, hipMemcpyDeviceToHost));
+    #undef PR_TRY
+    (void)hipFree(d_mm);
+    (void)hipFree(d_sel);
+    return CSTRIPE_OK;
+}
+
+/* =====================================================================
  * staging
  * ===================================================================== */
 

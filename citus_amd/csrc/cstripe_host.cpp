@@ -1136,6 +1136,31 @@ extern "C" cstripe_scan *cstripe_scan_begin(cstripe_reader *r, uint64_t cols_mas
                      });
     if (s->cols_mask == 0)
         s->cols_mask = 1;   /* pure count(*): still scan one column's chunks */
+    /* device pruning (SelectedChunkMask on GPU): same CNF
+     * refutation evaluated one-thread-per-chunk; worthwhile when the
+     * footer directory holds many thousands of chunk groups. Identical
+     * semantics pinned by tests (CSTRIPE_DEVICE_PRUNE=1/0 differential). */
+    {
+        uint64_t total_chunks = 0;
+        for (const auto &st : r->stripes) total_chunks += st.meta.chunk_count;
+        const char *dp = getenv("CSTRIPE_DEVICE_PRUNE");
+        const int dpv = dp ? atoi(dp) : -1;
+        std::vector<uint8_t> selmask;
+        if (!s->preds.empty() && dpv != 0 &&
+            (dpv == 1 || total_chunks >= 8192) &&
+            csgpu_prune(r, s->preds, selmask) == CSTRIPE_OK &&
+            selmask.size() == total_chunks) {
+            uint64_t ci = 0;
+            for (uint32_t si = 0; si < r->stripes.size(); si++) {
+                const cs_stripe_info &st = r->stripes[si];
+                for (uint32_t k = 0; k < st.meta.chunk_count; k++, ci++) {
+                    if (selmask[ci]) s->sel.push_back({si, k});
+                    else s->chunk_groups_filtered++;
+                }
+            }
+            return s;
+        }
+    }
     for (uint32_t si = 0; si < r->stripes.size(); si++) {
         const cs_stripe_info &st = r->stripes[si];
         for (uint32_t k = 0; k < st.meta.chunk_count; k++) {

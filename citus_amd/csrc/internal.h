@@ -78,6 +78,13 @@ struct cstripe_scan {
 };
 
 /* implemented in cstripe_gpu.hip */
+/* device chunk-group pruning (SelectedChunkMask on GPU):
+ * evaluates the CNF min/max refutation for every chunk in one launch;
+ * selected[global chunk index] = 1 to keep. Returns CSTRIPE_OK, or
+ * CSTRIPE_ERR_NOGPU when no device is visible (caller falls back to the
+ * host loop). preds must be normalized + group-sorted (scan_begin's form). */
+int  csgpu_prune(cstripe_reader *r, const std::vector<cstripe_pred> &preds,
+                 std::vector<uint8_t> &selected);
 int  csgpu_stage(cstripe_scan *s, int device_id);
 void csgpu_release(cstripe_scan *s);
 uint64_t csgpu_staged_bytes(const cstripe_scan *s);

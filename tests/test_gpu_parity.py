@@ -988,3 +988,52 @@ def test_batch_and_read_row_zstd_canonical(tmp_path):
             vals = s.read_row(rn)
             assert vals[0] == q[rn], rn
             assert (vals[1] & 0xFFFFFFFF) == int(fl[rn]), rn
+
+
+def test_device_prune_matches_host(tmp_path):
+    """chunk_prune_kernel (SelectedChunkMask on device) must remove EXACTLY
+    the chunks the host CNF loop removes — OR groups, PG-NaN float ordering
+    and all-NULL chunks included — and yield identical aggregates."""
+    import os as _os
+    n = 120_000
+    a = RNG.integers(0, 10_000, n).astype(np.int64)
+    f = RNG.normal(size=n)
+    f[: n // 6] = np.nan                        # NaN-heavy early chunks
+    b = RNG.integers(-500, 500, n).astype(np.int64)
+    nb = np.zeros(n, dtype=np.uint8)
+    nb[n // 3: n // 2] = 1                      # a run of all-NULL chunks
+    path = str(tmp_path / "dp.cs")
+    ca.write_table(path, [("a", ca.I64, 0), ("f", ca.F64, 0), ("b", ca.I64, 0)],
+                   [np.sort(a), f, b], nulls=[None, None, nb],
+                   compression=ca.COMP_LZ4, chunk_group_row_limit=1000,
+                   stripe_row_limit=10000)
+    aggs = [(ca.AGG_COUNT_STAR, -1), (ca.AGG_SUM_I64, 0), (ca.AGG_SUM_F64, 1)]
+    cases = [
+        [(0, ca.PRED_LT, 2500)],                          # sorted col: prunes
+        [(0, ca.PRED_GE, 9000), (2, ca.PRED_GT, 0)],
+        [(1, ca.PRED_GT, 10.0)],                          # NaN > c matches
+        [(2, ca.PRED_EQ, 77)],                            # NULL-run chunks
+        # OR group: a<500 OR a>9500 (same group id), AND b<=0
+        [(0, ca.PRED_LT, 500, 7), (0, ca.PRED_GT, 9500, 7),
+         (2, ca.PRED_LE, 0)],
+    ]
+    for preds in cases:
+        res = {}
+        for mode in ("1", "0"):
+            _os.environ["CSTRIPE_DEVICE_PRUNE"] = mode
+            try:
+                with ca.Reader(path) as r, \
+                     r.scan(cols_mask=0b111, preds=preds) as s:
+                    filt = s.chunk_groups_filtered
+                    s.stage()
+                    parts = s.agg(aggs)
+                res[mode] = (filt, [(p.i128, p.count, p.is_null,
+                                     None if p.is_null else
+                                     ("nan" if p.f64 != p.f64 else round(p.f64, 9)))
+                                    for p in parts])
+            finally:
+                del _os.environ["CSTRIPE_DEVICE_PRUNE"]
+        assert res["1"][0] == res["0"][0], preds    # identical chunk removal
+        assert res["1"][1] == res["0"][1], preds
+        if preds is cases[0]:
+            assert res["1"][0] > 0                  # pruning actually fired
