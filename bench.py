@@ -241,7 +241,8 @@ def main():
     if scan.last_fused:
         dominant = "fused_agg_kernel" if not group_cols else "fused_grouped_kernel"
     elif avg_decode >= avg_agg:
-        dominant = "lz4_decode_lane_kernel"
+        dominant = ("zr_decode_lds2_kernel" if args.compression == "zstd"
+                    else "lz4_decode_lane_kernel")
     else:
         # all-dense scans (canonical writer default) run the multi-row kernels
         dominant = "multi_grouped_kernel" if group_cols else "multi_agg_kernel"
