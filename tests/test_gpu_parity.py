@@ -1065,8 +1065,8 @@ def test_zstd_canonical_with_nulls(tmp_path):
     with ca.Reader(path) as r, r.scan(cols_mask=0b11, preds=[]) as s:
         s.stage()
         gg = s.agg_grouped(aggs, (1,))
-    sk = lambda k: (k is None, k if k is not None else 0)
-    assert sorted(og.keys(), key=sk) == sorted(gg.keys(), key=sk)
+    assert sorted(og.keys(), key=ca.encode_group_key) == \
+           sorted(gg.keys(), key=ca.encode_group_key)
     assert len(og) == 4                       # A, N, R, NULL
     for k in og:
         for i in range(len(aggs)):
