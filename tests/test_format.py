@@ -739,3 +739,58 @@ def test_zstd_canonical_boundary_sizes(tmp_path):
             with oracle.OracleTable(path) as t:
                 v, e = read_all(t, 0, n, dt, 10000, stripe_rows=150000)
                 np.testing.assert_array_equal(v, raw[:n].astype(dt) if dt != np.int32 else raw[:n])
+
+
+def test_zstd_canonical_random_fuzz(tmp_path):
+    """Seeded random sweep over canonical-zstd eligibility shapes: random
+    L in 1..4 (width-8), varying-byte k in 0..3 (width-4), random sizes,
+    optional NULLs — every zstd chunk the writer emits must reproduce its
+    raw stream through libzstd, and the oracle read must be exact."""
+    zstd = C.CDLL("libzstd.so.1")
+    zstd.ZSTD_decompress.restype = C.c_size_t
+    zstd.ZSTD_decompress.argtypes = [C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t]
+    zstd.ZSTD_isError.restype = C.c_uint
+    zstd.ZSTD_isError.argtypes = [C.c_size_t]
+    rng = np.random.default_rng(99)
+    for trial in range(60):
+        n = int(rng.integers(3, 30000))
+        wide = rng.random() < 0.5
+        with_nulls = rng.random() < 0.3
+        if wide:
+            L = int(rng.integers(1, 5))
+            base = int(rng.integers(-2**40, 2**40)) & ~((1 << (8 * L)) - 1)
+            a = (base | rng.integers(0, 1 << (8 * L), n, dtype=np.int64)).astype(np.int64)
+            defs, dt, w = [("a", ca.I64, 0)], np.int64, 8
+        else:
+            k = int(rng.integers(0, 4))
+            base = int(rng.integers(0, 2**32)) & ~(0xFF << (8 * k))
+            a = (base | (rng.integers(0, 256, n, dtype=np.uint32) << (8 * k))) \
+                .astype(np.uint32).view(np.int32)
+            defs, dt, w = [("a", ca.I32, 0)], np.int32, 4
+        nulls = [(rng.random(n) < 0.2).astype(np.uint8)] if with_nulls else None
+        path = str(tmp_path / f"fz{trial}.cs")
+        ca.write_table(path, defs, [np.ascontiguousarray(a)], nulls=nulls,
+                       compression=ca.COMP_ZSTD,
+                       chunk_group_row_limit=int(rng.choice([1000, 3000, 10000])))
+        foot = futil.read_footer(path)
+        for st in foot["stripes"]:
+            for nd in st["nodes"][0]:
+                if nd["comp_type"] != ca.COMP_ZSTD:
+                    continue
+                comp = futil.chunk_stream(path, nd)
+                for sg in nd["segs"]:
+                    fr = comp[sg["comp_off"]:sg["comp_off"] + sg["comp_len"]]
+                    buf = C.create_string_buffer(sg["decomp_len"])
+                    r = zstd.ZSTD_decompress(buf, sg["decomp_len"],
+                                             bytes(fr), len(fr))
+                    assert not zstd.ZSTD_isError(r) and r == sg["decomp_len"], \
+                        (trial, hex(sg["mode"]))
+        with oracle.OracleTable(path) as t:
+            v, e = read_all(t, 0, n, dt, foot["chunk_row_limit"],
+                            stripe_rows=150000)
+            if nulls is None:
+                np.testing.assert_array_equal(v, a)
+            else:
+                keep = nulls[0] == 0
+                np.testing.assert_array_equal(v[keep], a[keep])
+                np.testing.assert_array_equal(e, 1 - nulls[0])
