@@ -210,3 +210,38 @@ def test_bench_expected_goldens():
         got = ca.expected_q1(rows, seed)
         for k, v in got.items():
             assert exp[f"{k[0]}_{k[1]}"] == v, (key, k)
+
+
+def test_in_list_pushdown_reference_fixture(tmp_path):
+    """The reference pushes IN-lists (ScalarArrayOpExpr) into chunk
+    filtering whole (columnar_customscan.c:840-850); the ABI expresses
+    `col IN (...)` as one OR group of EQ atoms. Reproduces the reference's
+    own pushdown_test (columnar_chunk_filtering.out:1085-1121): 8 rows in
+    4 two-row chunk groups, `country IN ('USA','BR','ZW')` -> rows {3,7,8}.
+
+    Row results are EXACTLY the reference's. The chunk-group REMOVAL
+    count differs by design: the reference compares text min/max in
+    collation order and removes 2 groups; this writer records NO min/max
+    for TEXT chunks (conservative: never prunes on text, so results are
+    always exact) — collation-ordered text pruning is a documented next
+    step."""
+    ids = np.arange(1, 9, dtype=np.int64)
+    countries = ["AL", "AU", "BR", "BT", "PK", "PA", "USA", "ZW"]
+    slots = ca.text_slots(countries)
+    path = str(tmp_path / "push.cs")
+    ca.write_table(path, [("id", ca.I64, 0), ("country", ca.TEXT, 0)],
+                   [ids, slots.view(np.int32)], compression=ca.COMP_LZ4,
+                   stripe_row_limit=2, chunk_group_row_limit=2)
+    g = 42                                   # one OR group: the IN list
+    preds = [(1, ca.PRED_EQ, int(ca.text_slot("USA")), g),
+             (1, ca.PRED_EQ, int(ca.text_slot("BR")), g),
+             (1, ca.PRED_EQ, int(ca.text_slot("ZW")), g)]
+    with oracle.OracleTable(path) as t:
+        parts, filtered = t.scan_agg(preds, [(ca.AGG_COUNT_STAR, -1),
+                                             (ca.AGG_SUM_I64, 0),
+                                             (ca.AGG_MIN_I64, 0),
+                                             (ca.AGG_MAX_I64, 0)])
+    assert parts[0].count == 3               # rows 3, 7, 8 — reference answer
+    assert parts[1].i128 == 3 + 7 + 8
+    assert parts[2].i128 == 3 and parts[3].i128 == 8
+    assert filtered == 0                     # no TEXT min/max -> no pruning
