@@ -4501,7 +4501,10 @@ int csgpu_compress_chunk(const void *const *dev_vals, const uint8_t *types,
         WTRY(hipStreamSynchronize(st));
         oc.min_i = hs.mn;
         oc.max_i = hs.mx;
-        /* TEXT columns carry no skip-node min/max (host path does the same) */
+        /* TEXT: the host writer records C-collation lex-key min/max
+         * (format.h csf_text_lex_key); this stats kernel reduces raw
+         * integer order, so device-written TEXT chunks stay conservatively
+         * unprunable rather than carry a wrong-order range */
         if (types[c] == CSTRIPE_TEXT) oc.has_min_max = false;
 
         const uint64_t raw_bytes = (uint64_t)rows * width;

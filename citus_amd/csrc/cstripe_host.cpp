@@ -206,6 +206,20 @@ static void update_minmax(ColCur &c, uint8_t type, const uint8_t *vals, size_t n
             }
             break;
         }
+        case CSTRIPE_TEXT: {
+            /* min/max in the C-collation lex-key order (format.h
+             * csf_text_lex_key) — used for EQ/NE refutation only */
+            const uint32_t *v = (const uint32_t *)vals;
+            int64_t mn = csf_text_lex_key(v[0]), mx = mn;
+            for (size_t i = 1; i < n; i++) {
+                const int64_t k = csf_text_lex_key(v[i]);
+                if (k < mn) mn = k;
+                if (k > mx) mx = k;
+            }
+            if (!c.has_min_max) { c.min_i = mn; c.max_i = mx; c.has_min_max = true; }
+            else { if (mn < c.min_i) c.min_i = mn; if (mx > c.max_i) c.max_i = mx; }
+            break;
+        }
     }
 }
 
@@ -1100,6 +1114,18 @@ static bool pred_refutes(const cstripe_pred &p, uint8_t type, int64_t min_i, int
         }
         return false;
     }
+    if (type == CSTRIPE_TEXT) {
+        /* TEXT min/max hold the C-collation lex key (format.h); EQ/NE are
+         * order-independent, so refutation through ANY consistent order is
+         * sound — range operators are never refuted (the ABI's row-level
+         * text ordering is whole-slot, not collation) */
+        const int64_t c = csf_text_lex_key((uint32_t)p.ival);
+        switch (p.op) {
+            case CSTRIPE_PRED_EQ: return c < min_i || c > max_i;
+            case CSTRIPE_PRED_NE: return min_i == c && max_i == c;
+            default:              return false;
+        }
+    }
     int64_t c = p.ival;
     switch (p.op) {
         case CSTRIPE_PRED_LT: return min_i >= c;
@@ -1145,8 +1171,13 @@ extern "C" cstripe_scan *cstripe_scan_begin(cstripe_reader *r, uint64_t cols_mas
         for (const auto &st : r->stripes) total_chunks += st.meta.chunk_count;
         const char *dp = getenv("CSTRIPE_DEVICE_PRUNE");
         const int dpv = dp ? atoi(dp) : -1;
+        bool any_text = false;   /* TEXT refutation runs host-side (lex-key
+                                  * EQ/NE only; kernel not taught the
+                                  * transform) */
+        for (const auto &p : s->preds)
+            any_text |= r->cols[p.column].type == CSTRIPE_TEXT;
         std::vector<uint8_t> selmask;
-        if (!s->preds.empty() && dpv != 0 &&
+        if (!s->preds.empty() && !any_text && dpv != 0 &&
             (dpv == 1 || total_chunks >= 8192) &&
             csgpu_prune(r, s->preds, selmask) == CSTRIPE_OK &&
             selmask.size() == total_chunks) {

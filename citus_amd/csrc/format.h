@@ -172,6 +172,20 @@ static inline uint32_t csf_canon_zr4b_pos(uint32_t j, uint32_t k)
     return 15u + (j == 0 ? k : j + 3u + k);
 }
 
+/* C-collation (memcmp) order key of a short-varlena TEXT slot: the payload
+ * chars big-endian, so integer order == byte-lexicographic order of the
+ * text (zero padding ranks shorter strings first; PostgreSQL text cannot
+ * contain NUL, so the key is injective). TEXT skip-node min/max store THIS
+ * key, and refutation uses it only for the order-independent EQ/NE ops —
+ * matching the reference's predicate_refuted_by pruning under "C"
+ * collation for those operators (columnar_reader.c:1132-1187). */
+static inline int64_t csf_text_lex_key(uint32_t slot)
+{
+    return (int64_t)((((slot >> 8) & 0xFFu) << 16) |
+                     (((slot >> 16) & 0xFFu) << 8) |
+                     ((slot >> 24) & 0xFFu));
+}
+
 /* literal-run header size for a mode-LIT segment of len decompressed bytes */
 static inline uint32_t csf_canon_lit_hdr(uint32_t len)
 {

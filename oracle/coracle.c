@@ -296,6 +296,16 @@ static int o_pred_refutes(const cstripe_pred *p, uint8_t type, int64_t min_i, in
         }
         return 0;
     }
+    if (type == CSTRIPE_TEXT) {
+        /* TEXT min/max hold the C-collation lex key (format.h
+         * csf_text_lex_key); only order-independent EQ/NE refute */
+        const int64_t c = csf_text_lex_key((uint32_t)p->ival);
+        switch (p->op) {
+            case CSTRIPE_PRED_EQ: return c < min_i || c > max_i;
+            case CSTRIPE_PRED_NE: return min_i == c && max_i == c;
+            default:              return 0;
+        }
+    }
     int64_t c = p->ival;
     switch (p->op) {
         case CSTRIPE_PRED_LT: return min_i >= c;

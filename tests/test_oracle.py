@@ -219,12 +219,12 @@ def test_in_list_pushdown_reference_fixture(tmp_path):
     own pushdown_test (columnar_chunk_filtering.out:1085-1121): 8 rows in
     4 two-row chunk groups, `country IN ('USA','BR','ZW')` -> rows {3,7,8}.
 
-    Row results are EXACTLY the reference's. The chunk-group REMOVAL
-    count differs by design: the reference compares text min/max in
-    collation order and removes 2 groups; this writer records NO min/max
-    for TEXT chunks (conservative: never prunes on text, so results are
-    always exact) — collation-ordered text pruning is a documented next
-    step."""
+    Row results AND the chunk-group removal count are EXACTLY the
+    reference's: TEXT skip nodes store min/max as the C-collation lex key
+    (csf_text_lex_key) and EQ/NE predicates refute through it, so the
+    same two groups ({AL,AU} and {PK,PA}) are removed. Range operators on
+    TEXT are never refuted (the ABI's row-level text order is whole-slot,
+    not collation) — conservative, results always exact."""
     ids = np.arange(1, 9, dtype=np.int64)
     countries = ["AL", "AU", "BR", "BT", "PK", "PA", "USA", "ZW"]
     slots = ca.text_slots(countries)
@@ -244,4 +244,4 @@ def test_in_list_pushdown_reference_fixture(tmp_path):
     assert parts[0].count == 3               # rows 3, 7, 8 — reference answer
     assert parts[1].i128 == 3 + 7 + 8
     assert parts[2].i128 == 3 and parts[3].i128 == 8
-    assert filtered == 0                     # no TEXT min/max -> no pruning
+    assert filtered == 2                     # {AL,AU}, {PK,PA} — as the ref
