@@ -245,3 +245,6 @@ def test_in_list_pushdown_reference_fixture(tmp_path):
     assert parts[1].i128 == 3 + 7 + 8
     assert parts[2].i128 == 3 and parts[3].i128 == 8
     assert filtered == 2                     # {AL,AU}, {PK,PA} — as the ref
+    # the PRODUCT's host refutation (scan_begin, no GPU needed) agrees
+    with ca.Reader(path) as r, r.scan(cols_mask=0b11, preds=preds) as s:
+        assert s.chunk_groups_filtered == 2
